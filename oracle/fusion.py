@@ -1,0 +1,181 @@
+"""CPU restatement of mvrecon's block-wise affine fusion.
+
+Restates net.preibisch.mvrecon.process.fusion.blk.BlkAffineFusion
+.initWithIntensityCoefficients + the materializing copy, as invoked at
+reference SparkAffineFusion.java:602-615, :627, with interpolation order 1
+(:611), FusionType in {AVG, AVG_BLEND, MAX_INTENSITY} (:124-125) and
+RealUnsignedByte/ShortConverter min/max intensity scaling (:497-517).
+
+Per output voxel at world coordinate w = block_min + index:
+  for each view v whose transformed bbox overlaps the block:
+    p = A_v^{-1} . w            (A_v: view-local -> world, 3x4 double)
+    if p inside [0, dim_d - 1] for all d:  (see [PIN-BOUNDS])
+      value  = trilinear(view_v, p)
+      weight = blend_v(p)       (AVG: 1; AVG_BLEND: cosine border ramp)
+  AVG/AVG_BLEND: out = sum(w_v * value_v) / sum(w_v)   (0 where sum w = 0)
+  MAX_INTENSITY: out = max(value_v)                    (0 where no view)
+
+Restatement choices (artifact source unavailable — oracle/__init__ note):
+  [PIN-BOUNDS] a view contributes iff its inverse-mapped point lies in
+               [0, dim_d - 1] per axis (trilinear interior; the upper
+               neighbor index is clamped so p == dim-1 is exact).
+  [PIN-BLEND]  AVG_BLEND weight = prod_d f((dist_d - border_d)/range_d),
+               dist_d = min(p_d, dim_d - 1 - p_d) + 1 (px from outside),
+               f(t) = 0 for t<=0, 1 for t>=1, else 0.5 - 0.5 cos(pi t).
+               Defaults border=0, range=40 (mvrecon Blending defaults).
+  [PIN-CONV]   uint8/uint16 conversion: round((v - minI)/(maxI - minI)
+               * type_max), clamped to the type range (round half away
+               from zero, matching C lround / CUDA lrintf-free path).
+  [PIN-ORDER]  accumulation in the given view order (determinism).
+
+Oracle computes in float64; the HIP path accumulates float32 — the parity
+bar is 1e-4 relative on float32 output (north_star).
+"""
+
+from __future__ import annotations
+
+import numpy as np
+
+FUSION_AVG = 0
+FUSION_AVG_BLEND = 1
+FUSION_MAX = 2
+
+__all__ = [
+    "fuse_block",
+    "blend_weight",
+    "FUSION_AVG",
+    "FUSION_AVG_BLEND",
+    "FUSION_MAX",
+]
+
+
+def blend_weight(p, dims, border, rng):
+    """[PIN-BLEND] per-point blend weight. p: (..., 3) float (x,y,z order),
+    dims/border/rng: (3,) in x,y,z order."""
+    w = np.ones(p.shape[:-1], dtype=np.float64)
+    for d in range(3):
+        dist = np.minimum(p[..., d], dims[d] - 1 - p[..., d]) + 1.0
+        t = (dist - border[d]) / rng[d] if rng[d] > 0 else np.where(
+            dist > border[d], np.inf, -np.inf
+        )
+        fd = np.clip(t, 0.0, 1.0)
+        fd = np.where(
+            (t > 0) & (t < 1), 0.5 - 0.5 * np.cos(np.pi * t), fd
+        )
+        w = w * fd
+    return w
+
+
+def _trilinear(vol: np.ndarray, p: np.ndarray):
+    """Trilinear sample of (nz,ny,nx) vol at p (..., 3) x,y,z order.
+    Caller guarantees p in-bounds per [PIN-BOUNDS]."""
+    nz, ny, nx = vol.shape
+    x, y, z = p[..., 0], p[..., 1], p[..., 2]
+    x0 = np.floor(x).astype(np.int64)
+    y0 = np.floor(y).astype(np.int64)
+    z0 = np.floor(z).astype(np.int64)
+    x1 = np.minimum(x0 + 1, nx - 1)
+    y1 = np.minimum(y0 + 1, ny - 1)
+    z1 = np.minimum(z0 + 1, nz - 1)
+    fx, fy, fz = x - x0, y - y0, z - z0
+    v = vol.astype(np.float64)
+    c000 = v[z0, y0, x0]
+    c100 = v[z0, y0, x1]
+    c010 = v[z0, y1, x0]
+    c110 = v[z0, y1, x1]
+    c001 = v[z1, y0, x0]
+    c101 = v[z1, y0, x1]
+    c011 = v[z1, y1, x0]
+    c111 = v[z1, y1, x1]
+    c00 = c000 * (1 - fx) + c100 * fx
+    c10 = c010 * (1 - fx) + c110 * fx
+    c01 = c001 * (1 - fx) + c101 * fx
+    c11 = c011 * (1 - fx) + c111 * fx
+    c0 = c00 * (1 - fy) + c10 * fy
+    c1 = c01 * (1 - fy) + c11 * fy
+    return c0 * (1 - fz) + c1 * fz
+
+
+def _invert_affine(m: np.ndarray) -> np.ndarray:
+    """Invert a row-major 3x4 affine (returns 3x4)."""
+    a = m[:, :3]
+    t = m[:, 3]
+    ai = np.linalg.inv(a)
+    return np.hstack([ai, (-ai @ t)[:, None]])
+
+
+def fuse_block(
+    views,
+    block_min,
+    block_size,
+    fusion_type: int = FUSION_AVG_BLEND,
+    out_dtype=np.float32,
+    min_intensity: float = 0.0,
+    max_intensity: float = 65535.0,
+):
+    """Fuse one output block. Mirrors bs_fuse_blocks for a single block.
+
+    views: list of dicts with keys
+        data   : (nz, ny, nx) uint16
+        affine : (3, 4) float64 row-major, view-local -> world (x,y,z)
+        border : (3,) float, blend border px (default 0)
+        range  : (3,) float, blend range px (default 40)
+    block_min/block_size: (3,) int, x,y,z order, world coords.
+    Returns (nz, ny, nx)-shaped array of out_dtype (block_size reversed)."""
+    bs = np.asarray(block_size, dtype=np.int64)
+    bm = np.asarray(block_min, dtype=np.float64)
+    zz, yy, xx = np.meshgrid(
+        np.arange(bs[2]), np.arange(bs[1]), np.arange(bs[0]), indexing="ij"
+    )
+    w = np.stack(
+        [xx + bm[0], yy + bm[1], zz + bm[2]], axis=-1
+    ).astype(np.float64)
+
+    sum_wv = np.zeros(w.shape[:-1], dtype=np.float64)
+    sum_w = np.zeros(w.shape[:-1], dtype=np.float64)
+    vmax = np.zeros(w.shape[:-1], dtype=np.float64)
+    any_view = np.zeros(w.shape[:-1], dtype=bool)
+
+    for v in views:
+        vol = v["data"]
+        nz, ny, nx = vol.shape
+        dims = np.array([nx, ny, nz], dtype=np.float64)
+        inv = _invert_affine(np.asarray(v["affine"], dtype=np.float64))
+        p = w @ inv[:, :3].T + inv[:, 3]
+        inside = np.ones(p.shape[:-1], dtype=bool)
+        for d in range(3):
+            inside &= (p[..., d] >= 0) & (p[..., d] <= dims[d] - 1)
+        if not inside.any():
+            continue
+        pc = np.where(inside[..., None], p, 0.0)
+        val = _trilinear(vol, pc)
+        if fusion_type == FUSION_AVG_BLEND:
+            border = np.asarray(v.get("border", (0.0, 0.0, 0.0)))
+            rng = np.asarray(v.get("range", (40.0, 40.0, 40.0)))
+            wt = blend_weight(pc, dims, border, rng)
+        else:
+            wt = np.ones(p.shape[:-1], dtype=np.float64)
+        wt = np.where(inside, wt, 0.0)
+        if fusion_type == FUSION_MAX:
+            vmax = np.where(inside & (val > vmax), val, vmax)
+            any_view |= inside
+        else:
+            sum_wv += wt * val
+            sum_w += wt
+
+    if fusion_type == FUSION_MAX:
+        out = np.where(any_view, vmax, 0.0)
+    else:
+        with np.errstate(invalid="ignore", divide="ignore"):
+            out = np.where(sum_w > 0, sum_wv / np.maximum(sum_w, 1e-300), 0.0)
+
+    if out_dtype == np.float32:
+        return out.astype(np.float32)
+    tmax = 255.0 if out_dtype == np.uint8 else 65535.0
+    scaled = (out - min_intensity) / (max_intensity - min_intensity) * tmax
+    # [PIN-CONV] round half away from zero, clamp
+    scaled = np.where(
+        sum_w > 0 if fusion_type != FUSION_MAX else any_view, scaled, 0.0
+    )
+    q = np.floor(np.abs(scaled) + 0.5) * np.sign(scaled)
+    return np.clip(q, 0, tmax).astype(out_dtype)

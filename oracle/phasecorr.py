@@ -1,0 +1,218 @@
+"""CPU restatement of BigStitcher's pairwise phase-correlation stitching.
+
+Restates net.preibisch.stitcher.algorithm.PairwiseStitching.getShift as
+invoked through TransformationTools.computeStitching at reference
+SparkPairwiseStitching.java:247-255 with
+PairwiseStitchingParameters{doSubpixel, peaksToCheck} (:200-202) and
+downsampling ds (:77, default {2,2,1}).
+
+Algorithm (Preibisch et al. 2009; Hörl et al. 2019): pad the two
+(downsampled) overlap regions to a common FFT size; Q = conj(F1)*F2 /
+|conj(F1)*F2|; PCM = real(IFFT(Q)); take the `peaks_to_check` highest
+local maxima of the PCM; expand each to its 2^3 periodic-shift candidates;
+for each candidate with sufficient overlap compute the real-space Pearson
+cross-correlation r over the implied overlap; winner = max r; optional
+per-axis quadratic sub-pixel fit on the PCM around the winning peak.
+
+Restatement choices, each pinned by tests (the artifact source is not
+available — see oracle/__init__ parity note):
+  [PIN-DS]   downsampling = per-axis box mean over ds-blocks (remainder
+             voxels truncated), np.rint back to uint16.
+  [PIN-PAD]  common FFT size per axis = next power of two >= max of the
+             two (downsampled) sizes; images zero-padded at the high end.
+  [PIN-EPS]  cross-power magnitude < 1e-20 -> Q component set to 0.
+  [PIN-MAX]  PCM local maxima use the full 26-neighborhood with periodic
+             wrap, strict '>' comparison.
+  [PIN-CAND] all 2^3 periodic candidates {p_d, p_d - n_d} are r-tested,
+             subject to min_overlap_ratio vs the smaller region's volume.
+  [PIN-R]    r computed from exact int64 sums of the uint16 overlap voxels
+             (bit-identical between oracle and HIP path by construction).
+  [PIN-SUB]  sub-pixel = independent per-axis 3-point quadratic fit at the
+             winning PCM peak (wrapped neighbors), offset clamped to
+             [-0.5, 0.5]; applied after candidate selection; degenerate
+             curvature (|fm - 2 f0 + fp| < 1e-12) -> offset 0.
+  [PIN-SIGN] the returned shift s satisfies  B(x) ~= A(x - s)  over the
+             overlap-interval coordinates: a feature at A-coordinate u
+             appears at B-coordinate u + s.  s is returned in
+             FULL-RESOLUTION pixels (candidate+subpixel scaled by ds).
+"""
+
+from __future__ import annotations
+
+import numpy as np
+
+__all__ = ["downsample", "pcm", "phase_correlation_shift", "cross_corr_sums"]
+
+
+def downsample(vol: np.ndarray, ds) -> np.ndarray:
+    """[PIN-DS] box-mean downsample. vol is (nz, ny, nx) uint16; ds is
+    (dsx, dsy, dsz) (x,y,z order, matching the CLI flag order)."""
+    dsx, dsy, dsz = int(ds[0]), int(ds[1]), int(ds[2])
+    if dsx == 1 and dsy == 1 and dsz == 1:
+        return vol
+    nz, ny, nx = vol.shape
+    mz, my, mx = nz // dsz, ny // dsy, nx // dsx
+    v = vol[: mz * dsz, : my * dsy, : mx * dsx].astype(np.float64)
+    v = v.reshape(mz, dsz, my, dsy, mx, dsx).mean(axis=(1, 3, 5))
+    return np.rint(v).astype(np.uint16)
+
+
+def _next_pow2(n: int) -> int:
+    p = 1
+    while p < n:
+        p *= 2
+    return p
+
+
+def pcm(a: np.ndarray, b: np.ndarray):
+    """Phase-correlation matrix of two (nz,ny,nx) uint16 arrays.
+
+    Returns (pcm float64 (pz,py,px), padded shape). [PIN-PAD] [PIN-EPS]"""
+    shape = tuple(
+        _next_pow2(max(sa, sb)) for sa, sb in zip(a.shape, b.shape)
+    )
+    fa = np.fft.rfftn(a, s=shape)
+    fb = np.fft.rfftn(b, s=shape)
+    q = np.conj(fa) * fb
+    mag = np.abs(q)
+    with np.errstate(invalid="ignore", divide="ignore"):
+        q = np.where(mag < 1e-20, 0.0, q / mag)
+    return np.fft.irfftn(q, s=shape), shape
+
+
+def _local_maxima_topk(p: np.ndarray, k: int):
+    """[PIN-MAX] top-k strict local maxima (26-neighborhood, periodic).
+
+    Returns list of (value, (pz, py, px)) sorted by value descending,
+    ties broken by ascending linear index."""
+    is_max = np.ones(p.shape, dtype=bool)
+    for dz in (-1, 0, 1):
+        for dy in (-1, 0, 1):
+            for dx in (-1, 0, 1):
+                if dz == 0 and dy == 0 and dx == 0:
+                    continue
+                is_max &= p > np.roll(p, (dz, dy, dx), axis=(0, 1, 2))
+    idx = np.flatnonzero(is_max)
+    if idx.size == 0:
+        return []
+    vals = p.ravel()[idx]
+    # sort by (-value, index): stable sort on index then stable sort on -value
+    order = np.argsort(-vals, kind="stable")
+    order = order[: int(k)]
+    out = []
+    for o in order:
+        out.append((float(vals[o]), np.unravel_index(idx[o], p.shape)))
+    return out
+
+
+def cross_corr_sums(a: np.ndarray, b: np.ndarray, shift):
+    """[PIN-R] exact int64 sums over the overlap implied by integer shift.
+
+    shift is (sz, sy, sx) in downsampled px with the [PIN-SIGN] meaning
+    B(x) ~= A(x - s). Returns (n, sa, sb, saa, sbb, sab) as Python ints,
+    n == 0 when the overlap is empty."""
+    s = [int(v) for v in shift]
+    lo = [max(0, -sv) for sv in s]
+    hi = [
+        min(na, nb - sv) for na, nb, sv in zip(a.shape, b.shape, s)
+    ]
+    if any(h <= l for l, h in zip(lo, hi)):
+        return 0, 0, 0, 0, 0, 0
+    asub = a[lo[0]:hi[0], lo[1]:hi[1], lo[2]:hi[2]].astype(np.int64)
+    bsub = b[
+        lo[0] + s[0]:hi[0] + s[0],
+        lo[1] + s[1]:hi[1] + s[1],
+        lo[2] + s[2]:hi[2] + s[2],
+    ].astype(np.int64)
+    n = asub.size
+    return (
+        n,
+        int(asub.sum()),
+        int(bsub.sum()),
+        int((asub * asub).sum()),
+        int((bsub * bsub).sum()),
+        int((asub * bsub).sum()),
+    )
+
+
+def _r_from_sums(n, sa, sb, saa, sbb, sab) -> float:
+    if n == 0:
+        return -2.0
+    num = sab - sa * sb / n
+    da = saa - sa * sa / n
+    db = sbb - sb * sb / n
+    if da <= 0 or db <= 0:
+        return -2.0
+    return float(num / np.sqrt(da * db))
+
+
+def _subpixel_offset(p: np.ndarray, peak) -> np.ndarray:
+    """[PIN-SUB] per-axis quadratic fit with periodic wrap."""
+    off = np.zeros(3)
+    for d in range(3):
+        im = list(peak)
+        ip = list(peak)
+        im[d] = (peak[d] - 1) % p.shape[d]
+        ip[d] = (peak[d] + 1) % p.shape[d]
+        fm = p[tuple(im)]
+        f0 = p[tuple(peak)]
+        fp = p[tuple(ip)]
+        denom = fm - 2.0 * f0 + fp
+        if abs(denom) < 1e-12:
+            continue
+        o = 0.5 * (fm - fp) / denom
+        off[d] = min(0.5, max(-0.5, o))
+    return off
+
+
+def phase_correlation_shift(
+    a: np.ndarray,
+    b: np.ndarray,
+    ds=(2, 2, 1),
+    peaks_to_check: int = 5,
+    do_subpixel: bool = True,
+    min_overlap_ratio: float = 0.25,
+):
+    """Full restatement of PairwiseStitching.getShift for one tile pair.
+
+    a, b: (nz, ny, nx) uint16 overlap regions (full resolution).
+    Returns dict(shift=(sx, sy, sz) full-res px per [PIN-SIGN], r=float,
+    valid=bool). Matches bs_stitch_batch's bs_shift_result contract
+    (include/bigstitch.h)."""
+    ad = downsample(a, ds)
+    bd = downsample(b, ds)
+    p, _shape = pcm(ad, bd)
+    peaks = _local_maxima_topk(p, peaks_to_check)
+    min_n = min_overlap_ratio * min(ad.size, bd.size)
+    best = None  # (r, peak_rank, cand_idx, cand_shift, peak)
+    for rank, (_val, peak) in enumerate(peaks):
+        for ci in range(8):
+            cand = []
+            for d in range(3):
+                c = int(peak[d])
+                if (ci >> d) & 1:
+                    c -= p.shape[d]
+                cand.append(c)
+            sums = cross_corr_sums(ad, bd, cand)
+            if sums[0] < max(min_n, 1):
+                continue
+            r = _r_from_sums(*sums)
+            if r <= -2.0:
+                continue
+            key = (-r, rank, ci)
+            if best is None or key < best[0]:
+                best = (key, cand, peak, r)
+    if best is None:
+        return {"shift": np.zeros(3), "r": 0.0, "valid": False}
+    _key, cand, peak, r = best
+    shift_zyx = np.array(cand, dtype=np.float64)
+    if do_subpixel:
+        shift_zyx = shift_zyx + _subpixel_offset(p, peak)
+    # back to full resolution, and to (x, y, z) order for the ABI contract
+    dszyx = np.array([ds[2], ds[1], ds[0]], dtype=np.float64)
+    full = shift_zyx * dszyx
+    return {
+        "shift": full[::-1].copy(),  # (sx, sy, sz)
+        "r": r,
+        "valid": True,
+    }
