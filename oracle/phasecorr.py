@@ -71,13 +71,13 @@ def pcm(a: np.ndarray, b: np.ndarray):
     shape = tuple(
         _next_pow2(max(sa, sb)) for sa, sb in zip(a.shape, b.shape)
     )
-    fa = np.fft.rfftn(a, s=shape)
-    fb = np.fft.rfftn(b, s=shape)
+    fa = np.fft.rfftn(a, s=shape, axes=(0, 1, 2))
+    fb = np.fft.rfftn(b, s=shape, axes=(0, 1, 2))
     q = np.conj(fa) * fb
     mag = np.abs(q)
     with np.errstate(invalid="ignore", divide="ignore"):
         q = np.where(mag < 1e-20, 0.0, q / mag)
-    return np.fft.irfftn(q, s=shape), shape
+    return np.fft.irfftn(q, s=shape, axes=(0, 1, 2)), shape
 
 
 def _local_maxima_topk(p: np.ndarray, k: int):
