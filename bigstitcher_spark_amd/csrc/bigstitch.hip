@@ -1,0 +1,1304 @@
+/* libbigstitch — MI355X-native (gfx950/CDNA4) implementation of the
+ * BigStitcher-Spark hot-path math layer behind the C ABI declared in
+ * include/bigstitch.h.
+ *
+ * Replaces (from scratch, not a port):
+ *   - PairwiseStitching.getShift / TransformationTools.computeStitching
+ *     (reference SparkPairwiseStitching.java:247-255) -> bs_stitch_batch
+ *   - BlkAffineFusion.initWithIntensityCoefficients + materializing copy
+ *     (reference SparkAffineFusion.java:602-615,627)  -> bs_fuse_blocks
+ *
+ * Arithmetic contract = oracle/phasecorr.py and oracle/fusion.py (the CPU
+ * restatement; see its [PIN-*] notes). Shifts must match the oracle within
+ * 1e-3 px; fused float32 within 1e-4 relative; the candidate r-test uses
+ * exact int64 sums and matches the oracle bit-for-bit.
+ *
+ * Design notes (MI355X):
+ *   - Both paths are HBM-bound (no dense contraction; MFMA idle by design,
+ *     see BASELINE.json north_star). Kernels are organised as full-line LDS
+ *     FFT passes with coalesced global access on the x-fastest layout:
+ *       x-pass: 4 lines/WG x 64 threads-per-line (contiguous lines),
+ *       y/z-pass: 16 adjacent x-columns/WG x 16 threads-per-line so every
+ *       global transaction is 16 consecutive complex values (128 B).
+ *   - Cross-power normalisation is fused into the load of the inverse
+ *     z-pass (saves one full volume read+write).
+ *   - Half-spectrum (Hermitian) storage: after the inverse y/z passes each
+ *     x half-line is per-line Hermitian, so the final pass is a per-line
+ *     C2R; rows padded to 16-complex multiples for 128-B row alignment.
+ *   - Grids are >> 256 workgroups for every volume kernel (fills 8 XCDs).
+ */
+
+#include "../../include/bigstitch.h"
+
+#include <hip/hip_runtime.h>
+
+#include <cmath>
+#include <cstdio>
+#include <cstdlib>
+#include <cstring>
+#include <map>
+#include <mutex>
+#include <string>
+#include <vector>
+
+#define LPB_X 4    /* lines per block, contiguous-line passes */
+#define TPL_X 64
+#define LPB_S 16   /* lines per block, strided passes */
+#define TPL_S 16
+#define PK_TX 64   /* peak-scan tile */
+#define PK_TY 4
+#define PK_TZ 4
+
+typedef unsigned long long u64;
+
+struct f2 { float x, y; };
+
+__device__ __forceinline__ f2 cmul(f2 a, f2 b) {
+  return {a.x * b.x - a.y * b.y, a.x * b.y + a.y * b.x};
+}
+__device__ __forceinline__ f2 conjmul(f2 a, f2 b) { /* conj(a)*b */
+  return {a.x * b.x + a.y * b.y, a.x * b.y - a.y * b.x};
+}
+__device__ __forceinline__ unsigned brev_n(unsigned j, int log2n) {
+  return __brev(j) >> (32 - log2n);
+}
+
+/* -------------------------------------------------------- region descs */
+
+struct bs_region { /* a (possibly strided) uint16 sub-volume */
+  const unsigned short *ptr;
+  long sx, sxy;      /* row and slice strides, in elements */
+  long ox, oy, oz;   /* offset of the region inside ptr's volume */
+  int mx, my, mz;    /* region dims */
+};
+
+/* ------------------------------------------------------------ downsample */
+
+__global__ __launch_bounds__(256) void k_downsample(
+    bs_region in, unsigned short *out, int mx, int my, int mz,
+    int dsx, int dsy, int dsz) {
+  long n = (long)mx * my * mz;
+  float inv = 1.0f / (dsx * dsy * dsz);
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x) {
+    int x = (int)(i % mx);
+    long t = i / mx;
+    int y = (int)(t % my);
+    int z = (int)(t / my);
+    float s = 0.0f;
+    for (int kz = 0; kz < dsz; ++kz)
+      for (int ky = 0; ky < dsy; ++ky)
+        for (int kx = 0; kx < dsx; ++kx) {
+          long a = (in.oz + (long)z * dsz + kz) * in.sxy +
+                   (in.oy + (long)y * dsy + ky) * in.sx +
+                   (in.ox + (long)x * dsx + kx);
+          s += (float)in.ptr[a];
+        }
+    out[i] = (unsigned short)__float2int_rn(s * inv); /* [PIN-DS] rint */
+  }
+}
+
+/* ---------------------------------------------------------- FFT passes */
+
+/* Contiguous-line forward pass: u16 region lines (x) -> half spectrum.
+ * One workgroup = LPB_X lines x TPL_X threads. Dynamic LDS:
+ * [ tw: n/2 f2 | data: LPB_X * n f2 ]. */
+__global__ __launch_bounds__(LPB_X *TPL_X) void k_fft_x_fwd(
+    bs_region in, f2 *out, int n, int log2n, int cx, long cxp,
+    const f2 *twg) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  f2 *tw = (f2 *)smem;
+  f2 *data = tw + (n >> 1);
+  const int tid = threadIdx.x;
+  const int tl = tid % TPL_X, line = tid / TPL_X;
+  for (int i = tid; i < (n >> 1); i += LPB_X * TPL_X) tw[i] = twg[i];
+
+  long lid = (long)blockIdx.x * LPB_X + line;
+  long nlines = (long)in.my * in.mz;
+  bool active = lid < nlines;
+  int y = active ? (int)(lid % in.my) : 0;
+  int z = active ? (int)(lid / in.my) : 0;
+  const unsigned short *src =
+      in.ptr + (in.oz + z) * in.sxy + (in.oy + y) * in.sx + in.ox;
+  f2 *ld = data + (long)line * n;
+  for (int j = tl; j < n; j += TPL_X) {
+    float v = (active && j < in.mx) ? (float)src[j] : 0.0f;
+    ld[brev_n(j, log2n)] = {v, 0.0f};
+  }
+  __syncthreads();
+  const int nbf = n >> 1;
+  for (int len = 2; len <= n; len <<= 1) {
+    int half = len >> 1, shift = n / len;
+    for (int bf = tl; bf < nbf; bf += TPL_X) {
+      int blk = bf / half, off = bf % half;
+      int i0 = blk * len + off, i1 = i0 + half;
+      f2 w = tw[off * shift]; /* e^{-2pi i off/len} */
+      f2 u = ld[i0], v = cmul(ld[i1], w);
+      ld[i0] = {u.x + v.x, u.y + v.y};
+      ld[i1] = {u.x - v.x, u.y - v.y};
+    }
+    __syncthreads();
+  }
+  if (active) {
+    f2 *o = out + lid * cxp;
+    for (int k = tl; k < cx; k += TPL_X) o[k] = ld[k];
+  }
+}
+
+/* Strided C2C pass along y or z (and the fused cross-power inverse-z).
+ * One workgroup = LPB_S adjacent x-columns x TPL_S threads/line.
+ * blockIdx.x = group * nchunks + chunk. Line base = group*gstride + x.
+ * Elements e >= valid load as zero. dir=+1 forward, -1 inverse.
+ * in2 != nullptr: load q = conj(in)*in2 normalised * scale (cross-power,
+ * [PIN-EPS]) instead of in. In-place safe (each WG owns its lines). */
+__global__ __launch_bounds__(LPB_S *TPL_S) void k_fft_pass(
+    const f2 *in, const f2 *in2, f2 *out, int n, int log2n, long estride,
+    long gstride, int nlines, int nchunks, int valid, int dir, float scale,
+    const f2 *twg) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  f2 *tw = (f2 *)smem;
+  f2 *data = tw + (n >> 1);
+  const int tid = threadIdx.x;
+  const int line = tid % LPB_S, tl = tid / LPB_S;
+  for (int i = tid; i < (n >> 1); i += LPB_S * TPL_S) tw[i] = twg[i];
+
+  const int group = blockIdx.x / nchunks;
+  const int x = (blockIdx.x % nchunks) * LPB_S + line;
+  const bool active = x < nlines;
+  const long base = (long)group * gstride + x;
+  for (int e = tl; e < n; e += TPL_S) {
+    f2 v = {0.0f, 0.0f};
+    if (active && e < valid) {
+      if (in2) {
+        f2 q = conjmul(in[base + e * estride], in2[base + e * estride]);
+        float m2 = q.x * q.x + q.y * q.y;
+        if (m2 > 1e-40f) {
+          float s = scale / sqrtf(m2);
+          v = {q.x * s, q.y * s};
+        }
+      } else {
+        v = in[base + e * estride];
+      }
+    }
+    data[(long)brev_n(e, log2n) * LPB_S + line] = v;
+  }
+  __syncthreads();
+  const int nbf = n >> 1;
+  for (int len = 2; len <= n; len <<= 1) {
+    int half = len >> 1, shift = n / len;
+    for (int bf = tl; bf < nbf; bf += TPL_S) {
+      int blk = bf / half, off = bf % half;
+      long i0 = (long)(blk * len + off) * LPB_S + line;
+      long i1 = i0 + (long)half * LPB_S;
+      f2 w = tw[off * shift];
+      if (dir < 0) w.y = -w.y;
+      f2 u = data[i0], v = cmul(data[i1], w);
+      data[i0] = {u.x + v.x, u.y + v.y};
+      data[i1] = {u.x - v.x, u.y - v.y};
+    }
+    __syncthreads();
+  }
+  if (active)
+    for (int e = tl; e < n; e += TPL_S)
+      out[base + e * estride] = data[(long)e * LPB_S + line];
+}
+
+/* Inverse x pass: Hermitian half-line -> full line (per-line mirror),
+ * inverse FFT, write real part (the PCM). */
+__global__ __launch_bounds__(LPB_X *TPL_X) void k_fft_x_inv(
+    const f2 *in, float *out, int n, int log2n, int cx, long cxp,
+    long nlines, const f2 *twg) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  f2 *tw = (f2 *)smem;
+  f2 *data = tw + (n >> 1);
+  const int tid = threadIdx.x;
+  const int tl = tid % TPL_X, line = tid / TPL_X;
+  for (int i = tid; i < (n >> 1); i += LPB_X * TPL_X) tw[i] = twg[i];
+
+  long lid = (long)blockIdx.x * LPB_X + line;
+  bool active = lid < nlines;
+  const f2 *src = in + lid * cxp;
+  f2 *ld = data + (long)line * n;
+  for (int e = tl; e < n; e += TPL_X) {
+    f2 v = {0.0f, 0.0f};
+    if (active) {
+      if (e < cx) v = src[e];
+      else { v = src[n - e]; v.y = -v.y; }
+    }
+    ld[brev_n(e, log2n)] = v;
+  }
+  __syncthreads();
+  const int nbf = n >> 1;
+  for (int len = 2; len <= n; len <<= 1) {
+    int half = len >> 1, shift = n / len;
+    for (int bf = tl; bf < nbf; bf += TPL_X) {
+      int blk = bf / half, off = bf % half;
+      int i0 = blk * len + off, i1 = i0 + half;
+      f2 w = tw[off * shift];
+      w.y = -w.y; /* inverse */
+      f2 u = ld[i0], v = cmul(ld[i1], w);
+      ld[i0] = {u.x + v.x, u.y + v.y};
+      ld[i1] = {u.x - v.x, u.y - v.y};
+    }
+    __syncthreads();
+  }
+  if (active) {
+    float *o = out + lid * n;
+    for (int k = tl; k < n; k += TPL_X) o[k] = ld[k].x;
+  }
+}
+
+/* ------------------------------------------------------------- peak scan */
+
+struct bs_peak {
+  float v;
+  int pad;
+  long long idx;
+};
+
+__device__ __forceinline__ bool pk_better(float v, long long i, float v2,
+                                          long long i2) {
+  return (v > v2) || (v == v2 && i < i2); /* oracle tie-break [PIN-MAX] */
+}
+
+__device__ void pk_insert(float (&tv)[5], long long (&ti)[5], float v,
+                          long long i) {
+  if (!pk_better(v, i, tv[4], ti[4])) return;
+  tv[4] = v; ti[4] = i;
+  for (int k = 4; k > 0 && pk_better(tv[k], ti[k], tv[k - 1], ti[k - 1]);
+       --k) {
+    float fv = tv[k]; tv[k] = tv[k - 1]; tv[k - 1] = fv;
+    long long fi = ti[k]; ti[k] = ti[k - 1]; ti[k - 1] = fi;
+  }
+}
+
+__device__ void pk_merge_shfl(float (&tv)[5], long long (&ti)[5]) {
+  for (int off = 32; off >= 1; off >>= 1) {
+    float ov[5]; long long oi[5];
+    for (int k = 0; k < 5; ++k) {
+      ov[k] = __shfl_down(tv[k], off);
+      oi[k] = __shfl_down(ti[k], off);
+    }
+    for (int k = 0; k < 5; ++k) pk_insert(tv, ti, ov[k], oi[k]);
+  }
+}
+
+/* Tile local-maxima scan: strict 26-neighborhood maxima with periodic
+ * wrap [PIN-MAX]; per-WG top-5 -> wgbuf. */
+__global__ __launch_bounds__(256) void k_peak_tile(
+    const float *pcm, int px, int py, int pz, bs_peak *wgbuf) {
+  __shared__ float tile[(PK_TZ + 2) * (PK_TY + 2) * (PK_TX + 2)];
+  __shared__ float wv[4][5];
+  __shared__ long long wi[4][5];
+  const int tid = threadIdx.x;
+  int ntx = (px + PK_TX - 1) / PK_TX;
+  int nty = (py + PK_TY - 1) / PK_TY;
+  int bx = blockIdx.x % ntx;
+  int by = (blockIdx.x / ntx) % nty;
+  int bz = blockIdx.x / (ntx * nty);
+  int x0 = bx * PK_TX, y0 = by * PK_TY, z0 = bz * PK_TZ;
+  const int HX = PK_TX + 2, HY = PK_TY + 2, HZ = PK_TZ + 2;
+  for (int i = tid; i < HX * HY * HZ; i += 256) {
+    int lx = i % HX, t = i / HX, ly = t % HY, lz = t / HY;
+    int gx = (x0 + lx - 1 + px) % px;
+    int gy = (y0 + ly - 1 + py) % py;
+    int gz = (z0 + lz - 1 + pz) % pz;
+    tile[i] = pcm[((long)gz * py + gy) * px + gx];
+  }
+  __syncthreads();
+  float tv[5];
+  long long ti[5];
+  for (int k = 0; k < 5; ++k) { tv[k] = -3.0e38f; ti[k] = 0x7fffffffffffffffLL; }
+  for (int i = tid; i < PK_TX * PK_TY * PK_TZ; i += 256) {
+    int lx = i % PK_TX, t = i / PK_TX, ly = t % PK_TY, lz = t / PK_TY;
+    int gx = x0 + lx, gy = y0 + ly, gz = z0 + lz;
+    if (gx >= px || gy >= py || gz >= pz) continue;
+    float v = tile[((lz + 1) * HY + ly + 1) * HX + lx + 1];
+    bool ismax = true;
+    for (int dz = 0; dz <= 2 && ismax; ++dz)
+      for (int dy = 0; dy <= 2 && ismax; ++dy)
+        for (int dx = 0; dx <= 2; ++dx) {
+          if (dz == 1 && dy == 1 && dx == 1) continue;
+          if (!(v > tile[((lz + dz) * HY + ly + dy) * HX + lx + dx])) {
+            ismax = false;
+            break;
+          }
+        }
+    if (ismax)
+      pk_insert(tv, ti, v, ((long long)gz * py + gy) * px + gx);
+  }
+  pk_merge_shfl(tv, ti);
+  int lane = tid & 63, wave = tid >> 6;
+  if (lane == 0)
+    for (int k = 0; k < 5; ++k) { wv[wave][k] = tv[k]; wi[wave][k] = ti[k]; }
+  __syncthreads();
+  if (tid == 0) {
+    for (int w = 1; w < 4; ++w)
+      for (int k = 0; k < 5; ++k) pk_insert(tv, ti, wv[w][k], wi[w][k]);
+    bs_peak *o = wgbuf + (long)blockIdx.x * 5;
+    for (int k = 0; k < 5; ++k) o[k] = {tv[k], 0, ti[k]};
+  }
+}
+
+/* Merge all per-WG top-5 lists into the global top-5. Single workgroup. */
+__global__ __launch_bounds__(256) void k_peak_merge(const bs_peak *wgbuf,
+                                                    long n, bs_peak *out) {
+  __shared__ float wv[4][5];
+  __shared__ long long wi[4][5];
+  const int tid = threadIdx.x;
+  float tv[5];
+  long long ti[5];
+  for (int k = 0; k < 5; ++k) { tv[k] = -3.0e38f; ti[k] = 0x7fffffffffffffffLL; }
+  for (long i = tid; i < n; i += 256) {
+    bs_peak p = wgbuf[i];
+    if (p.v > -2.0e38f) pk_insert(tv, ti, p.v, p.idx);
+  }
+  pk_merge_shfl(tv, ti);
+  int lane = tid & 63, wave = tid >> 6;
+  if (lane == 0)
+    for (int k = 0; k < 5; ++k) { wv[wave][k] = tv[k]; wi[wave][k] = ti[k]; }
+  __syncthreads();
+  if (tid == 0) {
+    for (int w = 1; w < 4; ++w)
+      for (int k = 0; k < 5; ++k) pk_insert(tv, ti, wv[w][k], wi[w][k]);
+    for (int k = 0; k < 5; ++k) out[k] = {tv[k], 0, ti[k]};
+  }
+}
+
+/* 7-point PCM gather per peak for the sub-pixel fit [PIN-SUB]:
+ * out[p*7 + {0..6}] = f0, xm, xp, ym, yp, zm, zp (periodic wrap). */
+__global__ void k_gather_subpix(const float *pcm, int px, int py, int pz,
+                                const long long *peaks, int npeaks,
+                                float *out) {
+  int p = blockIdx.x * blockDim.x + threadIdx.x;
+  if (p >= npeaks) return;
+  long long idx = peaks[p];
+  int x = (int)(idx % px);
+  long t = idx / px;
+  int y = (int)(t % py), z = (int)(t / py);
+  float *o = out + p * 7;
+  o[0] = pcm[((long)z * py + y) * px + x];
+  o[1] = pcm[((long)z * py + y) * px + (x - 1 + px) % px];
+  o[2] = pcm[((long)z * py + y) * px + (x + 1) % px];
+  o[3] = pcm[((long)z * py + (y - 1 + py) % py) * px + x];
+  o[4] = pcm[((long)z * py + (y + 1) % py) * px + x];
+  o[5] = pcm[((long)((z - 1 + pz) % pz) * py + y) * px + x];
+  o[6] = pcm[((long)((z + 1) % pz) * py + y) * px + x];
+}
+
+/* --------------------------------------------------------------- r-test */
+
+struct bs_cand { /* overlap of A/B under integer shift, A coords [PIN-R] */
+  int lox, loy, loz;
+  int nx, ny, nz;     /* overlap dims */
+  int sx, sy, sz;     /* candidate shift (B coord = A coord + s) */
+};
+
+__device__ __forceinline__ u64 wave_sum_u64(u64 v) {
+  for (int off = 32; off >= 1; off >>= 1) v += __shfl_down(v, off);
+  return v;
+}
+
+__global__ __launch_bounds__(256) void k_rtest(
+    bs_region a, bs_region b, const bs_cand *cands, u64 *sums /* [nc][5] */) {
+  __shared__ u64 ws[4][5];
+  const bs_cand c = cands[blockIdx.y];
+  long n = (long)c.nx * c.ny * c.nz;
+  u64 pa = 0, pb = 0, paa = 0, pbb = 0, pab = 0;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x) {
+    int x = (int)(i % c.nx);
+    long t = i / c.nx;
+    int y = (int)(t % c.ny), z = (int)(t / c.ny);
+    u64 av = a.ptr[(a.oz + c.loz + z) * a.sxy + (a.oy + c.loy + y) * a.sx +
+                   a.ox + c.lox + x];
+    u64 bv = b.ptr[(b.oz + c.loz + c.sz + z) * b.sxy +
+                   (b.oy + c.loy + c.sy + y) * b.sx + b.ox + c.lox + c.sx + x];
+    pa += av; pb += bv; paa += av * av; pbb += bv * bv; pab += av * bv;
+  }
+  pa = wave_sum_u64(pa); pb = wave_sum_u64(pb); paa = wave_sum_u64(paa);
+  pbb = wave_sum_u64(pbb); pab = wave_sum_u64(pab);
+  int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  if (lane == 0) {
+    ws[wave][0] = pa; ws[wave][1] = pb; ws[wave][2] = paa;
+    ws[wave][3] = pbb; ws[wave][4] = pab;
+  }
+  __syncthreads();
+  if (threadIdx.x < 5) {
+    u64 s = ws[0][threadIdx.x] + ws[1][threadIdx.x] + ws[2][threadIdx.x] +
+            ws[3][threadIdx.x];
+    atomicAdd(&sums[(long)blockIdx.y * 5 + threadIdx.x], s);
+  }
+}
+
+/* ---------------------------------------------------------------- fusion */
+
+struct bs_dev_view {
+  const unsigned short *ptr;
+  int nx, ny, nz;
+  float inv[12];     /* world -> view-local, row-major 3x4 (fp32) */
+  float border[3], range[3];
+};
+
+__device__ __forceinline__ float blend_w(float p, int dim, float border,
+                                         float range) {
+  float dist = fminf(p, (float)(dim - 1) - p) + 1.0f; /* [PIN-BLEND] */
+  if (range <= 0.0f) return dist > border ? 1.0f : 0.0f;
+  float t = (dist - border) / range;
+  if (t <= 0.0f) return 0.0f;
+  if (t >= 1.0f) return 1.0f;
+  return 0.5f - 0.5f * __cosf(t * (float)M_PI);
+}
+
+#define BS_MAX_BLK_VIEWS 64
+
+__global__ __launch_bounds__(256) void k_fuse(
+    const bs_dev_view *views, const int *vidx, int nv, long bmx, long bmy,
+    long bmz, int bx, int by, int bz, int ftype, int dtype, float minI,
+    float invRange /* type_max/(maxI-minI) */, void *out) {
+  /* stage the (culled, small) per-block view table in LDS once */
+  __shared__ bs_dev_view sv[BS_MAX_BLK_VIEWS];
+  int nvs = min(nv, BS_MAX_BLK_VIEWS);
+  for (int i = threadIdx.x; i < nvs * (int)(sizeof(bs_dev_view) / 4);
+       i += 256) {
+    ((int *)sv)[i] = ((const int *)&views[vidx[i / (sizeof(bs_dev_view) / 4)]])
+        [i % (sizeof(bs_dev_view) / 4)];
+  }
+  __syncthreads();
+  long n = (long)bx * by * bz;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x) {
+    int x = (int)(i % bx);
+    long t = i / bx;
+    int y = (int)(t % by), z = (int)(t / by);
+    float wx = (float)(bmx + x), wy = (float)(bmy + y), wz = (float)(bmz + z);
+    float sum_wv = 0.0f, sum_w = 0.0f, vmax = 0.0f;
+    bool any = false;
+    for (int k = 0; k < nv; ++k) {
+      const bs_dev_view &v = k < nvs ? sv[k] : views[vidx[k]];
+      float px = v.inv[0] * wx + v.inv[1] * wy + v.inv[2] * wz + v.inv[3];
+      float py = v.inv[4] * wx + v.inv[5] * wy + v.inv[6] * wz + v.inv[7];
+      float pz = v.inv[8] * wx + v.inv[9] * wy + v.inv[10] * wz + v.inv[11];
+      if (!(px >= 0.0f && px <= (float)(v.nx - 1) && py >= 0.0f &&
+            py <= (float)(v.ny - 1) && pz >= 0.0f && pz <= (float)(v.nz - 1)))
+        continue; /* [PIN-BOUNDS] */
+      int x0 = (int)floorf(px), y0 = (int)floorf(py), z0 = (int)floorf(pz);
+      int x1 = min(x0 + 1, v.nx - 1), y1 = min(y0 + 1, v.ny - 1),
+          z1 = min(z0 + 1, v.nz - 1);
+      float fx = px - (float)x0, fy = py - (float)y0, fz = pz - (float)z0;
+      const unsigned short *p = v.ptr;
+      long s0 = (long)z0 * v.ny, s1 = (long)z1 * v.ny;
+      float c000 = p[(s0 + y0) * v.nx + x0], c100 = p[(s0 + y0) * v.nx + x1];
+      float c010 = p[(s0 + y1) * v.nx + x0], c110 = p[(s0 + y1) * v.nx + x1];
+      float c001 = p[(s1 + y0) * v.nx + x0], c101 = p[(s1 + y0) * v.nx + x1];
+      float c011 = p[(s1 + y1) * v.nx + x0], c111 = p[(s1 + y1) * v.nx + x1];
+      float c00 = c000 + (c100 - c000) * fx;
+      float c10 = c010 + (c110 - c010) * fx;
+      float c01 = c001 + (c101 - c001) * fx;
+      float c11 = c011 + (c111 - c011) * fx;
+      float c0 = c00 + (c10 - c00) * fy;
+      float c1 = c01 + (c11 - c01) * fy;
+      float val = c0 + (c1 - c0) * fz;
+      float w = 1.0f;
+      if (ftype == BS_FUSION_AVG_BLEND)
+        w = blend_w(px, v.nx, v.border[0], v.range[0]) *
+            blend_w(py, v.ny, v.border[1], v.range[1]) *
+            blend_w(pz, v.nz, v.border[2], v.range[2]);
+      if (ftype == BS_FUSION_MAX_INTENSITY) {
+        vmax = (any && vmax > val) ? vmax : val;
+        any = true;
+      } else {
+        sum_wv += w * val;
+        sum_w += w;
+      }
+    }
+    float o;
+    bool covered;
+    if (ftype == BS_FUSION_MAX_INTENSITY) {
+      o = any ? vmax : 0.0f;
+      covered = any;
+    } else {
+      covered = sum_w > 0.0f;
+      o = covered ? sum_wv / sum_w : 0.0f;
+    }
+    if (dtype == BS_OUT_FLOAT32) {
+      ((float *)out)[i] = o;
+    } else {
+      float s = covered ? (o - minI) * invRange : 0.0f;
+      float q = copysignf(floorf(fabsf(s) + 0.5f), s); /* [PIN-CONV] */
+      if (dtype == BS_OUT_UINT16)
+        ((unsigned short *)out)[i] =
+            (unsigned short)fminf(fmaxf(q, 0.0f), 65535.0f);
+      else
+        ((unsigned char *)out)[i] =
+            (unsigned char)fminf(fmaxf(q, 0.0f), 255.0f);
+    }
+  }
+}
+
+/* ----------------------------------------------------------- synthetic */
+
+__device__ __forceinline__ u64 splitmix64(u64 x) {
+  x += 0x9E3779B97F4A7C15ULL;
+  x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ULL;
+  x = (x ^ (x >> 27)) * 0x94D049BB133111EBULL;
+  return x ^ (x >> 31);
+}
+
+/* blobs: packed {cx,cy,cz,sigma,amp} per blob; one workgroup per blob,
+ * atomicAdd into the float accumulator over the 3-sigma support. */
+__global__ __launch_bounds__(256) void k_synth_blobs(
+    float *acc, int nx, int ny, int nz, const float *blobs) {
+  const float *b = blobs + (long)blockIdx.x * 5;
+  float cx = b[0], cy = b[1], cz = b[2], sg = b[3], am = b[4];
+  float r = 3.0f * sg;
+  int x0 = max(0, (int)floorf(cx - r)), x1 = min(nx, (int)ceilf(cx + r) + 1);
+  int y0 = max(0, (int)floorf(cy - r)), y1 = min(ny, (int)ceilf(cy + r) + 1);
+  int z0 = max(0, (int)floorf(cz - r)), z1 = min(nz, (int)ceilf(cz + r) + 1);
+  if (x0 >= x1 || y0 >= y1 || z0 >= z1) return;
+  int dx = x1 - x0, dy = y1 - y0, dz = z1 - z0;
+  long n = (long)dx * dy * dz;
+  float inv2s2 = -1.0f / (2.0f * sg * sg);
+  for (long i = threadIdx.x; i < n; i += 256) {
+    int x = (int)(i % dx);
+    long t = i / dx;
+    int y = (int)(t % dy), z = (int)(t / dy);
+    float fx = (float)(x0 + x) - cx, fy = (float)(y0 + y) - cy,
+          fz = (float)(z0 + z) - cz;
+    float v = am * __expf((fx * fx + fy * fy + fz * fz) * inv2s2);
+    atomicAdd(&acc[((long)(z0 + z) * ny + y0 + y) * nx + x0 + x], v);
+  }
+}
+
+__global__ __launch_bounds__(256) void k_synth_quant(
+    const float *acc, unsigned short *out, long n, u64 seed,
+    unsigned short floorv, unsigned short amp) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x) {
+    float noise = (float)(splitmix64(seed ^ (u64)i) % (u64)amp);
+    float v = acc[i] + (float)floorv + noise;
+    v = fminf(fmaxf(rintf(v), 0.0f), 65535.0f);
+    out[i] = (unsigned short)v;
+  }
+}
+
+/* ======================================================== host side ==== */
+
+struct bs_view_rec {
+  unsigned short *dptr;
+  long dims[3]; /* x,y,z */
+};
+
+struct bs_ev {
+  hipEvent_t a, b;
+  int kid;
+};
+
+struct bs_ctx {
+  int dev;
+  hipStream_t stream;
+  std::string err;
+  std::map<int32_t, bs_view_rec> views;
+  std::map<int, f2 *> twiddles;
+  /* workspace */
+  f2 *spec = nullptr;        /* 2 x specsize */
+  size_t spec_cap = 0;       /* in f2 units (per half) */
+  float *pcm = nullptr;
+  size_t pcm_cap = 0;
+  unsigned short *regbuf[2] = {nullptr, nullptr};
+  size_t reg_cap[2] = {0, 0};
+  bs_peak *wgpk = nullptr;
+  size_t wgpk_cap = 0;
+  float *synth_acc = nullptr;
+  size_t synth_cap = 0;
+  void *fuse_out = nullptr;
+  size_t fuse_cap = 0;
+  bs_dev_view *dviews = nullptr;
+  size_t dviews_cap = 0;
+  int *dvidx = nullptr;
+  size_t dvidx_cap = 0;
+  float *dblobs = nullptr;
+  size_t dblobs_cap = 0;
+  /* small device results + pinned mirrors */
+  bs_peak *dtop5 = nullptr;
+  bs_cand *dcands = nullptr;
+  u64 *dsums = nullptr;
+  long long *dpkidx = nullptr;
+  float *dsubpix = nullptr;
+  bs_peak *htop5 = nullptr; /* pinned */
+  u64 *hsums = nullptr;
+  float *hsubpix = nullptr;
+  /* stats */
+  bs_batch_stats stats{};
+  std::vector<bs_ev> evs;
+  std::vector<std::pair<hipEvent_t, hipEvent_t>> evpool;
+  size_t evused = 0;
+  std::mutex mu;
+};
+
+#define CHK(ctx, call)                                                      \
+  do {                                                                      \
+    hipError_t e_ = (call);                                                 \
+    if (e_ != hipSuccess) {                                                 \
+      (ctx)->err = std::string(#call) + ": " + hipGetErrorString(e_);       \
+      return BS_EHIP;                                                       \
+    }                                                                       \
+  } while (0)
+
+static thread_local std::string g_err;
+
+extern "C" const char *bs_last_error(const bs_ctx *ctx) {
+  return ctx ? ctx->err.c_str() : g_err.c_str();
+}
+
+static int ensure_dev(bs_ctx *c, void **p, size_t *cap, size_t need) {
+  if (need <= *cap) return BS_OK;
+  if (*p) (void)hipFree(*p);
+  *p = nullptr;
+  *cap = 0;
+  hipError_t e = hipMalloc(p, need);
+  if (e != hipSuccess) {
+    c->err = std::string("hipMalloc: ") + hipGetErrorString(e);
+    return BS_ENOMEM;
+  }
+  *cap = need;
+  return BS_OK;
+}
+
+extern "C" int bs_ctx_create(bs_ctx **out, int device_id) {
+  if (!out) return BS_EINVAL;
+  int n = 0;
+  if (hipGetDeviceCount(&n) != hipSuccess || device_id >= n) {
+    g_err = "no HIP device (libbigstitch requires a GPU; the product path "
+            "has no CPU fallback)";
+    return BS_ENODEV;
+  }
+  bs_ctx *c = new bs_ctx();
+  c->dev = device_id;
+  if (hipSetDevice(device_id) != hipSuccess ||
+      hipStreamCreate(&c->stream) != hipSuccess) {
+    delete c;
+    g_err = "hip init failed";
+    return BS_ENODEV;
+  }
+  /* FFT passes carve up to ~136 KB dynamic LDS (N=1024); opt in past the
+   * 64 KB default cap. */
+  (void)hipFuncSetAttribute((const void *)k_fft_pass,
+                            hipFuncAttributeMaxDynamicSharedMemorySize,
+                            160 * 1024);
+  (void)hipFuncSetAttribute((const void *)k_fft_x_fwd,
+                            hipFuncAttributeMaxDynamicSharedMemorySize,
+                            64 * 1024);
+  (void)hipFuncSetAttribute((const void *)k_fft_x_inv,
+                            hipFuncAttributeMaxDynamicSharedMemorySize,
+                            64 * 1024);
+  /* small fixed buffers */
+  if (hipMalloc(&c->dtop5, 5 * sizeof(bs_peak)) != hipSuccess ||
+      hipMalloc(&c->dcands, 64 * sizeof(bs_cand)) != hipSuccess ||
+      hipMalloc(&c->dsums, 64 * 5 * sizeof(u64)) != hipSuccess ||
+      hipMalloc(&c->dpkidx, 8 * sizeof(long long)) != hipSuccess ||
+      hipMalloc(&c->dsubpix, 8 * 7 * sizeof(float)) != hipSuccess ||
+      hipHostMalloc(&c->htop5, 5 * sizeof(bs_peak)) != hipSuccess ||
+      hipHostMalloc(&c->hsums, 64 * 5 * sizeof(u64)) != hipSuccess ||
+      hipHostMalloc(&c->hsubpix, 8 * 7 * sizeof(float)) != hipSuccess) {
+    delete c;
+    g_err = "alloc failed";
+    return BS_ENOMEM;
+  }
+  *out = c;
+  return BS_OK;
+}
+
+extern "C" void bs_ctx_destroy(bs_ctx *c) {
+  if (!c) return;
+  (void)hipSetDevice(c->dev);
+  (void)hipStreamSynchronize(c->stream);
+  for (auto &kv : c->views) (void)hipFree(kv.second.dptr);
+  for (auto &kv : c->twiddles) (void)hipFree(kv.second);
+  for (auto &pr : c->evpool) {
+    (void)hipEventDestroy(pr.first);
+    (void)hipEventDestroy(pr.second);
+  }
+  (void)hipFree(c->spec);
+  (void)hipFree(c->pcm);
+  (void)hipFree(c->regbuf[0]);
+  (void)hipFree(c->regbuf[1]);
+  (void)hipFree(c->wgpk);
+  (void)hipFree(c->synth_acc);
+  (void)hipFree(c->fuse_out);
+  (void)hipFree(c->dviews);
+  (void)hipFree(c->dvidx);
+  (void)hipFree(c->dblobs);
+  (void)hipFree(c->dtop5);
+  (void)hipFree(c->dcands);
+  (void)hipFree(c->dsums);
+  (void)hipFree(c->dpkidx);
+  (void)hipFree(c->dsubpix);
+  (void)hipHostFree(c->htop5);
+  (void)hipHostFree(c->hsums);
+  (void)hipHostFree(c->hsubpix);
+  (void)hipStreamDestroy(c->stream);
+  delete c;
+}
+
+/* ---- event-timed launch helper ---- */
+struct bs_tim {
+  bs_ctx *c;
+  int kid;
+  bs_tim(bs_ctx *c_, int kid_) : c(c_), kid(kid_) {
+    if (c->evused == c->evpool.size()) {
+      hipEvent_t a, b;
+      (void)hipEventCreate(&a);
+      (void)hipEventCreate(&b);
+      c->evpool.push_back({a, b});
+    }
+    (void)hipEventRecord(c->evpool[c->evused].first, c->stream);
+  }
+  ~bs_tim() {
+    (void)hipEventRecord(c->evpool[c->evused].second, c->stream);
+    c->evs.push_back({c->evpool[c->evused].first,
+                      c->evpool[c->evused].second, kid});
+    c->evused++;
+  }
+};
+
+static void flush_stats(bs_ctx *c) { /* call after stream sync */
+  for (auto &e : c->evs) {
+    float ms = 0.0f;
+    if (hipEventElapsedTime(&ms, e.a, e.b) == hipSuccess) {
+      c->stats.total_ms[e.kid] += ms;
+      c->stats.launches[e.kid]++;
+    }
+  }
+  c->evs.clear();
+  c->evused = 0;
+}
+
+extern "C" int bs_get_stats(bs_ctx *c, bs_batch_stats *out) {
+  if (!c || !out) return BS_EINVAL;
+  *out = c->stats;
+  return BS_OK;
+}
+extern "C" int bs_reset_stats(bs_ctx *c) {
+  if (!c) return BS_EINVAL;
+  std::memset(&c->stats, 0, sizeof(c->stats));
+  return BS_OK;
+}
+
+/* ---- views ---- */
+
+extern "C" int bs_view_upload(bs_ctx *c, int32_t id, const uint16_t *data,
+                              const int64_t dims[3]) {
+  if (!c || !data || !dims) return BS_EINVAL;
+  std::lock_guard<std::mutex> g(c->mu);
+  CHK(c, hipSetDevice(c->dev));
+  size_t n = (size_t)dims[0] * dims[1] * dims[2];
+  auto it = c->views.find(id);
+  if (it != c->views.end()) {
+    (void)hipFree(it->second.dptr);
+    c->views.erase(it);
+  }
+  unsigned short *d;
+  CHK(c, hipMalloc(&d, n * 2));
+  CHK(c, hipMemcpy(d, data, n * 2, hipMemcpyHostToDevice));
+  c->views[id] = {d, {dims[0], dims[1], dims[2]}};
+  return BS_OK;
+}
+
+extern "C" int bs_view_release(bs_ctx *c, int32_t id) {
+  if (!c) return BS_EINVAL;
+  std::lock_guard<std::mutex> g(c->mu);
+  auto it = c->views.find(id);
+  if (it == c->views.end()) return BS_ENOVIEW;
+  (void)hipFree(it->second.dptr);
+  c->views.erase(it);
+  return BS_OK;
+}
+
+extern "C" int bs_view_download(bs_ctx *c, int32_t id, uint16_t *out) {
+  if (!c || !out) return BS_EINVAL;
+  std::lock_guard<std::mutex> g(c->mu);
+  CHK(c, hipSetDevice(c->dev));
+  auto it = c->views.find(id);
+  if (it == c->views.end()) return BS_ENOVIEW;
+  size_t n = (size_t)it->second.dims[0] * it->second.dims[1] *
+             it->second.dims[2];
+  CHK(c, hipMemcpy(out, it->second.dptr, n * 2, hipMemcpyDeviceToHost));
+  return BS_OK;
+}
+
+extern "C" int bs_view_synth(bs_ctx *c, int32_t id, const int64_t dims[3],
+                             const float *blobs, int32_t n_blobs,
+                             uint32_t noise_seed, uint16_t noise_floor,
+                             uint16_t noise_amp) {
+  if (!c || !dims || (n_blobs > 0 && !blobs) || noise_amp == 0)
+    return BS_EINVAL;
+  std::lock_guard<std::mutex> g(c->mu);
+  CHK(c, hipSetDevice(c->dev));
+  long nx = dims[0], ny = dims[1], nz = dims[2];
+  size_t n = (size_t)nx * ny * nz;
+  int rc = ensure_dev(c, (void **)&c->synth_acc, &c->synth_cap, n * 4);
+  if (rc) return rc;
+  rc = ensure_dev(c, (void **)&c->dblobs, &c->dblobs_cap,
+                  (size_t)std::max(1, n_blobs) * 5 * 4);
+  if (rc) return rc;
+  auto it = c->views.find(id);
+  if (it != c->views.end()) {
+    (void)hipFree(it->second.dptr);
+    c->views.erase(it);
+  }
+  unsigned short *d;
+  CHK(c, hipMalloc(&d, n * 2));
+  CHK(c, hipMemsetAsync(c->synth_acc, 0, n * 4, c->stream));
+  if (n_blobs > 0) {
+    CHK(c, hipMemcpyAsync(c->dblobs, blobs, (size_t)n_blobs * 5 * 4,
+                          hipMemcpyHostToDevice, c->stream));
+    bs_tim t(c, BS_K_SYNTH);
+    hipLaunchKernelGGL(k_synth_blobs, dim3(n_blobs), dim3(256), 0, c->stream,
+                       c->synth_acc, (int)nx, (int)ny, (int)nz, c->dblobs);
+  }
+  {
+    bs_tim t(c, BS_K_SYNTH);
+    long blocks = std::min((long)2048, (long)((n + 255) / 256));
+    hipLaunchKernelGGL(k_synth_quant, dim3(blocks), dim3(256), 0, c->stream,
+                       c->synth_acc, d, (long)n, (u64)noise_seed * 0x100000001ULL,
+                       noise_floor, noise_amp);
+  }
+  CHK(c, hipStreamSynchronize(c->stream));
+  flush_stats(c);
+  c->views[id] = {d, {nx, ny, nz}};
+  return BS_OK;
+}
+
+/* ---- twiddles ---- */
+
+static f2 *get_twiddle(bs_ctx *c, int n) {
+  auto it = c->twiddles.find(n);
+  if (it != c->twiddles.end()) return it->second;
+  std::vector<f2> h(n / 2);
+  for (int k = 0; k < n / 2; ++k) {
+    double a = -2.0 * M_PI * k / n;
+    h[k] = {(float)std::cos(a), (float)std::sin(a)};
+  }
+  f2 *d = nullptr;
+  if (hipMalloc(&d, h.size() * sizeof(f2)) != hipSuccess) return nullptr;
+  (void)hipMemcpy(d, h.data(), h.size() * sizeof(f2), hipMemcpyHostToDevice);
+  c->twiddles[n] = d;
+  return d;
+}
+
+static int ilog2(int n) {
+  int l = 0;
+  while ((1 << l) < n) ++l;
+  return l;
+}
+static int next_pow2(int n) {
+  int p = 1;
+  while (p < n) p <<= 1;
+  return p;
+}
+
+/* ---- stitching ---- */
+
+extern "C" int bs_stitch_batch(bs_ctx *c, const bs_pair_desc *pairs, size_t np,
+                               const bs_stitch_params *prm,
+                               bs_shift_result *out) {
+  if (!c || !pairs || !prm || !out) return BS_EINVAL;
+  std::lock_guard<std::mutex> g(c->mu);
+  CHK(c, hipSetDevice(c->dev));
+  const int K = prm->peaks_to_check > 0 ? prm->peaks_to_check : 5;
+  if (K > 5) {
+    c->err = "peaks_to_check > 5 unsupported in this build";
+    return BS_EUNSUP;
+  }
+  hipEvent_t bev0, bev1;
+  if (c->evused == c->evpool.size()) {
+    hipEvent_t a, b;
+    CHK(c, hipEventCreate(&a));
+    CHK(c, hipEventCreate(&b));
+    c->evpool.push_back({a, b});
+  }
+  bev0 = c->evpool[c->evused].first;
+  bev1 = c->evpool[c->evused].second;
+  c->evused++;
+  CHK(c, hipEventRecord(bev0, c->stream));
+
+  for (size_t ip = 0; ip < np; ++ip) {
+    const bs_pair_desc &pd = pairs[ip];
+    out[ip] = {{0, 0, 0}, 0.0, 0};
+    auto ita = c->views.find(pd.view_a);
+    auto itb = c->views.find(pd.view_b);
+    if (ita == c->views.end() || itb == c->views.end()) {
+      c->err = "view not uploaded";
+      return BS_ENOVIEW;
+    }
+    int ds[3] = {prm->ds[0] > 0 ? prm->ds[0] : 1,
+                 prm->ds[1] > 0 ? prm->ds[1] : 1,
+                 prm->ds[2] > 0 ? prm->ds[2] : 1};
+    bool dsall1 = ds[0] == 1 && ds[1] == 1 && ds[2] == 1;
+    /* region descriptors (A=0, B=1) */
+    bs_region reg[2];
+    int m[2][3]; /* downsampled dims x,y,z */
+    for (int t = 0; t < 2; ++t) {
+      const bs_view_rec &vr = t == 0 ? ita->second : itb->second;
+      const int64_t *off = t == 0 ? pd.off_a : pd.off_b;
+      const int64_t *size = t == 0 ? pd.size_a : pd.size_b;
+      for (int d = 0; d < 3; ++d) {
+        if (off[d] < 0 || size[d] <= 0 || off[d] + size[d] > vr.dims[d]) {
+          c->err = "pair interval out of view bounds";
+          return BS_EINVAL;
+        }
+        m[t][d] = (int)(size[d] / ds[d]);
+        if (m[t][d] < 1) m[t][d] = 1;
+      }
+      if (dsall1) {
+        reg[t] = {vr.dptr, vr.dims[0], vr.dims[0] * vr.dims[1],
+                  off[0], off[1], off[2], m[t][0], m[t][1], m[t][2]};
+      } else {
+        size_t need = (size_t)m[t][0] * m[t][1] * m[t][2] * 2;
+        int rc = ensure_dev(c, (void **)&c->regbuf[t], &c->reg_cap[t], need);
+        if (rc) return rc;
+        bs_region src = {vr.dptr, vr.dims[0], vr.dims[0] * vr.dims[1],
+                         off[0], off[1], off[2], 0, 0, 0};
+        long nvox = (long)m[t][0] * m[t][1] * m[t][2];
+        bs_tim tt(c, BS_K_DOWNSAMPLE);
+        hipLaunchKernelGGL(k_downsample,
+                           dim3(std::min(8192L, (nvox + 255) / 256)),
+                           dim3(256), 0, c->stream, src, c->regbuf[t],
+                           m[t][0], m[t][1], m[t][2], ds[0], ds[1], ds[2]);
+        reg[t] = {c->regbuf[t], m[t][0], (long)m[t][0] * m[t][1], 0, 0, 0,
+                  m[t][0], m[t][1], m[t][2]};
+      }
+    }
+    /* padded FFT dims [PIN-PAD] */
+    int Px = next_pow2(std::max(m[0][0], m[1][0]));
+    int Py = next_pow2(std::max(m[0][1], m[1][1]));
+    int Pz = next_pow2(std::max(m[0][2], m[1][2]));
+    if (Px > 1024 || Py > 1024 || Pz > 1024) {
+      c->err = "FFT size > 1024 unsupported";
+      return BS_EUNSUP;
+    }
+    int Cx = Px / 2 + 1;
+    long Cxp = (Cx + 15) & ~15L;
+    size_t spec_half = (size_t)Pz * Py * Cxp;
+    int rc = ensure_dev(c, (void **)&c->spec, &c->spec_cap,
+                        2 * spec_half * sizeof(f2));
+    if (rc) return rc;
+    size_t pcm_n = (size_t)Pz * Py * Px;
+    rc = ensure_dev(c, (void **)&c->pcm, &c->pcm_cap, pcm_n * sizeof(float));
+    if (rc) return rc;
+    f2 *twx = get_twiddle(c, Px), *twy = get_twiddle(c, Py),
+       *twz = get_twiddle(c, Pz);
+    if (!twx || !twy || !twz) {
+      c->err = "twiddle alloc failed";
+      return BS_ENOMEM;
+    }
+    f2 *spec[2] = {c->spec, c->spec + spec_half};
+    /* forward x (R2C) per tile */
+    for (int t = 0; t < 2; ++t) {
+      long nlines = (long)reg[t].my * reg[t].mz;
+      size_t lds = ((Px / 2) + (size_t)LPB_X * Px) * sizeof(f2);
+      bs_tim tt(c, BS_K_FFT_X_FWD);
+      hipLaunchKernelGGL(k_fft_x_fwd, dim3((nlines + LPB_X - 1) / LPB_X),
+                         dim3(LPB_X * TPL_X), lds, c->stream, reg[t], spec[t],
+                         Px, ilog2(Px), Cx, Cxp, twx);
+    }
+    /* forward y: groups = z slices (z < mz), lines = Cx x-columns */
+    int nchunks = (Cx + LPB_S - 1) / LPB_S;
+    for (int t = 0; t < 2; ++t) {
+      size_t lds = ((Py / 2) + (size_t)LPB_S * Py) * sizeof(f2);
+      bs_tim tt(c, BS_K_FFT_Y_FWD);
+      hipLaunchKernelGGL(k_fft_pass, dim3(reg[t].mz * nchunks),
+                         dim3(LPB_S * TPL_S), lds, c->stream, spec[t],
+                         (const f2 *)nullptr, spec[t], Py, ilog2(Py), Cxp,
+                         (long)Py * Cxp, Cx, nchunks, reg[t].my, +1, 1.0f,
+                         twy);
+    }
+    /* forward z: groups = y rows (all Py), valid = mz */
+    for (int t = 0; t < 2; ++t) {
+      size_t lds = ((Pz / 2) + (size_t)LPB_S * Pz) * sizeof(f2);
+      bs_tim tt(c, BS_K_FFT_Z_FWD);
+      hipLaunchKernelGGL(k_fft_pass, dim3(Py * nchunks), dim3(LPB_S * TPL_S),
+                         lds, c->stream, spec[t], (const f2 *)nullptr,
+                         spec[t], Pz, ilog2(Pz), (long)Py * Cxp, Cxp, Cx,
+                         nchunks, reg[t].mz, +1, 1.0f, twz);
+    }
+    /* inverse z fused with cross-power normalise [PIN-EPS] */
+    {
+      float scale = 1.0f / ((float)Px * (float)Py * (float)Pz);
+      size_t lds = ((Pz / 2) + (size_t)LPB_S * Pz) * sizeof(f2);
+      bs_tim tt(c, BS_K_FFT_Z_INV);
+      hipLaunchKernelGGL(k_fft_pass, dim3(Py * nchunks), dim3(LPB_S * TPL_S),
+                         lds, c->stream, spec[0], spec[1], spec[0], Pz,
+                         ilog2(Pz), (long)Py * Cxp, Cxp, Cx, nchunks, Pz, -1,
+                         scale, twz);
+    }
+    /* inverse y */
+    {
+      size_t lds = ((Py / 2) + (size_t)LPB_S * Py) * sizeof(f2);
+      bs_tim tt(c, BS_K_FFT_Y_INV);
+      hipLaunchKernelGGL(k_fft_pass, dim3(Pz * nchunks), dim3(LPB_S * TPL_S),
+                         lds, c->stream, spec[0], (const f2 *)nullptr,
+                         spec[0], Py, ilog2(Py), Cxp, (long)Py * Cxp, Cx,
+                         nchunks, Py, -1, 1.0f, twy);
+    }
+    /* inverse x (C2R) -> PCM */
+    {
+      long nlines = (long)Pz * Py;
+      size_t lds = ((Px / 2) + (size_t)LPB_X * Px) * sizeof(f2);
+      bs_tim tt(c, BS_K_FFT_X_INV);
+      hipLaunchKernelGGL(k_fft_x_inv, dim3((nlines + LPB_X - 1) / LPB_X),
+                         dim3(LPB_X * TPL_X), lds, c->stream, spec[0], c->pcm,
+                         Px, ilog2(Px), Cx, Cxp, nlines, twx);
+    }
+    /* peak scan [PIN-MAX] */
+    long ntiles = (long)((Px + PK_TX - 1) / PK_TX) *
+                  ((Py + PK_TY - 1) / PK_TY) * ((Pz + PK_TZ - 1) / PK_TZ);
+    rc = ensure_dev(c, (void **)&c->wgpk, &c->wgpk_cap,
+                    (size_t)ntiles * 5 * sizeof(bs_peak));
+    if (rc) return rc;
+    {
+      bs_tim tt(c, BS_K_PEAK);
+      hipLaunchKernelGGL(k_peak_tile, dim3(ntiles), dim3(256), 0, c->stream,
+                         c->pcm, Px, Py, Pz, c->wgpk);
+    }
+    {
+      bs_tim tt(c, BS_K_PEAK_MERGE);
+      hipLaunchKernelGGL(k_peak_merge, dim3(1), dim3(256), 0, c->stream,
+                         c->wgpk, ntiles * 5, c->dtop5);
+    }
+    CHK(c, hipMemcpyAsync(c->htop5, c->dtop5, 5 * sizeof(bs_peak),
+                          hipMemcpyDeviceToHost, c->stream));
+    CHK(c, hipStreamSynchronize(c->stream));
+
+    /* host: candidates [PIN-CAND] */
+    struct HostCand {
+      bs_cand gc;
+      int rank, ci;
+      long n;
+    };
+    std::vector<HostCand> hc;
+    std::vector<long long> pkidx;
+    double min_n = prm->min_overlap_ratio *
+                   std::min((double)m[0][0] * m[0][1] * m[0][2],
+                            (double)m[1][0] * m[1][1] * m[1][2]);
+    int npk = 0;
+    for (int k = 0; k < K; ++k) {
+      if (c->htop5[k].v <= -2.0e38f) break;
+      long long idx = c->htop5[k].idx;
+      int px_ = (int)(idx % Px);
+      long tt_ = idx / Px;
+      int py_ = (int)(tt_ % Py), pz_ = (int)(tt_ / Py);
+      pkidx.push_back(idx);
+      npk = k + 1;
+      for (int ci = 0; ci < 8; ++ci) {
+        int sxyz[3] = {px_, py_, pz_};
+        if (ci & 1) sxyz[0] -= Px;
+        if (ci & 2) sxyz[1] -= Py;
+        if (ci & 4) sxyz[2] -= Pz;
+        /* overlap in A coords: lo = max(0,-s), hi = min(na, nb - s) */
+        int lo[3], hi[3];
+        bool ok = true;
+        for (int d = 0; d < 3; ++d) {
+          lo[d] = std::max(0, -sxyz[d]);
+          hi[d] = std::min(m[0][d], m[1][d] - sxyz[d]);
+          if (hi[d] <= lo[d]) ok = false;
+        }
+        if (!ok) continue;
+        long n = (long)(hi[0] - lo[0]) * (hi[1] - lo[1]) * (hi[2] - lo[2]);
+        if ((double)n < std::max(min_n, 1.0)) continue;
+        HostCand h;
+        h.gc = {lo[0], lo[1], lo[2], hi[0] - lo[0], hi[1] - lo[1],
+                hi[2] - lo[2], sxyz[0], sxyz[1], sxyz[2]};
+        h.rank = k;
+        h.ci = ci;
+        h.n = n;
+        hc.push_back(h);
+      }
+    }
+    double best_r = -3.0;
+    int best_i = -1;
+    if (!hc.empty()) {
+      std::vector<bs_cand> gc(hc.size());
+      for (size_t i = 0; i < hc.size(); ++i) gc[i] = hc[i].gc;
+      CHK(c, hipMemcpyAsync(c->dcands, gc.data(),
+                            gc.size() * sizeof(bs_cand),
+                            hipMemcpyHostToDevice, c->stream));
+      CHK(c, hipMemsetAsync(c->dsums, 0, gc.size() * 5 * sizeof(u64),
+                            c->stream));
+      {
+        bs_tim tt(c, BS_K_CORR);
+        hipLaunchKernelGGL(k_rtest, dim3(1024, (int)gc.size()), dim3(256), 0,
+                           c->stream, reg[0], reg[1], c->dcands, c->dsums);
+      }
+      CHK(c, hipMemcpyAsync(c->hsums, c->dsums, gc.size() * 5 * sizeof(u64),
+                            hipMemcpyDeviceToHost, c->stream));
+      /* subpixel gather for all peaks while r-test runs */
+      CHK(c, hipMemcpyAsync(c->dpkidx, pkidx.data(),
+                            pkidx.size() * sizeof(long long),
+                            hipMemcpyHostToDevice, c->stream));
+      {
+        bs_tim tt(c, BS_K_SUBPIX);
+        hipLaunchKernelGGL(k_gather_subpix, dim3(1), dim3(8), 0, c->stream,
+                           c->pcm, Px, Py, Pz, c->dpkidx, npk, c->dsubpix);
+      }
+      CHK(c, hipMemcpyAsync(c->hsubpix, c->dsubpix, npk * 7 * sizeof(float),
+                            hipMemcpyDeviceToHost, c->stream));
+      CHK(c, hipStreamSynchronize(c->stream));
+      /* winner: max r, ties by (rank, ci) — identical to oracle order */
+      for (size_t i = 0; i < hc.size(); ++i) {
+        const u64 *s = c->hsums + i * 5;
+        double n = (double)hc[i].n;
+        double sa = (double)s[0], sb = (double)s[1];
+        double num = (double)s[4] - sa * sb / n;
+        double da = (double)s[2] - sa * sa / n;
+        double db = (double)s[3] - sb * sb / n;
+        if (da <= 0 || db <= 0) continue;
+        double r = num / std::sqrt(da * db);
+        /* iteration is (rank asc, ci asc): keeping the first max matches
+         * the oracle's (-r, rank, ci) key ordering exactly */
+        if (best_i < 0 || r > best_r) {
+          best_r = r;
+          best_i = (int)i;
+        }
+      }
+    }
+    if (best_i < 0) continue; /* result stays invalid */
+    const HostCand &w = hc[best_i];
+    double shift[3] = {(double)w.gc.sx, (double)w.gc.sy, (double)w.gc.sz};
+    if (prm->do_subpixel) { /* [PIN-SUB] */
+      const float *f = c->hsubpix + w.rank * 7;
+      for (int d = 0; d < 3; ++d) {
+        double fm = f[1 + 2 * d], f0 = f[0], fp = f[2 + 2 * d];
+        double den = fm - 2.0 * f0 + fp;
+        if (std::fabs(den) < 1e-12) continue;
+        double o = 0.5 * (fm - fp) / den;
+        shift[d] += std::min(0.5, std::max(-0.5, o));
+      }
+    }
+    out[ip].shift[0] = shift[0] * ds[0];
+    out[ip].shift[1] = shift[1] * ds[1];
+    out[ip].shift[2] = shift[2] * ds[2];
+    out[ip].r = best_r;
+    out[ip].valid = 1;
+  }
+  CHK(c, hipEventRecord(bev1, c->stream));
+  CHK(c, hipStreamSynchronize(c->stream));
+  float bms = 0.0f;
+  (void)hipEventElapsedTime(&bms, bev0, bev1);
+  c->stats.batch_ms = bms;
+  c->stats.pairs += (long long)np;
+  flush_stats(c);
+  return BS_OK;
+}
+
+/* ---- fusion ---- */
+
+static int invert34(const double *m, float *inv) {
+  double a[9] = {m[0], m[1], m[2], m[4], m[5], m[6], m[8], m[9], m[10]};
+  double det = a[0] * (a[4] * a[8] - a[5] * a[7]) -
+               a[1] * (a[3] * a[8] - a[5] * a[6]) +
+               a[2] * (a[3] * a[7] - a[4] * a[6]);
+  if (std::fabs(det) < 1e-300) return BS_EINVAL;
+  double id = 1.0 / det;
+  double ai[9] = {(a[4] * a[8] - a[5] * a[7]) * id,
+                  (a[2] * a[7] - a[1] * a[8]) * id,
+                  (a[1] * a[5] - a[2] * a[4]) * id,
+                  (a[5] * a[6] - a[3] * a[8]) * id,
+                  (a[0] * a[8] - a[2] * a[6]) * id,
+                  (a[2] * a[3] - a[0] * a[5]) * id,
+                  (a[3] * a[7] - a[4] * a[6]) * id,
+                  (a[1] * a[6] - a[0] * a[7]) * id,
+                  (a[0] * a[4] - a[1] * a[3]) * id};
+  double t[3] = {m[3], m[7], m[11]};
+  for (int r = 0; r < 3; ++r) {
+    inv[r * 4 + 0] = (float)ai[r * 3 + 0];
+    inv[r * 4 + 1] = (float)ai[r * 3 + 1];
+    inv[r * 4 + 2] = (float)ai[r * 3 + 2];
+    inv[r * 4 + 3] = (float)-(ai[r * 3 + 0] * t[0] + ai[r * 3 + 1] * t[1] +
+                             ai[r * 3 + 2] * t[2]);
+  }
+  return BS_OK;
+}
+
+extern "C" int bs_fuse_blocks(bs_ctx *c, const bs_fuse_view *views,
+                              size_t nviews, const bs_block_desc *blocks,
+                              size_t nb, const int32_t *view_idx_per_block,
+                              const int64_t *view_idx_offsets,
+                              const bs_fuse_params *prm, void **out_blocks) {
+  if (!c || !views || !blocks || !prm || !out_blocks || !view_idx_offsets ||
+      nviews == 0)
+    return BS_EINVAL;
+  if (prm->interp != 1) {
+    c->err = "only interp=1 (trilinear) supported (reference default)";
+    return BS_EUNSUP;
+  }
+  std::lock_guard<std::mutex> g(c->mu);
+  CHK(c, hipSetDevice(c->dev));
+  /* device view table */
+  std::vector<bs_dev_view> dv(nviews);
+  for (size_t i = 0; i < nviews; ++i) {
+    auto it = c->views.find(views[i].view_id);
+    if (it == c->views.end()) {
+      c->err = "fusion view not uploaded";
+      return BS_ENOVIEW;
+    }
+    dv[i].ptr = it->second.dptr;
+    dv[i].nx = (int)it->second.dims[0];
+    dv[i].ny = (int)it->second.dims[1];
+    dv[i].nz = (int)it->second.dims[2];
+    if (invert34(views[i].affine, dv[i].inv) != BS_OK) {
+      c->err = "singular view affine";
+      return BS_EINVAL;
+    }
+    for (int d = 0; d < 3; ++d) {
+      dv[i].border[d] = views[i].blend_border[d];
+      dv[i].range[d] = views[i].blend_range[d];
+    }
+  }
+  int rc = ensure_dev(c, (void **)&c->dviews, &c->dviews_cap,
+                      nviews * sizeof(bs_dev_view));
+  if (rc) return rc;
+  CHK(c, hipMemcpyAsync(c->dviews, dv.data(), nviews * sizeof(bs_dev_view),
+                        hipMemcpyHostToDevice, c->stream));
+  long nidx = view_idx_offsets[nb];
+  rc = ensure_dev(c, (void **)&c->dvidx, &c->dvidx_cap,
+                  std::max(1L, nidx) * sizeof(int));
+  if (rc) return rc;
+  if (nidx > 0)
+    CHK(c, hipMemcpyAsync(c->dvidx, view_idx_per_block, nidx * sizeof(int),
+                          hipMemcpyHostToDevice, c->stream));
+  int esz = prm->out_dtype == BS_OUT_FLOAT32 ? 4
+            : prm->out_dtype == BS_OUT_UINT16 ? 2 : 1;
+  double denom = prm->max_intensity - prm->min_intensity;
+  float invRange =
+      (float)((prm->out_dtype == BS_OUT_UINT8 ? 255.0 : 65535.0) /
+              (denom != 0.0 ? denom : 1.0));
+  size_t maxvox = 0;
+  for (size_t i = 0; i < nb; ++i)
+    maxvox = std::max(maxvox, (size_t)blocks[i].size[0] * blocks[i].size[1] *
+                                  blocks[i].size[2]);
+  rc = ensure_dev(c, (void **)&c->fuse_out, &c->fuse_cap, maxvox * esz);
+  if (rc) return rc;
+  for (size_t i = 0; i < nb; ++i) {
+    const bs_block_desc &bd = blocks[i];
+    long nvox = (long)bd.size[0] * bd.size[1] * bd.size[2];
+    int nvb = (int)(view_idx_offsets[i + 1] - view_idx_offsets[i]);
+    {
+      bs_tim tt(c, BS_K_FUSE);
+      hipLaunchKernelGGL(k_fuse, dim3(std::min(16384L, (nvox + 255) / 256)),
+                         dim3(256), 0, c->stream, c->dviews,
+                         c->dvidx + view_idx_offsets[i], nvb, bd.min[0],
+                         bd.min[1], bd.min[2], (int)bd.size[0],
+                         (int)bd.size[1], (int)bd.size[2], prm->fusion_type,
+                         prm->out_dtype, (float)prm->min_intensity, invRange,
+                         c->fuse_out);
+    }
+    CHK(c, hipMemcpyAsync(out_blocks[i], c->fuse_out, nvox * esz,
+                          hipMemcpyDeviceToHost, c->stream));
+  }
+  CHK(c, hipStreamSynchronize(c->stream));
+  c->stats.blocks += (long long)nb;
+  flush_stats(c);
+  return BS_OK;
+}

@@ -173,19 +173,36 @@ int bs_fuse_blocks(bs_ctx *ctx, const bs_fuse_view *views, size_t nviews,
 
 /* ------------------------------------------------------- instrumentation */
 
-/* Per-batch timing of the dominant kernels, HIP-event measured on the
- * launch stream (bench.py roofline evidence). Values are averages over the
- * most recent bs_stitch_batch / bs_fuse_blocks call on this ctx. */
+/* Per-kernel timing, HIP-event measured on the launch stream (bench.py
+ * roofline evidence: average launch duration of the dominant kernel).
+ * Accumulated since the last bs_reset_stats on this ctx. */
+enum bs_kernel_id {
+  BS_K_DOWNSAMPLE = 0,
+  BS_K_FFT_X_FWD,  /* R2C load + x-line FFT */
+  BS_K_FFT_Y_FWD,
+  BS_K_FFT_Z_FWD,
+  BS_K_FFT_Z_INV,  /* fused cross-power normalize + inverse z pass */
+  BS_K_FFT_Y_INV,
+  BS_K_FFT_X_INV,  /* inverse x pass + C2R (PCM write) */
+  BS_K_PEAK,       /* PCM local-maxima top-5 tile scan */
+  BS_K_PEAK_MERGE,
+  BS_K_CORR,       /* candidate cross-correlation integer sums */
+  BS_K_SUBPIX,     /* 7-point PCM gather */
+  BS_K_FUSE,       /* inverse-affine trilinear blend fusion */
+  BS_K_SYNTH,      /* bench-only synthetic tile render */
+  BS_K_COUNT
+};
+
 typedef struct {
-  double fft_ms_per_pair;     /* sum of all FFT-pass kernel time / pairs */
-  double xpow_ms_per_pair;    /* cross-power normalize */
-  double peak_ms_per_pair;    /* PCM peak scan + merge */
-  double corr_ms_per_pair;    /* candidate cross-correlation r-tests */
-  double total_ms;            /* whole batch, event-bracketed */
-  double fuse_ms_per_block;   /* K7 kernel time / blocks (fusion calls) */
+  double total_ms[BS_K_COUNT];
+  long long launches[BS_K_COUNT];
+  double batch_ms;   /* whole-batch event-bracketed time (last batch) */
+  long long pairs;   /* pairs processed since reset */
+  long long blocks;  /* fusion blocks since reset */
 } bs_batch_stats;
 
 int bs_get_stats(bs_ctx *ctx, bs_batch_stats *out);
+int bs_reset_stats(bs_ctx *ctx);
 
 #ifdef __cplusplus
 }
