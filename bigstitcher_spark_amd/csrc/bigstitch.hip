@@ -104,7 +104,7 @@ __global__ __launch_bounds__(256) void k_downsample(
  * One workgroup = LPB_X lines x TPL_X threads. Dynamic LDS:
  * [ tw: n/2 f2 | data: LPB_X * n f2 ]. */
 __global__ __launch_bounds__(LPB_X *TPL_X) void k_fft_x_fwd(
-    bs_region in, f2 *out, int n, int log2n, int cx, long cxp,
+    bs_region in, f2 *out, int n, int log2n, int cx, long cxp, int py,
     const f2 *twg) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   f2 *tw = (f2 *)smem;
@@ -140,7 +140,7 @@ __global__ __launch_bounds__(LPB_X *TPL_X) void k_fft_x_fwd(
     __syncthreads();
   }
   if (active) {
-    f2 *o = out + lid * cxp;
+    f2 *o = out + ((long)z * py + y) * cxp; /* spectrum rows are Py-strided */
     for (int k = tl; k < cx; k += TPL_X) o[k] = ld[k];
   }
 }
@@ -1001,7 +1001,7 @@ extern "C" int bs_stitch_batch(bs_ctx *c, const bs_pair_desc *pairs, size_t np,
       bs_tim tt(c, BS_K_FFT_X_FWD);
       hipLaunchKernelGGL(k_fft_x_fwd, dim3((nlines + LPB_X - 1) / LPB_X),
                          dim3(LPB_X * TPL_X), lds, c->stream, reg[t], spec[t],
-                         Px, ilog2(Px), Cx, Cxp, twx);
+                         Px, ilog2(Px), Cx, Cxp, Py, twx);
     }
     /* forward y: groups = z slices (z < mz), lines = Cx x-columns */
     int nchunks = (Cx + LPB_S - 1) / LPB_S;
