@@ -21,7 +21,20 @@ __all__ = [
     "grid_create",
     "transformed_bbox",
     "find_overlapping_views",
+    "shard",
 ]
+
+
+def shard(n_units: int, world: int, rank: int):
+    """Deterministic hash-shard of independent work-unit ids over ranks
+    (`device = hash(unit) % nGPUs`; the reference treats
+    pairs/blocks as unordered independent RDD elements). Disjoint and
+    covering by construction."""
+    mask = np.arange(n_units, dtype=np.uint64)
+    # splitmix64-style mix so adjacent ids spread across ranks
+    h = (mask + np.uint64(0x9E3779B97F4A7C15)) * np.uint64(0xBF58476D1CE4E5B9)
+    h ^= h >> np.uint64(27)
+    return np.flatnonzero((h % np.uint64(world)) == np.uint64(rank))
 
 
 def overlap_interval(dims_a, pos_a, dims_b, pos_b):

@@ -64,20 +64,32 @@ def _next_pow2(n: int) -> int:
     return p
 
 
-def pcm(a: np.ndarray, b: np.ndarray):
+def pcm(a: np.ndarray, b: np.ndarray, workers: int = 1):
     """Phase-correlation matrix of two (nz,ny,nx) uint16 arrays.
 
-    Returns (pcm float64 (pz,py,px), padded shape). [PIN-PAD] [PIN-EPS]"""
+    Returns (pcm float64 (pz,py,px), padded shape). [PIN-PAD] [PIN-EPS]
+    workers != 1 switches to scipy.fft's multithreaded pocketfft — same
+    algorithm; used only by bench.py's timed cpu_baseline leg."""
     shape = tuple(
         _next_pow2(max(sa, sb)) for sa, sb in zip(a.shape, b.shape)
     )
-    fa = np.fft.rfftn(a, s=shape, axes=(0, 1, 2))
-    fb = np.fft.rfftn(b, s=shape, axes=(0, 1, 2))
+    if workers == 1:
+        fa = np.fft.rfftn(a, s=shape, axes=(0, 1, 2))
+        fb = np.fft.rfftn(b, s=shape, axes=(0, 1, 2))
+    else:
+        from scipy import fft as sfft
+
+        fa = sfft.rfftn(a, s=shape, axes=(0, 1, 2), workers=workers)
+        fb = sfft.rfftn(b, s=shape, axes=(0, 1, 2), workers=workers)
     q = np.conj(fa) * fb
     mag = np.abs(q)
     with np.errstate(invalid="ignore", divide="ignore"):
         q = np.where(mag < 1e-20, 0.0, q / mag)
-    return np.fft.irfftn(q, s=shape, axes=(0, 1, 2)), shape
+    if workers == 1:
+        return np.fft.irfftn(q, s=shape, axes=(0, 1, 2)), shape
+    from scipy import fft as sfft
+
+    return sfft.irfftn(q, s=shape, axes=(0, 1, 2), workers=workers), shape
 
 
 def _local_maxima_topk(p: np.ndarray, k: int):
@@ -85,13 +97,13 @@ def _local_maxima_topk(p: np.ndarray, k: int):
 
     Returns list of (value, (pz, py, px)) sorted by value descending,
     ties broken by ascending linear index."""
-    is_max = np.ones(p.shape, dtype=bool)
-    for dz in (-1, 0, 1):
-        for dy in (-1, 0, 1):
-            for dx in (-1, 0, 1):
-                if dz == 0 and dy == 0 and dx == 0:
-                    continue
-                is_max &= p > np.roll(p, (dz, dy, dx), axis=(0, 1, 2))
+    from scipy.ndimage import maximum_filter
+
+    # max over the 26 neighbors (footprint excludes the center), periodic
+    # wrap; strict '>' against it == the straightforward 26-roll compare.
+    fp = np.ones((3, 3, 3), bool)
+    fp[1, 1, 1] = False
+    is_max = p > maximum_filter(p, footprint=fp, mode="wrap")
     idx = np.flatnonzero(is_max)
     if idx.size == 0:
         return []
@@ -172,6 +184,7 @@ def phase_correlation_shift(
     peaks_to_check: int = 5,
     do_subpixel: bool = True,
     min_overlap_ratio: float = 0.25,
+    workers: int = 1,
 ):
     """Full restatement of PairwiseStitching.getShift for one tile pair.
 
@@ -181,7 +194,7 @@ def phase_correlation_shift(
     (include/bigstitch.h)."""
     ad = downsample(a, ds)
     bd = downsample(b, ds)
-    p, _shape = pcm(ad, bd)
+    p, _shape = pcm(ad, bd, workers=workers)
     peaks = _local_maxima_topk(p, peaks_to_check)
     min_n = min_overlap_ratio * min(ad.size, bd.size)
     best = None  # (r, peak_rank, cand_idx, cand_shift, peak)

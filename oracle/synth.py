@@ -92,6 +92,34 @@ def pair_blobs(shape_zyx, true_shift_xyz, seed, margin=12.0):
     return blobs_a, blobs_b
 
 
+def pair_blobs_union(shape_zyx, true_shift_xyz, seed, margin=12.0,
+                     speck_density=SPECK_DENSITY, blob_density=BLOB_DENSITY):
+    """Like pair_blobs, but the scene spans the UNION of the two tile
+    windows (needed for large shifts, e.g. the 10%-overlap bench pairs of
+    BASELINE.json configs[1]: s_x ~ 0.9*nx). Returns (blobs_a, blobs_b)."""
+    nz, ny, nx = shape_zyx
+    s = np.asarray(true_shift_xyz, np.float64)  # x,y,z
+    dims = np.array([nx, ny, nz], np.float64)
+    lo = np.minimum(0.0, -s) - margin
+    hi = np.maximum(dims, dims - s) + margin
+    vol = float(np.prod(hi - lo))
+    rng = np.random.default_rng(seed)
+    n_speck = max(8, int(round(speck_density * vol)))
+    n_blob = int(round(blob_density * vol))
+    cs = rng.uniform(lo, hi, size=(n_speck, 3))
+    ss = rng.uniform(1.0, 2.2, size=(n_speck, 1))
+    as_ = rng.uniform(500.0, 3000.0, size=(n_speck, 1))
+    cb = rng.uniform(lo, hi, size=(n_blob, 3))
+    sb = rng.uniform(2.0, 8.0, size=(n_blob, 1))
+    ab = rng.uniform(2000.0, 20000.0, size=(n_blob, 1))
+    blobs_a = np.vstack(
+        [np.hstack([cs, ss, as_]), np.hstack([cb, sb, ab])]
+    ).astype(np.float32)
+    blobs_b = blobs_a.copy()
+    blobs_b[:, :3] += s.astype(np.float32)
+    return blobs_a, blobs_b
+
+
 def make_pair(shape_zyx, true_shift_xyz, seed=17):
     """Render a tile pair with known shift. Returns (a, b) uint16."""
     blobs_a, blobs_b = pair_blobs(shape_zyx, true_shift_xyz, seed)
