@@ -44,10 +44,10 @@
 #define LPB_X 4    /* lines per block, contiguous-line passes */
 #define TPL_X 64
 #define LPB_S 16   /* lines per block, strided passes */
-#define TPL_S 16
-#define PK_TX 64   /* peak-scan tile */
-#define PK_TY 4
-#define PK_TZ 4
+#define TPL_S 32   /* 512-thread blocks: 2x the waves of the v1 16x16 */
+#define PK_TX 128  /* peak-scan tile */
+#define PK_TY 8
+#define PK_TZ 8
 
 typedef unsigned long long u64;
 
@@ -61,6 +61,55 @@ __device__ __forceinline__ f2 conjmul(f2 a, f2 b) { /* conj(a)*b */
 }
 __device__ __forceinline__ unsigned brev_n(unsigned j, int log2n) {
   return __brev(j) >> (32 - log2n);
+}
+
+/* In-LDS radix-2^2 DIT FFT over bit-reversed-loaded data. Two radix-2
+ * stages are fused per LDS round (4 elems in registers), halving LDS
+ * traffic and barriers vs plain radix-2; odd log2n does one radix-2
+ * stage first (twiddle-free: w=1). Element e lives at data[base + e*ES].
+ * tw holds e^{-2*pi*i*k/n}, k < n/2; dir<0 conjugates (inverse). Every
+ * thread of the block must call this (it contains __syncthreads). */
+template <int ES, int TPL>
+__device__ __forceinline__ void fft_lds(f2 *data, long base, int n,
+                                        int log2n, int tl, const f2 *tw,
+                                        int dir) {
+#define D_(e) data[base + (long)(e) * ES]
+  int h = 1;
+  if (log2n & 1) {
+    for (int bf = tl; bf < (n >> 1); bf += TPL) {
+      f2 u = D_(2 * bf), v = D_(2 * bf + 1);
+      D_(2 * bf) = {u.x + v.x, u.y + v.y};
+      D_(2 * bf + 1) = {u.x - v.x, u.y - v.y};
+    }
+    h = 2;
+    __syncthreads();
+  }
+  for (; h < n; h <<= 2) {
+    const int q = n >> 2;
+    const int s1 = n / (2 * h), s2 = n / (4 * h);
+    for (int g = tl; g < q; g += TPL) {
+      int off = g % h, blk = g / h;
+      int i = blk * 4 * h + off;
+      f2 a = D_(i), b = D_(i + h), c = D_(i + 2 * h), d = D_(i + 3 * h);
+      f2 w1 = tw[off * s1];
+      if (dir < 0) w1.y = -w1.y;
+      f2 t1 = cmul(b, w1), t2 = cmul(d, w1);
+      f2 A = {a.x + t1.x, a.y + t1.y}, B = {a.x - t1.x, a.y - t1.y};
+      f2 Cc = {c.x + t2.x, c.y + t2.y}, Dd = {c.x - t2.x, c.y - t2.y};
+      f2 w2a = tw[off * s2], w2b = tw[(off + h) * s2];
+      if (dir < 0) {
+        w2a.y = -w2a.y;
+        w2b.y = -w2b.y;
+      }
+      f2 u1 = cmul(Cc, w2a), u2 = cmul(Dd, w2b);
+      D_(i) = {A.x + u1.x, A.y + u1.y};
+      D_(i + 2 * h) = {A.x - u1.x, A.y - u1.y};
+      D_(i + h) = {B.x + u2.x, B.y + u2.y};
+      D_(i + 3 * h) = {B.x - u2.x, B.y - u2.y};
+    }
+    __syncthreads();
+  }
+#undef D_
 }
 
 /* -------------------------------------------------------- region descs */
@@ -126,19 +175,7 @@ __global__ __launch_bounds__(LPB_X *TPL_X) void k_fft_x_fwd(
     ld[brev_n(j, log2n)] = {v, 0.0f};
   }
   __syncthreads();
-  const int nbf = n >> 1;
-  for (int len = 2; len <= n; len <<= 1) {
-    int half = len >> 1, shift = n / len;
-    for (int bf = tl; bf < nbf; bf += TPL_X) {
-      int blk = bf / half, off = bf % half;
-      int i0 = blk * len + off, i1 = i0 + half;
-      f2 w = tw[off * shift]; /* e^{-2pi i off/len} */
-      f2 u = ld[i0], v = cmul(ld[i1], w);
-      ld[i0] = {u.x + v.x, u.y + v.y};
-      ld[i1] = {u.x - v.x, u.y - v.y};
-    }
-    __syncthreads();
-  }
+  fft_lds<1, TPL_X>(data, (long)line * n, n, log2n, tl, tw, +1);
   if (active) {
     f2 *o = out + ((long)z * py + y) * cxp; /* spectrum rows are Py-strided */
     for (int k = tl; k < cx; k += TPL_X) o[k] = ld[k];
@@ -183,21 +220,7 @@ __global__ __launch_bounds__(LPB_S *TPL_S) void k_fft_pass(
     data[(long)brev_n(e, log2n) * LPB_S + line] = v;
   }
   __syncthreads();
-  const int nbf = n >> 1;
-  for (int len = 2; len <= n; len <<= 1) {
-    int half = len >> 1, shift = n / len;
-    for (int bf = tl; bf < nbf; bf += TPL_S) {
-      int blk = bf / half, off = bf % half;
-      long i0 = (long)(blk * len + off) * LPB_S + line;
-      long i1 = i0 + (long)half * LPB_S;
-      f2 w = tw[off * shift];
-      if (dir < 0) w.y = -w.y;
-      f2 u = data[i0], v = cmul(data[i1], w);
-      data[i0] = {u.x + v.x, u.y + v.y};
-      data[i1] = {u.x - v.x, u.y - v.y};
-    }
-    __syncthreads();
-  }
+  fft_lds<LPB_S, TPL_S>(data, (long)line, n, log2n, tl, tw, dir);
   if (active)
     for (int e = tl; e < n; e += TPL_S)
       out[base + e * estride] = data[(long)e * LPB_S + line];
@@ -228,20 +251,7 @@ __global__ __launch_bounds__(LPB_X *TPL_X) void k_fft_x_inv(
     ld[brev_n(e, log2n)] = v;
   }
   __syncthreads();
-  const int nbf = n >> 1;
-  for (int len = 2; len <= n; len <<= 1) {
-    int half = len >> 1, shift = n / len;
-    for (int bf = tl; bf < nbf; bf += TPL_X) {
-      int blk = bf / half, off = bf % half;
-      int i0 = blk * len + off, i1 = i0 + half;
-      f2 w = tw[off * shift];
-      w.y = -w.y; /* inverse */
-      f2 u = ld[i0], v = cmul(ld[i1], w);
-      ld[i0] = {u.x + v.x, u.y + v.y};
-      ld[i1] = {u.x - v.x, u.y - v.y};
-    }
-    __syncthreads();
-  }
+  fft_lds<1, TPL_X>(data, (long)line * n, n, log2n, tl, tw, -1);
   if (active) {
     float *o = out + lid * n;
     for (int k = tl; k < n; k += TPL_X) o[k] = ld[k].x;
@@ -284,7 +294,8 @@ __device__ void pk_merge_shfl(float (&tv)[5], long long (&ti)[5]) {
 }
 
 /* Tile local-maxima scan: strict 26-neighborhood maxima with periodic
- * wrap [PIN-MAX]; per-WG top-5 -> wgbuf. */
+ * wrap [PIN-MAX]; per-WG top-5 -> wgbuf. Interior tiles (the common
+ * case) load their halo without any modulo index math. */
 __global__ __launch_bounds__(256) void k_peak_tile(
     const float *pcm, int px, int py, int pz, bs_peak *wgbuf) {
   __shared__ float tile[(PK_TZ + 2) * (PK_TY + 2) * (PK_TX + 2)];
@@ -298,12 +309,23 @@ __global__ __launch_bounds__(256) void k_peak_tile(
   int bz = blockIdx.x / (ntx * nty);
   int x0 = bx * PK_TX, y0 = by * PK_TY, z0 = bz * PK_TZ;
   const int HX = PK_TX + 2, HY = PK_TY + 2, HZ = PK_TZ + 2;
-  for (int i = tid; i < HX * HY * HZ; i += 256) {
-    int lx = i % HX, t = i / HX, ly = t % HY, lz = t / HY;
-    int gx = (x0 + lx - 1 + px) % px;
-    int gy = (y0 + ly - 1 + py) % py;
-    int gz = (z0 + lz - 1 + pz) % pz;
-    tile[i] = pcm[((long)gz * py + gy) * px + gx];
+  const bool interior = x0 > 0 && y0 > 0 && z0 > 0 && x0 + PK_TX < px &&
+                        y0 + PK_TY < py && z0 + PK_TZ < pz;
+  if (interior) {
+    const float *base =
+        pcm + ((long)(z0 - 1) * py + (y0 - 1)) * px + (x0 - 1);
+    for (int i = tid; i < HX * HY * HZ; i += 256) {
+      int lx = i % HX, t = i / HX, ly = t % HY, lz = t / HY;
+      tile[i] = base[((long)lz * py + ly) * px + lx];
+    }
+  } else {
+    for (int i = tid; i < HX * HY * HZ; i += 256) {
+      int lx = i % HX, t = i / HX, ly = t % HY, lz = t / HY;
+      int gx = (x0 + lx - 1 + px) % px;
+      int gy = (y0 + ly - 1 + py) % py;
+      int gz = (z0 + lz - 1 + pz) % pz;
+      tile[i] = pcm[((long)gz * py + gy) * px + gx];
+    }
   }
   __syncthreads();
   float tv[5];
@@ -340,16 +362,20 @@ __global__ __launch_bounds__(256) void k_peak_tile(
   }
 }
 
-/* Merge all per-WG top-5 lists into the global top-5. Single workgroup. */
+/* Merge per-WG top-5 lists: each block covers a contiguous slice of
+ * wgbuf and writes its own top-5 to out + 5*blockIdx.x. Launched twice
+ * (hierarchical): many blocks -> 1 block. */
 __global__ __launch_bounds__(256) void k_peak_merge(const bs_peak *wgbuf,
                                                     long n, bs_peak *out) {
   __shared__ float wv[4][5];
   __shared__ long long wi[4][5];
   const int tid = threadIdx.x;
+  long per = (n + gridDim.x - 1) / gridDim.x;
+  long lo = (long)blockIdx.x * per, hi = min(n, lo + per);
   float tv[5];
   long long ti[5];
   for (int k = 0; k < 5; ++k) { tv[k] = -3.0e38f; ti[k] = 0x7fffffffffffffffLL; }
-  for (long i = tid; i < n; i += 256) {
+  for (long i = lo + tid; i < hi; i += 256) {
     bs_peak p = wgbuf[i];
     if (p.v > -2.0e38f) pk_insert(tv, ti, p.v, p.idx);
   }
@@ -361,7 +387,8 @@ __global__ __launch_bounds__(256) void k_peak_merge(const bs_peak *wgbuf,
   if (tid == 0) {
     for (int w = 1; w < 4; ++w)
       for (int k = 0; k < 5; ++k) pk_insert(tv, ti, wv[w][k], wi[w][k]);
-    for (int k = 0; k < 5; ++k) out[k] = {tv[k], 0, ti[k]};
+    bs_peak *o = out + (long)blockIdx.x * 5;
+    for (int k = 0; k < 5; ++k) o[k] = {tv[k], 0, ti[k]};
   }
 }
 
@@ -620,6 +647,7 @@ struct bs_ctx {
   float *dblobs = nullptr;
   size_t dblobs_cap = 0;
   /* small device results + pinned mirrors */
+  bs_peak *dmerge = nullptr; /* 64*5 intermediate merge slots */
   bs_peak *dtop5 = nullptr;
   bs_cand *dcands = nullptr;
   u64 *dsums = nullptr;
@@ -693,7 +721,8 @@ extern "C" int bs_ctx_create(bs_ctx **out, int device_id) {
                             hipFuncAttributeMaxDynamicSharedMemorySize,
                             64 * 1024);
   /* small fixed buffers */
-  if (hipMalloc(&c->dtop5, 5 * sizeof(bs_peak)) != hipSuccess ||
+  if (hipMalloc(&c->dmerge, 64 * 5 * sizeof(bs_peak)) != hipSuccess ||
+      hipMalloc(&c->dtop5, 5 * sizeof(bs_peak)) != hipSuccess ||
       hipMalloc(&c->dcands, 64 * sizeof(bs_cand)) != hipSuccess ||
       hipMalloc(&c->dsums, 64 * 5 * sizeof(u64)) != hipSuccess ||
       hipMalloc(&c->dpkidx, 8 * sizeof(long long)) != hipSuccess ||
@@ -729,6 +758,7 @@ extern "C" void bs_ctx_destroy(bs_ctx *c) {
   (void)hipFree(c->dviews);
   (void)hipFree(c->dvidx);
   (void)hipFree(c->dblobs);
+  (void)hipFree(c->dmerge);
   (void)hipFree(c->dtop5);
   (void)hipFree(c->dcands);
   (void)hipFree(c->dsums);
@@ -1064,8 +1094,12 @@ extern "C" int bs_stitch_batch(bs_ctx *c, const bs_pair_desc *pairs, size_t np,
     }
     {
       bs_tim tt(c, BS_K_PEAK_MERGE);
+      long nent = ntiles * 5;
+      int nb1 = (int)std::min(64L, (nent + 1279) / 1280);
+      hipLaunchKernelGGL(k_peak_merge, dim3(nb1), dim3(256), 0, c->stream,
+                         c->wgpk, nent, c->dmerge);
       hipLaunchKernelGGL(k_peak_merge, dim3(1), dim3(256), 0, c->stream,
-                         c->wgpk, ntiles * 5, c->dtop5);
+                         c->dmerge, (long)nb1 * 5, c->dtop5);
     }
     CHK(c, hipMemcpyAsync(c->htop5, c->dtop5, 5 * sizeof(bs_peak),
                           hipMemcpyDeviceToHost, c->stream));
