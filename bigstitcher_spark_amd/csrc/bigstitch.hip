@@ -126,24 +126,23 @@ struct bs_region { /* a (possibly strided) uint16 sub-volume */
 __global__ __launch_bounds__(256) void k_downsample(
     bs_region in, unsigned short *out, int mx, int my, int mz,
     int dsx, int dsy, int dsz) {
-  long n = (long)mx * my * mz;
+  long nrows = (long)my * mz;
   float inv = 1.0f / (dsx * dsy * dsz);
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += (long)gridDim.x * blockDim.x) {
-    int x = (int)(i % mx);
-    long t = i / mx;
-    int y = (int)(t % my);
-    int z = (int)(t / my);
-    float s = 0.0f;
-    for (int kz = 0; kz < dsz; ++kz)
-      for (int ky = 0; ky < dsy; ++ky)
-        for (int kx = 0; kx < dsx; ++kx) {
-          long a = (in.oz + (long)z * dsz + kz) * in.sxy +
-                   (in.oy + (long)y * dsy + ky) * in.sx +
-                   (in.ox + (long)x * dsx + kx);
-          s += (float)in.ptr[a];
-        }
-    out[i] = (unsigned short)__float2int_rn(s * inv); /* [PIN-DS] rint */
+  for (long row = blockIdx.x; row < nrows; row += gridDim.x) {
+    int y = (int)(row % my), z = (int)(row / my);
+    unsigned short *orow = out + row * mx;
+    for (int x = threadIdx.x; x < mx; x += blockDim.x) {
+      float s = 0.0f;
+      for (int kz = 0; kz < dsz; ++kz)
+        for (int ky = 0; ky < dsy; ++ky)
+          for (int kx = 0; kx < dsx; ++kx) {
+            long a = (in.oz + (long)z * dsz + kz) * in.sxy +
+                     (in.oy + (long)y * dsy + ky) * in.sx +
+                     (in.ox + (long)x * dsx + kx);
+            s += (float)in.ptr[a];
+          }
+      orow[x] = (unsigned short)__float2int_rn(s * inv); /* [PIN-DS] */
+    }
   }
 }
 
@@ -161,24 +160,29 @@ __global__ __launch_bounds__(LPB_X *TPL_X) void k_fft_x_fwd(
   const int tid = threadIdx.x;
   const int tl = tid % TPL_X, line = tid / TPL_X;
   for (int i = tid; i < (n >> 1); i += LPB_X * TPL_X) tw[i] = twg[i];
-
-  long lid = (long)blockIdx.x * LPB_X + line;
-  long nlines = (long)in.my * in.mz;
-  bool active = lid < nlines;
-  int y = active ? (int)(lid % in.my) : 0;
-  int z = active ? (int)(lid / in.my) : 0;
-  const unsigned short *src =
-      in.ptr + (in.oz + z) * in.sxy + (in.oy + y) * in.sx + in.ox;
-  f2 *ld = data + (long)line * n;
-  for (int j = tl; j < n; j += TPL_X) {
-    float v = (active && j < in.mx) ? (float)src[j] : 0.0f;
-    ld[brev_n(j, log2n)] = {v, 0.0f};
-  }
   __syncthreads();
-  fft_lds<1, TPL_X>(data, (long)line * n, n, log2n, tl, tw, +1);
-  if (active) {
-    f2 *o = out + ((long)z * py + y) * cxp; /* spectrum rows are Py-strided */
-    for (int k = tl; k < cx; k += TPL_X) o[k] = ld[k];
+
+  long nlines = (long)in.my * in.mz;
+  long ngroups = (nlines + LPB_X - 1) / LPB_X;
+  f2 *ld = data + (long)line * n;
+  for (long grp = blockIdx.x; grp < ngroups; grp += gridDim.x) {
+    long lid = grp * LPB_X + line;
+    bool active = lid < nlines;
+    int y = active ? (int)(lid % in.my) : 0;
+    int z = active ? (int)(lid / in.my) : 0;
+    const unsigned short *src =
+        in.ptr + (in.oz + z) * in.sxy + (in.oy + y) * in.sx + in.ox;
+    for (int j = tl; j < n; j += TPL_X) {
+      float v = (active && j < in.mx) ? (float)src[j] : 0.0f;
+      ld[brev_n(j, log2n)] = {v, 0.0f};
+    }
+    __syncthreads();
+    fft_lds<1, TPL_X>(data, (long)line * n, n, log2n, tl, tw, +1);
+    if (active) {
+      f2 *o = out + ((long)z * py + y) * cxp; /* rows are Py-strided */
+      for (int k = tl; k < cx; k += TPL_X) o[k] = ld[k];
+    }
+    __syncthreads(); /* LDS reused next group */
   }
 }
 
@@ -190,40 +194,45 @@ __global__ __launch_bounds__(LPB_X *TPL_X) void k_fft_x_fwd(
  * [PIN-EPS]) instead of in. In-place safe (each WG owns its lines). */
 __global__ __launch_bounds__(LPB_S *TPL_S) void k_fft_pass(
     const f2 *in, const f2 *in2, f2 *out, int n, int log2n, long estride,
-    long gstride, int nlines, int nchunks, int valid, int dir, float scale,
-    const f2 *twg) {
+    long gstride, int nlines, int nchunks, int ngroups, int valid, int dir,
+    float scale, const f2 *twg) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   f2 *tw = (f2 *)smem;
   f2 *data = tw + (n >> 1);
   const int tid = threadIdx.x;
   const int line = tid % LPB_S, tl = tid / LPB_S;
   for (int i = tid; i < (n >> 1); i += LPB_S * TPL_S) tw[i] = twg[i];
-
-  const int group = blockIdx.x / nchunks;
-  const int x = (blockIdx.x % nchunks) * LPB_S + line;
-  const bool active = x < nlines;
-  const long base = (long)group * gstride + x;
-  for (int e = tl; e < n; e += TPL_S) {
-    f2 v = {0.0f, 0.0f};
-    if (active && e < valid) {
-      if (in2) {
-        f2 q = conjmul(in[base + e * estride], in2[base + e * estride]);
-        float m2 = q.x * q.x + q.y * q.y;
-        if (m2 > 1e-40f) {
-          float s = scale / sqrtf(m2);
-          v = {q.x * s, q.y * s};
-        }
-      } else {
-        v = in[base + e * estride];
-      }
-    }
-    data[(long)brev_n(e, log2n) * LPB_S + line] = v;
-  }
   __syncthreads();
-  fft_lds<LPB_S, TPL_S>(data, (long)line, n, log2n, tl, tw, dir);
-  if (active)
-    for (int e = tl; e < n; e += TPL_S)
-      out[base + e * estride] = data[(long)e * LPB_S + line];
+
+  const long nwg = (long)ngroups * nchunks;
+  for (long wg = blockIdx.x; wg < nwg; wg += gridDim.x) {
+    const int group = (int)(wg / nchunks);
+    const int x = (int)(wg % nchunks) * LPB_S + line;
+    const bool active = x < nlines;
+    const long base = (long)group * gstride + x;
+    for (int e = tl; e < n; e += TPL_S) {
+      f2 v = {0.0f, 0.0f};
+      if (active && e < valid) {
+        if (in2) {
+          f2 q = conjmul(in[base + e * estride], in2[base + e * estride]);
+          float m2 = q.x * q.x + q.y * q.y;
+          if (m2 > 1e-40f) {
+            float s = scale / sqrtf(m2);
+            v = {q.x * s, q.y * s};
+          }
+        } else {
+          v = in[base + e * estride];
+        }
+      }
+      data[(long)brev_n(e, log2n) * LPB_S + line] = v;
+    }
+    __syncthreads();
+    fft_lds<LPB_S, TPL_S>(data, (long)line, n, log2n, tl, tw, dir);
+    if (active)
+      for (int e = tl; e < n; e += TPL_S)
+        out[base + e * estride] = data[(long)e * LPB_S + line];
+    __syncthreads(); /* LDS reused next group */
+  }
 }
 
 /* Inverse x pass: Hermitian half-line -> full line (per-line mirror),
@@ -302,52 +311,57 @@ __global__ __launch_bounds__(256) void k_peak_tile(
   __shared__ float wv[4][5];
   __shared__ long long wi[4][5];
   const int tid = threadIdx.x;
-  int ntx = (px + PK_TX - 1) / PK_TX;
-  int nty = (py + PK_TY - 1) / PK_TY;
-  int bx = blockIdx.x % ntx;
-  int by = (blockIdx.x / ntx) % nty;
-  int bz = blockIdx.x / (ntx * nty);
-  int x0 = bx * PK_TX, y0 = by * PK_TY, z0 = bz * PK_TZ;
+  const int ntx = (px + PK_TX - 1) / PK_TX;
+  const int nty = (py + PK_TY - 1) / PK_TY;
+  const int ntz = (pz + PK_TZ - 1) / PK_TZ;
   const int HX = PK_TX + 2, HY = PK_TY + 2, HZ = PK_TZ + 2;
-  const bool interior = x0 > 0 && y0 > 0 && z0 > 0 && x0 + PK_TX < px &&
-                        y0 + PK_TY < py && z0 + PK_TZ < pz;
-  if (interior) {
-    const float *base =
-        pcm + ((long)(z0 - 1) * py + (y0 - 1)) * px + (x0 - 1);
-    for (int i = tid; i < HX * HY * HZ; i += 256) {
-      int lx = i % HX, t = i / HX, ly = t % HY, lz = t / HY;
-      tile[i] = base[((long)lz * py + ly) * px + lx];
-    }
-  } else {
-    for (int i = tid; i < HX * HY * HZ; i += 256) {
-      int lx = i % HX, t = i / HX, ly = t % HY, lz = t / HY;
-      int gx = (x0 + lx - 1 + px) % px;
-      int gy = (y0 + ly - 1 + py) % py;
-      int gz = (z0 + lz - 1 + pz) % pz;
-      tile[i] = pcm[((long)gz * py + gy) * px + gx];
-    }
-  }
-  __syncthreads();
+  const long ntiles = (long)ntx * nty * ntz;
   float tv[5];
   long long ti[5];
   for (int k = 0; k < 5; ++k) { tv[k] = -3.0e38f; ti[k] = 0x7fffffffffffffffLL; }
-  for (int i = tid; i < PK_TX * PK_TY * PK_TZ; i += 256) {
-    int lx = i % PK_TX, t = i / PK_TX, ly = t % PK_TY, lz = t / PK_TY;
-    int gx = x0 + lx, gy = y0 + ly, gz = z0 + lz;
-    if (gx >= px || gy >= py || gz >= pz) continue;
-    float v = tile[((lz + 1) * HY + ly + 1) * HX + lx + 1];
-    bool ismax = true;
-    for (int dz = 0; dz <= 2 && ismax; ++dz)
-      for (int dy = 0; dy <= 2 && ismax; ++dy)
-        for (int dx = 0; dx <= 2; ++dx) {
-          if (dz == 1 && dy == 1 && dx == 1) continue;
-          if (!(v > tile[((lz + dz) * HY + ly + dy) * HX + lx + dx])) {
-            ismax = false;
-            break;
+  for (long t0 = blockIdx.x; t0 < ntiles; t0 += gridDim.x) {
+    int bx = (int)(t0 % ntx);
+    int by = (int)((t0 / ntx) % nty);
+    int bz = (int)(t0 / ((long)ntx * nty));
+    int x0 = bx * PK_TX, y0 = by * PK_TY, z0 = bz * PK_TZ;
+    const bool interior = x0 > 0 && y0 > 0 && z0 > 0 && x0 + PK_TX < px &&
+                          y0 + PK_TY < py && z0 + PK_TZ < pz;
+    if (interior) {
+      const float *base =
+          pcm + ((long)(z0 - 1) * py + (y0 - 1)) * px + (x0 - 1);
+      for (int i = tid; i < HX * HY * HZ; i += 256) {
+        int lx = i % HX, t = i / HX, ly = t % HY, lz = t / HY;
+        tile[i] = base[((long)lz * py + ly) * px + lx];
+      }
+    } else {
+      for (int i = tid; i < HX * HY * HZ; i += 256) {
+        int lx = i % HX, t = i / HX, ly = t % HY, lz = t / HY;
+        int gx = (x0 + lx - 1 + px) % px;
+        int gy = (y0 + ly - 1 + py) % py;
+        int gz = (z0 + lz - 1 + pz) % pz;
+        tile[i] = pcm[((long)gz * py + gy) * px + gx];
+      }
+    }
+    __syncthreads();
+    for (int i = tid; i < PK_TX * PK_TY * PK_TZ; i += 256) {
+      int lx = i % PK_TX, t = i / PK_TX, ly = t % PK_TY, lz = t / PK_TY;
+      int gx = x0 + lx, gy = y0 + ly, gz = z0 + lz;
+      if (gx >= px || gy >= py || gz >= pz) continue;
+      float v = tile[((lz + 1) * HY + ly + 1) * HX + lx + 1];
+      bool ismax = true;
+      for (int dz = 0; dz <= 2 && ismax; ++dz)
+        for (int dy = 0; dy <= 2 && ismax; ++dy)
+          for (int dx = 0; dx <= 2; ++dx) {
+            if (dz == 1 && dy == 1 && dx == 1) continue;
+            if (!(v > tile[((lz + dz) * HY + ly + dy) * HX + lx + dx])) {
+              ismax = false;
+              break;
+            }
           }
-        }
-    if (ismax)
-      pk_insert(tv, ti, v, ((long long)gz * py + gy) * px + gx);
+      if (ismax)
+        pk_insert(tv, ti, v, ((long long)gz * py + gy) * px + gx);
+    }
+    __syncthreads(); /* tile LDS reused next iteration */
   }
   pk_merge_shfl(tv, ti);
   int lane = tid & 63, wave = tid >> 6;
@@ -430,18 +444,19 @@ __global__ __launch_bounds__(256) void k_rtest(
     bs_region a, bs_region b, const bs_cand *cands, u64 *sums /* [nc][5] */) {
   __shared__ u64 ws[4][5];
   const bs_cand c = cands[blockIdx.y];
-  long n = (long)c.nx * c.ny * c.nz;
+  long nrows = (long)c.ny * c.nz;
   u64 pa = 0, pb = 0, paa = 0, pbb = 0, pab = 0;
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += (long)gridDim.x * blockDim.x) {
-    int x = (int)(i % c.nx);
-    long t = i / c.nx;
-    int y = (int)(t % c.ny), z = (int)(t / c.ny);
-    u64 av = a.ptr[(a.oz + c.loz + z) * a.sxy + (a.oy + c.loy + y) * a.sx +
-                   a.ox + c.lox + x];
-    u64 bv = b.ptr[(b.oz + c.loz + c.sz + z) * b.sxy +
-                   (b.oy + c.loy + c.sy + y) * b.sx + b.ox + c.lox + c.sx + x];
-    pa += av; pb += bv; paa += av * av; pbb += bv * bv; pab += av * bv;
+  for (long row = blockIdx.x; row < nrows; row += gridDim.x) {
+    int y = (int)(row % c.ny), z = (int)(row / c.ny);
+    const unsigned short *ar = a.ptr + (a.oz + c.loz + z) * a.sxy +
+                               (a.oy + c.loy + y) * a.sx + a.ox + c.lox;
+    const unsigned short *br = b.ptr + (b.oz + c.loz + c.sz + z) * b.sxy +
+                               (b.oy + c.loy + c.sy + y) * b.sx + b.ox +
+                               c.lox + c.sx;
+    for (int x = threadIdx.x; x < c.nx; x += 256) {
+      u64 av = ar[x], bv = br[x];
+      pa += av; pb += bv; paa += av * av; pbb += bv * bv; pab += av * bv;
+    }
   }
   pa = wave_sum_u64(pa); pb = wave_sum_u64(pb); paa = wave_sum_u64(paa);
   pbb = wave_sum_u64(pbb); pab = wave_sum_u64(pab);
@@ -492,12 +507,11 @@ __global__ __launch_bounds__(256) void k_fuse(
         [i % (sizeof(bs_dev_view) / 4)];
   }
   __syncthreads();
-  long n = (long)bx * by * bz;
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += (long)gridDim.x * blockDim.x) {
-    int x = (int)(i % bx);
-    long t = i / bx;
-    int y = (int)(t % by), z = (int)(t / by);
+  long nrows = (long)by * bz;
+  for (long row = blockIdx.x; row < nrows; row += gridDim.x) {
+   int y = (int)(row % by), z = (int)(row / by);
+   for (int x = threadIdx.x; x < bx; x += blockDim.x) {
+    long i = row * bx + x;
     float wx = (float)(bmx + x), wy = (float)(bmy + y), wz = (float)(bmz + z);
     float sum_wv = 0.0f, sum_w = 0.0f, vmax = 0.0f;
     bool any = false;
@@ -560,6 +574,7 @@ __global__ __launch_bounds__(256) void k_fuse(
         ((unsigned char *)out)[i] =
             (unsigned char)fminf(fmaxf(q, 0.0f), 255.0f);
     }
+   }
   }
 }
 
@@ -990,10 +1005,9 @@ extern "C" int bs_stitch_batch(bs_ctx *c, const bs_pair_desc *pairs, size_t np,
         if (rc) return rc;
         bs_region src = {vr.dptr, vr.dims[0], vr.dims[0] * vr.dims[1],
                          off[0], off[1], off[2], 0, 0, 0};
-        long nvox = (long)m[t][0] * m[t][1] * m[t][2];
+        long nrows = (long)m[t][1] * m[t][2];
         bs_tim tt(c, BS_K_DOWNSAMPLE);
-        hipLaunchKernelGGL(k_downsample,
-                           dim3(std::min(8192L, (nvox + 255) / 256)),
+        hipLaunchKernelGGL(k_downsample, dim3(std::min(4096L, nrows)),
                            dim3(256), 0, c->stream, src, c->regbuf[t],
                            m[t][0], m[t][1], m[t][2], ds[0], ds[1], ds[2]);
         reg[t] = {c->regbuf[t], m[t][0], (long)m[t][0] * m[t][1], 0, 0, 0,
@@ -1027,9 +1041,10 @@ extern "C" int bs_stitch_batch(bs_ctx *c, const bs_pair_desc *pairs, size_t np,
     /* forward x (R2C) per tile */
     for (int t = 0; t < 2; ++t) {
       long nlines = (long)reg[t].my * reg[t].mz;
+      long ngrp = (nlines + LPB_X - 1) / LPB_X;
       size_t lds = ((Px / 2) + (size_t)LPB_X * Px) * sizeof(f2);
       bs_tim tt(c, BS_K_FFT_X_FWD);
-      hipLaunchKernelGGL(k_fft_x_fwd, dim3((nlines + LPB_X - 1) / LPB_X),
+      hipLaunchKernelGGL(k_fft_x_fwd, dim3(std::min(4096L, ngrp)),
                          dim3(LPB_X * TPL_X), lds, c->stream, reg[t], spec[t],
                          Px, ilog2(Px), Cx, Cxp, Py, twx);
     }
@@ -1038,63 +1053,71 @@ extern "C" int bs_stitch_batch(bs_ctx *c, const bs_pair_desc *pairs, size_t np,
     for (int t = 0; t < 2; ++t) {
       size_t lds = ((Py / 2) + (size_t)LPB_S * Py) * sizeof(f2);
       bs_tim tt(c, BS_K_FFT_Y_FWD);
-      hipLaunchKernelGGL(k_fft_pass, dim3(reg[t].mz * nchunks),
+      hipLaunchKernelGGL(k_fft_pass,
+                         dim3(std::min(4096L, (long)reg[t].mz * nchunks)),
                          dim3(LPB_S * TPL_S), lds, c->stream, spec[t],
                          (const f2 *)nullptr, spec[t], Py, ilog2(Py), Cxp,
-                         (long)Py * Cxp, Cx, nchunks, reg[t].my, +1, 1.0f,
-                         twy);
+                         (long)Py * Cxp, Cx, nchunks, reg[t].mz, reg[t].my,
+                         +1, 1.0f, twy);
     }
     /* forward z: groups = y rows (all Py), valid = mz */
     for (int t = 0; t < 2; ++t) {
       size_t lds = ((Pz / 2) + (size_t)LPB_S * Pz) * sizeof(f2);
       bs_tim tt(c, BS_K_FFT_Z_FWD);
-      hipLaunchKernelGGL(k_fft_pass, dim3(Py * nchunks), dim3(LPB_S * TPL_S),
-                         lds, c->stream, spec[t], (const f2 *)nullptr,
-                         spec[t], Pz, ilog2(Pz), (long)Py * Cxp, Cxp, Cx,
-                         nchunks, reg[t].mz, +1, 1.0f, twz);
+      hipLaunchKernelGGL(k_fft_pass,
+                         dim3(std::min(4096L, (long)Py * nchunks)),
+                         dim3(LPB_S * TPL_S), lds, c->stream, spec[t],
+                         (const f2 *)nullptr, spec[t], Pz, ilog2(Pz),
+                         (long)Py * Cxp, Cxp, Cx, nchunks, Py, reg[t].mz,
+                         +1, 1.0f, twz);
     }
     /* inverse z fused with cross-power normalise [PIN-EPS] */
     {
       float scale = 1.0f / ((float)Px * (float)Py * (float)Pz);
       size_t lds = ((Pz / 2) + (size_t)LPB_S * Pz) * sizeof(f2);
       bs_tim tt(c, BS_K_FFT_Z_INV);
-      hipLaunchKernelGGL(k_fft_pass, dim3(Py * nchunks), dim3(LPB_S * TPL_S),
-                         lds, c->stream, spec[0], spec[1], spec[0], Pz,
-                         ilog2(Pz), (long)Py * Cxp, Cxp, Cx, nchunks, Pz, -1,
-                         scale, twz);
+      hipLaunchKernelGGL(k_fft_pass,
+                         dim3(std::min(4096L, (long)Py * nchunks)),
+                         dim3(LPB_S * TPL_S), lds, c->stream, spec[0],
+                         spec[1], spec[0], Pz, ilog2(Pz), (long)Py * Cxp,
+                         Cxp, Cx, nchunks, Py, Pz, -1, scale, twz);
     }
     /* inverse y */
     {
       size_t lds = ((Py / 2) + (size_t)LPB_S * Py) * sizeof(f2);
       bs_tim tt(c, BS_K_FFT_Y_INV);
-      hipLaunchKernelGGL(k_fft_pass, dim3(Pz * nchunks), dim3(LPB_S * TPL_S),
-                         lds, c->stream, spec[0], (const f2 *)nullptr,
-                         spec[0], Py, ilog2(Py), Cxp, (long)Py * Cxp, Cx,
-                         nchunks, Py, -1, 1.0f, twy);
+      hipLaunchKernelGGL(k_fft_pass,
+                         dim3(std::min(4096L, (long)Pz * nchunks)),
+                         dim3(LPB_S * TPL_S), lds, c->stream, spec[0],
+                         (const f2 *)nullptr, spec[0], Py, ilog2(Py), Cxp,
+                         (long)Py * Cxp, Cx, nchunks, Pz, Py, -1, 1.0f,
+                         twy);
     }
     /* inverse x (C2R) -> PCM */
     {
       long nlines = (long)Pz * Py;
+      long ngrp = (nlines + LPB_X - 1) / LPB_X;
       size_t lds = ((Px / 2) + (size_t)LPB_X * Px) * sizeof(f2);
       bs_tim tt(c, BS_K_FFT_X_INV);
-      hipLaunchKernelGGL(k_fft_x_inv, dim3((nlines + LPB_X - 1) / LPB_X),
+      hipLaunchKernelGGL(k_fft_x_inv, dim3(std::min(4096L, ngrp)),
                          dim3(LPB_X * TPL_X), lds, c->stream, spec[0], c->pcm,
                          Px, ilog2(Px), Cx, Cxp, nlines, twx);
     }
     /* peak scan [PIN-MAX] */
     long ntiles = (long)((Px + PK_TX - 1) / PK_TX) *
                   ((Py + PK_TY - 1) / PK_TY) * ((Pz + PK_TZ - 1) / PK_TZ);
+    long npkwg = std::min(2048L, ntiles);
     rc = ensure_dev(c, (void **)&c->wgpk, &c->wgpk_cap,
-                    (size_t)ntiles * 5 * sizeof(bs_peak));
+                    (size_t)npkwg * 5 * sizeof(bs_peak));
     if (rc) return rc;
     {
       bs_tim tt(c, BS_K_PEAK);
-      hipLaunchKernelGGL(k_peak_tile, dim3(ntiles), dim3(256), 0, c->stream,
+      hipLaunchKernelGGL(k_peak_tile, dim3(npkwg), dim3(256), 0, c->stream,
                          c->pcm, Px, Py, Pz, c->wgpk);
     }
     {
       bs_tim tt(c, BS_K_PEAK_MERGE);
-      long nent = ntiles * 5;
+      long nent = npkwg * 5;
       int nb1 = (int)std::min(64L, (nent + 1279) / 1280);
       hipLaunchKernelGGL(k_peak_merge, dim3(nb1), dim3(256), 0, c->stream,
                          c->wgpk, nent, c->dmerge);
@@ -1161,9 +1184,15 @@ extern "C" int bs_stitch_batch(bs_ctx *c, const bs_pair_desc *pairs, size_t np,
       CHK(c, hipMemsetAsync(c->dsums, 0, gc.size() * 5 * sizeof(u64),
                             c->stream));
       {
+        long maxrows = 1;
+        for (auto &h : hc)
+          maxrows = std::max(maxrows, (long)h.gc.ny * h.gc.nz);
         bs_tim tt(c, BS_K_CORR);
-        hipLaunchKernelGGL(k_rtest, dim3(1024, (int)gc.size()), dim3(256), 0,
-                           c->stream, reg[0], reg[1], c->dcands, c->dsums);
+        hipLaunchKernelGGL(k_rtest,
+                           dim3((unsigned)std::min(2048L, maxrows),
+                                (unsigned)gc.size()),
+                           dim3(256), 0, c->stream, reg[0], reg[1],
+                           c->dcands, c->dsums);
       }
       CHK(c, hipMemcpyAsync(c->hsums, c->dsums, gc.size() * 5 * sizeof(u64),
                             hipMemcpyDeviceToHost, c->stream));
@@ -1319,8 +1348,9 @@ extern "C" int bs_fuse_blocks(bs_ctx *c, const bs_fuse_view *views,
     long nvox = (long)bd.size[0] * bd.size[1] * bd.size[2];
     int nvb = (int)(view_idx_offsets[i + 1] - view_idx_offsets[i]);
     {
+      long nrows_f = (long)bd.size[1] * bd.size[2];
       bs_tim tt(c, BS_K_FUSE);
-      hipLaunchKernelGGL(k_fuse, dim3(std::min(16384L, (nvox + 255) / 256)),
+      hipLaunchKernelGGL(k_fuse, dim3(std::min(4096L, nrows_f)),
                          dim3(256), 0, c->stream, c->dviews,
                          c->dvidx + view_idx_offsets[i], nvb, bd.min[0],
                          bd.min[1], bd.min[2], (int)bd.size[0],
