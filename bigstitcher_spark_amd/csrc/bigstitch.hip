@@ -1127,6 +1127,14 @@ extern "C" int bs_stitch_batch(bs_ctx *c, const bs_pair_desc *pairs, size_t np,
     CHK(c, hipMemcpyAsync(c->htop5, c->dtop5, 5 * sizeof(bs_peak),
                           hipMemcpyDeviceToHost, c->stream));
     CHK(c, hipStreamSynchronize(c->stream));
+    if (getenv("BS_DEBUG_PEAKS")) {
+      for (int k = 0; k < 5; ++k) {
+        long long idx = c->htop5[k].idx;
+        fprintf(stderr, "[bs] pair %zu peak%d v=%.6g zyx=(%lld,%lld,%lld)\n",
+                ip, k, c->htop5[k].v, idx / ((long)Px * Py),
+                (idx / Px) % Py, idx % Px);
+      }
+    }
 
     /* host: candidates [PIN-CAND] */
     struct HostCand {
