@@ -671,6 +671,7 @@ struct bs_ctx {
   bs_peak *htop5 = nullptr; /* pinned */
   u64 *hsums = nullptr;
   float *hsubpix = nullptr;
+  long dbg_px = 0, dbg_py = 0, dbg_pz = 0; /* last pair's PCM dims */
   /* stats */
   bs_batch_stats stats{};
   std::vector<bs_ev> evs;
@@ -1103,6 +1104,9 @@ extern "C" int bs_stitch_batch(bs_ctx *c, const bs_pair_desc *pairs, size_t np,
                          dim3(LPB_X * TPL_X), lds, c->stream, spec[0], c->pcm,
                          Px, ilog2(Px), Cx, Cxp, nlines, twx);
     }
+    c->dbg_px = Px;
+    c->dbg_py = Py;
+    c->dbg_pz = Pz;
     /* peak scan [PIN-MAX] */
     long ntiles = (long)((Px + PK_TX - 1) / PK_TX) *
                   ((Py + PK_TY - 1) / PK_TY) * ((Pz + PK_TZ - 1) / PK_TZ);
@@ -1260,6 +1264,22 @@ extern "C" int bs_stitch_batch(bs_ctx *c, const bs_pair_desc *pairs, size_t np,
   c->stats.batch_ms = bms;
   c->stats.pairs += (long long)np;
   flush_stats(c);
+  return BS_OK;
+}
+
+/* Debug-only: download the PCM of the LAST pair processed by
+ * bs_stitch_batch (dims out_dims = {Px, Py, Pz}). Not part of the
+ * product surface. */
+extern "C" int bs_debug_pcm(bs_ctx *c, float *out, int64_t out_dims[3]) {
+  if (!c || !out) return BS_EINVAL;
+  std::lock_guard<std::mutex> g(c->mu);
+  CHK(c, hipSetDevice(c->dev));
+  out_dims[0] = c->dbg_px;
+  out_dims[1] = c->dbg_py;
+  out_dims[2] = c->dbg_pz;
+  size_t n = (size_t)c->dbg_px * c->dbg_py * c->dbg_pz;
+  if (!n) return BS_EINVAL;
+  CHK(c, hipMemcpy(out, c->pcm, n * 4, hipMemcpyDeviceToHost));
   return BS_OK;
 }
 
