@@ -246,24 +246,29 @@ __global__ __launch_bounds__(LPB_X *TPL_X) void k_fft_x_inv(
   const int tid = threadIdx.x;
   const int tl = tid % TPL_X, line = tid / TPL_X;
   for (int i = tid; i < (n >> 1); i += LPB_X * TPL_X) tw[i] = twg[i];
-
-  long lid = (long)blockIdx.x * LPB_X + line;
-  bool active = lid < nlines;
-  const f2 *src = in + lid * cxp;
-  f2 *ld = data + (long)line * n;
-  for (int e = tl; e < n; e += TPL_X) {
-    f2 v = {0.0f, 0.0f};
-    if (active) {
-      if (e < cx) v = src[e];
-      else { v = src[n - e]; v.y = -v.y; }
-    }
-    ld[brev_n(e, log2n)] = v;
-  }
   __syncthreads();
-  fft_lds<1, TPL_X>(data, (long)line * n, n, log2n, tl, tw, -1);
-  if (active) {
-    float *o = out + lid * n;
-    for (int k = tl; k < n; k += TPL_X) o[k] = ld[k].x;
+
+  long ngroups = (nlines + LPB_X - 1) / LPB_X;
+  f2 *ld = data + (long)line * n;
+  for (long grp = blockIdx.x; grp < ngroups; grp += gridDim.x) {
+    long lid = grp * LPB_X + line;
+    bool active = lid < nlines;
+    const f2 *src = in + lid * cxp;
+    for (int e = tl; e < n; e += TPL_X) {
+      f2 v = {0.0f, 0.0f};
+      if (active) {
+        if (e < cx) v = src[e];
+        else { v = src[n - e]; v.y = -v.y; }
+      }
+      ld[brev_n(e, log2n)] = v;
+    }
+    __syncthreads();
+    fft_lds<1, TPL_X>(data, (long)line * n, n, log2n, tl, tw, -1);
+    if (active) {
+      float *o = out + lid * n;
+      for (int k = tl; k < n; k += TPL_X) o[k] = ld[k].x;
+    }
+    __syncthreads(); /* LDS reused next group */
   }
 }
 

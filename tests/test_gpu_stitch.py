@@ -123,6 +123,34 @@ def test_stitch_no_subpixel_integer(ctx):
     assert got["r"] == pytest.approx(ref["r"], abs=1e-12)
 
 
+def test_stitch_512_corner_peak(ctx):
+    """Full bench-scale pair whose PCM peak lands in the far-corner tile
+    (negative y/z shifts wrap to the last rows/slices). Regression for a
+    grid-stride bug where lines beyond the first grid sweep were never
+    computed (PCM zero for z>=32)."""
+    from oracle import synth as osynth
+
+    size = 512
+    shape = (size, size, size)
+    true_shift = (460.4, -5.5, -4.25)
+    ba, bb = osynth.pair_blobs_union(shape, true_shift, seed=77)
+    ctx.synth(50, shape, ba, noise_seed=3)
+    ctx.synth(51, shape, bb, noise_seed=4)
+    a = ctx.download(50, shape)
+    b = ctx.download(51, shape)
+    ref = phasecorr.phase_correlation_shift(
+        a, b, ds=(1, 1, 1), min_overlap_ratio=0.05, workers=-1
+    )
+    pair = dict(view_a=50, view_b=51, off_a=(0, 0, 0), size_a=shape,
+                off_b=(0, 0, 0), size_b=shape)
+    got = ctx.stitch_batch([pair], ds=(1, 1, 1), min_overlap_ratio=0.05)[0]
+    assert got["valid"] and ref["valid"]
+    assert np.all(np.abs(got["shift"] - ref["shift"]) < 1e-3), (
+        got["shift"], ref["shift"])
+    assert got["r"] == pytest.approx(ref["r"], abs=1e-9)
+    assert np.all(np.abs(got["shift"] - np.array(true_shift)) < 0.75)
+
+
 def test_synth_views_stitchable(ctx):
     """Device-side synth (bench input path): render a pair on the GPU,
     download, and check the GPU pipeline and oracle agree on it."""
