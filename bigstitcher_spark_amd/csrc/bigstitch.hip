@@ -351,19 +351,21 @@ __global__ __launch_bounds__(256) void k_peak_tile(
     for (int i = tid; i < PK_TX * PK_TY * PK_TZ; i += 256) {
       int lx = i % PK_TX, t = i / PK_TX, ly = t % PK_TY, lz = t / PK_TY;
       int gx = x0 + lx, gy = y0 + ly, gz = z0 + lz;
-      if (gx >= px || gy >= py || gz >= pz) continue;
-      float v = tile[((lz + 1) * HY + ly + 1) * HX + lx + 1];
-      bool ismax = true;
-      for (int dz = 0; dz <= 2 && ismax; ++dz)
-        for (int dy = 0; dy <= 2 && ismax; ++dy)
-          for (int dx = 0; dx <= 2; ++dx) {
-            if (dz == 1 && dy == 1 && dx == 1) continue;
-            if (!(v > tile[((lz + dz) * HY + ly + dy) * HX + lx + dx])) {
-              ismax = false;
-              break;
-            }
-          }
-      if (ismax)
+      bool inb = gx < px && gy < py && gz < pz;
+      /* branchless 26-neighbor max: every LDS offset is base + constant
+       * (the #pragma unroll makes dz/dy compile-time), no divergence */
+      const int base = ((lz + 1) * HY + ly + 1) * HX + lx + 1;
+      float v = tile[base];
+      float m = fmaxf(tile[base - 1], tile[base + 1]);
+#pragma unroll
+      for (int dz = 0; dz <= 2; ++dz)
+#pragma unroll
+        for (int dy = 0; dy <= 2; ++dy) {
+          if (dz == 1 && dy == 1) continue;
+          const int b2 = base + (dz - 1) * HY * HX + (dy - 1) * HX;
+          m = fmaxf(m, fmaxf(fmaxf(tile[b2 - 1], tile[b2]), tile[b2 + 1]));
+        }
+      if (inb && v > m)
         pk_insert(tv, ti, v, ((long long)gz * py + gy) * px + gx);
     }
     __syncthreads(); /* tile LDS reused next iteration */
