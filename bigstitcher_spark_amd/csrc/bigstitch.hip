@@ -148,23 +148,27 @@ __global__ __launch_bounds__(256) void k_downsample(
 
 /* ---------------------------------------------------------- FFT passes */
 
-/* Contiguous-line forward pass: u16 region lines (x) -> half spectrum.
- * One workgroup = LPB_X lines x TPL_X threads. Dynamic LDS:
- * [ tw: n/2 f2 | data: LPB_X * n f2 ]. */
+/* Contiguous-line forward pass: u16 region lines (x) -> half spectrum,
+ * via the packed-real trick: the n-point R2C is an (n/2)-point complex
+ * FFT of z[j] = x[2j] + i x[2j+1] plus an unpack (halves LDS + compute).
+ * One workgroup = LPB_X lines x TPL_X threads, grid-strided over line
+ * groups. Dynamic LDS: [ tw_h: h/2 f2 | data: LPB_X * h f2 ], h = n/2;
+ * tw_h[k] = twg[2k]; the unpack reads twg (size-n table) from global. */
 __global__ __launch_bounds__(LPB_X *TPL_X) void k_fft_x_fwd(
     bs_region in, f2 *out, int n, int log2n, int cx, long cxp, int py,
     const f2 *twg) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int h = n >> 1, log2h = log2n - 1;
   f2 *tw = (f2 *)smem;
-  f2 *data = tw + (n >> 1);
+  f2 *data = tw + (h >> 1);
   const int tid = threadIdx.x;
   const int tl = tid % TPL_X, line = tid / TPL_X;
-  for (int i = tid; i < (n >> 1); i += LPB_X * TPL_X) tw[i] = twg[i];
+  for (int i = tid; i < (h >> 1); i += LPB_X * TPL_X) tw[i] = twg[2 * i];
   __syncthreads();
 
   long nlines = (long)in.my * in.mz;
   long ngroups = (nlines + LPB_X - 1) / LPB_X;
-  f2 *ld = data + (long)line * n;
+  f2 *ld = data + (long)line * h;
   for (long grp = blockIdx.x; grp < ngroups; grp += gridDim.x) {
     long lid = grp * LPB_X + line;
     bool active = lid < nlines;
@@ -172,15 +176,29 @@ __global__ __launch_bounds__(LPB_X *TPL_X) void k_fft_x_fwd(
     int z = active ? (int)(lid / in.my) : 0;
     const unsigned short *src =
         in.ptr + (in.oz + z) * in.sxy + (in.oy + y) * in.sx + in.ox;
-    for (int j = tl; j < n; j += TPL_X) {
-      float v = (active && j < in.mx) ? (float)src[j] : 0.0f;
-      ld[brev_n(j, log2n)] = {v, 0.0f};
+    for (int j = tl; j < h; j += TPL_X) {
+      float xa = (active && 2 * j < in.mx) ? (float)src[2 * j] : 0.0f;
+      float xb = (active && 2 * j + 1 < in.mx) ? (float)src[2 * j + 1] : 0.0f;
+      ld[brev_n(j, log2h)] = {xa, xb};
     }
     __syncthreads();
-    fft_lds<1, TPL_X>(data, (long)line * n, n, log2n, tl, tw, +1);
+    fft_lds<1, TPL_X>(data, (long)line * h, h, log2h, tl, tw, +1);
     if (active) {
       f2 *o = out + ((long)z * py + y) * cxp; /* rows are Py-strided */
-      for (int k = tl; k < cx; k += TPL_X) o[k] = ld[k];
+      for (int k = tl; k < h; k += TPL_X) {
+        if (k == 0) {
+          f2 z0 = ld[0];
+          o[0] = {z0.x + z0.y, 0.0f};
+          o[h] = {z0.x - z0.y, 0.0f};
+        } else {
+          f2 zk = ld[k], zm = ld[h - k];
+          f2 ze = {0.5f * (zk.x + zm.x), 0.5f * (zk.y - zm.y)};
+          f2 dd = {zk.x - zm.x, zk.y + zm.y};   /* Zk - conj(Zmk) */
+          f2 zo = {0.5f * dd.y, -0.5f * dd.x};  /* -i/2 * dd */
+          f2 wzo = cmul(twg[k], zo);
+          o[k] = {ze.x + wzo.x, ze.y + wzo.y};
+        }
+      }
     }
     __syncthreads(); /* LDS reused next group */
   }
@@ -235,38 +253,48 @@ __global__ __launch_bounds__(LPB_S *TPL_S) void k_fft_pass(
   }
 }
 
-/* Inverse x pass: Hermitian half-line -> full line (per-line mirror),
- * inverse FFT, write real part (the PCM). */
+/* Inverse x pass (C2R, packed): per-line Hermitian half-line -> the real
+ * PCM line via an (n/2)-point inverse complex FFT. Build
+ * Z2[k] = A + i*conj(W^k)*B with A = X[k]+conj(X[h-k]),
+ * B = X[k]-conj(X[h-k]) (the dropped 1/2 makes the result exactly the
+ * UNNORMALIZED length-n inverse, matching the other passes); after the
+ * FFT, z[j] = (pcm[2j], pcm[2j+1]) -> vectorized float2 row writes. */
 __global__ __launch_bounds__(LPB_X *TPL_X) void k_fft_x_inv(
     const f2 *in, float *out, int n, int log2n, int cx, long cxp,
     long nlines, const f2 *twg) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int h = n >> 1, log2h = log2n - 1;
   f2 *tw = (f2 *)smem;
-  f2 *data = tw + (n >> 1);
+  f2 *data = tw + (h >> 1);
   const int tid = threadIdx.x;
   const int tl = tid % TPL_X, line = tid / TPL_X;
-  for (int i = tid; i < (n >> 1); i += LPB_X * TPL_X) tw[i] = twg[i];
+  for (int i = tid; i < (h >> 1); i += LPB_X * TPL_X) tw[i] = twg[2 * i];
   __syncthreads();
 
   long ngroups = (nlines + LPB_X - 1) / LPB_X;
-  f2 *ld = data + (long)line * n;
+  f2 *ld = data + (long)line * h;
   for (long grp = blockIdx.x; grp < ngroups; grp += gridDim.x) {
     long lid = grp * LPB_X + line;
     bool active = lid < nlines;
     const f2 *src = in + lid * cxp;
-    for (int e = tl; e < n; e += TPL_X) {
+    for (int k = tl; k < h; k += TPL_X) {
       f2 v = {0.0f, 0.0f};
       if (active) {
-        if (e < cx) v = src[e];
-        else { v = src[n - e]; v.y = -v.y; }
+        f2 xk = src[k], xm = src[h - k];
+        f2 A = {xk.x + xm.x, xk.y - xm.y};  /* X[k] + conj(X[h-k]) */
+        f2 B = {xk.x - xm.x, xk.y + xm.y};  /* X[k] - conj(X[h-k]) */
+        f2 wc = twg[k];
+        wc.y = -wc.y;
+        f2 wb = cmul(wc, B);
+        v = {A.x - wb.y, A.y + wb.x}; /* A + i*wb */
       }
-      ld[brev_n(e, log2n)] = v;
+      ld[brev_n(k, log2h)] = v;
     }
     __syncthreads();
-    fft_lds<1, TPL_X>(data, (long)line * n, n, log2n, tl, tw, -1);
+    fft_lds<1, TPL_X>(data, (long)line * h, h, log2h, tl, tw, -1);
     if (active) {
-      float *o = out + lid * n;
-      for (int k = tl; k < n; k += TPL_X) o[k] = ld[k].x;
+      f2 *o = (f2 *)(out + lid * n); /* rows are 8B-aligned (n >= 4) */
+      for (int j = tl; j < h; j += TPL_X) o[j] = ld[j];
     }
     __syncthreads(); /* LDS reused next group */
   }
@@ -1050,7 +1078,7 @@ extern "C" int bs_stitch_batch(bs_ctx *c, const bs_pair_desc *pairs, size_t np,
     for (int t = 0; t < 2; ++t) {
       long nlines = (long)reg[t].my * reg[t].mz;
       long ngrp = (nlines + LPB_X - 1) / LPB_X;
-      size_t lds = ((Px / 2) + (size_t)LPB_X * Px) * sizeof(f2);
+      size_t lds = ((Px / 4) + (size_t)LPB_X * (Px / 2)) * sizeof(f2);
       bs_tim tt(c, BS_K_FFT_X_FWD);
       hipLaunchKernelGGL(k_fft_x_fwd, dim3(std::min(4096L, ngrp)),
                          dim3(LPB_X * TPL_X), lds, c->stream, reg[t], spec[t],
@@ -1105,7 +1133,7 @@ extern "C" int bs_stitch_batch(bs_ctx *c, const bs_pair_desc *pairs, size_t np,
     {
       long nlines = (long)Pz * Py;
       long ngrp = (nlines + LPB_X - 1) / LPB_X;
-      size_t lds = ((Px / 2) + (size_t)LPB_X * Px) * sizeof(f2);
+      size_t lds = ((Px / 4) + (size_t)LPB_X * (Px / 2)) * sizeof(f2);
       bs_tim tt(c, BS_K_FFT_X_INV);
       hipLaunchKernelGGL(k_fft_x_inv, dim3(std::min(4096L, ngrp)),
                          dim3(LPB_X * TPL_X), lds, c->stream, spec[0], c->pcm,
