@@ -32,6 +32,7 @@
 
 #include <hip/hip_runtime.h>
 
+#include <chrono>
 #include <cmath>
 #include <cstdio>
 #include <cstdlib>
@@ -481,15 +482,35 @@ __global__ __launch_bounds__(256) void k_rtest(
   const bs_cand c = cands[blockIdx.y];
   long nrows = (long)c.ny * c.nz;
   u64 pa = 0, pb = 0, paa = 0, pbb = 0, pab = 0;
-  for (long row = blockIdx.x; row < nrows; row += gridDim.x) {
+  auto rowptr_a = [&](long row) {
     int y = (int)(row % c.ny), z = (int)(row / c.ny);
-    const unsigned short *ar = a.ptr + (a.oz + c.loz + z) * a.sxy +
-                               (a.oy + c.loy + y) * a.sx + a.ox + c.lox;
-    const unsigned short *br = b.ptr + (b.oz + c.loz + c.sz + z) * b.sxy +
-                               (b.oy + c.loy + c.sy + y) * b.sx + b.ox +
-                               c.lox + c.sx;
+    return a.ptr + (a.oz + c.loz + z) * a.sxy + (a.oy + c.loy + y) * a.sx +
+           a.ox + c.lox;
+  };
+  auto rowptr_b = [&](long row) {
+    int y = (int)(row % c.ny), z = (int)(row / c.ny);
+    return b.ptr + (b.oz + c.loz + c.sz + z) * b.sxy +
+           (b.oy + c.loy + c.sy + y) * b.sx + b.ox + c.lox + c.sx;
+  };
+  long row = blockIdx.x;
+  /* two rows in flight doubles outstanding loads (latency-bound kernel) */
+  for (; row + gridDim.x < nrows; row += 2L * gridDim.x) {
+    const unsigned short *a0 = rowptr_a(row), *b0 = rowptr_b(row);
+    const unsigned short *a1 = rowptr_a(row + gridDim.x);
+    const unsigned short *b1 = rowptr_b(row + gridDim.x);
     for (int x = threadIdx.x; x < c.nx; x += 256) {
-      u64 av = ar[x], bv = br[x];
+      u64 av0 = a0[x], bv0 = b0[x], av1 = a1[x], bv1 = b1[x];
+      pa += av0 + av1;
+      pb += bv0 + bv1;
+      paa += av0 * av0 + av1 * av1;
+      pbb += bv0 * bv0 + bv1 * bv1;
+      pab += av0 * bv0 + av1 * bv1;
+    }
+  }
+  for (; row < nrows; row += gridDim.x) {
+    const unsigned short *a0 = rowptr_a(row), *b0 = rowptr_b(row);
+    for (int x = threadIdx.x; x < c.nx; x += 256) {
+      u64 av = a0[x], bv = b0[x];
       pa += av; pb += bv; paa += av * av; pbb += bv * bv; pab += av * bv;
     }
   }
@@ -671,21 +692,55 @@ struct bs_ev {
   int kid;
 };
 
-struct bs_ctx {
-  int dev;
-  hipStream_t stream;
-  std::string err;
-  std::map<int32_t, bs_view_rec> views;
-  std::map<int, f2 *> twiddles;
-  /* workspace */
+struct bs_hostcand {
+  bs_cand gc;
+  int rank, ci;
+  long n;
+};
+
+/* One pipeline slot: its own stream + workspace so two pairs are in
+ * flight (pair i's FFT/peak chain overlaps pair i-1's r-test and the
+ * host-side candidate building). */
+struct bs_slot {
+  hipStream_t stream = nullptr;
+  hipEvent_t peaks_ready = nullptr, done = nullptr;
   f2 *spec = nullptr;        /* 2 x specsize */
-  size_t spec_cap = 0;       /* in f2 units (per half) */
+  size_t spec_cap = 0;
   float *pcm = nullptr;
   size_t pcm_cap = 0;
   unsigned short *regbuf[2] = {nullptr, nullptr};
   size_t reg_cap[2] = {0, 0};
   bs_peak *wgpk = nullptr;
   size_t wgpk_cap = 0;
+  bs_peak *dmerge = nullptr;
+  bs_peak *dtop5 = nullptr;
+  bs_cand *dcands = nullptr;
+  u64 *dsums = nullptr;
+  long long *dpkidx = nullptr;
+  float *dsubpix = nullptr;
+  bs_peak *htop5 = nullptr; /* pinned */
+  u64 *hsums = nullptr;
+  float *hsubpix = nullptr;
+  /* per-pair host context carried between phases */
+  int Px = 0, Py = 0, Pz = 0, Cx = 0;
+  long Cxp = 0;
+  bs_region reg[2];
+  int m[2][3];
+  int dsf[3];
+  std::vector<bs_hostcand> hc;
+  long long pkidx_h[8];
+  int npk = 0;
+  size_t out_idx = 0;
+  int stage = 0; /* 0 idle, 1 fft+peak issued, 2 rtest issued */
+};
+
+struct bs_ctx {
+  int dev;
+  hipStream_t stream; /* default stream: views, synth, fusion */
+  std::string err;
+  std::map<int32_t, bs_view_rec> views;
+  std::map<int, f2 *> twiddles;
+  bs_slot slot[2];
   float *synth_acc = nullptr;
   size_t synth_cap = 0;
   void *fuse_out = nullptr;
@@ -696,17 +751,8 @@ struct bs_ctx {
   size_t dvidx_cap = 0;
   float *dblobs = nullptr;
   size_t dblobs_cap = 0;
-  /* small device results + pinned mirrors */
-  bs_peak *dmerge = nullptr; /* 64*5 intermediate merge slots */
-  bs_peak *dtop5 = nullptr;
-  bs_cand *dcands = nullptr;
-  u64 *dsums = nullptr;
-  long long *dpkidx = nullptr;
-  float *dsubpix = nullptr;
-  bs_peak *htop5 = nullptr; /* pinned */
-  u64 *hsums = nullptr;
-  float *hsubpix = nullptr;
-  long dbg_px = 0, dbg_py = 0, dbg_pz = 0; /* last pair's PCM dims */
+  float *dbg_pcm = nullptr; /* last pair's PCM (points into a slot) */
+  long dbg_px = 0, dbg_py = 0, dbg_pz = 0;
   /* stats */
   bs_batch_stats stats{};
   std::vector<bs_ev> evs;
@@ -771,16 +817,26 @@ extern "C" int bs_ctx_create(bs_ctx **out, int device_id) {
   (void)hipFuncSetAttribute((const void *)k_fft_x_inv,
                             hipFuncAttributeMaxDynamicSharedMemorySize,
                             64 * 1024);
-  /* small fixed buffers */
-  if (hipMalloc(&c->dmerge, 64 * 5 * sizeof(bs_peak)) != hipSuccess ||
-      hipMalloc(&c->dtop5, 5 * sizeof(bs_peak)) != hipSuccess ||
-      hipMalloc(&c->dcands, 64 * sizeof(bs_cand)) != hipSuccess ||
-      hipMalloc(&c->dsums, 64 * 5 * sizeof(u64)) != hipSuccess ||
-      hipMalloc(&c->dpkidx, 8 * sizeof(long long)) != hipSuccess ||
-      hipMalloc(&c->dsubpix, 8 * 7 * sizeof(float)) != hipSuccess ||
-      hipHostMalloc(&c->htop5, 5 * sizeof(bs_peak)) != hipSuccess ||
-      hipHostMalloc(&c->hsums, 64 * 5 * sizeof(u64)) != hipSuccess ||
-      hipHostMalloc(&c->hsubpix, 8 * 7 * sizeof(float)) != hipSuccess) {
+  /* per-slot streams, events and small fixed buffers */
+  bool ok = true;
+  for (int s = 0; s < 2; ++s) {
+    bs_slot &sl = c->slot[s];
+    ok = ok && hipStreamCreate(&sl.stream) == hipSuccess &&
+         hipEventCreateWithFlags(&sl.peaks_ready, hipEventDisableTiming) ==
+             hipSuccess &&
+         hipEventCreateWithFlags(&sl.done, hipEventDisableTiming) ==
+             hipSuccess &&
+         hipMalloc(&sl.dmerge, 64 * 5 * sizeof(bs_peak)) == hipSuccess &&
+         hipMalloc(&sl.dtop5, 5 * sizeof(bs_peak)) == hipSuccess &&
+         hipMalloc(&sl.dcands, 64 * sizeof(bs_cand)) == hipSuccess &&
+         hipMalloc(&sl.dsums, 64 * 5 * sizeof(u64)) == hipSuccess &&
+         hipMalloc(&sl.dpkidx, 8 * sizeof(long long)) == hipSuccess &&
+         hipMalloc(&sl.dsubpix, 8 * 7 * sizeof(float)) == hipSuccess &&
+         hipHostMalloc(&sl.htop5, 5 * sizeof(bs_peak)) == hipSuccess &&
+         hipHostMalloc(&sl.hsums, 64 * 5 * sizeof(u64)) == hipSuccess &&
+         hipHostMalloc(&sl.hsubpix, 8 * 7 * sizeof(float)) == hipSuccess;
+  }
+  if (!ok) {
     delete c;
     g_err = "alloc failed";
     return BS_ENOMEM;
@@ -799,25 +855,32 @@ extern "C" void bs_ctx_destroy(bs_ctx *c) {
     (void)hipEventDestroy(pr.first);
     (void)hipEventDestroy(pr.second);
   }
-  (void)hipFree(c->spec);
-  (void)hipFree(c->pcm);
-  (void)hipFree(c->regbuf[0]);
-  (void)hipFree(c->regbuf[1]);
-  (void)hipFree(c->wgpk);
+  for (int s = 0; s < 2; ++s) {
+    bs_slot &sl = c->slot[s];
+    if (sl.stream) (void)hipStreamSynchronize(sl.stream);
+    (void)hipFree(sl.spec);
+    (void)hipFree(sl.pcm);
+    (void)hipFree(sl.regbuf[0]);
+    (void)hipFree(sl.regbuf[1]);
+    (void)hipFree(sl.wgpk);
+    (void)hipFree(sl.dmerge);
+    (void)hipFree(sl.dtop5);
+    (void)hipFree(sl.dcands);
+    (void)hipFree(sl.dsums);
+    (void)hipFree(sl.dpkidx);
+    (void)hipFree(sl.dsubpix);
+    (void)hipHostFree(sl.htop5);
+    (void)hipHostFree(sl.hsums);
+    (void)hipHostFree(sl.hsubpix);
+    if (sl.peaks_ready) (void)hipEventDestroy(sl.peaks_ready);
+    if (sl.done) (void)hipEventDestroy(sl.done);
+    if (sl.stream) (void)hipStreamDestroy(sl.stream);
+  }
   (void)hipFree(c->synth_acc);
   (void)hipFree(c->fuse_out);
   (void)hipFree(c->dviews);
   (void)hipFree(c->dvidx);
   (void)hipFree(c->dblobs);
-  (void)hipFree(c->dmerge);
-  (void)hipFree(c->dtop5);
-  (void)hipFree(c->dcands);
-  (void)hipFree(c->dsums);
-  (void)hipFree(c->dpkidx);
-  (void)hipFree(c->dsubpix);
-  (void)hipHostFree(c->htop5);
-  (void)hipHostFree(c->hsums);
-  (void)hipHostFree(c->hsubpix);
   (void)hipStreamDestroy(c->stream);
   delete c;
 }
@@ -826,17 +889,19 @@ extern "C" void bs_ctx_destroy(bs_ctx *c) {
 struct bs_tim {
   bs_ctx *c;
   int kid;
-  bs_tim(bs_ctx *c_, int kid_) : c(c_), kid(kid_) {
+  hipStream_t st;
+  bs_tim(bs_ctx *c_, int kid_, hipStream_t st_ = nullptr)
+      : c(c_), kid(kid_), st(st_ ? st_ : c_->stream) {
     if (c->evused == c->evpool.size()) {
       hipEvent_t a, b;
       (void)hipEventCreate(&a);
       (void)hipEventCreate(&b);
       c->evpool.push_back({a, b});
     }
-    (void)hipEventRecord(c->evpool[c->evused].first, c->stream);
+    (void)hipEventRecord(c->evpool[c->evused].first, st);
   }
   ~bs_tim() {
-    (void)hipEventRecord(c->evpool[c->evused].second, c->stream);
+    (void)hipEventRecord(c->evpool[c->evused].second, st);
     c->evs.push_back({c->evpool[c->evused].first,
                       c->evpool[c->evused].second, kid});
     c->evused++;
@@ -981,325 +1046,341 @@ static int next_pow2(int n) {
 
 /* ---- stitching ---- */
 
+/* ---- stitching pipeline (two slots, see bs_slot) ---- */
+
+/* Phase A: validate + downsample + FFT chain + peak scan for one pair;
+ * records sl->peaks_ready after the top-5 D2H. */
+static int stitch_phaseA(bs_ctx *c, bs_slot *sl, const bs_pair_desc &pd,
+                         const bs_stitch_params *prm, size_t out_idx) {
+  auto ita = c->views.find(pd.view_a);
+  auto itb = c->views.find(pd.view_b);
+  if (ita == c->views.end() || itb == c->views.end()) {
+    c->err = "view not uploaded";
+    return BS_ENOVIEW;
+  }
+  int *ds = sl->dsf;
+  ds[0] = prm->ds[0] > 0 ? prm->ds[0] : 1;
+  ds[1] = prm->ds[1] > 0 ? prm->ds[1] : 1;
+  ds[2] = prm->ds[2] > 0 ? prm->ds[2] : 1;
+  bool dsall1 = ds[0] == 1 && ds[1] == 1 && ds[2] == 1;
+  for (int t = 0; t < 2; ++t) {
+    const bs_view_rec &vr = t == 0 ? ita->second : itb->second;
+    const int64_t *off = t == 0 ? pd.off_a : pd.off_b;
+    const int64_t *size = t == 0 ? pd.size_a : pd.size_b;
+    for (int d = 0; d < 3; ++d) {
+      if (off[d] < 0 || size[d] <= 0 || off[d] + size[d] > vr.dims[d]) {
+        c->err = "pair interval out of view bounds";
+        return BS_EINVAL;
+      }
+      sl->m[t][d] = (int)(size[d] / ds[d]);
+      if (sl->m[t][d] < 1) sl->m[t][d] = 1;
+    }
+    if (dsall1) {
+      sl->reg[t] = {vr.dptr, vr.dims[0], vr.dims[0] * vr.dims[1],
+                    off[0], off[1], off[2],
+                    sl->m[t][0], sl->m[t][1], sl->m[t][2]};
+    } else {
+      size_t need = (size_t)sl->m[t][0] * sl->m[t][1] * sl->m[t][2] * 2;
+      int rc = ensure_dev(c, (void **)&sl->regbuf[t], &sl->reg_cap[t], need);
+      if (rc) return rc;
+      bs_region src = {vr.dptr, vr.dims[0], vr.dims[0] * vr.dims[1],
+                       off[0], off[1], off[2], 0, 0, 0};
+      long nrows = (long)sl->m[t][1] * sl->m[t][2];
+      bs_tim tt(c, BS_K_DOWNSAMPLE, sl->stream);
+      hipLaunchKernelGGL(k_downsample, dim3(std::min(4096L, nrows)),
+                         dim3(256), 0, sl->stream, src, sl->regbuf[t],
+                         sl->m[t][0], sl->m[t][1], sl->m[t][2], ds[0], ds[1],
+                         ds[2]);
+      sl->reg[t] = {sl->regbuf[t], sl->m[t][0],
+                    (long)sl->m[t][0] * sl->m[t][1], 0, 0, 0,
+                    sl->m[t][0], sl->m[t][1], sl->m[t][2]};
+    }
+  }
+  /* padded FFT dims [PIN-PAD] */
+  int Px = next_pow2(std::max(sl->m[0][0], sl->m[1][0]));
+  int Py = next_pow2(std::max(sl->m[0][1], sl->m[1][1]));
+  int Pz = next_pow2(std::max(sl->m[0][2], sl->m[1][2]));
+  if (Px > 1024 || Py > 1024 || Pz > 1024 || Px < 8) {
+    c->err = "FFT size unsupported (need 8..1024 per axis)";
+    return BS_EUNSUP;
+  }
+  int Cx = Px / 2 + 1;
+  long Cxp = (Cx + 15) & ~15L;
+  sl->Px = Px; sl->Py = Py; sl->Pz = Pz; sl->Cx = Cx; sl->Cxp = Cxp;
+  sl->out_idx = out_idx;
+  size_t spec_half = (size_t)Pz * Py * Cxp;
+  int rc = ensure_dev(c, (void **)&sl->spec, &sl->spec_cap,
+                      2 * spec_half * sizeof(f2));
+  if (rc) return rc;
+  size_t pcm_n = (size_t)Pz * Py * Px;
+  rc = ensure_dev(c, (void **)&sl->pcm, &sl->pcm_cap, pcm_n * sizeof(float));
+  if (rc) return rc;
+  f2 *twx = get_twiddle(c, Px), *twy = get_twiddle(c, Py),
+     *twz = get_twiddle(c, Pz);
+  if (!twx || !twy || !twz) {
+    c->err = "twiddle alloc failed";
+    return BS_ENOMEM;
+  }
+  f2 *spec[2] = {sl->spec, sl->spec + spec_half};
+  for (int t = 0; t < 2; ++t) {
+    long nlines = (long)sl->reg[t].my * sl->reg[t].mz;
+    long ngrp = (nlines + LPB_X - 1) / LPB_X;
+    size_t lds = ((Px / 4) + (size_t)LPB_X * (Px / 2)) * sizeof(f2);
+    bs_tim tt(c, BS_K_FFT_X_FWD, sl->stream);
+    hipLaunchKernelGGL(k_fft_x_fwd, dim3(std::min(4096L, ngrp)),
+                       dim3(LPB_X * TPL_X), lds, sl->stream, sl->reg[t],
+                       spec[t], Px, ilog2(Px), Cx, Cxp, Py, twx);
+  }
+  int nchunks = (Cx + LPB_S - 1) / LPB_S;
+  for (int t = 0; t < 2; ++t) {
+    size_t lds = ((Py / 2) + (size_t)LPB_S * Py) * sizeof(f2);
+    bs_tim tt(c, BS_K_FFT_Y_FWD, sl->stream);
+    hipLaunchKernelGGL(k_fft_pass,
+                       dim3(std::min(4096L, (long)sl->reg[t].mz * nchunks)),
+                       dim3(LPB_S * TPL_S), lds, sl->stream, spec[t],
+                       (const f2 *)nullptr, spec[t], Py, ilog2(Py), Cxp,
+                       (long)Py * Cxp, Cx, nchunks, sl->reg[t].mz,
+                       sl->reg[t].my, +1, 1.0f, twy);
+  }
+  for (int t = 0; t < 2; ++t) {
+    size_t lds = ((Pz / 2) + (size_t)LPB_S * Pz) * sizeof(f2);
+    bs_tim tt(c, BS_K_FFT_Z_FWD, sl->stream);
+    hipLaunchKernelGGL(k_fft_pass, dim3(std::min(4096L, (long)Py * nchunks)),
+                       dim3(LPB_S * TPL_S), lds, sl->stream, spec[t],
+                       (const f2 *)nullptr, spec[t], Pz, ilog2(Pz),
+                       (long)Py * Cxp, Cxp, Cx, nchunks, Py, sl->reg[t].mz,
+                       +1, 1.0f, twz);
+  }
+  { /* inverse z fused with cross-power normalise [PIN-EPS] */
+    float scale = 1.0f / ((float)Px * (float)Py * (float)Pz);
+    size_t lds = ((Pz / 2) + (size_t)LPB_S * Pz) * sizeof(f2);
+    bs_tim tt(c, BS_K_FFT_Z_INV, sl->stream);
+    hipLaunchKernelGGL(k_fft_pass, dim3(std::min(4096L, (long)Py * nchunks)),
+                       dim3(LPB_S * TPL_S), lds, sl->stream, spec[0],
+                       spec[1], spec[0], Pz, ilog2(Pz), (long)Py * Cxp, Cxp,
+                       Cx, nchunks, Py, Pz, -1, scale, twz);
+  }
+  {
+    size_t lds = ((Py / 2) + (size_t)LPB_S * Py) * sizeof(f2);
+    bs_tim tt(c, BS_K_FFT_Y_INV, sl->stream);
+    hipLaunchKernelGGL(k_fft_pass, dim3(std::min(4096L, (long)Pz * nchunks)),
+                       dim3(LPB_S * TPL_S), lds, sl->stream, spec[0],
+                       (const f2 *)nullptr, spec[0], Py, ilog2(Py), Cxp,
+                       (long)Py * Cxp, Cx, nchunks, Pz, Py, -1, 1.0f, twy);
+  }
+  {
+    long nlines = (long)Pz * Py;
+    long ngrp = (nlines + LPB_X - 1) / LPB_X;
+    size_t lds = ((Px / 4) + (size_t)LPB_X * (Px / 2)) * sizeof(f2);
+    bs_tim tt(c, BS_K_FFT_X_INV, sl->stream);
+    hipLaunchKernelGGL(k_fft_x_inv, dim3(std::min(4096L, ngrp)),
+                       dim3(LPB_X * TPL_X), lds, sl->stream, spec[0],
+                       sl->pcm, Px, ilog2(Px), Cx, Cxp, nlines, twx);
+  }
+  c->dbg_pcm = sl->pcm;
+  c->dbg_px = Px;
+  c->dbg_py = Py;
+  c->dbg_pz = Pz;
+  /* peak scan [PIN-MAX] */
+  long ntiles = (long)((Px + PK_TX - 1) / PK_TX) *
+                ((Py + PK_TY - 1) / PK_TY) * ((Pz + PK_TZ - 1) / PK_TZ);
+  long npkwg = std::min(2048L, ntiles);
+  int rc2 = ensure_dev(c, (void **)&sl->wgpk, &sl->wgpk_cap,
+                       (size_t)npkwg * 5 * sizeof(bs_peak));
+  if (rc2) return rc2;
+  {
+    bs_tim tt(c, BS_K_PEAK, sl->stream);
+    hipLaunchKernelGGL(k_peak_tile, dim3(npkwg), dim3(256), 0, sl->stream,
+                       sl->pcm, Px, Py, Pz, sl->wgpk);
+  }
+  {
+    bs_tim tt(c, BS_K_PEAK_MERGE, sl->stream);
+    long nent = npkwg * 5;
+    int nb1 = (int)std::min(64L, (nent + 1279) / 1280);
+    hipLaunchKernelGGL(k_peak_merge, dim3(nb1), dim3(256), 0, sl->stream,
+                       sl->wgpk, nent, sl->dmerge);
+    hipLaunchKernelGGL(k_peak_merge, dim3(1), dim3(256), 0, sl->stream,
+                       sl->dmerge, (long)nb1 * 5, sl->dtop5);
+  }
+  CHK(c, hipMemcpyAsync(sl->htop5, sl->dtop5, 5 * sizeof(bs_peak),
+                        hipMemcpyDeviceToHost, sl->stream));
+  CHK(c, hipEventRecord(sl->peaks_ready, sl->stream));
+  sl->stage = 1;
+  return BS_OK;
+}
+
+/* Phase B: wait for the top-5, build the periodic candidate set
+ * [PIN-CAND], launch r-test + sub-pixel gather; records sl->done. */
+static int stitch_phaseB(bs_ctx *c, bs_slot *sl,
+                         const bs_stitch_params *prm) {
+  CHK(c, hipEventSynchronize(sl->peaks_ready));
+  const int K = prm->peaks_to_check > 0 ? prm->peaks_to_check : 5;
+  if (getenv("BS_DEBUG_PEAKS")) {
+    for (int k = 0; k < 5; ++k) {
+      long long idx = sl->htop5[k].idx;
+      fprintf(stderr, "[bs] pair %zu peak%d v=%.6g zyx=(%lld,%lld,%lld)\n",
+              sl->out_idx, k, sl->htop5[k].v,
+              idx / ((long)sl->Px * sl->Py), (idx / sl->Px) % sl->Py,
+              idx % sl->Px);
+    }
+  }
+  sl->hc.clear();
+  long long *pkidx = sl->pkidx_h; /* must outlive the async H2D below */
+  sl->npk = 0;
+  double min_n = prm->min_overlap_ratio *
+                 std::min((double)sl->m[0][0] * sl->m[0][1] * sl->m[0][2],
+                          (double)sl->m[1][0] * sl->m[1][1] * sl->m[1][2]);
+  for (int k = 0; k < K && k < 5; ++k) {
+    if (sl->htop5[k].v <= -2.0e38f) break;
+    long long idx = sl->htop5[k].idx;
+    int px_ = (int)(idx % sl->Px);
+    long tt_ = idx / sl->Px;
+    int py_ = (int)(tt_ % sl->Py), pz_ = (int)(tt_ / sl->Py);
+    pkidx[sl->npk++] = idx;
+    for (int ci = 0; ci < 8; ++ci) {
+      int sxyz[3] = {px_, py_, pz_};
+      if (ci & 1) sxyz[0] -= sl->Px;
+      if (ci & 2) sxyz[1] -= sl->Py;
+      if (ci & 4) sxyz[2] -= sl->Pz;
+      int lo[3], hi[3];
+      bool ok = true;
+      for (int d = 0; d < 3; ++d) {
+        lo[d] = std::max(0, -sxyz[d]);
+        hi[d] = std::min(sl->m[0][d], sl->m[1][d] - sxyz[d]);
+        if (hi[d] <= lo[d]) ok = false;
+      }
+      if (!ok) continue;
+      long n = (long)(hi[0] - lo[0]) * (hi[1] - lo[1]) * (hi[2] - lo[2]);
+      if ((double)n < std::max(min_n, 1.0)) continue;
+      bs_hostcand h;
+      h.gc = {lo[0], lo[1], lo[2], hi[0] - lo[0], hi[1] - lo[1],
+              hi[2] - lo[2], sxyz[0], sxyz[1], sxyz[2]};
+      h.rank = k;
+      h.ci = ci;
+      h.n = n;
+      sl->hc.push_back(h);
+    }
+  }
+  if (!sl->hc.empty()) {
+    std::vector<bs_cand> gc(sl->hc.size());
+    for (size_t i = 0; i < sl->hc.size(); ++i) gc[i] = sl->hc[i].gc;
+    CHK(c, hipMemcpyAsync(sl->dcands, gc.data(), gc.size() * sizeof(bs_cand),
+                          hipMemcpyHostToDevice, sl->stream));
+    CHK(c, hipMemsetAsync(sl->dsums, 0, gc.size() * 5 * sizeof(u64),
+                          sl->stream));
+    CHK(c, hipStreamSynchronize(sl->stream)); /* gc dies at scope end */
+    {
+      long maxrows = 1;
+      for (auto &h : sl->hc)
+        maxrows = std::max(maxrows, (long)h.gc.ny * h.gc.nz);
+      bs_tim tt(c, BS_K_CORR, sl->stream);
+      hipLaunchKernelGGL(k_rtest,
+                         dim3((unsigned)std::min(2048L, maxrows),
+                              (unsigned)gc.size()),
+                         dim3(256), 0, sl->stream, sl->reg[0], sl->reg[1],
+                         sl->dcands, sl->dsums);
+    }
+    CHK(c, hipMemcpyAsync(sl->hsums, sl->dsums,
+                          sl->hc.size() * 5 * sizeof(u64),
+                          hipMemcpyDeviceToHost, sl->stream));
+    CHK(c, hipMemcpyAsync(sl->dpkidx, pkidx, sl->npk * sizeof(long long),
+                          hipMemcpyHostToDevice, sl->stream));
+    {
+      bs_tim tt(c, BS_K_SUBPIX, sl->stream);
+      hipLaunchKernelGGL(k_gather_subpix, dim3(1), dim3(8), 0, sl->stream,
+                         sl->pcm, sl->Px, sl->Py, sl->Pz, sl->dpkidx,
+                         sl->npk, sl->dsubpix);
+    }
+    CHK(c, hipMemcpyAsync(sl->hsubpix, sl->dsubpix,
+                          sl->npk * 7 * sizeof(float),
+                          hipMemcpyDeviceToHost, sl->stream));
+  }
+  CHK(c, hipEventRecord(sl->done, sl->stream));
+  sl->stage = 2;
+  return BS_OK;
+}
+
+/* Phase C: wait for sums, select the winner exactly as the oracle does
+ * ((-r, rank, ci) order over identical integer sums), apply the
+ * sub-pixel fit [PIN-SUB], scale by ds. */
+static int stitch_phaseC(bs_ctx *c, bs_slot *sl,
+                         const bs_stitch_params *prm, bs_shift_result *out) {
+  CHK(c, hipEventSynchronize(sl->done));
+  bs_shift_result &o = out[sl->out_idx];
+  o = {{0, 0, 0}, 0.0, 0};
+  double best_r = -3.0;
+  int best_i = -1;
+  for (size_t i = 0; i < sl->hc.size(); ++i) {
+    const u64 *s = sl->hsums + i * 5;
+    double n = (double)sl->hc[i].n;
+    double sa = (double)s[0], sb = (double)s[1];
+    double num = (double)s[4] - sa * sb / n;
+    double da = (double)s[2] - sa * sa / n;
+    double db = (double)s[3] - sb * sb / n;
+    if (da <= 0 || db <= 0) continue;
+    double r = num / std::sqrt(da * db);
+    /* iteration is (rank asc, ci asc): first max == oracle's key order */
+    if (best_i < 0 || r > best_r) {
+      best_r = r;
+      best_i = (int)i;
+    }
+  }
+  sl->stage = 0;
+  if (best_i < 0) return BS_OK; /* stays invalid */
+  const bs_hostcand &w = sl->hc[best_i];
+  double shift[3] = {(double)w.gc.sx, (double)w.gc.sy, (double)w.gc.sz};
+  if (prm->do_subpixel) {
+    const float *f = sl->hsubpix + w.rank * 7;
+    for (int d = 0; d < 3; ++d) {
+      double fm = f[1 + 2 * d], f0 = f[0], fp = f[2 + 2 * d];
+      double den = fm - 2.0 * f0 + fp;
+      if (std::fabs(den) < 1e-12) continue;
+      double sp = 0.5 * (fm - fp) / den;
+      shift[d] += std::min(0.5, std::max(-0.5, sp));
+    }
+  }
+  o.shift[0] = shift[0] * sl->dsf[0];
+  o.shift[1] = shift[1] * sl->dsf[1];
+  o.shift[2] = shift[2] * sl->dsf[2];
+  o.r = best_r;
+  o.valid = 1;
+  return BS_OK;
+}
+
 extern "C" int bs_stitch_batch(bs_ctx *c, const bs_pair_desc *pairs, size_t np,
                                const bs_stitch_params *prm,
                                bs_shift_result *out) {
   if (!c || !pairs || !prm || !out) return BS_EINVAL;
   std::lock_guard<std::mutex> g(c->mu);
   CHK(c, hipSetDevice(c->dev));
-  const int K = prm->peaks_to_check > 0 ? prm->peaks_to_check : 5;
-  if (K > 5) {
+  if (prm->peaks_to_check > 5) {
     c->err = "peaks_to_check > 5 unsupported in this build";
     return BS_EUNSUP;
   }
-  hipEvent_t bev0, bev1;
-  if (c->evused == c->evpool.size()) {
-    hipEvent_t a, b;
-    CHK(c, hipEventCreate(&a));
-    CHK(c, hipEventCreate(&b));
-    c->evpool.push_back({a, b});
+  for (size_t i = 0; i < np; ++i) out[i] = {{0, 0, 0}, 0.0, 0};
+  auto t0 = std::chrono::steady_clock::now();
+  int rc = BS_OK;
+  for (size_t i = 0; i < np && rc == BS_OK; ++i) {
+    bs_slot *sl = &c->slot[i & 1];
+    if (sl->stage == 1) rc = stitch_phaseB(c, sl, prm);
+    if (rc == BS_OK && sl->stage == 2) rc = stitch_phaseC(c, sl, prm, out);
+    if (rc == BS_OK) rc = stitch_phaseA(c, sl, pairs[i], prm, i);
   }
-  bev0 = c->evpool[c->evused].first;
-  bev1 = c->evpool[c->evused].second;
-  c->evused++;
-  CHK(c, hipEventRecord(bev0, c->stream));
-
-  for (size_t ip = 0; ip < np; ++ip) {
-    const bs_pair_desc &pd = pairs[ip];
-    out[ip] = {{0, 0, 0}, 0.0, 0};
-    auto ita = c->views.find(pd.view_a);
-    auto itb = c->views.find(pd.view_b);
-    if (ita == c->views.end() || itb == c->views.end()) {
-      c->err = "view not uploaded";
-      return BS_ENOVIEW;
-    }
-    int ds[3] = {prm->ds[0] > 0 ? prm->ds[0] : 1,
-                 prm->ds[1] > 0 ? prm->ds[1] : 1,
-                 prm->ds[2] > 0 ? prm->ds[2] : 1};
-    bool dsall1 = ds[0] == 1 && ds[1] == 1 && ds[2] == 1;
-    /* region descriptors (A=0, B=1) */
-    bs_region reg[2];
-    int m[2][3]; /* downsampled dims x,y,z */
-    for (int t = 0; t < 2; ++t) {
-      const bs_view_rec &vr = t == 0 ? ita->second : itb->second;
-      const int64_t *off = t == 0 ? pd.off_a : pd.off_b;
-      const int64_t *size = t == 0 ? pd.size_a : pd.size_b;
-      for (int d = 0; d < 3; ++d) {
-        if (off[d] < 0 || size[d] <= 0 || off[d] + size[d] > vr.dims[d]) {
-          c->err = "pair interval out of view bounds";
-          return BS_EINVAL;
-        }
-        m[t][d] = (int)(size[d] / ds[d]);
-        if (m[t][d] < 1) m[t][d] = 1;
-      }
-      if (dsall1) {
-        reg[t] = {vr.dptr, vr.dims[0], vr.dims[0] * vr.dims[1],
-                  off[0], off[1], off[2], m[t][0], m[t][1], m[t][2]};
-      } else {
-        size_t need = (size_t)m[t][0] * m[t][1] * m[t][2] * 2;
-        int rc = ensure_dev(c, (void **)&c->regbuf[t], &c->reg_cap[t], need);
-        if (rc) return rc;
-        bs_region src = {vr.dptr, vr.dims[0], vr.dims[0] * vr.dims[1],
-                         off[0], off[1], off[2], 0, 0, 0};
-        long nrows = (long)m[t][1] * m[t][2];
-        bs_tim tt(c, BS_K_DOWNSAMPLE);
-        hipLaunchKernelGGL(k_downsample, dim3(std::min(4096L, nrows)),
-                           dim3(256), 0, c->stream, src, c->regbuf[t],
-                           m[t][0], m[t][1], m[t][2], ds[0], ds[1], ds[2]);
-        reg[t] = {c->regbuf[t], m[t][0], (long)m[t][0] * m[t][1], 0, 0, 0,
-                  m[t][0], m[t][1], m[t][2]};
-      }
-    }
-    /* padded FFT dims [PIN-PAD] */
-    int Px = next_pow2(std::max(m[0][0], m[1][0]));
-    int Py = next_pow2(std::max(m[0][1], m[1][1]));
-    int Pz = next_pow2(std::max(m[0][2], m[1][2]));
-    if (Px > 1024 || Py > 1024 || Pz > 1024) {
-      c->err = "FFT size > 1024 unsupported";
-      return BS_EUNSUP;
-    }
-    int Cx = Px / 2 + 1;
-    long Cxp = (Cx + 15) & ~15L;
-    size_t spec_half = (size_t)Pz * Py * Cxp;
-    int rc = ensure_dev(c, (void **)&c->spec, &c->spec_cap,
-                        2 * spec_half * sizeof(f2));
-    if (rc) return rc;
-    size_t pcm_n = (size_t)Pz * Py * Px;
-    rc = ensure_dev(c, (void **)&c->pcm, &c->pcm_cap, pcm_n * sizeof(float));
-    if (rc) return rc;
-    f2 *twx = get_twiddle(c, Px), *twy = get_twiddle(c, Py),
-       *twz = get_twiddle(c, Pz);
-    if (!twx || !twy || !twz) {
-      c->err = "twiddle alloc failed";
-      return BS_ENOMEM;
-    }
-    f2 *spec[2] = {c->spec, c->spec + spec_half};
-    /* forward x (R2C) per tile */
-    for (int t = 0; t < 2; ++t) {
-      long nlines = (long)reg[t].my * reg[t].mz;
-      long ngrp = (nlines + LPB_X - 1) / LPB_X;
-      size_t lds = ((Px / 4) + (size_t)LPB_X * (Px / 2)) * sizeof(f2);
-      bs_tim tt(c, BS_K_FFT_X_FWD);
-      hipLaunchKernelGGL(k_fft_x_fwd, dim3(std::min(4096L, ngrp)),
-                         dim3(LPB_X * TPL_X), lds, c->stream, reg[t], spec[t],
-                         Px, ilog2(Px), Cx, Cxp, Py, twx);
-    }
-    /* forward y: groups = z slices (z < mz), lines = Cx x-columns */
-    int nchunks = (Cx + LPB_S - 1) / LPB_S;
-    for (int t = 0; t < 2; ++t) {
-      size_t lds = ((Py / 2) + (size_t)LPB_S * Py) * sizeof(f2);
-      bs_tim tt(c, BS_K_FFT_Y_FWD);
-      hipLaunchKernelGGL(k_fft_pass,
-                         dim3(std::min(4096L, (long)reg[t].mz * nchunks)),
-                         dim3(LPB_S * TPL_S), lds, c->stream, spec[t],
-                         (const f2 *)nullptr, spec[t], Py, ilog2(Py), Cxp,
-                         (long)Py * Cxp, Cx, nchunks, reg[t].mz, reg[t].my,
-                         +1, 1.0f, twy);
-    }
-    /* forward z: groups = y rows (all Py), valid = mz */
-    for (int t = 0; t < 2; ++t) {
-      size_t lds = ((Pz / 2) + (size_t)LPB_S * Pz) * sizeof(f2);
-      bs_tim tt(c, BS_K_FFT_Z_FWD);
-      hipLaunchKernelGGL(k_fft_pass,
-                         dim3(std::min(4096L, (long)Py * nchunks)),
-                         dim3(LPB_S * TPL_S), lds, c->stream, spec[t],
-                         (const f2 *)nullptr, spec[t], Pz, ilog2(Pz),
-                         (long)Py * Cxp, Cxp, Cx, nchunks, Py, reg[t].mz,
-                         +1, 1.0f, twz);
-    }
-    /* inverse z fused with cross-power normalise [PIN-EPS] */
-    {
-      float scale = 1.0f / ((float)Px * (float)Py * (float)Pz);
-      size_t lds = ((Pz / 2) + (size_t)LPB_S * Pz) * sizeof(f2);
-      bs_tim tt(c, BS_K_FFT_Z_INV);
-      hipLaunchKernelGGL(k_fft_pass,
-                         dim3(std::min(4096L, (long)Py * nchunks)),
-                         dim3(LPB_S * TPL_S), lds, c->stream, spec[0],
-                         spec[1], spec[0], Pz, ilog2(Pz), (long)Py * Cxp,
-                         Cxp, Cx, nchunks, Py, Pz, -1, scale, twz);
-    }
-    /* inverse y */
-    {
-      size_t lds = ((Py / 2) + (size_t)LPB_S * Py) * sizeof(f2);
-      bs_tim tt(c, BS_K_FFT_Y_INV);
-      hipLaunchKernelGGL(k_fft_pass,
-                         dim3(std::min(4096L, (long)Pz * nchunks)),
-                         dim3(LPB_S * TPL_S), lds, c->stream, spec[0],
-                         (const f2 *)nullptr, spec[0], Py, ilog2(Py), Cxp,
-                         (long)Py * Cxp, Cx, nchunks, Pz, Py, -1, 1.0f,
-                         twy);
-    }
-    /* inverse x (C2R) -> PCM */
-    {
-      long nlines = (long)Pz * Py;
-      long ngrp = (nlines + LPB_X - 1) / LPB_X;
-      size_t lds = ((Px / 4) + (size_t)LPB_X * (Px / 2)) * sizeof(f2);
-      bs_tim tt(c, BS_K_FFT_X_INV);
-      hipLaunchKernelGGL(k_fft_x_inv, dim3(std::min(4096L, ngrp)),
-                         dim3(LPB_X * TPL_X), lds, c->stream, spec[0], c->pcm,
-                         Px, ilog2(Px), Cx, Cxp, nlines, twx);
-    }
-    c->dbg_px = Px;
-    c->dbg_py = Py;
-    c->dbg_pz = Pz;
-    /* peak scan [PIN-MAX] */
-    long ntiles = (long)((Px + PK_TX - 1) / PK_TX) *
-                  ((Py + PK_TY - 1) / PK_TY) * ((Pz + PK_TZ - 1) / PK_TZ);
-    long npkwg = std::min(2048L, ntiles);
-    rc = ensure_dev(c, (void **)&c->wgpk, &c->wgpk_cap,
-                    (size_t)npkwg * 5 * sizeof(bs_peak));
-    if (rc) return rc;
-    {
-      bs_tim tt(c, BS_K_PEAK);
-      hipLaunchKernelGGL(k_peak_tile, dim3(npkwg), dim3(256), 0, c->stream,
-                         c->pcm, Px, Py, Pz, c->wgpk);
-    }
-    {
-      bs_tim tt(c, BS_K_PEAK_MERGE);
-      long nent = npkwg * 5;
-      int nb1 = (int)std::min(64L, (nent + 1279) / 1280);
-      hipLaunchKernelGGL(k_peak_merge, dim3(nb1), dim3(256), 0, c->stream,
-                         c->wgpk, nent, c->dmerge);
-      hipLaunchKernelGGL(k_peak_merge, dim3(1), dim3(256), 0, c->stream,
-                         c->dmerge, (long)nb1 * 5, c->dtop5);
-    }
-    CHK(c, hipMemcpyAsync(c->htop5, c->dtop5, 5 * sizeof(bs_peak),
-                          hipMemcpyDeviceToHost, c->stream));
-    CHK(c, hipStreamSynchronize(c->stream));
-    if (getenv("BS_DEBUG_PEAKS")) {
-      for (int k = 0; k < 5; ++k) {
-        long long idx = c->htop5[k].idx;
-        fprintf(stderr, "[bs] pair %zu peak%d v=%.6g zyx=(%lld,%lld,%lld)\n",
-                ip, k, c->htop5[k].v, idx / ((long)Px * Py),
-                (idx / Px) % Py, idx % Px);
-      }
-    }
-
-    /* host: candidates [PIN-CAND] */
-    struct HostCand {
-      bs_cand gc;
-      int rank, ci;
-      long n;
-    };
-    std::vector<HostCand> hc;
-    std::vector<long long> pkidx;
-    double min_n = prm->min_overlap_ratio *
-                   std::min((double)m[0][0] * m[0][1] * m[0][2],
-                            (double)m[1][0] * m[1][1] * m[1][2]);
-    int npk = 0;
-    for (int k = 0; k < K; ++k) {
-      if (c->htop5[k].v <= -2.0e38f) break;
-      long long idx = c->htop5[k].idx;
-      int px_ = (int)(idx % Px);
-      long tt_ = idx / Px;
-      int py_ = (int)(tt_ % Py), pz_ = (int)(tt_ / Py);
-      pkidx.push_back(idx);
-      npk = k + 1;
-      for (int ci = 0; ci < 8; ++ci) {
-        int sxyz[3] = {px_, py_, pz_};
-        if (ci & 1) sxyz[0] -= Px;
-        if (ci & 2) sxyz[1] -= Py;
-        if (ci & 4) sxyz[2] -= Pz;
-        /* overlap in A coords: lo = max(0,-s), hi = min(na, nb - s) */
-        int lo[3], hi[3];
-        bool ok = true;
-        for (int d = 0; d < 3; ++d) {
-          lo[d] = std::max(0, -sxyz[d]);
-          hi[d] = std::min(m[0][d], m[1][d] - sxyz[d]);
-          if (hi[d] <= lo[d]) ok = false;
-        }
-        if (!ok) continue;
-        long n = (long)(hi[0] - lo[0]) * (hi[1] - lo[1]) * (hi[2] - lo[2]);
-        if ((double)n < std::max(min_n, 1.0)) continue;
-        HostCand h;
-        h.gc = {lo[0], lo[1], lo[2], hi[0] - lo[0], hi[1] - lo[1],
-                hi[2] - lo[2], sxyz[0], sxyz[1], sxyz[2]};
-        h.rank = k;
-        h.ci = ci;
-        h.n = n;
-        hc.push_back(h);
-      }
-    }
-    double best_r = -3.0;
-    int best_i = -1;
-    if (!hc.empty()) {
-      std::vector<bs_cand> gc(hc.size());
-      for (size_t i = 0; i < hc.size(); ++i) gc[i] = hc[i].gc;
-      CHK(c, hipMemcpyAsync(c->dcands, gc.data(),
-                            gc.size() * sizeof(bs_cand),
-                            hipMemcpyHostToDevice, c->stream));
-      CHK(c, hipMemsetAsync(c->dsums, 0, gc.size() * 5 * sizeof(u64),
-                            c->stream));
-      {
-        long maxrows = 1;
-        for (auto &h : hc)
-          maxrows = std::max(maxrows, (long)h.gc.ny * h.gc.nz);
-        bs_tim tt(c, BS_K_CORR);
-        hipLaunchKernelGGL(k_rtest,
-                           dim3((unsigned)std::min(2048L, maxrows),
-                                (unsigned)gc.size()),
-                           dim3(256), 0, c->stream, reg[0], reg[1],
-                           c->dcands, c->dsums);
-      }
-      CHK(c, hipMemcpyAsync(c->hsums, c->dsums, gc.size() * 5 * sizeof(u64),
-                            hipMemcpyDeviceToHost, c->stream));
-      /* subpixel gather for all peaks while r-test runs */
-      CHK(c, hipMemcpyAsync(c->dpkidx, pkidx.data(),
-                            pkidx.size() * sizeof(long long),
-                            hipMemcpyHostToDevice, c->stream));
-      {
-        bs_tim tt(c, BS_K_SUBPIX);
-        hipLaunchKernelGGL(k_gather_subpix, dim3(1), dim3(8), 0, c->stream,
-                           c->pcm, Px, Py, Pz, c->dpkidx, npk, c->dsubpix);
-      }
-      CHK(c, hipMemcpyAsync(c->hsubpix, c->dsubpix, npk * 7 * sizeof(float),
-                            hipMemcpyDeviceToHost, c->stream));
-      CHK(c, hipStreamSynchronize(c->stream));
-      /* winner: max r, ties by (rank, ci) — identical to oracle order */
-      for (size_t i = 0; i < hc.size(); ++i) {
-        const u64 *s = c->hsums + i * 5;
-        double n = (double)hc[i].n;
-        double sa = (double)s[0], sb = (double)s[1];
-        double num = (double)s[4] - sa * sb / n;
-        double da = (double)s[2] - sa * sa / n;
-        double db = (double)s[3] - sb * sb / n;
-        if (da <= 0 || db <= 0) continue;
-        double r = num / std::sqrt(da * db);
-        /* iteration is (rank asc, ci asc): keeping the first max matches
-         * the oracle's (-r, rank, ci) key ordering exactly */
-        if (best_i < 0 || r > best_r) {
-          best_r = r;
-          best_i = (int)i;
-        }
-      }
-    }
-    if (best_i < 0) continue; /* result stays invalid */
-    const HostCand &w = hc[best_i];
-    double shift[3] = {(double)w.gc.sx, (double)w.gc.sy, (double)w.gc.sz};
-    if (prm->do_subpixel) { /* [PIN-SUB] */
-      const float *f = c->hsubpix + w.rank * 7;
-      for (int d = 0; d < 3; ++d) {
-        double fm = f[1 + 2 * d], f0 = f[0], fp = f[2 + 2 * d];
-        double den = fm - 2.0 * f0 + fp;
-        if (std::fabs(den) < 1e-12) continue;
-        double o = 0.5 * (fm - fp) / den;
-        shift[d] += std::min(0.5, std::max(-0.5, o));
-      }
-    }
-    out[ip].shift[0] = shift[0] * ds[0];
-    out[ip].shift[1] = shift[1] * ds[1];
-    out[ip].shift[2] = shift[2] * ds[2];
-    out[ip].r = best_r;
-    out[ip].valid = 1;
+  /* drain both slots, older first */
+  for (size_t k = np >= 2 ? np - 2 : 0; k < np && rc == BS_OK; ++k) {
+    bs_slot *sl = &c->slot[k & 1];
+    if (sl->stage == 1) rc = stitch_phaseB(c, sl, prm);
+    if (rc == BS_OK && sl->stage == 2) rc = stitch_phaseC(c, sl, prm, out);
   }
-  CHK(c, hipEventRecord(bev1, c->stream));
-  CHK(c, hipStreamSynchronize(c->stream));
-  float bms = 0.0f;
-  (void)hipEventElapsedTime(&bms, bev0, bev1);
-  c->stats.batch_ms = bms;
+  for (int s = 0; s < 2 && rc == BS_OK; ++s)
+    CHK(c, hipStreamSynchronize(c->slot[s].stream));
+  c->stats.batch_ms =
+      std::chrono::duration<double, std::milli>(
+          std::chrono::steady_clock::now() - t0)
+          .count();
   c->stats.pairs += (long long)np;
   flush_stats(c);
-  return BS_OK;
+  return rc;
 }
 
 /* Debug-only: download the PCM of the LAST pair processed by
@@ -1314,7 +1395,8 @@ extern "C" int bs_debug_pcm(bs_ctx *c, float *out, int64_t out_dims[3]) {
   out_dims[2] = c->dbg_pz;
   size_t n = (size_t)c->dbg_px * c->dbg_py * c->dbg_pz;
   if (!n) return BS_EINVAL;
-  CHK(c, hipMemcpy(out, c->pcm, n * 4, hipMemcpyDeviceToHost));
+  if (!c->dbg_pcm) return BS_EINVAL;
+  CHK(c, hipMemcpy(out, c->dbg_pcm, n * 4, hipMemcpyDeviceToHost));
   return BS_OK;
 }
 
