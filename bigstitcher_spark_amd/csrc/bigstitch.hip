@@ -287,13 +287,13 @@ __global__ __launch_bounds__(LPB_S *TPL_S) void k_fft_pass(
           v = in[base + e * estride];
         }
       }
-      data[(long)brev_n(e, log2n) * (LPB_S + 1) + line] = v;
+      data[(long)brev_n(e, log2n) * LPB_S + line] = v;
     }
     __syncthreads();
-    fft_lds<LPB_S + 1, TPL_S>(data, (long)line, n, log2n, tl, tw, dir);
+    fft_lds<LPB_S, TPL_S>(data, (long)line, n, log2n, tl, tw, dir);
     if (active)
       for (int e = tl; e < n; e += TPL_S)
-        out[base + e * estride] = data[(long)e * (LPB_S + 1) + line];
+        out[base + e * estride] = data[(long)e * LPB_S + line];
     __syncthreads(); /* LDS reused next group */
   }
 }
@@ -1177,7 +1177,7 @@ static int stitch_phaseA(bs_ctx *c, bs_slot *sl, const bs_pair_desc &pd,
   }
   int nchunks = (Cx + LPB_S - 1) / LPB_S;
   for (int t = 0; t < 2; ++t) {
-    size_t lds = ((Py / 2) + (size_t)(LPB_S + 1) * Py) * sizeof(f2);
+    size_t lds = ((Py / 2) + (size_t)LPB_S * Py) * sizeof(f2);
     bs_tim tt(c, BS_K_FFT_Y_FWD, sl->stream);
     hipLaunchKernelGGL(k_fft_pass,
                        dim3(std::min(4096L, (long)sl->reg[t].mz * nchunks)),
@@ -1187,7 +1187,7 @@ static int stitch_phaseA(bs_ctx *c, bs_slot *sl, const bs_pair_desc &pd,
                        sl->reg[t].my, +1, 1.0f, twy);
   }
   for (int t = 0; t < 2; ++t) {
-    size_t lds = ((Pz / 2) + (size_t)(LPB_S + 1) * Pz) * sizeof(f2);
+    size_t lds = ((Pz / 2) + (size_t)LPB_S * Pz) * sizeof(f2);
     bs_tim tt(c, BS_K_FFT_Z_FWD, sl->stream);
     hipLaunchKernelGGL(k_fft_pass, dim3(std::min(4096L, (long)Py * nchunks)),
                        dim3(LPB_S * TPL_S), lds, sl->stream, spec[t],
@@ -1197,7 +1197,7 @@ static int stitch_phaseA(bs_ctx *c, bs_slot *sl, const bs_pair_desc &pd,
   }
   { /* inverse z fused with cross-power normalise [PIN-EPS] */
     float scale = 1.0f / ((float)Px * (float)Py * (float)Pz);
-    size_t lds = ((Pz / 2) + (size_t)(LPB_S + 1) * Pz) * sizeof(f2);
+    size_t lds = ((Pz / 2) + (size_t)LPB_S * Pz) * sizeof(f2);
     bs_tim tt(c, BS_K_FFT_Z_INV, sl->stream);
     hipLaunchKernelGGL(k_fft_pass, dim3(std::min(4096L, (long)Py * nchunks)),
                        dim3(LPB_S * TPL_S), lds, sl->stream, spec[0],
@@ -1205,7 +1205,7 @@ static int stitch_phaseA(bs_ctx *c, bs_slot *sl, const bs_pair_desc &pd,
                        Cx, nchunks, Py, Pz, -1, scale, twz);
   }
   {
-    size_t lds = ((Py / 2) + (size_t)(LPB_S + 1) * Py) * sizeof(f2);
+    size_t lds = ((Py / 2) + (size_t)LPB_S * Py) * sizeof(f2);
     bs_tim tt(c, BS_K_FFT_Y_INV, sl->stream);
     hipLaunchKernelGGL(k_fft_pass, dim3(std::min(4096L, (long)Pz * nchunks)),
                        dim3(LPB_S * TPL_S), lds, sl->stream, spec[0],
