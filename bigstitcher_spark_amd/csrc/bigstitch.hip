@@ -64,27 +64,19 @@ __device__ __forceinline__ unsigned brev_n(unsigned j, int log2n) {
   return __brev(j) >> (32 - log2n);
 }
 
-/* In-LDS DIT FFT over bit-reversed-loaded data, radix-2^3: three radix-2
- * stages are fused per LDS round (8 elems in registers), so n=512 takes
- * 3 LDS read/write rounds + 3 barriers (vs 9 for plain radix-2). A
- * leading radix-2 or radix-2^2 stage handles log2n % 3. Element e lives
- * at data[base + e*ES]. tw holds e^{-2*pi*i*k/n}, k < n/2; dir<0
- * conjugates (inverse). Every thread of the block must call this (it
- * contains __syncthreads). */
+/* In-LDS radix-2^2 DIT FFT over bit-reversed-loaded data. Two radix-2
+ * stages are fused per LDS round (4 elems in registers), halving LDS
+ * traffic and barriers vs plain radix-2; odd log2n does one radix-2
+ * stage first (twiddle-free: w=1). Element e lives at data[base + e*ES].
+ * tw holds e^{-2*pi*i*k/n}, k < n/2; dir<0 conjugates (inverse). Every
+ * thread of the block must call this (it contains __syncthreads). */
 template <int ES, int TPL>
 __device__ __forceinline__ void fft_lds(f2 *data, long base, int n,
                                         int log2n, int tl, const f2 *tw,
                                         int dir) {
 #define D_(e) data[base + (long)(e) * ES]
-#define BF_(u, v, w, lo, hi)                                                \
-  {                                                                         \
-    f2 t_ = cmul(v, w);                                                     \
-    lo = {u.x + t_.x, u.y + t_.y};                                          \
-    hi = {u.x - t_.x, u.y - t_.y};                                          \
-  }
   int h = 1;
-  const int lead = log2n % 3;
-  if (lead == 1) {
+  if (log2n & 1) {
     for (int bf = tl; bf < (n >> 1); bf += TPL) {
       f2 u = D_(2 * bf), v = D_(2 * bf + 1);
       D_(2 * bf) = {u.x + v.x, u.y + v.y};
@@ -92,68 +84,32 @@ __device__ __forceinline__ void fft_lds(f2 *data, long base, int n,
     }
     h = 2;
     __syncthreads();
-  } else if (lead == 2) {
-    const int s2 = n >> 2;
-    for (int g = tl; g < (n >> 2); g += TPL) {
-      int i = g * 4;
-      f2 a = D_(i), b = D_(i + 1), c = D_(i + 2), d = D_(i + 3);
-      f2 A = {a.x + b.x, a.y + b.y}, B = {a.x - b.x, a.y - b.y};
-      f2 Cc = {c.x + d.x, c.y + d.y}, Dd = {c.x - d.x, c.y - d.y};
-      f2 w2b = tw[s2];
-      if (dir < 0) w2b.y = -w2b.y;
-      f2 u2 = cmul(Dd, w2b);
-      D_(i) = {A.x + Cc.x, A.y + Cc.y};
-      D_(i + 2) = {A.x - Cc.x, A.y - Cc.y};
-      D_(i + 1) = {B.x + u2.x, B.y + u2.y};
-      D_(i + 3) = {B.x - u2.x, B.y - u2.y};
-    }
-    h = 4;
-    __syncthreads();
   }
-  for (; h < n; h <<= 3) {
-    const int q = n >> 3;
-    const int s1 = n / (2 * h), s2 = n / (4 * h), s3 = n / (8 * h);
+  for (; h < n; h <<= 2) {
+    const int q = n >> 2;
+    const int s1 = n / (2 * h), s2 = n / (4 * h);
     for (int g = tl; g < q; g += TPL) {
       int off = g % h, blk = g / h;
-      int i = blk * 8 * h + off;
-      f2 a0 = D_(i), a1 = D_(i + h), a2 = D_(i + 2 * h), a3 = D_(i + 3 * h);
-      f2 a4 = D_(i + 4 * h), a5 = D_(i + 5 * h), a6 = D_(i + 6 * h),
-         a7 = D_(i + 7 * h);
+      int i = blk * 4 * h + off;
+      f2 a = D_(i), b = D_(i + h), c = D_(i + 2 * h), d = D_(i + 3 * h);
       f2 w1 = tw[off * s1];
+      if (dir < 0) w1.y = -w1.y;
+      f2 t1 = cmul(b, w1), t2 = cmul(d, w1);
+      f2 A = {a.x + t1.x, a.y + t1.y}, B = {a.x - t1.x, a.y - t1.y};
+      f2 Cc = {c.x + t2.x, c.y + t2.y}, Dd = {c.x - t2.x, c.y - t2.y};
       f2 w2a = tw[off * s2], w2b = tw[(off + h) * s2];
-      f2 w3a = tw[off * s3], w3b = tw[(off + h) * s3],
-         w3c = tw[(off + 2 * h) * s3], w3d = tw[(off + 3 * h) * s3];
       if (dir < 0) {
-        w1.y = -w1.y; w2a.y = -w2a.y; w2b.y = -w2b.y;
-        w3a.y = -w3a.y; w3b.y = -w3b.y; w3c.y = -w3c.y; w3d.y = -w3d.y;
+        w2a.y = -w2a.y;
+        w2b.y = -w2b.y;
       }
-      f2 b0, b1, b2, b3, b4, b5, b6, b7;
-      BF_(a0, a1, w1, b0, b1);
-      BF_(a2, a3, w1, b2, b3);
-      BF_(a4, a5, w1, b4, b5);
-      BF_(a6, a7, w1, b6, b7);
-      f2 c0, c1, c2, c3, c4, c5, c6, c7;
-      BF_(b0, b2, w2a, c0, c2);
-      BF_(b1, b3, w2b, c1, c3);
-      BF_(b4, b6, w2a, c4, c6);
-      BF_(b5, b7, w2b, c5, c7);
-      f2 d0, d1, d2, d3, d4, d5, d6, d7;
-      BF_(c0, c4, w3a, d0, d4);
-      BF_(c1, c5, w3b, d1, d5);
-      BF_(c2, c6, w3c, d2, d6);
-      BF_(c3, c7, w3d, d3, d7);
-      D_(i) = d0;
-      D_(i + h) = d1;
-      D_(i + 2 * h) = d2;
-      D_(i + 3 * h) = d3;
-      D_(i + 4 * h) = d4;
-      D_(i + 5 * h) = d5;
-      D_(i + 6 * h) = d6;
-      D_(i + 7 * h) = d7;
+      f2 u1 = cmul(Cc, w2a), u2 = cmul(Dd, w2b);
+      D_(i) = {A.x + u1.x, A.y + u1.y};
+      D_(i + 2 * h) = {A.x - u1.x, A.y - u1.y};
+      D_(i + h) = {B.x + u2.x, B.y + u2.y};
+      D_(i + 3 * h) = {B.x - u2.x, B.y - u2.y};
     }
     __syncthreads();
   }
-#undef BF_
 #undef D_
 }
 
