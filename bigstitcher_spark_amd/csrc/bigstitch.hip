@@ -223,33 +223,82 @@ __global__ __launch_bounds__(LPB_S *TPL_S) void k_fft_pass(
   for (int i = tid; i < (n >> 1); i += LPB_S * TPL_S) tw[i] = twg[i];
   __syncthreads();
 
+  /* I/O-phase mapping: pairs of adjacent lines -> one float4 (two
+   * complex) per global transaction; 16-B aligned by construction
+   * (gstride/estride are multiples of 16 f2, x even). */
+  const int pl = tid & 7, t2 = tid >> 3; /* pair-line, 64 elems/thread */
   const long nwg = (long)ngroups * nchunks;
   for (long wg = blockIdx.x; wg < nwg; wg += gridDim.x) {
     const int group = (int)(wg / nchunks);
-    const int x = (int)(wg % nchunks) * LPB_S + line;
-    const bool active = x < nlines;
-    const long base = (long)group * gstride + x;
-    for (int e = tl; e < n; e += TPL_S) {
-      f2 v = {0.0f, 0.0f};
-      if (active && e < valid) {
-        if (in2) {
-          f2 q = conjmul(in[base + e * estride], in2[base + e * estride]);
-          float m2 = q.x * q.x + q.y * q.y;
-          if (m2 > 1e-40f) {
-            float s = scale / sqrtf(m2);
-            v = {q.x * s, q.y * s};
+    const int x2 = (int)(wg % nchunks) * LPB_S + 2 * pl;
+    const bool pair_ok = x2 + 1 < nlines;
+    const long base2 = (long)group * gstride + x2;
+    if (pair_ok) {
+      for (int e = t2; e < n; e += 64) {
+        float4 v = {0.0f, 0.0f, 0.0f, 0.0f};
+        if (e < valid) {
+          if (in2) {
+            const float4 va = *(const float4 *)&in[base2 + e * estride];
+            const float4 vb = *(const float4 *)&in2[base2 + e * estride];
+            f2 q0 = conjmul({va.x, va.y}, {vb.x, vb.y});
+            f2 q1 = conjmul({va.z, va.w}, {vb.z, vb.w});
+            float m0 = q0.x * q0.x + q0.y * q0.y;
+            float m1 = q1.x * q1.x + q1.y * q1.y;
+            if (m0 > 1e-40f) {
+              float s = scale / sqrtf(m0);
+              v.x = q0.x * s;
+              v.y = q0.y * s;
+            }
+            if (m1 > 1e-40f) {
+              float s = scale / sqrtf(m1);
+              v.z = q1.x * s;
+              v.w = q1.y * s;
+            }
+          } else {
+            v = *(const float4 *)&in[base2 + e * estride];
           }
-        } else {
-          v = in[base + e * estride];
+        }
+        *(float4 *)&data[(long)brev_n(e, log2n) * LPB_S + 2 * pl] = v;
+      }
+    } else { /* partial last chunk: per-line scalar */
+      for (int l = 0; l < 2; ++l) {
+        const int x = x2 + l;
+        const bool active = x < nlines;
+        const long base = (long)group * gstride + x;
+        for (int e = t2; e < n; e += 64) {
+          f2 v = {0.0f, 0.0f};
+          if (active && e < valid) {
+            if (in2) {
+              f2 q =
+                  conjmul(in[base + e * estride], in2[base + e * estride]);
+              float m2 = q.x * q.x + q.y * q.y;
+              if (m2 > 1e-40f) {
+                float s = scale / sqrtf(m2);
+                v = {q.x * s, q.y * s};
+              }
+            } else {
+              v = in[base + e * estride];
+            }
+          }
+          data[(long)brev_n(e, log2n) * LPB_S + 2 * pl + l] = v;
         }
       }
-      data[(long)brev_n(e, log2n) * LPB_S + line] = v;
     }
     __syncthreads();
     fft_lds<LPB_S, TPL_S>(data, (long)line, n, log2n, tl, tw, dir);
-    if (active)
-      for (int e = tl; e < n; e += TPL_S)
-        out[base + e * estride] = data[(long)e * LPB_S + line];
+    if (pair_ok) {
+      for (int e = t2; e < n; e += 64)
+        *(float4 *)&out[base2 + e * estride] =
+            *(const float4 *)&data[(long)e * LPB_S + 2 * pl];
+    } else {
+      for (int l = 0; l < 2; ++l) {
+        const int x = x2 + l;
+        if (x >= nlines) continue;
+        const long base = (long)group * gstride + x;
+        for (int e = t2; e < n; e += 64)
+          out[base + e * estride] = data[(long)e * LPB_S + 2 * pl + l];
+      }
+    }
     __syncthreads(); /* LDS reused next group */
   }
 }
