@@ -6,7 +6,7 @@ CXXFLAGS = --offload-arch=$(ARCH) -O3 -std=c++17 -fPIC -Wall
 
 LIB = bigstitcher_spark_amd/libbigstitch.so
 
-all: $(LIB)
+all: $(LIB) cli
 
 $(LIB): bigstitcher_spark_amd/csrc/bigstitch.hip include/bigstitch.h
 	$(HIPCC) $(CXXFLAGS) -shared -Iinclude $< -o $@
@@ -18,3 +18,32 @@ clean:
 	rm -f $(LIB)
 
 .PHONY: all clean resource-report
+
+# ---- host surface: N5/XML/SpimData + CLI binaries ----
+HOSTDIR = bigstitcher_spark_amd/csrc/host
+HOSTOBJS = $(HOSTDIR)/bs_json.o $(HOSTDIR)/bs_n5.o $(HOSTDIR)/bs_xml.o $(HOSTDIR)/bs_spimdata.o
+BINDIR = bigstitcher_spark_amd/bin
+CXX_HOST = g++
+HOSTFLAGS = -O2 -std=c++17 -fPIC -Wall
+
+$(HOSTDIR)/%.o: $(HOSTDIR)/%.cpp $(HOSTDIR)/%.h
+	$(CXX_HOST) $(HOSTFLAGS) -c $< -o $@
+
+cli: $(BINDIR)/stitching $(BINDIR)/create-fusion-container $(BINDIR)/affine-fusion
+
+$(BINDIR)/stitching: $(HOSTDIR)/cli_stitching.cpp $(HOSTOBJS) $(LIB)
+	@mkdir -p $(BINDIR)
+	$(CXX_HOST) $(HOSTFLAGS) -Iinclude $< $(HOSTOBJS) -o $@ -Lbigstitcher_spark_amd -lbigstitch -lz -L/opt/rocm/lib -lamdhip64 -Wl,-rpath,'$$ORIGIN/..' -Wl,-rpath,/opt/rocm/lib
+
+$(BINDIR)/create-fusion-container: $(HOSTDIR)/cli_container.cpp $(HOSTOBJS)
+	@mkdir -p $(BINDIR)
+	$(CXX_HOST) $(HOSTFLAGS) -Iinclude $< $(HOSTOBJS) -o $@ -lz
+
+$(BINDIR)/affine-fusion: $(HOSTDIR)/cli_fusion.cpp $(HOSTOBJS) $(LIB)
+	@mkdir -p $(BINDIR)
+	$(CXX_HOST) $(HOSTFLAGS) -Iinclude $< $(HOSTOBJS) -o $@ -Lbigstitcher_spark_amd -lbigstitch -lz -L/opt/rocm/lib -lamdhip64 -Wl,-rpath,'$$ORIGIN/..' -Wl,-rpath,/opt/rocm/lib
+
+clean-cli:
+	rm -f $(HOSTOBJS) $(BINDIR)/*
+
+.PHONY: cli clean-cli

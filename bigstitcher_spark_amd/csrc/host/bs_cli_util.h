@@ -1,0 +1,142 @@
+/* Shared helpers for the CLI binaries (stitching / create-fusion-container
+ * / affine-fusion) — flag parsing plus the driver-layer geometry the
+ * reference host performs. */
+#ifndef BS_CLI_UTIL_H
+#define BS_CLI_UTIL_H
+
+#include <array>
+#include <algorithm>
+#include <cmath>
+#include <cstdio>
+#include <cstdlib>
+#include <cstring>
+#include <map>
+#include <string>
+#include <vector>
+
+namespace bscli {
+
+struct Args {
+  std::map<std::string, std::string> kv;
+  std::vector<std::string> flags;
+  bool parse(int argc, char **argv,
+             const std::map<std::string, std::string> &aliases,
+             const std::vector<std::string> &boolean_flags) {
+    for (int i = 1; i < argc; ++i) {
+      std::string a = argv[i];
+      auto it = aliases.find(a);
+      if (it != aliases.end()) a = it->second;
+      if (a.rfind("--", 0) != 0) {
+        fprintf(stderr, "unknown argument: %s\n", argv[i]);
+        return false;
+      }
+      std::string key = a.substr(2);
+      bool isbool = false;
+      for (auto &b : boolean_flags)
+        if (b == key) isbool = true;
+      if (isbool) {
+        flags.push_back(key);
+      } else {
+        if (i + 1 >= argc) {
+          fprintf(stderr, "missing value for %s\n", argv[i]);
+          return false;
+        }
+        kv[key] = argv[++i];
+      }
+    }
+    return true;
+  }
+  bool has(const std::string &k) const {
+    if (kv.count(k)) return true;
+    for (auto &f : flags)
+      if (f == k) return true;
+    return false;
+  }
+  std::string get(const std::string &k, const std::string &dflt = "") const {
+    auto it = kv.find(k);
+    return it == kv.end() ? dflt : it->second;
+  }
+  double getd(const std::string &k, double dflt) const {
+    auto it = kv.find(k);
+    return it == kv.end() ? dflt : atof(it->second.c_str());
+  }
+  long getl(const std::string &k, long dflt) const {
+    auto it = kv.find(k);
+    return it == kv.end() ? dflt : atol(it->second.c_str());
+  }
+};
+
+inline std::vector<long long> parse_ints(const std::string &s) {
+  std::vector<long long> out;
+  std::string cur;
+  for (char c : s + ",") {
+    if (c == ',' || c == ' ') {
+      if (!cur.empty()) out.push_back(atoll(cur.c_str()));
+      cur.clear();
+    } else {
+      cur += c;
+    }
+  }
+  return out;
+}
+
+/* ---- 3x4 affine helpers (row-major, world = L x + t) ---- */
+using M34 = std::array<double, 12>;
+
+inline void decompose(const M34 &m, double L[9], double t[3]) {
+  for (int r = 0; r < 3; ++r) {
+    for (int c = 0; c < 3; ++c) L[r * 3 + c] = m[r * 4 + c];
+    t[r] = m[r * 4 + 3];
+  }
+}
+
+inline bool inv3(const double L[9], double inv[9]) {
+  double det = L[0] * (L[4] * L[8] - L[5] * L[7]) -
+               L[1] * (L[3] * L[8] - L[5] * L[6]) +
+               L[2] * (L[3] * L[7] - L[4] * L[6]);
+  if (std::fabs(det) < 1e-300) return false;
+  double id = 1.0 / det;
+  inv[0] = (L[4] * L[8] - L[5] * L[7]) * id;
+  inv[1] = (L[2] * L[7] - L[1] * L[8]) * id;
+  inv[2] = (L[1] * L[5] - L[2] * L[4]) * id;
+  inv[3] = (L[5] * L[6] - L[3] * L[8]) * id;
+  inv[4] = (L[0] * L[8] - L[2] * L[6]) * id;
+  inv[5] = (L[2] * L[3] - L[0] * L[5]) * id;
+  inv[6] = (L[3] * L[7] - L[4] * L[6]) * id;
+  inv[7] = (L[1] * L[6] - L[0] * L[7]) * id;
+  inv[8] = (L[0] * L[4] - L[1] * L[3]) * id;
+  return true;
+}
+
+inline bool linear_equal(const M34 &a, const M34 &b, double tol = 1e-9) {
+  for (int r = 0; r < 3; ++r)
+    for (int c = 0; c < 3; ++c)
+      if (std::fabs(a[r * 4 + c] - b[r * 4 + c]) > tol) return false;
+  return true;
+}
+
+/* transformed bbox of [0, dims-1]^3 under model */
+inline void tbbox(const M34 &m, const long long dims[3], double lo[3],
+                  double hi[3]) {
+  for (int d = 0; d < 3; ++d) {
+    lo[d] = 1e300;
+    hi[d] = -1e300;
+  }
+  for (int cz = 0; cz < 2; ++cz)
+    for (int cy = 0; cy < 2; ++cy)
+      for (int cx = 0; cx < 2; ++cx) {
+        double p[3] = {cx ? (double)(dims[0] - 1) : 0.0,
+                       cy ? (double)(dims[1] - 1) : 0.0,
+                       cz ? (double)(dims[2] - 1) : 0.0};
+        for (int r = 0; r < 3; ++r) {
+          double w = m[r * 4 + 0] * p[0] + m[r * 4 + 1] * p[1] +
+                     m[r * 4 + 2] * p[2] + m[r * 4 + 3];
+          if (w < lo[r]) lo[r] = w;
+          if (w > hi[r]) hi[r] = w;
+        }
+      }
+}
+
+}  // namespace bscli
+
+#endif

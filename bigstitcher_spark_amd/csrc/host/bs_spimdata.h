@@ -1,0 +1,80 @@
+/* SpimData2 dataset.xml reader/writer — the subset of the spim_data
+ * schema the hot path touches (SURVEY.md §2 util/Spark verdict):
+ * BasePath, SequenceDescription (ImageLoader bdv.n5, ViewSetups sizes,
+ * Timepoints), ViewRegistrations (transform chains concatenated to one
+ * 3x4 model, world = T0(T1(...(x))) in list order), StitchingResults.
+ *
+ * StitchingResults entry fields mirror the reference's exact result
+ * contract (SerializablePairwiseStitchingResult, util/Spark.java:201-233:
+ * pair of grouped ViewIds, double[3][4] matrix, bbox min/max double[],
+ * r, hash). The XML element/tag NAMES are a restatement [PIN-XML] — the
+ * reference serializes via mvrecon's XmlIoStitchingResults whose source
+ * is not available here (oracle/__init__.py parity note). [PIN-HASH]:
+ * hash restates PairwiseStitchingResult.calculateHash(vrA, vrB)
+ * (call site SparkPairwiseStitching.java:287-289) as the sum of all 12
+ * model entries of both registrations — the artifact's exact formula is
+ * likewise unavailable; the round-trip property (solver drops entries
+ * whose hash no longer matches the current registrations,
+ * Solver.java:404-415) is preserved by ANY deterministic function of
+ * both models, which this is. */
+#ifndef BS_SPIMDATA_H
+#define BS_SPIMDATA_H
+
+#include <array>
+#include <map>
+#include <string>
+#include <utility>
+#include <vector>
+
+#include "bs_xml.h"
+
+namespace bssd {
+
+using ViewId = std::pair<int, int>; /* (timepoint, setup) */
+
+struct ViewSetup {
+  int id = 0;
+  long long dims[3] = {0, 0, 0}; /* x,y,z */
+  std::string name;
+};
+
+struct StitchEntry {
+  std::vector<ViewId> views_a, views_b;
+  double matrix[12] = {1, 0, 0, 0, 0, 1, 0, 0, 0, 0, 1, 0};
+  double bbox_min[3] = {0, 0, 0}, bbox_max[3] = {0, 0, 0};
+  double r = 0, hash = 0;
+};
+
+struct SpimData {
+  std::string xml_path, base_dir;
+  bsx::NodePtr root;
+  std::string n5_path; /* resolved image container (bdv.n5 loader) */
+  std::vector<ViewSetup> setups;
+  std::vector<int> timepoints;
+  /* (tp,setup) -> concatenated 3x4 model, row-major, world = M x local */
+  std::map<ViewId, std::array<double, 12>> regs;
+
+  bool load(const std::string &path, std::string *err);
+  bool save(const std::string &path) const;
+
+  const ViewSetup *setup(int id) const;
+  /* image dataset path inside the n5 container (bdv.n5 layout) */
+  static std::string image_dataset(int setup, int tp, int level = 0);
+
+  std::vector<StitchEntry> stitching_results() const;
+  void set_stitching_results(const std::vector<StitchEntry> &entries);
+
+  static double calculate_hash(const std::array<double, 12> &a,
+                               const std::array<double, 12> &b);
+};
+
+/* Create a minimal dataset.xml + registrations for tests/tools. */
+bsx::NodePtr make_dataset_xml(const std::string &n5_rel,
+                              const std::vector<ViewSetup> &setups,
+                              const std::vector<int> &tps,
+                              const std::map<ViewId, std::array<double, 12>>
+                                  &regs);
+
+}  // namespace bssd
+
+#endif

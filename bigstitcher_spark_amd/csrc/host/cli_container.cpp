@@ -1,0 +1,131 @@
+/* `create-fusion-container` — drop-in for the reference's
+ * CreateFusionContainer (plain host tool, no Spark/GPU; reference
+ * CreateFusionContainer.java). Creates the output N5 container with the
+ * exact Bigstitcher-Spark root-attribute contract the fusion step
+ * reads back (written :302-320, read SparkAffineFusion.java:239-307):
+ * FusionFormat, InputXML, NumTimepoints, NumChannels, Boundingbox_min/
+ * max, PreserveAnisotropy [,AnisotropyFactor], DataType, BlockSize,
+ * [Min/MaxIntensity], MultiResolutionInfos. Datasets are the plain-N5
+ * per-(channel,timepoint) layout "ch{c}tp{t}/s{l}" (:490-516).
+ * Round-1 scope: N5 only (no ZARR/HDF5), level s0 only (the pyramid is
+ * SURVEY.md §8(f) row 1), no anisotropy split. */
+#include <cstdio>
+
+#include "bs_cli_util.h"
+#include "bs_n5.h"
+#include "bs_spimdata.h"
+
+int main(int argc, char **argv) {
+  bscli::Args args;
+  std::map<std::string, std::string> alias = {{"-x", "--xml"},
+                                              {"-o", "--n5Path"}};
+  if (!args.parse(argc, argv, alias, {"preserveAnisotropy"}) ||
+      !args.has("xml") || !args.has("n5Path")) {
+    fprintf(stderr,
+            "usage: create-fusion-container -x dataset.xml -o out.n5 "
+            "[--blockSize 128,128,128] [--dataType UINT16|UINT8|FLOAT32] "
+            "[--minIntensity v --maxIntensity v] [--bbMin x,y,z --bbMax "
+            "x,y,z] [--compression GZIP|RAW]\n");
+    return 2;
+  }
+  bssd::SpimData sd;
+  std::string err;
+  if (!sd.load(args.get("xml"), &err)) {
+    fprintf(stderr, "error: %s\n", err.c_str());
+    return 1;
+  }
+  auto bs = bscli::parse_ints(args.get("blockSize", "128,128,128"));
+  std::string dt = args.get("dataType", "UINT16");
+  std::string n5dt = dt == "UINT8" ? "uint8"
+                     : dt == "FLOAT32" ? "float32" : "uint16";
+  std::string comp =
+      args.get("compression", "GZIP") == "RAW" ? "raw" : "gzip";
+
+  /* bounding box: explicit or the union of transformed view bboxes
+   * (the reference's default "estimate bounding box", :122-211) */
+  long long bbmin[3], bbmax[3];
+  if (args.has("bbMin") && args.has("bbMax")) {
+    auto mn = bscli::parse_ints(args.get("bbMin"));
+    auto mx = bscli::parse_ints(args.get("bbMax"));
+    for (int d = 0; d < 3; ++d) {
+      bbmin[d] = mn[d];
+      bbmax[d] = mx[d];
+    }
+  } else {
+    double lo[3] = {1e300, 1e300, 1e300}, hi[3] = {-1e300, -1e300, -1e300};
+    int tp0 = sd.timepoints.empty() ? 0 : sd.timepoints[0];
+    for (auto &s : sd.setups) {
+      auto r = sd.regs.find({tp0, s.id});
+      if (r == sd.regs.end()) continue;
+      double l[3], h[3];
+      bscli::tbbox(r->second, s.dims, l, h);
+      for (int d = 0; d < 3; ++d) {
+        lo[d] = std::min(lo[d], l[d]);
+        hi[d] = std::max(hi[d], h[d]);
+      }
+    }
+    for (int d = 0; d < 3; ++d) {
+      bbmin[d] = (long long)std::floor(lo[d]);
+      bbmax[d] = (long long)std::ceil(hi[d]);
+    }
+  }
+  long long dims[3] = {bbmax[0] - bbmin[0] + 1, bbmax[1] - bbmin[1] + 1,
+                       bbmax[2] - bbmin[2] + 1};
+  int numTp = (int)sd.timepoints.size(), numCh = 1;
+
+  bsn5::Container n5(args.get("n5Path"));
+  if (!n5.create()) {
+    fprintf(stderr, "cannot create %s\n", args.get("n5Path").c_str());
+    return 1;
+  }
+  auto set = [&](const std::string &k, bsj::ValuePtr v) {
+    n5.set_attr("", "Bigstitcher-Spark/" + k, v);
+  };
+  set("FusionFormat", bsj::Value::mkstr("N5"));
+  set("InputXML", bsj::Value::mkstr(args.get("xml")));
+  set("NumTimepoints", bsj::Value::mkint(numTp));
+  set("NumChannels", bsj::Value::mkint(numCh));
+  set("Boundingbox_min",
+      bsj::Value::mkints(std::vector<long long>{bbmin[0], bbmin[1], bbmin[2]}));
+  set("Boundingbox_max",
+      bsj::Value::mkints(std::vector<long long>{bbmax[0], bbmax[1], bbmax[2]}));
+  set("PreserveAnisotropy", bsj::Value::mkbool(false));
+  set("DataType", bsj::Value::mkstr(dt));
+  set("BlockSize", bsj::Value::mkints(std::vector<long long>{bs[0], bs[1], bs[2]}));
+  if (args.has("minIntensity") && args.has("maxIntensity")) {
+    set("MinIntensity", bsj::Value::mknum(args.getd("minIntensity", 0)));
+    set("MaxIntensity", bsj::Value::mknum(args.getd("maxIntensity", 65535)));
+  }
+  auto mri_all = bsj::Value::mkarr();
+  for (int t = 0; t < numTp; ++t)
+    for (int ch = 0; ch < numCh; ++ch) {
+      char dsname[64];
+      snprintf(dsname, sizeof dsname, "ch%dtp%d/s0", ch, t);
+      bsn5::DatasetAttrs da;
+      da.dims = {dims[0], dims[1], dims[2]};
+      da.block = {(int)bs[0], (int)bs[1], (int)bs[2]};
+      da.dtype = n5dt;
+      da.compression = comp;
+      if (!n5.create_dataset(dsname, da)) {
+        fprintf(stderr, "cannot create dataset %s\n", dsname);
+        return 1;
+      }
+      auto levels = bsj::Value::mkarr();
+      auto l0 = bsj::Value::mkobj();
+      l0->obj["dataset"] = bsj::Value::mkstr(dsname);
+      l0->obj["dimensions"] = bsj::Value::mkints(da.dims);
+      l0->obj["blockSize"] =
+          bsj::Value::mkints(std::vector<int>{da.block[0], da.block[1],
+                                              da.block[2]});
+      l0->obj["absoluteDownsampling"] =
+          bsj::Value::mkints(std::vector<int>{1, 1, 1});
+      levels->arr.push_back(l0);
+      mri_all->arr.push_back(levels);
+    }
+  set("MultiResolutionInfos", mri_all);
+  printf("created %s: %d tp x %d ch, bbox [%lld,%lld,%lld]..[%lld,%lld,%lld]"
+         ", %s %s\n",
+         args.get("n5Path").c_str(), numTp, numCh, bbmin[0], bbmin[1],
+         bbmin[2], bbmax[0], bbmax[1], bbmax[2], dt.c_str(), comp.c_str());
+  return 0;
+}

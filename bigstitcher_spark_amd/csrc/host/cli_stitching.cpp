@@ -1,0 +1,237 @@
+/* `stitching` — drop-in for the reference's SparkPairwiseStitching CLI
+ * (flag surface: SparkPairwiseStitching.java:76-107; pipeline: :110-393).
+ * Host drives libbigstitch's bs_stitch_batch (the computeStitching
+ * replacement, :247-255) over all overlapping view pairs, applies the
+ * FilteredStitchingResults filters (:369-380: minR/maxR/maxShift*), and
+ * writes <StitchingResults> back into dataset.xml (:389).
+ *
+ * Round-1 scope (documented limitations vs reference):
+ *  - one group per view (channel/illum grouping flags accepted, only
+ *    defaults supported),
+ *  - pairs whose registrations differ in their non-translational part
+ *    are skipped (the reference falls back to
+ *    computeStitchingNonEqualTransformations, :259-267). */
+#include <cinttypes>
+#include <cstdio>
+#include <set>
+
+#include "../../../include/bigstitch.h"
+#include "bs_cli_util.h"
+#include "bs_n5.h"
+#include "bs_spimdata.h"
+
+using bscli::M34;
+
+int main(int argc, char **argv) {
+  bscli::Args args;
+  std::map<std::string, std::string> alias = {
+      {"-x", "--xml"}, {"-ds", "--downsampling"}, {"-p", "--peaksToCheck"}};
+  if (!args.parse(argc, argv, alias,
+                  {"disableSubpixelResolution", "dryRun"}) ||
+      !args.has("xml")) {
+    fprintf(stderr,
+            "usage: stitching -x dataset.xml [-ds 2,2,1] [-p 5] "
+            "[--minR 0.3] [--maxR 1.0] [--maxShiftX px] [--maxShiftY px] "
+            "[--maxShiftZ px] [--maxShiftTotal px] "
+            "[--disableSubpixelResolution] [--channelCombine AVERAGE] "
+            "[--illumCombine PICK_BRIGHTEST] [--device N] [--dryRun]\n");
+    return 2;
+  }
+  if (args.get("channelCombine", "AVERAGE") != "AVERAGE" ||
+      args.get("illumCombine", "PICK_BRIGHTEST") != "PICK_BRIGHTEST") {
+    fprintf(stderr, "only default channel/illum grouping supported\n");
+    return 2;
+  }
+  auto ds = bscli::parse_ints(args.get("downsampling", "2,2,1"));
+  if (ds.size() != 3) {
+    fprintf(stderr, "bad -ds\n");
+    return 2;
+  }
+  double minR = args.getd("minR", 0.3), maxR = args.getd("maxR", 1.0);
+  double maxShift[3] = {args.getd("maxShiftX", 1e300),
+                        args.getd("maxShiftY", 1e300),
+                        args.getd("maxShiftZ", 1e300)};
+  double maxShiftTotal = args.getd("maxShiftTotal", 1e300);
+
+  bssd::SpimData sd;
+  std::string err;
+  if (!sd.load(args.get("xml"), &err)) {
+    fprintf(stderr, "error: %s\n", err.c_str());
+    return 1;
+  }
+  printf("stitching: %zu setups, %zu timepoints, container %s\n",
+         sd.setups.size(), sd.timepoints.size(), sd.n5_path.c_str());
+
+  bsn5::Container n5(sd.n5_path);
+  bs_ctx *ctx = nullptr;
+  if (bs_ctx_create(&ctx, (int)args.getl("device", 0)) != BS_OK) {
+    fprintf(stderr, "error: %s\n", bs_last_error(nullptr));
+    return 1;
+  }
+
+  std::vector<bssd::StitchEntry> entries;
+  for (int tp : sd.timepoints) {
+    /* upload views of this timepoint on demand */
+    std::set<int> uploaded;
+    auto ensure_view = [&](int setup) -> bool {
+      if (uploaded.count(setup)) return true;
+      std::vector<uint16_t> vox;
+      std::vector<long long> dims;
+      if (!n5.read_volume_u16(bssd::SpimData::image_dataset(setup, tp),
+                              &vox, &dims)) {
+        fprintf(stderr, "cannot read view tp=%d setup=%d from %s\n", tp,
+                setup, sd.n5_path.c_str());
+        return false;
+      }
+      int64_t d[3] = {dims[0], dims[1], dims[2]};
+      if (bs_view_upload(ctx, setup, vox.data(), d) != BS_OK) {
+        fprintf(stderr, "upload failed: %s\n", bs_last_error(ctx));
+        return false;
+      }
+      uploaded.insert(setup);
+      return true;
+    };
+
+    /* enumerate overlapping pairs (transformed-bbox intersection, the
+     * filterNonOverlappingPairs step, ref :165) */
+    struct PairPlan {
+      int sa, sb;
+      bs_pair_desc pd;
+      M34 ma, mb;
+    };
+    std::vector<PairPlan> plans;
+    for (size_t i = 0; i < sd.setups.size(); ++i) {
+      for (size_t j = i + 1; j < sd.setups.size(); ++j) {
+        const auto &A = sd.setups[i], &B = sd.setups[j];
+        auto ra = sd.regs.find({tp, A.id}), rb = sd.regs.find({tp, B.id});
+        if (ra == sd.regs.end() || rb == sd.regs.end()) continue;
+        double loA[3], hiA[3], loB[3], hiB[3];
+        bscli::tbbox(ra->second, A.dims, loA, hiA);
+        bscli::tbbox(rb->second, B.dims, loB, hiB);
+        bool ovl = true;
+        double lo[3], hi[3];
+        for (int d = 0; d < 3; ++d) {
+          lo[d] = std::max(loA[d], loB[d]);
+          hi[d] = std::min(hiA[d], hiB[d]);
+          if (hi[d] <= lo[d]) ovl = false;
+        }
+        if (!ovl) continue;
+        if (!bscli::linear_equal(ra->second, rb->second)) {
+          fprintf(stderr,
+                  "skipping pair (%d,%d): non-translational registration "
+                  "parts differ (unsupported this round)\n",
+                  A.id, B.id);
+          continue;
+        }
+        /* world overlap -> local interval per view */
+        double L[9], Li[9], t[3];
+        bscli::decompose(ra->second, L, t);
+        if (!bscli::inv3(L, Li)) continue;
+        PairPlan pp;
+        pp.sa = A.id;
+        pp.sb = B.id;
+        pp.ma = ra->second;
+        pp.mb = rb->second;
+        pp.pd.view_a = A.id;
+        pp.pd.view_b = B.id;
+        bool valid = true;
+        for (int v = 0; v < 2; ++v) {
+          const auto &m = v == 0 ? ra->second : rb->second;
+          const long long *dims = v == 0 ? A.dims : B.dims;
+          double tv[3] = {m[3], m[7], m[11]};
+          int64_t *off = v == 0 ? pp.pd.off_a : pp.pd.off_b;
+          int64_t *size = v == 0 ? pp.pd.size_a : pp.pd.size_b;
+          for (int d = 0; d < 3; ++d) {
+            /* local = Li (world - t); Li diag-ish handled generally */
+            double l0 = 0, l1 = 0;
+            for (int k = 0; k < 3; ++k) {
+              l0 += Li[d * 3 + k] * (lo[k] - tv[k]);
+              l1 += Li[d * 3 + k] * (hi[k] - tv[k]);
+            }
+            double a = std::min(l0, l1), b = std::max(l0, l1);
+            int64_t o = (int64_t)std::floor(a);
+            int64_t e = (int64_t)std::ceil(b) + 1;
+            if (o < 0) o = 0;
+            if (e > dims[d]) e = dims[d];
+            if (e <= o) valid = false;
+            off[d] = o;
+            size[d] = e - o;
+          }
+        }
+        if (valid) plans.push_back(pp);
+      }
+    }
+    printf("timepoint %d: %zu overlapping pairs\n", tp, plans.size());
+    if (args.has("dryRun") || plans.empty()) continue;
+
+    std::vector<bs_pair_desc> pds;
+    for (auto &pp : plans) {
+      if (!ensure_view(pp.sa) || !ensure_view(pp.sb)) return 1;
+      pds.push_back(pp.pd);
+    }
+    bs_stitch_params prm{};
+    prm.ds[0] = (int)ds[0];
+    prm.ds[1] = (int)ds[1];
+    prm.ds[2] = (int)ds[2];
+    prm.peaks_to_check = (int)args.getl("peaksToCheck", 5);
+    prm.do_subpixel = args.has("disableSubpixelResolution") ? 0 : 1;
+    prm.min_overlap_ratio = args.getd("minOverlapRatio", 0.25);
+    std::vector<bs_shift_result> res(pds.size());
+    int rc = bs_stitch_batch(ctx, pds.data(), pds.size(), &prm, res.data());
+    if (rc != BS_OK) {
+      fprintf(stderr, "stitch failed: %s\n", bs_last_error(ctx));
+      return 1;
+    }
+    for (size_t k = 0; k < plans.size(); ++k) {
+      const auto &pp = plans[k];
+      const auto &r = res[k];
+      if (!r.valid) {
+        printf("pair (%d,%d): no shift found\n", pp.sa, pp.sb);
+        continue;
+      }
+      /* world shift = L x local shift; filters per
+       * FilteredStitchingResults (ref :369-380) */
+      double L[9], t[3];
+      bscli::decompose(pp.ma, L, t);
+      double ws[3];
+      for (int d = 0; d < 3; ++d)
+        ws[d] = L[d * 3 + 0] * r.shift[0] + L[d * 3 + 1] * r.shift[1] +
+                L[d * 3 + 2] * r.shift[2];
+      printf("pair (%d,%d): shift=(%.3f, %.3f, %.3f) r=%.4f\n", pp.sa,
+             pp.sb, ws[0], ws[1], ws[2], r.r);
+      if (r.r < minR || r.r > maxR) continue;
+      if (std::fabs(ws[0]) > maxShift[0] || std::fabs(ws[1]) > maxShift[1] ||
+          std::fabs(ws[2]) > maxShift[2])
+        continue;
+      if (std::sqrt(ws[0] * ws[0] + ws[1] * ws[1] + ws[2] * ws[2]) >
+          maxShiftTotal)
+        continue;
+      bssd::StitchEntry e;
+      e.views_a = {{tp, pp.sa}};
+      e.views_b = {{tp, pp.sb}};
+      for (int d = 0; d < 3; ++d) e.matrix[d * 4 + 3] = ws[d];
+      /* bbox: world overlap of the two transformed views */
+      double loA[3], hiA[3], loB[3], hiB[3];
+      bscli::tbbox(pp.ma, sd.setup(pp.sa)->dims, loA, hiA);
+      bscli::tbbox(pp.mb, sd.setup(pp.sb)->dims, loB, hiB);
+      for (int d = 0; d < 3; ++d) {
+        e.bbox_min[d] = std::max(loA[d], loB[d]);
+        e.bbox_max[d] = std::min(hiA[d], hiB[d]);
+      }
+      e.r = r.r;
+      e.hash = bssd::SpimData::calculate_hash(pp.ma, pp.mb);
+      entries.push_back(e);
+    }
+  }
+  bs_ctx_destroy(ctx);
+  if (!args.has("dryRun")) {
+    sd.set_stitching_results(entries);
+    if (!sd.save(sd.xml_path)) {
+      fprintf(stderr, "cannot write %s\n", sd.xml_path.c_str());
+      return 1;
+    }
+    printf("wrote %zu stitching results to %s\n", entries.size(),
+           sd.xml_path.c_str());
+  }
+  return 0;
+}

@@ -1,0 +1,123 @@
+"""Tiny pure-python N5 reader/writer used by tests to create input
+containers for the CLI tools and to verify their output — an independent
+implementation of the same public N5 spec as csrc/host/bs_n5.cpp (so the
+C++ and python sides cross-check each other)."""
+
+import gzip
+import json
+import os
+import struct
+
+import numpy as np
+
+_DTYPES = {"uint8": np.uint8, "uint16": np.uint16, "float32": np.float32}
+
+
+def write_dataset(root, name, arr_zyx, block_xyz, compression="gzip"):
+    """arr is (nz,ny,nx); stored as N5 3-D dataset with dims (nx,ny,nz)."""
+    nz, ny, nx = arr_zyx.shape
+    dt = {np.dtype(np.uint8): "uint8", np.dtype(np.uint16): "uint16",
+          np.dtype(np.float32): "float32"}[arr_zyx.dtype]
+    ds = os.path.join(root, name)
+    os.makedirs(ds, exist_ok=True)
+    comp = {"type": compression}
+    if compression == "gzip":
+        comp.update(level=-1, useZlib=False)
+    with open(os.path.join(ds, "attributes.json"), "w") as f:
+        json.dump({"dimensions": [nx, ny, nz],
+                   "blockSize": list(block_xyz),
+                   "dataType": dt, "compression": comp}, f)
+    bx, by, bz = block_xyz
+    for gz in range((nz + bz - 1) // bz):
+        for gy in range((ny + by - 1) // by):
+            for gx in range((nx + bx - 1) // bx):
+                blk = arr_zyx[gz * bz:(gz + 1) * bz, gy * by:(gy + 1) * by,
+                              gx * bx:(gx + 1) * bx]
+                cz, cy, cx = blk.shape
+                payload = blk.astype(blk.dtype.newbyteorder(">")).tobytes()
+                if compression == "gzip":
+                    payload = gzip.compress(payload)
+                hdr = struct.pack(">HH", 0, 3) + struct.pack(
+                    ">III", cx, cy, cz)
+                d = os.path.join(ds, str(gx), str(gy))
+                os.makedirs(d, exist_ok=True)
+                with open(os.path.join(d, str(gz)), "wb") as f:
+                    f.write(hdr + payload)
+
+
+def read_dataset(root, name):
+    ds = os.path.join(root, name)
+    with open(os.path.join(ds, "attributes.json")) as f:
+        attrs = json.load(f)
+    nx, ny, nz = attrs["dimensions"]
+    bx, by, bz = attrs["blockSize"]
+    dt = np.dtype(_DTYPES[attrs["dataType"]])
+    comp = attrs["compression"]["type"]
+    out = np.zeros((nz, ny, nx), dt)
+    for gz in range((nz + bz - 1) // bz):
+        for gy in range((ny + by - 1) // by):
+            for gx in range((nx + bx - 1) // bx):
+                p = os.path.join(ds, str(gx), str(gy), str(gz))
+                if not os.path.exists(p):
+                    continue
+                raw = open(p, "rb").read()
+                mode, nd = struct.unpack(">HH", raw[:4])
+                dims = struct.unpack(">" + "I" * nd, raw[4:4 + 4 * nd])
+                cx, cy, cz = dims
+                body = raw[4 + 4 * nd:]
+                if comp == "gzip":
+                    body = gzip.decompress(body)
+                blk = np.frombuffer(body, dt.newbyteorder(">")).reshape(
+                    cz, cy, cx).astype(dt)
+                out[gz * bz:gz * bz + cz, gy * by:gy * by + cy,
+                    gx * bx:gx * bx + cx] = blk
+    return out, attrs
+
+
+def root_attrs(root):
+    with open(os.path.join(root, "attributes.json")) as f:
+        return json.load(f)
+
+
+DATASET_XML = """<?xml version="1.0" encoding="UTF-8"?>
+<SpimData version="0.2">
+  <BasePath type="relative">.</BasePath>
+  <SequenceDescription>
+    <ImageLoader format="bdv.n5" version="1.0">
+      <n5 type="relative">{n5}</n5>
+    </ImageLoader>
+    <ViewSetups>
+{setups}
+    </ViewSetups>
+    <Timepoints type="pattern">
+      <integerpattern>0</integerpattern>
+    </Timepoints>
+  </SequenceDescription>
+  <ViewRegistrations>
+{regs}
+  </ViewRegistrations>
+</SpimData>
+"""
+
+
+def make_dataset_xml(path, n5_rel, setups):
+    """setups: list of dicts {id, dims (x,y,z), pos (x,y,z) translation}."""
+    s_xml, r_xml = "", ""
+    for s in setups:
+        s_xml += (
+            f"      <ViewSetup>\n        <id>{s['id']}</id>\n"
+            f"        <name>setup {s['id']}</name>\n"
+            f"        <size>{s['dims'][0]} {s['dims'][1]} {s['dims'][2]}"
+            f"</size>\n      </ViewSetup>\n"
+        )
+        p = s["pos"]
+        aff = f"1.0 0.0 0.0 {p[0]} 0.0 1.0 0.0 {p[1]} 0.0 0.0 1.0 {p[2]}"
+        r_xml += (
+            f"    <ViewRegistration timepoint=\"0\" setup=\"{s['id']}\">\n"
+            f"      <ViewTransform type=\"affine\">\n"
+            f"        <Name>Translation to Regular Grid</Name>\n"
+            f"        <affine>{aff}</affine>\n"
+            f"      </ViewTransform>\n    </ViewRegistration>\n"
+        )
+    with open(path, "w") as f:
+        f.write(DATASET_XML.format(n5=n5_rel, setups=s_xml, regs=r_xml))
