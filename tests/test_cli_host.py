@@ -108,13 +108,14 @@ def test_cli_stitching_end_to_end(tmp_path):
     rv = float(pr.find("Correlation").text)
     assert rv > 0.8
     # and it matches the oracle run on the same overlap intervals
+    # (ws == the library shift s == oracle shift; s = -err by construction)
     sub_a = a[:, :, 64 - 24:]
     sub_b = b[:, :, :24]
     ref = phasecorr.phase_correlation_shift(sub_a, sub_b, ds=(1, 1, 1),
                                             min_overlap_ratio=0.05)
-    assert abs(-ref["shift"][0] - ws[0]) < 1e-3
-    assert abs(-ref["shift"][1] - ws[1]) < 1e-3
-    assert abs(-ref["shift"][2] - ws[2]) < 1e-3
+    assert abs(ref["shift"][0] - ws[0]) < 1e-3
+    assert abs(ref["shift"][1] - ws[1]) < 1e-3
+    assert abs(ref["shift"][2] - ws[2]) < 1e-3
     assert rv == pytest.approx(ref["r"], abs=1e-9)
 
 
