@@ -53,9 +53,11 @@ def algorithmic_bytes_per_launch(size, pairs_kernel_dims):
 
 
 def load_traffic_sidecar(size):
-    """PMC-measured HBM bytes per launch for the dominant kernel, collected
-    by tools/profile_bench.sh into profiles/ (rocprofv3 --pmc FETCH_SIZE /
-    WRITE_SIZE in separate passes, gfx950 corrections applied there)."""
+    """PMC-measured HBM bytes per launch per kernel, collected by
+    tools/parse_rocprof.py into profiles/traffic.json (rocprofv3
+    --pmc FETCH_SIZE / WRITE_SIZE in separate passes — the TCC slots
+    cannot hold both; FETCH doubled per the gfx950 wide-coalesced-read
+    calibration, MI355X_MICROARCH.md §HBM)."""
     path = os.path.join(ROOT, "profiles", "traffic.json")
     if not os.path.exists(path):
         return None
@@ -151,7 +153,11 @@ def main():
 
     from bigstitcher_spark_amd import Context
 
-    ctx = Context(local)
+    # one rank per GPU; tolerate oversubscribed debug runs (2 ranks on a
+    # 1-GPU box) by wrapping into the available device count
+    ndev = torch.cuda.device_count() if torch.cuda.is_available() else 1
+    dev = local % max(1, ndev)
+    ctx = Context(dev)
     n_distinct = min(args.distinct, args.pairs)
     shifts = make_rank_pairs(ctx, rank, args.size, n_distinct, args.overlap)
     sz = (args.size, args.size, args.size)
@@ -170,14 +176,14 @@ def main():
     ctx.reset_stats()
 
     if torch.cuda.is_available():
-        torch.cuda.synchronize(local)
+        torch.cuda.synchronize(dev)
     if dist:
         dist.barrier()
     t0 = time.perf_counter()
     for _ in range(args.steps):
         res = step()
     if torch.cuda.is_available():
-        torch.cuda.synchronize(local)
+        torch.cuda.synchronize(dev)
     elapsed = time.perf_counter() - t0
     if dist:
         t = torch.tensor([elapsed], dtype=torch.float64)
@@ -209,8 +215,14 @@ def main():
         achieved = ab[dom] / (avg_ms * 1e-3) / 1e9  # GB/s
         traffic = None
         sidecar = load_traffic_sidecar(args.size)
-        if sidecar and sidecar.get("kernel") == dom:
-            traffic = sidecar.get("bytes_per_launch")
+        if sidecar and dom in sidecar.get("kernels", {}):
+            traffic = sidecar["kernels"][dom]
+        total_alg_bytes_per_pair = (
+            2 * ab["fft_x_fwd"] + 2 * ab["fft_y_fwd"] + 2 * ab["fft_z_fwd"]
+            + ab["fft_z_inv"] + ab["fft_y_inv"] + ab["fft_x_inv"]
+            + ab["peak"]
+        )
+        agg = total_alg_bytes_per_pair * value / 1e9  # GB/s whole-path
         roofline = {
             "bound": "hbm",
             "kernel": dom,
@@ -221,6 +233,13 @@ def main():
             "traffic": traffic,
             "algorithmic_bytes_per_launch": ab[dom],
             "avg_launch_ms": round(avg_ms, 4),
+            "note": (
+                "kernel avg_launch_ms includes cross-stream contention "
+                "(two pairs pipelined); aggregate is the whole-path "
+                "algorithmic rate"
+            ),
+            "aggregate_achieved": round(agg, 1),
+            "aggregate_frac": round(agg / HBM_PEAK_GBS, 4),
         }
 
     cpu = None
