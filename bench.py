@@ -144,11 +144,16 @@ def main():
     import torch
 
     dist = None
+    ndev0 = torch.cuda.device_count() if torch.cuda.is_available() else 0
     if world > 1:
         import torch.distributed as dist_mod
 
         dist = dist_mod
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        # RCCL needs one distinct device per rank; oversubscribed debug
+        # runs (world > device count) use gloo for the control plane
+        backend = "nccl" if ndev0 >= world else "gloo"
+        if ndev0:
+            torch.cuda.set_device(local % ndev0)
         dist.init_process_group(backend=backend)
 
     from bigstitcher_spark_amd import Context
