@@ -44,8 +44,9 @@
 
 #define LPB_X 4    /* lines per block, contiguous-line passes */
 #define TPL_X 64
-#define LPB_S 16   /* lines per block, strided passes */
-#define TPL_S 32   /* 512-thread blocks: 2x the waves of the v1 16x16 */
+#define LPB_S 8    /* lines per block, strided passes */
+#define TPL_S 64   /* 512-thread blocks; 35 KB LDS -> 4 WGs/CU (full
+                      32-wave occupancy) */
 #define PK_TX 128  /* peak-scan tile */
 #define PK_TY 8
 #define PK_TZ 8
@@ -177,10 +178,20 @@ __global__ __launch_bounds__(LPB_X *TPL_X) void k_fft_x_fwd(
     int z = active ? (int)(lid / in.my) : 0;
     const unsigned short *src =
         in.ptr + (in.oz + z) * in.sxy + (in.oy + y) * in.sx + in.ox;
-    for (int j = tl; j < h; j += TPL_X) {
-      float xa = (active && 2 * j < in.mx) ? (float)src[2 * j] : 0.0f;
-      float xb = (active && 2 * j + 1 < in.mx) ? (float)src[2 * j + 1] : 0.0f;
-      ld[brev_n(j, log2h)] = {xa, xb};
+    const bool al4 = active && ((size_t)src & 3) == 0 && 2 * h <= in.mx;
+    if (al4) { /* fast path: one dword per packed pair */
+      const unsigned *src32 = (const unsigned *)src;
+      for (int j = tl; j < h; j += TPL_X) {
+        unsigned w = src32[j];
+        ld[brev_n(j, log2h)] = {(float)(w & 0xFFFF), (float)(w >> 16)};
+      }
+    } else {
+      for (int j = tl; j < h; j += TPL_X) {
+        float xa = (active && 2 * j < in.mx) ? (float)src[2 * j] : 0.0f;
+        float xb =
+            (active && 2 * j + 1 < in.mx) ? (float)src[2 * j + 1] : 0.0f;
+        ld[brev_n(j, log2h)] = {xa, xb};
+      }
     }
     __syncthreads();
     fft_lds<1, TPL_X>(data, (long)line * h, h, log2h, tl, tw, +1);
@@ -226,7 +237,9 @@ __global__ __launch_bounds__(LPB_S *TPL_S) void k_fft_pass(
   /* I/O-phase mapping: pairs of adjacent lines -> one float4 (two
    * complex) per global transaction; 16-B aligned by construction
    * (gstride/estride are multiples of 16 f2, x even). */
-  const int pl = tid & 7, t2 = tid >> 3; /* pair-line, 64 elems/thread */
+  constexpr int NPAIR = LPB_S / 2;
+  constexpr int ESTR = (LPB_S * TPL_S) / NPAIR;
+  const int pl = tid & (NPAIR - 1), t2 = tid / NPAIR;
   const long nwg = (long)ngroups * nchunks;
   for (long wg = blockIdx.x; wg < nwg; wg += gridDim.x) {
     const int group = (int)(wg / nchunks);
@@ -234,7 +247,7 @@ __global__ __launch_bounds__(LPB_S *TPL_S) void k_fft_pass(
     const bool pair_ok = x2 + 1 < nlines;
     const long base2 = (long)group * gstride + x2;
     if (pair_ok) {
-      for (int e = t2; e < n; e += 64) {
+      for (int e = t2; e < n; e += ESTR) {
         float4 v = {0.0f, 0.0f, 0.0f, 0.0f};
         if (e < valid) {
           if (in2) {
@@ -265,7 +278,7 @@ __global__ __launch_bounds__(LPB_S *TPL_S) void k_fft_pass(
         const int x = x2 + l;
         const bool active = x < nlines;
         const long base = (long)group * gstride + x;
-        for (int e = t2; e < n; e += 64) {
+        for (int e = t2; e < n; e += ESTR) {
           f2 v = {0.0f, 0.0f};
           if (active && e < valid) {
             if (in2) {
@@ -287,7 +300,7 @@ __global__ __launch_bounds__(LPB_S *TPL_S) void k_fft_pass(
     __syncthreads();
     fft_lds<LPB_S, TPL_S>(data, (long)line, n, log2n, tl, tw, dir);
     if (pair_ok) {
-      for (int e = t2; e < n; e += 64)
+      for (int e = t2; e < n; e += ESTR)
         *(float4 *)&out[base2 + e * estride] =
             *(const float4 *)&data[(long)e * LPB_S + 2 * pl];
     } else {
@@ -295,7 +308,7 @@ __global__ __launch_bounds__(LPB_S *TPL_S) void k_fft_pass(
         const int x = x2 + l;
         if (x >= nlines) continue;
         const long base = (long)group * gstride + x;
-        for (int e = t2; e < n; e += 64)
+        for (int e = t2; e < n; e += ESTR)
           out[base + e * estride] = data[(long)e * LPB_S + 2 * pl + l];
       }
     }
