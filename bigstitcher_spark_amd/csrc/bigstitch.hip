@@ -239,56 +239,20 @@ __global__ __launch_bounds__(LPB_S *TPL_S) void k_fft_pass(
    * (gstride/estride are multiples of 16 f2, x even). */
   constexpr int NPAIR = LPB_S / 2;
   constexpr int ESTR = (LPB_S * TPL_S) / NPAIR;
-  constexpr int MAXE = 512 / ESTR; /* prefetch depth: covers n<=512;
-                                      larger n loads the tail in the
-                                      scatter (register budget) */
   const int pl = tid & (NPAIR - 1), t2 = tid / NPAIR;
   const long nwg = (long)ngroups * nchunks;
-  const int ne = (n + ESTR - 1) / ESTR;
-
-  /* T14 async-stage split: the fast-path (full pair) loads of group g+1
-   * are ISSUED before the FFT of group g (plain register loads survive
-   * __syncthreads); the LDS scatter happens at the top of the next
-   * iteration. Statically-unrolled register arrays (rule 20: runtime
-   * indexing would spill to scratch). */
-  float4 ra[MAXE], rb[MAXE];
-  auto issue_loads = [&](long wg, bool &pok, long &pbase2) {
-    const int group = (int)(wg / nchunks);
-    const int x2 = (int)(wg % nchunks) * LPB_S + 2 * pl;
-    pok = x2 + 1 < nlines;
-    pbase2 = (long)group * gstride + x2;
-    if (!pok) return;
-#pragma unroll
-    for (int i = 0; i < MAXE; ++i) {
-      int e = t2 + i * ESTR;
-      if (i < ne && e < valid) {
-        ra[i] = *(const float4 *)&in[pbase2 + e * estride];
-        if (in2) rb[i] = *(const float4 *)&in2[pbase2 + e * estride];
-      }
-    }
-  };
-  bool pref_ok = false;
-  long pref_base2 = 0;
-  if (blockIdx.x < nwg) issue_loads(blockIdx.x, pref_ok, pref_base2);
-
   for (long wg = blockIdx.x; wg < nwg; wg += gridDim.x) {
     const int group = (int)(wg / nchunks);
     const int x2 = (int)(wg % nchunks) * LPB_S + 2 * pl;
-    const bool pair_ok = pref_ok;
-    const long base2 = pref_base2;
+    const bool pair_ok = x2 + 1 < nlines;
+    const long base2 = (long)group * gstride + x2;
     if (pair_ok) {
-#pragma unroll
-      for (int i = 0; i < 8; ++i) {
-        int e = t2 + i * ESTR;
-        if (i >= ne || e >= n) continue;
+      for (int e = t2; e < n; e += ESTR) {
         float4 v = {0.0f, 0.0f, 0.0f, 0.0f};
         if (e < valid) {
           if (in2) {
-            const float4 va =
-                i < MAXE ? ra[i] : *(const float4 *)&in[base2 + e * estride];
-            const float4 vb =
-                i < MAXE ? rb[i]
-                         : *(const float4 *)&in2[base2 + e * estride];
+            const float4 va = *(const float4 *)&in[base2 + e * estride];
+            const float4 vb = *(const float4 *)&in2[base2 + e * estride];
             f2 q0 = conjmul({va.x, va.y}, {vb.x, vb.y});
             f2 q1 = conjmul({va.z, va.w}, {vb.z, vb.w});
             float m0 = q0.x * q0.x + q0.y * q0.y;
@@ -304,8 +268,7 @@ __global__ __launch_bounds__(LPB_S *TPL_S) void k_fft_pass(
               v.w = q1.y * s;
             }
           } else {
-            v = i < MAXE ? ra[i]
-                         : *(const float4 *)&in[base2 + e * estride];
+            v = *(const float4 *)&in[base2 + e * estride];
           }
         }
         *(float4 *)&data[(long)brev_n(e, log2n) * LPB_S + 2 * pl] = v;
@@ -335,8 +298,6 @@ __global__ __launch_bounds__(LPB_S *TPL_S) void k_fft_pass(
       }
     }
     __syncthreads();
-    if (wg + gridDim.x < nwg)
-      issue_loads(wg + gridDim.x, pref_ok, pref_base2);
     fft_lds<LPB_S, TPL_S>(data, (long)line, n, log2n, tl, tw, dir);
     if (pair_ok) {
       for (int e = t2; e < n; e += ESTR)
