@@ -541,10 +541,7 @@ __device__ __forceinline__ u64 wave_sum_u64(u64 v) {
 __global__ __launch_bounds__(256) void k_rtest(
     bs_region a, bs_region b, const bs_cand *cands, u64 *sums /* [nc][5] */) {
   __shared__ u64 ws[4][5];
-  /* grid is (candidate, row-chunk): dispatch walks candidates fastest,
-   * so the same A rows are processed by all candidates back-to-back and
-   * the shared A reads hit L2 instead of re-fetching from HBM */
-  const bs_cand c = cands[blockIdx.x];
+  const bs_cand c = cands[blockIdx.y];
   long nrows = (long)c.ny * c.nz;
   u64 pa = 0, pb = 0, paa = 0, pbb = 0, pab = 0;
   auto rowptr_a = [&](long row) {
@@ -557,12 +554,12 @@ __global__ __launch_bounds__(256) void k_rtest(
     return b.ptr + (b.oz + c.loz + c.sz + z) * b.sxy +
            (b.oy + c.loy + c.sy + y) * b.sx + b.ox + c.lox + c.sx;
   };
-  long row = blockIdx.y;
+  long row = blockIdx.x;
   /* two rows in flight doubles outstanding loads (latency-bound kernel) */
-  for (; row + gridDim.y < nrows; row += 2L * gridDim.y) {
+  for (; row + gridDim.x < nrows; row += 2L * gridDim.x) {
     const unsigned short *a0 = rowptr_a(row), *b0 = rowptr_b(row);
-    const unsigned short *a1 = rowptr_a(row + gridDim.y);
-    const unsigned short *b1 = rowptr_b(row + gridDim.y);
+    const unsigned short *a1 = rowptr_a(row + gridDim.x);
+    const unsigned short *b1 = rowptr_b(row + gridDim.x);
     for (int x = threadIdx.x; x < c.nx; x += 256) {
       u64 av0 = a0[x], bv0 = b0[x], av1 = a1[x], bv1 = b1[x];
       pa += av0 + av1;
@@ -572,7 +569,7 @@ __global__ __launch_bounds__(256) void k_rtest(
       pab += av0 * bv0 + av1 * bv1;
     }
   }
-  for (; row < nrows; row += gridDim.y) {
+  for (; row < nrows; row += gridDim.x) {
     const unsigned short *a0 = rowptr_a(row), *b0 = rowptr_b(row);
     for (int x = threadIdx.x; x < c.nx; x += 256) {
       u64 av = a0[x], bv = b0[x];
@@ -590,7 +587,7 @@ __global__ __launch_bounds__(256) void k_rtest(
   if (threadIdx.x < 5) {
     u64 s = ws[0][threadIdx.x] + ws[1][threadIdx.x] + ws[2][threadIdx.x] +
             ws[3][threadIdx.x];
-    atomicAdd(&sums[(long)blockIdx.x * 5 + threadIdx.x], s);
+    atomicAdd(&sums[(long)blockIdx.y * 5 + threadIdx.x], s);
   }
 }
 
@@ -1340,8 +1337,8 @@ static int stitch_phaseB(bs_ctx *c, bs_slot *sl,
         maxrows = std::max(maxrows, (long)h.gc.ny * h.gc.nz);
       bs_tim tt(c, BS_K_CORR, sl->stream);
       hipLaunchKernelGGL(k_rtest,
-                         dim3((unsigned)gc.size(),
-                              (unsigned)std::min(2048L, maxrows)),
+                         dim3((unsigned)std::min(2048L, maxrows),
+                              (unsigned)gc.size()),
                          dim3(256), 0, sl->stream, sl->reg[0], sl->reg[1],
                          sl->dcands, sl->dsums);
     }
