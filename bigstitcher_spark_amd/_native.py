@@ -63,7 +63,7 @@ class _FuseParams(C.Structure):
 BS_K_NAMES = [
     "downsample", "fft_x_fwd", "fft_y_fwd", "fft_z_fwd", "fft_z_inv",
     "fft_y_inv", "fft_x_inv", "peak", "peak_merge", "corr", "subpix",
-    "fuse", "synth",
+    "fuse", "synth", "pyramid",
 ]
 _NK = len(BS_K_NAMES)
 
@@ -113,6 +113,12 @@ def load_lib():
                                    C.c_size_t, C.POINTER(C.c_int32),
                                    C.POINTER(C.c_int64),
                                    C.POINTER(_FuseParams),
+                                   C.POINTER(C.c_void_p)]
+    lib.bs_fuse_volume.argtypes = [C.c_void_p, C.POINTER(_FuseView),
+                                   C.c_size_t, C.c_int64 * 3, C.c_int64 * 3,
+                                   C.POINTER(_FuseParams), C.c_int32,
+                                   C.POINTER(C.c_int32),
+                                   C.POINTER(C.c_int64),
                                    C.POINTER(C.c_void_p)]
     lib.bs_get_stats.argtypes = [C.c_void_p, C.POINTER(_Stats)]
     lib.bs_reset_stats.argtypes = [C.c_void_p]
@@ -262,6 +268,45 @@ class Context:
             self._lib.bs_fuse_blocks(self._h, cv, nv, cb, nb, cidx, offs,
                                      C.byref(prm), outptrs),
             "bs_fuse_blocks",
+        )
+        return outs
+
+    def fuse_volume(self, views, vol_min, vol_dims, downsamplings=None,
+                    fusion_type=FUSION_AVG_BLEND, out_dtype=np.float32,
+                    min_intensity=0.0, max_intensity=65535.0):
+        """Whole-volume fusion + pyramid. downsamplings: list of (dx,dy,dz)
+        absolute factors per level (level 0 must be (1,1,1)). Returns a
+        list of (nz,ny,nx) arrays, one per level."""
+        if downsamplings is None:
+            downsamplings = [(1, 1, 1)]
+        nv, nl = len(views), len(downsamplings)
+        cv = (_FuseView * nv)()
+        for i, v in enumerate(views):
+            cv[i].view_id = v["view_id"]
+            aff = np.asarray(v["affine"], np.float64).reshape(12)
+            for d in range(12):
+                cv[i].affine[d] = aff[d]
+            for d in range(3):
+                cv[i].blend_border[d] = v.get("border", (0, 0, 0))[d]
+                cv[i].blend_range[d] = v.get("range", (40, 40, 40))[d]
+        dt = np.dtype(out_dtype)
+        prm = _FuseParams(fusion_type, OUT_DTYPES[dt], min_intensity,
+                          max_intensity, 1)
+        vmin = (C.c_int64 * 3)(*[int(x) for x in vol_min])
+        vdim = (C.c_int64 * 3)(*[int(x) for x in vol_dims])
+        cds = (C.c_int32 * (3 * nl))(
+            *[int(f) for lvl in downsamplings for f in lvl])
+        ldims = (C.c_int64 * (3 * nl))()
+        outs, outptrs = [], (C.c_void_p * nl)()
+        for l, lvl in enumerate(downsamplings):
+            d = [(int(vol_dims[k]) + lvl[k] - 1) // lvl[k] for k in range(3)]
+            a = np.empty((d[2], d[1], d[0]), dt)
+            outs.append(a)
+            outptrs[l] = a.ctypes.data
+        self._check(
+            self._lib.bs_fuse_volume(self._h, cv, nv, vmin, vdim,
+                                     C.byref(prm), nl, cds, ldims, outptrs),
+            "bs_fuse_volume",
         )
         return outs
 

@@ -615,7 +615,8 @@ __device__ __forceinline__ float blend_w(float p, int dim, float border,
 __global__ __launch_bounds__(256) void k_fuse(
     const bs_dev_view *views, const int *vidx, int nv, long bmx, long bmy,
     long bmz, int bx, int by, int bz, int ftype, int dtype, float minI,
-    float invRange /* type_max/(maxI-minI) */, void *out) {
+    float invRange /* type_max/(maxI-minI) */, void *out, long out_off,
+    long out_row, long out_slice) {
   /* stage the (culled, small) per-block view table in LDS once */
   __shared__ bs_dev_view sv[BS_MAX_BLK_VIEWS];
   int nvs = min(nv, BS_MAX_BLK_VIEWS);
@@ -629,7 +630,7 @@ __global__ __launch_bounds__(256) void k_fuse(
   for (long row = blockIdx.x; row < nrows; row += gridDim.x) {
    int y = (int)(row % by), z = (int)(row / by);
    for (int x = threadIdx.x; x < bx; x += blockDim.x) {
-    long i = row * bx + x;
+    long i = out_off + (long)z * out_slice + (long)y * out_row + x;
     float wx = (float)(bmx + x), wy = (float)(bmy + y), wz = (float)(bmz + z);
     float sum_wv = 0.0f, sum_w = 0.0f, vmax = 0.0f;
     bool any = false;
@@ -693,6 +694,37 @@ __global__ __launch_bounds__(256) void k_fuse(
             (unsigned char)fminf(fmaxf(q, 0.0f), 255.0f);
     }
    }
+  }
+}
+
+/* ---- multi-resolution pyramid: box-mean downsample ---- */
+
+template <typename T>
+__global__ __launch_bounds__(256) void k_pyr(
+    const T *src, T *dst, int sxd, int syd, int szd, int dxd, int dyd,
+    int dzd, int rx, int ry, int rz) {
+  long nrows = (long)dyd * dzd;
+  for (long row = blockIdx.x; row < nrows; row += gridDim.x) {
+    int y = (int)(row % dyd), z = (int)(row / dyd);
+    int y0 = y * ry, y1 = min(y0 + ry, syd);
+    int z0 = z * rz, z1 = min(z0 + rz, szd);
+    T *drow = dst + row * dxd;
+    for (int x = threadIdx.x; x < dxd; x += blockDim.x) {
+      int x0 = x * rx, x1 = min(x0 + rx, sxd);
+      float sum = 0.0f;
+      int cnt = 0;
+      for (int zz = z0; zz < z1; ++zz)
+        for (int yy = y0; yy < y1; ++yy)
+          for (int xx = x0; xx < x1; ++xx) {
+            sum += (float)src[((long)zz * syd + yy) * sxd + xx];
+            ++cnt;
+          }
+      float m = sum / (float)cnt;
+      if (sizeof(T) == 4)
+        drow[x] = (T)m; /* float32 */
+      else
+        drow[x] = (T)__float2int_rn(m); /* [PIN-PYR] round to nearest */
+    }
   }
 }
 
@@ -813,6 +845,8 @@ struct bs_ctx {
   size_t dvidx_cap = 0;
   float *dblobs = nullptr;
   size_t dblobs_cap = 0;
+  void *hstage[2] = {nullptr, nullptr}; /* pinned D2H staging (64 MB) */
+  hipEvent_t stage_ev[2] = {nullptr, nullptr};
   float *dbg_pcm = nullptr; /* last pair's PCM (points into a slot) */
   long dbg_px = 0, dbg_py = 0, dbg_pz = 0;
   /* stats */
@@ -938,6 +972,10 @@ extern "C" void bs_ctx_destroy(bs_ctx *c) {
     if (sl.done) (void)hipEventDestroy(sl.done);
     if (sl.stream) (void)hipStreamDestroy(sl.stream);
   }
+  (void)hipHostFree(c->hstage[0]);
+  (void)hipHostFree(c->hstage[1]);
+  if (c->stage_ev[0]) (void)hipEventDestroy(c->stage_ev[0]);
+  if (c->stage_ev[1]) (void)hipEventDestroy(c->stage_ev[1]);
   (void)hipFree(c->synth_acc);
   (void)hipFree(c->fuse_out);
   (void)hipFree(c->dviews);
@@ -1491,6 +1529,69 @@ static int invert34(const double *m, float *inv) {
   return BS_OK;
 }
 
+static int build_dev_views(bs_ctx *c, const bs_fuse_view *views,
+                           size_t nviews, const int64_t vol_min[3],
+                           std::vector<bs_dev_view> *dv) {
+  dv->resize(nviews);
+  for (size_t i = 0; i < nviews; ++i) {
+    auto it = c->views.find(views[i].view_id);
+    if (it == c->views.end()) {
+      c->err = "fusion view not uploaded";
+      return BS_ENOVIEW;
+    }
+    bs_dev_view &v = (*dv)[i];
+    v.ptr = it->second.dptr;
+    v.nx = (int)it->second.dims[0];
+    v.ny = (int)it->second.dims[1];
+    v.nz = (int)it->second.dims[2];
+    double adj[12];
+    for (int k = 0; k < 12; ++k) adj[k] = views[i].affine[k];
+    if (vol_min) {
+      adj[3] -= (double)vol_min[0];
+      adj[7] -= (double)vol_min[1];
+      adj[11] -= (double)vol_min[2];
+    }
+    if (invert34(adj, v.inv) != BS_OK) {
+      c->err = "singular view affine";
+      return BS_EINVAL;
+    }
+    for (int d = 0; d < 3; ++d) {
+      v.border[d] = views[i].blend_border[d];
+      v.range[d] = views[i].blend_range[d];
+    }
+  }
+  return BS_OK;
+}
+
+/* transformed-bbox culling against a block (OverlappingViews semantics,
+ * reference fusion/OverlappingViews.java:28-47, +2 px guard), in the
+ * (possibly vol_min-shifted) world frame. */
+static bool view_overlaps_block(const bs_fuse_view &fv,
+                                const bs_dev_view &dv,
+                                const int64_t vmin[3], const long bmin[3],
+                                const long bsize[3]) {
+  double lo[3] = {1e300, 1e300, 1e300}, hi[3] = {-1e300, -1e300, -1e300};
+  long long dims[3] = {dv.nx, dv.ny, dv.nz};
+  for (int cz = 0; cz < 2; ++cz)
+    for (int cy = 0; cy < 2; ++cy)
+      for (int cx = 0; cx < 2; ++cx) {
+        double px = cx ? (double)(dims[0] - 1) : 0.0;
+        double py = cy ? (double)(dims[1] - 1) : 0.0;
+        double pz = cz ? (double)(dims[2] - 1) : 0.0;
+        for (int r = 0; r < 3; ++r) {
+          const double *m = fv.affine;
+          double w = m[r * 4 + 0] * px + m[r * 4 + 1] * py +
+                     m[r * 4 + 2] * pz + m[r * 4 + 3] -
+                     (vmin ? (double)vmin[r] : 0.0);
+          if (w < lo[r]) lo[r] = w;
+          if (w > hi[r]) hi[r] = w;
+        }
+      }
+  for (int d = 0; d < 3; ++d)
+    if (hi[d] + 2 < bmin[d] || lo[d] - 2 > bmin[d] + bsize[d]) return false;
+  return true;
+}
+
 extern "C" int bs_fuse_blocks(bs_ctx *c, const bs_fuse_view *views,
                               size_t nviews, const bs_block_desc *blocks,
                               size_t nb, const int32_t *view_idx_per_block,
@@ -1505,29 +1606,11 @@ extern "C" int bs_fuse_blocks(bs_ctx *c, const bs_fuse_view *views,
   }
   std::lock_guard<std::mutex> g(c->mu);
   CHK(c, hipSetDevice(c->dev));
-  /* device view table */
-  std::vector<bs_dev_view> dv(nviews);
-  for (size_t i = 0; i < nviews; ++i) {
-    auto it = c->views.find(views[i].view_id);
-    if (it == c->views.end()) {
-      c->err = "fusion view not uploaded";
-      return BS_ENOVIEW;
-    }
-    dv[i].ptr = it->second.dptr;
-    dv[i].nx = (int)it->second.dims[0];
-    dv[i].ny = (int)it->second.dims[1];
-    dv[i].nz = (int)it->second.dims[2];
-    if (invert34(views[i].affine, dv[i].inv) != BS_OK) {
-      c->err = "singular view affine";
-      return BS_EINVAL;
-    }
-    for (int d = 0; d < 3; ++d) {
-      dv[i].border[d] = views[i].blend_border[d];
-      dv[i].range[d] = views[i].blend_range[d];
-    }
-  }
-  int rc = ensure_dev(c, (void **)&c->dviews, &c->dviews_cap,
-                      nviews * sizeof(bs_dev_view));
+  std::vector<bs_dev_view> dv;
+  int rc = build_dev_views(c, views, nviews, nullptr, &dv);
+  if (rc) return rc;
+  rc = ensure_dev(c, (void **)&c->dviews, &c->dviews_cap,
+                  nviews * sizeof(bs_dev_view));
   if (rc) return rc;
   CHK(c, hipMemcpyAsync(c->dviews, dv.data(), nviews * sizeof(bs_dev_view),
                         hipMemcpyHostToDevice, c->stream));
@@ -1563,7 +1646,8 @@ extern "C" int bs_fuse_blocks(bs_ctx *c, const bs_fuse_view *views,
                          bd.min[1], bd.min[2], (int)bd.size[0],
                          (int)bd.size[1], (int)bd.size[2], prm->fusion_type,
                          prm->out_dtype, (float)prm->min_intensity, invRange,
-                         c->fuse_out);
+                         c->fuse_out, 0L, (long)bd.size[0],
+                         (long)bd.size[0] * bd.size[1]);
     }
     CHK(c, hipMemcpyAsync(out_blocks[i], c->fuse_out, nvox * esz,
                           hipMemcpyDeviceToHost, c->stream));
@@ -1571,5 +1655,194 @@ extern "C" int bs_fuse_blocks(bs_ctx *c, const bs_fuse_view *views,
   CHK(c, hipStreamSynchronize(c->stream));
   c->stats.blocks += (long long)nb;
   flush_stats(c);
+  return BS_OK;
+}
+
+/* ---- volume-mode fusion + pyramid ---- */
+
+#define BS_STAGE_BYTES (64L << 20)
+
+/* pinned double-buffered device->host copy (pageable-dest hipMemcpy is
+ * ~3x slower and blocks; this overlaps the PCIe copy with the host-side
+ * memcpy out of the staging buffer). */
+static int staged_d2h(bs_ctx *c, const void *dsrc, void *hdst,
+                      size_t bytes) {
+  for (int i = 0; i < 2; ++i) {
+    if (!c->hstage[i]) {
+      if (hipHostMalloc(&c->hstage[i], BS_STAGE_BYTES) != hipSuccess) {
+        c->err = "pinned staging alloc failed";
+        return BS_ENOMEM;
+      }
+      CHK(c, hipEventCreateWithFlags(&c->stage_ev[i],
+                                     hipEventDisableTiming));
+    }
+  }
+  size_t nchunks = (bytes + BS_STAGE_BYTES - 1) / BS_STAGE_BYTES;
+  for (size_t k = 0; k < nchunks + 1; ++k) {
+    if (k < nchunks) {
+      size_t off = k * BS_STAGE_BYTES;
+      size_t len = std::min((size_t)BS_STAGE_BYTES, bytes - off);
+      CHK(c, hipMemcpyAsync(c->hstage[k & 1], (const char *)dsrc + off, len,
+                            hipMemcpyDeviceToHost, c->stream));
+      CHK(c, hipEventRecord(c->stage_ev[k & 1], c->stream));
+    }
+    if (k > 0) {
+      size_t off = (k - 1) * BS_STAGE_BYTES;
+      size_t len = std::min((size_t)BS_STAGE_BYTES, bytes - off);
+      CHK(c, hipEventSynchronize(c->stage_ev[(k - 1) & 1]));
+      memcpy((char *)hdst + off, c->hstage[(k - 1) & 1], len);
+    }
+  }
+  return BS_OK;
+}
+
+extern "C" int bs_fuse_volume(bs_ctx *c, const bs_fuse_view *views,
+                              size_t nviews, const int64_t vol_min[3],
+                              const int64_t vol_dims[3],
+                              const bs_fuse_params *prm, int32_t nlevels,
+                              const int32_t *abs_ds,
+                              int64_t *level_dims_out,
+                              void **level_buffers) {
+  if (!c || !views || !vol_min || !vol_dims || !prm || nlevels < 1 ||
+      !abs_ds || !level_buffers || nviews == 0)
+    return BS_EINVAL;
+  if (prm->interp != 1) {
+    c->err = "only interp=1 supported";
+    return BS_EUNSUP;
+  }
+  for (int d = 0; d < 3; ++d)
+    if (abs_ds[d] != 1) {
+      c->err = "level 0 downsampling must be 1,1,1";
+      return BS_EINVAL;
+    }
+  std::lock_guard<std::mutex> g(c->mu);
+  CHK(c, hipSetDevice(c->dev));
+  std::vector<bs_dev_view> dv;
+  int rc = build_dev_views(c, views, nviews, vol_min, &dv);
+  if (rc) return rc;
+  rc = ensure_dev(c, (void **)&c->dviews, &c->dviews_cap,
+                  nviews * sizeof(bs_dev_view));
+  if (rc) return rc;
+  CHK(c, hipMemcpyAsync(c->dviews, dv.data(), nviews * sizeof(bs_dev_view),
+                        hipMemcpyHostToDevice, c->stream));
+  int esz = prm->out_dtype == BS_OUT_FLOAT32 ? 4
+            : prm->out_dtype == BS_OUT_UINT16 ? 2 : 1;
+  double denom = prm->max_intensity - prm->min_intensity;
+  float invRange =
+      (float)((prm->out_dtype == BS_OUT_UINT8 ? 255.0 : 65535.0) /
+              (denom != 0.0 ? denom : 1.0));
+  /* level dims + device buffers */
+  std::vector<std::array<long, 3>> ldims(nlevels);
+  std::vector<void *> dlvl(nlevels, nullptr);
+  auto cleanup = [&]() {
+    for (auto p : dlvl)
+      if (p) (void)hipFree(p);
+  };
+  for (int l = 0; l < nlevels; ++l) {
+    for (int d = 0; d < 3; ++d) {
+      int f = abs_ds[l * 3 + d];
+      if (f < 1 || (l > 0 && f % abs_ds[(l - 1) * 3 + d] != 0)) {
+        c->err = "bad downsampling ladder";
+        return BS_EINVAL;
+      }
+      ldims[l][d] = (vol_dims[d] + f - 1) / f; /* [PIN-PYR] ceil */
+      if (level_dims_out) level_dims_out[l * 3 + d] = ldims[l][d];
+    }
+    size_t bytes = (size_t)ldims[l][0] * ldims[l][1] * ldims[l][2] * esz;
+    if (hipMalloc(&dlvl[l], bytes) != hipSuccess) {
+      cleanup();
+      c->err = "device volume alloc failed (volume too large this round)";
+      return BS_ENOMEM;
+    }
+  }
+  /* fusion of level 0 over an internal 256x128x128 grid with culling */
+  const long FB[3] = {256, 128, 128};
+  std::vector<int32_t> flat;
+  std::vector<std::array<long, 6>> fblocks; /* min[3], size[3] */
+  std::vector<int64_t> offs;
+  for (long z0 = 0; z0 < ldims[0][2]; z0 += FB[2])
+    for (long y0 = 0; y0 < ldims[0][1]; y0 += FB[1])
+      for (long x0 = 0; x0 < ldims[0][0]; x0 += FB[0]) {
+        long bmin[3] = {x0, y0, z0};
+        long bsz[3] = {std::min(FB[0], ldims[0][0] - x0),
+                       std::min(FB[1], ldims[0][1] - y0),
+                       std::min(FB[2], ldims[0][2] - z0)};
+        offs.push_back((int64_t)flat.size());
+        for (size_t v = 0; v < nviews; ++v)
+          if (view_overlaps_block(views[v], dv[v], vol_min, bmin, bsz))
+            flat.push_back((int32_t)v);
+        fblocks.push_back({bmin[0], bmin[1], bmin[2], bsz[0], bsz[1],
+                           bsz[2]});
+      }
+  offs.push_back((int64_t)flat.size());
+  rc = ensure_dev(c, (void **)&c->dvidx, &c->dvidx_cap,
+                  std::max((size_t)1, flat.size()) * sizeof(int32_t));
+  if (rc) {
+    cleanup();
+    return rc;
+  }
+  if (!flat.empty())
+    CHK(c, hipMemcpyAsync(c->dvidx, flat.data(),
+                          flat.size() * sizeof(int32_t),
+                          hipMemcpyHostToDevice, c->stream));
+  const long vrow = ldims[0][0];
+  const long vslice = ldims[0][0] * ldims[0][1];
+  for (size_t b = 0; b < fblocks.size(); ++b) {
+    auto &fb = fblocks[b];
+    int nvb = (int)(offs[b + 1] - offs[b]);
+    long nrows_f = fb[4] * fb[5];
+    bs_tim tt(c, BS_K_FUSE, c->stream);
+    hipLaunchKernelGGL(k_fuse, dim3(std::min(4096L, nrows_f)), dim3(256),
+                       0, c->stream, c->dviews, c->dvidx + offs[b], nvb,
+                       fb[0], fb[1], fb[2], (int)fb[3], (int)fb[4],
+                       (int)fb[5], prm->fusion_type, prm->out_dtype,
+                       (float)prm->min_intensity, invRange, dlvl[0],
+                       (long)fb[2] * vslice + fb[1] * vrow + fb[0], vrow,
+                       vslice);
+  }
+  /* pyramid levels */
+  for (int l = 1; l < nlevels; ++l) {
+    int rx = abs_ds[l * 3 + 0] / abs_ds[(l - 1) * 3 + 0];
+    int ry = abs_ds[l * 3 + 1] / abs_ds[(l - 1) * 3 + 1];
+    int rz = abs_ds[l * 3 + 2] / abs_ds[(l - 1) * 3 + 2];
+    long nrows = ldims[l][1] * ldims[l][2];
+    bs_tim tt(c, BS_K_PYRAMID, c->stream);
+    if (esz == 4)
+      hipLaunchKernelGGL(k_pyr<float>, dim3(std::min(4096L, nrows)),
+                         dim3(256), 0, c->stream, (const float *)dlvl[l - 1],
+                         (float *)dlvl[l], (int)ldims[l - 1][0],
+                         (int)ldims[l - 1][1], (int)ldims[l - 1][2],
+                         (int)ldims[l][0], (int)ldims[l][1],
+                         (int)ldims[l][2], rx, ry, rz);
+    else if (esz == 2)
+      hipLaunchKernelGGL(k_pyr<unsigned short>,
+                         dim3(std::min(4096L, nrows)), dim3(256), 0,
+                         c->stream, (const unsigned short *)dlvl[l - 1],
+                         (unsigned short *)dlvl[l], (int)ldims[l - 1][0],
+                         (int)ldims[l - 1][1], (int)ldims[l - 1][2],
+                         (int)ldims[l][0], (int)ldims[l][1],
+                         (int)ldims[l][2], rx, ry, rz);
+    else
+      hipLaunchKernelGGL(k_pyr<unsigned char>,
+                         dim3(std::min(4096L, nrows)), dim3(256), 0,
+                         c->stream, (const unsigned char *)dlvl[l - 1],
+                         (unsigned char *)dlvl[l], (int)ldims[l - 1][0],
+                         (int)ldims[l - 1][1], (int)ldims[l - 1][2],
+                         (int)ldims[l][0], (int)ldims[l][1],
+                         (int)ldims[l][2], rx, ry, rz);
+  }
+  /* staged D2H of every level */
+  for (int l = 0; l < nlevels; ++l) {
+    size_t bytes = (size_t)ldims[l][0] * ldims[l][1] * ldims[l][2] * esz;
+    rc = staged_d2h(c, dlvl[l], level_buffers[l], bytes);
+    if (rc) {
+      cleanup();
+      return rc;
+    }
+  }
+  CHK(c, hipStreamSynchronize(c->stream));
+  c->stats.blocks += (long long)fblocks.size();
+  flush_stats(c);
+  cleanup();
   return BS_OK;
 }

@@ -25,7 +25,8 @@ int main(int argc, char **argv) {
             "usage: create-fusion-container -x dataset.xml -o out.n5 "
             "[--blockSize 128,128,128] [--dataType UINT16|UINT8|FLOAT32] "
             "[--minIntensity v --maxIntensity v] [--bbMin x,y,z --bbMax "
-            "x,y,z] [--compression GZIP|RAW]\n");
+            "x,y,z] [--compression GZIP|RAW] "
+            "[--downsamplings \"1,1,1;2,2,2;4,4,4\"]\n");
     return 2;
   }
   bssd::SpimData sd;
@@ -96,30 +97,57 @@ int main(int argc, char **argv) {
     set("MinIntensity", bsj::Value::mknum(args.getd("minIntensity", 0)));
     set("MaxIntensity", bsj::Value::mknum(args.getd("maxIntensity", 65535)));
   }
+  /* pyramid ladder (reference
+   * CreateFusionContainer.java:260-273 estimates it — here explicit via
+   * --downsamplings, default s0 only) */
+  std::vector<std::array<long long, 3>> ladder;
+  {
+    std::string spec = args.get("downsamplings", "1,1,1");
+    std::string cur;
+    for (char ch2 : spec + ";") {
+      if (ch2 == ';') {
+        auto f = bscli::parse_ints(cur);
+        if (f.size() == 3) ladder.push_back({f[0], f[1], f[2]});
+        cur.clear();
+      } else {
+        cur += ch2;
+      }
+    }
+    if (ladder.empty() || ladder[0][0] != 1 || ladder[0][1] != 1 ||
+        ladder[0][2] != 1) {
+      fprintf(stderr, "--downsamplings must start with 1,1,1\n");
+      return 2;
+    }
+  }
   auto mri_all = bsj::Value::mkarr();
   for (int t = 0; t < numTp; ++t)
     for (int ch = 0; ch < numCh; ++ch) {
-      char dsname[64];
-      snprintf(dsname, sizeof dsname, "ch%dtp%d/s0", ch, t);
-      bsn5::DatasetAttrs da;
-      da.dims = {dims[0], dims[1], dims[2]};
-      da.block = {(int)bs[0], (int)bs[1], (int)bs[2]};
-      da.dtype = n5dt;
-      da.compression = comp;
-      if (!n5.create_dataset(dsname, da)) {
-        fprintf(stderr, "cannot create dataset %s\n", dsname);
-        return 1;
-      }
       auto levels = bsj::Value::mkarr();
-      auto l0 = bsj::Value::mkobj();
-      l0->obj["dataset"] = bsj::Value::mkstr(dsname);
-      l0->obj["dimensions"] = bsj::Value::mkints(da.dims);
-      l0->obj["blockSize"] =
-          bsj::Value::mkints(std::vector<int>{da.block[0], da.block[1],
-                                              da.block[2]});
-      l0->obj["absoluteDownsampling"] =
-          bsj::Value::mkints(std::vector<int>{1, 1, 1});
-      levels->arr.push_back(l0);
+      for (size_t l = 0; l < ladder.size(); ++l) {
+        char dsname[64];
+        snprintf(dsname, sizeof dsname, "ch%dtp%d/s%zu", ch, t, l);
+        bsn5::DatasetAttrs da;
+        da.dims = {(dims[0] + ladder[l][0] - 1) / ladder[l][0],
+                   (dims[1] + ladder[l][1] - 1) / ladder[l][1],
+                   (dims[2] + ladder[l][2] - 1) / ladder[l][2]};
+        da.block = {(int)bs[0], (int)bs[1], (int)bs[2]};
+        da.dtype = n5dt;
+        da.compression = comp;
+        if (!n5.create_dataset(dsname, da)) {
+          fprintf(stderr, "cannot create dataset %s\n", dsname);
+          return 1;
+        }
+        auto lv = bsj::Value::mkobj();
+        lv->obj["dataset"] = bsj::Value::mkstr(dsname);
+        lv->obj["dimensions"] = bsj::Value::mkints(da.dims);
+        lv->obj["blockSize"] =
+            bsj::Value::mkints(std::vector<int>{da.block[0], da.block[1],
+                                                da.block[2]});
+        lv->obj["absoluteDownsampling"] = bsj::Value::mkints(
+            std::vector<long long>{ladder[l][0], ladder[l][1],
+                                   ladder[l][2]});
+        levels->arr.push_back(lv);
+      }
       mri_all->arr.push_back(levels);
     }
   set("MultiResolutionInfos", mri_all);

@@ -8,6 +8,7 @@
  * Round-1 scope: plain-N5 containers, level s0, FusionTypes
  * AVG/AVG_BLEND/MAX_INTENSITY, no intensity coefficients / masks. */
 #include <cstdio>
+#include <cstring>
 #include <set>
 
 #include "../../../include/bigstitch.h"
@@ -54,12 +55,11 @@ int main(int argc, char **argv) {
     return 1;
   }
   long long bbmin[3], bbmax[3];
-  int blk[3];
   for (int d = 0; d < 3; ++d) {
     bbmin[d] = bbmin_a->arr[d]->inum;
     bbmax[d] = bbmax_a->arr[d]->inum;
-    blk[d] = (int)bs_a->arr[d]->inum;
   }
+  (void)bs_a;
   long long dims[3] = {bbmax[0] - bbmin[0] + 1, bbmax[1] - bbmin[1] + 1,
                        bbmax[2] - bbmin[2] + 1};
   std::string dt = dt_a->str;
@@ -92,7 +92,6 @@ int main(int argc, char **argv) {
     return 1;
   }
   int num_tp = ntp_a ? (int)ntp_a->inum : (int)sd.timepoints.size();
-  size_t batch = (size_t)args.getl("batchSize", 64);
 
   for (int ti = 0; ti < num_tp; ++ti) {
     int tp = sd.timepoints[ti % sd.timepoints.size()];
@@ -128,96 +127,87 @@ int main(int argc, char **argv) {
       fviews.push_back(fv);
       fsetups.push_back(&s);
     }
-    char dsname[64];
-    snprintf(dsname, sizeof dsname, "ch0tp%d/s0", ti);
-    bsn5::DatasetAttrs da;
-    if (!n5.get_dataset_attrs(dsname, &da)) {
-      fprintf(stderr, "missing dataset %s in container\n", dsname);
+    /* levels from the container's MultiResolutionInfos */
+    auto mri = geta("MultiResolutionInfos");
+    int vol_idx = ti; /* ch0 only this round */
+    if (!mri || (size_t)vol_idx >= mri->arr.size()) {
+      fprintf(stderr, "missing MultiResolutionInfos for volume %d\n",
+              vol_idx);
       return 1;
     }
-    /* output grid (Grid.create semantics) + per-block culling */
-    size_t esz = bsn5::dtype_size(da.dtype);
-    long long ngx = (dims[0] + blk[0] - 1) / blk[0];
-    long long ngy = (dims[1] + blk[1] - 1) / blk[1];
-    long long ngz = (dims[2] + blk[2] - 1) / blk[2];
-    std::vector<bs_block_desc> blocks;
-    std::vector<std::vector<long long>> gps;
-    std::vector<int32_t> vidx;
-    std::vector<int64_t> voffs;
-    std::vector<std::vector<char>> outbufs;
-    std::vector<void *> outptrs;
-    long long nblocks_total = ngx * ngy * ngz, done = 0;
-    auto flush = [&]() -> bool {
-      if (blocks.empty()) return true;
-      voffs.push_back((int64_t)vidx.size());
-      bs_fuse_params prm{};
-      prm.fusion_type = fusion_type;
-      prm.out_dtype = out_dtype;
-      prm.min_intensity = minI;
-      prm.max_intensity = maxI;
-      prm.interp = 1;
-      if (bs_fuse_blocks(ctx, fviews.data(), fviews.size(), blocks.data(),
-                         blocks.size(), vidx.data(), voffs.data(), &prm,
-                         outptrs.data()) != BS_OK) {
-        fprintf(stderr, "fusion failed: %s\n", bs_last_error(ctx));
-        return false;
+    auto levels = mri->arr[vol_idx];
+    int nlevels = (int)levels->arr.size();
+    std::vector<int32_t> abs_ds(3 * nlevels);
+    std::vector<std::string> dsnames(nlevels);
+    for (int l = 0; l < nlevels; ++l) {
+      auto lv = levels->arr[l];
+      dsnames[l] = bsj::get_path(lv, "dataset")->str;
+      auto ad = bsj::get_path(lv, "absoluteDownsampling");
+      for (int d = 0; d < 3; ++d)
+        abs_ds[l * 3 + d] = (int32_t)ad->arr[d]->inum;
+    }
+    bs_fuse_params prm{};
+    prm.fusion_type = fusion_type;
+    prm.out_dtype = out_dtype;
+    prm.min_intensity = minI;
+    prm.max_intensity = maxI;
+    prm.interp = 1;
+    size_t esz2 = out_dtype == BS_OUT_FLOAT32 ? 4
+                  : out_dtype == BS_OUT_UINT16 ? 2 : 1;
+    std::vector<std::vector<char>> hostlvl(nlevels);
+    std::vector<void *> lvlptr(nlevels);
+    std::vector<int64_t> ldims(3 * nlevels);
+    for (int l = 0; l < nlevels; ++l) {
+      long long b = esz2;
+      for (int d = 0; d < 3; ++d)
+        b *= (dims[d] + abs_ds[l * 3 + d] - 1) / abs_ds[l * 3 + d];
+      hostlvl[l].resize((size_t)b);
+      lvlptr[l] = hostlvl[l].data();
+    }
+    int64_t vmin[3] = {bbmin[0], bbmin[1], bbmin[2]};
+    int64_t vdim[3] = {dims[0], dims[1], dims[2]};
+    if (bs_fuse_volume(ctx, fviews.data(), fviews.size(), vmin, vdim, &prm,
+                       nlevels, abs_ds.data(), ldims.data(),
+                       lvlptr.data()) != BS_OK) {
+      fprintf(stderr, "fusion failed: %s\n", bs_last_error(ctx));
+      return 1;
+    }
+    /* write every level's N5 chunks */
+    for (int l = 0; l < nlevels; ++l) {
+      bsn5::DatasetAttrs da;
+      if (!n5.get_dataset_attrs(dsnames[l], &da)) {
+        fprintf(stderr, "missing dataset %s in container\n",
+                dsnames[l].c_str());
+        return 1;
       }
-      for (size_t b = 0; b < blocks.size(); ++b) {
-        std::vector<int> clipped = {(int)blocks[b].size[0],
-                                    (int)blocks[b].size[1],
-                                    (int)blocks[b].size[2]};
-        if (!n5.write_block(dsname, da, gps[b], outbufs[b].data(),
-                            clipped)) {
-          fprintf(stderr, "block write failed\n");
-          return false;
-        }
-      }
-      done += (long long)blocks.size();
-      printf("fused %lld / %lld blocks\n", done, nblocks_total);
-      blocks.clear();
-      gps.clear();
-      vidx.clear();
-      voffs.clear();
-      outbufs.clear();
-      outptrs.clear();
-      return true;
-    };
-    for (long long gz = 0; gz < ngz; ++gz)
-      for (long long gy = 0; gy < ngy; ++gy)
-        for (long long gx = 0; gx < ngx; ++gx) {
-          bs_block_desc bd{};
-          bd.min[0] = gx * blk[0];
-          bd.min[1] = gy * blk[1];
-          bd.min[2] = gz * blk[2];
-          bd.size[0] = std::min((long long)blk[0], dims[0] - bd.min[0]);
-          bd.size[1] = std::min((long long)blk[1], dims[1] - bd.min[1]);
-          bd.size[2] = std::min((long long)blk[2], dims[2] - bd.min[2]);
-          /* cull: transformed view bbox (+2 px guard) vs block
-           * (OverlappingViews.findOverlappingViews, ref :28-47) */
-          if (blocks.empty()) voffs.push_back(0);
-          else voffs.push_back((int64_t)vidx.size());
-          for (size_t v = 0; v < fviews.size(); ++v) {
-            double lo[3], hi[3];
-            bscli::M34 m;
-            for (int i = 0; i < 12; ++i) m[i] = fviews[v].affine[i];
-            bscli::tbbox(m, fsetups[v]->dims, lo, hi);
-            bool ov = true;
-            for (int d = 0; d < 3; ++d)
-              if (hi[d] + 2 < bd.min[d] ||
-                  lo[d] - 2 > bd.min[d] + bd.size[d])
-                ov = false;
-            if (ov) vidx.push_back((int32_t)v);
+      long long lx = ldims[l * 3], ly = ldims[l * 3 + 1],
+                lz = ldims[l * 3 + 2];
+      int bx = da.block[0], by = da.block[1], bz = da.block[2];
+      std::vector<char> blk((size_t)bx * by * bz * esz2);
+      long long nwritten = 0;
+      for (long long gz = 0; gz * bz < lz; ++gz)
+        for (long long gy = 0; gy * by < ly; ++gy)
+          for (long long gx = 0; gx * bx < lx; ++gx) {
+            int cx = (int)std::min((long long)bx, lx - gx * bx);
+            int cy = (int)std::min((long long)by, ly - gy * by);
+            int cz = (int)std::min((long long)bz, lz - gz * bz);
+            const char *src = hostlvl[l].data();
+            for (int z = 0; z < cz; ++z)
+              for (int y = 0; y < cy; ++y)
+                memcpy(&blk[((size_t)z * cy + y) * cx * esz2],
+                       src + (((gz * bz + z) * ly + gy * by + y) * lx +
+                              gx * bx) * esz2,
+                       (size_t)cx * esz2);
+            if (!n5.write_block(dsnames[l], da, {gx, gy, gz}, blk.data(),
+                                {cx, cy, cz})) {
+              fprintf(stderr, "block write failed\n");
+              return 1;
+            }
+            ++nwritten;
           }
-          blocks.push_back(bd);
-          gps.push_back({gx, gy, gz});
-          outbufs.emplace_back((size_t)bd.size[0] * bd.size[1] *
-                               bd.size[2] * esz);
-          outptrs.push_back(outbufs.back().data());
-          if (blocks.size() >= batch) {
-            if (!flush()) return 1;
-          }
-        }
-    if (!flush()) return 1;
+      printf("level %d (%s): wrote %lld blocks (%lldx%lldx%lld)\n", l,
+             dsnames[l].c_str(), nwritten, lx, ly, lz);
+    }
     for (auto &fv : fviews) bs_view_release(ctx, fv.view_id);
   }
   bs_ctx_destroy(ctx);

@@ -171,6 +171,26 @@ int bs_fuse_blocks(bs_ctx *ctx, const bs_fuse_view *views, size_t nviews,
                    const int64_t *view_idx_offsets, /* nb+1 prefix offsets */
                    const bs_fuse_params *params, void **out_blocks);
 
+/* Volume-mode fusion + multi-resolution pyramid (SURVEY.md §8(f) row 1:
+ * N5ApiTools.setupMultiResolutionPyramid / writeDownsampledBlock,
+ * reference SparkAffineFusion.java:703-782, CreateFusionContainer.java:
+ * 351-358). Fuses the whole bounding-box volume on-device (internal
+ * block grid + OverlappingViews culling), then computes each pyramid
+ * level from the previous by box-mean over the RELATIVE downsampling
+ * factors (values rounded to nearest for integer dtypes), and stages
+ * every level back into the caller's host buffers through pinned
+ * double-buffering.
+ *   abs_downsampling: nlevels x 3 ints, level 0 must be {1,1,1}; each
+ *     level's factors must be integer multiples of the previous level's.
+ *   level_dims_out (optional): nlevels x 3, receives ceil(dim/ds).
+ *   level_buffers: caller-owned host buffers, one per level, each
+ *     prod(level_dims)*dtype_size bytes. */
+int bs_fuse_volume(bs_ctx *ctx, const bs_fuse_view *views, size_t nviews,
+                   const int64_t vol_min[3], const int64_t vol_dims[3],
+                   const bs_fuse_params *params, int32_t nlevels,
+                   const int32_t *abs_downsampling, int64_t *level_dims_out,
+                   void **level_buffers);
+
 /* ------------------------------------------------------- instrumentation */
 
 /* Per-kernel timing, HIP-event measured on the launch stream (bench.py
@@ -190,6 +210,7 @@ enum bs_kernel_id {
   BS_K_SUBPIX,     /* 7-point PCM gather */
   BS_K_FUSE,       /* inverse-affine trilinear blend fusion */
   BS_K_SYNTH,      /* bench-only synthetic tile render */
+  BS_K_PYRAMID,    /* 2x (or general) box-mean pyramid level */
   BS_K_COUNT
 };
 

@@ -43,6 +43,7 @@ FUSION_MAX = 2
 __all__ = [
     "fuse_block",
     "blend_weight",
+    "downsample_level",
     "FUSION_AVG",
     "FUSION_AVG_BLEND",
     "FUSION_MAX",
@@ -179,3 +180,29 @@ def fuse_block(
     )
     q = np.floor(np.abs(scaled) + 0.5) * np.sign(scaled)
     return np.clip(q, 0, tmax).astype(out_dtype)
+
+
+def downsample_level(vol, rel):
+    """[PIN-PYR] pyramid level: box-mean over rel=(rx,ry,rz) (x,y,z
+    order), output dims ceil(dim/rel), edge boxes average their (fewer)
+    in-bounds voxels; integer dtypes round to nearest (ties to even,
+    matching the HIP __float2int_rn). Restates
+    N5ApiTools.writeDownsampledBlock (reference SparkAffineFusion.java:
+    736-753 call site; artifact-side implementation — see the parity
+    note in oracle/__init__)."""
+    rx, ry, rz = int(rel[0]), int(rel[1]), int(rel[2])
+    nz, ny, nx = vol.shape
+    dz, dy, dx = -(-nz // rz), -(-ny // ry), -(-nx // rx)
+    acc = np.zeros((dz, dy, dx), np.float64)
+    cnt = np.zeros((dz, dy, dx), np.float64)
+    for oz in range(rz):
+        for oy in range(ry):
+            for ox in range(rx):
+                sub = vol[oz::rz, oy::ry, ox::rx].astype(np.float64)
+                acc[:sub.shape[0], :sub.shape[1], :sub.shape[2]] += sub
+                cnt[:sub.shape[0], :sub.shape[1], :sub.shape[2]] += 1.0
+    m = acc / cnt
+    if np.issubdtype(vol.dtype, np.integer):
+        info = np.iinfo(vol.dtype)
+        return np.clip(np.rint(m), info.min, info.max).astype(vol.dtype)
+    return m.astype(vol.dtype)

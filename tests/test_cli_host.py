@@ -124,7 +124,8 @@ def test_cli_fusion_end_to_end(tmp_path):
     xml, n5, err, (a, b) = make_grid_dataset(str(tmp_path))
     out = os.path.join(str(tmp_path), "fused.n5")
     r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml, "-o",
-             out, "--blockSize", "32,32,32", "--dataType", "FLOAT32"])
+             out, "--blockSize", "32,32,32", "--dataType", "FLOAT32",
+             "--downsamplings", "1,1,1;2,2,2"])
     assert r.returncode == 0, r.stderr
     r = run([os.path.join(BIN, "affine-fusion"), "-o", out,
              "--fusionType", "AVG_BLEND", "--blendingRange", "8"])
@@ -142,3 +143,8 @@ def test_cli_fusion_end_to_end(tmp_path):
                         of.FUSION_AVG_BLEND, out_dtype=np.float32)
     denom = np.maximum(np.abs(ref), 1.0)
     assert np.max(np.abs(fused - ref) / denom) < 1e-4
+    # pyramid level s1 written and matching the oracle box-mean of s0
+    s1, _ = n5util.read_dataset(out, "ch0tp0/s1")
+    assert s1.shape == (32, 32, 52)
+    ref1 = of.downsample_level(fused, (2, 2, 2))
+    assert np.max(np.abs(s1 - ref1)) < 1e-3

@@ -115,3 +115,48 @@ def test_fuse_identity_roundtrip(ctx):
         [[0]], fusion_type=of.FUSION_AVG, out_dtype=np.float32,
     )[0]
     assert np.array_equal(got, vol.astype(np.float32))
+
+
+def test_fuse_volume_with_pyramid(ctx):
+    """Volume-mode fusion + pyramid levels vs the per-block oracle chain
+    (SURVEY.md §8(f) row 1)."""
+    views = fusion_views(13)
+    for i, v in enumerate(views):
+        v["affine"][0, 3] += 12.0 * i
+    gviews = upload_views(ctx, views, base=500)
+    vol_min, vol_dims = (1, 2, 0), (52, 30, 26)
+    ds = [(1, 1, 1), (2, 2, 2), (4, 4, 4)]
+    levels = ctx.fuse_volume(gviews, vol_min, vol_dims, downsamplings=ds,
+                             fusion_type=of.FUSION_AVG_BLEND,
+                             out_dtype=np.float32)
+    assert len(levels) == 3
+    assert levels[0].shape == (26, 30, 52)
+    assert levels[1].shape == (13, 15, 26)
+    assert levels[2].shape == (7, 8, 13)
+    ref0 = of.fuse_block(views, vol_min, vol_dims, of.FUSION_AVG_BLEND,
+                         out_dtype=np.float32)
+    denom = np.maximum(np.abs(ref0), 1.0)
+    assert np.max(np.abs(levels[0] - ref0) / denom) < 1e-4
+    # pyramid: oracle box-mean chain applied to the GPU's own level 0
+    ref1 = of.downsample_level(levels[0], (2, 2, 2))
+    assert np.max(np.abs(levels[1] - ref1)) < 1e-3
+    ref2 = of.downsample_level(levels[1], (2, 2, 2))
+    assert np.max(np.abs(levels[2] - ref2)) < 1e-3
+
+
+def test_fuse_volume_uint16_pyramid(ctx):
+    views = fusion_views(17)
+    gviews = upload_views(ctx, views, base=600)
+    levels = ctx.fuse_volume(gviews, (0, 0, 0), (32, 28, 24),
+                             downsamplings=[(1, 1, 1), (2, 2, 1)],
+                             fusion_type=of.FUSION_AVG,
+                             out_dtype=np.uint16, min_intensity=0,
+                             max_intensity=40000)
+    ref0 = of.fuse_block(views, (0, 0, 0), (32, 28, 24), of.FUSION_AVG,
+                         out_dtype=np.uint16, min_intensity=0,
+                         max_intensity=40000)
+    d = np.abs(levels[0].astype(np.int64) - ref0.astype(np.int64))
+    assert d.max() <= 1 and (d == 0).mean() > 0.99
+    ref1 = of.downsample_level(levels[0], (2, 2, 1))
+    d1 = np.abs(levels[1].astype(np.int64) - ref1.astype(np.int64))
+    assert d1.max() <= 1

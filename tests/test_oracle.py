@@ -248,3 +248,20 @@ def test_golden_fuse(path):
         assert np.allclose(out, g["out"], rtol=1e-12, atol=1e-12)
     else:
         assert np.array_equal(out, g["out"])
+
+
+def test_pyramid_downsample_level():
+    """[PIN-PYR] closed forms: constant stays constant; 2x of a ramp;
+    ceil dims with clamped edge boxes."""
+    c = np.full((6, 6, 6), 7.0, np.float32)
+    d = fusion.downsample_level(c, (2, 2, 2))
+    assert d.shape == (3, 3, 3) and np.all(d == 7.0)
+    ramp = np.tile(np.arange(8, dtype=np.uint16), (4, 4, 1))
+    d = fusion.downsample_level(ramp, (2, 1, 1))
+    assert d.shape == (4, 4, 4)
+    assert np.array_equal(d[0, 0], [0, 2, 4, 6])  # rint(mean of pairs)=0.5->0? no: (0+1)/2=0.5 -> rint=0; (2+3)/2=2.5 -> 2
+    # odd extent: last box averages the single remaining element
+    v = np.arange(5, dtype=np.float32).reshape(1, 1, 5)
+    d = fusion.downsample_level(v, (2, 1, 1))
+    assert d.shape == (1, 1, 3)
+    assert np.allclose(d[0, 0], [0.5, 2.5, 4.0])
