@@ -46,7 +46,8 @@ def main():
         vlists.append(host.find_overlapping_views(cull, off, bsz))
     dt = np.float32 if args.dtype == "float32" else np.uint16
     def run():
-        return ctx.fuse_blocks(views, blocks, vlists,
+        return ctx.fuse_volume(views, (0, 0, 0), dims,
+                               downsamplings=[(1, 1, 1)],
                                fusion_type=FUSION_AVG_BLEND, out_dtype=dt,
                                min_intensity=0, max_intensity=65535)
     for _ in range(args.warmup):
@@ -80,7 +81,7 @@ def main():
     )
     print(json.dumps(line))
     # light sanity: fused interior not all zero
-    assert float(np.asarray(out[len(out) // 2], np.float64).mean()) > 0
+    assert float(np.asarray(out[0][dims[2] // 2], np.float64).mean()) > 0
     ctx.close()
 
 if __name__ == "__main__":
