@@ -847,6 +847,8 @@ struct bs_ctx {
   float *dblobs = nullptr;
   size_t dblobs_cap = 0;
   void *hstage[2] = {nullptr, nullptr}; /* pinned D2H staging (64 MB) */
+  void *dvol_arena = nullptr; /* cached fuse_volume level buffers */
+  size_t dvol_cap = 0;
   hipEvent_t stage_ev[2] = {nullptr, nullptr};
   float *dbg_pcm = nullptr; /* last pair's PCM (points into a slot) */
   long dbg_px = 0, dbg_py = 0, dbg_pz = 0;
@@ -973,6 +975,7 @@ extern "C" void bs_ctx_destroy(bs_ctx *c) {
     if (sl.done) (void)hipEventDestroy(sl.done);
     if (sl.stream) (void)hipStreamDestroy(sl.stream);
   }
+  (void)hipFree(c->dvol_arena);
   (void)hipHostFree(c->hstage[0]);
   (void)hipHostFree(c->hstage[1]);
   if (c->stage_ev[0]) (void)hipEventDestroy(c->stage_ev[0]);
@@ -1749,13 +1752,13 @@ extern "C" int bs_fuse_volume(bs_ctx *c, const bs_fuse_view *views,
   float invRange =
       (float)((prm->out_dtype == BS_OUT_UINT8 ? 255.0 : 65535.0) /
               (denom != 0.0 ? denom : 1.0));
-  /* level dims + device buffers */
+  /* level dims + device buffers (carved from a cached grow-only arena:
+   * a per-call hipMalloc/hipFree of multi-GB volumes costs ~100 ms) */
   std::vector<std::array<long, 3>> ldims(nlevels);
   std::vector<void *> dlvl(nlevels, nullptr);
-  auto cleanup = [&]() {
-    for (auto p : dlvl)
-      if (p) (void)hipFree(p);
-  };
+  auto cleanup = [&]() {};
+  std::vector<size_t> lbytes(nlevels);
+  size_t total = 0;
   for (int l = 0; l < nlevels; ++l) {
     for (int d = 0; d < 3; ++d) {
       int f = abs_ds[l * 3 + d];
@@ -1766,11 +1769,19 @@ extern "C" int bs_fuse_volume(bs_ctx *c, const bs_fuse_view *views,
       ldims[l][d] = (vol_dims[d] + f - 1) / f; /* [PIN-PYR] ceil */
       if (level_dims_out) level_dims_out[l * 3 + d] = ldims[l][d];
     }
-    size_t bytes = (size_t)ldims[l][0] * ldims[l][1] * ldims[l][2] * esz;
-    if (hipMalloc(&dlvl[l], bytes) != hipSuccess) {
-      cleanup();
-      c->err = "device volume alloc failed (volume too large this round)";
-      return BS_ENOMEM;
+    lbytes[l] = (size_t)ldims[l][0] * ldims[l][1] * ldims[l][2] * esz;
+    total += (lbytes[l] + 255) & ~(size_t)255;
+  }
+  rc = ensure_dev(c, &c->dvol_arena, &c->dvol_cap, total);
+  if (rc) {
+    c->err = "device volume alloc failed (volume too large this round)";
+    return rc;
+  }
+  {
+    size_t off = 0;
+    for (int l = 0; l < nlevels; ++l) {
+      dlvl[l] = (char *)c->dvol_arena + off;
+      off += (lbytes[l] + 255) & ~(size_t)255;
     }
   }
   /* fusion of level 0 over an internal 256x128x128 grid with culling */
@@ -1861,6 +1872,5 @@ extern "C" int bs_fuse_volume(bs_ctx *c, const bs_fuse_view *views,
   CHK(c, hipStreamSynchronize(c->stream));
   c->stats.blocks += (long long)fblocks.size();
   flush_stats(c);
-  cleanup();
   return BS_OK;
 }
