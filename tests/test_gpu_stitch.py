@@ -39,6 +39,8 @@ def full_pair(ctx, a, b, ida=0, idb=1):
     [
         ((64, 64, 64), (5.25, -3.5, 2.0), (1, 1, 1)),
         ((64, 64, 64), (5.25, -3.5, 2.0), (2, 2, 1)),
+        ((64, 64, 64), (5.25, -3.5, 2.0), (2, 2, 2)),
+        ((64, 64, 64), (-4.5, 2.0, 6.25), (4, 4, 2)),
         ((48, 96, 80), (-6.5, 2.25, 0.75), (1, 1, 1)),
         ((128, 128, 128), (7.3, -4.8, 3.1), (2, 2, 1)),
         ((32, 48, 64), (0.0, 0.0, 0.0), (1, 1, 1)),
