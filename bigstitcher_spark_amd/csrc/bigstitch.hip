@@ -399,83 +399,68 @@ __device__ void pk_merge_shfl(float (&tv)[5], long long (&ti)[5]) {
   }
 }
 
-/* Slab local-maxima scan: strict 26-neighborhood maxima with periodic
- * wrap [PIN-MAX]; per-WG top-5 -> wgbuf.
- * Separable structure: per z-slice keep a 3-slice ring of VALues and of
- * R3 = 3-wide x-window maxima; a voxel's 26-neighbor max is then
- *   max( colmax3(R3[z-1]), colmax3(R3[z+1]),            (18 neighbors)
- *        R3[z][y-1], R3[z][y+1],                        (6 neighbors)
- *        VAL[z][y][x-1], VAL[z][y][x+1] )               (2 neighbors)
- * -> ~10 LDS reads/voxel instead of 26, and HBM reads ~1.25x algorithmic
- * (y-halo only; the z dimension is reused through the ring). */
-#define PKS_TX 256
-#define PKS_TY 8
-#define PKS_ZC 32
-
+/* Tile local-maxima scan: strict 26-neighborhood maxima with periodic
+ * wrap [PIN-MAX]; per-WG top-5 -> wgbuf. Interior tiles (the common
+ * case) load their halo without any modulo index math. */
 __global__ __launch_bounds__(256) void k_peak_tile(
     const float *pcm, int px, int py, int pz, bs_peak *wgbuf) {
-  __shared__ float val[3][PKS_TY + 2][PKS_TX + 2];
-  __shared__ float r3[3][PKS_TY + 2][PKS_TX];
+  __shared__ float tile[(PK_TZ + 2) * (PK_TY + 2) * (PK_TX + 2)];
   __shared__ float wv[4][5];
   __shared__ long long wi[4][5];
   const int tid = threadIdx.x;
-  const int ntx = (px + PKS_TX - 1) / PKS_TX;
-  const int nty = (py + PKS_TY - 1) / PKS_TY;
-  const int ntz = (pz + PKS_ZC - 1) / PKS_ZC;
+  const int ntx = (px + PK_TX - 1) / PK_TX;
+  const int nty = (py + PK_TY - 1) / PK_TY;
+  const int ntz = (pz + PK_TZ - 1) / PK_TZ;
+  const int HX = PK_TX + 2, HY = PK_TY + 2, HZ = PK_TZ + 2;
   const long ntiles = (long)ntx * nty * ntz;
   float tv[5];
   long long ti[5];
   for (int k = 0; k < 5; ++k) { tv[k] = -3.0e38f; ti[k] = 0x7fffffffffffffffLL; }
   for (long t0 = blockIdx.x; t0 < ntiles; t0 += gridDim.x) {
-    const int bx = (int)(t0 % ntx);
-    const int by = (int)((t0 / ntx) % nty);
-    const int bz = (int)(t0 / ((long)ntx * nty));
-    const int x0 = bx * PKS_TX, y0 = by * PKS_TY, z0 = bz * PKS_ZC;
-    const int tx = min(PKS_TX, px - x0);   /* eval extent this tile */
-    const int ty = min(PKS_TY, py - y0);
-    const int zend = min(pz, z0 + PKS_ZC);
-    const int nrow = ty + 2, ncol = tx + 2;
-    /* load slice (counter zc, slot zc%3) + compute its R3 */
-    auto load_slice = [&](int zc) {
-      const int slot = ((zc % 3) + 3) % 3;
-      const int zs = ((zc % pz) + pz) % pz;
-      const float *base = pcm + (long)zs * py * px;
-      for (int i = tid; i < nrow * ncol; i += 256) {
-        int r = i / ncol, xh = i - r * ncol;
-        int gy = y0 + r - 1, gx = x0 + xh - 1;
-        gy = gy < 0 ? gy + py : (gy >= py ? gy - py : gy);
-        gx = gx < 0 ? gx + px : (gx >= px ? gx - px : gx);
-        val[slot][r][xh] = base[(long)gy * px + gx];
+    int bx = (int)(t0 % ntx);
+    int by = (int)((t0 / ntx) % nty);
+    int bz = (int)(t0 / ((long)ntx * nty));
+    int x0 = bx * PK_TX, y0 = by * PK_TY, z0 = bz * PK_TZ;
+    const bool interior = x0 > 0 && y0 > 0 && z0 > 0 && x0 + PK_TX < px &&
+                          y0 + PK_TY < py && z0 + PK_TZ < pz;
+    if (interior) {
+      const float *base =
+          pcm + ((long)(z0 - 1) * py + (y0 - 1)) * px + (x0 - 1);
+      for (int i = tid; i < HX * HY * HZ; i += 256) {
+        int lx = i % HX, t = i / HX, ly = t % HY, lz = t / HY;
+        tile[i] = base[((long)lz * py + ly) * px + lx];
       }
-      __syncthreads();
-      for (int i = tid; i < nrow * tx; i += 256) {
-        int r = i / tx, x = i - r * tx;
-        r3[slot][r][x] = fmaxf(fmaxf(val[slot][r][x], val[slot][r][x + 1]),
-                               val[slot][r][x + 2]);
+    } else {
+      for (int i = tid; i < HX * HY * HZ; i += 256) {
+        int lx = i % HX, t = i / HX, ly = t % HY, lz = t / HY;
+        int gx = (x0 + lx - 1 + px) % px;
+        int gy = (y0 + ly - 1 + py) % py;
+        int gz = (z0 + lz - 1 + pz) % pz;
+        tile[i] = pcm[((long)gz * py + gy) * px + gx];
       }
-      __syncthreads();
-    };
-    load_slice(z0 - 1);
-    load_slice(z0);
-    for (int ze = z0; ze < zend; ++ze) {
-      load_slice(ze + 1);
-      const int sm = ((ze - 1) % 3 + 3) % 3, sc = ze % 3,
-                sp = (ze + 1) % 3;
-      for (int i = tid; i < ty * tx; i += 256) {
-        int y = i / tx, x = i - y * tx; /* local; eval row = y+1 */
-        float v = val[sc][y + 1][x + 1];
-        float m = fmaxf(val[sc][y + 1][x], val[sc][y + 1][x + 2]);
-        m = fmaxf(m, fmaxf(r3[sc][y][x], r3[sc][y + 2][x]));
-        m = fmaxf(m, fmaxf(fmaxf(r3[sm][y][x], r3[sm][y + 1][x]),
-                           r3[sm][y + 2][x]));
-        m = fmaxf(m, fmaxf(fmaxf(r3[sp][y][x], r3[sp][y + 1][x]),
-                           r3[sp][y + 2][x]));
-        if (v > m)
-          pk_insert(tv, ti, v,
-                    ((long long)ze * py + (y0 + y)) * px + (x0 + x));
-      }
-      __syncthreads(); /* ring slot reused by the next load */
     }
+    __syncthreads();
+    for (int i = tid; i < PK_TX * PK_TY * PK_TZ; i += 256) {
+      int lx = i % PK_TX, t = i / PK_TX, ly = t % PK_TY, lz = t / PK_TY;
+      int gx = x0 + lx, gy = y0 + ly, gz = z0 + lz;
+      bool inb = gx < px && gy < py && gz < pz;
+      /* branchless 26-neighbor max: every LDS offset is base + constant
+       * (the #pragma unroll makes dz/dy compile-time), no divergence */
+      const int base = ((lz + 1) * HY + ly + 1) * HX + lx + 1;
+      float v = tile[base];
+      float m = fmaxf(tile[base - 1], tile[base + 1]);
+#pragma unroll
+      for (int dz = 0; dz <= 2; ++dz)
+#pragma unroll
+        for (int dy = 0; dy <= 2; ++dy) {
+          if (dz == 1 && dy == 1) continue;
+          const int b2 = base + (dz - 1) * HY * HX + (dy - 1) * HX;
+          m = fmaxf(m, fmaxf(fmaxf(tile[b2 - 1], tile[b2]), tile[b2 + 1]));
+        }
+      if (inb && v > m)
+        pk_insert(tv, ti, v, ((long long)gz * py + gy) * px + gx);
+    }
+    __syncthreads(); /* tile LDS reused next iteration */
   }
   pk_merge_shfl(tv, ti);
   int lane = tid & 63, wave = tid >> 6;
@@ -1301,8 +1286,8 @@ static int stitch_phaseA(bs_ctx *c, bs_slot *sl, const bs_pair_desc &pd,
   c->dbg_py = Py;
   c->dbg_pz = Pz;
   /* peak scan [PIN-MAX] */
-  long ntiles = (long)((Px + PKS_TX - 1) / PKS_TX) *
-                ((Py + PKS_TY - 1) / PKS_TY) * ((Pz + PKS_ZC - 1) / PKS_ZC);
+  long ntiles = (long)((Px + PK_TX - 1) / PK_TX) *
+                ((Py + PK_TY - 1) / PK_TY) * ((Pz + PK_TZ - 1) / PK_TZ);
   long npkwg = std::min(2048L, ntiles);
   int rc2 = ensure_dev(c, (void **)&sl->wgpk, &sl->wgpk_cap,
                        (size_t)npkwg * 5 * sizeof(bs_peak));
