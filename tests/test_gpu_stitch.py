@@ -173,3 +173,23 @@ def test_synth_views_stitchable(ctx):
     # and the recovered shift is near the injected ground truth (sub-pixel
     # quadratic-fit bias on a small noisy tile can approach ~0.7 px)
     assert np.all(np.abs(got["shift"] - np.array(true_shift)) < 0.75)
+
+
+def test_stitch_1024_axis(ctx):
+    """Exercises the N=1024 FFT path (z axis; strided pass at 135 KB LDS,
+    1 WG/CU) with a non-cubic volume."""
+    shape = (1024, 256, 256)  # (nz, ny, nx)
+    true_shift = (3.5, -2.25, 10.0)
+    ba, bb = synth.pair_blobs_union(shape, true_shift, seed=9)
+    ctx.synth(60, shape, ba, noise_seed=11)
+    ctx.synth(61, shape, bb, noise_seed=12)
+    a = ctx.download(60, shape)
+    b = ctx.download(61, shape)
+    ref = phasecorr.phase_correlation_shift(a, b, ds=(1, 1, 1), workers=-1)
+    pair = dict(view_a=60, view_b=61, off_a=(0, 0, 0),
+                size_a=(256, 256, 1024), off_b=(0, 0, 0),
+                size_b=(256, 256, 1024))
+    got = ctx.stitch_batch([pair], ds=(1, 1, 1))[0]
+    assert got["valid"] and ref["valid"]
+    assert np.all(np.abs(got["shift"] - ref["shift"]) < 1e-3)
+    assert got["r"] == pytest.approx(ref["r"], abs=1e-9)
