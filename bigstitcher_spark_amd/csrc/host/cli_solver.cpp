@@ -1,0 +1,192 @@
+/* `solver` — drop-in for the reference's Solver CLI (reference
+ * Solver.java; a multi-threaded host tool, not Spark — README.md:220).
+ * Consumes the <StitchingResults> links written by `stitching`:
+ *   1. drops links whose stored hash no longer matches the current
+ *      registrations (Solver.java:404-415 semantics; hash rule
+ *      [PIN-HASH] in bs_spimdata.h),
+ *   2. solves the global translation adjustment minimising
+ *      sum_links || (t_B - t_A) - d_link ||^2 with the first (or
+ *      --fixedViews) view fixed — the translation case of the
+ *      reference's GlobalOpt tile optimisation (Solver.java:301-337;
+ *      the mpicbg TranslationModel3D path) as a direct graph-Laplacian
+ *      least squares,
+ *   3. prepends a "Stitching Transform" translation to each view's
+ *      registration chain and saves the XML.
+ * d_link: the stored link matrix's translation ws satisfies
+ * "B's content appears at +ws relative to where the current
+ * registrations place it" is ws = -correction, i.e. the solved
+ * adjustment should satisfy t_B - t_A = -ws. */
+#include <cmath>
+#include <cstdio>
+#include <set>
+
+#include "bs_cli_util.h"
+#include "bs_spimdata.h"
+
+int main(int argc, char **argv) {
+  bscli::Args args;
+  std::map<std::string, std::string> alias = {{"-x", "--xml"}};
+  if (!args.parse(argc, argv, alias, {"dryRun"}) || !args.has("xml")) {
+    fprintf(stderr,
+            "usage: solver -x dataset.xml [--fixedViews tp,setup[;..]] "
+            "[--minR 0.3] [--dryRun]\n");
+    return 2;
+  }
+  bssd::SpimData sd;
+  std::string err;
+  if (!sd.load(args.get("xml"), &err)) {
+    fprintf(stderr, "error: %s\n", err.c_str());
+    return 1;
+  }
+  double minR = args.getd("minR", 0.3);
+  auto entries = sd.stitching_results();
+  printf("solver: %zu stitching links loaded\n", entries.size());
+
+  /* view index map over (tp, setup) present in registrations */
+  std::vector<bssd::ViewId> ids;
+  std::map<bssd::ViewId, int> idx;
+  for (auto &kv : sd.regs) {
+    idx[kv.first] = (int)ids.size();
+    ids.push_back(kv.first);
+  }
+  int n = (int)ids.size();
+  if (n == 0) {
+    fprintf(stderr, "no registrations\n");
+    return 1;
+  }
+  struct Link {
+    int a, b;
+    double d[3];
+  };
+  std::vector<Link> links;
+  size_t dropped_hash = 0, dropped_r = 0;
+  for (auto &e : entries) {
+    if (e.views_a.empty() || e.views_b.empty()) continue;
+    auto ia = idx.find(e.views_a[0]);
+    auto ib = idx.find(e.views_b[0]);
+    if (ia == idx.end() || ib == idx.end()) continue;
+    /* hash check against CURRENT registrations (stale links are silently
+     * discarded, exactly the reference solver's behaviour) */
+    double h = bssd::SpimData::calculate_hash(sd.regs[e.views_a[0]],
+                                              sd.regs[e.views_b[0]]);
+    if (std::fabs(h - e.hash) > 1e-6 * (1.0 + std::fabs(h))) {
+      ++dropped_hash;
+      continue;
+    }
+    if (e.r < minR) {
+      ++dropped_r;
+      continue;
+    }
+    Link l;
+    l.a = ia->second;
+    l.b = ib->second;
+    for (int d = 0; d < 3; ++d) l.d[d] = -e.matrix[d * 4 + 3];
+    links.push_back(l);
+  }
+  printf("solver: %zu usable links (%zu stale-hash, %zu below minR)\n",
+         links.size(), dropped_hash, dropped_r);
+  if (links.empty()) {
+    fprintf(stderr, "nothing to solve\n");
+    return 1;
+  }
+
+  /* fixed views: default = first view id */
+  std::set<int> fixed;
+  if (args.has("fixedViews")) {
+    std::string spec = args.get("fixedViews");
+    std::string cur;
+    for (char c : spec + ";") {
+      if (c == ';') {
+        int tp, su;
+        if (sscanf(cur.c_str(), "%d,%d", &tp, &su) == 2) {
+          auto it = idx.find({tp, su});
+          if (it != idx.end()) fixed.insert(it->second);
+        }
+        cur.clear();
+      } else {
+        cur += c;
+      }
+    }
+  }
+  if (fixed.empty()) fixed.insert(0);
+
+  /* graph-Laplacian normal equations, solved per axis by dense Gaussian
+   * elimination (n = #views is metadata-sized) */
+  std::vector<double> t(3 * n, 0.0);
+  for (int ax = 0; ax < 3; ++ax) {
+    std::vector<double> A((size_t)n * n, 0.0), rhs(n, 0.0);
+    for (auto &l : links) {
+      A[(size_t)l.a * n + l.a] += 1.0;
+      A[(size_t)l.b * n + l.b] += 1.0;
+      A[(size_t)l.a * n + l.b] -= 1.0;
+      A[(size_t)l.b * n + l.a] -= 1.0;
+      rhs[l.a] -= l.d[ax];
+      rhs[l.b] += l.d[ax];
+    }
+    for (int f : fixed) { /* pin t_f = 0 */
+      for (int j = 0; j < n; ++j) A[(size_t)f * n + j] = 0.0;
+      A[(size_t)f * n + f] = 1.0;
+      rhs[f] = 0.0;
+    }
+    /* tiny ridge keeps disconnected components solvable */
+    for (int i = 0; i < n; ++i) A[(size_t)i * n + i] += 1e-9;
+    for (int col = 0; col < n; ++col) { /* partial-pivot elimination */
+      int piv = col;
+      for (int r = col + 1; r < n; ++r)
+        if (std::fabs(A[(size_t)r * n + col]) >
+            std::fabs(A[(size_t)piv * n + col]))
+          piv = r;
+      if (piv != col) {
+        for (int j = 0; j < n; ++j)
+          std::swap(A[(size_t)col * n + j], A[(size_t)piv * n + j]);
+        std::swap(rhs[col], rhs[piv]);
+      }
+      double p = A[(size_t)col * n + col];
+      for (int r = 0; r < n; ++r) {
+        if (r == col) continue;
+        double f = A[(size_t)r * n + col] / p;
+        if (f == 0.0) continue;
+        for (int j = col; j < n; ++j)
+          A[(size_t)r * n + j] -= f * A[(size_t)col * n + j];
+        rhs[r] -= f * rhs[col];
+      }
+    }
+    for (int i = 0; i < n; ++i) t[ax * n + i] = rhs[i] / A[(size_t)i * n + i];
+  }
+
+  for (int i = 0; i < n; ++i)
+    printf("view (%d,%d): adjustment (%.4f, %.4f, %.4f)\n", ids[i].first,
+           ids[i].second, t[0 * n + i], t[1 * n + i], t[2 * n + i]);
+  if (args.has("dryRun")) return 0;
+
+  /* prepend the Stitching Transform (outermost) to each registration */
+  auto vrs = sd.root->child("ViewRegistrations");
+  if (!vrs) {
+    fprintf(stderr, "missing ViewRegistrations\n");
+    return 1;
+  }
+  for (auto &vr : vrs->all("ViewRegistration")) {
+    int tp = atoi(vr->attrs["timepoint"].c_str());
+    int su = atoi(vr->attrs["setup"].c_str());
+    auto it = idx.find({tp, su});
+    if (it == idx.end()) continue;
+    int i = it->second;
+    char aff[256];
+    snprintf(aff, sizeof aff,
+             "1.0 0.0 0.0 %.17g 0.0 1.0 0.0 %.17g 0.0 0.0 1.0 %.17g",
+             t[0 * n + i], t[1 * n + i], t[2 * n + i]);
+    auto vt = std::make_shared<bsx::Node>();
+    vt->tag = "ViewTransform";
+    vt->attrs["type"] = "affine";
+    vt->add_text("Name", "Stitching Transform");
+    vt->add_text("affine", aff);
+    vr->children.insert(vr->children.begin(), vt);
+  }
+  if (!sd.save(sd.xml_path)) {
+    fprintf(stderr, "cannot write %s\n", sd.xml_path.c_str());
+    return 1;
+  }
+  printf("solver: updated registrations written to %s\n",
+         sd.xml_path.c_str());
+  return 0;
+}

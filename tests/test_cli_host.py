@@ -148,3 +148,50 @@ def test_cli_fusion_end_to_end(tmp_path):
     assert s1.shape == (32, 32, 52)
     ref1 = of.downsample_level(fused, (2, 2, 2))
     assert np.max(np.abs(s1 - ref1)) < 1e-3
+
+
+@pytest.mark.gpu
+def test_cli_full_pipeline_stitch_solve_fuse(tmp_path):
+    """stitch -> solver -> fuse, self-contained:
+    the solver turns our XML links into corrected registrations and the
+    fusion of the corrected dataset matches the oracle fused with the
+    same (solved) affines; the solved position lands on the injected
+    ground truth."""
+    from tests.test_cli_solver import model_translations
+
+    xml, n5, err, (a, b) = make_grid_dataset(str(tmp_path))
+    r = run([os.path.join(BIN, "stitching"), "-x", xml, "-ds", "1,1,1",
+             "--minOverlapRatio", "0.05"])
+    assert r.returncode == 0, r.stderr + r.stdout
+    r = run([os.path.join(BIN, "solver"), "-x", xml])
+    assert r.returncode == 0, r.stderr + r.stdout
+    t = model_translations(xml)
+    # solver moved B to its true content position (nominal 40 + err)
+    true_pos = np.array([40 + err[0], err[1], err[2]])
+    assert np.all(np.abs(t[1] - true_pos) < 0.75), (t[1], true_pos)
+    out = os.path.join(str(tmp_path), "fused.n5")
+    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml, "-o",
+             out, "--blockSize", "32,32,32", "--dataType", "FLOAT32"])
+    assert r.returncode == 0, r.stderr
+    r = run([os.path.join(BIN, "affine-fusion"), "-o", out,
+             "--fusionType", "AVG_BLEND", "--blendingRange", "8"])
+    assert r.returncode == 0, r.stderr + r.stdout
+    # bbox moved with the solved registration; read back its metadata
+    attrs = n5util.root_attrs(out)["Bigstitcher-Spark"]
+    bbmin = attrs["Boundingbox_min"]
+    fused, _ = n5util.read_dataset(out, "ch0tp0/s0")
+    ident = np.hstack([np.eye(3), np.zeros((3, 1))])
+    affA = ident.copy()
+    affA[:, 3] -= np.array(bbmin, float)
+    affB = ident.copy()
+    affB[:, 3] = t[1] - np.array(bbmin, float)
+    views = [
+        dict(data=a, affine=affA, border=(0, 0, 0), range=(8, 8, 8)),
+        dict(data=b, affine=affB, border=(0, 0, 0), range=(8, 8, 8)),
+    ]
+    dims_zyx = fused.shape
+    ref = of.fuse_block(views, (0, 0, 0),
+                        (dims_zyx[2], dims_zyx[1], dims_zyx[0]),
+                        of.FUSION_AVG_BLEND, out_dtype=np.float32)
+    denom = np.maximum(np.abs(ref), 1.0)
+    assert np.max(np.abs(fused - ref) / denom) < 1e-4

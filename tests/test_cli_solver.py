@@ -1,0 +1,128 @@
+"""CPU tests of the `solver` CLI (§8(f) row 3): translation global
+optimisation over StitchingResults links, the stale-hash drop rule
+(reference Solver.java:404-415), and the registration update side
+effect."""
+
+import os
+import subprocess
+import xml.etree.ElementTree as ET
+
+import numpy as np
+
+from tests import n5util
+from tests.test_cli_host import BIN, run
+
+
+def write_xml_with_links(path, entries, posB=40.0):
+    n5util.make_dataset_xml(
+        path, "input.n5",
+        [dict(id=0, dims=(64, 64, 64), pos=(0.0, 0.0, 0.0)),
+         dict(id=1, dims=(64, 64, 64), pos=(posB, 0.0, 0.0))],
+    )
+    text = open(path).read()
+    sr = "  <StitchingResults>\n"
+    for e in entries:
+        sr += (
+            "    <PairwiseResult>\n"
+            f"      <ViewIdsA>{e['a']}</ViewIdsA>\n"
+            f"      <ViewIdsB>{e['b']}</ViewIdsB>\n"
+            f"      <Matrix>1 0 0 {e['ws'][0]} 0 1 0 {e['ws'][1]} "
+            f"0 0 1 {e['ws'][2]}</Matrix>\n"
+            "      <BoundingBoxMin>0 0 0</BoundingBoxMin>\n"
+            "      <BoundingBoxMax>1 1 1</BoundingBoxMax>\n"
+            f"      <Correlation>{e.get('r', 0.95)}</Correlation>\n"
+            f"      <Hash>{e['hash']}</Hash>\n"
+            "    </PairwiseResult>\n"
+        )
+    sr += "  </StitchingResults>\n"
+    text = text.replace("</SpimData>", sr + "</SpimData>")
+    open(path, "w").write(text)
+
+
+def model_translations(path):
+    """Effective (outermost-first concatenated) translation per setup."""
+    tree = ET.parse(path)
+    out = {}
+    for vr in tree.getroot().iter("ViewRegistration"):
+        setup = int(vr.get("setup"))
+        m = np.eye(4)
+        first = True
+        for vt in vr.findall("ViewTransform"):
+            a = np.fromstring(vt.find("affine").text, sep=" ").reshape(3, 4)
+            m4 = np.vstack([a, [0, 0, 0, 1]])
+            m = m4 if first else m @ m4
+            first = False
+        out[setup] = m[:3, 3]
+    return out
+
+
+def test_solver_translation_solve(tmp_path):
+    xml = os.path.join(str(tmp_path), "dataset.xml")
+    # link says B's content appears at ws = -e relative to registrations;
+    # solver should move B by +e = -ws
+    e = (2.5, -1.5, 1.0)
+    hash_ok = 3.0 + (3.0 + 40.0)  # [PIN-HASH] sums of both 3x4 models
+    write_xml_with_links(
+        xml,
+        [dict(a="0,0", b="0,1", ws=(-e[0], -e[1], -e[2]), hash=hash_ok)],
+    )
+    r = run([os.path.join(BIN, "solver"), "-x", xml])
+    assert r.returncode == 0, r.stderr + r.stdout
+    t = model_translations(xml)
+    assert np.allclose(t[0], [0, 0, 0], atol=1e-9)  # fixed view
+    assert np.allclose(t[1], [40 + e[0], e[1], e[2]], atol=1e-6), t[1]
+
+
+def test_solver_drops_stale_hash(tmp_path):
+    xml = os.path.join(str(tmp_path), "dataset.xml")
+    write_xml_with_links(
+        xml, [dict(a="0,0", b="0,1", ws=(-2.0, 0.0, 0.0), hash=123.456)]
+    )
+    r = run([os.path.join(BIN, "solver"), "-x", xml])
+    assert r.returncode == 1  # nothing to solve after the stale drop
+    assert "1 stale-hash" in r.stdout
+
+
+def test_solver_drops_low_r(tmp_path):
+    xml = os.path.join(str(tmp_path), "dataset.xml")
+    hash_ok = 3.0 + 43.0
+    write_xml_with_links(
+        xml,
+        [dict(a="0,0", b="0,1", ws=(-2.0, 0.0, 0.0), hash=hash_ok, r=0.1)],
+    )
+    r = run([os.path.join(BIN, "solver"), "-x", xml])
+    assert r.returncode == 1
+    assert "1 below minR" in r.stdout
+
+
+def test_solver_three_view_chain(tmp_path):
+    """0-1 and 1-2 links chain; adjustments accumulate."""
+    xml = os.path.join(str(tmp_path), "dataset.xml")
+    n5util.make_dataset_xml(
+        xml, "input.n5",
+        [dict(id=0, dims=(64, 64, 64), pos=(0.0, 0.0, 0.0)),
+         dict(id=1, dims=(64, 64, 64), pos=(40.0, 0.0, 0.0)),
+         dict(id=2, dims=(64, 64, 64), pos=(80.0, 0.0, 0.0))],
+    )
+    h01 = 3.0 + 43.0
+    h12 = 43.0 + 83.0
+    text = open(xml).read()
+    sr = "  <StitchingResults>\n"
+    for a, b, ws, h in [("0,0", "0,1", -1.0, h01),
+                        ("0,1", "0,2", -2.0, h12)]:
+        sr += (
+            "    <PairwiseResult>\n"
+            f"      <ViewIdsA>{a}</ViewIdsA>\n      <ViewIdsB>{b}</ViewIdsB>\n"
+            f"      <Matrix>1 0 0 {ws} 0 1 0 0 0 0 1 0</Matrix>\n"
+            "      <BoundingBoxMin>0 0 0</BoundingBoxMin>\n"
+            "      <BoundingBoxMax>1 1 1</BoundingBoxMax>\n"
+            "      <Correlation>0.9</Correlation>\n"
+            f"      <Hash>{h}</Hash>\n    </PairwiseResult>\n"
+        )
+    sr += "  </StitchingResults>\n"
+    open(xml, "w").write(text.replace("</SpimData>", sr + "</SpimData>"))
+    r = run([os.path.join(BIN, "solver"), "-x", xml])
+    assert r.returncode == 0, r.stderr + r.stdout
+    t = model_translations(xml)
+    assert np.allclose(t[1], [41.0, 0, 0], atol=1e-6)
+    assert np.allclose(t[2], [83.0, 0, 0], atol=1e-6)
