@@ -115,11 +115,8 @@ int main(int argc, char **argv) {
       }
       bs_fuse_view fv{};
       fv.view_id = s.id;
-      /* world = bbox coords: shift model by -bbmin */
+      /* world coords; bs_fuse_volume shifts by -vol_min itself */
       for (int i = 0; i < 12; ++i) fv.affine[i] = r->second[i];
-      fv.affine[3] -= bbmin[0];
-      fv.affine[7] -= bbmin[1];
-      fv.affine[11] -= bbmin[2];
       for (int d2 = 0; d2 < 3; ++d2) {
         fv.blend_border[d2] = bborder;
         fv.blend_range[d2] = brange;
