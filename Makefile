@@ -21,7 +21,7 @@ clean:
 
 # ---- host surface: N5/XML/SpimData + CLI binaries ----
 HOSTDIR = bigstitcher_spark_amd/csrc/host
-HOSTOBJS = $(HOSTDIR)/bs_json.o $(HOSTDIR)/bs_n5.o $(HOSTDIR)/bs_xml.o $(HOSTDIR)/bs_spimdata.o
+HOSTOBJS = $(HOSTDIR)/bs_json.o $(HOSTDIR)/bs_n5.o $(HOSTDIR)/bs_zarr.o $(HOSTDIR)/bs_xml.o $(HOSTDIR)/bs_spimdata.o
 BINDIR = bigstitcher_spark_amd/bin
 CXX_HOST = g++
 HOSTFLAGS = -O2 -std=c++17 -fPIC -Wall

@@ -9,9 +9,7 @@
 #include <cstring>
 #include <fstream>
 
-namespace bsn5 {
-
-namespace {
+namespace bsio {
 
 bool mkdirs(const std::string &path) {
   std::string cur;
@@ -77,9 +75,10 @@ void byteswap(std::string &buf, size_t esz) {
   }
 }
 
-bool gzip_deflate(const std::string &in, std::string *out) {
+bool gzip_deflate(const std::string &in, std::string *out, int level,
+                  bool raw_zlib) {
   z_stream zs{};
-  if (deflateInit2(&zs, Z_DEFAULT_COMPRESSION, Z_DEFLATED, 15 + 16, 8,
+  if (deflateInit2(&zs, level, Z_DEFLATED, raw_zlib ? 15 : 15 + 16, 8,
                    Z_DEFAULT_STRATEGY) != Z_OK)
     return false;
   out->resize(deflateBound(&zs, in.size()));
@@ -110,7 +109,19 @@ bool gzip_inflate(const unsigned char *in, size_t n, std::string *out,
   return true;
 }
 
-}  // namespace
+}  // namespace bsio
+
+namespace bsn5 {
+using bsio::gzip_deflate;
+using bsio::gzip_inflate;
+using bsio::mkdirs;
+using bsio::read_file;
+using bsio::write_file;
+using bsio::put_be16;
+using bsio::put_be32;
+using bsio::get_be16;
+using bsio::get_be32;
+using bsio::byteswap;
 
 size_t dtype_size(const std::string &dtype) {
   if (dtype == "uint8" || dtype == "int8") return 1;

@@ -20,6 +20,16 @@
 
 #include "bs_json.h"
 
+namespace bsio { /* shared by the N5 and ZARR writers */
+bool mkdirs(const std::string &path);
+bool read_file(const std::string &p, std::string *out);
+bool write_file(const std::string &p, const std::string &data);
+bool gzip_deflate(const std::string &in, std::string *out, int level = -1,
+                  bool raw_zlib = false);
+bool gzip_inflate(const unsigned char *in, size_t n, std::string *out,
+                  size_t expected);
+}  // namespace bsio
+
 namespace bsn5 {
 
 struct DatasetAttrs {

@@ -1,0 +1,135 @@
+#include "bs_zarr.h"
+
+#include <cstring>
+
+#include "bs_n5.h" /* bsio helpers */
+
+namespace bszarr {
+
+using bsio::gzip_deflate;
+using bsio::mkdirs;
+using bsio::read_file;
+using bsio::write_file;
+
+size_t dtype_size(const std::string &dtype) {
+  if (dtype == "|u1" || dtype == "|i1") return 1;
+  if (dtype == "<u2" || dtype == "<i2") return 2;
+  if (dtype == "<u4" || dtype == "<i4" || dtype == "<f4") return 4;
+  if (dtype == "<u8" || dtype == "<i8" || dtype == "<f8") return 8;
+  return 0;
+}
+
+bool Container::create() {
+  if (!mkdirs(root_)) return false;
+  auto g = bsj::Value::mkobj();
+  g->obj["zarr_format"] = bsj::Value::mkint(2);
+  std::string existing;
+  if (read_file(root_ + "/.zgroup", &existing)) return true;
+  return write_file(root_ + "/.zgroup", bsj::dump(g));
+}
+
+bool Container::set_root_attr(const std::string &key, bsj::ValuePtr v) {
+  std::string p = root_ + "/.zattrs";
+  std::string text;
+  bsj::ValuePtr a;
+  if (read_file(p, &text)) a = bsj::parse(text);
+  if (!a || a->type != bsj::Value::OBJ) a = bsj::Value::mkobj();
+  bsj::set_path(a, key, v);
+  return write_file(p, bsj::dump(a));
+}
+
+bsj::ValuePtr Container::get_root_attr(const std::string &key) const {
+  std::string text;
+  if (!read_file(root_ + "/.zattrs", &text)) return nullptr;
+  auto a = bsj::parse(text);
+  if (!a) return nullptr;
+  return bsj::get_path(a, key);
+}
+
+bool Container::create_array(const std::string &name, const ArrayAttrs &a) {
+  std::string dir = root_ + "/" + name;
+  if (!mkdirs(dir)) return false;
+  auto z = bsj::Value::mkobj();
+  z->obj["zarr_format"] = bsj::Value::mkint(2);
+  z->obj["shape"] = bsj::Value::mkints(a.shape);
+  z->obj["chunks"] = bsj::Value::mkints(a.chunks);
+  z->obj["dtype"] = bsj::Value::mkstr(a.dtype);
+  z->obj["order"] = bsj::Value::mkstr("C");
+  z->obj["fill_value"] = bsj::Value::mkint(0);
+  z->obj["filters"] = bsj::Value::mknull();
+  z->obj["dimension_separator"] = bsj::Value::mkstr(".");
+  if (a.gzip) {
+    auto comp = bsj::Value::mkobj();
+    comp->obj["id"] = bsj::Value::mkstr("gzip");
+    comp->obj["level"] = bsj::Value::mkint(1);
+    z->obj["compressor"] = comp;
+  } else {
+    z->obj["compressor"] = bsj::Value::mknull();
+  }
+  return write_file(dir + "/.zarray", bsj::dump(z));
+}
+
+bool Container::get_array_attrs(const std::string &name,
+                                ArrayAttrs *out) const {
+  std::string text;
+  if (!read_file(root_ + "/" + name + "/.zarray", &text)) return false;
+  auto z = bsj::parse(text);
+  if (!z) return false;
+  auto sh = bsj::get_path(z, "shape"), ch = bsj::get_path(z, "chunks");
+  auto dt = bsj::get_path(z, "dtype");
+  if (!sh || !ch || !dt) return false;
+  out->shape.clear();
+  for (auto &e : sh->arr) out->shape.push_back(e->inum);
+  out->chunks.clear();
+  for (auto &e : ch->arr) out->chunks.push_back((int)e->inum);
+  out->dtype = dt->str;
+  auto comp = bsj::get_path(z, "compressor");
+  out->gzip = comp && comp->type == bsj::Value::OBJ;
+  return true;
+}
+
+bool Container::write_chunk(const std::string &name, const ArrayAttrs &a,
+                            const std::vector<long long> &grid_pos,
+                            const void *data,
+                            const std::vector<int> &clipped) {
+  size_t esz = dtype_size(a.dtype);
+  size_t nd = a.shape.size();
+  if (!esz || clipped.size() != nd || grid_pos.size() != nd) return false;
+  /* zarr edge chunks are full-size, zero-padded: scatter the clipped
+   * C-order payload into the full chunk buffer */
+  size_t full = esz;
+  for (int c : a.chunks) full *= (size_t)c;
+  std::string payload(full, '\0');
+  /* innermost (last) dim is contiguous in both */
+  size_t row = (size_t)clipped[nd - 1] * esz;
+  size_t nrows = 1;
+  for (size_t d = 0; d + 1 < nd; ++d) nrows *= (size_t)clipped[d];
+  for (size_t r = 0; r < nrows; ++r) {
+    /* decode r over the outer clipped dims -> full-chunk element offset */
+    size_t rem = r, off = 0;
+    for (size_t d = 0; d + 1 < nd; ++d) {
+      size_t stride_c = 1, stride_f = 1;
+      for (size_t k = d + 1; k + 1 < nd; ++k) stride_c *= (size_t)clipped[k];
+      for (size_t k = d + 1; k < nd; ++k) stride_f *= (size_t)a.chunks[k];
+      size_t idx = rem / stride_c;
+      rem %= stride_c;
+      off += idx * stride_f;
+    }
+    memcpy(&payload[off * esz], (const char *)data + r * row, row);
+  }
+  std::string body;
+  if (a.gzip) {
+    /* zarr "gzip" codec = gzip container */
+    if (!gzip_deflate(payload, &body, 1, false)) return false;
+  } else {
+    body = payload;
+  }
+  std::string key;
+  for (size_t d = 0; d < nd; ++d) {
+    if (d) key += ".";
+    key += std::to_string(grid_pos[d]);
+  }
+  return write_file(root_ + "/" + name + "/" + key, body);
+}
+
+}  // namespace bszarr

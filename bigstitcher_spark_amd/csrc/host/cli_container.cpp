@@ -13,6 +13,7 @@
 
 #include "bs_cli_util.h"
 #include "bs_n5.h"
+#include "bs_zarr.h"
 #include "bs_spimdata.h"
 
 int main(int argc, char **argv) {
@@ -25,7 +26,7 @@ int main(int argc, char **argv) {
             "usage: create-fusion-container -x dataset.xml -o out.n5 "
             "[--blockSize 128,128,128] [--dataType UINT16|UINT8|FLOAT32] "
             "[--minIntensity v --maxIntensity v] [--bbMin x,y,z --bbMax "
-            "x,y,z] [--compression GZIP|RAW] "
+            "x,y,z] [--compression GZIP|RAW] [--storage N5|ZARR] "
             "[--downsamplings \"1,1,1;2,2,2;4,4,4\"]\n");
     return 2;
   }
@@ -73,16 +74,21 @@ int main(int argc, char **argv) {
   long long dims[3] = {bbmax[0] - bbmin[0] + 1, bbmax[1] - bbmin[1] + 1,
                        bbmax[2] - bbmin[2] + 1};
   int numTp = (int)sd.timepoints.size(), numCh = 1;
+  const bool zarr = args.get("storage", "N5") == "ZARR";
 
   bsn5::Container n5(args.get("n5Path"));
-  if (!n5.create()) {
+  bszarr::Container zr(args.get("n5Path"));
+  if (zarr ? !zr.create() : !n5.create()) {
     fprintf(stderr, "cannot create %s\n", args.get("n5Path").c_str());
     return 1;
   }
   auto set = [&](const std::string &k, bsj::ValuePtr v) {
-    n5.set_attr("", "Bigstitcher-Spark/" + k, v);
+    if (zarr)
+      zr.set_root_attr("Bigstitcher-Spark/" + k, v);
+    else
+      n5.set_attr("", "Bigstitcher-Spark/" + k, v);
   };
-  set("FusionFormat", bsj::Value::mkstr("N5"));
+  set("FusionFormat", bsj::Value::mkstr(zarr ? "OME-ZARR" : "N5"));
   set("InputXML", bsj::Value::mkstr(args.get("xml")));
   set("NumTimepoints", bsj::Value::mkint(numTp));
   set("NumChannels", bsj::Value::mkint(numCh));
@@ -120,36 +126,108 @@ int main(int argc, char **argv) {
     }
   }
   auto mri_all = bsj::Value::mkarr();
-  for (int t = 0; t < numTp; ++t)
-    for (int ch = 0; ch < numCh; ++ch) {
-      auto levels = bsj::Value::mkarr();
-      for (size_t l = 0; l < ladder.size(); ++l) {
-        char dsname[64];
-        snprintf(dsname, sizeof dsname, "ch%dtp%d/s%zu", ch, t, l);
-        bsn5::DatasetAttrs da;
-        da.dims = {(dims[0] + ladder[l][0] - 1) / ladder[l][0],
-                   (dims[1] + ladder[l][1] - 1) / ladder[l][1],
-                   (dims[2] + ladder[l][2] - 1) / ladder[l][2]};
-        da.block = {(int)bs[0], (int)bs[1], (int)bs[2]};
-        da.dtype = n5dt;
-        da.compression = comp;
-        if (!n5.create_dataset(dsname, da)) {
-          fprintf(stderr, "cannot create dataset %s\n", dsname);
-          return 1;
-        }
+  if (zarr) {
+    /* ONE 5-D [t,c,z,y,x] array per level, OME-NGFF v0.4 multiscales
+     * (reference CreateFusionContainer.java:331-389) */
+    std::string zdt = n5dt == "uint8" ? "|u1"
+                      : n5dt == "float32" ? "<f4" : "<u2";
+    auto msets = bsj::Value::mkarr();
+    for (size_t l = 0; l < ladder.size(); ++l) {
+      char dsname[32];
+      snprintf(dsname, sizeof dsname, "s%zu", l);
+      bszarr::ArrayAttrs za;
+      za.shape = {numTp, numCh,
+                  (dims[2] + ladder[l][2] - 1) / ladder[l][2],
+                  (dims[1] + ladder[l][1] - 1) / ladder[l][1],
+                  (dims[0] + ladder[l][0] - 1) / ladder[l][0]};
+      za.chunks = {1, 1, (int)bs[2], (int)bs[1], (int)bs[0]};
+      za.dtype = zdt;
+      za.gzip = comp == "gzip";
+      if (!zr.create_array(dsname, za)) {
+        fprintf(stderr, "cannot create array %s\n", dsname);
+        return 1;
+      }
+      auto dset = bsj::Value::mkobj();
+      dset->obj["path"] = bsj::Value::mkstr(dsname);
+      auto cts = bsj::Value::mkarr();
+      auto sc = bsj::Value::mkobj();
+      sc->obj["type"] = bsj::Value::mkstr("scale");
+      auto scale = bsj::Value::mkarr();
+      scale->arr.push_back(bsj::Value::mknum(1.0));
+      scale->arr.push_back(bsj::Value::mknum(1.0));
+      scale->arr.push_back(bsj::Value::mknum((double)ladder[l][2]));
+      scale->arr.push_back(bsj::Value::mknum((double)ladder[l][1]));
+      scale->arr.push_back(bsj::Value::mknum((double)ladder[l][0]));
+      sc->obj["scale"] = scale;
+      cts->arr.push_back(sc);
+      dset->obj["coordinateTransformations"] = cts;
+      msets->arr.push_back(dset);
+      /* one shared MultiResolutionInfos entry set (all t,c share it) */
+      if (l == 0)
+        for (int vi = 0; vi < numTp * numCh; ++vi)
+          mri_all->arr.push_back(bsj::Value::mkarr());
+      for (int vi = 0; vi < numTp * numCh; ++vi) {
         auto lv = bsj::Value::mkobj();
         lv->obj["dataset"] = bsj::Value::mkstr(dsname);
-        lv->obj["dimensions"] = bsj::Value::mkints(da.dims);
-        lv->obj["blockSize"] =
-            bsj::Value::mkints(std::vector<int>{da.block[0], da.block[1],
-                                                da.block[2]});
+        lv->obj["dimensions"] = bsj::Value::mkints(
+            std::vector<long long>{za.shape[4], za.shape[3], za.shape[2]});
+        lv->obj["blockSize"] = bsj::Value::mkints(
+            std::vector<int>{(int)bs[0], (int)bs[1], (int)bs[2]});
         lv->obj["absoluteDownsampling"] = bsj::Value::mkints(
             std::vector<long long>{ladder[l][0], ladder[l][1],
                                    ladder[l][2]});
-        levels->arr.push_back(lv);
+        mri_all->arr[vi]->arr.push_back(lv);
       }
-      mri_all->arr.push_back(levels);
     }
+    auto ms = bsj::Value::mkobj();
+    ms->obj["version"] = bsj::Value::mkstr("0.4");
+    ms->obj["name"] = bsj::Value::mkstr("fused");
+    auto axes = bsj::Value::mkarr();
+    const char *axn[5] = {"t", "c", "z", "y", "x"};
+    const char *axt[5] = {"time", "channel", "space", "space", "space"};
+    for (int i = 0; i < 5; ++i) {
+      auto ax = bsj::Value::mkobj();
+      ax->obj["name"] = bsj::Value::mkstr(axn[i]);
+      ax->obj["type"] = bsj::Value::mkstr(axt[i]);
+      axes->arr.push_back(ax);
+    }
+    ms->obj["axes"] = axes;
+    ms->obj["datasets"] = msets;
+    auto msl = bsj::Value::mkarr();
+    msl->arr.push_back(ms);
+    zr.set_root_attr("multiscales", msl);
+  } else {
+    for (int t = 0; t < numTp; ++t)
+      for (int ch = 0; ch < numCh; ++ch) {
+        auto levels = bsj::Value::mkarr();
+        for (size_t l = 0; l < ladder.size(); ++l) {
+          char dsname[64];
+          snprintf(dsname, sizeof dsname, "ch%dtp%d/s%zu", ch, t, l);
+          bsn5::DatasetAttrs da;
+          da.dims = {(dims[0] + ladder[l][0] - 1) / ladder[l][0],
+                     (dims[1] + ladder[l][1] - 1) / ladder[l][1],
+                     (dims[2] + ladder[l][2] - 1) / ladder[l][2]};
+          da.block = {(int)bs[0], (int)bs[1], (int)bs[2]};
+          da.dtype = n5dt;
+          da.compression = comp;
+          if (!n5.create_dataset(dsname, da)) {
+            fprintf(stderr, "cannot create dataset %s\n", dsname);
+            return 1;
+          }
+          auto lv = bsj::Value::mkobj();
+          lv->obj["dataset"] = bsj::Value::mkstr(dsname);
+          lv->obj["dimensions"] = bsj::Value::mkints(da.dims);
+          lv->obj["blockSize"] =
+              bsj::Value::mkints(std::vector<int>{da.block[0], da.block[1],
+                                                  da.block[2]});
+          lv->obj["absoluteDownsampling"] = bsj::Value::mkints(
+              std::vector<long long>{ladder[l][0], ladder[l][1],
+                                     ladder[l][2]});
+          levels->arr.push_back(lv);
+        }
+        mri_all->arr.push_back(levels);
+      }
+  }
   set("MultiResolutionInfos", mri_all);
   printf("created %s: %d tp x %d ch, bbox [%lld,%lld,%lld]..[%lld,%lld,%lld]"
          ", %s %s\n",

@@ -14,6 +14,7 @@
 #include "../../../include/bigstitch.h"
 #include "bs_cli_util.h"
 #include "bs_n5.h"
+#include "bs_zarr.h"
 #include "bs_spimdata.h"
 
 int main(int argc, char **argv) {
@@ -29,8 +30,11 @@ int main(int argc, char **argv) {
     return 2;
   }
   bsn5::Container n5(args.get("n5Path"));
+  bszarr::Container zr(args.get("n5Path"));
   auto geta = [&](const std::string &k) {
-    return n5.get_attr("", "Bigstitcher-Spark/" + k);
+    auto v = n5.get_attr("", "Bigstitcher-Spark/" + k);
+    if (!v) v = zr.get_root_attr("Bigstitcher-Spark/" + k);
+    return v;
   };
   auto fmt = geta("FusionFormat");
   if (!fmt) {
@@ -46,6 +50,7 @@ int main(int argc, char **argv) {
     auto x = geta("InputXML");
     if (x) xml = x->str;
   }
+  const bool zarr = fmt->str.find("ZARR") != std::string::npos;
   auto bbmin_a = geta("Boundingbox_min"), bbmax_a = geta("Boundingbox_max");
   auto bs_a = geta("BlockSize");
   auto dt_a = geta("DataType");
@@ -169,17 +174,34 @@ int main(int argc, char **argv) {
       fprintf(stderr, "fusion failed: %s\n", bs_last_error(ctx));
       return 1;
     }
-    /* write every level's N5 chunks */
+    /* write every level's chunks (N5 3-D datasets or OME-ZARR 5-D
+     * arrays — the reference's 3-D-block-into-5-D lift,
+     * SparkAffineFusion.java:630-643) */
     for (int l = 0; l < nlevels; ++l) {
       bsn5::DatasetAttrs da;
-      if (!n5.get_dataset_attrs(dsnames[l], &da)) {
-        fprintf(stderr, "missing dataset %s in container\n",
-                dsnames[l].c_str());
-        return 1;
+      bszarr::ArrayAttrs za;
+      int bx, by, bz;
+      if (zarr) {
+        if (!zr.get_array_attrs(dsnames[l], &za)) {
+          fprintf(stderr, "missing array %s in container\n",
+                  dsnames[l].c_str());
+          return 1;
+        }
+        bx = za.chunks[4];
+        by = za.chunks[3];
+        bz = za.chunks[2];
+      } else {
+        if (!n5.get_dataset_attrs(dsnames[l], &da)) {
+          fprintf(stderr, "missing dataset %s in container\n",
+                  dsnames[l].c_str());
+          return 1;
+        }
+        bx = da.block[0];
+        by = da.block[1];
+        bz = da.block[2];
       }
       long long lx = ldims[l * 3], ly = ldims[l * 3 + 1],
                 lz = ldims[l * 3 + 2];
-      int bx = da.block[0], by = da.block[1], bz = da.block[2];
       std::vector<char> blk((size_t)bx * by * bz * esz2);
       long long nwritten = 0;
       for (long long gz = 0; gz * bz < lz; ++gz)
@@ -195,8 +217,14 @@ int main(int argc, char **argv) {
                        src + (((gz * bz + z) * ly + gy * by + y) * lx +
                               gx * bx) * esz2,
                        (size_t)cx * esz2);
-            if (!n5.write_block(dsnames[l], da, {gx, gy, gz}, blk.data(),
-                                {cx, cy, cz})) {
+            bool ok;
+            if (zarr)
+              ok = zr.write_chunk(dsnames[l], za, {ti, 0, gz, gy, gx},
+                                  blk.data(), {1, 1, cz, cy, cx});
+            else
+              ok = n5.write_block(dsnames[l], da, {gx, gy, gz}, blk.data(),
+                                  {cx, cy, cz});
+            if (!ok) {
               fprintf(stderr, "block write failed\n");
               return 1;
             }

@@ -121,3 +121,43 @@ def make_dataset_xml(path, n5_rel, setups):
         )
     with open(path, "w") as f:
         f.write(DATASET_XML.format(n5=n5_rel, setups=s_xml, regs=r_xml))
+
+
+def zarr_root_attrs(root):
+    with open(os.path.join(root, ".zattrs")) as f:
+        return json.load(f)
+
+
+_ZDTYPES = {"|u1": np.uint8, "<u2": np.dtype("<u2"), "<f4": np.dtype("<f4")}
+
+
+def read_zarr(root, name):
+    """Read a zarr v2 array written by bs_zarr.cpp (gzip or raw,
+    dimension_separator '.', full-size edge chunks)."""
+    ds = os.path.join(root, name)
+    with open(os.path.join(ds, ".zarray")) as f:
+        za = json.load(f)
+    shape, chunks = za["shape"], za["chunks"]
+    dt = np.dtype(_ZDTYPES[za["dtype"]])
+    comp = za.get("compressor")
+    out = np.zeros(shape, dt)
+    ngrid = [(s + c - 1) // c for s, c in zip(shape, chunks)]
+    import itertools
+
+    for idx in itertools.product(*[range(n) for n in ngrid]):
+        key = ".".join(str(i) for i in idx)
+        p = os.path.join(ds, key)
+        if not os.path.exists(p):
+            continue
+        raw = open(p, "rb").read()
+        if comp and comp.get("id") == "gzip":
+            raw = gzip.decompress(raw)
+        blk = np.frombuffer(raw, dt).reshape(chunks)
+        sl_out, sl_blk = [], []
+        for d, (i, c, s) in enumerate(zip(idx, chunks, shape)):
+            lo = i * c
+            hi = min(lo + c, s)
+            sl_out.append(slice(lo, hi))
+            sl_blk.append(slice(0, hi - lo))
+        out[tuple(sl_out)] = blk[tuple(sl_blk)]
+    return out, za

@@ -195,3 +195,62 @@ def test_cli_full_pipeline_stitch_solve_fuse(tmp_path):
                         of.FUSION_AVG_BLEND, out_dtype=np.float32)
     denom = np.maximum(np.abs(ref), 1.0)
     assert np.max(np.abs(fused - ref) / denom) < 1e-4
+
+
+def test_container_cli_zarr(tmp_path):
+    """OME-ZARR storage: .zgroup/.zattrs metadata contract, 5-D
+    [t,c,z,y,x] arrays with OME-NGFF v0.4 multiscales (reference
+    CreateFusionContainer.java:331-389)."""
+    xml, n5, _err, _ = make_grid_dataset(str(tmp_path))
+    out = os.path.join(str(tmp_path), "fused.zarr")
+    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml, "-o",
+             out, "--blockSize", "32,32,32", "--dataType", "UINT16",
+             "--storage", "ZARR", "--downsamplings", "1,1,1;2,2,2"])
+    assert r.returncode == 0, r.stderr
+    za = n5util.zarr_root_attrs(out)
+    bs = za["Bigstitcher-Spark"]
+    assert bs["FusionFormat"] == "OME-ZARR"
+    assert bs["Boundingbox_max"] == [103, 63, 63]
+    ms = za["multiscales"][0]
+    assert ms["version"] == "0.4"
+    assert [a["name"] for a in ms["axes"]] == ["t", "c", "z", "y", "x"]
+    assert [d["path"] for d in ms["datasets"]] == ["s0", "s1"]
+    assert ms["datasets"][1]["coordinateTransformations"][0]["scale"] == \
+        [1.0, 1.0, 2.0, 2.0, 2.0]
+    import json as _json
+
+    zarr_meta = _json.load(open(os.path.join(out, "s0", ".zarray")))
+    assert zarr_meta["shape"] == [1, 1, 64, 64, 104]
+    assert zarr_meta["chunks"] == [1, 1, 32, 32, 32]
+    assert zarr_meta["dtype"] == "<u2"
+
+
+@pytest.mark.gpu
+def test_cli_fusion_zarr_end_to_end(tmp_path):
+    xml, n5, err, (a, b) = make_grid_dataset(str(tmp_path))
+    out = os.path.join(str(tmp_path), "fused.zarr")
+    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml, "-o",
+             out, "--blockSize", "32,32,32", "--dataType", "FLOAT32",
+             "--storage", "ZARR", "--downsamplings", "1,1,1;2,2,2"])
+    assert r.returncode == 0, r.stderr
+    r = run([os.path.join(BIN, "affine-fusion"), "-o", out,
+             "--fusionType", "AVG_BLEND", "--blendingRange", "8"])
+    assert r.returncode == 0, r.stderr + r.stdout
+    s0, za = n5util.read_zarr(out, "s0")
+    assert s0.shape == (1, 1, 64, 64, 104)
+    fused = s0[0, 0]
+    ident = np.hstack([np.eye(3), np.zeros((3, 1))])
+    affB = ident.copy()
+    affB[0, 3] = 40.0
+    views = [
+        dict(data=a, affine=ident, border=(0, 0, 0), range=(8, 8, 8)),
+        dict(data=b, affine=affB, border=(0, 0, 0), range=(8, 8, 8)),
+    ]
+    ref = of.fuse_block(views, (0, 0, 0), (104, 64, 64),
+                        of.FUSION_AVG_BLEND, out_dtype=np.float32)
+    denom = np.maximum(np.abs(ref), 1.0)
+    assert np.max(np.abs(fused - ref) / denom) < 1e-4
+    s1, _ = n5util.read_zarr(out, "s1")
+    ref1 = of.downsample_level(fused, (2, 2, 2))
+    rd = np.abs(s1[0, 0] - ref1) / np.maximum(np.abs(ref1), 1.0)
+    assert rd.max() < 1e-5
