@@ -29,7 +29,7 @@ HOSTFLAGS = -O2 -std=c++17 -fPIC -Wall
 $(HOSTDIR)/%.o: $(HOSTDIR)/%.cpp $(HOSTDIR)/%.h
 	$(CXX_HOST) $(HOSTFLAGS) -c $< -o $@
 
-cli: $(BINDIR)/stitching $(BINDIR)/create-fusion-container $(BINDIR)/affine-fusion $(BINDIR)/solver
+cli: $(BINDIR)/stitching $(BINDIR)/create-fusion-container $(BINDIR)/affine-fusion $(BINDIR)/solver $(BINDIR)/resave
 
 $(BINDIR)/stitching: $(HOSTDIR)/cli_stitching.cpp $(HOSTOBJS) $(LIB)
 	@mkdir -p $(BINDIR)
@@ -46,6 +46,10 @@ $(BINDIR)/affine-fusion: $(HOSTDIR)/cli_fusion.cpp $(HOSTOBJS) $(LIB)
 $(BINDIR)/solver: $(HOSTDIR)/cli_solver.cpp $(HOSTOBJS)
 	@mkdir -p $(BINDIR)
 	$(CXX_HOST) $(HOSTFLAGS) -Iinclude $< $(HOSTOBJS) -o $@ -lz
+
+$(BINDIR)/resave: $(HOSTDIR)/cli_resave.cpp $(HOSTOBJS) $(LIB)
+	@mkdir -p $(BINDIR)
+	$(CXX_HOST) $(HOSTFLAGS) -Iinclude $< $(HOSTOBJS) -o $@ -Lbigstitcher_spark_amd -lbigstitch -lz -L/opt/rocm/lib -lamdhip64 -Wl,-rpath,'$$ORIGIN/..' -Wl,-rpath,/opt/rocm/lib
 
 clean-cli:
 	rm -f $(HOSTOBJS) $(BINDIR)/*

@@ -254,3 +254,37 @@ def test_cli_fusion_zarr_end_to_end(tmp_path):
     ref1 = of.downsample_level(fused, (2, 2, 2))
     rd = np.abs(s1[0, 0] - ref1) / np.maximum(np.abs(ref1), 1.0)
     assert rd.max() < 1e-5
+
+
+def test_resave_cli_missing_args():
+    r = run([os.path.join(BIN, "resave")])
+    assert r.returncode == 2
+
+
+@pytest.mark.gpu
+def test_cli_resave_end_to_end(tmp_path):
+    """resave (§8(f) row 2): re-chunk + per-view GPU pyramid into a new
+    bdv.n5 container; level voxels match the oracle box-mean chain and
+    s0 is an exact round trip."""
+    xml, n5, err, (a, b) = make_grid_dataset(str(tmp_path))
+    out = os.path.join(str(tmp_path), "resaved.n5")
+    xo = os.path.join(str(tmp_path), "resaved.xml")
+    r = run([os.path.join(BIN, "resave"), "-x", xml, "-o", out, "-xo", xo,
+             "--blockSize", "32,32,16", "--downsamplings", "1,1,1;2,2,1"])
+    assert r.returncode == 0, r.stderr + r.stdout
+    s0, attrs = n5util.read_dataset(out, "setup0/timepoint0/s0")
+    assert attrs["blockSize"] == [32, 32, 16]
+    assert np.array_equal(s0, a)  # exact round trip
+    s1, _ = n5util.read_dataset(out, "setup1/timepoint0/s1")
+    ref1 = of.downsample_level(b, (2, 2, 1))
+    d = np.abs(s1.astype(np.int64) - ref1.astype(np.int64))
+    assert d.max() <= 1  # .5-boundary rounding fp32 vs fp64
+    # the rewritten XML points at the new container and still stitches
+    tree = ET.parse(xo)
+    n5node = tree.getroot().find(".//ImageLoader/n5")
+    assert n5node.get("type") == "absolute" and n5node.text == out
+    r = run([os.path.join(BIN, "stitching"), "-x", xo, "-ds", "1,1,1",
+             "--minOverlapRatio", "0.05"])
+    assert r.returncode == 0, r.stderr + r.stdout
+    prs = ET.parse(xo).getroot().findall(".//StitchingResults/PairwiseResult")
+    assert len(prs) == 1
