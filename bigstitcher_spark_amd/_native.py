@@ -79,6 +79,9 @@ class _Stats(C.Structure):
 FUSION_AVG = 0
 FUSION_AVG_BLEND = 1
 FUSION_MAX_INTENSITY = 2
+FUSION_LOWEST_VIEWID_WINS = 3
+FUSION_HIGHEST_VIEWID_WINS = 4
+FUSION_CLOSEST_PIXEL_WINS = 5
 OUT_DTYPES = {np.dtype(np.float32): 0, np.dtype(np.uint16): 1,
               np.dtype(np.uint8): 2}
 

@@ -633,7 +633,8 @@ __global__ __launch_bounds__(256) void k_fuse(
    for (int x = threadIdx.x; x < bx; x += blockDim.x) {
     long i = out_off + (long)z * out_slice + (long)y * out_row + x;
     float wx = (float)(bmx + x), wy = (float)(bmy + y), wz = (float)(bmz + z);
-    float sum_wv = 0.0f, sum_w = 0.0f, vmax = 0.0f;
+    float sum_wv = 0.0f, sum_w = 0.0f, vmax = 0.0f, pick = 0.0f;
+    float best_dist = -1.0f;
     bool any = false;
     for (int k = 0; k < nv; ++k) {
       const bs_dev_view &v = k < nvs ? sv[k] : views[vidx[k]];
@@ -668,6 +669,22 @@ __global__ __launch_bounds__(256) void k_fuse(
       if (ftype == BS_FUSION_MAX_INTENSITY) {
         vmax = (any && vmax > val) ? vmax : val;
         any = true;
+      } else if (ftype == BS_FUSION_LOWEST_VIEWID_WINS) {
+        if (!any) pick = val;
+        any = true;
+      } else if (ftype == BS_FUSION_HIGHEST_VIEWID_WINS) {
+        pick = val;
+        any = true;
+      } else if (ftype == BS_FUSION_CLOSEST_PIXEL_WINS) {
+        float dx = fminf(px, (float)(v.nx - 1) - px);
+        float dy2 = fminf(py, (float)(v.ny - 1) - py);
+        float dz2 = fminf(pz, (float)(v.nz - 1) - pz);
+        float dist = fminf(dx, fminf(dy2, dz2));
+        if (dist > best_dist) {
+          best_dist = dist;
+          pick = val;
+        }
+        any = true;
       } else {
         sum_wv += w * val;
         sum_w += w;
@@ -677,6 +694,9 @@ __global__ __launch_bounds__(256) void k_fuse(
     bool covered;
     if (ftype == BS_FUSION_MAX_INTENSITY) {
       o = any ? vmax : 0.0f;
+      covered = any;
+    } else if (ftype >= BS_FUSION_LOWEST_VIEWID_WINS) {
+      o = any ? pick : 0.0f;
       covered = any;
     } else {
       covered = sum_w > 0.0f;

@@ -77,9 +77,13 @@ int main(int argc, char **argv) {
   if (out_dtype == BS_OUT_UINT8 && !maxI_a) maxI = 255;
 
   std::string ft = args.get("fusionType", "AVG_BLEND");
-  int fusion_type = ft == "AVG" ? BS_FUSION_AVG
-                    : ft == "MAX_INTENSITY" ? BS_FUSION_MAX_INTENSITY
-                                            : BS_FUSION_AVG_BLEND;
+  int fusion_type =
+      ft == "AVG" ? BS_FUSION_AVG
+      : ft == "MAX_INTENSITY" ? BS_FUSION_MAX_INTENSITY
+      : ft == "LOWEST_VIEWID_WINS" ? BS_FUSION_LOWEST_VIEWID_WINS
+      : ft == "HIGHEST_VIEWID_WINS" ? BS_FUSION_HIGHEST_VIEWID_WINS
+      : ft == "CLOSEST_PIXEL_WINS" ? BS_FUSION_CLOSEST_PIXEL_WINS
+                                   : BS_FUSION_AVG_BLEND;
   float brange = (float)args.getd("blendingRange", 40.0);
   float bborder = (float)args.getd("blendingBorder", 0.0);
 

@@ -122,11 +122,16 @@ int bs_stitch_batch(bs_ctx *ctx, const bs_pair_desc *pairs, size_t n,
 
 /* ------------------------------------------------------------------ fusion */
 
-/* FusionType enum — reference SparkAffineFusion.java:124-125 (subset; the
- * remaining enum values are SURVEY.md §8(f) row 4). */
+/* FusionType enum — reference SparkAffineFusion.java:124-125 (complete
+ * except intensity-coefficient and --masks modes; CLOSEST_PIXEL_WINS
+ * picks the view whose inverse-mapped point has the largest min-axis
+ * distance to its view border [PIN], ties to the first view). */
 #define BS_FUSION_AVG 0
 #define BS_FUSION_AVG_BLEND 1
 #define BS_FUSION_MAX_INTENSITY 2
+#define BS_FUSION_LOWEST_VIEWID_WINS 3  /* first contributing view */
+#define BS_FUSION_HIGHEST_VIEWID_WINS 4 /* last contributing view */
+#define BS_FUSION_CLOSEST_PIXEL_WINS 5  /* largest border distance */
 
 #define BS_OUT_FLOAT32 0
 #define BS_OUT_UINT16 1

@@ -39,6 +39,9 @@ import numpy as np
 FUSION_AVG = 0
 FUSION_AVG_BLEND = 1
 FUSION_MAX = 2
+FUSION_LOWEST_VIEWID = 3   # first contributing view wins
+FUSION_HIGHEST_VIEWID = 4  # last contributing view wins
+FUSION_CLOSEST_PIXEL = 5   # largest min-axis border distance wins [PIN]
 
 __all__ = [
     "fuse_block",
@@ -135,6 +138,8 @@ def fuse_block(
     sum_wv = np.zeros(w.shape[:-1], dtype=np.float64)
     sum_w = np.zeros(w.shape[:-1], dtype=np.float64)
     vmax = np.zeros(w.shape[:-1], dtype=np.float64)
+    pick = np.zeros(w.shape[:-1], dtype=np.float64)
+    best_dist = np.full(w.shape[:-1], -1.0)
     any_view = np.zeros(w.shape[:-1], dtype=bool)
 
     for v in views:
@@ -160,12 +165,31 @@ def fuse_block(
         if fusion_type == FUSION_MAX:
             vmax = np.where(inside & (val > vmax), val, vmax)
             any_view |= inside
+        elif fusion_type == FUSION_LOWEST_VIEWID:
+            pick = np.where(inside & ~any_view, val, pick)
+            any_view |= inside
+        elif fusion_type == FUSION_HIGHEST_VIEWID:
+            pick = np.where(inside, val, pick)
+            any_view |= inside
+        elif fusion_type == FUSION_CLOSEST_PIXEL:
+            dist = np.minimum(
+                np.minimum(pc[..., 0], dims[0] - 1 - pc[..., 0]),
+                np.minimum(np.minimum(pc[..., 1], dims[1] - 1 - pc[..., 1]),
+                           np.minimum(pc[..., 2],
+                                      dims[2] - 1 - pc[..., 2])))
+            better = inside & (dist > best_dist)
+            pick = np.where(better, val, pick)
+            best_dist = np.where(better, dist, best_dist)
+            any_view |= inside
         else:
             sum_wv += wt * val
             sum_w += wt
 
     if fusion_type == FUSION_MAX:
         out = np.where(any_view, vmax, 0.0)
+    elif fusion_type in (FUSION_LOWEST_VIEWID, FUSION_HIGHEST_VIEWID,
+                         FUSION_CLOSEST_PIXEL):
+        out = np.where(any_view, pick, 0.0)
     else:
         with np.errstate(invalid="ignore", divide="ignore"):
             out = np.where(sum_w > 0, sum_wv / np.maximum(sum_w, 1e-300), 0.0)
@@ -175,9 +199,9 @@ def fuse_block(
     tmax = 255.0 if out_dtype == np.uint8 else 65535.0
     scaled = (out - min_intensity) / (max_intensity - min_intensity) * tmax
     # [PIN-CONV] round half away from zero, clamp
-    scaled = np.where(
-        sum_w > 0 if fusion_type != FUSION_MAX else any_view, scaled, 0.0
-    )
+    covered = sum_w > 0 if fusion_type in (FUSION_AVG, FUSION_AVG_BLEND) \
+        else any_view
+    scaled = np.where(covered, scaled, 0.0)
     q = np.floor(np.abs(scaled) + 0.5) * np.sign(scaled)
     return np.clip(q, 0, tmax).astype(out_dtype)
 

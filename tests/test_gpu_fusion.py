@@ -162,3 +162,18 @@ def test_fuse_volume_uint16_pyramid(ctx):
     ref1 = of.downsample_level(levels[0], (2, 2, 1))
     d1 = np.abs(levels[1].astype(np.int64) - ref1.astype(np.int64))
     assert d1.max() <= 1
+
+
+@pytest.mark.parametrize("ftype", [3, 4, 5])
+def test_fuse_winner_types_parity(ctx, ftype):
+    """LOWEST/HIGHEST_VIEWID_WINS and CLOSEST_PIXEL_WINS (§8(f) row 4)."""
+    views = fusion_views(21)
+    for i, v in enumerate(views):
+        v["affine"][0, 3] += 9.0 * i
+    gviews = upload_views(ctx, views, base=700)
+    bmin, bsize = (0, 0, 0), (40, 24, 20)
+    ref = of.fuse_block(views, bmin, bsize, ftype, out_dtype=np.float32)
+    got = ctx.fuse_blocks(gviews, [(bmin, bsize)], [[0, 1, 2]],
+                          fusion_type=ftype, out_dtype=np.float32)[0]
+    denom = np.maximum(np.abs(ref), 1.0)
+    assert np.max(np.abs(got - ref) / denom) < 1e-4

@@ -265,3 +265,21 @@ def test_pyramid_downsample_level():
     d = fusion.downsample_level(v, (2, 1, 1))
     assert d.shape == (1, 1, 3)
     assert np.allclose(d[0, 0], [0.5, 2.5, 4.0])
+
+
+def test_fuse_viewid_and_closest_wins():
+    a = np.full((8, 8, 8), 100, np.uint16)
+    bvol = np.full((8, 8, 8), 900, np.uint16)
+    vs = [dict(data=a, affine=IDENT), dict(data=bvol, affine=IDENT)]
+    low = fusion.fuse_block(vs, (0, 0, 0), (4, 4, 4),
+                            fusion.FUSION_LOWEST_VIEWID)
+    high = fusion.fuse_block(vs, (0, 0, 0), (4, 4, 4),
+                             fusion.FUSION_HIGHEST_VIEWID)
+    assert np.all(low == 100.0) and np.all(high == 900.0)
+    # closest-pixel: B shifted so A is closer to its border near x=0
+    affB = IDENT.copy()
+    affB[0, 3] = -4.0  # B-local x = world x + 4 -> B farther from border
+    vs2 = [dict(data=a, affine=IDENT), dict(data=bvol, affine=affB)]
+    cl = fusion.fuse_block(vs2, (0, 0, 0), (2, 8, 8),
+                           fusion.FUSION_CLOSEST_PIXEL)
+    assert np.all(cl[2:6, 2:6, :] == 900.0)  # B wins in A's border zone
