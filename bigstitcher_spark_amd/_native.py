@@ -108,6 +108,8 @@ def load_lib():
     lib.bs_view_synth.argtypes = [C.c_void_p, C.c_int32, C.c_int64 * 3,
                                   C.c_void_p, C.c_int32, C.c_uint32,
                                   C.c_uint16, C.c_uint16]
+    lib.bs_view_set_coefficients.argtypes = [C.c_void_p, C.c_int32,
+                                             C.c_void_p, C.c_int32 * 3]
     lib.bs_stitch_batch.argtypes = [C.c_void_p, C.POINTER(_Pair), C.c_size_t,
                                     C.POINTER(_StitchParams),
                                     C.POINTER(_ShiftResult)]
@@ -199,6 +201,22 @@ class Context:
             ),
             "bs_view_synth",
         )
+
+    def set_coefficients(self, view_id: int, ab):
+        """ab: (2, gz, gy, gx) float32 (a-plane, b-plane) or None."""
+        if ab is None:
+            self._check(
+                self._lib.bs_view_set_coefficients(
+                    self._h, view_id, None, (C.c_int32 * 3)(0, 0, 0)),
+                "bs_view_set_coefficients")
+            return
+        ab = np.ascontiguousarray(ab, np.float32)
+        _, gz, gy, gx = ab.shape
+        self._check(
+            self._lib.bs_view_set_coefficients(
+                self._h, view_id, ab.ctypes.data_as(C.c_void_p),
+                (C.c_int32 * 3)(gx, gy, gz)),
+            "bs_view_set_coefficients")
 
     # -- stitching --------------------------------------------------------
     def stitch_batch(self, pairs, ds=(2, 2, 1), peaks_to_check=5,

@@ -599,6 +599,8 @@ struct bs_dev_view {
   int nx, ny, nz;
   float inv[12];     /* world -> view-local, row-major 3x4 (fp32) */
   float border[3], range[3];
+  const float *coeff; /* nullptr = no intensity correction */
+  int cgx, cgy, cgz;
 };
 
 __device__ __forceinline__ float blend_w(float p, int dim, float border,
@@ -661,6 +663,39 @@ __global__ __launch_bounds__(256) void k_fuse(
       float c0 = c00 + (c10 - c00) * fy;
       float c1 = c01 + (c11 - c01) * fy;
       float val = c0 + (c1 - c0) * fz;
+      if (v.coeff) { /* [PIN-COEFF] trilinear over cell centers */
+        float tx2 = px * (float)v.cgx / (float)v.nx - 0.5f;
+        float ty2 = py * (float)v.cgy / (float)v.ny - 0.5f;
+        float tz2 = pz * (float)v.cgz / (float)v.nz - 0.5f;
+        tx2 = fminf(fmaxf(tx2, 0.0f), (float)(v.cgx - 1));
+        ty2 = fminf(fmaxf(ty2, 0.0f), (float)(v.cgy - 1));
+        tz2 = fminf(fmaxf(tz2, 0.0f), (float)(v.cgz - 1));
+        int cx0 = (int)tx2, cy0 = (int)ty2, cz0 = (int)tz2;
+        int cx1 = min(cx0 + 1, v.cgx - 1), cy1 = min(cy0 + 1, v.cgy - 1),
+            cz1 = min(cz0 + 1, v.cgz - 1);
+        float gx2 = tx2 - cx0, gy2 = ty2 - cy0, gz2 = tz2 - cz0;
+        long npl = (long)v.cgx * v.cgy * v.cgz;
+        float ab2[2];
+        for (int pl2 = 0; pl2 < 2; ++pl2) {
+          const float *g = v.coeff + pl2 * npl;
+          float d000 = g[((long)cz0 * v.cgy + cy0) * v.cgx + cx0];
+          float d100 = g[((long)cz0 * v.cgy + cy0) * v.cgx + cx1];
+          float d010 = g[((long)cz0 * v.cgy + cy1) * v.cgx + cx0];
+          float d110 = g[((long)cz0 * v.cgy + cy1) * v.cgx + cx1];
+          float d001 = g[((long)cz1 * v.cgy + cy0) * v.cgx + cx0];
+          float d101 = g[((long)cz1 * v.cgy + cy0) * v.cgx + cx1];
+          float d011 = g[((long)cz1 * v.cgy + cy1) * v.cgx + cx0];
+          float d111 = g[((long)cz1 * v.cgy + cy1) * v.cgx + cx1];
+          float e00 = d000 + (d100 - d000) * gx2;
+          float e10 = d010 + (d110 - d010) * gx2;
+          float e01 = d001 + (d101 - d001) * gx2;
+          float e11 = d011 + (d111 - d011) * gx2;
+          float e0 = e00 + (e10 - e00) * gy2;
+          float e1 = e01 + (e11 - e01) * gy2;
+          ab2[pl2] = e0 + (e1 - e0) * gz2;
+        }
+        val = ab2[0] * val + ab2[1];
+      }
       float w = 1.0f;
       if (ftype == BS_FUSION_AVG_BLEND)
         w = blend_w(px, v.nx, v.border[0], v.range[0]) *
@@ -800,6 +835,8 @@ __global__ __launch_bounds__(256) void k_synth_quant(
 struct bs_view_rec {
   unsigned short *dptr;
   long dims[3]; /* x,y,z */
+  float *coeff = nullptr; /* 2*cg[0]*cg[1]*cg[2], a-plane then b-plane */
+  int cg[3] = {0, 0, 0};
 };
 
 struct bs_ev {
@@ -968,7 +1005,10 @@ extern "C" void bs_ctx_destroy(bs_ctx *c) {
   if (!c) return;
   (void)hipSetDevice(c->dev);
   (void)hipStreamSynchronize(c->stream);
-  for (auto &kv : c->views) (void)hipFree(kv.second.dptr);
+  for (auto &kv : c->views) {
+    (void)hipFree(kv.second.dptr);
+    if (kv.second.coeff) (void)hipFree(kv.second.coeff);
+  }
   for (auto &kv : c->twiddles) (void)hipFree(kv.second);
   for (auto &pr : c->evpool) {
     (void)hipEventDestroy(pr.first);
@@ -1066,6 +1106,7 @@ extern "C" int bs_view_upload(bs_ctx *c, int32_t id, const uint16_t *data,
   auto it = c->views.find(id);
   if (it != c->views.end()) {
     (void)hipFree(it->second.dptr);
+    if (it->second.coeff) (void)hipFree(it->second.coeff);
     c->views.erase(it);
   }
   unsigned short *d;
@@ -1081,7 +1122,32 @@ extern "C" int bs_view_release(bs_ctx *c, int32_t id) {
   auto it = c->views.find(id);
   if (it == c->views.end()) return BS_ENOVIEW;
   (void)hipFree(it->second.dptr);
+  if (it->second.coeff) (void)hipFree(it->second.coeff);
   c->views.erase(it);
+  return BS_OK;
+}
+
+extern "C" int bs_view_set_coefficients(bs_ctx *c, int32_t id,
+                                        const float *ab,
+                                        const int32_t grid_dims[3]) {
+  if (!c) return BS_EINVAL;
+  std::lock_guard<std::mutex> g(c->mu);
+  CHK(c, hipSetDevice(c->dev));
+  auto it = c->views.find(id);
+  if (it == c->views.end()) return BS_ENOVIEW;
+  if (it->second.coeff) {
+    (void)hipFree(it->second.coeff);
+    it->second.coeff = nullptr;
+    it->second.cg[0] = it->second.cg[1] = it->second.cg[2] = 0;
+  }
+  if (!ab) return BS_OK;
+  if (!grid_dims || grid_dims[0] < 1 || grid_dims[1] < 1 ||
+      grid_dims[2] < 1)
+    return BS_EINVAL;
+  size_t n = 2ull * grid_dims[0] * grid_dims[1] * grid_dims[2];
+  CHK(c, hipMalloc(&it->second.coeff, n * 4));
+  CHK(c, hipMemcpy(it->second.coeff, ab, n * 4, hipMemcpyHostToDevice));
+  for (int d = 0; d < 3; ++d) it->second.cg[d] = grid_dims[d];
   return BS_OK;
 }
 
@@ -1115,6 +1181,7 @@ extern "C" int bs_view_synth(bs_ctx *c, int32_t id, const int64_t dims[3],
   auto it = c->views.find(id);
   if (it != c->views.end()) {
     (void)hipFree(it->second.dptr);
+    if (it->second.coeff) (void)hipFree(it->second.coeff);
     c->views.erase(it);
   }
   unsigned short *d;
@@ -1568,6 +1635,10 @@ static int build_dev_views(bs_ctx *c, const bs_fuse_view *views,
     v.nx = (int)it->second.dims[0];
     v.ny = (int)it->second.dims[1];
     v.nz = (int)it->second.dims[2];
+    v.coeff = it->second.coeff;
+    v.cgx = it->second.cg[0];
+    v.cgy = it->second.cg[1];
+    v.cgz = it->second.cg[2];
     double adj[12];
     for (int k = 0; k < 12; ++k) adj[k] = views[i].affine[k];
     if (vol_min) {

@@ -82,6 +82,17 @@ int bs_view_synth(bs_ctx *ctx, int32_t view_id, const int64_t dims[3],
 /* Download an uploaded/synth view back to host (tests). */
 int bs_view_download(bs_ctx *ctx, int32_t view_id, uint16_t *out);
 
+/* Optional per-view linear intensity coefficients for fusion
+ * (reference SparkAffineFusion.java:545-559: Coefficients read per view
+ * and applied inside BlkAffineFusion.initWithIntensityCoefficients).
+ * ab: 2 * gx*gy*gz floats — the multiplicative plane (a), then the
+ * additive plane (b), each a coarse grid covering the view uniformly;
+ * the fusion kernel samples the grid trilinearly at cell centers
+ * ([PIN-COEFF]) and applies value' = a*value + b before weighting.
+ * Passing NULL clears the view's coefficients. */
+int bs_view_set_coefficients(bs_ctx *ctx, int32_t view_id, const float *ab,
+                             const int32_t grid_dims[3]);
+
 /* --------------------------------------------------------------- stitching */
 
 /* One tile-pair work unit. The overlap interval inside each view is computed

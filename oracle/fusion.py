@@ -155,6 +155,19 @@ def fuse_block(
             continue
         pc = np.where(inside[..., None], p, 0.0)
         val = _trilinear(vol, pc)
+        if v.get("coeff") is not None:
+            # [PIN-COEFF] linear intensity correction a*val + b, with
+            # (a, b) trilinearly sampled from a coarse grid covering the
+            # view uniformly (cell centers); restates the Coefficients
+            # application at reference SparkAffineFusion.java:545-559.
+            ab = np.asarray(v["coeff"], np.float64)  # (2, gz, gy, gx)
+            g = np.array([ab.shape[3], ab.shape[2], ab.shape[1]], float)
+            t = pc * (g / dims) - 0.5
+            t = np.clip(t, 0.0, g - 1.0)
+            planes = []
+            for pl in range(2):
+                planes.append(_trilinear(ab[pl], t))
+            val = planes[0] * val + planes[1]
         if fusion_type == FUSION_AVG_BLEND:
             border = np.asarray(v.get("border", (0.0, 0.0, 0.0)))
             rng = np.asarray(v.get("range", (40.0, 40.0, 40.0)))

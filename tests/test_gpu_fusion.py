@@ -177,3 +177,39 @@ def test_fuse_winner_types_parity(ctx, ftype):
                           fusion_type=ftype, out_dtype=np.float32)[0]
     denom = np.maximum(np.abs(ref), 1.0)
     assert np.max(np.abs(got - ref) / denom) < 1e-4
+
+
+def test_fuse_intensity_coefficients_parity(ctx):
+    """Per-view linear intensity coefficients (§8(f) row 4; reference
+    SparkAffineFusion.java:545-559): coarse-grid (a,b) applied before
+    blending, GPU vs oracle."""
+    rng = np.random.default_rng(31)
+    views = fusion_views(23)
+    coeffs = []
+    for v in views:
+        ab = rng.uniform(0.5, 2.0, size=(2, 2, 3, 4))
+        ab[1] = rng.uniform(-200, 200, size=(2, 3, 4))[None][0]
+        v["coeff"] = ab
+        coeffs.append(ab)
+    gviews = upload_views(ctx, views, base=800)
+    for gv, ab in zip(gviews, coeffs):
+        ctx.set_coefficients(gv["view_id"], ab)
+    bmin, bsize = (1, 2, 0), (20, 16, 12)
+    ref = of.fuse_block(views, bmin, bsize, of.FUSION_AVG_BLEND,
+                        out_dtype=np.float32)
+    got = ctx.fuse_blocks(gviews, [(bmin, bsize)], [[0, 1, 2]],
+                          fusion_type=of.FUSION_AVG_BLEND,
+                          out_dtype=np.float32)[0]
+    denom = np.maximum(np.abs(ref), 1.0)
+    assert np.max(np.abs(got - ref) / denom) < 1e-4
+    # clearing coefficients restores the uncorrected result
+    for gv in gviews:
+        ctx.set_coefficients(gv["view_id"], None)
+    for v in views:
+        v.pop("coeff")
+    ref2 = of.fuse_block(views, bmin, bsize, of.FUSION_AVG_BLEND,
+                         out_dtype=np.float32)
+    got2 = ctx.fuse_blocks(gviews, [(bmin, bsize)], [[0, 1, 2]],
+                           fusion_type=of.FUSION_AVG_BLEND,
+                           out_dtype=np.float32)[0]
+    assert np.max(np.abs(got2 - ref2) / np.maximum(np.abs(ref2), 1.0)) < 1e-4

@@ -283,3 +283,16 @@ def test_fuse_viewid_and_closest_wins():
     cl = fusion.fuse_block(vs2, (0, 0, 0), (2, 8, 8),
                            fusion.FUSION_CLOSEST_PIXEL)
     assert np.all(cl[2:6, 2:6, :] == 900.0)  # B wins in A's border zone
+
+
+def test_fuse_intensity_coefficients():
+    """[PIN-COEFF] constant grid -> exact linear transform; 1x1x1 grid."""
+    vol = np.full((8, 8, 8), 100, np.uint16)
+    ab = np.zeros((2, 1, 1, 1))
+    ab[0] = 2.0
+    ab[1] = 30.0
+    out = fusion.fuse_block(
+        [dict(data=vol, affine=IDENT, coeff=ab)], (0, 0, 0), (4, 4, 4),
+        fusion.FUSION_AVG,
+    )
+    assert np.all(out == 230.0)
