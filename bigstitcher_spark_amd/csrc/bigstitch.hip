@@ -430,7 +430,21 @@ __global__ __launch_bounds__(256) void k_peak_tile(
         int lx = i % HX, t = i / HX, ly = t % HY, lz = t / HY;
         tile[i] = base[((long)lz * py + ly) * px + lx];
       }
-    } else {
+    } else if (px >= HX && py >= HY && pz >= HZ) {
+      /* boundary tile, but every halo coordinate wraps at most one
+       * period: branchless select instead of 3 runtime idivs/element */
+      for (int i = tid; i < HX * HY * HZ; i += 256) {
+        int lx = i % HX, t = i / HX, ly = t % HY, lz = t / HY;
+        int gx = x0 + lx - 1, gy = y0 + ly - 1, gz = z0 + lz - 1;
+        gx += gx < 0 ? px : 0;
+        gx -= gx >= px ? px : 0;
+        gy += gy < 0 ? py : 0;
+        gy -= gy >= py ? py : 0;
+        gz += gz < 0 ? pz : 0;
+        gz -= gz >= pz ? pz : 0;
+        tile[i] = pcm[((long)gz * py + gy) * px + gx];
+      }
+    } else { /* tiny volumes: full modulo wrap */
       for (int i = tid; i < HX * HY * HZ; i += 256) {
         int lx = i % HX, t = i / HX, ly = t % HY, lz = t / HY;
         int gx = (x0 + lx - 1 + px) % px;
