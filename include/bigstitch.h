@@ -207,6 +207,12 @@ int bs_fuse_volume(bs_ctx *ctx, const bs_fuse_view *views, size_t nviews,
                    const int32_t *abs_downsampling, int64_t *level_dims_out,
                    void **level_buffers);
 
+/* Debug-only: download the PCM volume of the LAST pair processed by
+ * bs_stitch_batch on this ctx (out must hold prod(out_dims) floats after
+ * a first call with out=NULL is not supported — query dims via the pair
+ * geometry). Not part of the drop-in surface; used by parity tooling. */
+int bs_debug_pcm(bs_ctx *ctx, float *out, int64_t out_dims[3]);
+
 /* ------------------------------------------------------- instrumentation */
 
 /* Per-kernel timing, HIP-event measured on the launch stream (bench.py

@@ -193,3 +193,33 @@ def test_stitch_1024_axis(ctx):
     assert got["valid"] and ref["valid"]
     assert np.all(np.abs(got["shift"] - ref["shift"]) < 1e-3)
     assert got["r"] == pytest.approx(ref["r"], abs=1e-9)
+
+
+def test_stitch_repeat_determinism(ctx):
+    """Repeated batches over the same resident views return bit-identical
+    results (fixed view-order accumulation, deterministic peak
+    tie-breaks — SURVEY.md §5 'deterministic-order accumulate')."""
+    pairs = []
+    for i, shift in enumerate([(5.0, -2.0, 1.5), (-3.25, 4.0, 0.0),
+                               (0.5, 0.5, 0.5), (7.75, -6.5, 3.25)]):
+        a, b = synth.make_pair((96, 96, 96), shift, seed=200 + i)
+        ctx.upload(70 + 2 * i, a)
+        ctx.upload(71 + 2 * i, b)
+        pairs.append(dict(view_a=70 + 2 * i, view_b=71 + 2 * i,
+                          off_a=(0, 0, 0), size_a=(96, 96, 96),
+                          off_b=(0, 0, 0), size_b=(96, 96, 96)))
+    runs = [ctx.stitch_batch(pairs, ds=(1, 1, 1)) for _ in range(3)]
+    for rep in runs[1:]:
+        for r0, r1 in zip(runs[0], rep):
+            assert r0["valid"] == r1["valid"]
+            assert np.array_equal(r0["shift"], r1["shift"])
+            assert r0["r"] == r1["r"]
+
+
+def test_stitch_missing_view_errors(ctx):
+    import pytest as _pytest
+
+    pair = dict(view_a=9991, view_b=9992, off_a=(0, 0, 0),
+                size_a=(16, 16, 16), off_b=(0, 0, 0), size_b=(16, 16, 16))
+    with _pytest.raises(RuntimeError, match="view not uploaded"):
+        ctx.stitch_batch([pair], ds=(1, 1, 1))
