@@ -900,13 +900,15 @@ struct bs_slot {
   int stage = 0; /* 0 idle, 1 fft+peak issued, 2 rtest issued */
 };
 
+#define BS_NSLOTS 3
+
 struct bs_ctx {
   int dev;
   hipStream_t stream; /* default stream: views, synth, fusion */
   std::string err;
   std::map<int32_t, bs_view_rec> views;
   std::map<int, f2 *> twiddles;
-  bs_slot slot[2];
+  bs_slot slot[BS_NSLOTS];
   float *synth_acc = nullptr;
   size_t synth_cap = 0;
   void *fuse_out = nullptr;
@@ -989,7 +991,7 @@ extern "C" int bs_ctx_create(bs_ctx **out, int device_id) {
                             64 * 1024);
   /* per-slot streams, events and small fixed buffers */
   bool ok = true;
-  for (int s = 0; s < 2; ++s) {
+  for (int s = 0; s < BS_NSLOTS; ++s) {
     bs_slot &sl = c->slot[s];
     ok = ok && hipStreamCreate(&sl.stream) == hipSuccess &&
          hipEventCreateWithFlags(&sl.peaks_ready, hipEventDisableTiming) ==
@@ -1028,7 +1030,7 @@ extern "C" void bs_ctx_destroy(bs_ctx *c) {
     (void)hipEventDestroy(pr.first);
     (void)hipEventDestroy(pr.second);
   }
-  for (int s = 0; s < 2; ++s) {
+  for (int s = 0; s < BS_NSLOTS; ++s) {
     bs_slot &sl = c->slot[s];
     if (sl.stream) (void)hipStreamSynchronize(sl.stream);
     (void)hipFree(sl.spec);
@@ -1566,18 +1568,19 @@ extern "C" int bs_stitch_batch(bs_ctx *c, const bs_pair_desc *pairs, size_t np,
   auto t0 = std::chrono::steady_clock::now();
   int rc = BS_OK;
   for (size_t i = 0; i < np && rc == BS_OK; ++i) {
-    bs_slot *sl = &c->slot[i & 1];
+    bs_slot *sl = &c->slot[i % BS_NSLOTS];
     if (sl->stage == 1) rc = stitch_phaseB(c, sl, prm);
     if (rc == BS_OK && sl->stage == 2) rc = stitch_phaseC(c, sl, prm, out);
     if (rc == BS_OK) rc = stitch_phaseA(c, sl, pairs[i], prm, i);
   }
-  /* drain both slots, older first */
-  for (size_t k = np >= 2 ? np - 2 : 0; k < np && rc == BS_OK; ++k) {
-    bs_slot *sl = &c->slot[k & 1];
+  /* drain the open slots, oldest first */
+  for (size_t k = np >= BS_NSLOTS ? np - BS_NSLOTS : 0;
+       k < np && rc == BS_OK; ++k) {
+    bs_slot *sl = &c->slot[k % BS_NSLOTS];
     if (sl->stage == 1) rc = stitch_phaseB(c, sl, prm);
     if (rc == BS_OK && sl->stage == 2) rc = stitch_phaseC(c, sl, prm, out);
   }
-  for (int s = 0; s < 2 && rc == BS_OK; ++s)
+  for (int s = 0; s < BS_NSLOTS && rc == BS_OK; ++s)
     CHK(c, hipStreamSynchronize(c->slot[s].stream));
   c->stats.batch_ms =
       std::chrono::duration<double, std::milli>(
