@@ -288,3 +288,68 @@ def test_cli_resave_end_to_end(tmp_path):
     assert r.returncode == 0, r.stderr + r.stdout
     prs = ET.parse(xo).getroot().findall(".//StitchingResults/PairwiseResult")
     assert len(prs) == 1
+
+
+@pytest.mark.gpu
+def test_cli_grid6_stitch_solve_fuse(tmp_path):
+    """3x2 grid of 6 tiles (7 overlap links), the configs[2]-shaped
+    scenario: stitch all pairs, solve globally, fuse; every solved
+    position lands on its injected ground truth."""
+    from tests.test_cli_solver import model_translations
+
+    size, ov = 48, 16
+    step = size - ov
+    rng = np.random.default_rng(99)
+    setups, tiles, true_pos = [], {}, {}
+    sid = 0
+    for gy in range(2):
+        for gx in range(3):
+            nominal = np.array([gx * step, gy * step, 0.0])
+            errv = np.zeros(3) if sid == 0 else rng.uniform(-2.5, 2.5, 3)
+            pos = nominal + errv
+            true_pos[sid] = pos
+            # scene shared across the grid: blobs in world coords,
+            # shifted into each tile's local frame (tile content at
+            # world position `pos`): feature world w -> local w - pos
+            setups.append(dict(id=sid, dims=(size, size, size),
+                               pos=tuple(nominal)))
+            sid += 1
+    world_rng = np.random.default_rng(5)
+    scene = synth.make_scene((size, size + step, size + 2 * step),
+                             world_rng, margin=10.0)
+    for s in range(6):
+        local = scene.copy()
+        local[:, 0] -= np.float32(true_pos[s][0])
+        local[:, 1] -= np.float32(true_pos[s][1])
+        local[:, 2] -= np.float32(true_pos[s][2])
+        tiles[s] = synth.render_tile((size, size, size), local,
+                                     noise_seed=1000 + s)
+    n5 = os.path.join(str(tmp_path), "input.n5")
+    xml = os.path.join(str(tmp_path), "dataset.xml")
+    for s in range(6):
+        n5util.write_dataset(n5, f"setup{s}/timepoint0/s0", tiles[s],
+                             (32, 32, 32))
+    n5util.make_dataset_xml(xml, "input.n5", setups)
+    r = run([os.path.join(BIN, "stitching"), "-x", xml, "-ds", "1,1,1",
+             "--minOverlapRatio", "0.05"])
+    assert r.returncode == 0, r.stderr + r.stdout
+    prs = ET.parse(xml).getroot().findall(
+        ".//StitchingResults/PairwiseResult")
+    assert len(prs) >= 7  # 3x2 grid: 7 edge links minimum
+    r = run([os.path.join(BIN, "solver"), "-x", xml])
+    assert r.returncode == 0, r.stderr + r.stdout
+    t = model_translations(xml)
+    for s in range(6):
+        err_px = np.abs(t[s] - true_pos[s])
+        assert np.all(err_px < 0.8), (s, t[s], true_pos[s])
+    out = os.path.join(str(tmp_path), "fused.n5")
+    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml, "-o",
+             out, "--blockSize", "32,32,32", "--dataType", "UINT16",
+             "--minIntensity", "0", "--maxIntensity", "65535"])
+    assert r.returncode == 0, r.stderr
+    r = run([os.path.join(BIN, "affine-fusion"), "-o", out,
+             "--fusionType", "AVG_BLEND", "--blendingRange", "8"])
+    assert r.returncode == 0, r.stderr + r.stdout
+    fused, _ = n5util.read_dataset(out, "ch0tp0/s0")
+    assert fused.mean() > 100  # populated everywhere (no black seams)
+    assert (fused == 0).mean() < 0.02
