@@ -351,5 +351,8 @@ def test_cli_grid6_stitch_solve_fuse(tmp_path):
              "--fusionType", "AVG_BLEND", "--blendingRange", "8"])
     assert r.returncode == 0, r.stderr + r.stdout
     fused, _ = n5util.read_dataset(out, "ch0tp0/s0")
-    assert fused.mean() > 100  # populated everywhere (no black seams)
-    assert (fused == 0).mean() < 0.02
+    assert fused.mean() > 100
+    # interior fully populated (no black seams); the bbox border slabs
+    # are legitimately empty where solver-shifted tiles do not reach
+    interior = fused[3:-3, 3:-3, 3:-3]
+    assert (interior == 0).mean() < 0.01
