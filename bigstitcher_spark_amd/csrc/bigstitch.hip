@@ -553,80 +553,56 @@ __device__ __forceinline__ u64 wave_sum_u64(u64 v) {
   return v;
 }
 
-/* Candidate cross-correlation sums [PIN-R]: stage a chunk of A rows in
- * LDS ONCE and let every candidate read it from there — the candidates'
- * overlap regions cover nearly the same A voxels, so the naive
- * one-kernel-per-candidate structure re-reads A from HBM k times
- * (~2 GB/pair measured at 512^3); this structure reads A once and B per
- * candidate. Exact u64 arithmetic, order-independent (deterministic and
- * bit-identical to the oracle's int64 sums). */
-#define RT_MAXC 64
-
 __global__ __launch_bounds__(256) void k_rtest(
-    bs_region a, bs_region b, const bs_cand *cands, int nc, int rows_per,
-    u64 *sums /* [nc][5] */) {
-  extern __shared__ __attribute__((aligned(16))) char smem[];
-  unsigned short *arow = (unsigned short *)smem; /* [rows_per][a.mx] */
+    bs_region a, bs_region b, const bs_cand *cands, u64 *sums /* [nc][5] */) {
   __shared__ u64 ws[4][5];
-  const int tid = threadIdx.x;
-  const int mx = a.mx;
-  const long nrows = (long)a.my * a.mz;
-  const long nchunks = (nrows + rows_per - 1) / rows_per;
-  for (long ch = blockIdx.x; ch < nchunks; ch += gridDim.x) {
-    const long r0 = ch * rows_per;
-    const int nr =
-        (int)((nrows - r0) < (long)rows_per ? (nrows - r0) : rows_per);
-    /* stage A rows r0..r0+nr (region-strided source, coalesced in x) */
-    for (int i = tid; i < nr * mx; i += 256) {
-      int rl = i / mx, x = i - rl * mx;
-      long r = r0 + rl;
-      int y = (int)(r % a.my), z = (int)(r / a.my);
-      arow[(size_t)rl * mx + x] =
-          a.ptr[(a.oz + z) * a.sxy + (a.oy + y) * a.sx + a.ox + x];
+  const bs_cand c = cands[blockIdx.y];
+  long nrows = (long)c.ny * c.nz;
+  u64 pa = 0, pb = 0, paa = 0, pbb = 0, pab = 0;
+  auto rowptr_a = [&](long row) {
+    int y = (int)(row % c.ny), z = (int)(row / c.ny);
+    return a.ptr + (a.oz + c.loz + z) * a.sxy + (a.oy + c.loy + y) * a.sx +
+           a.ox + c.lox;
+  };
+  auto rowptr_b = [&](long row) {
+    int y = (int)(row % c.ny), z = (int)(row / c.ny);
+    return b.ptr + (b.oz + c.loz + c.sz + z) * b.sxy +
+           (b.oy + c.loy + c.sy + y) * b.sx + b.ox + c.lox + c.sx;
+  };
+  long row = blockIdx.x;
+  /* two rows in flight doubles outstanding loads (latency-bound kernel) */
+  for (; row + gridDim.x < nrows; row += 2L * gridDim.x) {
+    const unsigned short *a0 = rowptr_a(row), *b0 = rowptr_b(row);
+    const unsigned short *a1 = rowptr_a(row + gridDim.x);
+    const unsigned short *b1 = rowptr_b(row + gridDim.x);
+    for (int x = threadIdx.x; x < c.nx; x += 256) {
+      u64 av0 = a0[x], bv0 = b0[x], av1 = a1[x], bv1 = b1[x];
+      pa += av0 + av1;
+      pb += bv0 + bv1;
+      paa += av0 * av0 + av1 * av1;
+      pbb += bv0 * bv0 + bv1 * bv1;
+      pab += av0 * bv0 + av1 * bv1;
     }
-    __syncthreads();
-    for (int c = 0; c < nc; ++c) {
-      const bs_cand cd = cands[c];
-      u64 pa = 0, pb = 0, paa = 0, pbb = 0, pab = 0;
-      for (int rl = 0; rl < nr; ++rl) {
-        long r = r0 + rl;
-        int y = (int)(r % a.my), z = (int)(r / a.my);
-        if (y < cd.loy || y >= cd.loy + cd.ny || z < cd.loz ||
-            z >= cd.loz + cd.nz)
-          continue;
-        const unsigned short *al = arow + (size_t)rl * mx;
-        const unsigned short *br = b.ptr + (b.oz + z + cd.sz) * b.sxy +
-                                   (b.oy + y + cd.sy) * b.sx + b.ox + cd.sx;
-        for (int x = cd.lox + tid; x < cd.lox + cd.nx; x += 256) {
-          u64 av = al[x], bv = br[x];
-          pa += av;
-          pb += bv;
-          paa += av * av;
-          pbb += bv * bv;
-          pab += av * bv;
-        }
-      }
-      pa = wave_sum_u64(pa);
-      pb = wave_sum_u64(pb);
-      paa = wave_sum_u64(paa);
-      pbb = wave_sum_u64(pbb);
-      pab = wave_sum_u64(pab);
-      int lane = tid & 63, wave = tid >> 6;
-      if (lane == 0) {
-        ws[wave][0] = pa;
-        ws[wave][1] = pb;
-        ws[wave][2] = paa;
-        ws[wave][3] = pbb;
-        ws[wave][4] = pab;
-      }
-      __syncthreads();
-      if (tid < 5) {
-        u64 t = ws[0][tid] + ws[1][tid] + ws[2][tid] + ws[3][tid];
-        if (t) atomicAdd(&sums[(long)c * 5 + tid], t);
-      }
-      __syncthreads();
+  }
+  for (; row < nrows; row += gridDim.x) {
+    const unsigned short *a0 = rowptr_a(row), *b0 = rowptr_b(row);
+    for (int x = threadIdx.x; x < c.nx; x += 256) {
+      u64 av = a0[x], bv = b0[x];
+      pa += av; pb += bv; paa += av * av; pbb += bv * bv; pab += av * bv;
     }
-    __syncthreads(); /* arow reused next chunk */
+  }
+  pa = wave_sum_u64(pa); pb = wave_sum_u64(pb); paa = wave_sum_u64(paa);
+  pbb = wave_sum_u64(pbb); pab = wave_sum_u64(pab);
+  int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  if (lane == 0) {
+    ws[wave][0] = pa; ws[wave][1] = pb; ws[wave][2] = paa;
+    ws[wave][3] = pbb; ws[wave][4] = pab;
+  }
+  __syncthreads();
+  if (threadIdx.x < 5) {
+    u64 s = ws[0][threadIdx.x] + ws[1][threadIdx.x] + ws[2][threadIdx.x] +
+            ws[3][threadIdx.x];
+    atomicAdd(&sums[(long)blockIdx.y * 5 + threadIdx.x], s);
   }
 }
 
@@ -1013,9 +989,6 @@ extern "C" int bs_ctx_create(bs_ctx **out, int device_id) {
   (void)hipFuncSetAttribute((const void *)k_fft_x_inv,
                             hipFuncAttributeMaxDynamicSharedMemorySize,
                             64 * 1024);
-  (void)hipFuncSetAttribute((const void *)k_rtest,
-                            hipFuncAttributeMaxDynamicSharedMemorySize,
-                            100 * 1024);
   /* per-slot streams, events and small fixed buffers */
   bool ok = true;
   for (int s = 0; s < BS_NSLOTS; ++s) {
@@ -1504,18 +1477,15 @@ static int stitch_phaseB(bs_ctx *c, bs_slot *sl,
                           sl->stream));
     CHK(c, hipStreamSynchronize(sl->stream)); /* gc dies at scope end */
     {
-      /* A rows staged per chunk; size the chunk for ~72 KB LDS */
-      int rows_per = std::max(
-          8, (int)std::min(
-                 96L, (72L * 1024) /
-                          std::max(1L, (long)sl->reg[0].mx * 2)));
-      long nrows = (long)sl->reg[0].my * sl->reg[0].mz;
-      long nchunks = (nrows + rows_per - 1) / rows_per;
-      size_t lds = (size_t)rows_per * sl->reg[0].mx * 2;
+      long maxrows = 1;
+      for (auto &h : sl->hc)
+        maxrows = std::max(maxrows, (long)h.gc.ny * h.gc.nz);
       bs_tim tt(c, BS_K_CORR, sl->stream);
-      hipLaunchKernelGGL(k_rtest, dim3((unsigned)std::min(2048L, nchunks)),
-                         dim3(256), lds, sl->stream, sl->reg[0], sl->reg[1],
-                         sl->dcands, (int)gc.size(), rows_per, sl->dsums);
+      hipLaunchKernelGGL(k_rtest,
+                         dim3((unsigned)std::min(2048L, maxrows),
+                              (unsigned)gc.size()),
+                         dim3(256), 0, sl->stream, sl->reg[0], sl->reg[1],
+                         sl->dcands, sl->dsums);
     }
     CHK(c, hipMemcpyAsync(sl->hsums, sl->dsums,
                           sl->hc.size() * 5 * sizeof(u64),
