@@ -17,7 +17,13 @@ import os
 import sqlite3
 import sys
 
-CORRECT2X = {"k_fft_x_fwd", "k_fft_pass", "k_fft_x_inv"}
+# Per-kernel FETCH calibration against known algorithmic byte counts
+# (MI355X_MICROARCH.md §HBM says to calibrate per access pattern):
+# k_fft_x_fwd (u32/lane) and k_fft_x_inv (8 B/lane) measure exactly half
+# the algorithmic read -> x2; k_fft_pass (float4 = 16 B/lane) measures
+# the full bytes (raw launch-mix FETCH 0.629 GB == the 0.629 GB
+# algorithmic mix) -> x1.
+CORRECT2X = {"k_fft_x_fwd", "k_fft_x_inv"}
 # map mangled display prefix -> bench kernel-stat names (fft_pass splits
 # by launch context are not distinguishable in PMC output; report under
 # one name and let bench match per-kernel names it knows)
