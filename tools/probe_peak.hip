@@ -1,0 +1,188 @@
+/* Isolated within-probe A/B of the PCM peak scan (see DESIGN.md §4 —
+ * the kernel whose measured time never matched the byte/instruction
+ * model). Variants co-run interleaved (§5.4 rules 13/24).
+ * Build: hipcc --offload-arch=gfx950 -O3 -std=c++17 tools/probe_peak.hip
+ *        -o tools/probe_peak */
+#include <hip/hip_runtime.h>
+#include <cmath>
+#include <cstdio>
+#include <cstdlib>
+#include <vector>
+
+struct bs_peak { float v; int pad; long long idx; };
+
+__device__ __forceinline__ bool pk_better(float v, long long i, float v2,
+                                          long long i2) {
+  return (v > v2) || (v == v2 && i < i2);
+}
+__device__ void pk_insert(float (&tv)[5], long long (&ti)[5], float v,
+                          long long i) {
+  if (!pk_better(v, i, tv[4], ti[4])) return;
+  tv[4] = v; ti[4] = i;
+  for (int k = 4; k > 0 && pk_better(tv[k], ti[k], tv[k - 1], ti[k - 1]);
+       --k) {
+    float fv = tv[k]; tv[k] = tv[k - 1]; tv[k - 1] = fv;
+    long long fi = ti[k]; ti[k] = ti[k - 1]; ti[k - 1] = fi;
+  }
+}
+__device__ void pk_merge_shfl(float (&tv)[5], long long (&ti)[5]) {
+  for (int off = 32; off >= 1; off >>= 1) {
+    float ov[5]; long long oi[5];
+    for (int k = 0; k < 5; ++k) {
+      ov[k] = __shfl_down(tv[k], off);
+      oi[k] = __shfl_down(ti[k], off);
+    }
+    for (int k = 0; k < 5; ++k) pk_insert(tv, ti, ov[k], oi[k]);
+  }
+}
+
+#define PK_TX 128
+#define PK_TY 8
+#define PK_TZ 8
+
+/* VAR 0: production (LDS halo tile, branchless 26-max)
+ * VAR 1: load-only ablation (no maxima math)
+ * VAR 2: maxima-only ablation (no halo load; garbage LDS)
+ * VAR 3: no-LDS, global-direct 26 reads (L1/L2 reliance) */
+template <int VAR>
+__global__ __launch_bounds__(256) void k_peak(const float *pcm, int px,
+                                              int py, int pz,
+                                              bs_peak *wgbuf) {
+  __shared__ float tile[(PK_TZ + 2) * (PK_TY + 2) * (PK_TX + 2)];
+  __shared__ float wv[4][5];
+  __shared__ long long wi[4][5];
+  const int tid = threadIdx.x;
+  const int ntx = (px + PK_TX - 1) / PK_TX;
+  const int nty = (py + PK_TY - 1) / PK_TY;
+  const int ntz = (pz + PK_TZ - 1) / PK_TZ;
+  const int HX = PK_TX + 2, HY = PK_TY + 2, HZ = PK_TZ + 2;
+  const long ntiles = (long)ntx * nty * ntz;
+  float tv[5];
+  long long ti[5];
+  for (int k = 0; k < 5; ++k) { tv[k] = -3.0e38f; ti[k] = 0x7fffffffffffffffLL; }
+  for (long t0 = blockIdx.x; t0 < ntiles; t0 += gridDim.x) {
+    const int bx = (int)(t0 % ntx);
+    const int by = (int)((t0 / ntx) % nty);
+    const int bz = (int)(t0 / ((long)ntx * nty));
+    const int x0 = bx * PK_TX, y0 = by * PK_TY, z0 = bz * PK_TZ;
+    const bool interior = x0 > 0 && y0 > 0 && z0 > 0 && x0 + PK_TX < px &&
+                          y0 + PK_TY < py && z0 + PK_TZ < pz;
+    if (VAR != 2) {
+      if (interior) {
+        const float *base =
+            pcm + ((long)(z0 - 1) * py + (y0 - 1)) * px + (x0 - 1);
+        for (int i = tid; i < HX * HY * HZ; i += 256) {
+          int lx = i % HX, t = i / HX, ly = t % HY, lz = t / HY;
+          tile[i] = base[((long)lz * py + ly) * px + lx];
+        }
+      } else {
+        for (int i = tid; i < HX * HY * HZ; i += 256) {
+          int lx = i % HX, t = i / HX, ly = t % HY, lz = t / HY;
+          int gx = x0 + lx - 1, gy = y0 + ly - 1, gz = z0 + lz - 1;
+          gx += gx < 0 ? px : 0;  gx -= gx >= px ? px : 0;
+          gy += gy < 0 ? py : 0;  gy -= gy >= py ? py : 0;
+          gz += gz < 0 ? pz : 0;  gz -= gz >= pz ? pz : 0;
+          tile[i] = pcm[((long)gz * py + gy) * px + gx];
+        }
+      }
+    }
+    __syncthreads();
+    if (VAR != 1) {
+      for (int i = tid; i < PK_TX * PK_TY * PK_TZ; i += 256) {
+        int lx = i % PK_TX, t = i / PK_TX, ly = t % PK_TY, lz = t / PK_TY;
+        int gx = x0 + lx, gy = y0 + ly, gz = z0 + lz;
+        bool inb = gx < px && gy < py && gz < pz;
+        float v, m;
+        if (VAR == 3) {
+          if (!inb) continue;
+          const float *c0 =
+              pcm + ((long)gz * py + gy) * px + gx;
+          v = *c0;
+          m = -3e38f;
+          for (int dz = -1; dz <= 1; ++dz)
+            for (int dy = -1; dy <= 1; ++dy) {
+              int zz = gz + dz; zz += zz < 0 ? pz : 0; zz -= zz >= pz ? pz : 0;
+              int yy = gy + dy; yy += yy < 0 ? py : 0; yy -= yy >= py ? py : 0;
+              const float *r = pcm + ((long)zz * py + yy) * px;
+              int xm = gx - 1; xm += xm < 0 ? px : 0;
+              int xp = gx + 1; xp -= xp >= px ? px : 0;
+              float a = r[xm], bq = r[gx], c = r[xp];
+              if (dz == 0 && dy == 0) bq = -3e38f;
+              m = fmaxf(m, fmaxf(fmaxf(a, bq), c));
+            }
+        } else {
+          const int base = ((lz + 1) * HY + ly + 1) * HX + lx + 1;
+          v = tile[base];
+          m = fmaxf(tile[base - 1], tile[base + 1]);
+#pragma unroll
+          for (int dz = 0; dz <= 2; ++dz)
+#pragma unroll
+            for (int dy = 0; dy <= 2; ++dy) {
+              if (dz == 1 && dy == 1) continue;
+              const int b2 = base + (dz - 1) * HY * HX + (dy - 1) * HX;
+              m = fmaxf(m,
+                        fmaxf(fmaxf(tile[b2 - 1], tile[b2]), tile[b2 + 1]));
+            }
+        }
+        if (inb && v > m)
+          pk_insert(tv, ti, v, ((long long)gz * py + gy) * px + gx);
+      }
+    }
+    __syncthreads();
+  }
+  pk_merge_shfl(tv, ti);
+  int lane = tid & 63, wave = tid >> 6;
+  if (lane == 0)
+    for (int k = 0; k < 5; ++k) { wv[wave][k] = tv[k]; wi[wave][k] = ti[k]; }
+  __syncthreads();
+  if (tid == 0) {
+    for (int w = 1; w < 4; ++w)
+      for (int k = 0; k < 5; ++k) pk_insert(tv, ti, wv[w][k], wi[w][k]);
+    bs_peak *o = wgbuf + (long)blockIdx.x * 5;
+    for (int k = 0; k < 5; ++k) o[k] = {tv[k], 0, ti[k]};
+  }
+}
+
+#define HIPCHK(x) if ((x) != hipSuccess) { printf("hiperr @%d\n", __LINE__); exit(1); }
+
+int main() {
+  const int px = 512, py = 512, pz = 512;
+  const long n = (long)px * py * pz;
+  float *d;
+  HIPCHK(hipMalloc(&d, n * 4));
+  std::vector<float> h(n);
+  srand(3);
+  for (long i = 0; i < n; ++i) h[i] = (float)(rand() % 10000) * 1e-4f;
+  HIPCHK(hipMemcpy(d, h.data(), n * 4, hipMemcpyHostToDevice));
+  bs_peak *wb;
+  HIPCHK(hipMalloc(&wb, 2048 * 5 * sizeof(bs_peak)));
+  hipEvent_t e0, e1;
+  HIPCHK(hipEventCreate(&e0));
+  HIPCHK(hipEventCreate(&e1));
+  const char *names[4] = {"prod (LDS tile)", "load-only", "maxima-only",
+                          "global-direct"};
+  const double bytes = n * 4.0;
+  std::vector<std::vector<float>> ms(4);
+  for (int r = 0; r < 7; ++r) {
+    for (int v = 0; v < 4; ++v) {
+      HIPCHK(hipEventRecord(e0, 0));
+      switch (v) {
+        case 0: hipLaunchKernelGGL(k_peak<0>, dim3(2048), dim3(256), 0, 0, d, px, py, pz, wb); break;
+        case 1: hipLaunchKernelGGL(k_peak<1>, dim3(2048), dim3(256), 0, 0, d, px, py, pz, wb); break;
+        case 2: hipLaunchKernelGGL(k_peak<2>, dim3(2048), dim3(256), 0, 0, d, px, py, pz, wb); break;
+        case 3: hipLaunchKernelGGL(k_peak<3>, dim3(2048), dim3(256), 0, 0, d, px, py, pz, wb); break;
+      }
+      HIPCHK(hipEventRecord(e1, 0));
+      HIPCHK(hipEventSynchronize(e1));
+      float m;
+      HIPCHK(hipEventElapsedTime(&m, e0, e1));
+      ms[v].push_back(m);
+    }
+  }
+  for (int v = 0; v < 4; ++v) {
+    std::sort(ms[v].begin(), ms[v].end());
+    printf("%-16s med=%.3f ms  alg_GB/s=%.0f\n", names[v], ms[v][3],
+           bytes / (ms[v][3] * 1e-3) / 1e9);
+  }
+  return 0;
+}
