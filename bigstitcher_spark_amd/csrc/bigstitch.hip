@@ -48,9 +48,9 @@
 #define LPB_S 8    /* lines per block, strided passes */
 #define TPL_S 64   /* 512-thread blocks; 35 KB LDS -> 4 WGs/CU (full
                       32-wave occupancy) */
-#define PK_TX 128  /* peak-scan tile */
+#define PK_TX 128 /* peak-scan (x,y) strip */
 #define PK_TY 8
-#define PK_TZ 8
+#define PK_CZ 64 /* z planes streamed per chunk */
 
 typedef unsigned long long u64;
 
@@ -399,82 +399,115 @@ __device__ void pk_merge_shfl(float (&tv)[5], long long (&ti)[5]) {
   }
 }
 
-/* Tile local-maxima scan: strict 26-neighborhood maxima with periodic
- * wrap [PIN-MAX]; per-WG top-5 -> wgbuf. Interior tiles (the common
- * case) load their halo without any modulo index math. */
+/* Streaming local-maxima scan: strict 26-neighborhood maxima with
+ * periodic wrap [PIN-MAX]; per-WG top-5 -> wgbuf. A WG owns an (x,y)
+ * strip of PK_TX x PK_TY and streams a z-chunk of PK_CZ planes through
+ * a 4-plane rolling LDS buffer: one barrier per plane, the load of
+ * plane z+2 issued before computing plane z (load/compute overlap),
+ * and read amplification only (PK_TX+2)(PK_TY+2)/(PK_TX*PK_TY) ~ 1.27
+ * plus two preload planes per chunk — vs 1.59x for a full 3-D halo
+ * tile, which also serializes load and compute at its barriers
+ * (ablation: tools/probe_peak.hip). */
 __global__ __launch_bounds__(256) void k_peak_tile(
     const float *pcm, int px, int py, int pz, bs_peak *wgbuf) {
-  __shared__ float tile[(PK_TZ + 2) * (PK_TY + 2) * (PK_TX + 2)];
+  const int HX = PK_TX + 2, HY = PK_TY + 2;
+  __shared__ float pl[4][HY * HX];
   __shared__ float wv[4][5];
   __shared__ long long wi[4][5];
   const int tid = threadIdx.x;
   const int ntx = (px + PK_TX - 1) / PK_TX;
   const int nty = (py + PK_TY - 1) / PK_TY;
-  const int ntz = (pz + PK_TZ - 1) / PK_TZ;
-  const int HX = PK_TX + 2, HY = PK_TY + 2, HZ = PK_TZ + 2;
-  const long ntiles = (long)ntx * nty * ntz;
+  const int ncz = (pz + PK_CZ - 1) / PK_CZ;
+  const long nchunks = (long)ntx * nty * ncz;
+  /* x/y halo coordinates wrap at most one period unless the volume is
+   * smaller than one strip+halo (tiny parity cases): uniform branch */
+  const bool wrap1 = px >= HX && py >= HY;
   float tv[5];
   long long ti[5];
   for (int k = 0; k < 5; ++k) { tv[k] = -3.0e38f; ti[k] = 0x7fffffffffffffffLL; }
-  for (long t0 = blockIdx.x; t0 < ntiles; t0 += gridDim.x) {
-    int bx = (int)(t0 % ntx);
-    int by = (int)((t0 / ntx) % nty);
-    int bz = (int)(t0 / ((long)ntx * nty));
-    int x0 = bx * PK_TX, y0 = by * PK_TY, z0 = bz * PK_TZ;
-    const bool interior = x0 > 0 && y0 > 0 && z0 > 0 && x0 + PK_TX < px &&
-                          y0 + PK_TY < py && z0 + PK_TZ < pz;
-    if (interior) {
-      const float *base =
-          pcm + ((long)(z0 - 1) * py + (y0 - 1)) * px + (x0 - 1);
-      for (int i = tid; i < HX * HY * HZ; i += 256) {
-        int lx = i % HX, t = i / HX, ly = t % HY, lz = t / HY;
-        tile[i] = base[((long)lz * py + ly) * px + lx];
-      }
-    } else if (px >= HX && py >= HY && pz >= HZ) {
-      /* boundary tile, but every halo coordinate wraps at most one
-       * period: branchless select instead of 3 runtime idivs/element */
-      for (int i = tid; i < HX * HY * HZ; i += 256) {
-        int lx = i % HX, t = i / HX, ly = t % HY, lz = t / HY;
-        int gx = x0 + lx - 1, gy = y0 + ly - 1, gz = z0 + lz - 1;
-        gx += gx < 0 ? px : 0;
-        gx -= gx >= px ? px : 0;
-        gy += gy < 0 ? py : 0;
-        gy -= gy >= py ? py : 0;
-        gz += gz < 0 ? pz : 0;
-        gz -= gz >= pz ? pz : 0;
-        tile[i] = pcm[((long)gz * py + gy) * px + gx];
-      }
-    } else { /* tiny volumes: full modulo wrap */
-      for (int i = tid; i < HX * HY * HZ; i += 256) {
-        int lx = i % HX, t = i / HX, ly = t % HY, lz = t / HY;
-        int gx = (x0 + lx - 1 + px) % px;
-        int gy = (y0 + ly - 1 + py) % py;
-        int gz = (z0 + lz - 1 + pz) % pz;
-        tile[i] = pcm[((long)gz * py + gy) * px + gx];
+  for (long t0 = blockIdx.x; t0 < nchunks; t0 += gridDim.x) {
+    const int bx = (int)(t0 % ntx);
+    const int by = (int)((t0 / ntx) % nty);
+    const int bz = (int)(t0 / ((long)ntx * nty));
+    const int x0 = bx * PK_TX, y0 = by * PK_TY, z0 = bz * PK_CZ;
+    const int zend = min(z0 + PK_CZ, pz);
+    /* preload planes z0-1, z0, z0+1 into slots 0,1,2 */
+    for (int p = -1; p <= 1; ++p) {
+      int gz = z0 + p;
+      gz += gz < 0 ? pz : 0;
+      gz -= gz >= pz ? pz : 0;
+      const float *src = pcm + (long)gz * py * px;
+      float *dst = pl[p + 1];
+      if (wrap1) {
+        for (int i = tid; i < HX * HY; i += 256) {
+          int lx = i % HX, ly = i / HX;
+          int gx = x0 + lx - 1, gy = y0 + ly - 1;
+          gx += gx < 0 ? px : 0;
+          gx -= gx >= px ? px : 0;
+          gy += gy < 0 ? py : 0;
+          gy -= gy >= py ? py : 0;
+          dst[i] = src[(long)gy * px + gx];
+        }
+      } else {
+        for (int i = tid; i < HX * HY; i += 256) {
+          int lx = i % HX, ly = i / HX;
+          int gx = (x0 + lx - 1 + px) % px, gy = (y0 + ly - 1 + py) % py;
+          dst[i] = src[(long)gy * px + gx];
+        }
       }
     }
     __syncthreads();
-    for (int i = tid; i < PK_TX * PK_TY * PK_TZ; i += 256) {
-      int lx = i % PK_TX, t = i / PK_TX, ly = t % PK_TY, lz = t / PK_TY;
-      int gx = x0 + lx, gy = y0 + ly, gz = z0 + lz;
-      bool inb = gx < px && gy < py && gz < pz;
-      /* branchless 26-neighbor max: every LDS offset is base + constant
-       * (the #pragma unroll makes dz/dy compile-time), no divergence */
-      const int base = ((lz + 1) * HY + ly + 1) * HX + lx + 1;
-      float v = tile[base];
-      float m = fmaxf(tile[base - 1], tile[base + 1]);
-#pragma unroll
-      for (int dz = 0; dz <= 2; ++dz)
-#pragma unroll
-        for (int dy = 0; dy <= 2; ++dy) {
-          if (dz == 1 && dy == 1) continue;
-          const int b2 = base + (dz - 1) * HY * HX + (dy - 1) * HX;
-          m = fmaxf(m, fmaxf(fmaxf(tile[b2 - 1], tile[b2]), tile[b2 + 1]));
+    for (int z = z0; z < zend; ++z) {
+      /* issue the load of plane z+2 into the free slot before the
+       * maxima math on plane z (z+2 wraps at most one period: z<pz) */
+      {
+        int gz = z + 2;
+        gz -= gz >= pz ? pz : 0;
+        gz -= gz >= pz ? pz : 0; /* pz<=2: two single-period steps */
+        const float *src = pcm + (long)gz * py * px;
+        float *dst = pl[(z - z0 + 3) & 3];
+        if (wrap1) {
+          for (int i = tid; i < HX * HY; i += 256) {
+            int lx = i % HX, ly = i / HX;
+            int gx = x0 + lx - 1, gy = y0 + ly - 1;
+            gx += gx < 0 ? px : 0;
+            gx -= gx >= px ? px : 0;
+            gy += gy < 0 ? py : 0;
+            gy -= gy >= py ? py : 0;
+            dst[i] = src[(long)gy * px + gx];
+          }
+        } else {
+          for (int i = tid; i < HX * HY; i += 256) {
+            int lx = i % HX, ly = i / HX;
+            int gx = (x0 + lx - 1 + px) % px, gy = (y0 + ly - 1 + py) % py;
+            dst[i] = src[(long)gy * px + gx];
+          }
         }
-      if (inb && v > m)
-        pk_insert(tv, ti, v, ((long long)gz * py + gy) * px + gx);
+      }
+      /* strict 26-max on plane z from slots (z-z0)%4 .. +2: every LDS
+       * offset is base + compile-time constant, no divergence */
+      const float *pm = pl[(z - z0) & 3];
+      const float *pc = pl[(z - z0 + 1) & 3];
+      const float *pp = pl[(z - z0 + 2) & 3];
+      for (int i = tid; i < PK_TX * PK_TY; i += 256) {
+        int lx = i % PK_TX, ly = i / PK_TX;
+        int gx = x0 + lx, gy = y0 + ly;
+        const int base = (ly + 1) * HX + lx + 1;
+        float v = pc[base];
+        float m = fmaxf(pc[base - 1], pc[base + 1]);
+#pragma unroll
+        for (int dy = -1; dy <= 1; ++dy) {
+          const int b2 = base + dy * HX;
+          m = fmaxf(m, fmaxf(fmaxf(pm[b2 - 1], pm[b2]), pm[b2 + 1]));
+          m = fmaxf(m, fmaxf(fmaxf(pp[b2 - 1], pp[b2]), pp[b2 + 1]));
+          if (dy != 0)
+            m = fmaxf(m, fmaxf(fmaxf(pc[b2 - 1], pc[b2]), pc[b2 + 1]));
+        }
+        if (gx < px && gy < py && v > m)
+          pk_insert(tv, ti, v, ((long long)z * py + gy) * px + gx);
+      }
+      __syncthreads(); /* load of z+2 complete; slots rotate */
     }
-    __syncthreads(); /* tile LDS reused next iteration */
   }
   pk_merge_shfl(tv, ti);
   int lane = tid & 63, wave = tid >> 6;
@@ -1390,7 +1423,7 @@ static int stitch_phaseA(bs_ctx *c, bs_slot *sl, const bs_pair_desc &pd,
   c->dbg_pz = Pz;
   /* peak scan [PIN-MAX] */
   long ntiles = (long)((Px + PK_TX - 1) / PK_TX) *
-                ((Py + PK_TY - 1) / PK_TY) * ((Pz + PK_TZ - 1) / PK_TZ);
+                ((Py + PK_TY - 1) / PK_TY) * ((Pz + PK_CZ - 1) / PK_CZ);
   long npkwg = std::min(2048L, ntiles);
   int rc2 = ensure_dev(c, (void **)&sl->wgpk, &sl->wgpk_cap,
                        (size_t)npkwg * 5 * sizeof(bs_peak));

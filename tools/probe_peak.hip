@@ -143,6 +143,106 @@ __global__ __launch_bounds__(256) void k_peak(const float *pcm, int px,
   }
 }
 
+
+/* VAR 4: streaming z-plane scan. WG owns an (x,y) strip of TX4 x TY4 and
+ * streams a z-chunk of CZ4 planes through a 4-plane rolling LDS buffer:
+ * one barrier per plane, load of plane z+1 issued before computing plane
+ * z (overlap), read amplification only (TX4+2)(TY4+2)/(TX4*TY4) ~ 1.27
+ * plus 2 planes per chunk boundary. */
+#define TX4 128
+#define TY4 8
+#define CZ4 64
+__global__ __launch_bounds__(256) void k_peak_stream(const float *pcm,
+                                                     int px, int py, int pz,
+                                                     bs_peak *wgbuf) {
+  const int HX = TX4 + 2, HY = TY4 + 2;
+  __shared__ float pl[4][HY * HX];
+  __shared__ float wv[4][5];
+  __shared__ long long wi[4][5];
+  const int tid = threadIdx.x;
+  const int ntx = (px + TX4 - 1) / TX4;
+  const int nty = (py + TY4 - 1) / TY4;
+  const int ncz = (pz + CZ4 - 1) / CZ4;
+  const long nchunks = (long)ntx * nty * ncz;
+  float tv[5];
+  long long ti[5];
+  for (int k = 0; k < 5; ++k) { tv[k] = -3.0e38f; ti[k] = 0x7fffffffffffffffLL; }
+  for (long c = blockIdx.x; c < nchunks; c += gridDim.x) {
+    const int bx = (int)(c % ntx);
+    const int by = (int)((c / ntx) % nty);
+    const int bz = (int)(c / ((long)ntx * nty));
+    const int x0 = bx * TX4, y0 = by * TY4, z0 = bz * CZ4;
+    const int zend = min(z0 + CZ4, pz);
+    /* preload planes z0-1, z0, z0+1 into slots 0,1,2 */
+    for (int p = -1; p <= 1; ++p) {
+      int gz = z0 + p;
+      gz += gz < 0 ? pz : 0;
+      gz -= gz >= pz ? pz : 0;
+      const float *src = pcm + (long)gz * py * px;
+      float *dst = pl[p + 1];
+      for (int i = tid; i < HX * HY; i += 256) {
+        int lx = i % HX, ly = i / HX;
+        int gx = x0 + lx - 1, gy = y0 + ly - 1;
+        gx += gx < 0 ? px : 0;  gx -= gx >= px ? px : 0;
+        gy += gy < 0 ? py : 0;  gy -= gy >= py ? py : 0;
+        dst[i] = src[(long)gy * px + gx];
+      }
+    }
+    __syncthreads();
+    for (int z = z0; z < zend; ++z) {
+      /* issue load of plane z+2 into the free slot (z+3)%4 */
+      {
+        int gz = z + 2;
+        gz += gz < 0 ? pz : 0;
+        gz -= gz >= pz ? pz : 0;
+        const float *src = pcm + (long)gz * py * px;
+        float *dst = pl[(z - z0 + 3) & 3];
+        for (int i = tid; i < HX * HY; i += 256) {
+          int lx = i % HX, ly = i / HX;
+          int gx = x0 + lx - 1, gy = y0 + ly - 1;
+          gx += gx < 0 ? px : 0;  gx -= gx >= px ? px : 0;
+          gy += gy < 0 ? py : 0;  gy -= gy >= py ? py : 0;
+          dst[i] = src[(long)gy * px + gx];
+        }
+      }
+      /* maxima on plane z from slots (z-z0)%4, +1, +2 */
+      const float *pm = pl[(z - z0) & 3];
+      const float *pc = pl[(z - z0 + 1) & 3];
+      const float *pp = pl[(z - z0 + 2) & 3];
+      for (int i = tid; i < TX4 * TY4; i += 256) {
+        int lx = i % TX4, ly = i / TX4;
+        int gx = x0 + lx, gy = y0 + ly;
+        if (gx >= px || gy >= py) continue;
+        const int b = (ly + 1) * HX + lx + 1;
+        float v = pc[b];
+        float m = fmaxf(pc[b - 1], pc[b + 1]);
+#pragma unroll
+        for (int dy = -1; dy <= 1; ++dy) {
+          int b2 = b + dy * HX;
+          m = fmaxf(m, fmaxf(fmaxf(pm[b2 - 1], pm[b2]), pm[b2 + 1]));
+          m = fmaxf(m, fmaxf(fmaxf(pp[b2 - 1], pp[b2]), pp[b2 + 1]));
+          if (dy != 0)
+            m = fmaxf(m, fmaxf(fmaxf(pc[b2 - 1], pc[b2]), pc[b2 + 1]));
+        }
+        if (v > m)
+          pk_insert(tv, ti, v, ((long long)z * py + gy) * px + gx);
+      }
+      __syncthreads();
+    }
+  }
+  pk_merge_shfl(tv, ti);
+  int lane = tid & 63, wave = tid >> 6;
+  if (lane == 0)
+    for (int k = 0; k < 5; ++k) { wv[wave][k] = tv[k]; wi[wave][k] = ti[k]; }
+  __syncthreads();
+  if (tid == 0) {
+    for (int w = 1; w < 4; ++w)
+      for (int k = 0; k < 5; ++k) pk_insert(tv, ti, wv[w][k], wi[w][k]);
+    bs_peak *o = wgbuf + (long)blockIdx.x * 5;
+    for (int k = 0; k < 5; ++k) o[k] = {tv[k], 0, ti[k]};
+  }
+}
+
 #define HIPCHK(x) if ((x) != hipSuccess) { printf("hiperr @%d\n", __LINE__); exit(1); }
 
 int main() {
@@ -159,18 +259,19 @@ int main() {
   hipEvent_t e0, e1;
   HIPCHK(hipEventCreate(&e0));
   HIPCHK(hipEventCreate(&e1));
-  const char *names[4] = {"prod (LDS tile)", "load-only", "maxima-only",
-                          "global-direct"};
+  const char *names[5] = {"prod (LDS tile)", "load-only", "maxima-only",
+                          "global-direct", "stream-z"};
   const double bytes = n * 4.0;
-  std::vector<std::vector<float>> ms(4);
+  std::vector<std::vector<float>> ms(5);
   for (int r = 0; r < 7; ++r) {
-    for (int v = 0; v < 4; ++v) {
+    for (int v = 0; v < 5; ++v) {
       HIPCHK(hipEventRecord(e0, 0));
       switch (v) {
         case 0: hipLaunchKernelGGL(k_peak<0>, dim3(2048), dim3(256), 0, 0, d, px, py, pz, wb); break;
         case 1: hipLaunchKernelGGL(k_peak<1>, dim3(2048), dim3(256), 0, 0, d, px, py, pz, wb); break;
         case 2: hipLaunchKernelGGL(k_peak<2>, dim3(2048), dim3(256), 0, 0, d, px, py, pz, wb); break;
         case 3: hipLaunchKernelGGL(k_peak<3>, dim3(2048), dim3(256), 0, 0, d, px, py, pz, wb); break;
+        case 4: hipLaunchKernelGGL(k_peak_stream, dim3(2048), dim3(256), 0, 0, d, px, py, pz, wb); break;
       }
       HIPCHK(hipEventRecord(e1, 0));
       HIPCHK(hipEventSynchronize(e1));
@@ -179,10 +280,25 @@ int main() {
       ms[v].push_back(m);
     }
   }
-  for (int v = 0; v < 4; ++v) {
+  for (int v = 0; v < 5; ++v) {
     std::sort(ms[v].begin(), ms[v].end());
     printf("%-16s med=%.3f ms  alg_GB/s=%.0f\n", names[v], ms[v][3],
            bytes / (ms[v][3] * 1e-3) / 1e9);
+  }
+  /* correctness: top-1 of prod vs stream over merged wg buffers */
+  {
+    std::vector<bs_peak> hb(2048 * 5);
+    float bv[2]; long long bi[2];
+    for (int v = 0; v < 2; ++v) {
+      if (v == 0) hipLaunchKernelGGL(k_peak<0>, dim3(2048), dim3(256), 0, 0, d, px, py, pz, wb);
+      else hipLaunchKernelGGL(k_peak_stream, dim3(2048), dim3(256), 0, 0, d, px, py, pz, wb);
+      HIPCHK(hipMemcpy(hb.data(), wb, hb.size() * sizeof(bs_peak), hipMemcpyDeviceToHost));
+      bv[v] = -3e38f; bi[v] = -1;
+      for (auto &p : hb)
+        if (p.v > bv[v] || (p.v == bv[v] && p.idx < bi[v])) { bv[v] = p.v; bi[v] = p.idx; }
+    }
+    printf("top1 prod=(%.6g,%lld) stream=(%.6g,%lld) %s\n", bv[0], bi[0],
+           bv[1], bi[1], (bv[0] == bv[1] && bi[0] == bi[1]) ? "MATCH" : "MISMATCH");
   }
   return 0;
 }
