@@ -364,6 +364,148 @@ __global__ __launch_bounds__(LPB_X *TPL_X) void k_fft_x_inv(
   }
 }
 
+/* ----------------------- wave-resident x passes (pow2 fast paths) */
+
+/* h-point complex FFT resident in one wave's registers, E = h/64
+ * elements per lane at positions p = e*64 + lane. Radix-2 DIT over
+ * bit-reversed-loaded data: strides < 64 exchange via __shfl_xor,
+ * strides >= 64 are in-register butterflies — no data LDS, no
+ * barriers, no cross-wave coupling (the generic fft_lds couples all
+ * lines of a block at every stage's __syncthreads). tw = the h-point
+ * half table (h/2 entries). */
+template <int E>
+__device__ __forceinline__ void ffth_wave(f2 (&v)[E], int lane,
+                                          const f2 *tw, int dir) {
+  constexpr int h = 64 * E;
+#pragma unroll
+  for (int s = 1; s < 64 && s < h; s <<= 1) {
+    f2 w = tw[(lane & (s - 1)) * ((h / 2) / s)];
+    if (dir < 0) w.y = -w.y;
+    const bool up = lane & s;
+    const float sg = up ? -1.0f : 1.0f;
+#pragma unroll
+    for (int e = 0; e < E; ++e) {
+      f2 t = {__shfl_xor(v[e].x, s), __shfl_xor(v[e].y, s)};
+      f2 b = up ? v[e] : t;
+      f2 a = up ? t : v[e];
+      f2 wb = cmul(b, w);
+      v[e] = {a.x + sg * wb.x, a.y + sg * wb.y};
+    }
+  }
+#pragma unroll
+  for (int s = 64; s < h; s <<= 1) {
+    const int es = s >> 6; /* element-index stride */
+#pragma unroll
+    for (int e = 0; e < E; ++e) {
+      if (e & es) continue; /* pair (e, e+es) handled once */
+      const int off = (e & (es - 1)) * 64 + lane;
+      f2 w = tw[off * ((h / 2) / s)];
+      if (dir < 0) w.y = -w.y;
+      f2 wb = cmul(v[e + es], w);
+      f2 a = v[e];
+      v[e] = {a.x + wb.x, a.y + wb.y};
+      v[e + es] = {a.x - wb.x, a.y - wb.y};
+    }
+  }
+}
+
+#define XW_WPB 4 /* lines (waves) per block */
+
+/* Specialization of k_fft_x_fwd for n = 128*E (E = 1,2,4,8 covers the
+ * production pads 128..1024): one wave per line, same packed-real
+ * math, the Hermitian unpack's ld[h-k] read via __shfl. */
+template <int E>
+__global__ __launch_bounds__(64 * XW_WPB) void k_fft_x_fwd_w(
+    bs_region in, f2 *out, long cxp, int py, const f2 *twg) {
+  constexpr int h = 64 * E, log2h = 6 + (E == 2) + 2 * (E == 4) + 3 * (E == 8);
+  __shared__ f2 tw[h / 2]; /* h-pt table: twg[2i] */
+  __shared__ f2 twn[h];    /* n-pt table (unpack) */
+  const int tid = threadIdx.x;
+  for (int i = tid; i < h / 2; i += 64 * XW_WPB) tw[i] = twg[2 * i];
+  for (int i = tid; i < h; i += 64 * XW_WPB) twn[i] = twg[i];
+  __syncthreads(); /* the only barrier: tables ready */
+  const int lane = tid & 63, wv = tid >> 6;
+  const long nlines = (long)in.my * in.mz;
+  for (long lid = (long)blockIdx.x * XW_WPB + wv; lid < nlines;
+       lid += (long)gridDim.x * XW_WPB) {
+    const int y = (int)(lid % in.my), z = (int)(lid / in.my);
+    const unsigned short *src =
+        in.ptr + (in.oz + z) * in.sxy + (in.oy + y) * in.sx + in.ox;
+    f2 v[E];
+    const bool al4 = ((size_t)src & 3) == 0 && in.mx >= 2 * h;
+#pragma unroll
+    for (int e = 0; e < E; ++e) {
+      const int j = (int)brev_n((unsigned)(e * 64 + lane), log2h);
+      if (al4) {
+        unsigned w = ((const unsigned *)src)[j];
+        v[e] = {(float)(w & 0xFFFF), (float)(w >> 16)};
+      } else {
+        float xa = (2 * j < in.mx) ? (float)src[2 * j] : 0.0f;
+        float xb = (2 * j + 1 < in.mx) ? (float)src[2 * j + 1] : 0.0f;
+        v[e] = {xa, xb};
+      }
+    }
+    ffth_wave<E>(v, lane, tw, +1);
+    f2 *o = out + ((long)z * py + y) * cxp;
+#pragma unroll
+    for (int e = 0; e < E; ++e) {
+      /* position k = e*64+lane; Z[h-k] lives at element E-1-e lane
+       * 64-lane (lane>0) or element (E-e)%E lane 0 (lane==0) */
+      f2 zr = {__shfl(v[E - 1 - e].x, (64 - lane) & 63),
+               __shfl(v[E - 1 - e].y, (64 - lane) & 63)};
+      f2 z0 = {__shfl(v[(E - e) % E].x, 0), __shfl(v[(E - e) % E].y, 0)};
+      f2 zm = lane ? zr : z0;
+      const int k = e * 64 + lane;
+      if (k == 0) {
+        o[0] = {v[0].x + v[0].y, 0.0f};
+        o[h] = {v[0].x - v[0].y, 0.0f};
+      } else {
+        f2 ze = {0.5f * (v[e].x + zm.x), 0.5f * (v[e].y - zm.y)};
+        f2 dd = {v[e].x - zm.x, v[e].y + zm.y};  /* Zk - conj(Zmk) */
+        f2 zo = {0.5f * dd.y, -0.5f * dd.x};     /* -i/2 * dd */
+        f2 wzo = cmul(twn[k], zo);
+        o[k] = {ze.x + wzo.x, ze.y + wzo.y};
+      }
+    }
+  }
+}
+
+/* Specialization of k_fft_x_inv for n = 128*E: identical Z2
+ * construction, one wave per line, natural-order float2 row writes. */
+template <int E>
+__global__ __launch_bounds__(64 * XW_WPB) void k_fft_x_inv_w(
+    const f2 *in, float *out, long cxp, long nlines, const f2 *twg) {
+  constexpr int h = 64 * E, log2h = 6 + (E == 2) + 2 * (E == 4) + 3 * (E == 8);
+  __shared__ f2 tw[h / 2];
+  __shared__ f2 twn[h];
+  const int tid = threadIdx.x;
+  for (int i = tid; i < h / 2; i += 64 * XW_WPB) tw[i] = twg[2 * i];
+  for (int i = tid; i < h; i += 64 * XW_WPB) twn[i] = twg[i];
+  __syncthreads();
+  const int lane = tid & 63, wv = tid >> 6;
+  for (long lid = (long)blockIdx.x * XW_WPB + wv; lid < nlines;
+       lid += (long)gridDim.x * XW_WPB) {
+    const f2 *src = in + lid * cxp;
+    f2 v[E];
+#pragma unroll
+    for (int e = 0; e < E; ++e) {
+      const int k = (int)brev_n((unsigned)(e * 64 + lane), log2h);
+      f2 xk = src[k], xm = src[h - k];
+      f2 A = {xk.x + xm.x, xk.y - xm.y}; /* X[k] + conj(X[h-k]) */
+      f2 B = {xk.x - xm.x, xk.y + xm.y}; /* X[k] - conj(X[h-k]) */
+      f2 wc = twn[k];
+      wc.y = -wc.y;
+      f2 wb = cmul(wc, B);
+      v[e] = {A.x - wb.y, A.y + wb.x}; /* A + i*wb */
+    }
+    ffth_wave<E>(v, lane, tw, -1);
+    f2 *o = (f2 *)(out + lid * 2 * h);
+#pragma unroll
+    for (int e = 0; e < E; ++e)
+      o[e * 64 + lane] = v[e]; /* (pcm[2p], pcm[2p+1]) */
+  }
+}
+
 /* ------------------------------------------------------------- peak scan */
 
 struct bs_peak {
@@ -1364,12 +1506,29 @@ static int stitch_phaseA(bs_ctx *c, bs_slot *sl, const bs_pair_desc &pd,
   f2 *spec[2] = {sl->spec, sl->spec + spec_half};
   for (int t = 0; t < 2; ++t) {
     long nlines = (long)sl->reg[t].my * sl->reg[t].mz;
-    long ngrp = (nlines + LPB_X - 1) / LPB_X;
-    size_t lds = ((Px / 4) + (size_t)LPB_X * (Px / 2)) * sizeof(f2);
     bs_tim tt(c, BS_K_FFT_X_FWD, sl->stream);
-    hipLaunchKernelGGL(k_fft_x_fwd, dim3(std::min(4096L, ngrp)),
-                       dim3(LPB_X * TPL_X), lds, sl->stream, sl->reg[t],
-                       spec[t], Px, ilog2(Px), Cx, Cxp, Py, twx);
+    if (Px >= 128 && Px <= 1024) { /* wave-resident fast path */
+      long ngrp = (nlines + XW_WPB - 1) / XW_WPB;
+      dim3 g(std::min(4096L, ngrp)), b(64 * XW_WPB);
+      if (Px == 128)
+        hipLaunchKernelGGL(k_fft_x_fwd_w<1>, g, b, 0, sl->stream,
+                           sl->reg[t], spec[t], Cxp, Py, twx);
+      else if (Px == 256)
+        hipLaunchKernelGGL(k_fft_x_fwd_w<2>, g, b, 0, sl->stream,
+                           sl->reg[t], spec[t], Cxp, Py, twx);
+      else if (Px == 512)
+        hipLaunchKernelGGL(k_fft_x_fwd_w<4>, g, b, 0, sl->stream,
+                           sl->reg[t], spec[t], Cxp, Py, twx);
+      else
+        hipLaunchKernelGGL(k_fft_x_fwd_w<8>, g, b, 0, sl->stream,
+                           sl->reg[t], spec[t], Cxp, Py, twx);
+    } else {
+      long ngrp = (nlines + LPB_X - 1) / LPB_X;
+      size_t lds = ((Px / 4) + (size_t)LPB_X * (Px / 2)) * sizeof(f2);
+      hipLaunchKernelGGL(k_fft_x_fwd, dim3(std::min(4096L, ngrp)),
+                         dim3(LPB_X * TPL_X), lds, sl->stream, sl->reg[t],
+                         spec[t], Px, ilog2(Px), Cx, Cxp, Py, twx);
+    }
   }
   int nchunks = (Cx + LPB_S - 1) / LPB_S;
   for (int t = 0; t < 2; ++t) {
@@ -1410,12 +1569,29 @@ static int stitch_phaseA(bs_ctx *c, bs_slot *sl, const bs_pair_desc &pd,
   }
   {
     long nlines = (long)Pz * Py;
-    long ngrp = (nlines + LPB_X - 1) / LPB_X;
-    size_t lds = ((Px / 4) + (size_t)LPB_X * (Px / 2)) * sizeof(f2);
     bs_tim tt(c, BS_K_FFT_X_INV, sl->stream);
-    hipLaunchKernelGGL(k_fft_x_inv, dim3(std::min(4096L, ngrp)),
-                       dim3(LPB_X * TPL_X), lds, sl->stream, spec[0],
-                       sl->pcm, Px, ilog2(Px), Cx, Cxp, nlines, twx);
+    if (Px >= 128 && Px <= 1024) { /* wave-resident fast path */
+      long ngrp = (nlines + XW_WPB - 1) / XW_WPB;
+      dim3 g(std::min(4096L, ngrp)), b(64 * XW_WPB);
+      if (Px == 128)
+        hipLaunchKernelGGL(k_fft_x_inv_w<1>, g, b, 0, sl->stream, spec[0],
+                           sl->pcm, Cxp, nlines, twx);
+      else if (Px == 256)
+        hipLaunchKernelGGL(k_fft_x_inv_w<2>, g, b, 0, sl->stream, spec[0],
+                           sl->pcm, Cxp, nlines, twx);
+      else if (Px == 512)
+        hipLaunchKernelGGL(k_fft_x_inv_w<4>, g, b, 0, sl->stream, spec[0],
+                           sl->pcm, Cxp, nlines, twx);
+      else
+        hipLaunchKernelGGL(k_fft_x_inv_w<8>, g, b, 0, sl->stream, spec[0],
+                           sl->pcm, Cxp, nlines, twx);
+    } else {
+      long ngrp = (nlines + LPB_X - 1) / LPB_X;
+      size_t lds = ((Px / 4) + (size_t)LPB_X * (Px / 2)) * sizeof(f2);
+      hipLaunchKernelGGL(k_fft_x_inv, dim3(std::min(4096L, ngrp)),
+                         dim3(LPB_X * TPL_X), lds, sl->stream, spec[0],
+                         sl->pcm, Px, ilog2(Px), Cx, Cxp, nlines, twx);
+    }
   }
   c->dbg_pcm = sl->pcm;
   c->dbg_px = Px;
