@@ -49,7 +49,7 @@
 #define TPL_S 64   /* 512-thread blocks; 35 KB LDS -> 4 WGs/CU (full
                       32-wave occupancy) */
 #define PK_TX 128 /* peak-scan (x,y) strip */
-#define PK_TY 8
+#define PK_TY 16
 #define PK_CZ 64 /* z planes streamed per chunk */
 
 typedef unsigned long long u64;
@@ -562,7 +562,9 @@ __global__ __launch_bounds__(256) void k_peak_tile(
   const int ncz = (pz + PK_CZ - 1) / PK_CZ;
   const long nchunks = (long)ntx * nty * ncz;
   /* x/y halo coordinates wrap at most one period unless the volume is
-   * smaller than one strip+halo (tiny parity cases): uniform branch */
+   * smaller than one strip+halo (tiny parity cases): uniform branch.
+   * Interior rows load as aligned float4 (x0 is a multiple of PK_TX,
+   * px a pow2): 32 lanes x 16B + 2 scalar edge lanes per row. */
   const bool wrap1 = px >= HX && py >= HY;
   float tv[5];
   long long ti[5];
@@ -573,30 +575,53 @@ __global__ __launch_bounds__(256) void k_peak_tile(
     const int bz = (int)(t0 / ((long)ntx * nty));
     const int x0 = bx * PK_TX, y0 = by * PK_TY, z0 = bz * PK_CZ;
     const int zend = min(z0 + PK_CZ, pz);
+    auto load_plane = [&](int gz, float *dst) {
+      const float *src = pcm + (long)gz * py * px;
+      if (wrap1) {
+        for (int i = tid; i < HY * 34; i += 256) {
+          int t = i % 34, ly = i / 34;
+          int gy = y0 + ly - 1;
+          gy += gy < 0 ? py : 0;
+          gy -= gy >= py ? py : 0;
+          const float *row = src + (long)gy * px;
+          if (t < 32) {
+            int gx = x0 + 4 * t;
+            float4 v;
+            if (gx + 3 < px) {
+              v = *(const float4 *)(row + gx);
+            } else { /* last strip of a px < x0+PK_TX volume: wrap */
+              float tmp[4];
+              for (int q = 0; q < 4; ++q) {
+                int xx = gx + q;
+                xx -= xx >= px ? px : 0;
+                tmp[q] = row[xx];
+              }
+              v = {tmp[0], tmp[1], tmp[2], tmp[3]};
+            }
+            float *d = dst + ly * HX + 1 + 4 * t;
+            d[0] = v.x; d[1] = v.y; d[2] = v.z; d[3] = v.w;
+          } else {
+            int gx = (t == 32) ? x0 - 1 : x0 + PK_TX;
+            gx += gx < 0 ? px : 0;
+            gx -= gx >= px ? px : 0;
+            dst[ly * HX + (t == 32 ? 0 : HX - 1)] = row[gx];
+          }
+        }
+      } else { /* tiny volumes: full modulo wrap */
+        for (int i = tid; i < HX * HY; i += 256) {
+          int lx = i % HX, ly = i / HX;
+          int gx = (x0 + lx - 1 + 4 * px) % px;
+          int gy = (y0 + ly - 1 + py) % py;
+          dst[i] = src[(long)gy * px + gx];
+        }
+      }
+    };
     /* preload planes z0-1, z0, z0+1 into slots 0,1,2 */
     for (int p = -1; p <= 1; ++p) {
       int gz = z0 + p;
       gz += gz < 0 ? pz : 0;
       gz -= gz >= pz ? pz : 0;
-      const float *src = pcm + (long)gz * py * px;
-      float *dst = pl[p + 1];
-      if (wrap1) {
-        for (int i = tid; i < HX * HY; i += 256) {
-          int lx = i % HX, ly = i / HX;
-          int gx = x0 + lx - 1, gy = y0 + ly - 1;
-          gx += gx < 0 ? px : 0;
-          gx -= gx >= px ? px : 0;
-          gy += gy < 0 ? py : 0;
-          gy -= gy >= py ? py : 0;
-          dst[i] = src[(long)gy * px + gx];
-        }
-      } else {
-        for (int i = tid; i < HX * HY; i += 256) {
-          int lx = i % HX, ly = i / HX;
-          int gx = (x0 + lx - 1 + px) % px, gy = (y0 + ly - 1 + py) % py;
-          dst[i] = src[(long)gy * px + gx];
-        }
-      }
+      load_plane(gz, pl[p + 1]);
     }
     __syncthreads();
     for (int z = z0; z < zend; ++z) {
@@ -606,25 +631,7 @@ __global__ __launch_bounds__(256) void k_peak_tile(
         int gz = z + 2;
         gz -= gz >= pz ? pz : 0;
         gz -= gz >= pz ? pz : 0; /* pz<=2: two single-period steps */
-        const float *src = pcm + (long)gz * py * px;
-        float *dst = pl[(z - z0 + 3) & 3];
-        if (wrap1) {
-          for (int i = tid; i < HX * HY; i += 256) {
-            int lx = i % HX, ly = i / HX;
-            int gx = x0 + lx - 1, gy = y0 + ly - 1;
-            gx += gx < 0 ? px : 0;
-            gx -= gx >= px ? px : 0;
-            gy += gy < 0 ? py : 0;
-            gy -= gy >= py ? py : 0;
-            dst[i] = src[(long)gy * px + gx];
-          }
-        } else {
-          for (int i = tid; i < HX * HY; i += 256) {
-            int lx = i % HX, ly = i / HX;
-            int gx = (x0 + lx - 1 + px) % px, gy = (y0 + ly - 1 + py) % py;
-            dst[i] = src[(long)gy * px + gx];
-          }
-        }
+        load_plane(gz, pl[(z - z0 + 3) & 3]);
       }
       /* strict 26-max on plane z from slots (z-z0)%4 .. +2: every LDS
        * offset is base + compile-time constant, no divergence */

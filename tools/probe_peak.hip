@@ -243,6 +243,238 @@ __global__ __launch_bounds__(256) void k_peak_stream(const float *pcm,
   }
 }
 
+
+/* VAR 5/7: stream-z with float4 interior plane loads (32 lanes x 16B +
+ * 2 scalar edge lanes per row) and TY parameterized (8 or 16). */
+template <int TY>
+__global__ __launch_bounds__(256) void k_peak_stream4(const float *pcm,
+                                                      int px, int py, int pz,
+                                                      bs_peak *wgbuf) {
+  const int TX = 128, CZ = 64;
+  const int HX = TX + 2, HY = TY + 2;
+  __shared__ float pl[4][HY * HX];
+  __shared__ float wv[4][5];
+  __shared__ long long wi[4][5];
+  const int tid = threadIdx.x;
+  const int ntx = (px + TX - 1) / TX;
+  const int nty = (py + TY - 1) / TY;
+  const int ncz = (pz + CZ - 1) / CZ;
+  const long nchunks = (long)ntx * nty * ncz;
+  float tv[5];
+  long long ti[5];
+  for (int k = 0; k < 5; ++k) { tv[k] = -3.0e38f; ti[k] = 0x7fffffffffffffffLL; }
+  /* row-task decomposition: 34 tasks per row (32 float4 + 2 edges) */
+  auto load_plane = [&](int gz, float *dst, int x0, int y0) {
+    const float *src = pcm + (long)gz * py * px;
+    for (int i = tid; i < HY * 34; i += 256) {
+      int t = i % 34, ly = i / 34;
+      int gy = y0 + ly - 1;
+      gy += gy < 0 ? py : 0;
+      gy -= gy >= py ? py : 0;
+      const float *row = src + (long)gy * px;
+      if (t < 32) {
+        int gx = x0 + 4 * t; /* aligned: x0 mult of 128, px pow2 */
+        float4 v;
+        if (gx + 3 < px) {
+          v = *(const float4 *)(row + gx);
+        } else { /* last x-strip of a non-multiple px: scalar wrap */
+          float tmp[4];
+          for (int q = 0; q < 4; ++q) {
+            int xx = gx + q;
+            xx -= xx >= px ? px : 0;
+            tmp[q] = row[xx];
+          }
+          v = {tmp[0], tmp[1], tmp[2], tmp[3]};
+        }
+        float *d = dst + ly * HX + 1 + 4 * t;
+        d[0] = v.x; d[1] = v.y; d[2] = v.z; d[3] = v.w;
+      } else {
+        int gx = (t == 32) ? x0 - 1 : x0 + TX;
+        gx += gx < 0 ? px : 0;
+        gx -= gx >= px ? px : 0;
+        dst[ly * HX + (t == 32 ? 0 : HX - 1)] = row[gx];
+      }
+    }
+  };
+  for (long t0 = blockIdx.x; t0 < nchunks; t0 += gridDim.x) {
+    const int bx = (int)(t0 % ntx);
+    const int by = (int)((t0 / ntx) % nty);
+    const int bz = (int)(t0 / ((long)ntx * nty));
+    const int x0 = bx * TX, y0 = by * TY, z0 = bz * CZ;
+    const int zend = min(z0 + CZ, pz);
+    for (int p = -1; p <= 1; ++p) {
+      int gz = z0 + p;
+      gz += gz < 0 ? pz : 0;
+      gz -= gz >= pz ? pz : 0;
+      load_plane(gz, pl[p + 1], x0, y0);
+    }
+    __syncthreads();
+    for (int z = z0; z < zend; ++z) {
+      {
+        int gz = z + 2;
+        gz -= gz >= pz ? pz : 0;
+        gz -= gz >= pz ? pz : 0;
+        load_plane(gz, pl[(z - z0 + 3) & 3], x0, y0);
+      }
+      const float *pm = pl[(z - z0) & 3];
+      const float *pc = pl[(z - z0 + 1) & 3];
+      const float *pp = pl[(z - z0 + 2) & 3];
+      for (int i = tid; i < TX * TY; i += 256) {
+        int lx = i % TX, ly = i / TX;
+        int gx = x0 + lx, gy = y0 + ly;
+        const int base = (ly + 1) * HX + lx + 1;
+        float v = pc[base];
+        float m = fmaxf(pc[base - 1], pc[base + 1]);
+#pragma unroll
+        for (int dy = -1; dy <= 1; ++dy) {
+          const int b2 = base + dy * HX;
+          m = fmaxf(m, fmaxf(fmaxf(pm[b2 - 1], pm[b2]), pm[b2 + 1]));
+          m = fmaxf(m, fmaxf(fmaxf(pp[b2 - 1], pp[b2]), pp[b2 + 1]));
+          if (dy != 0)
+            m = fmaxf(m, fmaxf(fmaxf(pc[b2 - 1], pc[b2]), pc[b2 + 1]));
+        }
+        if (gx < px && gy < py && v > m)
+          pk_insert(tv, ti, v, ((long long)z * py + gy) * px + gx);
+      }
+      __syncthreads();
+    }
+  }
+  pk_merge_shfl(tv, ti);
+  int lane = tid & 63, wave = tid >> 6;
+  if (lane == 0)
+    for (int k = 0; k < 5; ++k) { wv[wave][k] = tv[k]; wi[wave][k] = ti[k]; }
+  __syncthreads();
+  if (tid == 0) {
+    for (int w = 1; w < 4; ++w)
+      for (int k = 0; k < 5; ++k) pk_insert(tv, ti, wv[w][k], wi[w][k]);
+    bs_peak *o = wgbuf + (long)blockIdx.x * 5;
+    for (int k = 0; k < 5; ++k) o[k] = {tv[k], 0, ti[k]};
+  }
+}
+
+
+/* VAR 8: f4-ty16 + semi-separable maxima. Per plane keep xc[ly][lx] =
+ * max3 of the raw row (computed once, one plane behind the raw load);
+ * the 26-neighbor max becomes 2 column-max3 of xc (z+-1 planes), a
+ * column-max2 of xc plus a raw row-max2 (center plane): ~9 max ops
+ * instead of ~26. One barrier per plane; raw loads still overlapped. */
+__global__ __launch_bounds__(256) void k_peak_sep(const float *pcm, int px,
+                                                  int py, int pz,
+                                                  bs_peak *wgbuf) {
+  const int TX = 128, TY = 16, CZ = 64;
+  const int HX = TX + 2, HY = TY + 2;
+  extern __shared__ float lds[];
+  float *pl[4], *xc[4];
+  for (int q = 0; q < 4; ++q) {
+    pl[q] = lds + q * (HY * HX);
+    xc[q] = lds + 4 * (HY * HX) + q * (HY * TX);
+  }
+  __shared__ float wv[4][5];
+  __shared__ long long wi[4][5];
+  const int tid = threadIdx.x;
+  const int ntx = (px + TX - 1) / TX;
+  const int nty = (py + TY - 1) / TY;
+  const int ncz = (pz + CZ - 1) / CZ;
+  const long nchunks = (long)ntx * nty * ncz;
+  float tv[5];
+  long long ti[5];
+  for (int k = 0; k < 5; ++k) { tv[k] = -3.0e38f; ti[k] = 0x7fffffffffffffffLL; }
+  auto load_plane = [&](int gz, float *dst, int x0, int y0) {
+    const float *src = pcm + (long)gz * py * px;
+    for (int i = tid; i < HY * 34; i += 256) {
+      int t = i % 34, ly = i / 34;
+      int gy = y0 + ly - 1;
+      gy += gy < 0 ? py : 0;
+      gy -= gy >= py ? py : 0;
+      const float *row = src + (long)gy * px;
+      if (t < 32) {
+        int gx = x0 + 4 * t;
+        float4 v;
+        if (gx + 3 < px) {
+          v = *(const float4 *)(row + gx);
+        } else {
+          float tmp[4];
+          for (int q = 0; q < 4; ++q) {
+            int xx = gx + q;
+            xx -= xx >= px ? px : 0;
+            tmp[q] = row[xx];
+          }
+          v = {tmp[0], tmp[1], tmp[2], tmp[3]};
+        }
+        float *d = dst + ly * HX + 1 + 4 * t;
+        d[0] = v.x; d[1] = v.y; d[2] = v.z; d[3] = v.w;
+      } else {
+        int gx = (t == 32) ? x0 - 1 : x0 + TX;
+        gx += gx < 0 ? px : 0;
+        gx -= gx >= px ? px : 0;
+        dst[ly * HX + (t == 32 ? 0 : HX - 1)] = row[gx];
+      }
+    }
+  };
+  auto calc_xc = [&](const float *raw, float *dst) {
+    for (int i = tid; i < HY * TX; i += 256) {
+      int lx = i % TX, ly = i / TX;
+      const float *r = raw + ly * HX + lx; /* r[0..2] = x-1,x,x+1 */
+      dst[i] = fmaxf(fmaxf(r[0], r[1]), r[2]);
+    }
+  };
+  for (long t0 = blockIdx.x; t0 < nchunks; t0 += gridDim.x) {
+    const int bx = (int)(t0 % ntx);
+    const int by = (int)((t0 / ntx) % nty);
+    const int bz = (int)(t0 / ((long)ntx * nty));
+    const int x0 = bx * TX, y0 = by * TY, z0 = bz * CZ;
+    const int zend = min(z0 + CZ, pz);
+    for (int p = -1; p <= 1; ++p) {
+      int gz = z0 + p;
+      gz += gz < 0 ? pz : 0;
+      gz -= gz >= pz ? pz : 0;
+      load_plane(gz, pl[p + 1], x0, y0);
+    }
+    __syncthreads();
+    for (int p = -1; p <= 1; ++p) calc_xc(pl[p + 1], xc[p + 1]);
+    __syncthreads();
+    for (int z = z0; z < zend; ++z) {
+      { /* issue raw load of z+2 */
+        int gz = z + 2;
+        gz -= gz >= pz ? pz : 0;
+        gz -= gz >= pz ? pz : 0;
+        load_plane(gz, pl[(z - z0 + 3) & 3], x0, y0);
+      }
+      const int sm = (z - z0) & 3, sc = (z - z0 + 1) & 3,
+                sp = (z - z0 + 2) & 3;
+      const float *rc = pl[sc];
+      const float *xm = xc[sm], *xcc = xc[sc], *xp = xc[sp];
+      for (int i = tid; i < TX * TY; i += 256) {
+        int lx = i % TX, ly = i / TX;
+        int gx = x0 + lx, gy = y0 + ly;
+        const int base = (ly + 1) * HX + lx + 1;
+        const int cb = ly * TX + lx; /* xc row ly = raw row ly (halo incl) */
+        float v = rc[base];
+        float m = fmaxf(fmaxf(xm[cb], xm[cb + TX]), xm[cb + 2 * TX]);
+        m = fmaxf(m, fmaxf(fmaxf(xp[cb], xp[cb + TX]), xp[cb + 2 * TX]));
+        m = fmaxf(m, fmaxf(xcc[cb], xcc[cb + 2 * TX]));
+        m = fmaxf(m, fmaxf(rc[base - 1], rc[base + 1]));
+        if (gx < px && gy < py && v > m)
+          pk_insert(tv, ti, v, ((long long)z * py + gy) * px + gx);
+      }
+      __syncthreads(); /* load(z+2) done; xc slots rotate */
+      calc_xc(pl[(z - z0 + 3) & 3], xc[(z - z0 + 3) & 3]);
+      __syncthreads();
+    }
+  }
+  pk_merge_shfl(tv, ti);
+  int lane = tid & 63, wave = tid >> 6;
+  if (lane == 0)
+    for (int k = 0; k < 5; ++k) { wv[wave][k] = tv[k]; wi[wave][k] = ti[k]; }
+  __syncthreads();
+  if (tid == 0) {
+    for (int w = 1; w < 4; ++w)
+      for (int k = 0; k < 5; ++k) pk_insert(tv, ti, wv[w][k], wi[w][k]);
+    bs_peak *o = wgbuf + (long)blockIdx.x * 5;
+    for (int k = 0; k < 5; ++k) o[k] = {tv[k], 0, ti[k]};
+  }
+}
+
 #define HIPCHK(x) if ((x) != hipSuccess) { printf("hiperr @%d\n", __LINE__); exit(1); }
 
 int main() {
@@ -256,15 +488,17 @@ int main() {
   HIPCHK(hipMemcpy(d, h.data(), n * 4, hipMemcpyHostToDevice));
   bs_peak *wb;
   HIPCHK(hipMalloc(&wb, 2048 * 5 * sizeof(bs_peak)));
+  HIPCHK(hipFuncSetAttribute((const void *)k_peak_sep,
+      hipFuncAttributeMaxDynamicSharedMemorySize, (4*(18*130)+4*(18*128))*4));
   hipEvent_t e0, e1;
   HIPCHK(hipEventCreate(&e0));
   HIPCHK(hipEventCreate(&e1));
-  const char *names[5] = {"prod (LDS tile)", "load-only", "maxima-only",
-                          "global-direct", "stream-z"};
+  const char *names[8] = {"prod (LDS tile)", "load-only", "maxima-only",
+                          "global-direct", "stream-z", "stream-f4-ty8", "stream-f4-ty16", "stream-sep"};
   const double bytes = n * 4.0;
-  std::vector<std::vector<float>> ms(5);
+  std::vector<std::vector<float>> ms(8);
   for (int r = 0; r < 7; ++r) {
-    for (int v = 0; v < 5; ++v) {
+    for (int v = 0; v < 8; ++v) {
       HIPCHK(hipEventRecord(e0, 0));
       switch (v) {
         case 0: hipLaunchKernelGGL(k_peak<0>, dim3(2048), dim3(256), 0, 0, d, px, py, pz, wb); break;
@@ -272,6 +506,9 @@ int main() {
         case 2: hipLaunchKernelGGL(k_peak<2>, dim3(2048), dim3(256), 0, 0, d, px, py, pz, wb); break;
         case 3: hipLaunchKernelGGL(k_peak<3>, dim3(2048), dim3(256), 0, 0, d, px, py, pz, wb); break;
         case 4: hipLaunchKernelGGL(k_peak_stream, dim3(2048), dim3(256), 0, 0, d, px, py, pz, wb); break;
+        case 5: hipLaunchKernelGGL(k_peak_stream4<8>, dim3(2048), dim3(256), 0, 0, d, px, py, pz, wb); break;
+        case 6: hipLaunchKernelGGL(k_peak_stream4<16>, dim3(2048), dim3(256), 0, 0, d, px, py, pz, wb); break;
+        case 7: hipLaunchKernelGGL(k_peak_sep, dim3(2048), dim3(256), (4*(18*130)+4*(18*128))*4, 0, d, px, py, pz, wb); break;
       }
       HIPCHK(hipEventRecord(e1, 0));
       HIPCHK(hipEventSynchronize(e1));
@@ -280,7 +517,7 @@ int main() {
       ms[v].push_back(m);
     }
   }
-  for (int v = 0; v < 5; ++v) {
+  for (int v = 0; v < 8; ++v) {
     std::sort(ms[v].begin(), ms[v].end());
     printf("%-16s med=%.3f ms  alg_GB/s=%.0f\n", names[v], ms[v][3],
            bytes / (ms[v][3] * 1e-3) / 1e9);
@@ -288,17 +525,19 @@ int main() {
   /* correctness: top-1 of prod vs stream over merged wg buffers */
   {
     std::vector<bs_peak> hb(2048 * 5);
-    float bv[2]; long long bi[2];
-    for (int v = 0; v < 2; ++v) {
+    float bv[4]; long long bi[4];
+    for (int v = 0; v < 4; ++v) {
       if (v == 0) hipLaunchKernelGGL(k_peak<0>, dim3(2048), dim3(256), 0, 0, d, px, py, pz, wb);
-      else hipLaunchKernelGGL(k_peak_stream, dim3(2048), dim3(256), 0, 0, d, px, py, pz, wb);
+      else if (v == 1) hipLaunchKernelGGL(k_peak_stream4<8>, dim3(2048), dim3(256), 0, 0, d, px, py, pz, wb);
+      else if (v == 2) hipLaunchKernelGGL(k_peak_stream4<16>, dim3(2048), dim3(256), 0, 0, d, px, py, pz, wb);
+      else hipLaunchKernelGGL(k_peak_sep, dim3(2048), dim3(256), (4*(18*130)+4*(18*128))*4, 0, d, px, py, pz, wb);
       HIPCHK(hipMemcpy(hb.data(), wb, hb.size() * sizeof(bs_peak), hipMemcpyDeviceToHost));
       bv[v] = -3e38f; bi[v] = -1;
       for (auto &p : hb)
         if (p.v > bv[v] || (p.v == bv[v] && p.idx < bi[v])) { bv[v] = p.v; bi[v] = p.idx; }
     }
-    printf("top1 prod=(%.6g,%lld) stream=(%.6g,%lld) %s\n", bv[0], bi[0],
-           bv[1], bi[1], (bv[0] == bv[1] && bi[0] == bi[1]) ? "MATCH" : "MISMATCH");
+    printf("top1 prod=(%.6g,%lld) f4ty8=(%.6g,%lld) f4ty16=(%.6g,%lld) sep=(%.6g,%lld) %s\n", bv[0], bi[0],
+           bv[1], bi[1], bv[2], bi[2], bv[3], bi[3], (bv[0] == bv[1] && bi[0] == bi[1] && bv[1] == bv[2] && bi[1] == bi[2] && bv[2] == bv[3] && bi[2] == bi[3]) ? "MATCH" : "MISMATCH");
   }
   return 0;
 }
