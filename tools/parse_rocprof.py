@@ -22,19 +22,32 @@ import sys
 # k_fft_x_fwd (u32/lane) and k_fft_x_inv (8 B/lane) measure exactly half
 # the algorithmic read -> x2; k_fft_pass (float4 = 16 B/lane) measures
 # the full bytes (raw launch-mix FETCH 0.629 GB == the 0.629 GB
-# algorithmic mix) -> x1.
-CORRECT2X = {"k_fft_x_fwd", "k_fft_x_inv"}
+# algorithmic mix) -> x1. r01_final recalibration for the wave-resident
+# x passes (same per-lane widths): FETCH 0.134 vs 0.268 GB alg (fwd),
+# 0.285 vs 0.539 GB alg (inv) -> x2. k_peak_tile measures BELOW its
+# algorithmic read (0.474 vs 0.537 GB): float4 loads are full-counted
+# and halo rows shared between adjacent y-strips hit in L2 -> x1,
+# genuine reuse. k_rtest u16 gathers -> x1 (1.536 vs 1.48 GB alg,
+# cacheline rounding).
+CORRECT2X = {"k_fft_x_fwd", "k_fft_x_inv", "k_fft_x_fwd_w", "k_fft_x_inv_w"}
 # map mangled display prefix -> bench kernel-stat names (fft_pass splits
 # by launch context are not distinguishable in PMC output; report under
 # one name and let bench match per-kernel names it knows)
 NAME_MAP = {
     "k_fft_x_fwd": ["fft_x_fwd"],
     "k_fft_x_inv": ["fft_x_inv"],
+    "k_fft_x_fwd_w": ["fft_x_fwd"],
+    "k_fft_x_inv_w": ["fft_x_inv"],
     "k_fft_pass": ["fft_y_fwd", "fft_z_fwd", "fft_z_inv", "fft_y_inv"],
     "k_peak_tile": ["peak"],
     "k_rtest": ["corr"],
     "k_fuse": ["fuse"],
 }
+
+
+def norm(disp):
+    n = disp.split("(")[0].split("<")[0].strip()
+    return n[5:] if n.startswith("void ") else n
 
 
 def open_db(path):
@@ -52,7 +65,7 @@ def kernel_times(path):
             FROM rocpd_kernel_dispatch_{s} k
             JOIN rocpd_info_kernel_symbol_{s} ks ON k.kernel_id=ks.id
             GROUP BY 1 ORDER BY 3 DESC"""
-    return [(r[0].split("(")[0], r[1], r[2], r[3]) for r in db.execute(q)]
+    return [(norm(r[0]), r[1], r[2], r[3]) for r in db.execute(q)]
 
 
 def pmc_avg(path):
@@ -62,7 +75,7 @@ def pmc_avg(path):
             JOIN rocpd_kernel_dispatch_{s} k ON p.event_id=k.event_id
             JOIN rocpd_info_kernel_symbol_{s} ks ON k.kernel_id=ks.id
             GROUP BY 1"""
-    return {r[0].split("(")[0]: r[1] for r in db.execute(q)}
+    return {norm(r[0]): r[1] for r in db.execute(q)}
 
 
 def main():

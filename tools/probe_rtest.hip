@@ -228,6 +228,66 @@ __global__ __launch_bounds__(256) void k_rtest2(bs_region a, bs_region b,
   }
 }
 
+/* V3: 256-wide rows, FOUR rows in flight */
+__global__ __launch_bounds__(256) void k_rtest3(bs_region a, bs_region b,
+                                                const bs_cand *cands,
+                                                u64 *sums) {
+  __shared__ u64 ws[4][5];
+  const bs_cand c = cands[blockIdx.y];
+  long nrows = (long)c.ny * c.nz;
+  u64 pa = 0, pb = 0, paa = 0, pbb = 0, pab = 0;
+  auto rowptr_a = [&](long row) {
+    int y = (int)(row % c.ny), z = (int)(row / c.ny);
+    return a.ptr + (a.oz + c.loz + z) * a.sxy + (a.oy + c.loy + y) * a.sx +
+           a.ox + c.lox;
+  };
+  auto rowptr_b = [&](long row) {
+    int y = (int)(row % c.ny), z = (int)(row / c.ny);
+    return b.ptr + (b.oz + c.loz + c.sz + z) * b.sxy +
+           (b.oy + c.loy + c.sy + y) * b.sx + b.ox + c.lox + c.sx;
+  };
+  long row = blockIdx.x;
+  for (; row + 3L * gridDim.x < nrows; row += 4L * gridDim.x) {
+    const unsigned short *ap[4], *bp[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      ap[j] = rowptr_a(row + j * (long)gridDim.x);
+      bp[j] = rowptr_b(row + j * (long)gridDim.x);
+    }
+    for (int x = threadIdx.x; x < c.nx; x += 256) {
+      u64 av[4], bv[4];
+#pragma unroll
+      for (int j = 0; j < 4; ++j) { av[j] = ap[j][x]; bv[j] = bp[j][x]; }
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        pa += av[j]; pb += bv[j]; paa += av[j] * av[j];
+        pbb += bv[j] * bv[j]; pab += av[j] * bv[j];
+      }
+    }
+  }
+  for (; row < nrows; row += gridDim.x) {
+    const unsigned short *a0 = rowptr_a(row), *b0 = rowptr_b(row);
+    for (int x = threadIdx.x; x < c.nx; x += 256) {
+      u64 av = a0[x], bv = b0[x];
+      pa += av; pb += bv; paa += av * av; pbb += bv * bv; pab += av * bv;
+    }
+  }
+  pa = wave_sum_u64(pa); pb = wave_sum_u64(pb); paa = wave_sum_u64(paa);
+  pbb = wave_sum_u64(pbb); pab = wave_sum_u64(pab);
+  int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  if (lane == 0) {
+    ws[wave][0] = pa; ws[wave][1] = pb; ws[wave][2] = paa;
+    ws[wave][3] = pbb; ws[wave][4] = pab;
+  }
+  __syncthreads();
+  if (threadIdx.x < 5) {
+    u64 s = ws[0][threadIdx.x] + ws[1][threadIdx.x] + ws[2][threadIdx.x] +
+            ws[3][threadIdx.x];
+    atomicAdd(&sums[(long)blockIdx.y * 5 + threadIdx.x], s);
+  }
+}
+
+
 #define HIPCHK(x) if ((x) != hipSuccess) { printf("hiperr @%d\n", __LINE__); exit(1); }
 
 int main(int argc, char **argv) {
@@ -271,18 +331,19 @@ int main(int argc, char **argv) {
   hipEvent_t e0, e1;
   HIPCHK(hipEventCreate(&e0));
   HIPCHK(hipEventCreate(&e1));
-  const char *names[3] = {"prod 256-wide", "rw-adaptive x2", "rw-adaptive x4"};
-  std::vector<std::vector<float>> ms(3);
-  std::vector<std::vector<u64>> res(3);
+  const char *names[4] = {"prod 256-wide", "rw-adaptive x2", "rw-adaptive x4", "256-wide x4"};
+  std::vector<std::vector<float>> ms(4);
+  std::vector<std::vector<u64>> res(4);
   dim3 grid((unsigned)std::min(2048L, maxrows), (unsigned)hc.size());
   for (int r = 0; r < 7; ++r) {
-    for (int v = 0; v < 3; ++v) {
+    for (int v = 0; v < 4; ++v) {
       HIPCHK(hipMemset(ds, 0, hc.size() * 5 * sizeof(u64)));
       HIPCHK(hipEventRecord(e0, 0));
       switch (v) {
         case 0: hipLaunchKernelGGL(k_rtest0, grid, dim3(256), 0, 0, ra, rb, dc, ds); break;
         case 1: hipLaunchKernelGGL(k_rtest1, grid, dim3(256), 0, 0, ra, rb, dc, ds); break;
         case 2: hipLaunchKernelGGL(k_rtest2, grid, dim3(256), 0, 0, ra, rb, dc, ds); break;
+        case 3: hipLaunchKernelGGL(k_rtest3, grid, dim3(256), 0, 0, ra, rb, dc, ds); break;
       }
       HIPCHK(hipEventRecord(e1, 0));
       HIPCHK(hipEventSynchronize(e1));
@@ -296,8 +357,8 @@ int main(int argc, char **argv) {
       }
     }
   }
-  bool ok = res[0] == res[1] && res[0] == res[2];
-  for (int v = 0; v < 3; ++v) {
+  bool ok = res[0] == res[1] && res[0] == res[2] && res[0] == res[3];
+  for (int v = 0; v < 4; ++v) {
     std::sort(ms[v].begin(), ms[v].end());
     printf("%-16s med=%.3f ms  alg_GB/s=%.0f\n", names[v], ms[v][3],
            gb / (ms[v][3] * 1e-3));
