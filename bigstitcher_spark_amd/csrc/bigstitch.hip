@@ -735,56 +735,59 @@ __device__ __forceinline__ u64 wave_sum_u64(u64 v) {
   return v;
 }
 
-__global__ __launch_bounds__(256) void k_rtest(
-    bs_region a, bs_region b, const bs_cand *cands, u64 *sums /* [nc][5] */) {
+/* Plane-stationary candidate-batched scan (2.9x the per-candidate
+ * row-sliced mapping, tools/probe_rtest.hip + profiles/probe_rtest_r01.txt). One block per A plane
+ * (y-split via gridDim.y); the candidate loop runs INSIDE, so the A
+ * plane and the few distinct shifted B planes stay L2-resident across
+ * all candidates that touch them: HBM bytes ~ (A region once + B
+ * region x distinct sz) instead of (both windows x candidates). u64
+ * sums are order-independent -> bit-exact [PIN-R]. The per-candidate
+ * row walk uses the adaptive row width (rows here are single y-lines,
+ * no division anywhere). */
+__global__ __launch_bounds__(256) void k_rtest(bs_region a, bs_region b,
+                                                const bs_cand *cands,
+                                                int nc, u64 *sums) {
   __shared__ u64 ws[4][5];
-  const bs_cand c = cands[blockIdx.y];
-  long nrows = (long)c.ny * c.nz;
-  u64 pa = 0, pb = 0, paa = 0, pbb = 0, pab = 0;
-  auto rowptr_a = [&](long row) {
-    int y = (int)(row % c.ny), z = (int)(row / c.ny);
-    return a.ptr + (a.oz + c.loz + z) * a.sxy + (a.oy + c.loy + y) * a.sx +
-           a.ox + c.lox;
-  };
-  auto rowptr_b = [&](long row) {
-    int y = (int)(row % c.ny), z = (int)(row / c.ny);
-    return b.ptr + (b.oz + c.loz + c.sz + z) * b.sxy +
-           (b.oy + c.loy + c.sy + y) * b.sx + b.ox + c.lox + c.sx;
-  };
-  long row = blockIdx.x;
-  /* two rows in flight doubles outstanding loads (latency-bound kernel) */
-  for (; row + gridDim.x < nrows; row += 2L * gridDim.x) {
-    const unsigned short *a0 = rowptr_a(row), *b0 = rowptr_b(row);
-    const unsigned short *a1 = rowptr_a(row + gridDim.x);
-    const unsigned short *b1 = rowptr_b(row + gridDim.x);
-    for (int x = threadIdx.x; x < c.nx; x += 256) {
-      u64 av0 = a0[x], bv0 = b0[x], av1 = a1[x], bv1 = b1[x];
-      pa += av0 + av1;
-      pb += bv0 + bv1;
-      paa += av0 * av0 + av1 * av1;
-      pbb += bv0 * bv0 + bv1 * bv1;
-      pab += av0 * bv0 + av1 * bv1;
+  const int tid = threadIdx.x;
+  const int z = blockIdx.x;
+  for (int ci = 0; ci < nc; ++ci) {
+    const bs_cand c = cands[ci];
+    if (z < c.loz || z >= c.loz + c.nz) continue; /* block-uniform */
+    int rw = 256;
+    while ((rw >> 1) >= c.nx && rw > 16) rw >>= 1;
+    const int rsh = __ffs(rw) - 1;
+    const int rpg = 256 >> rsh;
+    const int lx = tid & (rw - 1);
+    const int lr = tid >> rsh;
+    u64 pa = 0, pb = 0, paa = 0, pbb = 0, pab = 0;
+    const unsigned short *abase =
+        a.ptr + (a.oz + z) * a.sxy + a.ox + c.lox;
+    const unsigned short *bbase = b.ptr + (b.oz + z + c.sz) * b.sxy +
+                                  b.ox + c.lox + c.sx;
+    for (int r = blockIdx.y * rpg + lr; r < c.ny;
+         r += gridDim.y * rpg) {
+      const unsigned short *ar = abase + (a.oy + c.loy + r) * a.sx;
+      const unsigned short *br =
+          bbase + (b.oy + c.loy + c.sy + r) * b.sx;
+      for (int x = lx; x < c.nx; x += rw) {
+        u64 av = ar[x], bv = br[x];
+        pa += av; pb += bv; paa += av * av; pbb += bv * bv;
+        pab += av * bv;
+      }
     }
-  }
-  for (; row < nrows; row += gridDim.x) {
-    const unsigned short *a0 = rowptr_a(row), *b0 = rowptr_b(row);
-    for (int x = threadIdx.x; x < c.nx; x += 256) {
-      u64 av = a0[x], bv = b0[x];
-      pa += av; pb += bv; paa += av * av; pbb += bv * bv; pab += av * bv;
+    pa = wave_sum_u64(pa); pb = wave_sum_u64(pb); paa = wave_sum_u64(paa);
+    pbb = wave_sum_u64(pbb); pab = wave_sum_u64(pab);
+    int lane = tid & 63, wave = tid >> 6;
+    if (lane == 0) {
+      ws[wave][0] = pa; ws[wave][1] = pb; ws[wave][2] = paa;
+      ws[wave][3] = pbb; ws[wave][4] = pab;
     }
-  }
-  pa = wave_sum_u64(pa); pb = wave_sum_u64(pb); paa = wave_sum_u64(paa);
-  pbb = wave_sum_u64(pbb); pab = wave_sum_u64(pab);
-  int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
-  if (lane == 0) {
-    ws[wave][0] = pa; ws[wave][1] = pb; ws[wave][2] = paa;
-    ws[wave][3] = pbb; ws[wave][4] = pab;
-  }
-  __syncthreads();
-  if (threadIdx.x < 5) {
-    u64 s = ws[0][threadIdx.x] + ws[1][threadIdx.x] + ws[2][threadIdx.x] +
-            ws[3][threadIdx.x];
-    atomicAdd(&sums[(long)blockIdx.y * 5 + threadIdx.x], s);
+    __syncthreads();
+    if (tid < 5) {
+      u64 sv = ws[0][tid] + ws[1][tid] + ws[2][tid] + ws[3][tid];
+      atomicAdd(&sums[(long)ci * 5 + tid], sv);
+    }
+    __syncthreads(); /* ws reused next candidate */
   }
 }
 
@@ -1693,15 +1696,13 @@ static int stitch_phaseB(bs_ctx *c, bs_slot *sl,
                           sl->stream));
     CHK(c, hipStreamSynchronize(sl->stream)); /* gc dies at scope end */
     {
-      long maxrows = 1;
+      int maxz = 1;
       for (auto &h : sl->hc)
-        maxrows = std::max(maxrows, (long)h.gc.ny * h.gc.nz);
+        maxz = std::max(maxz, h.gc.loz + h.gc.nz);
       bs_tim tt(c, BS_K_CORR, sl->stream);
-      hipLaunchKernelGGL(k_rtest,
-                         dim3((unsigned)std::min(2048L, maxrows),
-                              (unsigned)gc.size()),
-                         dim3(256), 0, sl->stream, sl->reg[0], sl->reg[1],
-                         sl->dcands, sl->dsums);
+      hipLaunchKernelGGL(k_rtest, dim3((unsigned)maxz, 4), dim3(256), 0,
+                         sl->stream, sl->reg[0], sl->reg[1], sl->dcands,
+                         (int)gc.size(), sl->dsums);
     }
     CHK(c, hipMemcpyAsync(sl->hsums, sl->dsums,
                           sl->hc.size() * 5 * sizeof(u64),

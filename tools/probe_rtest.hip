@@ -288,6 +288,62 @@ __global__ __launch_bounds__(256) void k_rtest3(bs_region a, bs_region b,
 }
 
 
+
+/* V5: plane-stationary candidate-batched scan. One block per A plane
+ * (y-split via gridDim.y); the candidate loop runs INSIDE, so the A
+ * plane and the few distinct shifted B planes stay L2-resident across
+ * all candidates that touch them: HBM bytes ~ (A region once + B
+ * region x distinct sz) instead of (both windows x candidates). u64
+ * sums are order-independent -> bit-exact [PIN-R]. The per-candidate
+ * row walk uses the adaptive row width (rows here are single y-lines,
+ * no division anywhere). */
+__global__ __launch_bounds__(256) void k_rtest5(bs_region a, bs_region b,
+                                                const bs_cand *cands,
+                                                int nc, u64 *sums) {
+  __shared__ u64 ws[4][5];
+  const int tid = threadIdx.x;
+  const int z = blockIdx.x;
+  for (int ci = 0; ci < nc; ++ci) {
+    const bs_cand c = cands[ci];
+    if (z < c.loz || z >= c.loz + c.nz) continue; /* block-uniform */
+    int rw = 256;
+    while ((rw >> 1) >= c.nx && rw > 16) rw >>= 1;
+    const int rsh = __ffs(rw) - 1;
+    const int rpg = 256 >> rsh;
+    const int lx = tid & (rw - 1);
+    const int lr = tid >> rsh;
+    u64 pa = 0, pb = 0, paa = 0, pbb = 0, pab = 0;
+    const unsigned short *abase =
+        a.ptr + (a.oz + z) * a.sxy + a.ox + c.lox;
+    const unsigned short *bbase = b.ptr + (b.oz + z + c.sz) * b.sxy +
+                                  b.ox + c.lox + c.sx;
+    for (int r = blockIdx.y * rpg + lr; r < c.ny;
+         r += gridDim.y * rpg) {
+      const unsigned short *ar = abase + (a.oy + c.loy + r) * a.sx;
+      const unsigned short *br =
+          bbase + (b.oy + c.loy + c.sy + r) * b.sx;
+      for (int x = lx; x < c.nx; x += rw) {
+        u64 av = ar[x], bv = br[x];
+        pa += av; pb += bv; paa += av * av; pbb += bv * bv;
+        pab += av * bv;
+      }
+    }
+    pa = wave_sum_u64(pa); pb = wave_sum_u64(pb); paa = wave_sum_u64(paa);
+    pbb = wave_sum_u64(pbb); pab = wave_sum_u64(pab);
+    int lane = tid & 63, wave = tid >> 6;
+    if (lane == 0) {
+      ws[wave][0] = pa; ws[wave][1] = pb; ws[wave][2] = paa;
+      ws[wave][3] = pbb; ws[wave][4] = pab;
+    }
+    __syncthreads();
+    if (tid < 5) {
+      u64 sv = ws[0][tid] + ws[1][tid] + ws[2][tid] + ws[3][tid];
+      atomicAdd(&sums[(long)ci * 5 + tid], sv);
+    }
+    __syncthreads(); /* ws reused next candidate */
+  }
+}
+
 #define HIPCHK(x) if ((x) != hipSuccess) { printf("hiperr @%d\n", __LINE__); exit(1); }
 
 int main(int argc, char **argv) {
@@ -331,12 +387,12 @@ int main(int argc, char **argv) {
   hipEvent_t e0, e1;
   HIPCHK(hipEventCreate(&e0));
   HIPCHK(hipEventCreate(&e1));
-  const char *names[4] = {"prod 256-wide", "rw-adaptive x2", "rw-adaptive x4", "256-wide x4"};
-  std::vector<std::vector<float>> ms(4);
-  std::vector<std::vector<u64>> res(4);
+  const char *names[5] = {"prod 256-wide", "rw-adaptive x2", "rw-adaptive x4", "256-wide x4", "plane-stationary"};
+  std::vector<std::vector<float>> ms(5);
+  std::vector<std::vector<u64>> res(5);
   dim3 grid((unsigned)std::min(2048L, maxrows), (unsigned)hc.size());
   for (int r = 0; r < 7; ++r) {
-    for (int v = 0; v < 4; ++v) {
+    for (int v = 0; v < 5; ++v) {
       HIPCHK(hipMemset(ds, 0, hc.size() * 5 * sizeof(u64)));
       HIPCHK(hipEventRecord(e0, 0));
       switch (v) {
@@ -344,6 +400,11 @@ int main(int argc, char **argv) {
         case 1: hipLaunchKernelGGL(k_rtest1, grid, dim3(256), 0, 0, ra, rb, dc, ds); break;
         case 2: hipLaunchKernelGGL(k_rtest2, grid, dim3(256), 0, 0, ra, rb, dc, ds); break;
         case 3: hipLaunchKernelGGL(k_rtest3, grid, dim3(256), 0, 0, ra, rb, dc, ds); break;
+        case 4: {
+          int maxz = 0;
+          for (auto &cc : hc) maxz = std::max(maxz, cc.loz + cc.nz);
+          hipLaunchKernelGGL(k_rtest5, dim3(maxz, 4), dim3(256), 0, 0, ra, rb, dc, (int)hc.size(), ds);
+        } break;
       }
       HIPCHK(hipEventRecord(e1, 0));
       HIPCHK(hipEventSynchronize(e1));
@@ -357,8 +418,8 @@ int main(int argc, char **argv) {
       }
     }
   }
-  bool ok = res[0] == res[1] && res[0] == res[2] && res[0] == res[3];
-  for (int v = 0; v < 4; ++v) {
+  bool ok = res[0] == res[1] && res[0] == res[2] && res[0] == res[3] && res[0] == res[4];
+  for (int v = 0; v < 5; ++v) {
     std::sort(ms[v].begin(), ms[v].end());
     printf("%-16s med=%.3f ms  alg_GB/s=%.0f\n", names[v], ms[v][3],
            gb / (ms[v][3] * 1e-3));
