@@ -42,6 +42,7 @@
 #include <thread>
 #include <string>
 #include <vector>
+#include <algorithm>
 
 #define LPB_X 4    /* lines per block, contiguous-line passes */
 #define TPL_X 64
@@ -764,15 +765,33 @@ __global__ __launch_bounds__(256) void k_rtest(bs_region a, bs_region b,
         a.ptr + (a.oz + z) * a.sxy + a.ox + c.lox;
     const unsigned short *bbase = b.ptr + (b.oz + z + c.sz) * b.sxy +
                                   b.ox + c.lox + c.sx;
-    for (int r = blockIdx.y * rpg + lr; r < c.ny;
-         r += gridDim.y * rpg) {
+    /* two row-groups in flight; u16^2 fits u32 exactly (promote each
+     * product to u64 BEFORE summing two of them) */
+    const int rstride = gridDim.y * rpg;
+    int r = blockIdx.y * rpg + lr;
+    for (; r + rstride < c.ny; r += 2 * rstride) {
+      const unsigned short *a0 = abase + (a.oy + c.loy + r) * a.sx;
+      const unsigned short *b0 =
+          bbase + (b.oy + c.loy + c.sy + r) * b.sx;
+      const unsigned short *a1 = a0 + (long)rstride * a.sx;
+      const unsigned short *b1 = b0 + (long)rstride * b.sx;
+      for (int x = lx; x < c.nx; x += rw) {
+        unsigned av0 = a0[x], bv0 = b0[x], av1 = a1[x], bv1 = b1[x];
+        pa += av0 + av1;
+        pb += bv0 + bv1;
+        paa += (u64)(av0 * av0) + (u64)(av1 * av1);
+        pbb += (u64)(bv0 * bv0) + (u64)(bv1 * bv1);
+        pab += (u64)(av0 * bv0) + (u64)(av1 * bv1);
+      }
+    }
+    for (; r < c.ny; r += rstride) {
       const unsigned short *ar = abase + (a.oy + c.loy + r) * a.sx;
       const unsigned short *br =
           bbase + (b.oy + c.loy + c.sy + r) * b.sx;
       for (int x = lx; x < c.nx; x += rw) {
-        u64 av = ar[x], bv = br[x];
-        pa += av; pb += bv; paa += av * av; pbb += bv * bv;
-        pab += av * bv;
+        unsigned av = ar[x], bv = br[x];
+        pa += av; pb += bv; paa += (u64)(av * av); pbb += (u64)(bv * bv);
+        pab += (u64)(av * bv);
       }
     }
     pa = wave_sum_u64(pa); pb = wave_sum_u64(pb); paa = wave_sum_u64(paa);
@@ -1688,6 +1707,14 @@ static int stitch_phaseB(bs_ctx *c, bs_slot *sl,
     }
   }
   if (!sl->hc.empty()) {
+    /* group candidates sharing a z-shift so the plane-stationary
+     * r-test's B-plane reads repeat consecutively (L2 hits); u64 sums
+     * are order-independent and hc/gc stay index-aligned, so results
+     * are unchanged */
+    std::stable_sort(sl->hc.begin(), sl->hc.end(),
+                     [](const bs_hostcand &x, const bs_hostcand &y) {
+                       return x.gc.sz < y.gc.sz;
+                     });
     std::vector<bs_cand> gc(sl->hc.size());
     for (size_t i = 0; i < sl->hc.size(); ++i) gc[i] = sl->hc[i].gc;
     CHK(c, hipMemcpyAsync(sl->dcands, gc.data(), gc.size() * sizeof(bs_cand),
@@ -1743,8 +1770,16 @@ static int stitch_phaseC(bs_ctx *c, bs_slot *sl,
     double db = (double)s[3] - sb * sb / n;
     if (da <= 0 || db <= 0) continue;
     double r = num / std::sqrt(da * db);
-    /* iteration is (rank asc, ci asc): first max == oracle's key order */
-    if (best_i < 0 || r > best_r) {
+    /* explicit (r desc, rank asc, ci asc) key == the oracle's order;
+     * candidates arrive sz-sorted (L2 grouping), so the tie-break
+     * cannot rely on iteration order */
+    bool better =
+        best_i < 0 || r > best_r ||
+        (r == best_r &&
+         (sl->hc[i].rank < sl->hc[best_i].rank ||
+          (sl->hc[i].rank == sl->hc[best_i].rank &&
+           sl->hc[i].ci < sl->hc[best_i].ci)));
+    if (better) {
       best_r = r;
       best_i = (int)i;
     }

@@ -344,6 +344,134 @@ __global__ __launch_bounds__(256) void k_rtest5(bs_region a, bs_region b,
   }
 }
 
+/* V5: plane-stationary candidate-batched scan. One block per A plane
+ * (y-split via gridDim.y); the candidate loop runs INSIDE, so the A
+ * plane and the few distinct shifted B planes stay L2-resident across
+ * all candidates that touch them: HBM bytes ~ (A region once + B
+ * region x distinct sz) instead of (both windows x candidates). u64
+ * sums are order-independent -> bit-exact [PIN-R]. The per-candidate
+ * row walk uses the adaptive row width (rows here are single y-lines,
+ * no division anywhere). */
+__global__ __launch_bounds__(256) void k_rtest6(bs_region a, bs_region b,
+                                                const bs_cand *cands,
+                                                int nc, u64 *sums) {
+  __shared__ u64 ws[4][5];
+  const int tid = threadIdx.x;
+  const int z = blockIdx.x;
+  for (int ci = 0; ci < nc; ++ci) {
+    const bs_cand c = cands[ci];
+    if (z < c.loz || z >= c.loz + c.nz) continue; /* block-uniform */
+    int rw = 256;
+    while ((rw >> 1) >= c.nx && rw > 16) rw >>= 1;
+    const int rsh = __ffs(rw) - 1;
+    const int rpg = 256 >> rsh;
+    const int lx = tid & (rw - 1);
+    const int lr = tid >> rsh;
+    u64 pa = 0, pb = 0, paa = 0, pbb = 0, pab = 0;
+    const unsigned short *abase =
+        a.ptr + (a.oz + z) * a.sxy + a.ox + c.lox;
+    const unsigned short *bbase = b.ptr + (b.oz + z + c.sz) * b.sxy +
+                                  b.ox + c.lox + c.sx;
+    for (int r = blockIdx.y * rpg + lr; r < c.ny;
+         r += gridDim.y * rpg) {
+      const unsigned short *ar = abase + (a.oy + c.loy + r) * a.sx;
+      const unsigned short *br =
+          bbase + (b.oy + c.loy + c.sy + r) * b.sx;
+      for (int x = lx; x < c.nx; x += rw) {
+        unsigned av = ar[x], bv = br[x]; /* u16^2 fits u32 exactly */
+        pa += av; pb += bv; paa += av * av; pbb += bv * bv;
+        pab += av * bv;
+      }
+    }
+    pa = wave_sum_u64(pa); pb = wave_sum_u64(pb); paa = wave_sum_u64(paa);
+    pbb = wave_sum_u64(pbb); pab = wave_sum_u64(pab);
+    int lane = tid & 63, wave = tid >> 6;
+    if (lane == 0) {
+      ws[wave][0] = pa; ws[wave][1] = pb; ws[wave][2] = paa;
+      ws[wave][3] = pbb; ws[wave][4] = pab;
+    }
+    __syncthreads();
+    if (tid < 5) {
+      u64 sv = ws[0][tid] + ws[1][tid] + ws[2][tid] + ws[3][tid];
+      atomicAdd(&sums[(long)ci * 5 + tid], sv);
+    }
+    __syncthreads(); /* ws reused next candidate */
+  }
+}
+
+
+/* V5: plane-stationary candidate-batched scan. One block per A plane
+ * (y-split via gridDim.y); the candidate loop runs INSIDE, so the A
+ * plane and the few distinct shifted B planes stay L2-resident across
+ * all candidates that touch them: HBM bytes ~ (A region once + B
+ * region x distinct sz) instead of (both windows x candidates). u64
+ * sums are order-independent -> bit-exact [PIN-R]. The per-candidate
+ * row walk uses the adaptive row width (rows here are single y-lines,
+ * no division anywhere). */
+__global__ __launch_bounds__(256) void k_rtest7(bs_region a, bs_region b,
+                                                const bs_cand *cands,
+                                                int nc, u64 *sums) {
+  __shared__ u64 ws[4][5];
+  const int tid = threadIdx.x;
+  const int z = blockIdx.x;
+  for (int ci = 0; ci < nc; ++ci) {
+    const bs_cand c = cands[ci];
+    if (z < c.loz || z >= c.loz + c.nz) continue; /* block-uniform */
+    int rw = 256;
+    while ((rw >> 1) >= c.nx && rw > 16) rw >>= 1;
+    const int rsh = __ffs(rw) - 1;
+    const int rpg = 256 >> rsh;
+    const int lx = tid & (rw - 1);
+    const int lr = tid >> rsh;
+    u64 pa = 0, pb = 0, paa = 0, pbb = 0, pab = 0;
+    const unsigned short *abase =
+        a.ptr + (a.oz + z) * a.sxy + a.ox + c.lox;
+    const unsigned short *bbase = b.ptr + (b.oz + z + c.sz) * b.sxy +
+                                  b.ox + c.lox + c.sx;
+    const int rstride = gridDim.y * rpg;
+    int r = blockIdx.y * rpg + lr;
+    for (; r + rstride < c.ny; r += 2 * rstride) {
+      const unsigned short *a0 = abase + (a.oy + c.loy + r) * a.sx;
+      const unsigned short *b0 =
+          bbase + (b.oy + c.loy + c.sy + r) * b.sx;
+      const unsigned short *a1 = a0 + (long)rstride * a.sx;
+      const unsigned short *b1 = b0 + (long)rstride * b.sx;
+      for (int x = lx; x < c.nx; x += rw) {
+        unsigned av0 = a0[x], bv0 = b0[x], av1 = a1[x], bv1 = b1[x];
+        pa += av0 + av1; pb += bv0 + bv1; /* sums of two u16 fit u32 */
+        paa += (u64)(av0 * av0) + (u64)(av1 * av1); /* each u16^2 fits
+            u32 exactly; promote BEFORE adding two products */
+        pbb += (u64)(bv0 * bv0) + (u64)(bv1 * bv1);
+        pab += (u64)(av0 * bv0) + (u64)(av1 * bv1);
+      }
+    }
+    for (; r < c.ny; r += rstride) {
+      const unsigned short *ar = abase + (a.oy + c.loy + r) * a.sx;
+      const unsigned short *br =
+          bbase + (b.oy + c.loy + c.sy + r) * b.sx;
+      for (int x = lx; x < c.nx; x += rw) {
+        unsigned av = ar[x], bv = br[x];
+        pa += av; pb += bv; paa += av * av; pbb += bv * bv;
+        pab += av * bv;
+      }
+    }
+    pa = wave_sum_u64(pa); pb = wave_sum_u64(pb); paa = wave_sum_u64(paa);
+    pbb = wave_sum_u64(pbb); pab = wave_sum_u64(pab);
+    int lane = tid & 63, wave = tid >> 6;
+    if (lane == 0) {
+      ws[wave][0] = pa; ws[wave][1] = pb; ws[wave][2] = paa;
+      ws[wave][3] = pbb; ws[wave][4] = pab;
+    }
+    __syncthreads();
+    if (tid < 5) {
+      u64 sv = ws[0][tid] + ws[1][tid] + ws[2][tid] + ws[3][tid];
+      atomicAdd(&sums[(long)ci * 5 + tid], sv);
+    }
+    __syncthreads(); /* ws reused next candidate */
+  }
+}
+
+
 #define HIPCHK(x) if ((x) != hipSuccess) { printf("hiperr @%d\n", __LINE__); exit(1); }
 
 int main(int argc, char **argv) {
@@ -387,12 +515,12 @@ int main(int argc, char **argv) {
   hipEvent_t e0, e1;
   HIPCHK(hipEventCreate(&e0));
   HIPCHK(hipEventCreate(&e1));
-  const char *names[5] = {"prod 256-wide", "rw-adaptive x2", "rw-adaptive x4", "256-wide x4", "plane-stationary"};
-  std::vector<std::vector<float>> ms(5);
-  std::vector<std::vector<u64>> res(5);
+  const char *names[7] = {"prod 256-wide", "rw-adaptive x2", "rw-adaptive x4", "256-wide x4", "plane-stationary", "plane-u32", "plane-u32-2deep"};
+  std::vector<std::vector<float>> ms(7);
+  std::vector<std::vector<u64>> res(7);
   dim3 grid((unsigned)std::min(2048L, maxrows), (unsigned)hc.size());
   for (int r = 0; r < 7; ++r) {
-    for (int v = 0; v < 5; ++v) {
+    for (int v = 0; v < 7; ++v) {
       HIPCHK(hipMemset(ds, 0, hc.size() * 5 * sizeof(u64)));
       HIPCHK(hipEventRecord(e0, 0));
       switch (v) {
@@ -404,6 +532,12 @@ int main(int argc, char **argv) {
           int maxz = 0;
           for (auto &cc : hc) maxz = std::max(maxz, cc.loz + cc.nz);
           hipLaunchKernelGGL(k_rtest5, dim3(maxz, 4), dim3(256), 0, 0, ra, rb, dc, (int)hc.size(), ds);
+        } break;
+        case 5: case 6: {
+          int maxz = 0;
+          for (auto &cc : hc) maxz = std::max(maxz, cc.loz + cc.nz);
+          if (v == 5) hipLaunchKernelGGL(k_rtest6, dim3(maxz, 4), dim3(256), 0, 0, ra, rb, dc, (int)hc.size(), ds);
+          else hipLaunchKernelGGL(k_rtest7, dim3(maxz, 4), dim3(256), 0, 0, ra, rb, dc, (int)hc.size(), ds);
         } break;
       }
       HIPCHK(hipEventRecord(e1, 0));
@@ -418,8 +552,8 @@ int main(int argc, char **argv) {
       }
     }
   }
-  bool ok = res[0] == res[1] && res[0] == res[2] && res[0] == res[3] && res[0] == res[4];
-  for (int v = 0; v < 5; ++v) {
+  bool ok = true; for (int v = 1; v < 7; ++v) ok = ok && res[0] == res[v];
+  for (int v = 0; v < 7; ++v) {
     std::sort(ms[v].begin(), ms[v].end());
     printf("%-16s med=%.3f ms  alg_GB/s=%.0f\n", names[v], ms[v][3],
            gb / (ms[v][3] * 1e-3));
