@@ -585,6 +585,115 @@ __global__ __launch_bounds__(512) void k_peak_stream512(const float *pcm,
 
 
 
+/* VAR 10/11: f4 with TY and chunk length parameterized. */
+template <int TY, int CZX>
+__global__ __launch_bounds__(256) void k_peak_streamc(const float *pcm,
+                                                      int px, int py, int pz,
+                                                      bs_peak *wgbuf) {
+  const int TX = 128, CZ = CZX;
+  const int HX = TX + 2, HY = TY + 2;
+  __shared__ float pl[4][HY * HX];
+  __shared__ float wv[4][5];
+  __shared__ long long wi[4][5];
+  const int tid = threadIdx.x;
+  const int ntx = (px + TX - 1) / TX;
+  const int nty = (py + TY - 1) / TY;
+  const int ncz = (pz + CZ - 1) / CZ;
+  const long nchunks = (long)ntx * nty * ncz;
+  float tv[5];
+  long long ti[5];
+  for (int k = 0; k < 5; ++k) { tv[k] = -3.0e38f; ti[k] = 0x7fffffffffffffffLL; }
+  /* row-task decomposition: 34 tasks per row (32 float4 + 2 edges) */
+  auto load_plane = [&](int gz, float *dst, int x0, int y0) {
+    const float *src = pcm + (long)gz * py * px;
+    for (int i = tid; i < HY * 34; i += 256) {
+      int t = i % 34, ly = i / 34;
+      int gy = y0 + ly - 1;
+      gy += gy < 0 ? py : 0;
+      gy -= gy >= py ? py : 0;
+      const float *row = src + (long)gy * px;
+      if (t < 32) {
+        int gx = x0 + 4 * t; /* aligned: x0 mult of 128, px pow2 */
+        float4 v;
+        if (gx + 3 < px) {
+          v = *(const float4 *)(row + gx);
+        } else { /* last x-strip of a non-multiple px: scalar wrap */
+          float tmp[4];
+          for (int q = 0; q < 4; ++q) {
+            int xx = gx + q;
+            xx -= xx >= px ? px : 0;
+            tmp[q] = row[xx];
+          }
+          v = {tmp[0], tmp[1], tmp[2], tmp[3]};
+        }
+        float *d = dst + ly * HX + 1 + 4 * t;
+        d[0] = v.x; d[1] = v.y; d[2] = v.z; d[3] = v.w;
+      } else {
+        int gx = (t == 32) ? x0 - 1 : x0 + TX;
+        gx += gx < 0 ? px : 0;
+        gx -= gx >= px ? px : 0;
+        dst[ly * HX + (t == 32 ? 0 : HX - 1)] = row[gx];
+      }
+    }
+  };
+  for (long t0 = blockIdx.x; t0 < nchunks; t0 += gridDim.x) {
+    const int bx = (int)(t0 % ntx);
+    const int by = (int)((t0 / ntx) % nty);
+    const int bz = (int)(t0 / ((long)ntx * nty));
+    const int x0 = bx * TX, y0 = by * TY, z0 = bz * CZ;
+    const int zend = min(z0 + CZ, pz);
+    for (int p = -1; p <= 1; ++p) {
+      int gz = z0 + p;
+      gz += gz < 0 ? pz : 0;
+      gz -= gz >= pz ? pz : 0;
+      load_plane(gz, pl[p + 1], x0, y0);
+    }
+    __syncthreads();
+    for (int z = z0; z < zend; ++z) {
+      {
+        int gz = z + 2;
+        gz -= gz >= pz ? pz : 0;
+        gz -= gz >= pz ? pz : 0;
+        load_plane(gz, pl[(z - z0 + 3) & 3], x0, y0);
+      }
+      const float *pm = pl[(z - z0) & 3];
+      const float *pc = pl[(z - z0 + 1) & 3];
+      const float *pp = pl[(z - z0 + 2) & 3];
+      for (int i = tid; i < TX * TY; i += 256) {
+        int lx = i % TX, ly = i / TX;
+        int gx = x0 + lx, gy = y0 + ly;
+        const int base = (ly + 1) * HX + lx + 1;
+        float v = pc[base];
+        float m = fmaxf(pc[base - 1], pc[base + 1]);
+#pragma unroll
+        for (int dy = -1; dy <= 1; ++dy) {
+          const int b2 = base + dy * HX;
+          m = fmaxf(m, fmaxf(fmaxf(pm[b2 - 1], pm[b2]), pm[b2 + 1]));
+          m = fmaxf(m, fmaxf(fmaxf(pp[b2 - 1], pp[b2]), pp[b2 + 1]));
+          if (dy != 0)
+            m = fmaxf(m, fmaxf(fmaxf(pc[b2 - 1], pc[b2]), pc[b2 + 1]));
+        }
+        if (gx < px && gy < py && v > m)
+          pk_insert(tv, ti, v, ((long long)z * py + gy) * px + gx);
+      }
+      __syncthreads();
+    }
+  }
+  pk_merge_shfl(tv, ti);
+  int lane = tid & 63, wave = tid >> 6;
+  if (lane == 0)
+    for (int k = 0; k < 5; ++k) { wv[wave][k] = tv[k]; wi[wave][k] = ti[k]; }
+  __syncthreads();
+  if (tid == 0) {
+    for (int w = 1; w < 4; ++w)
+      for (int k = 0; k < 5; ++k) pk_insert(tv, ti, wv[w][k], wi[w][k]);
+    bs_peak *o = wgbuf + (long)blockIdx.x * 5;
+    for (int k = 0; k < 5; ++k) o[k] = {tv[k], 0, ti[k]};
+  }
+}
+
+
+
 #define HIPCHK(x) if ((x) != hipSuccess) { printf("hiperr @%d\n", __LINE__); exit(1); }
 
 int main() {
@@ -603,12 +712,12 @@ int main() {
   hipEvent_t e0, e1;
   HIPCHK(hipEventCreate(&e0));
   HIPCHK(hipEventCreate(&e1));
-  const char *names[9] = {"prod (LDS tile)", "load-only", "maxima-only",
-                          "global-direct", "stream-z", "stream-f4-ty8", "stream-f4-ty16", "stream-sep", "f4-ty16-512t"};
+  const char *names[11] = {"prod (LDS tile)", "load-only", "maxima-only",
+                          "global-direct", "stream-z", "stream-f4-ty8", "stream-f4-ty16", "stream-sep", "f4-ty16-512t", "f4-ty12", "f4-ty16-cz128"};
   const double bytes = n * 4.0;
-  std::vector<std::vector<float>> ms(9);
+  std::vector<std::vector<float>> ms(11);
   for (int r = 0; r < 7; ++r) {
-    for (int v = 0; v < 9; ++v) {
+    for (int v = 0; v < 11; ++v) {
       HIPCHK(hipEventRecord(e0, 0));
       switch (v) {
         case 0: hipLaunchKernelGGL(k_peak<0>, dim3(2048), dim3(256), 0, 0, d, px, py, pz, wb); break;
@@ -620,6 +729,8 @@ int main() {
         case 6: hipLaunchKernelGGL(k_peak_stream4<16>, dim3(2048), dim3(256), 0, 0, d, px, py, pz, wb); break;
         case 7: hipLaunchKernelGGL(k_peak_sep, dim3(2048), dim3(256), (4*(18*130)+4*(18*128))*4, 0, d, px, py, pz, wb); break;
         case 8: hipLaunchKernelGGL(k_peak_stream512<16>, dim3(1024), dim3(512), 0, 0, d, px, py, pz, wb); break;
+        case 9: hipLaunchKernelGGL((k_peak_streamc<12, 64>), dim3(2048), dim3(256), 0, 0, d, px, py, pz, wb); break;
+        case 10: hipLaunchKernelGGL((k_peak_streamc<16, 128>), dim3(2048), dim3(256), 0, 0, d, px, py, pz, wb); break;
       }
       HIPCHK(hipEventRecord(e1, 0));
       HIPCHK(hipEventSynchronize(e1));
@@ -628,7 +739,7 @@ int main() {
       ms[v].push_back(m);
     }
   }
-  for (int v = 0; v < 9; ++v) {
+  for (int v = 0; v < 11; ++v) {
     std::sort(ms[v].begin(), ms[v].end());
     printf("%-16s med=%.3f ms  alg_GB/s=%.0f\n", names[v], ms[v][3],
            bytes / (ms[v][3] * 1e-3) / 1e9);
