@@ -44,6 +44,10 @@ def full_pair(ctx, a, b, ida=0, idb=1):
         ((48, 96, 80), (-6.5, 2.25, 0.75), (1, 1, 1)),
         ((128, 128, 128), (7.3, -4.8, 3.1), (2, 2, 1)),
         ((32, 48, 64), (0.0, 0.0, 0.0), (1, 1, 1)),
+        # pads one axis to 1024: pins the E=8 wave-resident x pass
+        # (and the generic path on the other axes)
+        ((600, 64, 48), (3.5, -2.25, 1.5), (1, 1, 1)),
+        ((48, 64, 600), (1.5, -2.25, 3.5), (1, 1, 1)),
     ],
 )
 def test_stitch_parity(ctx, shape, shift, ds):
