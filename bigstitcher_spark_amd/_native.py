@@ -57,6 +57,7 @@ class _FuseParams(C.Structure):
         ("fusion_type", C.c_int32), ("out_dtype", C.c_int32),
         ("min_intensity", C.c_double), ("max_intensity", C.c_double),
         ("interp", C.c_int32),
+        ("masks", C.c_int32), ("mask_offset", C.c_double * 3),
     ]
 
 
@@ -252,7 +253,8 @@ class Context:
     # -- fusion -----------------------------------------------------------
     def fuse_blocks(self, views, blocks, view_idx_per_block,
                     fusion_type=FUSION_AVG_BLEND, out_dtype=np.float32,
-                    min_intensity=0.0, max_intensity=65535.0):
+                    min_intensity=0.0, max_intensity=65535.0,
+                    masks=False, mask_offset=(0.0, 0.0, 0.0)):
         """views: list of dicts {view_id, affine (3,4), border, range};
         blocks: list of (min_xyz, size_xyz); view_idx_per_block: list of
         index lists into views. Returns list of (nz,ny,nx) arrays."""
@@ -284,7 +286,8 @@ class Context:
         cidx = (C.c_int32 * max(1, len(flat)))(*flat) if flat else \
             (C.c_int32 * 1)(0)
         prm = _FuseParams(fusion_type, OUT_DTYPES[dt], min_intensity,
-                          max_intensity, 1)
+                          max_intensity, 1, 1 if masks else 0,
+                          (C.c_double * 3)(*mask_offset))
         self._check(
             self._lib.bs_fuse_blocks(self._h, cv, nv, cb, nb, cidx, offs,
                                      C.byref(prm), outptrs),
@@ -294,7 +297,8 @@ class Context:
 
     def fuse_volume(self, views, vol_min, vol_dims, downsamplings=None,
                     fusion_type=FUSION_AVG_BLEND, out_dtype=np.float32,
-                    min_intensity=0.0, max_intensity=65535.0):
+                    min_intensity=0.0, max_intensity=65535.0,
+                    masks=False, mask_offset=(0.0, 0.0, 0.0)):
         """Whole-volume fusion + pyramid. downsamplings: list of (dx,dy,dz)
         absolute factors per level (level 0 must be (1,1,1)). Returns a
         list of (nz,ny,nx) arrays, one per level."""
@@ -312,7 +316,8 @@ class Context:
                 cv[i].blend_range[d] = v.get("range", (40, 40, 40))[d]
         dt = np.dtype(out_dtype)
         prm = _FuseParams(fusion_type, OUT_DTYPES[dt], min_intensity,
-                          max_intensity, 1)
+                          max_intensity, 1, 1 if masks else 0,
+                          (C.c_double * 3)(*mask_offset))
         vmin = (C.c_int64 * 3)(*[int(x) for x in vol_min])
         vdim = (C.c_int64 * 3)(*[int(x) for x in vol_dims])
         cds = (C.c_int32 * (3 * nl))(

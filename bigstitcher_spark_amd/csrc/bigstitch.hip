@@ -973,6 +973,52 @@ __global__ __launch_bounds__(256) void k_fuse(
 
 /* ---- multi-resolution pyramid: box-mean downsample ---- */
 
+/* Coverage-mask mode of the fusion stage [--masks]: per output voxel,
+ * type-max where ANY view's inverse affine lands inside
+ * [-moff, dim-1+moff] per axis (inclusive), else 0 — no interpolation,
+ * no intensity scaling. Restates
+ * fusion/GenerateComputeBlockMasks.java:85-151 (bounds: dim.min -
+ * maskOffset .. dim.max + maskOffset; uint8 255 / uint16 65535 /
+ * float 1.0 at :152-176). */
+__global__ __launch_bounds__(256) void k_mask(
+    const bs_dev_view *views, const int *vidx, int nv, long bmx, long bmy,
+    long bmz, int bx, int by, int bz, int dtype, float mofx, float mofy,
+    float mofz, void *out, long out_off, long out_row, long out_slice) {
+  __shared__ bs_dev_view sv[BS_MAX_BLK_VIEWS];
+  int nvs = min(nv, BS_MAX_BLK_VIEWS);
+  for (int i = threadIdx.x; i < nvs * (int)(sizeof(bs_dev_view) / 4);
+       i += 256) {
+    ((int *)sv)[i] = ((const int *)&views[vidx[i / (sizeof(bs_dev_view) / 4)]])
+        [i % (sizeof(bs_dev_view) / 4)];
+  }
+  __syncthreads();
+  long nrows = (long)by * bz;
+  for (long row = blockIdx.x; row < nrows; row += gridDim.x) {
+    int y = (int)(row % by), z = (int)(row / by);
+    for (int x = threadIdx.x; x < bx; x += blockDim.x) {
+      long i = out_off + (long)z * out_slice + (long)y * out_row + x;
+      float wx = (float)(bmx + x), wy = (float)(bmy + y),
+            wz = (float)(bmz + z);
+      bool any = false;
+      for (int k = 0; k < nv && !any; ++k) {
+        const bs_dev_view &v = k < nvs ? sv[k] : views[vidx[k]];
+        float px = v.inv[0] * wx + v.inv[1] * wy + v.inv[2] * wz + v.inv[3];
+        float py = v.inv[4] * wx + v.inv[5] * wy + v.inv[6] * wz + v.inv[7];
+        float pz = v.inv[8] * wx + v.inv[9] * wy + v.inv[10] * wz + v.inv[11];
+        any = px >= -mofx && px <= (float)(v.nx - 1) + mofx &&
+              py >= -mofy && py <= (float)(v.ny - 1) + mofy &&
+              pz >= -mofz && pz <= (float)(v.nz - 1) + mofz;
+      }
+      if (dtype == 2) /* BS_OUT_UINT8 */
+        ((unsigned char *)out)[i] = any ? 255 : 0;
+      else if (dtype == 1) /* BS_OUT_UINT16 */
+        ((unsigned short *)out)[i] = any ? 65535 : 0;
+      else /* BS_OUT_FLOAT32 */
+        ((float *)out)[i] = any ? 1.0f : 0.0f;
+    }
+  }
+}
+
 template <typename T>
 __global__ __launch_bounds__(256) void k_pyr(
     const T *src, T *dst, int sxd, int syd, int szd, int dxd, int dyd,
@@ -2004,14 +2050,26 @@ extern "C" int bs_fuse_blocks(bs_ctx *c, const bs_fuse_view *views,
     {
       long nrows_f = (long)bd.size[1] * bd.size[2];
       bs_tim tt(c, BS_K_FUSE);
-      hipLaunchKernelGGL(k_fuse, dim3(std::min(4096L, nrows_f)),
-                         dim3(256), 0, c->stream, c->dviews,
-                         c->dvidx + view_idx_offsets[i], nvb, bd.min[0],
-                         bd.min[1], bd.min[2], (int)bd.size[0],
-                         (int)bd.size[1], (int)bd.size[2], prm->fusion_type,
-                         prm->out_dtype, (float)prm->min_intensity, invRange,
-                         c->fuse_out, 0L, (long)bd.size[0],
-                         (long)bd.size[0] * bd.size[1]);
+      if (prm->masks)
+        hipLaunchKernelGGL(k_mask, dim3(std::min(4096L, nrows_f)),
+                           dim3(256), 0, c->stream, c->dviews,
+                           c->dvidx + view_idx_offsets[i], nvb, bd.min[0],
+                           bd.min[1], bd.min[2], (int)bd.size[0],
+                           (int)bd.size[1], (int)bd.size[2], prm->out_dtype,
+                           (float)prm->mask_offset[0],
+                           (float)prm->mask_offset[1],
+                           (float)prm->mask_offset[2], c->fuse_out, 0L,
+                           (long)bd.size[0], (long)bd.size[0] * bd.size[1]);
+      else
+        hipLaunchKernelGGL(k_fuse, dim3(std::min(4096L, nrows_f)),
+                           dim3(256), 0, c->stream, c->dviews,
+                           c->dvidx + view_idx_offsets[i], nvb, bd.min[0],
+                           bd.min[1], bd.min[2], (int)bd.size[0],
+                           (int)bd.size[1], (int)bd.size[2],
+                           prm->fusion_type, prm->out_dtype,
+                           (float)prm->min_intensity, invRange, c->fuse_out,
+                           0L, (long)bd.size[0],
+                           (long)bd.size[0] * bd.size[1]);
     }
     CHK(c, hipMemcpyAsync(out_blocks[i], c->fuse_out, nvox * esz,
                           hipMemcpyDeviceToHost, c->stream));
@@ -2181,13 +2239,24 @@ extern "C" int bs_fuse_volume(bs_ctx *c, const bs_fuse_view *views,
     int nvb = (int)(offs[b + 1] - offs[b]);
     long nrows_f = fb[4] * fb[5];
     bs_tim tt(c, BS_K_FUSE, c->stream);
-    hipLaunchKernelGGL(k_fuse, dim3(std::min(4096L, nrows_f)), dim3(256),
-                       0, c->stream, c->dviews, c->dvidx + offs[b], nvb,
-                       fb[0], fb[1], fb[2], (int)fb[3], (int)fb[4],
-                       (int)fb[5], prm->fusion_type, prm->out_dtype,
-                       (float)prm->min_intensity, invRange, dlvl[0],
-                       (long)fb[2] * vslice + fb[1] * vrow + fb[0], vrow,
-                       vslice);
+    if (prm->masks)
+      hipLaunchKernelGGL(k_mask, dim3(std::min(4096L, nrows_f)), dim3(256),
+                         0, c->stream, c->dviews, c->dvidx + offs[b], nvb,
+                         fb[0], fb[1], fb[2], (int)fb[3], (int)fb[4],
+                         (int)fb[5], prm->out_dtype,
+                         (float)prm->mask_offset[0],
+                         (float)prm->mask_offset[1],
+                         (float)prm->mask_offset[2], dlvl[0],
+                         (long)fb[2] * vslice + fb[1] * vrow + fb[0], vrow,
+                         vslice);
+    else
+      hipLaunchKernelGGL(k_fuse, dim3(std::min(4096L, nrows_f)), dim3(256),
+                         0, c->stream, c->dviews, c->dvidx + offs[b], nvb,
+                         fb[0], fb[1], fb[2], (int)fb[3], (int)fb[4],
+                         (int)fb[5], prm->fusion_type, prm->out_dtype,
+                         (float)prm->min_intensity, invRange, dlvl[0],
+                         (long)fb[2] * vslice + fb[1] * vrow + fb[0], vrow,
+                         vslice);
   }
   /* pyramid levels */
   for (int l = 1; l < nlevels; ++l) {

@@ -66,6 +66,20 @@ struct Args {
   }
 };
 
+inline std::vector<double> parse_floats(const std::string &s) {
+  std::vector<double> out;
+  std::string cur;
+  for (char c : s + ",") {
+    if (c == ',' || c == ' ') {
+      if (!cur.empty()) out.push_back(atof(cur.c_str()));
+      cur.clear();
+    } else {
+      cur += c;
+    }
+  }
+  return out;
+}
+
 inline std::vector<long long> parse_ints(const std::string &s) {
   std::vector<long long> out;
   std::string cur;

@@ -6,7 +6,9 @@
  * (bs_fuse_blocks = the BlkAffineFusion replacement, :602-615) and
  * writes the voxel blocks into the container (N5Utils.saveBlock, :670).
  * Round-1 scope: plain-N5 containers, level s0, FusionTypes
- * AVG/AVG_BLEND/MAX_INTENSITY, no intensity coefficients / masks. */
+ * AVG/AVG_BLEND/MAX_INTENSITY; --masks writes coverage masks instead of
+ * fused intensities (reference SparkAffineFusion.java:112-115, :565-578
+ * via fusion/GenerateComputeBlockMasks.java). */
 #include <cstdio>
 #include <cstring>
 #include <set>
@@ -26,7 +28,7 @@ int main(int argc, char **argv) {
             "usage: affine-fusion -o out.n5 [-x dataset.xml] "
             "[--fusionType AVG_BLEND|AVG|MAX_INTENSITY] "
             "[--blendingRange 40] [--blendingBorder 0] [--device N] "
-            "[--batchSize 64]\n");
+            "[--batchSize 64] [--masks] [--maskOffset 0.0,0.0,0.0]\n");
     return 2;
   }
   bsn5::Container n5(args.get("n5Path"));
@@ -158,6 +160,12 @@ int main(int argc, char **argv) {
     prm.min_intensity = minI;
     prm.max_intensity = maxI;
     prm.interp = 1;
+    prm.masks = args.has("masks") ? 1 : 0;
+    {
+      auto mo = bscli::parse_floats(args.get("maskOffset", "0,0,0"));
+      for (int d = 0; d < 3 && d < (int)mo.size(); ++d)
+        prm.mask_offset[d] = mo[d];
+    }
     size_t esz2 = out_dtype == BS_OUT_FLOAT32 ? 4
                   : out_dtype == BS_OUT_UINT16 ? 2 : 1;
     std::vector<std::vector<char>> hostlvl(nlevels);

@@ -174,6 +174,15 @@ typedef struct {
                           497-517) */
   double min_intensity, max_intensity;
   int32_t interp; /* 1 = trilinear; only value supported (reference :611) */
+  int32_t masks;  /* 1 = write coverage masks instead of fused intensities
+                     (--masks; reference SparkAffineFusion.java:112-115,
+                     :565-578 + fusion/GenerateComputeBlockMasks.java:
+                     85-151): out = type max (uint8 255 / uint16 65535 /
+                     float32 1.0) where ANY listed view's inverse affine
+                     maps the voxel into [-mask_offset, dim-1+mask_offset]
+                     per axis (inclusive), else 0. min/max_intensity
+                     scaling does not apply. */
+  double mask_offset[3]; /* --maskOffset, raw input px (default 0,0,0) */
 } bs_fuse_params;
 
 /* Fuse nb output blocks. views[] lists the views overlapping ANY of the

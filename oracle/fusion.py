@@ -45,6 +45,7 @@ FUSION_CLOSEST_PIXEL = 5   # largest min-axis border distance wins [PIN]
 
 __all__ = [
     "fuse_block",
+    "mask_block",
     "blend_weight",
     "downsample_level",
     "FUSION_AVG",
@@ -217,6 +218,46 @@ def fuse_block(
     scaled = np.where(covered, scaled, 0.0)
     q = np.floor(np.abs(scaled) + 0.5) * np.sign(scaled)
     return np.clip(q, 0, tmax).astype(out_dtype)
+
+
+def mask_block(
+    views,
+    block_min,
+    block_size,
+    mask_offset=(0.0, 0.0, 0.0),
+    out_dtype=np.uint8,
+):
+    """Coverage masks instead of fused intensities (--masks). Restates
+    fusion/GenerateComputeBlockMasks.java:85-151: an output voxel is set
+    (uint8 255 / uint16 65535 / float32 1.0, :152-176) iff ANY view's
+    inverse affine maps it into [dim.min - maskOffset, dim.max +
+    maskOffset] per axis, inclusive; no interpolation, no intensity
+    scaling. maskOffset is in raw input px (SparkAffineFusion.java:
+    112-115)."""
+    bs = np.asarray(block_size, dtype=np.int64)
+    bm = np.asarray(block_min, dtype=np.float64)
+    zz, yy, xx = np.meshgrid(
+        np.arange(bs[2]), np.arange(bs[1]), np.arange(bs[0]), indexing="ij"
+    )
+    w = np.stack(
+        [xx + bm[0], yy + bm[1], zz + bm[2]], axis=-1
+    ).astype(np.float64)
+    off = np.asarray(mask_offset, dtype=np.float64)
+    any_view = np.zeros(w.shape[:-1], dtype=bool)
+    for v in views:
+        nz, ny, nx = v["data"].shape
+        dims = np.array([nx, ny, nz], dtype=np.float64)
+        inv = _invert_affine(np.asarray(v["affine"], dtype=np.float64))
+        p = w @ inv[:, :3].T + inv[:, 3]
+        inside = np.ones(p.shape[:-1], dtype=bool)
+        for d in range(3):
+            inside &= (p[..., d] >= -off[d]) & (
+                p[..., d] <= dims[d] - 1 + off[d])
+        any_view |= inside
+    if out_dtype == np.float32:
+        return any_view.astype(np.float32)
+    tmax = 255 if out_dtype == np.uint8 else 65535
+    return np.where(any_view, tmax, 0).astype(out_dtype)
 
 
 def downsample_level(vol, rel):

@@ -356,3 +356,35 @@ def test_cli_grid6_stitch_solve_fuse(tmp_path):
     # are legitimately empty where solver-shifted tiles do not reach
     interior = fused[3:-3, 3:-3, 3:-3]
     assert (interior == 0).mean() < 0.01
+
+
+@pytest.mark.gpu
+def test_cli_fusion_masks(tmp_path):
+    """--masks writes coverage masks instead of fused intensities
+    (reference SparkAffineFusion.java:112-115, :565-578 via
+    GenerateComputeBlockMasks); same container layout, pyramid of the
+    mask content."""
+    xml, n5, err, (a, b) = make_grid_dataset(str(tmp_path))
+    out = os.path.join(str(tmp_path), "masks.n5")
+    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml, "-o",
+             out, "--blockSize", "32,32,32", "--dataType", "UINT8",
+             "--downsamplings", "1,1,1;2,2,2"])
+    assert r.returncode == 0, r.stderr
+    r = run([os.path.join(BIN, "affine-fusion"), "-o", out, "--masks",
+             "--maskOffset", "0.0,0.0,0.0"])
+    assert r.returncode == 0, r.stderr + r.stdout
+    m, _ = n5util.read_dataset(out, "ch0tp0/s0")
+    assert m.shape == (64, 64, 104) and m.dtype == np.uint8
+    ident = np.hstack([np.eye(3), np.zeros((3, 1))])
+    affB = ident.copy()
+    affB[0, 3] = 40.0
+    ref = of.mask_block(
+        [dict(data=a, affine=ident), dict(data=b, affine=affB)],
+        (0, 0, 0), (104, 64, 64), out_dtype=np.uint8)
+    # the stitched grid covers x in [0,63]+[40,103]: everything set
+    # except nothing — compare exactly (translation-only affines)
+    assert np.array_equal(m, ref)
+    assert ref.max() == 255
+    s1, _ = n5util.read_dataset(out, "ch0tp0/s1")
+    ref1 = of.downsample_level(ref, (2, 2, 2))
+    assert np.array_equal(s1, ref1)

@@ -213,3 +213,63 @@ def test_fuse_intensity_coefficients_parity(ctx):
                            fusion_type=of.FUSION_AVG_BLEND,
                            out_dtype=np.float32)[0]
     assert np.max(np.abs(got2 - ref2) / np.maximum(np.abs(ref2), 1.0)) < 1e-4
+
+
+@pytest.mark.parametrize("dtype", [np.uint8, np.uint16, np.float32])
+def test_mask_parity_translation(ctx, dtype):
+    """--masks coverage-mask mode vs oracle (GenerateComputeBlockMasks):
+    translation-only affines with fractional offsets keep every sample
+    >0.2 px from a containment boundary, so fp32 vs fp64 predicates agree
+    exactly."""
+    rng = np.random.default_rng(21)
+    views = []
+    for i in range(3):
+        data = rng.integers(0, 40000, size=(10, 12, 14)).astype(np.uint16)
+        aff = IDENT.copy()
+        aff[:, 3] = [3.37 + 5 * i, -2.63 + 3 * i, 1.21 + 2 * i]
+        views.append(dict(data=data, affine=aff))
+    gviews = upload_views(ctx, views, base=300)
+    bmin, bsize = (-2, -1, 0), (24, 20, 16)
+    moff = (1.3, 0.0, 0.7)
+    ref = of.mask_block(views, bmin, bsize, mask_offset=moff,
+                        out_dtype=dtype)
+    got = ctx.fuse_blocks(gviews, [(bmin, bsize)], [[0, 1, 2]],
+                          out_dtype=dtype, masks=True,
+                          mask_offset=moff)[0]
+    assert got.shape == ref.shape and got.dtype == ref.dtype
+    assert ref.max() > 0 and ref.min() == 0  # mask has both phases
+    assert np.array_equal(got, ref)
+
+
+def test_mask_parity_sheared(ctx):
+    """Sheared/scaled affines: fp32 inverse-affine may flip voxels that
+    land within ~1e-4 px of a containment boundary — allow a tiny
+    mismatch fraction, mirroring the fusion parity bar."""
+    views = fusion_views(7)
+    gviews = upload_views(ctx, views, base=310)
+    bmin, bsize = (0, 0, 0), (40, 36, 30)
+    ref = of.mask_block(views, bmin, bsize, out_dtype=np.uint8)
+    got = ctx.fuse_blocks(gviews, [(bmin, bsize)], [[0, 1, 2]],
+                          out_dtype=np.uint8, masks=True)[0]
+    assert ref.max() > 0
+    assert (got != ref).mean() < 1e-3
+
+
+def test_mask_volume_pyramid(ctx):
+    """Masks through bs_fuse_volume + pyramid: level 0 is the mask,
+    higher levels are its box-means [PIN-PYR] (the reference pyramids
+    whatever level 0 holds)."""
+    rng = np.random.default_rng(5)
+    data = rng.integers(0, 40000, size=(12, 12, 12)).astype(np.uint16)
+    aff = IDENT.copy()
+    aff[:, 3] = [1.41, 2.72, 0.58]
+    ctx.upload(320, data)
+    gv = [dict(view_id=320, affine=aff)]
+    lv = ctx.fuse_volume(gv, (0, 0, 0), (16, 16, 16),
+                         downsamplings=[(1, 1, 1), (2, 2, 2)],
+                         out_dtype=np.uint16, masks=True)
+    ref0 = of.mask_block([dict(data=data, affine=aff)], (0, 0, 0),
+                         (16, 16, 16), out_dtype=np.uint16)
+    assert np.array_equal(lv[0], ref0)
+    ref1 = of.downsample_level(ref0, (2, 2, 2))
+    assert np.array_equal(lv[1], ref1)

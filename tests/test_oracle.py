@@ -296,3 +296,30 @@ def test_fuse_intensity_coefficients():
         fusion.FUSION_AVG,
     )
     assert np.all(out == 230.0)
+
+
+def test_mask_block_semantics():
+    """mask_block restates GenerateComputeBlockMasks.java:85-151:
+    containment of the inverse-mapped point in [min-off, max+off],
+    inclusive, ANY view; 255/65535/1.0 fill."""
+    from oracle import fusion as of
+
+    data = np.zeros((4, 4, 4), np.uint16)
+    aff = np.hstack([np.eye(3), np.array([[2.0], [0.0], [0.0]])])
+    m = of.mask_block([dict(data=data, affine=aff)], (0, 0, 0),
+                      (8, 6, 5), out_dtype=np.uint8)
+    # view occupies x in [2, 5], y in [0, 3], z in [0, 3] (inclusive)
+    assert m.dtype == np.uint8 and m.shape == (5, 6, 8)
+    assert m[0, 0, 2] == 255 and m[0, 0, 5] == 255
+    assert m[0, 0, 1] == 0 and m[0, 0, 6] == 0
+    assert m[3, 3, 3] == 255 and m[4, 0, 3] == 0 and m[0, 4, 3] == 0
+    # maskOffset widens per axis (x only here)
+    m2 = of.mask_block([dict(data=data, affine=aff)], (0, 0, 0),
+                       (8, 6, 5), mask_offset=(1.0, 0.0, 0.0),
+                       out_dtype=np.uint16)
+    assert m2[0, 0, 1] == 65535 and m2[0, 0, 6] == 65535
+    assert m2[0, 4, 3] == 0
+    # float32 fill is 1.0
+    m3 = of.mask_block([dict(data=data, affine=aff)], (0, 0, 0),
+                       (8, 6, 5), out_dtype=np.float32)
+    assert m3.max() == 1.0 and m3.dtype == np.float32
