@@ -19,6 +19,67 @@
 #include "bs_zarr.h"
 #include "bs_spimdata.h"
 
+
+/* Load one view's intensity coefficients from an N5 container
+ * (reference SparkAffineFusion.java:158-166, :554:
+ * IntensityCorrection.readCoefficients at
+ * "{--intensityN5Group}/setup{s}/timepoint{t}/{--intensityN5Dataset}").
+ * The artifact-side array layout is not vendored; restated as a 4-D
+ * dataset dims {gx, gy, gz, 2} (dimension 0 fastest; the slowest axis
+ * indexes the multiplicative then the additive field), float32 or
+ * float64 — parity at this sub-boundary is unpinned (oracle/__init__
+ * note). Returns false if the dataset is absent (view fused without
+ * correction, matching the reference's per-view map lookup). */
+static bool load_coefficients(const bsn5::Container &cn,
+                              const std::string &group,
+                              const std::string &dsname, int setup, int tp,
+                              std::vector<float> *ab, int32_t gdims[3]) {
+  std::string name = (group.empty() ? std::string() : group + "/") +
+                     "setup" + std::to_string(setup) + "/timepoint" +
+                     std::to_string(tp) + "/" + dsname;
+  bsn5::DatasetAttrs a;
+  if (!cn.get_dataset_attrs(name, &a) || a.dims.size() != 4 ||
+      a.dims[3] != 2 || (a.dtype != "float32" && a.dtype != "float64"))
+    return false;
+  long long gx = a.dims[0], gy = a.dims[1], gz = a.dims[2];
+  size_t n = (size_t)gx * gy * gz * 2;
+  size_t esz = bsn5::dtype_size(a.dtype);
+  std::vector<char> buf(n * esz, 0);
+  std::vector<int> nb(4), bsz(4);
+  for (int d = 0; d < 4; ++d) {
+    bsz[d] = a.block[d];
+    nb[d] = (int)((a.dims[d] + a.block[d] - 1) / a.block[d]);
+  }
+  std::vector<char> blk((size_t)bsz[0] * bsz[1] * bsz[2] * bsz[3] * esz);
+  for (long long g3 = 0; g3 < nb[3]; ++g3)
+    for (long long g2 = 0; g2 < nb[2]; ++g2)
+      for (long long g1 = 0; g1 < nb[1]; ++g1)
+        for (long long g0 = 0; g0 < nb[0]; ++g0) {
+          std::vector<int> cd;
+          if (!cn.read_block(name, a, {g0, g1, g2, g3}, blk.data(), &cd))
+            continue; /* missing chunk = zeros */
+          for (int f = 0; f < cd[3]; ++f)
+            for (int z = 0; z < cd[2]; ++z)
+              for (int y = 0; y < cd[1]; ++y) {
+                long long dst = (((g3 * bsz[3] + f) * gz + g2 * bsz[2] + z) *
+                                     gy + g1 * bsz[1] + y) * gx + g0 * bsz[0];
+                memcpy(&buf[dst * esz],
+                       &blk[(((size_t)f * cd[2] + z) * cd[1] + y) * cd[0] *
+                            esz],
+                       (size_t)cd[0] * esz);
+              }
+        }
+  ab->resize(n);
+  if (a.dtype == "float64") {
+    const double *src = (const double *)buf.data();
+    for (size_t i = 0; i < n; ++i) (*ab)[i] = (float)src[i];
+  } else {
+    memcpy(ab->data(), buf.data(), n * 4);
+  }
+  gdims[0] = (int32_t)gx; gdims[1] = (int32_t)gy; gdims[2] = (int32_t)gz;
+  return true;
+}
+
 int main(int argc, char **argv) {
   bscli::Args args;
   std::map<std::string, std::string> alias = {{"-x", "--xml"},
@@ -28,7 +89,9 @@ int main(int argc, char **argv) {
             "usage: affine-fusion -o out.n5 [-x dataset.xml] "
             "[--fusionType AVG_BLEND|AVG|MAX_INTENSITY] "
             "[--blendingRange 40] [--blendingBorder 0] [--device N] "
-            "[--batchSize 64] [--masks] [--maskOffset 0.0,0.0,0.0]\n");
+            "[--batchSize 64] [--masks] [--maskOffset 0.0,0.0,0.0] "
+            "[--intensityN5Path coeff.n5 [--intensityN5Group g] "
+            "[--intensityN5Dataset intensity]]\n");
     return 2;
   }
   bsn5::Container n5(args.get("n5Path"));
@@ -131,6 +194,22 @@ int main(int argc, char **argv) {
       for (int d2 = 0; d2 < 3; ++d2) {
         fv.blend_border[d2] = bborder;
         fv.blend_range[d2] = brange;
+      }
+      if (args.has("intensityN5Path")) {
+        bsn5::Container coeff_n5(args.get("intensityN5Path"));
+        std::vector<float> ab;
+        int32_t gd[3];
+        if (load_coefficients(coeff_n5, args.get("intensityN5Group", ""),
+                              args.get("intensityN5Dataset", "intensity"),
+                              s.id, tp, &ab, gd)) {
+          if (bs_view_set_coefficients(ctx, s.id, ab.data(), gd) != BS_OK) {
+            fprintf(stderr, "set_coefficients failed: %s\n",
+                    bs_last_error(ctx));
+            return 1;
+          }
+          printf("loaded intensity coefficients for setup %d (%dx%dx%d)\n",
+                 s.id, gd[0], gd[1], gd[2]);
+        }
       }
       fviews.push_back(fv);
       fsetups.push_back(&s);

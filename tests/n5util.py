@@ -161,3 +161,30 @@ def read_zarr(root, name):
             sl_blk.append(slice(0, hi - lo))
         out[tuple(sl_out)] = blk[tuple(sl_blk)]
     return out, za
+
+
+def write_dataset_nd(root, name, arr, compression="gzip"):
+    """N-D N5 dataset, one chunk (blockSize == dimensions). arr axes in
+    numpy (slowest..fastest) order; stored dims are reversed (dimension
+    0 fastest), matching the 3-D convention above. Supports float32/
+    float64 (coefficient containers)."""
+    dt = {np.dtype(np.float32): "float32",
+          np.dtype(np.float64): "float64"}[arr.dtype]
+    dims = list(reversed(arr.shape))
+    ds = os.path.join(root, name)
+    os.makedirs(ds, exist_ok=True)
+    comp = {"type": compression}
+    if compression == "gzip":
+        comp.update(level=-1, useZlib=False)
+    with open(os.path.join(ds, "attributes.json"), "w") as f:
+        json.dump({"dimensions": dims, "blockSize": dims,
+                   "dataType": dt, "compression": comp}, f)
+    payload = arr.astype(arr.dtype.newbyteorder(">")).tobytes()
+    if compression == "gzip":
+        payload = gzip.compress(payload)
+    nd = len(dims)
+    hdr = struct.pack(">HH", 0, nd) + struct.pack(">%dI" % nd, *dims)
+    d = os.path.join(ds, *["0"] * (nd - 1))
+    os.makedirs(d, exist_ok=True)
+    with open(os.path.join(d, "0"), "wb") as f:
+        f.write(hdr + payload)
