@@ -425,29 +425,48 @@ __global__ __launch_bounds__(64 * XW_WPB) void k_fft_x_fwd_w(
   for (int i = tid; i < h / 2; i += 64 * XW_WPB) tw[i] = twg[2 * i];
   for (int i = tid; i < h; i += 64 * XW_WPB) twn[i] = twg[i];
   __syncthreads(); /* the only barrier: tables ready */
+  __shared__ unsigned stg[XW_WPB][64 * E]; /* bit-reversal staging */
   const int lane = tid & 63, wv = tid >> 6;
   const long nlines = (long)in.my * in.mz;
-  for (long lid = (long)blockIdx.x * XW_WPB + wv; lid < nlines;
-       lid += (long)gridDim.x * XW_WPB) {
-    const int y = (int)(lid % in.my), z = (int)(lid / in.my);
+  const long stride = (long)gridDim.x * XW_WPB;
+  const long niter = (nlines + stride - 1) / stride;
+  for (long it = 0; it < niter; ++it) {
+    const long lid = it * stride + (long)blockIdx.x * XW_WPB + wv;
+    const bool active = lid < nlines;
+    const int y = active ? (int)(lid % in.my) : 0;
+    const int z = active ? (int)(lid / in.my) : 0;
     const unsigned short *src =
         in.ptr + (in.oz + z) * in.sxy + (in.oy + y) * in.sx + in.ox;
     f2 v[E];
-    const bool al4 = ((size_t)src & 3) == 0 && in.mx >= 2 * h;
+    const bool al4 =
+        active && ((size_t)src & 3) == 0 && in.mx >= 2 * h;
+    if (al4) {
+      /* contiguous u32 loads (2 cachelines per wave-load instead of a
+       * 16-line bit-reversed gather), bit reversal via one LDS
+       * round-trip; all waves iterate niter times so the barrier is
+       * uniform */
+#pragma unroll
+      for (int e = 0; e < E; ++e)
+        stg[wv][e * 64 + lane] = ((const unsigned *)src)[e * 64 + lane];
+    }
+    __syncthreads();
 #pragma unroll
     for (int e = 0; e < E; ++e) {
       const int j = (int)brev_n((unsigned)(e * 64 + lane), log2h);
       if (al4) {
-        unsigned w = ((const unsigned *)src)[j];
+        unsigned w = stg[wv][j];
         v[e] = {(float)(w & 0xFFFF), (float)(w >> 16)};
-      } else {
+      } else if (active) {
         float xa = (2 * j < in.mx) ? (float)src[2 * j] : 0.0f;
         float xb = (2 * j + 1 < in.mx) ? (float)src[2 * j + 1] : 0.0f;
         v[e] = {xa, xb};
+      } else {
+        v[e] = {0.0f, 0.0f};
       }
     }
     ffth_wave<E>(v, lane, tw, +1);
     f2 *o = out + ((long)z * py + y) * cxp;
+    if (active)
 #pragma unroll
     for (int e = 0; e < E; ++e) {
       /* position k = e*64+lane; Z[h-k] lives at element E-1-e lane
@@ -468,6 +487,7 @@ __global__ __launch_bounds__(64 * XW_WPB) void k_fft_x_fwd_w(
         o[k] = {ze.x + wzo.x, ze.y + wzo.y};
       }
     }
+    __syncthreads(); /* stg reused next iteration */
   }
 }
 
