@@ -232,6 +232,116 @@ __global__ __launch_bounds__(LPB *TPL) void k_pass(
   }
 }
 
+
+/* wave-register hybrid: I/O transpose through LDS (stride LPB+1 to
+ * spread banks), the n-point FFT entirely in one wave's registers
+ * (E = n/64 f2 per lane at positions p = e*64 + lane): radix-2 DIT
+ * via __shfl_xor below stride 64, in-register butterflies above.
+ * 3 barriers per workgroup instead of the radix-2^2 path's ~7, and
+ * no LDS data traffic during the butterfly stages. */
+template <int E>
+__device__ __forceinline__ void ffthw(f2 (&v)[E], int lane, const f2 *tw,
+                                      int dir) {
+  constexpr int h = 64 * E;
+#pragma unroll
+  for (int s = 1; s < 64 && s < h; s <<= 1) {
+    f2 w = tw[(lane & (s - 1)) * ((h / 2) / s)];
+    if (dir < 0) w.y = -w.y;
+    const bool up = lane & s;
+    const float sg = up ? -1.0f : 1.0f;
+#pragma unroll
+    for (int e = 0; e < E; ++e) {
+      f2 t = {__shfl_xor(v[e].x, s), __shfl_xor(v[e].y, s)};
+      f2 b = up ? v[e] : t;
+      f2 a = up ? t : v[e];
+      f2 wb = cmul(b, w);
+      v[e] = {a.x + sg * wb.x, a.y + sg * wb.y};
+    }
+  }
+#pragma unroll
+  for (int s = 64; s < h; s <<= 1) {
+    const int es = s >> 6;
+#pragma unroll
+    for (int e = 0; e < E; ++e) {
+      if (e & es) continue;
+      const int off = (e & (es - 1)) * 64 + lane;
+      f2 w = tw[off * ((h / 2) / s)];
+      if (dir < 0) w.y = -w.y;
+      f2 wb = cmul(v[e + es], w);
+      f2 a = v[e];
+      v[e] = {a.x + wb.x, a.y + wb.y};
+      v[e + es] = {a.x - wb.x, a.y - wb.y};
+    }
+  }
+}
+
+template <int LPB, int TPL, int E>
+__global__ __launch_bounds__(LPB *TPL) void k_pass_w(
+    const f2 *__restrict__ in, f2 *__restrict__ out, int n, int log2n,
+    long estride, long gstride, int nlines, int nchunks, int ngroups,
+    const f2 *twg) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  constexpr int LS = LPB + 1; /* padded line stride (bank spread) */
+  f2 *tw = (f2 *)smem;
+  f2 *data = tw + (n >> 1);
+  const int tid = threadIdx.x;
+  for (int i = tid; i < (n >> 1); i += LPB * TPL) tw[i] = twg[i];
+  __syncthreads();
+  constexpr int NPAIR = LPB / 2;
+  constexpr int ESTR = (LPB * TPL) / NPAIR;
+  const int pl = tid & (NPAIR - 1), t2 = tid / NPAIR;
+  const int lane = tid & 63, wv = tid >> 6; /* wave = line (LPB==waves) */
+  const long nwg = (long)ngroups * nchunks;
+  for (long wg = blockIdx.x; wg < nwg; wg += gridDim.x) {
+    const int group = (int)(wg / nchunks);
+    const int x2 = (int)(wg % nchunks) * LPB + 2 * pl;
+    const bool pair_ok = x2 + 1 < nlines;
+    const long base2 = (long)group * gstride + x2;
+    if (pair_ok) {
+      for (int e = t2; e < n; e += ESTR) {
+        float4 v = *(const float4 *)&in[base2 + e * estride];
+        f2 *d = &data[(long)brev_n(e, log2n) * LS + 2 * pl];
+        d[0] = {v.x, v.y};
+        d[1] = {v.z, v.w};
+      }
+    } else {
+      for (int l = 0; l < 2; ++l) {
+        const int x = x2 + l;
+        const long base = (long)group * gstride + x;
+        for (int e = t2; e < n; e += ESTR) {
+          f2 v = {0, 0};
+          if (x < nlines) v = in[base + e * estride];
+          data[(long)brev_n(e, log2n) * LS + 2 * pl + l] = v;
+        }
+      }
+    }
+    __syncthreads();
+    f2 v[E];
+#pragma unroll
+    for (int e = 0; e < E; ++e) v[e] = data[(long)(e * 64 + lane) * LS + wv];
+    ffthw<E>(v, lane, tw, +1);
+#pragma unroll
+    for (int e = 0; e < E; ++e) data[(long)(e * 64 + lane) * LS + wv] = v[e];
+    __syncthreads();
+    if (pair_ok) {
+      for (int e = t2; e < n; e += ESTR) {
+        const f2 *d = &data[(long)e * LS + 2 * pl];
+        float4 v4 = {d[0].x, d[0].y, d[1].x, d[1].y};
+        *(float4 *)&out[base2 + e * estride] = v4;
+      }
+    } else {
+      for (int l = 0; l < 2; ++l) {
+        const int x = x2 + l;
+        if (x >= nlines) continue;
+        const long base = (long)group * gstride + x;
+        for (int e = t2; e < n; e += ESTR)
+          out[base + e * estride] = data[(long)e * LS + 2 * pl + l];
+      }
+    }
+    __syncthreads();
+  }
+}
+
 #define HIPCHK(x)                                                           \
   if ((x) != hipSuccess) {                                                  \
     printf("hip error %s @%d\n", hipGetErrorString(hipGetLastError()),      \
@@ -276,11 +386,22 @@ int main() {
       {"r2^2 LPB16xTPL32", 16, 32, 0},
       {"r2^2 LPB4xTPL64 (256t)", 4, 64, 0},
       {"r2^3 LPB16xTPL32", 16, 32, 1},
+      {"wave-reg LPB8xE8", 8, 64, 3},
   };
   auto launch = [&](const Var &v) {
     int nchunks = (Cx + v.lpb - 1) / v.lpb;
-    size_t lds = ((n / 2) + (size_t)(v.lpb) * n) * sizeof(f2);
+    size_t lds = ((n / 2) + (size_t)(v.var == 3 ? v.lpb + 1 : v.lpb) * n) *
+                 sizeof(f2);
     long grid = std::min(4096L, (long)512 * nchunks);
+    if (v.var == 3) {
+      hipFuncSetAttribute((const void *)k_pass_w<8, 64, 8>,
+                          hipFuncAttributeMaxDynamicSharedMemorySize,
+                          160 * 1024);
+      hipLaunchKernelGGL((k_pass_w<8, 64, 8>), dim3(grid), dim3(512), lds,
+                         0, din, dout, n, log2n, Cxp, (long)Py * Cxp, Cx,
+                         nchunks, 512, dtw);
+      return;
+    }
 #define CASE(L, T, V)                                                       \
   if (v.lpb == L && v.tpl == T && v.var == V) {                             \
     hipFuncSetAttribute((const void *)k_pass<L, T, V>,                      \
