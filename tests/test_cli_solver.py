@@ -126,3 +126,41 @@ def test_solver_three_view_chain(tmp_path):
     t = model_translations(xml)
     assert np.allclose(t[1], [41.0, 0, 0], atol=1e-6)
     assert np.allclose(t[2], [83.0, 0, 0], atol=1e-6)
+
+
+def test_solver_isolated_view_unchanged(tmp_path):
+    """A view with no (accepted) links must keep its registration and
+    the solver must not fail on the disconnected graph (the reference
+    optimises per connected component; an isolated tile stays put)."""
+    xml = os.path.join(str(tmp_path), "d.xml")
+    n5util.make_dataset_xml(
+        xml, "input.n5",
+        [dict(id=0, dims=(64, 64, 64), pos=(0.0, 0.0, 0.0)),
+         dict(id=1, dims=(64, 64, 64), pos=(40.0, 0.0, 0.0)),
+         dict(id=2, dims=(64, 64, 64), pos=(0.0, 40.0, 0.0))],
+    )
+    # link only 0<->1; setup 2 is isolated
+    text = open(xml).read()
+    sr = (
+        "  <StitchingResults>\n"
+        "    <PairwiseResult>\n"
+        "      <ViewIdsA>0,0</ViewIdsA>\n"
+        "      <ViewIdsB>0,1</ViewIdsB>\n"
+        "      <Matrix>1 0 0 -1.5 0 1 0 0 0 0 1 0</Matrix>\n"
+        "      <BoundingBoxMin>0 0 0</BoundingBoxMin>\n"
+        "      <BoundingBoxMax>1 1 1</BoundingBoxMax>\n"
+        "      <Correlation>0.95</Correlation>\n"
+        f"      <Hash>{3.0 + (3.0 + 40.0)}</Hash>\n"  # [PIN-HASH]
+        "    </PairwiseResult>\n"
+        "  </StitchingResults>\n"
+    )
+    open(xml, "w").write(text.replace("</SpimData>", sr + "</SpimData>"))
+    before = model_translations(xml)
+    r = run([os.path.join(BIN, "solver"), "-x", xml])
+    assert r.returncode == 0, r.stderr + r.stdout
+    after = model_translations(xml)
+    # isolated view 2: unchanged
+    assert np.allclose(after[2], before[2])
+    # linked pair: relative x positions corrected by the link (-1.5)
+    rel = after[1][0] - after[0][0]
+    assert abs(rel - (40.0 - 1.5)) < 1e-6

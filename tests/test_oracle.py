@@ -323,3 +323,37 @@ def test_mask_block_semantics():
     m3 = of.mask_block([dict(data=data, affine=aff)], (0, 0, 0),
                        (8, 6, 5), out_dtype=np.float32)
     assert m3.max() == 1.0 and m3.dtype == np.float32
+
+
+def test_downsample_level_odd_dims_brute_force():
+    """[PIN-PYR] edge boxes average only their in-bounds voxels; checked
+    against a direct per-cell brute force on awkward odd dims."""
+    from oracle import fusion as of
+
+    rng = np.random.default_rng(13)
+    vol = rng.integers(0, 65536, size=(5, 7, 3)).astype(np.uint16)
+    got = of.downsample_level(vol, (2, 2, 2))
+    nz, ny, nx = vol.shape
+    assert got.shape == ((nz + 1) // 2, (ny + 1) // 2, (nx + 1) // 2)
+    for z in range(got.shape[0]):
+        for y in range(got.shape[1]):
+            for x in range(got.shape[2]):
+                box = vol[2 * z:2 * z + 2, 2 * y:2 * y + 2,
+                          2 * x:2 * x + 2].astype(np.float64)
+                assert got[z, y, x] == np.clip(
+                    np.rint(box.mean()), 0, 65535).astype(np.uint16)
+
+
+def test_phase_correlation_min_overlap_rejects():
+    """min_overlap_ratio culls candidate shifts whose overlap volume is
+    below the threshold; with an impossible threshold the pair is
+    invalid (mirrors the GPU-side test at the oracle level)."""
+    from oracle import phasecorr, synth
+
+    a, b = synth.make_pair((48, 48, 48), (5.0, -3.0, 2.0), seed=3)
+    ok = phasecorr.phase_correlation_shift(a, b, ds=(1, 1, 1),
+                                           min_overlap_ratio=0.05)
+    assert ok["valid"] == 1
+    bad = phasecorr.phase_correlation_shift(a, b, ds=(1, 1, 1),
+                                            min_overlap_ratio=1.01)
+    assert bad["valid"] == 0
