@@ -161,6 +161,8 @@ def test_solver_isolated_view_unchanged(tmp_path):
     after = model_translations(xml)
     # isolated view 2: unchanged
     assert np.allclose(after[2], before[2])
-    # linked pair: relative x positions corrected by the link (-1.5)
+    # linked pair: ws = -e means B's content sits at -1.5 relative to
+    # the registrations; solver moves B by +e = -ws (the convention
+    # pinned by test_solver_translation_solve)
     rel = after[1][0] - after[0][0]
-    assert abs(rel - (40.0 - 1.5)) < 1e-6
+    assert abs(rel - (40.0 + 1.5)) < 1e-6
