@@ -342,6 +342,73 @@ __global__ __launch_bounds__(LPB *TPL) void k_pass_w(
   }
 }
 
+
+/* VAR 4: radix-2^2 with the LDS line array padded to stride LPB+1
+ * (element stride 9 f2 = 18 words: gcd(18,64)=2 -> 32 banks instead
+ * of 16-word stride's 4; SQ_LDS_BANK_CONFLICT measured 29% of LDS
+ * cycles on the production layout). I/O uses f2 pairs since the
+ * padded offsets lose float4 alignment on odd elements. */
+template <int LPB, int TPL>
+__global__ __launch_bounds__(LPB *TPL) void k_pass_p(
+    const f2 *__restrict__ in, f2 *__restrict__ out, int n, int log2n,
+    long estride, long gstride, int nlines, int nchunks, int ngroups,
+    const f2 *twg) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  constexpr int LS = LPB + 1;
+  f2 *tw = (f2 *)smem;
+  f2 *data = tw + (n >> 1);
+  const int tid = threadIdx.x;
+  const int line = tid % LPB, tl = tid / LPB;
+  for (int i = tid; i < (n >> 1); i += LPB * TPL) tw[i] = twg[i];
+  __syncthreads();
+  constexpr int NPAIR = LPB / 2;
+  constexpr int ESTR = (LPB * TPL) / NPAIR;
+  const int pl = tid & (NPAIR - 1), t2 = tid / NPAIR;
+  const long nwg = (long)ngroups * nchunks;
+  for (long wg = blockIdx.x; wg < nwg; wg += gridDim.x) {
+    const int group = (int)(wg / nchunks);
+    const int x2 = (int)(wg % nchunks) * LPB + 2 * pl;
+    const bool pair_ok = x2 + 1 < nlines;
+    const long base2 = (long)group * gstride + x2;
+    if (pair_ok) {
+      for (int e = t2; e < n; e += ESTR) {
+        float4 v = *(const float4 *)&in[base2 + e * estride];
+        f2 *d = &data[(long)brev_n(e, log2n) * LS + 2 * pl];
+        d[0] = {v.x, v.y};
+        d[1] = {v.z, v.w};
+      }
+    } else {
+      for (int l = 0; l < 2; ++l) {
+        const int x = x2 + l;
+        const long base = (long)group * gstride + x;
+        for (int e = t2; e < n; e += ESTR) {
+          f2 v = {0, 0};
+          if (x < nlines) v = in[base + e * estride];
+          data[(long)brev_n(e, log2n) * LS + 2 * pl + l] = v;
+        }
+      }
+    }
+    __syncthreads();
+    fft22<LS, TPL>(data, (long)line, n, log2n, tl, tw, +1);
+    if (pair_ok) {
+      for (int e = t2; e < n; e += ESTR) {
+        const f2 *d = &data[(long)e * LS + 2 * pl];
+        float4 v4 = {d[0].x, d[0].y, d[1].x, d[1].y};
+        *(float4 *)&out[base2 + e * estride] = v4;
+      }
+    } else {
+      for (int l = 0; l < 2; ++l) {
+        const int x = x2 + l;
+        if (x >= nlines) continue;
+        const long base = (long)group * gstride + x;
+        for (int e = t2; e < n; e += ESTR)
+          out[base + e * estride] = data[(long)e * LS + 2 * pl + l];
+      }
+    }
+    __syncthreads();
+  }
+}
+
 #define HIPCHK(x)                                                           \
   if ((x) != hipSuccess) {                                                  \
     printf("hip error %s @%d\n", hipGetErrorString(hipGetLastError()),      \
@@ -387,12 +454,23 @@ int main() {
       {"r2^2 LPB4xTPL64 (256t)", 4, 64, 0},
       {"r2^3 LPB16xTPL32", 16, 32, 1},
       {"wave-reg LPB8xE8", 8, 64, 3},
+      {"r2^2 pad9 LPB8", 8, 64, 4},
   };
   auto launch = [&](const Var &v) {
     int nchunks = (Cx + v.lpb - 1) / v.lpb;
     size_t lds = ((n / 2) + (size_t)(v.var == 3 ? v.lpb + 1 : v.lpb) * n) *
                  sizeof(f2);
     long grid = std::min(4096L, (long)512 * nchunks);
+    if (v.var == 4) {
+      size_t lds4 = ((n / 2) + (size_t)(v.lpb + 1) * n) * sizeof(f2);
+      hipFuncSetAttribute((const void *)k_pass_p<8, 64>,
+                          hipFuncAttributeMaxDynamicSharedMemorySize,
+                          160 * 1024);
+      hipLaunchKernelGGL((k_pass_p<8, 64>), dim3(grid), dim3(512), lds4,
+                         0, din, dout, n, log2n, Cxp, (long)Py * Cxp, Cx,
+                         nchunks, 512, dtw);
+      return;
+    }
     if (v.var == 3) {
       hipFuncSetAttribute((const void *)k_pass_w<8, 64, 8>,
                           hipFuncAttributeMaxDynamicSharedMemorySize,
