@@ -88,3 +88,19 @@ def test_find_overlapping_views():
     assert host.find_overlapping_views(views, (64, 0, 0), (32, 32, 32)) == [0]
     assert host.find_overlapping_views(views, (900, 900, 900),
                                        (200, 200, 200)) == [1]
+
+
+def test_shard_disjoint_covering_balanced():
+    """Hash-sharding is disjoint, covering, deterministic, and roughly
+    balanced."""
+    from bigstitcher_spark_amd.host import shard
+
+    n, world = 1024, 8
+    parts = [shard(n, world, r) for r in range(world)]
+    allidx = np.concatenate(parts)
+    assert len(allidx) == n and len(np.unique(allidx)) == n
+    sizes = [len(p) for p in parts]
+    assert min(sizes) > n / world * 0.7 and max(sizes) < n / world * 1.3
+    assert np.array_equal(shard(n, world, 3), shard(n, world, 3))
+    # world=1 gets everything in order
+    assert np.array_equal(shard(5, 1, 0), np.arange(5))

@@ -357,3 +357,34 @@ def test_phase_correlation_min_overlap_rejects():
     bad = phasecorr.phase_correlation_shift(a, b, ds=(1, 1, 1),
                                             min_overlap_ratio=1.01)
     assert bad["valid"] == 0
+
+
+def test_subpixel_offset_exact_parabola():
+    """[PIN-SUB] the per-axis quadratic fit recovers the vertex of an
+    exact parabola exactly, clamps to +-0.5, and wraps periodically."""
+    from oracle.phasecorr import _subpixel_offset
+
+    p = np.zeros((8, 8, 8))
+    true = np.array([0.3, -0.2, 0.45])  # vertex offsets per axis (z,y,x)
+    peak = (4, 4, 4)
+    # separable parabola f = sum_d -(i_d - peak_d - t_d)^2
+    for d, t in enumerate(true):
+        idx = [np.arange(8)[:, None, None], np.arange(8)[None, :, None],
+               np.arange(8)[None, None, :]][d]
+        p = p - (idx - peak[d] - t) ** 2
+    off = _subpixel_offset(p, peak)
+    assert np.allclose(off, true, atol=1e-12)
+    # vertex far away -> clamped to 0.5
+    p2 = np.zeros((8, 8, 8))
+    zz = np.arange(8)[:, None, None].astype(float)
+    p2 += -(zz - 4 - 2.0) ** 2 * 0.001
+    p2[4, 4, 4] += 1e-9  # keep center the max
+    off2 = _subpixel_offset(p2, (4, 4, 4))
+    assert off2[0] == 0.5
+    # periodic wrap: peak at index 0 uses p[-1] as the minus neighbor
+    p3 = np.zeros((8, 1, 1))
+    p3[7, 0, 0] = 0.5
+    p3[0, 0, 0] = 1.0
+    p3[1, 0, 0] = 0.4
+    off3 = _subpixel_offset(p3, (0, 0, 0))
+    assert -0.5 <= off3[0] <= 0.5 and off3[0] != 0.0
