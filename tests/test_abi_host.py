@@ -104,3 +104,25 @@ def test_shard_disjoint_covering_balanced():
     assert np.array_equal(shard(n, world, 3), shard(n, world, 3))
     # world=1 gets everything in order
     assert np.array_equal(shard(5, 1, 0), np.arange(5))
+
+
+def test_transformed_bbox_rotation():
+    """transformed_bbox bounds the image of [0, dim-1] under a rotated
+    affine (the +2 px culling guard is applied by the caller,
+    find_overlapping_views — reference ViewUtil.java:309-328)."""
+    from bigstitcher_spark_amd.host import transformed_bbox
+
+    th = np.deg2rad(30.0)
+    aff = np.array([
+        [np.cos(th), -np.sin(th), 0.0, 5.0],
+        [np.sin(th), np.cos(th), 0.0, -3.0],
+        [0.0, 0.0, 1.0, 2.0],
+    ])
+    dims = (10, 20, 4)
+    lo, hi = transformed_bbox(dims, aff)
+    rng = np.random.default_rng(2)
+    pts = rng.uniform([0, 0, 0], [9, 19, 3], size=(500, 3))
+    w = pts @ aff[:, :3].T + aff[:, 3]
+    assert np.all(w.min(0) >= lo - 1e-9) and np.all(w.max(0) <= hi + 1e-9)
+    # corners are attained (bbox is tight)
+    assert np.allclose(lo[2], 2.0) and np.allclose(hi[2], 5.0)
