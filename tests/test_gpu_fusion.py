@@ -273,3 +273,26 @@ def test_mask_volume_pyramid(ctx):
     assert np.array_equal(lv[0], ref0)
     ref1 = of.downsample_level(ref0, (2, 2, 2))
     assert np.array_equal(lv[1], ref1)
+
+
+def test_fuse_view_table_spill_over_64(ctx):
+    """More than BS_MAX_BLK_VIEWS (64) views on one block: views past
+    the LDS table are read from global memory (the k<nvs?sv[k]:... spill
+    path in k_fuse) — results must match the oracle exactly as usual."""
+    rng = np.random.default_rng(77)
+    views = []
+    for i in range(70):
+        data = rng.integers(0, 40000, size=(6, 6, 6)).astype(np.uint16)
+        aff = IDENT.copy()
+        aff[:, 3] = rng.uniform(-2, 10, 3).round(2) + 0.37
+        views.append(dict(data=data, affine=aff,
+                          border=(0.0, 0.0, 0.0), range=(3.0, 3.0, 3.0)))
+    gviews = upload_views(ctx, views, base=400)
+    bmin, bsize = (0, 0, 0), (14, 14, 14)
+    ref = of.fuse_block(views, bmin, bsize, of.FUSION_AVG_BLEND,
+                        out_dtype=np.float32)
+    got = ctx.fuse_blocks(gviews, [(bmin, bsize)], [list(range(70))],
+                          fusion_type=of.FUSION_AVG_BLEND,
+                          out_dtype=np.float32)[0]
+    denom = np.maximum(np.abs(ref), 1.0)
+    assert np.max(np.abs(got - ref) / denom) < 1e-4

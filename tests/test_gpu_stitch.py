@@ -227,3 +227,28 @@ def test_stitch_missing_view_errors(ctx):
                 size_a=(16, 16, 16), off_b=(0, 0, 0), size_b=(16, 16, 16))
     with _pytest.raises(RuntimeError, match="view not uploaded"):
         ctx.stitch_batch([pair], ds=(1, 1, 1))
+
+
+def test_stitch_differing_tile_sizes(ctx):
+    """A and B tiles of different dims: the PCM pads to the
+    element-wise max [PIN-PAD]; parity vs the oracle."""
+    shape_a, shape_b = (48, 64, 80), (64, 48, 72)  # (nz, ny, nx)
+    blobs_a, blobs_b = synth.pair_blobs((64, 64, 80), (5.25, -3.5, 2.0),
+                                        seed=23)
+    a = synth.render_tile(shape_a, blobs_a, noise_seed=101)
+    b = synth.render_tile(shape_b, blobs_b, noise_seed=102)
+    ref = phasecorr.phase_correlation_shift(a, b, ds=(1, 1, 1))
+    ida, idb = 77, 78
+    ctx.upload(ida, a)
+    ctx.upload(idb, b)
+    got = ctx.stitch_batch([
+        dict(view_a=ida, view_b=idb, off_a=(0, 0, 0),
+             size_a=(shape_a[2], shape_a[1], shape_a[0]),
+             off_b=(0, 0, 0),
+             size_b=(shape_b[2], shape_b[1], shape_b[0]))
+    ], ds=(1, 1, 1))[0]
+    assert got["valid"] == ref["valid"]
+    if ref["valid"]:
+        assert np.all(np.abs(got["shift"] - ref["shift"]) < 1e-3), (
+            got["shift"], ref["shift"])
+        assert got["r"] == pytest.approx(ref["r"], abs=1e-9)
