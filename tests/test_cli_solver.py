@@ -166,3 +166,19 @@ def test_solver_isolated_view_unchanged(tmp_path):
     # pinned by test_solver_translation_solve)
     rel = after[1][0] - after[0][0]
     assert abs(rel - (40.0 + 1.5)) < 1e-6
+
+
+def test_solver_malformed_xml_clean_error(tmp_path):
+    """Truncated/invalid XML must produce a clean nonzero exit, not a
+    crash (the host parsers are exercised by every CLI)."""
+    xml = os.path.join(str(tmp_path), "broken.xml")
+    open(xml, "w").write("<SpimData version=\"0.2\"><SequenceDescription>")
+    r = run([os.path.join(BIN, "solver"), "-x", xml])
+    assert r.returncode != 0
+    assert r.returncode < 128, "must exit, not die on a signal"
+
+
+def test_solver_missing_file_clean_error(tmp_path):
+    r = run([os.path.join(BIN, "solver"), "-x",
+             os.path.join(str(tmp_path), "nope.xml")])
+    assert r.returncode != 0 and r.returncode < 128
