@@ -33,23 +33,23 @@ cli: $(BINDIR)/stitching $(BINDIR)/create-fusion-container $(BINDIR)/affine-fusi
 
 $(BINDIR)/stitching: $(HOSTDIR)/cli_stitching.cpp $(HOSTOBJS) $(LIB)
 	@mkdir -p $(BINDIR)
-	$(CXX_HOST) $(HOSTFLAGS) -Iinclude $< $(HOSTOBJS) -o $@ -Lbigstitcher_spark_amd -lbigstitch -lz -L/opt/rocm/lib -lamdhip64 -Wl,-rpath,'$$ORIGIN/..' -Wl,-rpath,/opt/rocm/lib
+	$(CXX_HOST) $(HOSTFLAGS) -Iinclude $< $(HOSTOBJS) -o $@ -Lbigstitcher_spark_amd -lbigstitch -lz -l:libzstd.so.1 -L/opt/rocm/lib -lamdhip64 -Wl,-rpath,'$$ORIGIN/..' -Wl,-rpath,/opt/rocm/lib
 
 $(BINDIR)/create-fusion-container: $(HOSTDIR)/cli_container.cpp $(HOSTOBJS)
 	@mkdir -p $(BINDIR)
-	$(CXX_HOST) $(HOSTFLAGS) -Iinclude $< $(HOSTOBJS) -o $@ -lz
+	$(CXX_HOST) $(HOSTFLAGS) -Iinclude $< $(HOSTOBJS) -o $@ -lz -l:libzstd.so.1
 
 $(BINDIR)/affine-fusion: $(HOSTDIR)/cli_fusion.cpp $(HOSTOBJS) $(LIB)
 	@mkdir -p $(BINDIR)
-	$(CXX_HOST) $(HOSTFLAGS) -Iinclude $< $(HOSTOBJS) -o $@ -Lbigstitcher_spark_amd -lbigstitch -lz -L/opt/rocm/lib -lamdhip64 -Wl,-rpath,'$$ORIGIN/..' -Wl,-rpath,/opt/rocm/lib
+	$(CXX_HOST) $(HOSTFLAGS) -Iinclude $< $(HOSTOBJS) -o $@ -Lbigstitcher_spark_amd -lbigstitch -lz -l:libzstd.so.1 -L/opt/rocm/lib -lamdhip64 -Wl,-rpath,'$$ORIGIN/..' -Wl,-rpath,/opt/rocm/lib
 
 $(BINDIR)/solver: $(HOSTDIR)/cli_solver.cpp $(HOSTOBJS)
 	@mkdir -p $(BINDIR)
-	$(CXX_HOST) $(HOSTFLAGS) -Iinclude $< $(HOSTOBJS) -o $@ -lz
+	$(CXX_HOST) $(HOSTFLAGS) -Iinclude $< $(HOSTOBJS) -o $@ -lz -l:libzstd.so.1
 
 $(BINDIR)/resave: $(HOSTDIR)/cli_resave.cpp $(HOSTOBJS) $(LIB)
 	@mkdir -p $(BINDIR)
-	$(CXX_HOST) $(HOSTFLAGS) -Iinclude $< $(HOSTOBJS) -o $@ -Lbigstitcher_spark_amd -lbigstitch -lz -L/opt/rocm/lib -lamdhip64 -Wl,-rpath,'$$ORIGIN/..' -Wl,-rpath,/opt/rocm/lib
+	$(CXX_HOST) $(HOSTFLAGS) -Iinclude $< $(HOSTOBJS) -o $@ -Lbigstitcher_spark_amd -lbigstitch -lz -l:libzstd.so.1 -L/opt/rocm/lib -lamdhip64 -Wl,-rpath,'$$ORIGIN/..' -Wl,-rpath,/opt/rocm/lib
 
 clean-cli:
 	rm -f $(HOSTOBJS) $(BINDIR)/*

@@ -1550,7 +1550,10 @@ static int stitch_phaseA(bs_ctx *c, bs_slot *sl, const bs_pair_desc &pd,
         return BS_EINVAL;
       }
       sl->m[t][d] = (int)(size[d] / ds[d]);
-      if (sl->m[t][d] < 1) sl->m[t][d] = 1;
+      if (sl->m[t][d] < 1) { /* oracle: m = size//ds, empty = no pair */
+        c->err = "pair interval smaller than downsampling factor";
+        return BS_EINVAL;
+      }
     }
     if (dsall1) {
       sl->reg[t] = {vr.dptr, vr.dims[0], vr.dims[0] * vr.dims[1],

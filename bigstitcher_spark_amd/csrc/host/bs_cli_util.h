@@ -17,7 +17,10 @@
 namespace bscli {
 
 struct Args {
-  std::map<std::string, std::string> kv;
+  /* picocli semantics: repeated options accumulate (e.g. -ds 1,1,1
+   * -ds 2,2,1, -vi '0,0' -vi '0,1'); get() returns the LAST value for
+   * single-value lookups, getall() the full list in order. */
+  std::map<std::string, std::vector<std::string>> kv;
   std::vector<std::string> flags;
   bool parse(int argc, char **argv,
              const std::map<std::string, std::string> &aliases,
@@ -41,7 +44,7 @@ struct Args {
           fprintf(stderr, "missing value for %s\n", argv[i]);
           return false;
         }
-        kv[key] = argv[++i];
+        kv[key].push_back(argv[++i]);
       }
     }
     return true;
@@ -54,15 +57,19 @@ struct Args {
   }
   std::string get(const std::string &k, const std::string &dflt = "") const {
     auto it = kv.find(k);
-    return it == kv.end() ? dflt : it->second;
+    return it == kv.end() ? dflt : it->second.back();
+  }
+  std::vector<std::string> getall(const std::string &k) const {
+    auto it = kv.find(k);
+    return it == kv.end() ? std::vector<std::string>{} : it->second;
   }
   double getd(const std::string &k, double dflt) const {
     auto it = kv.find(k);
-    return it == kv.end() ? dflt : atof(it->second.c_str());
+    return it == kv.end() ? dflt : atof(it->second.back().c_str());
   }
   long getl(const std::string &k, long dflt) const {
     auto it = kv.find(k);
-    return it == kv.end() ? dflt : atol(it->second.c_str());
+    return it == kv.end() ? dflt : atol(it->second.back().c_str());
   }
 };
 

@@ -109,6 +109,37 @@ bool gzip_inflate(const unsigned char *in, size_t n, std::string *out,
   return true;
 }
 
+/* Zstandard via the system libzstd.so.1 (no dev header in this image:
+ * the stable simple API is declared here and resolved at link time,
+ * Makefile -l:libzstd.so.1). The reference's default container codec
+ * (CreateFusionContainer.java:71-73, n5-zstandard). */
+extern "C" {
+size_t ZSTD_compressBound(size_t srcSize);
+size_t ZSTD_compress(void *dst, size_t dstCap, const void *src,
+                     size_t srcSize, int level);
+size_t ZSTD_decompress(void *dst, size_t dstCap, const void *src,
+                       size_t srcSize);
+unsigned ZSTD_isError(size_t code);
+}
+
+bool zstd_compress(const std::string &in, std::string *out, int level) {
+  out->resize(ZSTD_compressBound(in.size()));
+  size_t n = ZSTD_compress(&(*out)[0], out->size(), in.data(), in.size(),
+                           level);
+  if (ZSTD_isError(n)) return false;
+  out->resize(n);
+  return true;
+}
+
+bool zstd_decompress(const unsigned char *in, size_t n, std::string *out,
+                     size_t expected) {
+  out->resize(expected);
+  size_t r = ZSTD_decompress(&(*out)[0], out->size(), in, n);
+  if (ZSTD_isError(r)) return false;
+  out->resize(r);
+  return true;
+}
+
 }  // namespace bsio
 
 namespace bsn5 {
@@ -183,6 +214,9 @@ bool Container::create_dataset(const std::string &name,
   if (a.compression == "gzip") {
     comp->obj["level"] = bsj::Value::mkint(-1);
     comp->obj["useZlib"] = bsj::Value::mkbool(false);
+  } else if (a.compression == "zstd") {
+    /* n5-zstandard ZstandardCompression attribute shape */
+    comp->obj["level"] = bsj::Value::mkint(a.level ? a.level : 3);
   }
   v->obj["compression"] = comp;
   return write_file(dir + "/attributes.json", bsj::dump(v));
@@ -206,6 +240,8 @@ bool Container::get_dataset_attrs(const std::string &name,
   for (auto &e : block->arr) out->block.push_back((int)e->inum);
   out->dtype = dt->str;
   out->compression = comp ? comp->str : "raw";
+  auto lvl = bsj::get_path(v, "compression/level");
+  out->level = lvl && lvl->is_int ? (int)lvl->inum : 0;
   return true;
 }
 
@@ -227,6 +263,9 @@ bool Container::write_block(const std::string &name,
   std::string body;
   if (a.compression == "gzip") {
     if (!gzip_deflate(payload, &body)) return false;
+  } else if (a.compression == "zstd") {
+    if (!bsio::zstd_compress(payload, &body, a.level ? a.level : 3))
+      return false;
   } else {
     body = payload;
   }
@@ -264,6 +303,10 @@ bool Container::read_block(const std::string &name, const DatasetAttrs &a,
   std::string payload;
   if (a.compression == "gzip") {
     if (!gzip_inflate(p + off, raw.size() - off, &payload, nelem * esz))
+      return false;
+  } else if (a.compression == "zstd") {
+    if (!bsio::zstd_decompress(p + off, raw.size() - off, &payload,
+                               nelem * esz))
       return false;
   } else {
     payload.assign((const char *)p + off, raw.size() - off);

@@ -7,9 +7,10 @@
  *     default-mode header: uint16 mode(0), uint16 ndim, uint32 dims[ndim]
  *     (big-endian), then the compressed payload of the block in
  *     column-major (dimension-0-fastest) element order, big-endian.
- * Codecs: raw and gzip (zlib). Zstd (the reference's container default,
- * CreateFusionContainer.java:71-73) is unavailable in this image (no
- * libzstd dev) — gzip is our default; readers of either work.
+ * Codecs: raw, gzip (zlib) and zstd (the reference's container default,
+ * CreateFusionContainer.java:71-73; linked against the system
+ * libzstd.so.1 via its stable simple API — no dev header in this image,
+ * so the four functions are declared locally in bs_n5.cpp).
  * uint8/uint16/float32 element types (the fusion output set). */
 #ifndef BS_N5_H
 #define BS_N5_H
@@ -28,6 +29,9 @@ bool gzip_deflate(const std::string &in, std::string *out, int level = -1,
                   bool raw_zlib = false);
 bool gzip_inflate(const unsigned char *in, size_t n, std::string *out,
                   size_t expected);
+bool zstd_compress(const std::string &in, std::string *out, int level = 3);
+bool zstd_decompress(const unsigned char *in, size_t n, std::string *out,
+                     size_t expected);
 }  // namespace bsio
 
 namespace bsn5 {
@@ -36,7 +40,8 @@ struct DatasetAttrs {
   std::vector<long long> dims;      /* dimension 0 fastest (x) */
   std::vector<int> block;
   std::string dtype;                /* "uint8"|"uint16"|"float32"|... */
-  std::string compression;          /* "raw"|"gzip" */
+  std::string compression;          /* "raw"|"gzip"|"zstd" */
+  int level = 0;                    /* codec level; 0 = codec default */
 };
 
 size_t dtype_size(const std::string &dtype);

@@ -1,5 +1,7 @@
 #include "bs_spimdata.h"
 
+#include <algorithm>
+#include <cmath>
 #include <cstdio>
 #include <cstring>
 #include <sstream>
@@ -107,6 +109,25 @@ bool SpimData::load(const std::string &path, std::string *err) {
       auto d = nums(sz->text);
       for (size_t i = 0; i < 3 && i < d.size(); ++i)
         s.dims[i] = (long long)d[i];
+    }
+    auto vx = vs->child("voxelSize");
+    if (vx) {
+      auto vsz = vx->child("size");
+      if (vsz) {
+        auto d = nums(vsz->text);
+        for (size_t i = 0; i < 3 && i < d.size(); ++i) s.voxel[i] = d[i];
+      }
+    }
+    auto at = vs->child("attributes");
+    if (at) {
+      auto rd = [&](const char *tag, int *dst) {
+        auto n = at->child(tag);
+        if (n) *dst = atoi(n->text.c_str());
+      };
+      rd("angle", &s.angle);
+      rd("tile", &s.tile);
+      rd("channel", &s.channel);
+      rd("illumination", &s.illumination);
     }
     setups.push_back(s);
   }
@@ -224,6 +245,84 @@ void SpimData::set_stitching_results(
     pr->add_text("Correlation", join(&e.r, 1));
     pr->add_text("Hash", join(&e.hash, 1));
   }
+}
+
+namespace {
+/* "0,1,2" -> set; empty string -> nullopt-like empty set + false */
+bool parse_idlist(const std::string &s, std::vector<int> *out) {
+  if (s.empty()) return false;
+  std::istringstream is(s);
+  std::string tok;
+  while (std::getline(is, tok, ','))
+    if (!tok.empty()) out->push_back(atoi(tok.c_str()));
+  return true;
+}
+bool contains(const std::vector<int> &v, int x) {
+  for (int e : v)
+    if (e == x) return true;
+  return false;
+}
+}  // namespace
+
+bool select_views(const SpimData &sd, const std::vector<std::string> &vi,
+                  const std::string &angle_ids, const std::string &tile_ids,
+                  const std::string &illum_ids,
+                  const std::string &channel_ids,
+                  const std::string &timepoint_ids,
+                  std::vector<ViewId> *out, std::string *err) {
+  out->clear();
+  if (!vi.empty()) {
+    /* explicit -vi 'tp,setup' list wins (Import.java:103-139) */
+    for (auto &s : vi) {
+      int tp, su;
+      if (sscanf(s.c_str(), "%d,%d", &tp, &su) != 2) {
+        if (err) *err = "bad -vi entry '" + s + "' (expected 'tp,setup')";
+        return false;
+      }
+      if (!sd.setup(su) || !sd.regs.count({tp, su})) {
+        if (err)
+          *err = "-vi view (" + std::to_string(tp) + "," +
+                 std::to_string(su) + ") not present in the XML";
+        return false;
+      }
+      out->push_back({tp, su});
+    }
+    return true;
+  }
+  std::vector<int> ang, til, ill, cha, tps;
+  bool f_ang = parse_idlist(angle_ids, &ang);
+  bool f_til = parse_idlist(tile_ids, &til);
+  bool f_ill = parse_idlist(illum_ids, &ill);
+  bool f_cha = parse_idlist(channel_ids, &cha);
+  bool f_tps = parse_idlist(timepoint_ids, &tps);
+  for (int tp : sd.timepoints) {
+    if (f_tps && !contains(tps, tp)) continue;
+    for (auto &s : sd.setups) {
+      if (!sd.regs.count({tp, s.id})) continue;
+      if (f_ang && !contains(ang, s.angle)) continue;
+      if (f_til && !contains(til, s.tile)) continue;
+      if (f_ill && !contains(ill, s.illumination)) continue;
+      if (f_cha && !contains(cha, s.channel)) continue;
+      out->push_back({tp, s.id});
+    }
+  }
+  if (out->empty() && err) *err = "no views selected";
+  return !out->empty();
+}
+
+double average_anisotropy(const SpimData &sd,
+                          const std::vector<ViewId> &views) {
+  double sum = 0;
+  int n = 0;
+  for (auto &v : views) {
+    const ViewSetup *s = sd.setup(v.second);
+    if (!s) continue;
+    double xy = std::min(s->voxel[0], s->voxel[1]);
+    if (xy <= 0) continue;
+    sum += s->voxel[2] / xy;
+    ++n;
+  }
+  return n ? sum / n : 1.0;
 }
 
 bsx::NodePtr make_dataset_xml(

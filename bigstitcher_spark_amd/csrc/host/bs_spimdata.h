@@ -36,6 +36,10 @@ struct ViewSetup {
   int id = 0;
   long long dims[3] = {0, 0, 0}; /* x,y,z */
   std::string name;
+  double voxel[3] = {1.0, 1.0, 1.0}; /* voxelSize/size, x y z */
+  /* setup attribute ids (<attributes>: angle/tile/channel/illumination);
+   * -1 = absent in the XML */
+  int angle = -1, tile = -1, channel = -1, illumination = -1;
 };
 
 struct StitchEntry {
@@ -67,6 +71,26 @@ struct SpimData {
   static double calculate_hash(const std::array<double, 12> &a,
                                const std::array<double, 12> &b);
 };
+
+/* View selection per the reference's Import.getViewIds
+ * (util/Import.java:94-204 + AbstractSelectableViews.java:38-54):
+ * if `vi` entries ("tp,setup") are given they win; otherwise every view
+ * (tp x setup with a registration) is filtered by the optional id lists
+ * (empty list = no filter on that attribute). Unknown -vi entries are
+ * reported via *err. */
+bool select_views(const SpimData &sd, const std::vector<std::string> &vi,
+                  const std::string &angle_ids, const std::string &tile_ids,
+                  const std::string &illum_ids,
+                  const std::string &channel_ids,
+                  const std::string &timepoint_ids,
+                  std::vector<ViewId> *out, std::string *err);
+
+/* [PIN-ANISO] restatement of mvrecon
+ * TransformationTools.getAverageAnisotropyFactor (called at reference
+ * CreateFusionContainer.java:195; artifact un-vendored): mean over the
+ * selected views of voxelZ / min(voxelX, voxelY). */
+double average_anisotropy(const SpimData &sd,
+                          const std::vector<ViewId> &views);
 
 /* Create a minimal dataset.xml + registrations for tests/tools. */
 bsx::NodePtr make_dataset_xml(const std::string &n5_rel,

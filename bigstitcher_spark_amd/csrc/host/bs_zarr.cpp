@@ -58,10 +58,16 @@ bool Container::create_array(const std::string &name, const ArrayAttrs &a) {
   z->obj["fill_value"] = bsj::Value::mkint(0);
   z->obj["filters"] = bsj::Value::mknull();
   z->obj["dimension_separator"] = bsj::Value::mkstr(".");
-  if (a.gzip) {
+  std::string codec = a.codec.empty() ? (a.gzip ? "gzip" : "raw") : a.codec;
+  if (codec == "gzip") {
     auto comp = bsj::Value::mkobj();
     comp->obj["id"] = bsj::Value::mkstr("gzip");
-    comp->obj["level"] = bsj::Value::mkint(1);
+    comp->obj["level"] = bsj::Value::mkint(a.level ? a.level : 1);
+    z->obj["compressor"] = comp;
+  } else if (codec == "zstd") { /* numcodecs zstd */
+    auto comp = bsj::Value::mkobj();
+    comp->obj["id"] = bsj::Value::mkstr("zstd");
+    comp->obj["level"] = bsj::Value::mkint(a.level ? a.level : 3);
     z->obj["compressor"] = comp;
   } else {
     z->obj["compressor"] = bsj::Value::mknull();
@@ -84,7 +90,16 @@ bool Container::get_array_attrs(const std::string &name,
   for (auto &e : ch->arr) out->chunks.push_back((int)e->inum);
   out->dtype = dt->str;
   auto comp = bsj::get_path(z, "compressor");
-  out->gzip = comp && comp->type == bsj::Value::OBJ;
+  if (comp && comp->type == bsj::Value::OBJ) {
+    auto id = bsj::get_path(comp, "id");
+    out->codec = id ? id->str : "gzip";
+    auto lvl = bsj::get_path(comp, "level");
+    out->level = lvl && lvl->is_int ? (int)lvl->inum : 0;
+  } else {
+    out->codec = "raw";
+    out->level = 0;
+  }
+  out->gzip = out->codec == "gzip";
   return true;
 }
 
@@ -118,9 +133,14 @@ bool Container::write_chunk(const std::string &name, const ArrayAttrs &a,
     memcpy(&payload[off * esz], (const char *)data + r * row, row);
   }
   std::string body;
-  if (a.gzip) {
+  std::string codec = a.codec.empty() ? (a.gzip ? "gzip" : "raw") : a.codec;
+  if (codec == "gzip") {
     /* zarr "gzip" codec = gzip container */
-    if (!gzip_deflate(payload, &body, 1, false)) return false;
+    if (!gzip_deflate(payload, &body, a.level ? a.level : 1, false))
+      return false;
+  } else if (codec == "zstd") {
+    if (!bsio::zstd_compress(payload, &body, a.level ? a.level : 3))
+      return false;
   } else {
     body = payload;
   }

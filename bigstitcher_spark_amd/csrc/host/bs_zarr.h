@@ -28,7 +28,9 @@ struct ArrayAttrs {
   std::vector<long long> shape;  /* C-order, slowest first */
   std::vector<int> chunks;
   std::string dtype;             /* "<u2" | "<f4" | "|u1" */
-  bool gzip = true;
+  std::string codec = "gzip";    /* "raw"|"gzip"|"zstd" (numcodecs ids) */
+  int level = 0;                 /* 0 = codec default */
+  bool gzip = true;              /* legacy view of codec (kept in sync) */
 };
 
 size_t dtype_size(const std::string &dtype);

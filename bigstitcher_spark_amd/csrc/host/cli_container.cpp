@@ -1,15 +1,25 @@
 /* `create-fusion-container` — drop-in for the reference's
  * CreateFusionContainer (plain host tool, no Spark/GPU; reference
- * CreateFusionContainer.java). Creates the output N5 container with the
- * exact Bigstitcher-Spark root-attribute contract the fusion step
- * reads back (written :302-320, read SparkAffineFusion.java:239-307):
+ * CreateFusionContainer.java). Creates the output N5/OME-ZARR container
+ * with the exact Bigstitcher-Spark root-attribute contract the fusion
+ * step reads back (written :302-320, read SparkAffineFusion.java:239-307):
  * FusionFormat, InputXML, NumTimepoints, NumChannels, Boundingbox_min/
  * max, PreserveAnisotropy [,AnisotropyFactor], DataType, BlockSize,
- * [Min/MaxIntensity], MultiResolutionInfos. Datasets are the plain-N5
- * per-(channel,timepoint) layout "ch{c}tp{t}/s{l}" (:490-516).
- * Round-1 scope: N5 only (no ZARR/HDF5), level s0 only (the pyramid is
- * SURVEY.md §8(f) row 1), no anisotropy split. */
+ * [Min/MaxIntensity], MultiResolutionInfos. Datasets: plain-N5
+ * per-(channel,timepoint) "ch{c}tp{t}/s{l}" (:490-516) or one 5-D
+ * OME-ZARR array per level (:331-389).
+ *
+ * Flag surface mirrors the reference (:64-117): -o/--outputPath,
+ * -s/--storage (default ZARR), -c/--compression (default Zstandard),
+ * -cl/--compressionLevel, -d/--dataType (default FLOAT32), --blockSize,
+ * --minIntensity/--maxIntensity, --multiRes, repeated
+ * -ds/--downsampling (";"-splittable), --preserveAnisotropy,
+ * --anisotropyFactor, plus view-selection (-vi/--angleId/...).
+ * Unsupported storage/codec values exit with an explicit error.
+ * HDF5/--bdv variants are not built (documented gap). */
+#include <cmath>
 #include <cstdio>
+#include <set>
 
 #include "bs_cli_util.h"
 #include "bs_n5.h"
@@ -18,16 +28,32 @@
 
 int main(int argc, char **argv) {
   bscli::Args args;
-  std::map<std::string, std::string> alias = {{"-x", "--xml"},
-                                              {"-o", "--n5Path"}};
-  if (!args.parse(argc, argv, alias, {"preserveAnisotropy"}) ||
-      !args.has("xml") || !args.has("n5Path")) {
+  std::map<std::string, std::string> alias = {
+      {"-x", "--xml"},         {"-o", "--outputPath"},
+      {"--n5Path", "--outputPath"}, /* pre-round-2 spelling */
+      {"-s", "--storage"},     {"-c", "--compression"},
+      {"-cl", "--compressionLevel"}, {"-d", "--dataType"},
+      {"-ds", "--downsampling"},
+      {"--downsamplings", "--downsampling"}, /* pre-round-2 spelling */
+      {"-tp", "--numTimepoints"}, {"-ch", "--numChannels"},
+      {"-vi", "--vi"}};
+  if (!args.parse(argc, argv, alias,
+                  {"preserveAnisotropy", "multiRes", "dryRun", "bdv"}) ||
+      !args.has("xml") || !args.has("outputPath")) {
     fprintf(stderr,
-            "usage: create-fusion-container -x dataset.xml -o out.n5 "
-            "[--blockSize 128,128,128] [--dataType UINT16|UINT8|FLOAT32] "
+            "usage: create-fusion-container -x dataset.xml -o out.zarr "
+            "[-s ZARR|N5] [-c Zstandard|Gzip|Raw] [-cl level] "
+            "[--blockSize 128,128,128] [-d FLOAT32|UINT16|UINT8] "
             "[--minIntensity v --maxIntensity v] [--bbMin x,y,z --bbMax "
-            "x,y,z] [--compression GZIP|RAW] [--storage N5|ZARR] "
-            "[--downsamplings \"1,1,1;2,2,2;4,4,4\"]\n");
+            "x,y,z] [--multiRes | -ds 1,1,1 -ds 2,2,1 ...] "
+            "[--preserveAnisotropy [--anisotropyFactor f]] "
+            "[-vi 'tp,setup' ... | --angleId/--tileId/--channelId/"
+            "--illuminationId/--timepointId '0,1,..']\n");
+    return 2;
+  }
+  if (args.has("bdv")) {
+    fprintf(stderr, "--bdv (BDV N5/HDF5 container variants) is not "
+                    "supported by this build\n");
     return 2;
   }
   bssd::SpimData sd;
@@ -36,15 +62,38 @@ int main(int argc, char **argv) {
     fprintf(stderr, "error: %s\n", err.c_str());
     return 1;
   }
+  std::vector<bssd::ViewId> views;
+  if (!bssd::select_views(sd, args.getall("vi"), args.get("angleId"),
+                          args.get("tileId"), args.get("illuminationId"),
+                          args.get("channelId"), args.get("timepointId"),
+                          &views, &err)) {
+    fprintf(stderr, "error: %s\n", err.c_str());
+    return 1;
+  }
   auto bs = bscli::parse_ints(args.get("blockSize", "128,128,128"));
-  std::string dt = args.get("dataType", "UINT16");
+  std::string dt = args.get("dataType", "FLOAT32"); /* reference default */
+  if (dt != "UINT8" && dt != "UINT16" && dt != "FLOAT32") {
+    fprintf(stderr, "unsupported --dataType %s\n", dt.c_str());
+    return 2;
+  }
   std::string n5dt = dt == "UINT8" ? "uint8"
                      : dt == "FLOAT32" ? "float32" : "uint16";
-  std::string comp =
-      args.get("compression", "GZIP") == "RAW" ? "raw" : "gzip";
+  /* reference default codec = Zstandard (CreateFusionContainer.java:71-73) */
+  std::string cname = args.get("compression", "Zstandard");
+  std::string comp = cname == "Raw" ? "raw"
+                     : cname == "Gzip" ? "gzip"
+                     : cname == "Zstandard" ? "zstd" : "";
+  if (comp.empty()) {
+    fprintf(stderr,
+            "unsupported --compression %s (supported: Zstandard, Gzip, "
+            "Raw)\n",
+            cname.c_str());
+    return 2;
+  }
+  int clevel = (int)args.getl("compressionLevel", 0); /* 0 = codec dflt */
 
-  /* bounding box: explicit or the union of transformed view bboxes
-   * (the reference's default "estimate bounding box", :122-211) */
+  /* bounding box: explicit or the union of the SELECTED transformed view
+   * bboxes (the reference's "estimate bounding box" default, :122-211) */
   long long bbmin[3], bbmax[3];
   if (args.has("bbMin") && args.has("bbMax")) {
     auto mn = bscli::parse_ints(args.get("bbMin"));
@@ -55,12 +104,12 @@ int main(int argc, char **argv) {
     }
   } else {
     double lo[3] = {1e300, 1e300, 1e300}, hi[3] = {-1e300, -1e300, -1e300};
-    int tp0 = sd.timepoints.empty() ? 0 : sd.timepoints[0];
-    for (auto &s : sd.setups) {
-      auto r = sd.regs.find({tp0, s.id});
-      if (r == sd.regs.end()) continue;
+    for (auto &v : views) {
+      auto r = sd.regs.find(v);
+      const bssd::ViewSetup *s = sd.setup(v.second);
+      if (r == sd.regs.end() || !s) continue;
       double l[3], h[3];
-      bscli::tbbox(r->second, s.dims, l, h);
+      bscli::tbbox(r->second, s->dims, l, h);
       for (int d = 0; d < 3; ++d) {
         lo[d] = std::min(lo[d], l[d]);
         hi[d] = std::max(hi[d], h[d]);
@@ -71,15 +120,41 @@ int main(int argc, char **argv) {
       bbmax[d] = (long long)std::ceil(hi[d]);
     }
   }
+  /* anisotropy (reference :189-211): factor from data when not given;
+   * the container's bbox z is divided by the factor (floor/ceil) and the
+   * fusion step adjusts all transforms by the same factor */
+  const bool preserveAniso = args.has("preserveAnisotropy");
+  double anisoF = args.getd("anisotropyFactor", NAN);
+  if (preserveAniso) {
+    if (std::isnan(anisoF)) {
+      anisoF = bssd::average_anisotropy(sd, views);
+      printf("Anisotropy factor [computed from data]: %g\n", anisoF);
+    } else {
+      printf("Anisotropy factor [provided]: %g\n", anisoF);
+    }
+    bbmin[2] = (long long)std::llround(std::floor(bbmin[2] / anisoF));
+    bbmax[2] = (long long)std::llround(std::ceil(bbmax[2] / anisoF));
+  }
   long long dims[3] = {bbmax[0] - bbmin[0] + 1, bbmax[1] - bbmin[1] + 1,
                        bbmax[2] - bbmin[2] + 1};
-  int numTp = (int)sd.timepoints.size(), numCh = 1;
-  const bool zarr = args.get("storage", "N5") == "ZARR";
+  int numTp = (int)args.getl(
+      "numTimepoints",
+      (long)std::max<size_t>(1, sd.timepoints.size()));
+  int numCh = (int)args.getl("numChannels", 1);
+  std::string storage = args.get("storage", "ZARR"); /* reference default */
+  if (storage != "ZARR" && storage != "N5") {
+    fprintf(stderr,
+            "unsupported --storage %s (supported: ZARR, N5; HDF5 is not "
+            "built)\n",
+            storage.c_str());
+    return 2;
+  }
+  const bool zarr = storage == "ZARR";
 
-  bsn5::Container n5(args.get("n5Path"));
-  bszarr::Container zr(args.get("n5Path"));
+  bsn5::Container n5(args.get("outputPath"));
+  bszarr::Container zr(args.get("outputPath"));
   if (zarr ? !zr.create() : !n5.create()) {
-    fprintf(stderr, "cannot create %s\n", args.get("n5Path").c_str());
+    fprintf(stderr, "cannot create %s\n", args.get("outputPath").c_str());
     return 1;
   }
   auto set = [&](const std::string &k, bsj::ValuePtr v) {
@@ -96,33 +171,61 @@ int main(int argc, char **argv) {
       bsj::Value::mkints(std::vector<long long>{bbmin[0], bbmin[1], bbmin[2]}));
   set("Boundingbox_max",
       bsj::Value::mkints(std::vector<long long>{bbmax[0], bbmax[1], bbmax[2]}));
-  set("PreserveAnisotropy", bsj::Value::mkbool(false));
+  set("PreserveAnisotropy", bsj::Value::mkbool(preserveAniso));
+  if (preserveAniso)
+    set("AnisotropyFactor", bsj::Value::mknum(anisoF));
   set("DataType", bsj::Value::mkstr(dt));
   set("BlockSize", bsj::Value::mkints(std::vector<long long>{bs[0], bs[1], bs[2]}));
   if (args.has("minIntensity") && args.has("maxIntensity")) {
     set("MinIntensity", bsj::Value::mknum(args.getd("minIntensity", 0)));
     set("MaxIntensity", bsj::Value::mknum(args.getd("maxIntensity", 65535)));
   }
-  /* pyramid ladder (reference
-   * CreateFusionContainer.java:260-273 estimates it — here explicit via
-   * --downsamplings, default s0 only) */
+  /* pyramid ladder (reference :260-273):
+   * repeated -ds (each entry ";"-splittable, picocli split=";"), or
+   * --multiRes [PIN-MULTIRES]: restating ExportN5Api's estimate — halve
+   * every axis whose CURRENT level extent exceeds its block size until
+   * all fit in one block or 8 levels, never downsampling an axis below
+   * one block. The artifact's exact ladder is un-vendored; the contract
+   * consumed downstream is only "ladder[0]==1,1,1, each level divides
+   * the next" which any reader of MultiResolutionInfos re-reads. */
   std::vector<std::array<long long, 3>> ladder;
   {
-    std::string spec = args.get("downsamplings", "1,1,1");
-    std::string cur;
-    for (char ch2 : spec + ";") {
-      if (ch2 == ';') {
-        auto f = bscli::parse_ints(cur);
-        if (f.size() == 3) ladder.push_back({f[0], f[1], f[2]});
-        cur.clear();
-      } else {
-        cur += ch2;
+    std::vector<std::string> specs = args.getall("downsampling");
+    std::string joined;
+    for (auto &s : specs) joined += (joined.empty() ? "" : ";") + s;
+    if (!joined.empty()) {
+      std::string cur;
+      for (char ch2 : joined + ";") {
+        if (ch2 == ';') {
+          auto f = bscli::parse_ints(cur);
+          if (f.size() == 3) ladder.push_back({f[0], f[1], f[2]});
+          cur.clear();
+        } else {
+          cur += ch2;
+        }
       }
-    }
-    if (ladder.empty() || ladder[0][0] != 1 || ladder[0][1] != 1 ||
-        ladder[0][2] != 1) {
-      fprintf(stderr, "--downsamplings must start with 1,1,1\n");
-      return 2;
+      if (ladder.empty() || ladder[0][0] != 1 || ladder[0][1] != 1 ||
+          ladder[0][2] != 1) {
+        fprintf(stderr, "-ds ladder must start with 1,1,1\n");
+        return 2;
+      }
+    } else if (args.has("multiRes")) {
+      long long f[3] = {1, 1, 1};
+      ladder.push_back({1, 1, 1});
+      for (int l = 0; l < 7; ++l) {
+        bool any = false;
+        long long nf[3];
+        for (int d = 0; d < 3; ++d) {
+          long long ext = (dims[d] + f[d] - 1) / f[d];
+          nf[d] = ext > bs[d] ? f[d] * 2 : f[d];
+          any = any || nf[d] != f[d];
+        }
+        if (!any) break;
+        ladder.push_back({nf[0], nf[1], nf[2]});
+        f[0] = nf[0]; f[1] = nf[1]; f[2] = nf[2];
+      }
+    } else {
+      ladder.push_back({1, 1, 1});
     }
   }
   auto mri_all = bsj::Value::mkarr();
@@ -142,6 +245,8 @@ int main(int argc, char **argv) {
                   (dims[0] + ladder[l][0] - 1) / ladder[l][0]};
       za.chunks = {1, 1, (int)bs[2], (int)bs[1], (int)bs[0]};
       za.dtype = zdt;
+      za.codec = comp;
+      za.level = clevel;
       za.gzip = comp == "gzip";
       if (!zr.create_array(dsname, za)) {
         fprintf(stderr, "cannot create array %s\n", dsname);
@@ -210,6 +315,7 @@ int main(int argc, char **argv) {
           da.block = {(int)bs[0], (int)bs[1], (int)bs[2]};
           da.dtype = n5dt;
           da.compression = comp;
+          da.level = clevel;
           if (!n5.create_dataset(dsname, da)) {
             fprintf(stderr, "cannot create dataset %s\n", dsname);
             return 1;
@@ -230,8 +336,9 @@ int main(int argc, char **argv) {
   }
   set("MultiResolutionInfos", mri_all);
   printf("created %s: %d tp x %d ch, bbox [%lld,%lld,%lld]..[%lld,%lld,%lld]"
-         ", %s %s\n",
-         args.get("n5Path").c_str(), numTp, numCh, bbmin[0], bbmin[1],
-         bbmin[2], bbmax[0], bbmax[1], bbmax[2], dt.c_str(), comp.c_str());
+         ", %zu level(s), %s %s\n",
+         args.get("outputPath").c_str(), numTp, numCh, bbmin[0], bbmin[1],
+         bbmin[2], bbmax[0], bbmax[1], bbmax[2], ladder.size(), dt.c_str(),
+         comp.c_str());
   return 0;
 }

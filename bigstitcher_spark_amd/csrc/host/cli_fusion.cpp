@@ -82,14 +82,22 @@ static bool load_coefficients(const bsn5::Container &cn,
 
 int main(int argc, char **argv) {
   bscli::Args args;
-  std::map<std::string, std::string> alias = {{"-x", "--xml"},
-                                              {"-o", "--n5Path"}};
-  if (!args.parse(argc, argv, alias, {"masks"}) || !args.has("n5Path")) {
+  std::map<std::string, std::string> alias = {
+      {"-x", "--xml"},
+      {"-o", "--n5Path"},
+      {"-f", "--fusion"}, /* reference spelling, SparkAffineFusion.java:124 */
+      {"--fusionType", "--fusion"}, /* pre-round-2 spelling */
+      {"-vi", "--vi"}};
+  if (!args.parse(argc, argv, alias, {"masks", "prefetch", "dryRun"}) ||
+      !args.has("n5Path")) {
     fprintf(stderr,
             "usage: affine-fusion -o out.n5 [-x dataset.xml] "
-            "[--fusionType AVG_BLEND|AVG|MAX_INTENSITY] "
+            "[-f AVG_BLEND|AVG|MAX_INTENSITY|LOWEST_VIEWID_WINS|"
+            "HIGHEST_VIEWID_WINS|CLOSEST_PIXEL_WINS] "
             "[--blendingRange 40] [--blendingBorder 0] [--device N] "
-            "[--batchSize 64] [--masks] [--maskOffset 0.0,0.0,0.0] "
+            "[--masks] [--maskOffset 0.0,0.0,0.0] "
+            "[-vi 'tp,setup' ... | --angleId/--tileId/--channelId/"
+            "--illuminationId/--timepointId '0,1,..'] "
             "[--intensityN5Path coeff.n5 [--intensityN5Group g] "
             "[--intensityN5Dataset intensity]]\n");
     return 2;
@@ -115,7 +123,17 @@ int main(int argc, char **argv) {
     auto x = geta("InputXML");
     if (x) xml = x->str;
   }
-  const bool zarr = fmt->str.find("ZARR") != std::string::npos;
+  /* exact-format whitelist: BDV-prefixed and HDF5 containers have different
+   * dataset layouts (reference SparkAffineFusion.java:239-307) — refuse
+   * them explicitly rather than guessing a layout */
+  if (fmt->str != "N5" && fmt->str != "OME-ZARR") {
+    fprintf(stderr,
+            "unsupported FusionFormat '%s' (supported: N5, OME-ZARR; "
+            "BDV/* and HDF5 containers are not supported by this build)\n",
+            fmt->str.c_str());
+    return 1;
+  }
+  const bool zarr = fmt->str == "OME-ZARR";
   auto bbmin_a = geta("Boundingbox_min"), bbmax_a = geta("Boundingbox_max");
   auto bs_a = geta("BlockSize");
   auto dt_a = geta("DataType");
@@ -135,13 +153,39 @@ int main(int argc, char **argv) {
   std::string dt = dt_a->str;
   int out_dtype = dt == "UINT8" ? BS_OUT_UINT8
                   : dt == "FLOAT32" ? BS_OUT_FLOAT32 : BS_OUT_UINT16;
+  /* anisotropy (reference SparkAffineFusion.java:271-272, :486-491):
+   * when the container was created with --preserveAnisotropy, every
+   * registration is adjusted by pre-concatenating a world-z scale of
+   * 1/AnisotropyFactor ([PIN-ANISO] restatement of mvrecon
+   * TransformVirtual.adjustAllTransforms(views, regs, factor, NaN) —
+   * artifact un-vendored; the container's bbox z was divided by the
+   * same factor at create time, keeping the two sides consistent). */
+  double anisoF = 1.0;
+  {
+    auto pa = geta("PreserveAnisotropy");
+    if (pa && pa->type == bsj::Value::BOOL && pa->b) {
+      auto af = geta("AnisotropyFactor");
+      if (!af) {
+        fprintf(stderr,
+                "container has PreserveAnisotropy=true but no "
+                "AnisotropyFactor attribute\n");
+        return 1;
+      }
+      anisoF = af->is_int ? (double)af->inum : af->num;
+      if (!(anisoF > 0)) {
+        fprintf(stderr, "bad AnisotropyFactor %g\n", anisoF);
+        return 1;
+      }
+      printf("preserving anisotropy: factor %g\n", anisoF);
+    }
+  }
   double minI = 0, maxI = 65535;
   auto minI_a = geta("MinIntensity"), maxI_a = geta("MaxIntensity");
   if (minI_a) minI = minI_a->is_int ? minI_a->inum : minI_a->num;
   if (maxI_a) maxI = maxI_a->is_int ? maxI_a->inum : maxI_a->num;
   if (out_dtype == BS_OUT_UINT8 && !maxI_a) maxI = 255;
 
-  std::string ft = args.get("fusionType", "AVG_BLEND");
+  std::string ft = args.get("fusion", "AVG_BLEND");
   int fusion_type =
       ft == "AVG" ? BS_FUSION_AVG
       : ft == "MAX_INTENSITY" ? BS_FUSION_MAX_INTENSITY
@@ -159,6 +203,15 @@ int main(int argc, char **argv) {
             err.c_str());
     return 1;
   }
+  std::vector<bssd::ViewId> selected;
+  if (!bssd::select_views(sd, args.getall("vi"), args.get("angleId"),
+                          args.get("tileId"), args.get("illuminationId"),
+                          args.get("channelId"), args.get("timepointId"),
+                          &selected, &err)) {
+    fprintf(stderr, "error: %s\n", err.c_str());
+    return 1;
+  }
+  std::set<std::pair<int, int>> selset(selected.begin(), selected.end());
   bsn5::Container in_n5(sd.n5_path);
   bs_ctx *ctx = nullptr;
   if (bs_ctx_create(&ctx, (int)args.getl("device", 0)) != BS_OK) {
@@ -175,6 +228,7 @@ int main(int argc, char **argv) {
     for (auto &s : sd.setups) {
       auto r = sd.regs.find({tp, s.id});
       if (r == sd.regs.end()) continue;
+      if (!selset.count({tp, s.id})) continue;
       std::vector<uint16_t> vox;
       std::vector<long long> vdims;
       if (!in_n5.read_volume_u16(bssd::SpimData::image_dataset(s.id, tp),
@@ -189,8 +243,12 @@ int main(int argc, char **argv) {
       }
       bs_fuse_view fv{};
       fv.view_id = s.id;
-      /* world coords; bs_fuse_volume shifts by -vol_min itself */
+      /* world coords; bs_fuse_volume shifts by -vol_min itself.
+       * [PIN-ANISO]: world z divided by the anisotropy factor
+       * (S(1,1,1/f) pre-concatenated, :486-491) */
       for (int i = 0; i < 12; ++i) fv.affine[i] = r->second[i];
+      if (anisoF != 1.0)
+        for (int i = 8; i < 12; ++i) fv.affine[i] /= anisoF;
       for (int d2 = 0; d2 < 3; ++d2) {
         fv.blend_border[d2] = bborder;
         fv.blend_range[d2] = brange;

@@ -25,7 +25,8 @@ using bscli::M34;
 int main(int argc, char **argv) {
   bscli::Args args;
   std::map<std::string, std::string> alias = {
-      {"-x", "--xml"}, {"-ds", "--downsampling"}, {"-p", "--peaksToCheck"}};
+      {"-x", "--xml"}, {"-ds", "--downsampling"}, {"-p", "--peaksToCheck"},
+      {"-vi", "--vi"}};
   if (!args.parse(argc, argv, alias,
                   {"disableSubpixelResolution", "dryRun"}) ||
       !args.has("xml")) {
@@ -34,7 +35,10 @@ int main(int argc, char **argv) {
             "[--minR 0.3] [--maxR 1.0] [--maxShiftX px] [--maxShiftY px] "
             "[--maxShiftZ px] [--maxShiftTotal px] "
             "[--disableSubpixelResolution] [--channelCombine AVERAGE] "
-            "[--illumCombine PICK_BRIGHTEST] [--device N] [--dryRun]\n");
+            "[--illumCombine PICK_BRIGHTEST] "
+            "[-vi 'tp,setup' ... | --angleId/--tileId/--channelId/"
+            "--illuminationId/--timepointId '0,1,..'] [--device N] "
+            "[--dryRun]\n");
     return 2;
   }
   if (args.get("channelCombine", "AVERAGE") != "AVERAGE" ||
@@ -61,13 +65,27 @@ int main(int argc, char **argv) {
   }
   printf("stitching: %zu setups, %zu timepoints, container %s\n",
          sd.setups.size(), sd.timepoints.size(), sd.n5_path.c_str());
-
-  bsn5::Container n5(sd.n5_path);
-  bs_ctx *ctx = nullptr;
-  if (bs_ctx_create(&ctx, (int)args.getl("device", 0)) != BS_OK) {
-    fprintf(stderr, "error: %s\n", bs_last_error(nullptr));
+  std::vector<bssd::ViewId> selected;
+  if (!bssd::select_views(sd, args.getall("vi"), args.get("angleId"),
+                          args.get("tileId"), args.get("illuminationId"),
+                          args.get("channelId"), args.get("timepointId"),
+                          &selected, &err)) {
+    fprintf(stderr, "error: %s\n", err.c_str());
     return 1;
   }
+  std::set<std::pair<int, int>> selset(selected.begin(), selected.end());
+
+  bsn5::Container n5(sd.n5_path);
+  /* ctx created lazily: --dryRun only enumerates pairs (no GPU) */
+  bs_ctx *ctx = nullptr;
+  auto ensure_ctx = [&]() -> bool {
+    if (ctx) return true;
+    if (bs_ctx_create(&ctx, (int)args.getl("device", 0)) != BS_OK) {
+      fprintf(stderr, "error: %s\n", bs_last_error(nullptr));
+      return false;
+    }
+    return true;
+  };
 
   std::vector<bssd::StitchEntry> entries;
   for (int tp : sd.timepoints) {
@@ -103,6 +121,8 @@ int main(int argc, char **argv) {
     for (size_t i = 0; i < sd.setups.size(); ++i) {
       for (size_t j = i + 1; j < sd.setups.size(); ++j) {
         const auto &A = sd.setups[i], &B = sd.setups[j];
+        if (!selset.count({tp, A.id}) || !selset.count({tp, B.id}))
+          continue;
         auto ra = sd.regs.find({tp, A.id}), rb = sd.regs.find({tp, B.id});
         if (ra == sd.regs.end() || rb == sd.regs.end()) continue;
         double loA[3], hiA[3], loB[3], hiB[3];
@@ -163,6 +183,7 @@ int main(int argc, char **argv) {
     }
     printf("timepoint %d: %zu overlapping pairs\n", tp, plans.size());
     if (args.has("dryRun") || plans.empty()) continue;
+    if (!ensure_ctx()) return 1;
 
     std::vector<bs_pair_desc> pds;
     for (auto &pp : plans) {

@@ -3,6 +3,7 @@ containers for the CLI tools and to verify their output — an independent
 implementation of the same public N5 spec as csrc/host/bs_n5.cpp (so the
 C++ and python sides cross-check each other)."""
 
+import ctypes
 import gzip
 import json
 import os
@@ -11,6 +12,45 @@ import struct
 import numpy as np
 
 _DTYPES = {"uint8": np.uint8, "uint16": np.uint16, "float32": np.float32}
+
+_zstd = None
+
+
+def _zstd_lib():
+    """libzstd via ctypes (no python zstd module in this image); the
+    same system library the C++ side links (-l:libzstd.so.1)."""
+    global _zstd
+    if _zstd is None:
+        _zstd = ctypes.CDLL("libzstd.so.1")
+        _zstd.ZSTD_compressBound.restype = ctypes.c_size_t
+        _zstd.ZSTD_compressBound.argtypes = [ctypes.c_size_t]
+        _zstd.ZSTD_compress.restype = ctypes.c_size_t
+        _zstd.ZSTD_compress.argtypes = [ctypes.c_char_p, ctypes.c_size_t,
+                                        ctypes.c_char_p, ctypes.c_size_t,
+                                        ctypes.c_int]
+        _zstd.ZSTD_decompress.restype = ctypes.c_size_t
+        _zstd.ZSTD_decompress.argtypes = [ctypes.c_char_p, ctypes.c_size_t,
+                                          ctypes.c_char_p, ctypes.c_size_t]
+        _zstd.ZSTD_isError.restype = ctypes.c_uint
+        _zstd.ZSTD_isError.argtypes = [ctypes.c_size_t]
+    return _zstd
+
+
+def zstd_compress(data, level=3):
+    z = _zstd_lib()
+    bound = z.ZSTD_compressBound(len(data))
+    buf = ctypes.create_string_buffer(bound)
+    n = z.ZSTD_compress(buf, bound, data, len(data), level)
+    assert not z.ZSTD_isError(n)
+    return buf.raw[:n]
+
+
+def zstd_decompress(data, expected):
+    z = _zstd_lib()
+    buf = ctypes.create_string_buffer(expected)
+    n = z.ZSTD_decompress(buf, expected, data, len(data))
+    assert not z.ZSTD_isError(n), "zstd decompress failed"
+    return buf.raw[:n]
 
 
 def write_dataset(root, name, arr_zyx, block_xyz, compression="gzip"):
@@ -23,6 +63,8 @@ def write_dataset(root, name, arr_zyx, block_xyz, compression="gzip"):
     comp = {"type": compression}
     if compression == "gzip":
         comp.update(level=-1, useZlib=False)
+    elif compression == "zstd":
+        comp.update(level=3)
     with open(os.path.join(ds, "attributes.json"), "w") as f:
         json.dump({"dimensions": [nx, ny, nz],
                    "blockSize": list(block_xyz),
@@ -37,6 +79,8 @@ def write_dataset(root, name, arr_zyx, block_xyz, compression="gzip"):
                 payload = blk.astype(blk.dtype.newbyteorder(">")).tobytes()
                 if compression == "gzip":
                     payload = gzip.compress(payload)
+                elif compression == "zstd":
+                    payload = zstd_compress(payload)
                 hdr = struct.pack(">HH", 0, 3) + struct.pack(
                     ">III", cx, cy, cz)
                 d = os.path.join(ds, str(gx), str(gy))
@@ -67,6 +111,9 @@ def read_dataset(root, name):
                 body = raw[4 + 4 * nd:]
                 if comp == "gzip":
                     body = gzip.decompress(body)
+                elif comp == "zstd":
+                    nel = cx * cy * cz
+                    body = zstd_decompress(body, nel * dt.itemsize)
                 blk = np.frombuffer(body, dt.newbyteorder(">")).reshape(
                     cz, cy, cx).astype(dt)
                 out[gz * bz:gz * bz + cz, gy * by:gy * by + cy,
@@ -101,14 +148,29 @@ DATASET_XML = """<?xml version="1.0" encoding="UTF-8"?>
 
 
 def make_dataset_xml(path, n5_rel, setups):
-    """setups: list of dicts {id, dims (x,y,z), pos (x,y,z) translation}."""
+    """setups: list of dicts {id, dims (x,y,z), pos (x,y,z) translation,
+    optional voxel (x,y,z sizes), optional attrs {angle,tile,channel,
+    illumination}}."""
     s_xml, r_xml = "", ""
     for s in setups:
+        extra = ""
+        if "voxel" in s:
+            v = s["voxel"]
+            extra += (
+                f"        <voxelSize>\n          <unit>um</unit>\n"
+                f"          <size>{v[0]} {v[1]} {v[2]}</size>\n"
+                f"        </voxelSize>\n"
+            )
+        if "attrs" in s:
+            extra += "        <attributes>\n"
+            for k, v in s["attrs"].items():
+                extra += f"          <{k}>{v}</{k}>\n"
+            extra += "        </attributes>\n"
         s_xml += (
             f"      <ViewSetup>\n        <id>{s['id']}</id>\n"
             f"        <name>setup {s['id']}</name>\n"
             f"        <size>{s['dims'][0]} {s['dims'][1]} {s['dims'][2]}"
-            f"</size>\n      </ViewSetup>\n"
+            f"</size>\n{extra}      </ViewSetup>\n"
         )
         p = s["pos"]
         aff = f"1.0 0.0 0.0 {p[0]} 0.0 1.0 0.0 {p[1]} 0.0 0.0 1.0 {p[2]}"
@@ -152,6 +214,9 @@ def read_zarr(root, name):
         raw = open(p, "rb").read()
         if comp and comp.get("id") == "gzip":
             raw = gzip.decompress(raw)
+        elif comp and comp.get("id") == "zstd":
+            nel = int(np.prod(chunks))
+            raw = zstd_decompress(raw, nel * dt.itemsize)
         blk = np.frombuffer(raw, dt).reshape(chunks)
         sl_out, sl_blk = [], []
         for d, (i, c, s) in enumerate(zip(idx, chunks, shape)):

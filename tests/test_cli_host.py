@@ -49,7 +49,7 @@ def make_grid_dataset(tmp, size=64, overlap=24, err=(2.5, -1.5, 1.0),
 def test_container_cli_attribute_contract(tmp_path):
     xml, n5, _err, _ = make_grid_dataset(str(tmp_path))
     out = os.path.join(str(tmp_path), "fused.n5")
-    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml, "-o",
+    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml, "-s", "N5", "-o",
              out, "--blockSize", "32,32,32", "--dataType", "UINT16",
              "--minIntensity", "0", "--maxIntensity", "40000"])
     assert r.returncode == 0, r.stderr
@@ -123,7 +123,7 @@ def test_cli_stitching_end_to_end(tmp_path):
 def test_cli_fusion_end_to_end(tmp_path):
     xml, n5, err, (a, b) = make_grid_dataset(str(tmp_path))
     out = os.path.join(str(tmp_path), "fused.n5")
-    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml, "-o",
+    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml, "-s", "N5", "-o",
              out, "--blockSize", "32,32,32", "--dataType", "FLOAT32",
              "--downsamplings", "1,1,1;2,2,2"])
     assert r.returncode == 0, r.stderr
@@ -170,7 +170,7 @@ def test_cli_full_pipeline_stitch_solve_fuse(tmp_path):
     true_pos = np.array([40 + err[0], err[1], err[2]])
     assert np.all(np.abs(t[1] - true_pos) < 0.75), (t[1], true_pos)
     out = os.path.join(str(tmp_path), "fused.n5")
-    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml, "-o",
+    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml, "-s", "N5", "-o",
              out, "--blockSize", "32,32,32", "--dataType", "FLOAT32"])
     assert r.returncode == 0, r.stderr
     r = run([os.path.join(BIN, "affine-fusion"), "-o", out,
@@ -343,7 +343,7 @@ def test_cli_grid6_stitch_solve_fuse(tmp_path):
         err_px = np.abs(t[s] - true_pos[s])
         assert np.all(err_px < 0.8), (s, t[s], true_pos[s])
     out = os.path.join(str(tmp_path), "fused.n5")
-    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml, "-o",
+    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml, "-s", "N5", "-o",
              out, "--blockSize", "32,32,32", "--dataType", "UINT16",
              "--minIntensity", "0", "--maxIntensity", "65535"])
     assert r.returncode == 0, r.stderr
@@ -366,7 +366,7 @@ def test_cli_fusion_masks(tmp_path):
     mask content."""
     xml, n5, err, (a, b) = make_grid_dataset(str(tmp_path))
     out = os.path.join(str(tmp_path), "masks.n5")
-    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml, "-o",
+    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml, "-s", "N5", "-o",
              out, "--blockSize", "32,32,32", "--dataType", "UINT8",
              "--downsamplings", "1,1,1;2,2,2"])
     assert r.returncode == 0, r.stderr
@@ -412,7 +412,7 @@ def test_cli_fusion_intensity_coefficients(tmp_path):
                                 ab)
         abs_.append(ab)
     out = os.path.join(str(tmp_path), "fused_coeff.n5")
-    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml, "-o",
+    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml, "-s", "N5", "-o",
              out, "--blockSize", "32,32,32", "--dataType", "FLOAT32",
              "--downsamplings", "1,1,1"])
     assert r.returncode == 0, r.stderr
@@ -441,3 +441,205 @@ def test_cli_fusion_intensity_coefficients(tmp_path):
         (0, 0, 0), (104, 64, 64), of.FUSION_AVG_BLEND,
         out_dtype=np.float32)
     assert np.max(np.abs(ref_plain - fused)) > 1.0
+
+
+def test_container_cli_zstd_default_and_roundtrip(tmp_path):
+    """Zstandard is the reference's default codec
+    (CreateFusionContainer.java:71-73): a container created with default
+    flags stores zstd chunks; the C++ reader reads back what the
+    independent python zstd writer produced and vice versa."""
+    xml, n5, _err, _ = make_grid_dataset(str(tmp_path))
+    out = os.path.join(str(tmp_path), "fused.n5")
+    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml,
+             "-s", "N5", "-o", out, "--blockSize", "32,32,32",
+             "-d", "UINT16"])
+    assert r.returncode == 0, r.stderr
+    _, dattrs = n5util.read_dataset(out, "ch0tp0/s0")
+    assert dattrs["compression"]["type"] == "zstd"
+    # python-written zstd input container readable by the C++ CLI side:
+    n5z = os.path.join(str(tmp_path), "zin.n5")
+    rng = np.random.default_rng(7)
+    vol = rng.integers(0, 60000, size=(16, 16, 16)).astype(np.uint16)
+    n5util.write_dataset(n5z, "setup0/timepoint0/s0", vol, (8, 8, 8),
+                         compression="zstd")
+    back, _ = n5util.read_dataset(n5z, "setup0/timepoint0/s0")
+    assert np.array_equal(back, vol)
+
+
+def test_container_cli_zarr_default_storage(tmp_path):
+    """Default storage is OME-ZARR (reference -s default), with zstd
+    chunks, readable back by the python zarr reader."""
+    xml, n5, _err, _ = make_grid_dataset(str(tmp_path))
+    out = os.path.join(str(tmp_path), "fused.zarr")
+    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml,
+             "-o", out, "--blockSize", "32,32,32", "-d", "UINT16"])
+    assert r.returncode == 0, r.stderr
+    bs = n5util.zarr_root_attrs(out)["Bigstitcher-Spark"]
+    assert bs["FusionFormat"] == "OME-ZARR"
+    import json as _json
+    za = _json.load(open(os.path.join(out, "s0", ".zarray")))
+    assert za["compressor"]["id"] == "zstd"
+
+
+def test_container_cli_anisotropy_attrs(tmp_path):
+    """--preserveAnisotropy: factor computed from voxel sizes when not
+    given (reference CreateFusionContainer.java:189-211), bbox z divided
+    by it (floor/ceil), attributes written for the fusion step."""
+    n5 = os.path.join(str(tmp_path), "input.n5")
+    xml = os.path.join(str(tmp_path), "dataset.xml")
+    rng = np.random.default_rng(3)
+    vol = rng.integers(0, 60000, size=(16, 32, 32)).astype(np.uint16)
+    n5util.write_dataset(n5, "setup0/timepoint0/s0", vol, (16, 16, 16))
+    n5util.make_dataset_xml(
+        xml, "input.n5",
+        [dict(id=0, dims=(32, 32, 16), pos=(0.0, 0.0, 0.0),
+              voxel=(0.5, 0.5, 2.0))])
+    out = os.path.join(str(tmp_path), "fused.n5")
+    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml,
+             "-s", "N5", "-o", out, "--blockSize", "16,16,16",
+             "-d", "UINT16", "--preserveAnisotropy"])
+    assert r.returncode == 0, r.stderr
+    attrs = n5util.root_attrs(out)["Bigstitcher-Spark"]
+    assert attrs["PreserveAnisotropy"] is True
+    assert abs(attrs["AnisotropyFactor"] - 4.0) < 1e-12  # 2.0/min(0.5,0.5)
+    # z extent [0,15] -> [floor(0/4), ceil(15/4)] = [0, 4]
+    assert attrs["Boundingbox_min"] == [0, 0, 0]
+    assert attrs["Boundingbox_max"][2] == 4
+    mri = attrs["MultiResolutionInfos"]
+    assert mri[0][0]["dimensions"] == [32, 32, 5]
+
+
+def test_container_cli_multires_and_repeated_ds(tmp_path):
+    """--multiRes auto-ladder and the reference's repeated -ds flags
+    (CreateFusionContainer.java:110-112, split=';')."""
+    xml, n5, _err, _ = make_grid_dataset(str(tmp_path))
+    out = os.path.join(str(tmp_path), "a.n5")
+    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml,
+             "-s", "N5", "-o", out, "--blockSize", "16,16,16",
+             "-d", "UINT16", "--multiRes"])
+    assert r.returncode == 0, r.stderr
+    mri = n5util.root_attrs(out)["Bigstitcher-Spark"]["MultiResolutionInfos"]
+    ladders = [lv["absoluteDownsampling"] for lv in mri[0]]
+    assert ladders[0] == [1, 1, 1]
+    assert len(ladders) >= 3  # 104x64x64 / 16-block needs >= 8x in x
+    for a, b in zip(ladders, ladders[1:]):
+        assert all(bb % aa == 0 for aa, bb in zip(a, b))
+    # last level fits one block
+    last = mri[0][-1]["dimensions"]
+    assert all(d <= 16 for d in last)
+    out2 = os.path.join(str(tmp_path), "b.n5")
+    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml,
+             "-s", "N5", "-o", out2, "--blockSize", "16,16,16",
+             "-d", "UINT16", "-ds", "1,1,1", "-ds", "2,2,1",
+             "-ds", "4,4,2"])
+    assert r.returncode == 0, r.stderr
+    mri2 = n5util.root_attrs(out2)["Bigstitcher-Spark"]["MultiResolutionInfos"]
+    assert [lv["absoluteDownsampling"] for lv in mri2[0]] == [
+        [1, 1, 1], [2, 2, 1], [4, 4, 2]]
+
+
+def test_container_cli_rejects_unsupported(tmp_path):
+    xml, n5, _err, _ = make_grid_dataset(str(tmp_path))
+    out = os.path.join(str(tmp_path), "x.n5")
+    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml,
+             "-o", out, "-s", "HDF5"])
+    assert r.returncode != 0 and "HDF5" in r.stderr
+    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml,
+             "-o", out, "-c", "Lz4"])
+    assert r.returncode != 0 and "Lz4" in r.stderr
+    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml,
+             "-o", out, "--bdv"])
+    assert r.returncode != 0
+
+
+def test_view_selection_flags(tmp_path):
+    """-vi / --tileId view selection (util/Import.java:94-204): the
+    container bbox covers only the selected views; stitching --dryRun
+    enumerates only selected pairs."""
+    n5 = os.path.join(str(tmp_path), "input.n5")
+    xml = os.path.join(str(tmp_path), "dataset.xml")
+    rng = np.random.default_rng(3)
+    for sid in range(3):
+        vol = rng.integers(0, 60000, size=(16, 16, 16)).astype(np.uint16)
+        n5util.write_dataset(n5, f"setup{sid}/timepoint0/s0", vol,
+                             (16, 16, 16))
+    n5util.make_dataset_xml(
+        xml, "input.n5",
+        [dict(id=0, dims=(16, 16, 16), pos=(0.0, 0.0, 0.0),
+              attrs=dict(tile=0, angle=0, channel=0, illumination=0)),
+         dict(id=1, dims=(16, 16, 16), pos=(8.0, 0.0, 0.0),
+              attrs=dict(tile=1, angle=0, channel=0, illumination=0)),
+         dict(id=2, dims=(16, 16, 16), pos=(16.0, 0.0, 0.0),
+              attrs=dict(tile=2, angle=0, channel=0, illumination=0))])
+    out = os.path.join(str(tmp_path), "sel.n5")
+    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml,
+             "-s", "N5", "-o", out, "--blockSize", "16,16,16",
+             "-d", "UINT16", "--tileId", "0,1"])
+    assert r.returncode == 0, r.stderr
+    attrs = n5util.root_attrs(out)["Bigstitcher-Spark"]
+    assert attrs["Boundingbox_max"][0] == 23  # setups 0,1 only: 8+16-1
+    out2 = os.path.join(str(tmp_path), "sel2.n5")
+    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml,
+             "-s", "N5", "-o", out2, "--blockSize", "16,16,16",
+             "-d", "UINT16", "-vi", "0,0", "-vi", "0,2"])
+    assert r.returncode == 0, r.stderr
+    attrs2 = n5util.root_attrs(out2)["Bigstitcher-Spark"]
+    assert attrs2["Boundingbox_max"][0] == 31
+    # bad -vi -> clean error
+    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml,
+             "-s", "N5", "-o", out2, "-vi", "0,9"])
+    assert r.returncode != 0 and "not present" in r.stderr
+    # stitching respects --tileId (dryRun: no GPU touched)
+    r = run([os.path.join(BIN, "stitching"), "-x", xml, "--tileId", "0,1",
+             "--dryRun"])
+    assert r.returncode == 0, r.stderr
+    assert "1 overlapping pairs" in r.stdout
+
+
+@pytest.mark.gpu
+def test_cli_fusion_anisotropy_parity(tmp_path):
+    """GPU parity for the anisotropy path: an
+    anisotropic dataset (voxel z = 4x xy) fused through
+    create-fusion-container --preserveAnisotropy + affine-fusion must
+    equal the oracle fusing with the z-divided transforms
+    (SparkAffineFusion.java:486-491, TransformVirtual.adjustAllTransforms
+    [PIN-ANISO])."""
+    n5 = os.path.join(str(tmp_path), "input.n5")
+    xml = os.path.join(str(tmp_path), "dataset.xml")
+    rng = np.random.default_rng(11)
+    vols = {}
+    for sid, pos in ((0, 0.0), (1, 20.0)):
+        vol = rng.integers(0, 60000, size=(12, 32, 32)).astype(np.uint16)
+        vols[sid] = vol
+        n5util.write_dataset(n5, f"setup{sid}/timepoint0/s0", vol,
+                             (16, 16, 16))
+    n5util.make_dataset_xml(
+        xml, "input.n5",
+        [dict(id=0, dims=(32, 32, 12), pos=(0.0, 0.0, 0.0),
+              voxel=(0.5, 0.5, 2.0)),
+         dict(id=1, dims=(32, 32, 12), pos=(20.0, 0.0, 0.0),
+              voxel=(0.5, 0.5, 2.0))])
+    out = os.path.join(str(tmp_path), "fused.n5")
+    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml,
+             "-s", "N5", "-o", out, "--blockSize", "16,16,16",
+             "-d", "FLOAT32", "--preserveAnisotropy"])
+    assert r.returncode == 0, r.stderr
+    attrs = n5util.root_attrs(out)["Bigstitcher-Spark"]
+    f = attrs["AnisotropyFactor"]
+    assert abs(f - 4.0) < 1e-12
+    bbmin = attrs["Boundingbox_min"]
+    r = run([os.path.join(BIN, "affine-fusion"), "-o", out,
+             "-f", "AVG_BLEND", "--blendingRange", "4"])
+    assert r.returncode == 0, r.stderr + r.stdout
+    fused, _ = n5util.read_dataset(out, "ch0tp0/s0")
+    # oracle: same fusion with world-z / 4 pre-concatenated transforms
+    views = []
+    for sid, pos in ((0, 0.0), (1, 20.0)):
+        aff = np.hstack([np.eye(3), np.array([[pos], [0.0], [0.0]])])
+        aff[2, :] /= f
+        views.append(dict(data=vols[sid], affine=aff, border=(0, 0, 0),
+                          range=(4, 4, 4)))
+    ref = of.fuse_block(views, tuple(bbmin), fused.shape[::-1],
+                        of.FUSION_AVG_BLEND)
+    denom = np.maximum(np.abs(ref), 1.0)
+    assert np.max(np.abs(fused.astype(np.float64) - ref) / denom) < 1e-4
