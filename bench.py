@@ -44,8 +44,10 @@ def algorithmic_bytes_per_launch(size, pairs_kernel_dims):
     return {
         "fft_x_fwd": 2.0 * mx * my * mz + c8 * cx * my * mz,
         "fft_y_fwd": c8 * cx * my * mz + c8 * cx * py * mz,
-        "fft_z_fwd": c8 * cx * py * mz + c8 * cx * py * pz,
-        "fft_z_inv": 3.0 * c8 * cx * py * pz,
+        # fused z chain (k_fft_z_fused, logged as fft_z_inv): reads both
+        # y-transformed spectra once (valid z-extent) + writes Q once —
+        # the z spectra never round-trip through HBM (round 2)
+        "fft_z_inv": 2.0 * c8 * cx * py * mz + c8 * cx * py * pz,
         "fft_y_inv": 2.0 * c8 * cx * py * pz,
         "fft_x_inv": c8 * cx * py * pz + 4.0 * px * py * pz,
         "peak": 4.0 * px * py * pz,
@@ -222,11 +224,11 @@ def main():
         sidecar = load_traffic_sidecar(args.size)
         if sidecar and dom in sidecar.get("kernels", {}):
             traffic = sidecar["kernels"][dom]
-        total_alg_bytes_per_pair = (
-            2 * ab["fft_x_fwd"] + 2 * ab["fft_y_fwd"] + 2 * ab["fft_z_fwd"]
-            + ab["fft_z_inv"] + ab["fft_y_inv"] + ab["fft_x_inv"]
-            + ab["peak"]
-        )
+        # whole-path algorithmic bytes from the ACTUAL launch mix
+        total_alg_bytes_per_pair = sum(
+            ab[kname] * stats["kernels"][kname]["launches"]
+            for kname in ab
+        ) / max(1, stats["pairs"])
         agg = total_alg_bytes_per_pair * value / 1e9  # GB/s whole-path
         roofline = {
             "bound": "hbm",

@@ -49,9 +49,6 @@
 #define LPB_S 8    /* lines per block, strided passes */
 #define TPL_S 64   /* 512-thread blocks; 35 KB LDS -> 4 WGs/CU (full
                       32-wave occupancy) */
-#define PK_TX 128 /* peak-scan (x,y) strip */
-#define PK_TY 16
-#define PK_CZ 64 /* z planes streamed per chunk */
 
 typedef unsigned long long u64;
 
@@ -318,6 +315,116 @@ __global__ __launch_bounds__(LPB_S *TPL_S) void k_fft_pass(
   }
 }
 
+/* Fused z chain: forward z-FFT of BOTH spectra + cross-power normalise
+ * + inverse z-FFT in ONE kernel (replaces two k_fft_pass(+1) launches
+ * and the fused-crosspower k_fft_pass(-1) launch). The z spectra are
+ * never materialised in HBM: per column pair this reads A and B once
+ * (y-transformed) and writes Q once — 3 column-volumes of traffic
+ * instead of 7 (−2.16 GB/pair at 512^3). Same line mapping as
+ * k_fft_pass (group = y, lines = adjacent x-columns, element = z);
+ * LDS holds TWO data buffers (66 KB at n=512 -> 2 WGs/CU; the tripled
+ * FFT compute per load keeps the CUs busy at the lower occupancy).
+ * Q for each thread's elements is staged in registers between the
+ * forward FFTs and the bit-reversed rewrite for the inverse FFT. */
+__global__ __launch_bounds__(LPB_S *TPL_S) void k_fft_z_fused(
+    f2 *a, f2 *b, int n, int log2n, long estride, long gstride, int nlines,
+    int nchunks, int ngroups, int valid_a, int valid_b, float scale,
+    const f2 *twg) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  f2 *tw = (f2 *)smem;
+  f2 *da = tw + (n >> 1);
+  f2 *db = da + (long)LPB_S * n;
+  const int tid = threadIdx.x;
+  const int line = tid % LPB_S, tl = tid / LPB_S;
+  for (int i = tid; i < (n >> 1); i += LPB_S * TPL_S) tw[i] = twg[i];
+  __syncthreads();
+  constexpr int NPAIR = LPB_S / 2;
+  constexpr int ESTR = (LPB_S * TPL_S) / NPAIR;
+  const int pl = tid & (NPAIR - 1), t2 = tid / NPAIR;
+  const long nwg = (long)ngroups * nchunks;
+  for (long wg = blockIdx.x; wg < nwg; wg += gridDim.x) {
+    const int group = (int)(wg / nchunks);
+    const int x2 = (int)(wg % nchunks) * LPB_S + 2 * pl;
+    const bool pair_ok = x2 + 1 < nlines;
+    const long base2 = (long)group * gstride + x2;
+    if (pair_ok) {
+      for (int e = t2; e < n; e += ESTR) {
+        float4 va = {0, 0, 0, 0}, vb = {0, 0, 0, 0};
+        if (e < valid_a) va = *(const float4 *)&a[base2 + e * estride];
+        if (e < valid_b) vb = *(const float4 *)&b[base2 + e * estride];
+        const long d = (long)brev_n(e, log2n) * LPB_S + 2 * pl;
+        *(float4 *)&da[d] = va;
+        *(float4 *)&db[d] = vb;
+      }
+    } else {
+      for (int l = 0; l < 2; ++l) {
+        const int x = x2 + l;
+        const bool active = x < nlines;
+        const long base = (long)group * gstride + x;
+        for (int e = t2; e < n; e += ESTR) {
+          f2 va = {0, 0}, vb = {0, 0};
+          if (active && e < valid_a) va = a[base + e * estride];
+          if (active && e < valid_b) vb = b[base + e * estride];
+          const long d = (long)brev_n(e, log2n) * LPB_S + 2 * pl + l;
+          da[d] = va;
+          db[d] = vb;
+        }
+      }
+    }
+    __syncthreads();
+    fft_lds<LPB_S, TPL_S>(da, (long)line, n, log2n, tl, tw, +1);
+    fft_lds<LPB_S, TPL_S>(db, (long)line, n, log2n, tl, tw, +1);
+    /* cross-power into registers (each thread its own elements, both
+     * lines of its float4 pair), then bit-reversed rewrite of da for
+     * the inverse DIT [PIN-EPS] */
+    {
+      float4 q[8]; /* up to n/ESTR = 8 elements at n<=1024 */
+      int ne = 0;
+      for (int e = t2; e < n; e += ESTR, ++ne) {
+        const long d = (long)e * LPB_S + 2 * pl;
+        const float4 va = *(const float4 *)&da[d];
+        const float4 vb = *(const float4 *)&db[d];
+        f2 q0 = conjmul({va.x, va.y}, {vb.x, vb.y});
+        f2 q1 = conjmul({va.z, va.w}, {vb.z, vb.w});
+        float m0 = q0.x * q0.x + q0.y * q0.y;
+        float m1 = q1.x * q1.x + q1.y * q1.y;
+        float4 v = {0, 0, 0, 0};
+        if (m0 > 1e-40f) {
+          float s = scale / sqrtf(m0);
+          v.x = q0.x * s;
+          v.y = q0.y * s;
+        }
+        if (m1 > 1e-40f) {
+          float s = scale / sqrtf(m1);
+          v.z = q1.x * s;
+          v.w = q1.y * s;
+        }
+        q[ne] = v;
+      }
+      __syncthreads();
+      ne = 0;
+      for (int e = t2; e < n; e += ESTR, ++ne)
+        *(float4 *)&da[(long)brev_n(e, log2n) * LPB_S + 2 * pl] = q[ne];
+    }
+    __syncthreads();
+    fft_lds<LPB_S, TPL_S>(da, (long)line, n, log2n, tl, tw, -1);
+    if (pair_ok) {
+      for (int e = t2; e < n; e += ESTR)
+        *(float4 *)&a[base2 + e * estride] =
+            *(const float4 *)&da[(long)e * LPB_S + 2 * pl];
+    } else {
+      for (int l = 0; l < 2; ++l) {
+        const int x = x2 + l;
+        if (x >= nlines) continue;
+        const long base = (long)group * gstride + x;
+        for (int e = t2; e < n; e += ESTR)
+          a[base + e * estride] = da[(long)e * LPB_S + 2 * pl + l];
+      }
+    }
+    __syncthreads();
+  }
+}
+
 /* Inverse x pass (C2R, packed): per-line Hermitian half-line -> the real
  * PCM line via an (n/2)-point inverse complex FFT. Build
  * Z2[k] = A + i*conj(W^k)*B with A = X[k]+conj(X[h-k]),
@@ -562,131 +669,115 @@ __device__ void pk_merge_shfl(float (&tv)[5], long long (&ti)[5]) {
   }
 }
 
-/* Streaming local-maxima scan: strict 26-neighborhood maxima with
- * periodic wrap [PIN-MAX]; per-WG top-5 -> wgbuf. A WG owns an (x,y)
- * strip of PK_TX x PK_TY and streams a z-chunk of PK_CZ planes through
- * a 4-plane rolling LDS buffer: one barrier per plane, the load of
- * plane z+2 issued before computing plane z (load/compute overlap),
- * and read amplification only (PK_TX+2)(PK_TY+2)/(PK_TX*PK_TY) ~ 1.27
- * plus two preload planes per chunk — vs 1.59x for a full 3-D halo
- * tile, which also serializes load and compute at its barriers
- * (ablation: tools/probe_peak.hip). */
+/* Register-window local-maxima scan: strict 26-neighborhood maxima
+ * with periodic wrap [PIN-MAX]; per-WG top-5 -> wgbuf.
+ *
+ * Wave-parallel, NO data LDS and NO per-plane barrier (the round-1
+ * LDS-halo version was barrier-serialization-bound: 0.55 ms solo vs a
+ * 0.25 ms load floor — DESIGN.md §8 item 2). Each wave owns one y-row
+ * of a 256-wide x-tile (4 voxels per lane as one aligned float4) and
+ * streams a z-chunk, holding the 3x3 (y,z) row window in registers:
+ * per z-step it loads 3 new rows (y-1,y,y+1 at z+1), gets the x-1/x+1
+ * columns from the neighbor lanes via __shfl_up/down (tile-edge lanes
+ * read the wrapped scalar), and evaluates v > max(8 rows' 3-window
+ * maxima, center row's left/right). The 3x y-row re-read is served by
+ * L1/L2 (waves y-1,y,y+1 of the same block touch the same rows in
+ * lockstep); HBM traffic stays ~= algorithmic. */
+struct pk_row {
+  float4 raw; /* the 4 voxels */
+  float4 nm;  /* per-voxel max over its 3-window (incl self) */
+  float4 cm;  /* per-voxel max over left/right only (excl self) */
+};
+
+__device__ __forceinline__ pk_row pk_load_row(const float *row, int gx,
+                                              int lane, int nlt, int px) {
+  pk_row o;
+  float4 r = *(const float4 *)(row + gx);
+  float lw = __shfl_up(r.w, 1);
+  float rw = __shfl_down(r.x, 1);
+  if (lane == 0) lw = row[(gx - 1 + px) % px];
+  if (lane == nlt - 1) rw = row[(gx + 4) % px];
+  o.raw = r;
+  o.nm.x = fmaxf(fmaxf(lw, r.x), r.y);
+  o.nm.y = fmaxf(fmaxf(r.x, r.y), r.z);
+  o.nm.z = fmaxf(fmaxf(r.y, r.z), r.w);
+  o.nm.w = fmaxf(fmaxf(r.z, r.w), rw);
+  o.cm.x = fmaxf(lw, r.y);
+  o.cm.y = fmaxf(r.x, r.z);
+  o.cm.z = fmaxf(r.y, r.w);
+  o.cm.w = fmaxf(r.z, rw);
+  return o;
+}
+
+__device__ __forceinline__ float4 f4max(float4 a, float4 b) {
+  return {fmaxf(a.x, b.x), fmaxf(a.y, b.y), fmaxf(a.z, b.z),
+          fmaxf(a.w, b.w)};
+}
+
+#define PKW_CZ 64 /* z planes streamed per chunk */
+
 __global__ __launch_bounds__(256) void k_peak_tile(
     const float *pcm, int px, int py, int pz, bs_peak *wgbuf) {
-  const int HX = PK_TX + 2, HY = PK_TY + 2;
-  __shared__ float pl[4][HY * HX];
-  __shared__ float wv[4][5];
-  __shared__ long long wi[4][5];
-  const int tid = threadIdx.x;
-  const int ntx = (px + PK_TX - 1) / PK_TX;
-  const int nty = (py + PK_TY - 1) / PK_TY;
-  const int ncz = (pz + PK_CZ - 1) / PK_CZ;
-  const long nchunks = (long)ntx * nty * ncz;
-  /* x/y halo coordinates wrap at most one period unless the volume is
-   * smaller than one strip+halo (tiny parity cases): uniform branch.
-   * Interior rows load as aligned float4 (x0 is a multiple of PK_TX,
-   * px a pow2): 32 lanes x 16B + 2 scalar edge lanes per row. */
-  const bool wrap1 = px >= HX && py >= HY;
+  __shared__ float wvs[4][5];
+  __shared__ long long wis[4][5];
+  const int tid = threadIdx.x, lane = tid & 63, wv = tid >> 6;
+  const int ntx = (px + 255) / 256;
+  const int nty = (py + 3) / 4; /* 4 y-rows per block, one per wave */
+  const int ncz = (pz + PKW_CZ - 1) / PKW_CZ;
   float tv[5];
   long long ti[5];
   for (int k = 0; k < 5; ++k) { tv[k] = -3.0e38f; ti[k] = 0x7fffffffffffffffLL; }
+  const long nchunks = (long)ntx * nty * ncz;
   for (long t0 = blockIdx.x; t0 < nchunks; t0 += gridDim.x) {
     const int bx = (int)(t0 % ntx);
     const int by = (int)((t0 / ntx) % nty);
     const int bz = (int)(t0 / ((long)ntx * nty));
-    const int x0 = bx * PK_TX, y0 = by * PK_TY, z0 = bz * PK_CZ;
-    const int zend = min(z0 + PK_CZ, pz);
-    auto load_plane = [&](int gz, float *dst) {
+    const int y = by * 4 + wv;
+    if (y >= py) continue; /* no barriers in the loop: safe */
+    const int x0 = bx * 256;
+    const int nlt = min(64, (px - x0) >> 2); /* active lanes this tile */
+    const bool lx = lane < nlt;
+    const int gx = x0 + 4 * (lx ? lane : 0);
+    const int ym = (y - 1 + py) % py, yp = (y + 1) % py;
+    const int z0 = bz * PKW_CZ, zend = min(z0 + PKW_CZ, pz);
+    auto plane_rows = [&](int gz, pk_row *rows) {
       const float *src = pcm + (long)gz * py * px;
-      if (wrap1) {
-        for (int i = tid; i < HY * 34; i += 256) {
-          int t = i % 34, ly = i / 34;
-          int gy = y0 + ly - 1;
-          gy += gy < 0 ? py : 0;
-          gy -= gy >= py ? py : 0;
-          const float *row = src + (long)gy * px;
-          if (t < 32) {
-            int gx = x0 + 4 * t;
-            float4 v;
-            if (gx + 3 < px) {
-              v = *(const float4 *)(row + gx);
-            } else { /* last strip of a px < x0+PK_TX volume: wrap */
-              float tmp[4];
-              for (int q = 0; q < 4; ++q) {
-                int xx = gx + q;
-                xx -= xx >= px ? px : 0;
-                tmp[q] = row[xx];
-              }
-              v = {tmp[0], tmp[1], tmp[2], tmp[3]};
-            }
-            float *d = dst + ly * HX + 1 + 4 * t;
-            d[0] = v.x; d[1] = v.y; d[2] = v.z; d[3] = v.w;
-          } else {
-            int gx = (t == 32) ? x0 - 1 : x0 + PK_TX;
-            gx += gx < 0 ? px : 0;
-            gx -= gx >= px ? px : 0;
-            dst[ly * HX + (t == 32 ? 0 : HX - 1)] = row[gx];
-          }
-        }
-      } else { /* tiny volumes: full modulo wrap */
-        for (int i = tid; i < HX * HY; i += 256) {
-          int lx = i % HX, ly = i / HX;
-          int gx = (x0 + lx - 1 + 4 * px) % px;
-          int gy = (y0 + ly - 1 + py) % py;
-          dst[i] = src[(long)gy * px + gx];
-        }
-      }
+      rows[0] = pk_load_row(src + (long)ym * px, gx, lane, nlt, px);
+      rows[1] = pk_load_row(src + (long)y * px, gx, lane, nlt, px);
+      rows[2] = pk_load_row(src + (long)yp * px, gx, lane, nlt, px);
     };
-    /* preload planes z0-1, z0, z0+1 into slots 0,1,2 */
-    for (int p = -1; p <= 1; ++p) {
-      int gz = z0 + p;
-      gz += gz < 0 ? pz : 0;
-      gz -= gz >= pz ? pz : 0;
-      load_plane(gz, pl[p + 1]);
-    }
-    __syncthreads();
+    pk_row rm[3], r0[3], rp[3];
+    plane_rows((z0 - 1 + pz) % pz, rm);
+    plane_rows(z0 % pz, r0);
     for (int z = z0; z < zend; ++z) {
-      /* issue the load of plane z+2 into the free slot before the
-       * maxima math on plane z (z+2 wraps at most one period: z<pz) */
-      {
-        int gz = z + 2;
-        gz -= gz >= pz ? pz : 0;
-        gz -= gz >= pz ? pz : 0; /* pz<=2: two single-period steps */
-        load_plane(gz, pl[(z - z0 + 3) & 3]);
+      plane_rows((z + 1) % pz, rp);
+      /* strict 26-max for plane z, center row r0[1] */
+      float4 m = f4max(rm[0].nm, f4max(rm[1].nm, rm[2].nm));
+      m = f4max(m, f4max(rp[0].nm, f4max(rp[1].nm, rp[2].nm)));
+      m = f4max(m, f4max(r0[0].nm, r0[2].nm));
+      m = f4max(m, r0[1].cm);
+      const float4 v = r0[1].raw;
+      if (lx) {
+        const long long ibase = ((long long)z * py + y) * px + gx;
+        if (v.x > m.x) pk_insert(tv, ti, v.x, ibase);
+        if (v.y > m.y) pk_insert(tv, ti, v.y, ibase + 1);
+        if (v.z > m.z) pk_insert(tv, ti, v.z, ibase + 2);
+        if (v.w > m.w) pk_insert(tv, ti, v.w, ibase + 3);
       }
-      /* strict 26-max on plane z from slots (z-z0)%4 .. +2: every LDS
-       * offset is base + compile-time constant, no divergence */
-      const float *pm = pl[(z - z0) & 3];
-      const float *pc = pl[(z - z0 + 1) & 3];
-      const float *pp = pl[(z - z0 + 2) & 3];
-      for (int i = tid; i < PK_TX * PK_TY; i += 256) {
-        int lx = i % PK_TX, ly = i / PK_TX;
-        int gx = x0 + lx, gy = y0 + ly;
-        const int base = (ly + 1) * HX + lx + 1;
-        float v = pc[base];
-        float m = fmaxf(pc[base - 1], pc[base + 1]);
 #pragma unroll
-        for (int dy = -1; dy <= 1; ++dy) {
-          const int b2 = base + dy * HX;
-          m = fmaxf(m, fmaxf(fmaxf(pm[b2 - 1], pm[b2]), pm[b2 + 1]));
-          m = fmaxf(m, fmaxf(fmaxf(pp[b2 - 1], pp[b2]), pp[b2 + 1]));
-          if (dy != 0)
-            m = fmaxf(m, fmaxf(fmaxf(pc[b2 - 1], pc[b2]), pc[b2 + 1]));
-        }
-        if (gx < px && gy < py && v > m)
-          pk_insert(tv, ti, v, ((long long)z * py + gy) * px + gx);
+      for (int k = 0; k < 3; ++k) {
+        rm[k] = r0[k];
+        r0[k] = rp[k];
       }
-      __syncthreads(); /* load of z+2 complete; slots rotate */
     }
   }
   pk_merge_shfl(tv, ti);
-  int lane = tid & 63, wave = tid >> 6;
   if (lane == 0)
-    for (int k = 0; k < 5; ++k) { wv[wave][k] = tv[k]; wi[wave][k] = ti[k]; }
+    for (int k = 0; k < 5; ++k) { wvs[wv][k] = tv[k]; wis[wv][k] = ti[k]; }
   __syncthreads();
   if (tid == 0) {
     for (int w = 1; w < 4; ++w)
-      for (int k = 0; k < 5; ++k) pk_insert(tv, ti, wv[w][k], wi[w][k]);
+      for (int k = 0; k < 5; ++k) pk_insert(tv, ti, wvs[w][k], wis[w][k]);
     bs_peak *o = wgbuf + (long)blockIdx.x * 5;
     for (int k = 0; k < 5; ++k) o[k] = {tv[k], 0, ti[k]};
   }
@@ -1253,6 +1344,9 @@ extern "C" int bs_ctx_create(bs_ctx **out, int device_id) {
   (void)hipFuncSetAttribute((const void *)k_fft_pass,
                             hipFuncAttributeMaxDynamicSharedMemorySize,
                             160 * 1024);
+  (void)hipFuncSetAttribute((const void *)k_fft_z_fused,
+                            hipFuncAttributeMaxDynamicSharedMemorySize,
+                            160 * 1024);
   (void)hipFuncSetAttribute((const void *)k_fft_x_fwd,
                             hipFuncAttributeMaxDynamicSharedMemorySize,
                             64 * 1024);
@@ -1639,23 +1733,17 @@ static int stitch_phaseA(bs_ctx *c, bs_slot *sl, const bs_pair_desc &pd,
                        (long)Py * Cxp, Cx, nchunks, sl->reg[t].mz,
                        sl->reg[t].my, +1, 1.0f, twy);
   }
-  for (int t = 0; t < 2; ++t) {
-    size_t lds = ((Pz / 2) + (size_t)LPB_S * Pz) * sizeof(f2);
-    bs_tim tt(c, BS_K_FFT_Z_FWD, sl->stream);
-    hipLaunchKernelGGL(k_fft_pass, dim3(std::min(4096L, (long)Py * nchunks)),
-                       dim3(LPB_S * TPL_S), lds, sl->stream, spec[t],
-                       (const f2 *)nullptr, spec[t], Pz, ilog2(Pz),
-                       (long)Py * Cxp, Cxp, Cx, nchunks, Py, sl->reg[t].mz,
-                       +1, 1.0f, twz);
-  }
-  { /* inverse z fused with cross-power normalise [PIN-EPS] */
+  { /* fused z chain: fwd z (A,B) + cross-power [PIN-EPS] + inv z in one
+     * launch — the z spectra never round-trip through HBM */
     float scale = 1.0f / ((float)Px * (float)Py * (float)Pz);
-    size_t lds = ((Pz / 2) + (size_t)LPB_S * Pz) * sizeof(f2);
+    size_t lds = ((Pz / 2) + 2 * (size_t)LPB_S * Pz) * sizeof(f2);
     bs_tim tt(c, BS_K_FFT_Z_INV, sl->stream);
-    hipLaunchKernelGGL(k_fft_pass, dim3(std::min(4096L, (long)Py * nchunks)),
+    hipLaunchKernelGGL(k_fft_z_fused,
+                       dim3(std::min(4096L, (long)Py * nchunks)),
                        dim3(LPB_S * TPL_S), lds, sl->stream, spec[0],
-                       spec[1], spec[0], Pz, ilog2(Pz), (long)Py * Cxp, Cxp,
-                       Cx, nchunks, Py, Pz, -1, scale, twz);
+                       spec[1], Pz, ilog2(Pz), (long)Py * Cxp, Cxp, Cx,
+                       nchunks, Py, sl->reg[0].mz, sl->reg[1].mz, scale,
+                       twz);
   }
   {
     size_t lds = ((Py / 2) + (size_t)LPB_S * Py) * sizeof(f2);
@@ -1696,8 +1784,8 @@ static int stitch_phaseA(bs_ctx *c, bs_slot *sl, const bs_pair_desc &pd,
   c->dbg_py = Py;
   c->dbg_pz = Pz;
   /* peak scan [PIN-MAX] */
-  long ntiles = (long)((Px + PK_TX - 1) / PK_TX) *
-                ((Py + PK_TY - 1) / PK_TY) * ((Pz + PK_CZ - 1) / PK_CZ);
+  long ntiles = (long)((Px + 255) / 256) * ((Py + 3) / 4) *
+                ((Pz + PKW_CZ - 1) / PKW_CZ);
   long npkwg = std::min(2048L, ntiles);
   int rc2 = ensure_dev(c, (void **)&sl->wgpk, &sl->wgpk_cap,
                        (size_t)npkwg * 5 * sizeof(bs_peak));
