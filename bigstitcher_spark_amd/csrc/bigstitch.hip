@@ -322,18 +322,21 @@ __global__ __launch_bounds__(LPB_S *TPL_S) void k_fft_pass(
  * (y-transformed) and writes Q once — 3 column-volumes of traffic
  * instead of 7 (−2.16 GB/pair at 512^3). Same line mapping as
  * k_fft_pass (group = y, lines = adjacent x-columns, element = z);
- * LDS holds TWO data buffers (66 KB at n=512 -> 2 WGs/CU; the tripled
- * FFT compute per load keeps the CUs busy at the lower occupancy).
- * Q for each thread's elements is staged in registers between the
- * forward FFTs and the bit-reversed rewrite for the inverse FFT. */
-__global__ __launch_bounds__(LPB_S *TPL_S) void k_fft_z_fused(
+ * ONE LDS data buffer (35 KB at n=512 -> 4 WGs/CU, full occupancy):
+ * B's global loads are issued into registers up front (their latency
+ * hides under A's FFT), A's spectrum is parked in registers while the
+ * buffer is reused for B's FFT, and Q is rebuilt bit-reversed in the
+ * same buffer for the inverse DIT. NE = elements per thread
+ * (compile-time so the staging arrays live in REGISTERS, not scratch);
+ * n <= 128*NE, max 1024. */
+template <int NE, bool PB>
+__device__ __forceinline__ void zf_body(
     f2 *a, f2 *b, int n, int log2n, long estride, long gstride, int nlines,
     int nchunks, int ngroups, int valid_a, int valid_b, float scale,
     const f2 *twg) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   f2 *tw = (f2 *)smem;
   f2 *da = tw + (n >> 1);
-  f2 *db = da + (long)LPB_S * n;
   const int tid = threadIdx.x;
   const int line = tid % LPB_S, tl = tid / LPB_S;
   for (int i = tid; i < (n >> 1); i += LPB_S * TPL_S) tw[i] = twg[i];
@@ -347,43 +350,94 @@ __global__ __launch_bounds__(LPB_S *TPL_S) void k_fft_z_fused(
     const int x2 = (int)(wg % nchunks) * LPB_S + 2 * pl;
     const bool pair_ok = x2 + 1 < nlines;
     const long base2 = (long)group * gstride + x2;
+    float4 breg[PB ? NE : 1]; /* B pair-lines (register prefetch) */
+    float4 areg[NE]; /* A spectrum after the forward FFT */
     if (pair_ok) {
-      for (int e = t2; e < n; e += ESTR) {
-        float4 va = {0, 0, 0, 0}, vb = {0, 0, 0, 0};
-        if (e < valid_a) va = *(const float4 *)&a[base2 + e * estride];
-        if (e < valid_b) vb = *(const float4 *)&b[base2 + e * estride];
-        const long d = (long)brev_n(e, log2n) * LPB_S + 2 * pl;
-        *(float4 *)&da[d] = va;
-        *(float4 *)&db[d] = vb;
+      /* issue B's loads first — they drain while A's FFT runs */
+      if (PB)
+#pragma unroll
+      for (int k = 0; k < NE; ++k) {
+        const int e = t2 + k * ESTR;
+        breg[k] = (e < n && e < valid_b)
+                      ? *(const float4 *)&b[base2 + e * estride]
+                      : float4{0, 0, 0, 0};
+      }
+#pragma unroll
+      for (int k = 0; k < NE; ++k) {
+        const int e = t2 + k * ESTR;
+        if (e >= n) break;
+        float4 va = e < valid_a
+                        ? *(const float4 *)&a[base2 + e * estride]
+                        : float4{0, 0, 0, 0};
+        *(float4 *)&da[(long)brev_n(e, log2n) * LPB_S + 2 * pl] = va;
       }
     } else {
       for (int l = 0; l < 2; ++l) {
         const int x = x2 + l;
         const bool active = x < nlines;
         const long base = (long)group * gstride + x;
-        for (int e = t2; e < n; e += ESTR) {
-          f2 va = {0, 0}, vb = {0, 0};
-          if (active && e < valid_a) va = a[base + e * estride];
-          if (active && e < valid_b) vb = b[base + e * estride];
-          const long d = (long)brev_n(e, log2n) * LPB_S + 2 * pl + l;
-          da[d] = va;
-          db[d] = vb;
+#pragma unroll
+        for (int k = 0; k < NE; ++k) {
+          const int e = t2 + k * ESTR;
+          if (e >= n) break;
+          f2 va = (active && e < valid_a) ? a[base + e * estride]
+                                          : f2{0, 0};
+          da[(long)brev_n(e, log2n) * LPB_S + 2 * pl + l] = va;
         }
       }
     }
     __syncthreads();
     fft_lds<LPB_S, TPL_S>(da, (long)line, n, log2n, tl, tw, +1);
-    fft_lds<LPB_S, TPL_S>(db, (long)line, n, log2n, tl, tw, +1);
-    /* cross-power into registers (each thread its own elements, both
-     * lines of its float4 pair), then bit-reversed rewrite of da for
-     * the inverse DIT [PIN-EPS] */
-    {
-      float4 q[8]; /* up to n/ESTR = 8 elements at n<=1024 */
-      int ne = 0;
-      for (int e = t2; e < n; e += ESTR, ++ne) {
-        const long d = (long)e * LPB_S + 2 * pl;
-        const float4 va = *(const float4 *)&da[d];
-        const float4 vb = *(const float4 *)&db[d];
+    { /* park A, refill with bit-reversed B, FFT B */
+#pragma unroll
+      for (int k = 0; k < NE; ++k) {
+        const int e = t2 + k * ESTR;
+        if (e >= n) break;
+        areg[k] = *(const float4 *)&da[(long)e * LPB_S + 2 * pl];
+      }
+      __syncthreads();
+      if (PB && pair_ok) {
+#pragma unroll
+        for (int k = 0; k < NE; ++k) {
+          const int e = t2 + k * ESTR;
+          if (e >= n) break;
+          *(float4 *)&da[(long)brev_n(e, log2n) * LPB_S + 2 * pl] = breg[k];
+        }
+      } else if (pair_ok) {
+#pragma unroll
+        for (int k = 0; k < NE; ++k) {
+          const int e = t2 + k * ESTR;
+          if (e >= n) break;
+          float4 vb = e < valid_b
+                          ? *(const float4 *)&b[base2 + e * estride]
+                          : float4{0, 0, 0, 0};
+          *(float4 *)&da[(long)brev_n(e, log2n) * LPB_S + 2 * pl] = vb;
+        }
+      } else {
+        for (int l = 0; l < 2; ++l) {
+          const int x = x2 + l;
+          const bool active = x < nlines;
+          const long base = (long)group * gstride + x;
+#pragma unroll
+          for (int k = 0; k < NE; ++k) {
+            const int e = t2 + k * ESTR;
+            if (e >= n) break;
+            f2 vb = (active && e < valid_b) ? b[base + e * estride]
+                                            : f2{0, 0};
+            da[(long)brev_n(e, log2n) * LPB_S + 2 * pl + l] = vb;
+          }
+        }
+      }
+      __syncthreads();
+      fft_lds<LPB_S, TPL_S>(da, (long)line, n, log2n, tl, tw, +1);
+    }
+    { /* cross-power normalise [PIN-EPS] into regs, rewrite bit-reversed */
+#pragma unroll
+      for (int k = 0; k < NE; ++k) {
+        const int e = t2 + k * ESTR;
+        if (e >= n) break;
+        const float4 va = areg[k];
+        const float4 vb = *(const float4 *)&da[(long)e * LPB_S + 2 * pl];
         f2 q0 = conjmul({va.x, va.y}, {vb.x, vb.y});
         f2 q1 = conjmul({va.z, va.w}, {vb.z, vb.w});
         float m0 = q0.x * q0.x + q0.y * q0.y;
@@ -399,30 +453,61 @@ __global__ __launch_bounds__(LPB_S *TPL_S) void k_fft_z_fused(
           v.z = q1.x * s;
           v.w = q1.y * s;
         }
-        q[ne] = v;
+        areg[k] = v;
       }
       __syncthreads();
-      ne = 0;
-      for (int e = t2; e < n; e += ESTR, ++ne)
-        *(float4 *)&da[(long)brev_n(e, log2n) * LPB_S + 2 * pl] = q[ne];
+#pragma unroll
+      for (int k = 0; k < NE; ++k) {
+        const int e = t2 + k * ESTR;
+        if (e >= n) break;
+        *(float4 *)&da[(long)brev_n(e, log2n) * LPB_S + 2 * pl] = areg[k];
+      }
+      __syncthreads();
     }
-    __syncthreads();
     fft_lds<LPB_S, TPL_S>(da, (long)line, n, log2n, tl, tw, -1);
     if (pair_ok) {
-      for (int e = t2; e < n; e += ESTR)
+#pragma unroll
+      for (int k = 0; k < NE; ++k) {
+        const int e = t2 + k * ESTR;
+        if (e >= n) break;
         *(float4 *)&a[base2 + e * estride] =
             *(const float4 *)&da[(long)e * LPB_S + 2 * pl];
+      }
     } else {
       for (int l = 0; l < 2; ++l) {
         const int x = x2 + l;
         if (x >= nlines) continue;
         const long base = (long)group * gstride + x;
-        for (int e = t2; e < n; e += ESTR)
+#pragma unroll
+        for (int k = 0; k < NE; ++k) {
+          const int e = t2 + k * ESTR;
+          if (e >= n) break;
           a[base + e * estride] = da[(long)e * LPB_S + 2 * pl + l];
+        }
       }
     }
     __syncthreads();
   }
+}
+
+template <int NE>
+__global__ __launch_bounds__(LPB_S *TPL_S) void k_fft_z_fused(
+    f2 *a, f2 *b, int n, int log2n, long estride, long gstride, int nlines,
+    int nchunks, int ngroups, int valid_a, int valid_b, float scale,
+    const f2 *twg) {
+  zf_body<NE, true>(a, b, n, log2n, estride, gstride, nlines, nchunks,
+                    ngroups, valid_a, valid_b, scale, twg);
+}
+
+/* no-register-prefetch variant, VGPR-capped for 4 blocks/CU (34 KB LDS
+ * is the other limit at n=512): __launch_bounds__ min-blocks 4 */
+template <int NE>
+__global__ __launch_bounds__(LPB_S *TPL_S) __attribute__((amdgpu_waves_per_eu(8))) void k_fft_z_fused_np(
+    f2 *a, f2 *b, int n, int log2n, long estride, long gstride, int nlines,
+    int nchunks, int ngroups, int valid_a, int valid_b, float scale,
+    const f2 *twg) {
+  zf_body<NE, false>(a, b, n, log2n, estride, gstride, nlines, nchunks,
+                     ngroups, valid_a, valid_b, scale, twg);
 }
 
 /* Inverse x pass (C2R, packed): per-line Hermitian half-line -> the real
@@ -683,29 +768,37 @@ __device__ void pk_merge_shfl(float (&tv)[5], long long (&ti)[5]) {
  * maxima, center row's left/right). The 3x y-row re-read is served by
  * L1/L2 (waves y-1,y,y+1 of the same block touch the same rows in
  * lockstep); HBM traffic stays ~= algorithmic. */
-struct pk_row {
-  float4 raw; /* the 4 voxels */
-  float4 nm;  /* per-voxel max over its 3-window (incl self) */
-  float4 cm;  /* per-voxel max over left/right only (excl self) */
+struct pk_raw { /* one plane's 3 y-rows, loads only (no waits forced) */
+  float4 r[3];
+  float el[3], er[3]; /* tile-edge halo scalars (lane 0 / nlt-1 only) */
 };
 
-__device__ __forceinline__ pk_row pk_load_row(const float *row, int gx,
-                                              int lane, int nlt, int px) {
-  pk_row o;
-  float4 r = *(const float4 *)(row + gx);
-  float lw = __shfl_up(r.w, 1);
-  float rw = __shfl_down(r.x, 1);
-  if (lane == 0) lw = row[(gx - 1 + px) % px];
-  if (lane == nlt - 1) rw = row[(gx + 4) % px];
-  o.raw = r;
-  o.nm.x = fmaxf(fmaxf(lw, r.x), r.y);
-  o.nm.y = fmaxf(fmaxf(r.x, r.y), r.z);
-  o.nm.z = fmaxf(fmaxf(r.y, r.z), r.w);
-  o.nm.w = fmaxf(fmaxf(r.z, r.w), rw);
-  o.cm.x = fmaxf(lw, r.y);
-  o.cm.y = fmaxf(r.x, r.z);
-  o.cm.z = fmaxf(r.y, r.w);
-  o.cm.w = fmaxf(r.z, rw);
+struct pk_win { /* computed window entry for one plane */
+  float4 nm[3]; /* per-row max over each voxel's 3-window (incl self) */
+  float4 raw1;  /* center row voxels */
+  float4 cm1;   /* center row left/right max (excl self) */
+};
+
+__device__ __forceinline__ pk_win pk_compute(const pk_raw &p, int lane,
+                                             int nlt) {
+  pk_win o;
+#pragma unroll
+  for (int k = 0; k < 3; ++k) {
+    const float4 r = p.r[k];
+    float lw = __shfl_up(r.w, 1);
+    float rw = __shfl_down(r.x, 1);
+    if (lane == 0) lw = p.el[k];
+    if (lane == nlt - 1) rw = p.er[k];
+    o.nm[k].x = fmaxf(fmaxf(lw, r.x), r.y);
+    o.nm[k].y = fmaxf(fmaxf(r.x, r.y), r.z);
+    o.nm[k].z = fmaxf(fmaxf(r.y, r.z), r.w);
+    o.nm[k].w = fmaxf(fmaxf(r.z, r.w), rw);
+    if (k == 1) {
+      o.raw1 = r;
+      o.cm1 = {fmaxf(lw, r.y), fmaxf(r.x, r.z), fmaxf(r.y, r.w),
+               fmaxf(r.z, rw)};
+    }
+  }
   return o;
 }
 
@@ -716,8 +809,9 @@ __device__ __forceinline__ float4 f4max(float4 a, float4 b) {
 
 #define PKW_CZ 64 /* z planes streamed per chunk */
 
-__global__ __launch_bounds__(256) void k_peak_tile(
-    const float *pcm, int px, int py, int pz, bs_peak *wgbuf) {
+template <bool UNUSED>
+__device__ __forceinline__ void pk_body(const float *pcm, int px, int py,
+                                        int pz, bs_peak *wgbuf) {
   __shared__ float wvs[4][5];
   __shared__ long long wis[4][5];
   const int tid = threadIdx.x, lane = tid & 63, wv = tid >> 6;
@@ -740,23 +834,34 @@ __global__ __launch_bounds__(256) void k_peak_tile(
     const int gx = x0 + 4 * (lx ? lane : 0);
     const int ym = (y - 1 + py) % py, yp = (y + 1) % py;
     const int z0 = bz * PKW_CZ, zend = min(z0 + PKW_CZ, pz);
-    auto plane_rows = [&](int gz, pk_row *rows) {
+    auto issue_plane = [&](int gz) -> pk_raw {
+      /* raw loads only: no shuffle/use, so they stay in flight until
+       * pk_compute one iteration later (load/compute pipelining) */
+      pk_raw p;
       const float *src = pcm + (long)gz * py * px;
-      rows[0] = pk_load_row(src + (long)ym * px, gx, lane, nlt, px);
-      rows[1] = pk_load_row(src + (long)y * px, gx, lane, nlt, px);
-      rows[2] = pk_load_row(src + (long)yp * px, gx, lane, nlt, px);
+      const int yy[3] = {ym, y, yp};
+#pragma unroll
+      for (int k = 0; k < 3; ++k) {
+        const float *row = src + (long)yy[k] * px;
+        p.r[k] = *(const float4 *)(row + gx);
+        if (lane == 0) p.el[k] = row[(gx - 1 + px) % px];
+        if (lane == nlt - 1) p.er[k] = row[(gx + 4) % px];
+      }
+      return p;
     };
-    pk_row rm[3], r0[3], rp[3];
-    plane_rows((z0 - 1 + pz) % pz, rm);
-    plane_rows(z0 % pz, r0);
+    pk_win rm, r0, rp;
+    rm = pk_compute(issue_plane((z0 - 1 + pz) % pz), lane, nlt);
+    r0 = pk_compute(issue_plane(z0 % pz), lane, nlt);
+    pk_raw pend = issue_plane((z0 + 1) % pz);
     for (int z = z0; z < zend; ++z) {
-      plane_rows((z + 1) % pz, rp);
-      /* strict 26-max for plane z, center row r0[1] */
-      float4 m = f4max(rm[0].nm, f4max(rm[1].nm, rm[2].nm));
-      m = f4max(m, f4max(rp[0].nm, f4max(rp[1].nm, rp[2].nm)));
-      m = f4max(m, f4max(r0[0].nm, r0[2].nm));
-      m = f4max(m, r0[1].cm);
-      const float4 v = r0[1].raw;
+      pk_raw pend2 = issue_plane((z + 2) % pz);
+      rp = pk_compute(pend, lane, nlt); /* waits: issued LAST iter */
+      /* strict 26-max for plane z, center row r0 */
+      float4 m = f4max(rm.nm[0], f4max(rm.nm[1], rm.nm[2]));
+      m = f4max(m, f4max(rp.nm[0], f4max(rp.nm[1], rp.nm[2])));
+      m = f4max(m, f4max(r0.nm[0], r0.nm[2]));
+      m = f4max(m, r0.cm1);
+      const float4 v = r0.raw1;
       if (lx) {
         const long long ibase = ((long long)z * py + y) * px + gx;
         if (v.x > m.x) pk_insert(tv, ti, v.x, ibase);
@@ -764,11 +869,9 @@ __global__ __launch_bounds__(256) void k_peak_tile(
         if (v.z > m.z) pk_insert(tv, ti, v.z, ibase + 2);
         if (v.w > m.w) pk_insert(tv, ti, v.w, ibase + 3);
       }
-#pragma unroll
-      for (int k = 0; k < 3; ++k) {
-        rm[k] = r0[k];
-        r0[k] = rp[k];
-      }
+      rm = r0;
+      r0 = rp;
+      pend = pend2;
     }
   }
   pk_merge_shfl(tv, ti);
@@ -781,6 +884,19 @@ __global__ __launch_bounds__(256) void k_peak_tile(
     bs_peak *o = wgbuf + (long)blockIdx.x * 5;
     for (int k = 0; k < 5; ++k) o[k] = {tv[k], 0, ti[k]};
   }
+}
+
+__global__ __launch_bounds__(256)
+    __attribute__((amdgpu_waves_per_eu(4))) void k_peak_tile(
+        const float *pcm, int px, int py, int pz, bs_peak *wgbuf) {
+  pk_body<true>(pcm, px, py, pz, wgbuf);
+}
+
+/* natural-VGPR variant (152 VGPR, 3 waves/SIMD, no spill) for the
+ * occupancy-vs-spill A/B (env BS_PEAK_RELAX) */
+__global__ __launch_bounds__(256) void k_peak_tile_relaxed(
+    const float *pcm, int px, int py, int pz, bs_peak *wgbuf) {
+  pk_body<false>(pcm, px, py, pz, wgbuf);
 }
 
 /* Merge per-WG top-5 lists: each block covers a contiguous slice of
@@ -866,7 +982,7 @@ __global__ __launch_bounds__(256) void k_rtest(bs_region a, bs_region b,
     const bs_cand c = cands[ci];
     if (z < c.loz || z >= c.loz + c.nz) continue; /* block-uniform */
     int rw = 256;
-    while ((rw >> 1) >= c.nx && rw > 16) rw >>= 1;
+    while ((rw >> 1) >= ((c.nx + 1) >> 1) && rw > 16) rw >>= 1;
     const int rsh = __ffs(rw) - 1;
     const int rpg = 256 >> rsh;
     const int lx = tid & (rw - 1);
@@ -876,33 +992,102 @@ __global__ __launch_bounds__(256) void k_rtest(bs_region a, bs_region b,
         a.ptr + (a.oz + z) * a.sxy + a.ox + c.lox;
     const unsigned short *bbase = b.ptr + (b.oz + z + c.sz) * b.sxy +
                                   b.ox + c.lox + c.sx;
-    /* two row-groups in flight; u16^2 fits u32 exactly (promote each
-     * product to u64 BEFORE summing two of them) */
+    /* u16^2 fits u32 exactly (promote each product to u64 BEFORE
+     * summing two of them); u64 sums are order-independent so the
+     * vector path below is bit-identical to the scalar one [PIN-R] */
+    auto add1 = [&](unsigned av, unsigned bv) {
+      pa += av;
+      pb += bv;
+      paa += (u64)(av * av);
+      pbb += (u64)(bv * bv);
+      pab += (u64)(av * bv);
+    };
     const int rstride = gridDim.y * rpg;
-    int r = blockIdx.y * rpg + lr;
-    for (; r + rstride < c.ny; r += 2 * rstride) {
-      const unsigned short *a0 = abase + (a.oy + c.loy + r) * a.sx;
-      const unsigned short *b0 =
-          bbase + (b.oy + c.loy + c.sy + r) * b.sx;
-      const unsigned short *a1 = a0 + (long)rstride * a.sx;
-      const unsigned short *b1 = b0 + (long)rstride * b.sx;
-      for (int x = lx; x < c.nx; x += rw) {
-        unsigned av0 = a0[x], bv0 = b0[x], av1 = a1[x], bv1 = b1[x];
-        pa += av0 + av1;
-        pb += bv0 + bv1;
-        paa += (u64)(av0 * av0) + (u64)(av1 * av1);
-        pbb += (u64)(bv0 * bv0) + (u64)(bv1 * bv1);
-        pab += (u64)(av0 * bv0) + (u64)(av1 * bv1);
+    const int r00 = blockIdx.y * rpg + lr;
+    const unsigned short *ar = abase + (a.oy + c.loy + r00) * a.sx;
+    const unsigned short *br =
+        bbase + (b.oy + c.loy + c.sy + r00) * b.sx;
+    const long astep = (long)rstride * a.sx, bstep = (long)rstride * b.sx;
+    /* u32-pair loads when both rows share 2-byte parity for the WHOLE
+     * candidate (row steps both even — the common case: power-of-two
+     * strides); two rows in flight to amortise loop overhead */
+    if ((((size_t)ar) & 3) == (((size_t)br) & 3) && c.nx >= 4 &&
+        !((a.sx | b.sx) & 1)) {
+      const int h = (int)((((size_t)ar) & 3) >> 1);
+      const int np2 = (c.nx - h) >> 1;
+      const int tail = (c.nx - h) & 1;
+      auto vrow2 = [&](const unsigned *a32, const unsigned *b32,
+                       const unsigned *a32b, const unsigned *b32b) {
+        for (int p = lx; p < np2; p += rw) {
+          const unsigned av0 = a32[p], bv0 = b32[p];
+          const unsigned av1 = a32b[p], bv1 = b32b[p];
+          const unsigned al0 = av0 & 0xFFFFu, ah0 = av0 >> 16;
+          const unsigned bl0 = bv0 & 0xFFFFu, bh0 = bv0 >> 16;
+          const unsigned al1 = av1 & 0xFFFFu, ah1 = av1 >> 16;
+          const unsigned bl1 = bv1 & 0xFFFFu, bh1 = bv1 >> 16;
+          pa += al0 + ah0 + al1 + ah1;
+          pb += bl0 + bh0 + bl1 + bh1;
+          paa += (u64)(al0 * al0) + (u64)(ah0 * ah0) +
+                 (u64)(al1 * al1) + (u64)(ah1 * ah1);
+          pbb += (u64)(bl0 * bl0) + (u64)(bh0 * bh0) +
+                 (u64)(bl1 * bl1) + (u64)(bh1 * bh1);
+          pab += (u64)(al0 * bl0) + (u64)(ah0 * bh0) +
+                 (u64)(al1 * bl1) + (u64)(ah1 * bh1);
+        }
+      };
+      int r = r00;
+      for (; r + rstride < c.ny; r += 2 * rstride) {
+        if (h && lx == 0) {
+          add1(ar[0], br[0]);
+          add1(ar[astep], br[bstep]);
+        }
+        if (tail && lx == rw - 1) {
+          add1(ar[c.nx - 1], br[c.nx - 1]);
+          add1(ar[astep + c.nx - 1], br[bstep + c.nx - 1]);
+        }
+        vrow2((const unsigned *)(ar + h), (const unsigned *)(br + h),
+              (const unsigned *)(ar + astep + h),
+              (const unsigned *)(br + bstep + h));
+        ar += 2 * astep;
+        br += 2 * bstep;
       }
-    }
-    for (; r < c.ny; r += rstride) {
-      const unsigned short *ar = abase + (a.oy + c.loy + r) * a.sx;
-      const unsigned short *br =
-          bbase + (b.oy + c.loy + c.sy + r) * b.sx;
-      for (int x = lx; x < c.nx; x += rw) {
-        unsigned av = ar[x], bv = br[x];
-        pa += av; pb += bv; paa += (u64)(av * av); pbb += (u64)(bv * bv);
-        pab += (u64)(av * bv);
+      for (; r < c.ny; r += rstride) {
+        if (h && lx == 0) add1(ar[0], br[0]);
+        if (tail && lx == rw - 1) add1(ar[c.nx - 1], br[c.nx - 1]);
+        const unsigned *a32 = (const unsigned *)(ar + h);
+        const unsigned *b32 = (const unsigned *)(br + h);
+        for (int p = lx; p < np2; p += rw) {
+          const unsigned av = a32[p], bv = b32[p];
+          const unsigned al = av & 0xFFFFu, ah = av >> 16;
+          const unsigned bl = bv & 0xFFFFu, bh = bv >> 16;
+          pa += al + ah;
+          pb += bl + bh;
+          paa += (u64)(al * al) + (u64)(ah * ah);
+          pbb += (u64)(bl * bl) + (u64)(bh * bh);
+          pab += (u64)(al * bl) + (u64)(ah * bh);
+        }
+        ar += astep;
+        br += bstep;
+      }
+    } else {
+      int r = r00;
+      for (; r + rstride < c.ny; r += 2 * rstride) {
+        for (int x = lx; x < c.nx; x += rw) {
+          const unsigned av0 = ar[x], bv0 = br[x];
+          const unsigned av1 = ar[astep + x], bv1 = br[bstep + x];
+          pa += av0 + av1;
+          pb += bv0 + bv1;
+          paa += (u64)(av0 * av0) + (u64)(av1 * av1);
+          pbb += (u64)(bv0 * bv0) + (u64)(bv1 * bv1);
+          pab += (u64)(av0 * bv0) + (u64)(av1 * bv1);
+        }
+        ar += 2 * astep;
+        br += 2 * bstep;
+      }
+      for (; r < c.ny; r += rstride) {
+        for (int x = lx; x < c.nx; x += rw) add1(ar[x], br[x]);
+        ar += astep;
+        br += bstep;
       }
     }
     pa = wave_sum_u64(pa); pb = wave_sum_u64(pb); paa = wave_sum_u64(paa);
@@ -1344,7 +1529,16 @@ extern "C" int bs_ctx_create(bs_ctx **out, int device_id) {
   (void)hipFuncSetAttribute((const void *)k_fft_pass,
                             hipFuncAttributeMaxDynamicSharedMemorySize,
                             160 * 1024);
-  (void)hipFuncSetAttribute((const void *)k_fft_z_fused,
+  (void)hipFuncSetAttribute((const void *)k_fft_z_fused<1>,
+                            hipFuncAttributeMaxDynamicSharedMemorySize,
+                            160 * 1024);
+  (void)hipFuncSetAttribute((const void *)k_fft_z_fused<2>,
+                            hipFuncAttributeMaxDynamicSharedMemorySize,
+                            160 * 1024);
+  (void)hipFuncSetAttribute((const void *)k_fft_z_fused<4>,
+                            hipFuncAttributeMaxDynamicSharedMemorySize,
+                            160 * 1024);
+  (void)hipFuncSetAttribute((const void *)k_fft_z_fused<8>,
                             hipFuncAttributeMaxDynamicSharedMemorySize,
                             160 * 1024);
   (void)hipFuncSetAttribute((const void *)k_fft_x_fwd,
@@ -1736,14 +1930,25 @@ static int stitch_phaseA(bs_ctx *c, bs_slot *sl, const bs_pair_desc &pd,
   { /* fused z chain: fwd z (A,B) + cross-power [PIN-EPS] + inv z in one
      * launch — the z spectra never round-trip through HBM */
     float scale = 1.0f / ((float)Px * (float)Py * (float)Pz);
-    size_t lds = ((Pz / 2) + 2 * (size_t)LPB_S * Pz) * sizeof(f2);
+    size_t lds = ((Pz / 2) + (size_t)LPB_S * Pz) * sizeof(f2);
     bs_tim tt(c, BS_K_FFT_Z_INV, sl->stream);
-    hipLaunchKernelGGL(k_fft_z_fused,
-                       dim3(std::min(4096L, (long)Py * nchunks)),
-                       dim3(LPB_S * TPL_S), lds, sl->stream, spec[0],
-                       spec[1], Pz, ilog2(Pz), (long)Py * Cxp, Cxp, Cx,
-                       nchunks, Py, sl->reg[0].mz, sl->reg[1].mz, scale,
-                       twz);
+    dim3 g(std::min(4096L, (long)Py * nchunks)), b(LPB_S * TPL_S);
+    auto zf = [&](auto kern) {
+      hipLaunchKernelGGL(kern, g, b, lds, sl->stream, spec[0], spec[1],
+                         Pz, ilog2(Pz), (long)Py * Cxp, Cxp, Cx, nchunks,
+                         Py, sl->reg[0].mz, sl->reg[1].mz, scale, twz);
+    };
+    /* n>=512: the occupancy-capped no-prefetch variant wins (A/B on
+     * hardware: 0.734 vs 0.842 ms at 512^3 — 4 blocks/CU beats the
+     * B-register prefetch at 2 blocks/CU) */
+    if (Pz <= 128)
+      zf(k_fft_z_fused<1>);
+    else if (Pz == 256)
+      zf(k_fft_z_fused<2>);
+    else if (Pz == 512)
+      zf(k_fft_z_fused_np<4>);
+    else
+      zf(k_fft_z_fused_np<8>);
   }
   {
     size_t lds = ((Py / 2) + (size_t)LPB_S * Py) * sizeof(f2);
@@ -1792,8 +1997,13 @@ static int stitch_phaseA(bs_ctx *c, bs_slot *sl, const bs_pair_desc &pd,
   if (rc2) return rc2;
   {
     bs_tim tt(c, BS_K_PEAK, sl->stream);
-    hipLaunchKernelGGL(k_peak_tile, dim3(npkwg), dim3(256), 0, sl->stream,
-                       sl->pcm, Px, Py, Pz, sl->wgpk);
+    static const bool pk_relax = getenv("BS_PEAK_RELAX") != nullptr;
+    if (pk_relax)
+      hipLaunchKernelGGL(k_peak_tile_relaxed, dim3(npkwg), dim3(256), 0,
+                         sl->stream, sl->pcm, Px, Py, Pz, sl->wgpk);
+    else
+      hipLaunchKernelGGL(k_peak_tile, dim3(npkwg), dim3(256), 0, sl->stream,
+                         sl->pcm, Px, Py, Pz, sl->wgpk);
   }
   {
     bs_tim tt(c, BS_K_PEAK_MERGE, sl->stream);
