@@ -315,6 +315,166 @@ __global__ __launch_bounds__(LPB_S *TPL_S) void k_fft_pass(
   }
 }
 
+/* ---- glds (async global->LDS DMA) double-buffered strided pass ---- */
+
+typedef __attribute__((address_space(3))) unsigned bs_lds_u32;
+typedef const __attribute__((address_space(1))) unsigned bs_glb_u32;
+
+__device__ __forceinline__ void bs_glds16(const void *g, void *l) {
+  __builtin_amdgcn_global_load_lds((bs_glb_u32 *)g, (bs_lds_u32 *)l, 16, 0,
+                                   0);
+}
+/* s_waitcnt immediates (gfx9 encoding): vmcnt[3:0|15:14], expcnt[6:4],
+ * lgkmcnt[13:8] */
+#define BS_WAIT_LGKM0 0xC07F /* lgkmcnt(0), vmcnt/expcnt unconstrained */
+#define BS_WAIT_VM0 0x3F70   /* vmcnt(0), lgkmcnt/expcnt unconstrained */
+
+/* fft_lds with RAW barriers (lgkmcnt(0) only): a glds issued for the
+ * OTHER buffer stays in flight across the FFT rounds — __syncthreads()
+ * would wait vmcnt(0) and drain it (cdna_hip_programming.md §5). */
+template <int ES, int TPL>
+__device__ __forceinline__ void fft_lds_raw(f2 *data, long base, int n,
+                                            int log2n, int tl, const f2 *tw,
+                                            int dir) {
+#define D_(e) data[base + (long)(e) * ES]
+#define RAWBAR()                                                            \
+  do {                                                                      \
+    __builtin_amdgcn_s_waitcnt(BS_WAIT_LGKM0);                              \
+    __builtin_amdgcn_s_barrier();                                           \
+  } while (0)
+  int h = 1;
+  if (log2n & 1) {
+    for (int bf = tl; bf < (n >> 1); bf += TPL) {
+      f2 u = D_(2 * bf), v = D_(2 * bf + 1);
+      D_(2 * bf) = {u.x + v.x, u.y + v.y};
+      D_(2 * bf + 1) = {u.x - v.x, u.y - v.y};
+    }
+    h = 2;
+    RAWBAR();
+  }
+  for (; h < n; h <<= 2) {
+    const int q = n >> 2;
+    const int s1 = n / (2 * h), s2 = n / (4 * h);
+    for (int g = tl; g < q; g += TPL) {
+      int off = g % h, blk = g / h;
+      int i = blk * 4 * h + off;
+      f2 a = D_(i), b = D_(i + h), c = D_(i + 2 * h), d = D_(i + 3 * h);
+      f2 w1 = tw[off * s1];
+      if (dir < 0) w1.y = -w1.y;
+      f2 t1 = cmul(b, w1), t2 = cmul(d, w1);
+      f2 A = {a.x + t1.x, a.y + t1.y}, B = {a.x - t1.x, a.y - t1.y};
+      f2 Cc = {c.x + t2.x, c.y + t2.y}, Dd = {c.x - t2.x, c.y - t2.y};
+      f2 w2a = tw[off * s2], w2b = tw[(off + h) * s2];
+      if (dir < 0) {
+        w2a.y = -w2a.y;
+        w2b.y = -w2b.y;
+      }
+      f2 u1 = cmul(Cc, w2a), u2 = cmul(Dd, w2b);
+      D_(i) = {A.x + u1.x, A.y + u1.y};
+      D_(i + 2 * h) = {A.x - u1.x, A.y - u1.y};
+      D_(i + h) = {B.x + u2.x, B.y + u2.y};
+      D_(i + 3 * h) = {B.x - u2.x, B.y - u2.y};
+    }
+    RAWBAR();
+  }
+#undef D_
+#undef RAWBAR
+}
+
+/* Double-buffered strided C2C pass: while the FFT rounds run on one
+ * LDS buffer, the NEXT line-group's transpose loads stream into the
+ * other via global_load_lds (async DMA — no staging registers, no
+ * early drain; the round-1 T14 register prefetch lost to its +90 VGPR,
+ * DESIGN.md §4). LDS dest is wave-uniform + lane*16, so the
+ * bit-reversal swizzle moves into each lane's GLOBAL source address:
+ * slot s holds element e = brev(s>>2) of line pair 2*(s&3)
+ * (cdna_hip_programming.md rule 21). 66 KB LDS -> 2 WGs/CU; the bet is
+ * intra-WG overlap over occupancy. Grid should be ~2 WGs/CU so every
+ * WG runs many iterations. In-place (in==out) safe: each WG's loads
+ * and stores touch only its own grid-strided line groups. */
+template <int NE>
+__global__ __launch_bounds__(LPB_S *TPL_S) void k_fft_pass_glds(
+    f2 *inout, int n, int log2n, long estride, long gstride, int nlines,
+    int nchunks, int ngroups, int valid, int dir, const f2 *twg) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  f2 *tw = (f2 *)smem;
+  f2 *buf[2] = {tw + (n >> 1), tw + (n >> 1) + (long)LPB_S * n};
+  const int tid = threadIdx.x;
+  const int line = tid % LPB_S, tl = tid / LPB_S;
+  const int lane = tid & 63, wv = tid >> 6;
+  for (int i = tid; i < (n >> 1); i += LPB_S * TPL_S) tw[i] = twg[i];
+  constexpr int NPAIR = LPB_S / 2;
+  constexpr int ESTR = (LPB_S * TPL_S) / NPAIR;
+  const int pl = tid & (NPAIR - 1), t2 = tid / NPAIR;
+  const long nwg = (long)ngroups * nchunks;
+  /* issue the glds transfers of one line-group into buf[which] */
+  auto issue = [&](long wg, int which) {
+    const int group = (int)(wg / nchunks);
+    const int x2b = (int)(wg % nchunks) * LPB_S;
+    const long gbase = (long)group * gstride + x2b;
+    f2 *db = buf[which];
+#pragma unroll
+    for (int k = 0; k < NE; ++k) {
+      const int sbase = (k * 8 + wv) * 64; /* 16B slots, wave-uniform */
+      const int s = sbase + lane;
+      if (s >> 2 >= n) continue; /* n < 128*NE tail */
+      const int e = (int)brev_n((unsigned)(s >> 2), log2n);
+      const int lp = s & 3;
+      /* wave-uniform LDS base; hardware lands lane i at base+16*i */
+      f2 *dst = db + (long)sbase * 2;
+      if (e < valid) {
+        bs_glds16(&inout[gbase + 2 * lp + (long)e * estride], dst);
+      } else {
+        *(float4 *)((char *)dst + (size_t)lane * 16) =
+            float4{0, 0, 0, 0};
+      }
+    }
+  };
+  /* first group */
+  long wg = blockIdx.x;
+  if (wg < nwg) issue(wg, 0);
+  __builtin_amdgcn_s_waitcnt(BS_WAIT_VM0);
+  __syncthreads(); /* tables + first buffer ready */
+  int cur = 0;
+  for (; wg < nwg; wg += gridDim.x) {
+    const long nxt = wg + gridDim.x;
+    if (nxt < nwg) issue(nxt, cur ^ 1); /* in flight during the FFT */
+    fft_lds_raw<LPB_S, TPL_S>(buf[cur], (long)line, n, log2n, tl, tw, dir);
+    { /* store (natural order; per-line guard for the partial chunk) */
+      const int group = (int)(wg / nchunks);
+      const int x2 = (int)(wg % nchunks) * LPB_S + 2 * pl;
+      const long base2 = (long)group * gstride + x2;
+      f2 *db = buf[cur];
+      if (x2 + 1 < nlines) {
+#pragma unroll
+        for (int k = 0; k < NE; ++k) {
+          const int e = t2 + k * ESTR;
+          if (e >= n) break;
+          *(float4 *)&inout[base2 + e * estride] =
+              *(const float4 *)&db[(long)e * LPB_S + 2 * pl];
+        }
+      } else {
+        for (int l = 0; l < 2; ++l) {
+          const int x = x2 + l;
+          if (x >= nlines) continue;
+          const long base = (long)group * gstride + x;
+#pragma unroll
+          for (int k = 0; k < NE; ++k) {
+            const int e = t2 + k * ESTR;
+            if (e >= n) break;
+            inout[base + e * estride] = db[(long)e * LPB_S + 2 * pl + l];
+          }
+        }
+      }
+    }
+    /* next buffer's DMA done + all lanes past the store reads */
+    __builtin_amdgcn_s_waitcnt(BS_WAIT_VM0);
+    __builtin_amdgcn_s_waitcnt(BS_WAIT_LGKM0);
+    __builtin_amdgcn_s_barrier();
+    cur ^= 1;
+  }
+}
+
 /* Fused z chain: forward z-FFT of BOTH spectra + cross-power normalise
  * + inverse z-FFT in ONE kernel (replaces two k_fft_pass(+1) launches
  * and the fused-crosspower k_fft_pass(-1) launch). The z spectra are
@@ -1529,6 +1689,15 @@ extern "C" int bs_ctx_create(bs_ctx **out, int device_id) {
   (void)hipFuncSetAttribute((const void *)k_fft_pass,
                             hipFuncAttributeMaxDynamicSharedMemorySize,
                             160 * 1024);
+  (void)hipFuncSetAttribute((const void *)k_fft_pass_glds<2>,
+                            hipFuncAttributeMaxDynamicSharedMemorySize,
+                            160 * 1024);
+  (void)hipFuncSetAttribute((const void *)k_fft_pass_glds<4>,
+                            hipFuncAttributeMaxDynamicSharedMemorySize,
+                            160 * 1024);
+  (void)hipFuncSetAttribute((const void *)k_fft_pass_glds<8>,
+                            hipFuncAttributeMaxDynamicSharedMemorySize,
+                            160 * 1024);
   (void)hipFuncSetAttribute((const void *)k_fft_z_fused<1>,
                             hipFuncAttributeMaxDynamicSharedMemorySize,
                             160 * 1024);
@@ -1917,15 +2086,39 @@ static int stitch_phaseA(bs_ctx *c, bs_slot *sl, const bs_pair_desc &pd,
     }
   }
   int nchunks = (Cx + LPB_S - 1) / LPB_S;
+  static const bool yglds = getenv("BS_Y_GLDS") != nullptr;
+  /* glds double-buffered y-pass launcher (n >= 256; see
+   * k_fft_pass_glds) — grid sized to residency (2 WGs/CU) */
+  auto launch_y = [&](f2 *sp, int ngroups, int valid, int dir) {
+    if (yglds && Py >= 256) {
+      size_t lds2 = ((Py / 2) + 2 * (size_t)LPB_S * Py) * sizeof(f2);
+      long nwg = (long)ngroups * nchunks;
+      dim3 g(std::min(512L, nwg)), b(LPB_S * TPL_S);
+      if (Py == 256)
+        hipLaunchKernelGGL(k_fft_pass_glds<2>, g, b, lds2, sl->stream, sp,
+                           Py, ilog2(Py), Cxp, (long)Py * Cxp, Cx, nchunks,
+                           ngroups, valid, dir, twy);
+      else if (Py == 512)
+        hipLaunchKernelGGL(k_fft_pass_glds<4>, g, b, lds2, sl->stream, sp,
+                           Py, ilog2(Py), Cxp, (long)Py * Cxp, Cx, nchunks,
+                           ngroups, valid, dir, twy);
+      else
+        hipLaunchKernelGGL(k_fft_pass_glds<8>, g, b, lds2, sl->stream, sp,
+                           Py, ilog2(Py), Cxp, (long)Py * Cxp, Cx, nchunks,
+                           ngroups, valid, dir, twy);
+    } else {
+      size_t lds = ((Py / 2) + (size_t)LPB_S * Py) * sizeof(f2);
+      hipLaunchKernelGGL(k_fft_pass,
+                         dim3(std::min(4096L, (long)ngroups * nchunks)),
+                         dim3(LPB_S * TPL_S), lds, sl->stream, sp,
+                         (const f2 *)nullptr, sp, Py, ilog2(Py), Cxp,
+                         (long)Py * Cxp, Cx, nchunks, ngroups, valid, dir,
+                         1.0f, twy);
+    }
+  };
   for (int t = 0; t < 2; ++t) {
-    size_t lds = ((Py / 2) + (size_t)LPB_S * Py) * sizeof(f2);
     bs_tim tt(c, BS_K_FFT_Y_FWD, sl->stream);
-    hipLaunchKernelGGL(k_fft_pass,
-                       dim3(std::min(4096L, (long)sl->reg[t].mz * nchunks)),
-                       dim3(LPB_S * TPL_S), lds, sl->stream, spec[t],
-                       (const f2 *)nullptr, spec[t], Py, ilog2(Py), Cxp,
-                       (long)Py * Cxp, Cx, nchunks, sl->reg[t].mz,
-                       sl->reg[t].my, +1, 1.0f, twy);
+    launch_y(spec[t], sl->reg[t].mz, sl->reg[t].my, +1);
   }
   { /* fused z chain: fwd z (A,B) + cross-power [PIN-EPS] + inv z in one
      * launch — the z spectra never round-trip through HBM */
@@ -1953,10 +2146,8 @@ static int stitch_phaseA(bs_ctx *c, bs_slot *sl, const bs_pair_desc &pd,
   {
     size_t lds = ((Py / 2) + (size_t)LPB_S * Py) * sizeof(f2);
     bs_tim tt(c, BS_K_FFT_Y_INV, sl->stream);
-    hipLaunchKernelGGL(k_fft_pass, dim3(std::min(4096L, (long)Pz * nchunks)),
-                       dim3(LPB_S * TPL_S), lds, sl->stream, spec[0],
-                       (const f2 *)nullptr, spec[0], Py, ilog2(Py), Cxp,
-                       (long)Py * Cxp, Cx, nchunks, Pz, Py, -1, 1.0f, twy);
+    (void)lds;
+    launch_y(spec[0], Pz, Py, -1);
   }
   {
     long nlines = (long)Pz * Py;
