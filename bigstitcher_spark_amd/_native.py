@@ -298,10 +298,13 @@ class Context:
     def fuse_volume(self, views, vol_min, vol_dims, downsamplings=None,
                     fusion_type=FUSION_AVG_BLEND, out_dtype=np.float32,
                     min_intensity=0.0, max_intensity=65535.0,
-                    masks=False, mask_offset=(0.0, 0.0, 0.0)):
+                    masks=False, mask_offset=(0.0, 0.0, 0.0),
+                    out_buffers=None):
         """Whole-volume fusion + pyramid. downsamplings: list of (dx,dy,dz)
         absolute factors per level (level 0 must be (1,1,1)). Returns a
-        list of (nz,ny,nx) arrays, one per level."""
+        list of (nz,ny,nx) arrays, one per level. out_buffers: optional
+        pre-allocated per-level arrays to write into (avoids the
+        first-touch page-fault cost on the D2H path for repeat calls)."""
         if downsamplings is None:
             downsamplings = [(1, 1, 1)]
         nv, nl = len(views), len(downsamplings)
@@ -326,7 +329,12 @@ class Context:
         outs, outptrs = [], (C.c_void_p * nl)()
         for l, lvl in enumerate(downsamplings):
             d = [(int(vol_dims[k]) + lvl[k] - 1) // lvl[k] for k in range(3)]
-            a = np.empty((d[2], d[1], d[0]), dt)
+            if out_buffers is not None:
+                a = out_buffers[l]
+                assert a.shape == (d[2], d[1], d[0]) and a.dtype == dt \
+                    and a.flags["C_CONTIGUOUS"]
+            else:
+                a = np.empty((d[2], d[1], d[0]), dt)
             outs.append(a)
             outptrs[l] = a.ctypes.data
         self._check(

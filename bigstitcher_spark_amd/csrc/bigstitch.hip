@@ -37,6 +37,8 @@
 #include <cstdio>
 #include <cstdlib>
 #include <cstring>
+#include <condition_variable>
+#include <functional>
 #include <map>
 #include <mutex>
 #include <thread>
@@ -1293,7 +1295,7 @@ __global__ __launch_bounds__(256) void k_fuse(
     const bs_dev_view *views, const int *vidx, int nv, long bmx, long bmy,
     long bmz, int bx, int by, int bz, int ftype, int dtype, float minI,
     float invRange /* type_max/(maxI-minI) */, void *out, long out_off,
-    long out_row, long out_slice) {
+    long out_row, long out_slice, int use_riv) {
   /* stage the (culled, small) per-block view table in LDS once */
   __shared__ bs_dev_view sv[BS_MAX_BLK_VIEWS];
   int nvs = min(nv, BS_MAX_BLK_VIEWS);
@@ -1304,8 +1306,62 @@ __global__ __launch_bounds__(256) void k_fuse(
   }
   __syncthreads();
   long nrows = (long)by * bz;
-  for (long row = blockIdx.x; row < nrows; row += gridDim.x) {
+  /* XCD-aware row mapping: same-XCD blocks (hardware dispatches
+   * blockIdx round-robin over the 8 XCDs) cover contiguous row ranges
+   * so the trilinear stencil's row reuse stays in one XCD's L2
+   * (gridDim.x is launched as a multiple of 8). */
+  long vb = blockIdx.x;
+  if ((gridDim.x & 7) == 0)
+    vb = (long)(blockIdx.x & 7) * (gridDim.x >> 3) + (blockIdx.x >> 3);
+  /* per-row view clip intervals: for each (row, view) the CONSERVATIVE
+   * x-range where the inverse affine can land in bounds (each axis'
+   * c + s*x in [0, dim-1] solved as an interval, expanded by 1 px so
+   * the exact per-voxel [PIN-BOUNDS] test keeps the edge semantics),
+   * plus the hoisted y/z blend product when py/pz are x-independent
+   * (inv[4]==inv[8]==0, the axis-aligned common case). Kills the
+   * per-voxel affine+bounds work for every view that cannot touch the
+   * voxel — with superblock-granularity culling most can't. */
+  __shared__ float riv_lo[BS_MAX_BLK_VIEWS], riv_hi[BS_MAX_BLK_VIEWS];
+  __shared__ float riv_wyz[BS_MAX_BLK_VIEWS]; /* -1 = per-voxel */
+  for (long row = vb; row < nrows; row += gridDim.x) {
    int y = (int)(row % by), z = (int)(row / by);
+   if (use_riv) {
+     __syncthreads(); /* previous row's interval reads done */
+     const int k = threadIdx.x;
+     if (k < nvs) {
+       const bs_dev_view &v = sv[k];
+       const float wy = (float)(bmy + y), wz = (float)(bmz + z);
+       float lo = 0.0f, hi = (float)bx;
+       for (int d = 0; d < 3 && lo < hi; ++d) {
+         const float s = v.inv[d * 4 + 0];
+         const float cc = v.inv[d * 4 + 0] * (float)bmx +
+                          v.inv[d * 4 + 1] * wy + v.inv[d * 4 + 2] * wz +
+                          v.inv[d * 4 + 3];
+         const float D = (float)((d == 0 ? v.nx : d == 1 ? v.ny : v.nz) -
+                                 1);
+         if (s == 0.0f) {
+           if (!(cc >= 0.0f && cc <= D)) hi = lo; /* empty */
+         } else {
+           float t0 = (0.0f - cc) / s, t1 = (D - cc) / s;
+           float a = fminf(t0, t1) - 1.0f, b2 = fmaxf(t0, t1) + 1.0f;
+           lo = fmaxf(lo, fminf(a, 1.0e9f));
+           hi = fminf(hi, fmaxf(fminf(b2, 1.0e9f), -1.0e9f));
+         }
+       }
+       riv_lo[k] = lo;
+       riv_hi[k] = hi;
+       float wyz = -1.0f;
+       if (ftype == BS_FUSION_AVG_BLEND && v.inv[4] == 0.0f &&
+           v.inv[8] == 0.0f) {
+         const float py = v.inv[5] * wy + v.inv[6] * wz + v.inv[7];
+         const float pz = v.inv[9] * wy + v.inv[10] * wz + v.inv[11];
+         wyz = blend_w(py, v.ny, v.border[1], v.range[1]) *
+               blend_w(pz, v.nz, v.border[2], v.range[2]);
+       }
+       riv_wyz[k] = wyz;
+     }
+     __syncthreads();
+   }
    for (int x = threadIdx.x; x < bx; x += blockDim.x) {
     long i = out_off + (long)z * out_slice + (long)y * out_row + x;
     float wx = (float)(bmx + x), wy = (float)(bmy + y), wz = (float)(bmz + z);
@@ -1313,6 +1369,9 @@ __global__ __launch_bounds__(256) void k_fuse(
     float best_dist = -1.0f;
     bool any = false;
     for (int k = 0; k < nv; ++k) {
+      if (use_riv && k < nvs &&
+          !((float)x >= riv_lo[k] && (float)x < riv_hi[k]))
+        continue; /* conservatively outside this view for this row */
       const bs_dev_view &v = k < nvs ? sv[k] : views[vidx[k]];
       float px = v.inv[0] * wx + v.inv[1] * wy + v.inv[2] * wz + v.inv[3];
       float py = v.inv[4] * wx + v.inv[5] * wy + v.inv[6] * wz + v.inv[7];
@@ -1371,10 +1430,14 @@ __global__ __launch_bounds__(256) void k_fuse(
         val = ab2[0] * val + ab2[1];
       }
       float w = 1.0f;
-      if (ftype == BS_FUSION_AVG_BLEND)
-        w = blend_w(px, v.nx, v.border[0], v.range[0]) *
-            blend_w(py, v.ny, v.border[1], v.range[1]) *
-            blend_w(pz, v.nz, v.border[2], v.range[2]);
+      if (ftype == BS_FUSION_AVG_BLEND) {
+        const float wyz = use_riv && k < nvs && riv_wyz[k] >= 0.0f
+                              ? riv_wyz[k]
+                              : blend_w(py, v.ny, v.border[1], v.range[1]) *
+                                    blend_w(pz, v.nz, v.border[2],
+                                            v.range[2]);
+        w = blend_w(px, v.nx, v.border[0], v.range[0]) * wyz;
+      }
       if (ftype == BS_FUSION_MAX_INTENSITY) {
         vmax = (any && vmax > val) ? vmax : val;
         any = true;
@@ -1570,6 +1633,64 @@ struct bs_hostcand {
   long n;
 };
 
+/* Persistent host memcpy pool for the staged D2H copy-out leg: a
+ * per-chunk std::thread spawn (round 1) cost ~50-100 us x 12 threads x
+ * ~60 chunks per volume — milliseconds of pure spawn overhead on the
+ * D2H critical path. */
+struct bs_pool {
+  std::vector<std::thread> th;
+  std::mutex mu;
+  std::condition_variable cv, done;
+  char *dst = nullptr;
+  const char *src = nullptr;
+  size_t len = 0;
+  long epoch = 0;
+  int remaining = 0;
+  bool stop = false;
+  int n = 0;
+  void start(int nthreads) {
+    n = nthreads;
+    for (int i = 0; i < n; ++i)
+      th.emplace_back([this, i]() {
+        long seen = 0;
+        for (;;) {
+          std::unique_lock<std::mutex> lk(mu);
+          cv.wait(lk, [&] { return stop || epoch != seen; });
+          if (stop) return;
+          seen = epoch;
+          char *d = dst;
+          const char *s2 = src;
+          size_t l = len;
+          lk.unlock();
+          size_t part = (l + n - 1) / n;
+          size_t o = (size_t)i * part;
+          if (o < l) memcpy(d + o, s2 + o, std::min(part, l - o));
+          lk.lock();
+          if (--remaining == 0) done.notify_all();
+        }
+      });
+  }
+  void copy(char *d, const char *s2, size_t l) {
+    if (th.empty()) start(12);
+    std::unique_lock<std::mutex> lk(mu);
+    dst = d;
+    src = s2;
+    len = l;
+    remaining = n;
+    ++epoch;
+    cv.notify_all();
+    done.wait(lk, [&] { return remaining == 0; });
+  }
+  ~bs_pool() {
+    {
+      std::lock_guard<std::mutex> lk(mu);
+      stop = true;
+    }
+    cv.notify_all();
+    for (auto &t : th) t.join();
+  }
+};
+
 /* One pipeline slot: its own stream + workspace so two pairs are in
  * flight (pair i's FFT/peak chain overlaps pair i-1's r-test and the
  * host-side candidate building). */
@@ -1611,6 +1732,8 @@ struct bs_slot {
 struct bs_ctx {
   int dev;
   hipStream_t stream; /* default stream: views, synth, fusion */
+  hipStream_t copy_stream = nullptr; /* D2H overlap (fusion volume) */
+  bs_pool cpool; /* host memcpy workers for staged_d2h */
   std::string err;
   std::map<int32_t, bs_view_rec> views;
   std::map<int, f2 *> twiddles;
@@ -1679,7 +1802,8 @@ extern "C" int bs_ctx_create(bs_ctx **out, int device_id) {
   bs_ctx *c = new bs_ctx();
   c->dev = device_id;
   if (hipSetDevice(device_id) != hipSuccess ||
-      hipStreamCreate(&c->stream) != hipSuccess) {
+      hipStreamCreate(&c->stream) != hipSuccess ||
+      hipStreamCreate(&c->copy_stream) != hipSuccess) {
     delete c;
     g_err = "hip init failed";
     return BS_ENODEV;
@@ -1789,6 +1913,7 @@ extern "C" void bs_ctx_destroy(bs_ctx *c) {
   (void)hipFree(c->dvidx);
   (void)hipFree(c->dblobs);
   (void)hipStreamDestroy(c->stream);
+  if (c->copy_stream) (void)hipStreamDestroy(c->copy_stream);
   delete c;
 }
 
@@ -2561,9 +2686,12 @@ extern "C" int bs_fuse_blocks(bs_ctx *c, const bs_fuse_view *views,
     int nvb = (int)(view_idx_offsets[i + 1] - view_idx_offsets[i]);
     {
       long nrows_f = (long)bd.size[1] * bd.size[2];
+      long gfb = std::max(8L, std::min(4096L, nrows_f) & ~7L);
+      static const bool noswiz = getenv("BS_FUSE_NOSWIZ") != nullptr;
+      if (noswiz && gfb > 8) gfb -= 1; /* non-x8 grid = identity map */
       bs_tim tt(c, BS_K_FUSE);
       if (prm->masks)
-        hipLaunchKernelGGL(k_mask, dim3(std::min(4096L, nrows_f)),
+        hipLaunchKernelGGL(k_mask, dim3(gfb),
                            dim3(256), 0, c->stream, c->dviews,
                            c->dvidx + view_idx_offsets[i], nvb, bd.min[0],
                            bd.min[1], bd.min[2], (int)bd.size[0],
@@ -2573,7 +2701,7 @@ extern "C" int bs_fuse_blocks(bs_ctx *c, const bs_fuse_view *views,
                            (float)prm->mask_offset[2], c->fuse_out, 0L,
                            (long)bd.size[0], (long)bd.size[0] * bd.size[1]);
       else
-        hipLaunchKernelGGL(k_fuse, dim3(std::min(4096L, nrows_f)),
+        hipLaunchKernelGGL(k_fuse, dim3(gfb),
                            dim3(256), 0, c->stream, c->dviews,
                            c->dvidx + view_idx_offsets[i], nvb, bd.min[0],
                            bd.min[1], bd.min[2], (int)bd.size[0],
@@ -2581,7 +2709,8 @@ extern "C" int bs_fuse_blocks(bs_ctx *c, const bs_fuse_view *views,
                            prm->fusion_type, prm->out_dtype,
                            (float)prm->min_intensity, invRange, c->fuse_out,
                            0L, (long)bd.size[0],
-                           (long)bd.size[0] * bd.size[1]);
+                           (long)bd.size[0] * bd.size[1],
+                           getenv("BS_FUSE_NORIV") ? 0 : 1);
     }
     CHK(c, hipMemcpyAsync(out_blocks[i], c->fuse_out, nvox * esz,
                           hipMemcpyDeviceToHost, c->stream));
@@ -2600,7 +2729,9 @@ extern "C" int bs_fuse_blocks(bs_ctx *c, const bs_fuse_view *views,
  * ~3x slower and blocks; this overlaps the PCIe copy with the host-side
  * memcpy out of the staging buffer). */
 static int staged_d2h(bs_ctx *c, const void *dsrc, void *hdst,
-                      size_t bytes) {
+                      size_t bytes, hipStream_t st,
+                      const std::function<void(size_t, size_t)> &gate =
+                          nullptr) {
   for (int i = 0; i < 2; ++i) {
     if (!c->hstage[i]) {
       if (hipHostMalloc(&c->hstage[i], BS_STAGE_BYTES) != hipSuccess) {
@@ -2616,9 +2747,10 @@ static int staged_d2h(bs_ctx *c, const void *dsrc, void *hdst,
     if (k < nchunks) {
       size_t off = k * BS_STAGE_BYTES;
       size_t len = std::min((size_t)BS_STAGE_BYTES, bytes - off);
+      if (gate) gate(off, len); /* e.g. wait for the producing kernels */
       CHK(c, hipMemcpyAsync(c->hstage[k & 1], (const char *)dsrc + off, len,
-                            hipMemcpyDeviceToHost, c->stream));
-      CHK(c, hipEventRecord(c->stage_ev[k & 1], c->stream));
+                            hipMemcpyDeviceToHost, st));
+      CHK(c, hipEventRecord(c->stage_ev[k & 1], st));
     }
     if (k > 0) {
       size_t off = (k - 1) * BS_STAGE_BYTES;
@@ -2629,16 +2761,7 @@ static int staged_d2h(bs_ctx *c, const void *dsrc, void *hdst,
       const char *src = (const char *)c->hstage[(k - 1) & 1];
       char *dst = (char *)hdst + off;
       if (len >= (8 << 20)) {
-        const int NT = 6;
-        std::thread th[NT];
-        size_t part = (len + NT - 1) / NT;
-        for (int t = 0; t < NT; ++t) {
-          size_t o = t * part, l = std::min(part, len - std::min(len, o));
-          th[t] = std::thread([dst, src, o, l]() {
-            if (l) memcpy(dst + o, src + o, l);
-          });
-        }
-        for (int t = 0; t < NT; ++t) th[t].join();
+        c->cpool.copy(dst, src, len);
       } else {
         memcpy(dst, src, len);
       }
@@ -2746,13 +2869,22 @@ extern "C" int bs_fuse_volume(bs_ctx *c, const bs_fuse_view *views,
                           hipMemcpyHostToDevice, c->stream));
   const long vrow = ldims[0][0];
   const long vslice = ldims[0][0] * ldims[0][1];
+  /* per-z-slab completion events so the level-0 D2H (on copy_stream)
+   * starts as soon as the slab's blocks finished, overlapping the
+   * remaining fusion + pyramid kernels (the round-1 path serialized
+   * the whole D2H after all kernels) */
+  std::vector<hipEvent_t> slab_ev;
+  std::vector<long> slab_zend;
   for (size_t b = 0; b < fblocks.size(); ++b) {
     auto &fb = fblocks[b];
     int nvb = (int)(offs[b + 1] - offs[b]);
     long nrows_f = fb[4] * fb[5];
+    long gfb = std::max(8L, std::min(4096L, nrows_f) & ~7L);
+    static const bool noswiz = getenv("BS_FUSE_NOSWIZ") != nullptr;
+    if (noswiz && gfb > 8) gfb -= 1; /* non-x8 grid = identity map */
     bs_tim tt(c, BS_K_FUSE, c->stream);
     if (prm->masks)
-      hipLaunchKernelGGL(k_mask, dim3(std::min(4096L, nrows_f)), dim3(256),
+      hipLaunchKernelGGL(k_mask, dim3(gfb), dim3(256),
                          0, c->stream, c->dviews, c->dvidx + offs[b], nvb,
                          fb[0], fb[1], fb[2], (int)fb[3], (int)fb[4],
                          (int)fb[5], prm->out_dtype,
@@ -2762,13 +2894,20 @@ extern "C" int bs_fuse_volume(bs_ctx *c, const bs_fuse_view *views,
                          (long)fb[2] * vslice + fb[1] * vrow + fb[0], vrow,
                          vslice);
     else
-      hipLaunchKernelGGL(k_fuse, dim3(std::min(4096L, nrows_f)), dim3(256),
+      hipLaunchKernelGGL(k_fuse, dim3(gfb), dim3(256),
                          0, c->stream, c->dviews, c->dvidx + offs[b], nvb,
                          fb[0], fb[1], fb[2], (int)fb[3], (int)fb[4],
                          (int)fb[5], prm->fusion_type, prm->out_dtype,
                          (float)prm->min_intensity, invRange, dlvl[0],
                          (long)fb[2] * vslice + fb[1] * vrow + fb[0], vrow,
-                         vslice);
+                         vslice, getenv("BS_FUSE_NORIV") ? 0 : 1);
+    if (b + 1 == fblocks.size() || fblocks[b + 1][2] != fb[2]) {
+      hipEvent_t ev;
+      CHK(c, hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+      CHK(c, hipEventRecord(ev, c->stream));
+      slab_ev.push_back(ev);
+      slab_zend.push_back(fb[2] + fb[5]); /* z rows complete below this */
+    }
   }
   /* pyramid levels */
   for (int l = 1; l < nlevels; ++l) {
@@ -2801,16 +2940,38 @@ extern "C" int bs_fuse_volume(bs_ctx *c, const bs_fuse_view *views,
                          (int)ldims[l][0], (int)ldims[l][1],
                          (int)ldims[l][2], rx, ry, rz);
   }
-  /* staged D2H of every level */
-  for (int l = 0; l < nlevels; ++l) {
+  /* level 0 D2H on the copy stream, gated per chunk on the z-slab
+   * events — overlaps the remaining fusion blocks and the pyramid
+   * kernels still running on c->stream */
+  {
+    const size_t plane = (size_t)ldims[0][0] * ldims[0][1] * esz;
+    size_t bytes = lbytes[0];
+    auto gate = [&](size_t off, size_t len) {
+      const long zhi = (long)((off + len + plane - 1) / plane);
+      for (size_t i = 0; i < slab_ev.size(); ++i)
+        if (slab_zend[i] >= zhi || i + 1 == slab_ev.size()) {
+          (void)hipStreamWaitEvent(c->copy_stream, slab_ev[i], 0);
+          break;
+        }
+    };
+    rc = staged_d2h(c, dlvl[0], level_buffers[0], bytes, c->copy_stream,
+                    gate);
+    if (rc) {
+      cleanup();
+      return rc;
+    }
+  }
+  for (int l = 1; l < nlevels; ++l) {
     size_t bytes = (size_t)ldims[l][0] * ldims[l][1] * ldims[l][2] * esz;
-    rc = staged_d2h(c, dlvl[l], level_buffers[l], bytes);
+    rc = staged_d2h(c, dlvl[l], level_buffers[l], bytes, c->stream);
     if (rc) {
       cleanup();
       return rc;
     }
   }
   CHK(c, hipStreamSynchronize(c->stream));
+  CHK(c, hipStreamSynchronize(c->copy_stream));
+  for (auto ev : slab_ev) (void)hipEventDestroy(ev);
   c->stats.blocks += (long long)fblocks.size();
   flush_stats(c);
   return BS_OK;

@@ -45,11 +45,13 @@ def main():
         blocks.append((tuple(off), tuple(bsz)))
         vlists.append(host.find_overlapping_views(cull, off, bsz))
     dt = np.float32 if args.dtype == "float32" else np.uint16
+    outbuf = [np.empty((dims[2], dims[1], dims[0]), dt)]
     def run():
         return ctx.fuse_volume(views, (0, 0, 0), dims,
                                downsamplings=[(1, 1, 1)],
                                fusion_type=FUSION_AVG_BLEND, out_dtype=dt,
-                               min_intensity=0, max_intensity=65535)
+                               min_intensity=0, max_intensity=65535,
+                               out_buffers=outbuf)
     for _ in range(args.warmup):
         run()
     ctx.reset_stats()
