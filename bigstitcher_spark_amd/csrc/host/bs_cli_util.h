@@ -158,6 +158,64 @@ inline void tbbox(const M34 &m, const long long dims[3], double lo[3],
       }
 }
 
+/* ---- multi-resolution input levels (bdv.n5 pyramid) ---- */
+
+struct MipLevel {
+  int level = 0;
+  long long f[3] = {1, 1, 1};      /* absolute downsampling factors */
+  std::vector<long long> dims;     /* level dataset dims (x,y,z) */
+};
+
+/* [PIN-MIP] mipmap transform of a box-averaged level: level coords ->
+ * full-res local coords, x0 = f*xl + (f-1)/2 (the BDV default mipmap
+ * transform; artifact un-vendored). */
+
+/* Pick the best level for sampling under sourceToWorld, restating
+ * ViewUtil.ImgAndMipmapTransform.forBestResolution (reference
+ * ViewUtil.java:425-493, logic copied there from
+ * FusionTools.fuseVirtual): step size per axis = column norm of
+ * (sourceToWorld o mip); a level is valid when every step < 1.02 or
+ * approx equals (+-0.02) the level-0 step; among valid levels take the
+ * largest factor product. */
+inline int pick_level_for_transform(const M34 &model,
+                                    const std::vector<MipLevel> &levels) {
+  if (levels.size() <= 1) return 0;
+  const float acceptedError = 0.02f;
+  float size0[3] = {0, 0, 0};
+  int best = 0;
+  double bestScaling = 0;
+  for (size_t l = 0; l < levels.size(); ++l) {
+    const auto &lv = levels[l];
+    float size[3];
+    for (int d = 0; d < 3; ++d) {
+      double s2 = 0;
+      for (int i = 0; i < 3; ++i) {
+        double e = model[i * 4 + d] * (double)lv.f[d];
+        s2 += e * e;
+      }
+      size[d] = (float)std::sqrt(s2);
+    }
+    double total = (double)lv.f[0] * lv.f[1] * lv.f[2];
+    if (l == 0) {
+      for (int d = 0; d < 3; ++d) size0[d] = size[d];
+      bestScaling = total;
+      continue;
+    }
+    bool valid = true;
+    for (int d = 0; d < 3; ++d)
+      if (!(size[d] < 1.0f + acceptedError ||
+            std::fabs(size[d] - size0[d]) <= acceptedError)) {
+        valid = false;
+        break;
+      }
+    if (valid && total > bestScaling) {
+      bestScaling = total;
+      best = (int)l;
+    }
+  }
+  return best;
+}
+
 }  // namespace bscli
 
 #endif

@@ -16,6 +16,7 @@
 #include "../../../include/bigstitch.h"
 #include "bs_cli_util.h"
 #include "bs_n5.h"
+#include "bs_mip.h"
 #include "bs_zarr.h"
 #include "bs_spimdata.h"
 
@@ -229,18 +230,6 @@ int main(int argc, char **argv) {
       auto r = sd.regs.find({tp, s.id});
       if (r == sd.regs.end()) continue;
       if (!selset.count({tp, s.id})) continue;
-      std::vector<uint16_t> vox;
-      std::vector<long long> vdims;
-      if (!in_n5.read_volume_u16(bssd::SpimData::image_dataset(s.id, tp),
-                                 &vox, &vdims)) {
-        fprintf(stderr, "cannot read view tp=%d setup=%d\n", tp, s.id);
-        return 1;
-      }
-      int64_t d[3] = {vdims[0], vdims[1], vdims[2]};
-      if (bs_view_upload(ctx, s.id, vox.data(), d) != BS_OK) {
-        fprintf(stderr, "upload failed: %s\n", bs_last_error(ctx));
-        return 1;
-      }
       bs_fuse_view fv{};
       fv.view_id = s.id;
       /* world coords; bs_fuse_volume shifts by -vol_min itself.
@@ -249,9 +238,49 @@ int main(int argc, char **argv) {
       for (int i = 0; i < 12; ++i) fv.affine[i] = r->second[i];
       if (anisoF != 1.0)
         for (int i = 8; i < 12; ++i) fv.affine[i] /= anisoF;
+      /* multi-resolution input: pick the coarsest level whose sampling
+       * step under the adjusted transform stays <= ~1 px (ViewUtil's
+       * forBestResolution rule, bs_cli_util.h) and fold the [PIN-MIP]
+       * mipmap transform (x0 = f*xl + (f-1)/2) into the view affine;
+       * blending borders/ranges shrink into the level grid. */
+      auto levels = bscli::read_levels(in_n5, s.id, tp);
+      bscli::M34 model;
+      for (int i = 0; i < 12; ++i) model[i] = fv.affine[i];
+      int lvi = bscli::pick_level_for_transform(model, levels);
+      long long lf[3] = {1, 1, 1};
+      if (lvi > 0 && lvi < (int)levels.size()) {
+        for (int d = 0; d < 3; ++d) lf[d] = levels[lvi].f[d];
+        printf("view tp=%d setup=%d: fusing from pyramid level s%d "
+               "(factors %lld,%lld,%lld)\n",
+               tp, s.id, levels[lvi].level, lf[0], lf[1], lf[2]);
+        for (int rr = 0; rr < 3; ++rr) {
+          double off = 0;
+          for (int cc = 0; cc < 3; ++cc) {
+            off += fv.affine[rr * 4 + cc] * 0.5 * (double)(lf[cc] - 1);
+            fv.affine[rr * 4 + cc] *= (double)lf[cc];
+          }
+          fv.affine[rr * 4 + 3] += off;
+        }
+      }
+      std::vector<uint16_t> vox;
+      std::vector<long long> vdims;
+      if (!in_n5.read_volume_u16(
+              bssd::SpimData::image_dataset(
+                  s.id, tp, lvi < (int)levels.size() && lvi > 0
+                                ? levels[lvi].level
+                                : 0),
+              &vox, &vdims)) {
+        fprintf(stderr, "cannot read view tp=%d setup=%d\n", tp, s.id);
+        return 1;
+      }
+      int64_t d[3] = {vdims[0], vdims[1], vdims[2]};
+      if (bs_view_upload(ctx, s.id, vox.data(), d) != BS_OK) {
+        fprintf(stderr, "upload failed: %s\n", bs_last_error(ctx));
+        return 1;
+      }
       for (int d2 = 0; d2 < 3; ++d2) {
-        fv.blend_border[d2] = bborder;
-        fv.blend_range[d2] = brange;
+        fv.blend_border[d2] = bborder / (float)lf[d2];
+        fv.blend_range[d2] = brange / (float)lf[d2];
       }
       if (args.has("intensityN5Path")) {
         bsn5::Container coeff_n5(args.get("intensityN5Path"));

@@ -18,6 +18,7 @@
 #include "../../../include/bigstitch.h"
 #include "bs_cli_util.h"
 #include "bs_n5.h"
+#include "bs_mip.h"
 #include "bs_spimdata.h"
 
 using bscli::M34;
@@ -89,16 +90,66 @@ int main(int argc, char **argv) {
 
   std::vector<bssd::StitchEntry> entries;
   for (int tp : sd.timepoints) {
-    /* upload views of this timepoint on demand */
-    std::set<int> uploaded;
-    auto ensure_view = [&](int setup) -> bool {
-      if (uploaded.count(setup)) return true;
+    /* multi-resolution input: per setup pick the largest pyramid level
+     * whose factors divide -ds ([PIN-MIPSEL], bs_mip.h) and stitch the
+     * remainder; pairs whose two views land on different factors fall
+     * back to the largest COMMON dividing level. */
+    std::map<int, std::vector<bscli::MipLevel>> setup_levels;
+    auto levels_of = [&](int setup) -> const std::vector<bscli::MipLevel> & {
+      auto it = setup_levels.find(setup);
+      if (it == setup_levels.end())
+        it = setup_levels
+                 .emplace(setup, bscli::read_levels(n5, setup, tp))
+                 .first;
+      return it->second;
+    };
+    long long dsv[3] = {ds[0], ds[1], ds[2]};
+    auto common_level = [&](int sa, int sb, long long f[3], int *lva,
+                            int *lvb) {
+      const auto &la = levels_of(sa), &lb = levels_of(sb);
+      *lva = *lvb = 0;
+      f[0] = f[1] = f[2] = 1;
+      long long bestProd = 0;
+      for (size_t i = 0; i < la.size(); ++i) {
+        bool div = true;
+        for (int d = 0; d < 3; ++d)
+          if (la[i].f[d] <= 0 || dsv[d] % la[i].f[d] != 0) div = false;
+        if (!div) continue;
+        int bidx = -1;
+        for (size_t j = 0; j < lb.size(); ++j)
+          if (lb[j].f[0] == la[i].f[0] && lb[j].f[1] == la[i].f[1] &&
+              lb[j].f[2] == la[i].f[2])
+            bidx = (int)j;
+        long long prod = la[i].f[0] * la[i].f[1] * la[i].f[2];
+        if (bidx >= 0 && prod > bestProd) {
+          bestProd = prod;
+          *lva = (int)i;
+          *lvb = bidx;
+          for (int d = 0; d < 3; ++d) f[d] = la[i].f[d];
+        }
+      }
+    };
+    /* upload views of this timepoint on demand (at a chosen level) */
+    std::map<int, int> uploaded; /* setup -> uploaded level */
+    std::map<int, std::vector<long long>> updims;
+    auto ensure_view = [&](int setup, int level) -> bool {
+      auto it = uploaded.find(setup);
+      if (it != uploaded.end()) {
+        if (it->second != level) {
+          fprintf(stderr,
+                  "internal: setup %d needed at two pyramid levels\n",
+                  setup);
+          return false;
+        }
+        return true;
+      }
       std::vector<uint16_t> vox;
       std::vector<long long> dims;
-      if (!n5.read_volume_u16(bssd::SpimData::image_dataset(setup, tp),
-                              &vox, &dims)) {
-        fprintf(stderr, "cannot read view tp=%d setup=%d from %s\n", tp,
-                setup, sd.n5_path.c_str());
+      if (!n5.read_volume_u16(
+              bssd::SpimData::image_dataset(setup, tp, level), &vox,
+              &dims)) {
+        fprintf(stderr, "cannot read view tp=%d setup=%d s%d from %s\n",
+                tp, setup, level, sd.n5_path.c_str());
         return false;
       }
       int64_t d[3] = {dims[0], dims[1], dims[2]};
@@ -106,7 +157,8 @@ int main(int argc, char **argv) {
         fprintf(stderr, "upload failed: %s\n", bs_last_error(ctx));
         return false;
       }
-      uploaded.insert(setup);
+      uploaded[setup] = level;
+      updims[setup] = dims;
       return true;
     };
 
@@ -116,6 +168,9 @@ int main(int argc, char **argv) {
       int sa, sb;
       bs_pair_desc pd;
       M34 ma, mb;
+      long long f[3] = {1, 1, 1}; /* shared pyramid-level factors */
+      int lva = 0, lvb = 0;       /* level index per view */
+      int rem[3] = {1, 1, 1};     /* ds remainder stitched on the GPU */
     };
     std::vector<PairPlan> plans;
     for (size_t i = 0; i < sd.setups.size(); ++i) {
@@ -154,6 +209,14 @@ int main(int argc, char **argv) {
         pp.mb = rb->second;
         pp.pd.view_a = A.id;
         pp.pd.view_b = B.id;
+        common_level(A.id, B.id, pp.f, &pp.lva, &pp.lvb);
+        for (int d = 0; d < 3; ++d)
+          pp.rem[d] = (int)(dsv[d] / pp.f[d]);
+        if (pp.f[0] * pp.f[1] * pp.f[2] > 1)
+          printf("pair (%d,%d): reading pyramid level s%d "
+                 "(factors %lld,%lld,%lld, remainder %d,%d,%d)\n",
+                 A.id, B.id, pp.lva, pp.f[0], pp.f[1], pp.f[2], pp.rem[0],
+                 pp.rem[1], pp.rem[2]);
         bool valid = true;
         for (int v = 0; v < 2; ++v) {
           const auto &m = v == 0 ? ra->second : rb->second;
@@ -173,9 +236,21 @@ int main(int argc, char **argv) {
             int64_t e = (int64_t)std::ceil(b) + 1;
             if (o < 0) o = 0;
             if (e > dims[d]) e = dims[d];
-            if (e <= o) valid = false;
-            off[d] = o;
-            size[d] = e - o;
+            /* interval in the uploaded level's grid */
+            const long long fd = pp.f[d];
+            int64_t lev_dim = (dims[d] + fd - 1) / fd;
+            {
+              const auto &ls = levels_of(v == 0 ? A.id : B.id);
+              int li = v == 0 ? pp.lva : pp.lvb;
+              if (li < (int)ls.size() && (int)ls[li].dims.size() == 3)
+                lev_dim = ls[li].dims[d];
+            }
+            int64_t ol = o / fd;
+            int64_t el = (e + fd - 1) / fd;
+            if (el > lev_dim) el = lev_dim;
+            if (el <= ol) valid = false;
+            off[d] = ol;
+            size[d] = el - ol;
           }
         }
         if (valid) plans.push_back(pp);
@@ -185,23 +260,40 @@ int main(int argc, char **argv) {
     if (args.has("dryRun") || plans.empty()) continue;
     if (!ensure_ctx()) return 1;
 
-    std::vector<bs_pair_desc> pds;
-    for (auto &pp : plans) {
-      if (!ensure_view(pp.sa) || !ensure_view(pp.sb)) return 1;
-      pds.push_back(pp.pd);
-    }
-    bs_stitch_params prm{};
-    prm.ds[0] = (int)ds[0];
-    prm.ds[1] = (int)ds[1];
-    prm.ds[2] = (int)ds[2];
-    prm.peaks_to_check = (int)args.getl("peaksToCheck", 5);
-    prm.do_subpixel = args.has("disableSubpixelResolution") ? 0 : 1;
-    prm.min_overlap_ratio = args.getd("minOverlapRatio", 0.25);
-    std::vector<bs_shift_result> res(pds.size());
-    int rc = bs_stitch_batch(ctx, pds.data(), pds.size(), &prm, res.data());
-    if (rc != BS_OK) {
-      fprintf(stderr, "stitch failed: %s\n", bs_last_error(ctx));
-      return 1;
+    for (auto &pp : plans)
+      if (!ensure_view(pp.sa, pp.lva) || !ensure_view(pp.sb, pp.lvb))
+        return 1;
+    /* one bs_stitch_batch per distinct ds-remainder (pairs from
+     * different pyramid depths stitch at different residual factors) */
+    std::vector<bs_shift_result> res(plans.size());
+    std::map<std::array<int, 3>, std::vector<size_t>> groups;
+    for (size_t k = 0; k < plans.size(); ++k)
+      groups[{plans[k].rem[0], plans[k].rem[1], plans[k].rem[2]}]
+          .push_back(k);
+    for (auto &g : groups) {
+      std::vector<bs_pair_desc> pds;
+      for (size_t k : g.second) pds.push_back(plans[k].pd);
+      bs_stitch_params prm{};
+      prm.ds[0] = g.first[0];
+      prm.ds[1] = g.first[1];
+      prm.ds[2] = g.first[2];
+      prm.peaks_to_check = (int)args.getl("peaksToCheck", 5);
+      prm.do_subpixel = args.has("disableSubpixelResolution") ? 0 : 1;
+      prm.min_overlap_ratio = args.getd("minOverlapRatio", 0.25);
+      std::vector<bs_shift_result> gres(pds.size());
+      int rc =
+          bs_stitch_batch(ctx, pds.data(), pds.size(), &prm, gres.data());
+      if (rc != BS_OK) {
+        fprintf(stderr, "stitch failed: %s\n", bs_last_error(ctx));
+        return 1;
+      }
+      for (size_t i = 0; i < g.second.size(); ++i) {
+        /* back to full-resolution local px: the kernel already scaled
+           by the remainder; the pyramid level contributes f more */
+        for (int d = 0; d < 3; ++d)
+          gres[i].shift[d] *= (double)plans[g.second[i]].f[d];
+        res[g.second[i]] = gres[i];
+      }
     }
     for (size_t k = 0; k < plans.size(); ++k) {
       const auto &pp = plans[k];
