@@ -111,6 +111,10 @@ def load_lib():
                                   C.c_uint16, C.c_uint16]
     lib.bs_view_set_coefficients.argtypes = [C.c_void_p, C.c_int32,
                                              C.c_void_p, C.c_int32 * 3]
+    lib.bs_view_combine_avg.argtypes = [C.c_void_p, C.c_int32,
+                                        C.POINTER(C.c_int32), C.c_int32]
+    lib.bs_view_sum.argtypes = [C.c_void_p, C.c_int32,
+                                C.POINTER(C.c_uint64)]
     lib.bs_stitch_batch.argtypes = [C.c_void_p, C.POINTER(_Pair), C.c_size_t,
                                     C.POINTER(_StitchParams),
                                     C.POINTER(_ShiftResult)]

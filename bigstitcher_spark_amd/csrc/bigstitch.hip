@@ -1567,6 +1567,42 @@ __global__ __launch_bounds__(256) void k_pyr(
   }
 }
 
+/* ---------------------------------------------- view-group aggregation */
+
+struct bs_ptrs8 {
+  const unsigned short *p[8];
+};
+
+/* voxelwise mean of n equal-sized views, round-to-nearest [PIN-GROUP]
+ * (ActionType.AVERAGE of GroupedViewAggregator) */
+__global__ __launch_bounds__(256) void k_view_avg(bs_ptrs8 ins, int n,
+                                                  unsigned short *out,
+                                                  long nvox) {
+  const float inv = 1.0f / (float)n;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nvox;
+       i += (long)gridDim.x * blockDim.x) {
+    float s = 0.0f;
+    for (int k = 0; k < n; ++k) s += (float)ins.p[k][i];
+    out[i] = (unsigned short)__float2int_rn(s * inv);
+  }
+}
+
+/* exact u64 voxel sum (PICK_BRIGHTEST support) */
+__global__ __launch_bounds__(256) void k_view_sum(const unsigned short *in,
+                                                  long nvox, u64 *out) {
+  u64 acc = 0;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nvox;
+       i += (long)gridDim.x * blockDim.x)
+    acc += in[i];
+  for (int off = 32; off >= 1; off >>= 1) acc += __shfl_down(acc, off);
+  __shared__ u64 ws[4];
+  const int lane = threadIdx.x & 63, wv = threadIdx.x >> 6;
+  if (lane == 0) ws[wv] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0)
+    atomicAdd(out, ws[0] + ws[1] + ws[2] + ws[3]);
+}
+
 /* ----------------------------------------------------------- synthetic */
 
 __device__ __forceinline__ u64 splitmix64(u64 x) {
@@ -2072,6 +2108,71 @@ extern "C" int bs_view_synth(bs_ctx *c, int32_t id, const int64_t dims[3],
   CHK(c, hipStreamSynchronize(c->stream));
   flush_stats(c);
   c->views[id] = {d, {nx, ny, nz}};
+  return BS_OK;
+}
+
+extern "C" int bs_view_combine_avg(bs_ctx *c, int32_t out_id,
+                                   const int32_t *in_ids, int32_t n) {
+  if (!c || !in_ids || n < 1 || n > 8) return BS_EINVAL;
+  std::lock_guard<std::mutex> g(c->mu);
+  CHK(c, hipSetDevice(c->dev));
+  bs_ptrs8 ins{};
+  long dims[3] = {0, 0, 0};
+  for (int k = 0; k < n; ++k) {
+    auto it = c->views.find(in_ids[k]);
+    if (it == c->views.end()) return BS_ENOVIEW;
+    if (k == 0) {
+      for (int d = 0; d < 3; ++d) dims[d] = it->second.dims[d];
+    } else {
+      for (int d = 0; d < 3; ++d)
+        if (dims[d] != it->second.dims[d]) {
+          c->err = "combine: view dims differ";
+          return BS_EINVAL;
+        }
+    }
+    ins.p[k] = it->second.dptr;
+  }
+  size_t nvox = (size_t)dims[0] * dims[1] * dims[2];
+  unsigned short *dout;
+  CHK(c, hipMalloc(&dout, nvox * 2));
+  {
+    long blocks = std::min((long)4096, (long)((nvox + 255) / 256));
+    hipLaunchKernelGGL(k_view_avg, dim3(blocks), dim3(256), 0, c->stream,
+                       ins, n, dout, (long)nvox);
+  }
+  CHK(c, hipStreamSynchronize(c->stream));
+  auto it = c->views.find(out_id);
+  if (it != c->views.end()) {
+    (void)hipFree(it->second.dptr);
+    if (it->second.coeff) (void)hipFree(it->second.coeff);
+    c->views.erase(it);
+  }
+  c->views[out_id] = {dout, {dims[0], dims[1], dims[2]}};
+  return BS_OK;
+}
+
+extern "C" int bs_view_sum(bs_ctx *c, int32_t view_id, uint64_t *sum) {
+  if (!c || !sum) return BS_EINVAL;
+  std::lock_guard<std::mutex> g(c->mu);
+  CHK(c, hipSetDevice(c->dev));
+  auto it = c->views.find(view_id);
+  if (it == c->views.end()) return BS_ENOVIEW;
+  size_t nvox = (size_t)it->second.dims[0] * it->second.dims[1] *
+                it->second.dims[2];
+  u64 *dsum;
+  CHK(c, hipMalloc(&dsum, sizeof(u64)));
+  CHK(c, hipMemsetAsync(dsum, 0, sizeof(u64), c->stream));
+  {
+    long blocks = std::min((long)4096, (long)((nvox + 255) / 256));
+    hipLaunchKernelGGL(k_view_sum, dim3(blocks), dim3(256), 0, c->stream,
+                       it->second.dptr, (long)nvox, dsum);
+  }
+  u64 h = 0;
+  CHK(c, hipMemcpyAsync(&h, dsum, sizeof(u64), hipMemcpyDeviceToHost,
+                        c->stream));
+  CHK(c, hipStreamSynchronize(c->stream));
+  (void)hipFree(dsum);
+  *sum = h;
   return BS_OK;
 }
 

@@ -42,11 +42,15 @@ int main(int argc, char **argv) {
             "[--dryRun]\n");
     return 2;
   }
-  if (args.get("channelCombine", "AVERAGE") != "AVERAGE" ||
-      args.get("illumCombine", "PICK_BRIGHTEST") != "PICK_BRIGHTEST") {
-    fprintf(stderr, "only default channel/illum grouping supported\n");
-    return 2;
-  }
+  const std::string chComb = args.get("channelCombine", "AVERAGE");
+  const std::string ilComb = args.get("illumCombine", "PICK_BRIGHTEST");
+  for (auto &a : {chComb, ilComb})
+    if (a != "AVERAGE" && a != "PICK_BRIGHTEST") {
+      fprintf(stderr,
+              "unsupported combine action %s (AVERAGE|PICK_BRIGHTEST)\n",
+              a.c_str());
+      return 2;
+    }
   auto ds = bscli::parse_ints(args.get("downsampling", "2,2,1"));
   if (ds.size() != 3) {
     fprintf(stderr, "bad -ds\n");
@@ -162,22 +166,62 @@ int main(int argc, char **argv) {
       return true;
     };
 
-    /* enumerate overlapping pairs (transformed-bbox intersection, the
-     * filterNonOverlappingPairs step, ref :165) */
+    /* views of one Tile grouped over {Illumination, Channel}; Tiles
+     * compared; application over {TimePoint, Angle} — the reference's
+     * default SpimDataFilteringAndGrouping (ref :146-161). Setups
+     * without attributes form singleton groups. */
+    struct VGroup {
+      std::vector<const bssd::ViewSetup *> members; /* id-ascending */
+      int rep = -1; /* first member: registration + dims source */
+    };
+    std::vector<VGroup> vgroups;
+    {
+      std::map<std::pair<int, int>, VGroup> gm;
+      int solo = 0;
+      for (auto &sv : sd.setups) {
+        if (!selset.count({tp, sv.id}) || !sd.regs.count({tp, sv.id}))
+          continue;
+        auto key = sv.tile >= 0 ? std::make_pair(sv.tile, sv.angle)
+                                : std::make_pair(-1 - (solo++), -1);
+        gm[key].members.push_back(&sv);
+      }
+      for (auto &kv : gm) {
+        std::sort(kv.second.members.begin(), kv.second.members.end(),
+                  [](const bssd::ViewSetup *a, const bssd::ViewSetup *b) {
+                    return a->id < b->id;
+                  });
+        kv.second.rep = kv.second.members[0]->id;
+        /* grouped members must share the registration (reference TODO
+         * at :212 — unequal in-group transforms are not handled there
+         * either) */
+        for (auto *m : kv.second.members)
+          if (!(sd.regs.at({tp, m->id}) ==
+                sd.regs.at({tp, kv.second.rep})))
+            fprintf(stderr,
+                    "warning: grouped views %d and %d have different "
+                    "registrations; using setup %d's\n",
+                    m->id, kv.second.rep, kv.second.rep);
+        vgroups.push_back(kv.second);
+      }
+    }
+    /* enumerate overlapping group pairs (transformed-bbox intersection,
+     * the filterNonOverlappingPairs step, ref :165) */
     struct PairPlan {
-      int sa, sb;
+      int sa, sb; /* representative setups */
+      const VGroup *ga = nullptr, *gb = nullptr;
       bs_pair_desc pd;
       M34 ma, mb;
       long long f[3] = {1, 1, 1}; /* shared pyramid-level factors */
       int lva = 0, lvb = 0;       /* level index per view */
       int rem[3] = {1, 1, 1};     /* ds remainder stitched on the GPU */
+      bool nonequal = false;      /* resample-to-common-frame fallback */
+      double lo[3], hi[3];        /* world overlap box */
+      long long rdims[3] = {1, 1, 1}; /* fallback resample grid dims */
     };
     std::vector<PairPlan> plans;
-    for (size_t i = 0; i < sd.setups.size(); ++i) {
-      for (size_t j = i + 1; j < sd.setups.size(); ++j) {
-        const auto &A = sd.setups[i], &B = sd.setups[j];
-        if (!selset.count({tp, A.id}) || !selset.count({tp, B.id}))
-          continue;
+    for (size_t i = 0; i < vgroups.size(); ++i) {
+      for (size_t j = i + 1; j < vgroups.size(); ++j) {
+        const auto &A = *vgroups[i].members[0], &B = *vgroups[j].members[0];
         auto ra = sd.regs.find({tp, A.id}), rb = sd.regs.find({tp, B.id});
         if (ra == sd.regs.end() || rb == sd.regs.end()) continue;
         double loA[3], hiA[3], loB[3], hiB[3];
@@ -192,10 +236,29 @@ int main(int argc, char **argv) {
         }
         if (!ovl) continue;
         if (!bscli::linear_equal(ra->second, rb->second)) {
-          fprintf(stderr,
-                  "skipping pair (%d,%d): non-translational registration "
-                  "parts differ (unsupported this round)\n",
-                  A.id, B.id);
+          /* computeStitchingNonEqualTransformations (ref :259-267)
+           * [PIN-NONEQ]: both groups are resampled onto the world
+           * overlap box at -ds spacing (inverse-affine trilinear, the
+           * fusion sampler) and phase-correlated there; the shift is
+           * then in world units of ds. */
+          printf("pair (%d,%d): non-translations differ — using "
+                 "virtually fused views for stitching\n",
+                 A.id, B.id);
+          PairPlan pp;
+          pp.sa = A.id;
+          pp.sb = B.id;
+          pp.ga = &vgroups[i];
+          pp.gb = &vgroups[j];
+          pp.ma = ra->second;
+          pp.mb = rb->second;
+          pp.nonequal = true;
+          for (int d = 0; d < 3; ++d) {
+            pp.lo[d] = lo[d];
+            pp.hi[d] = hi[d];
+            pp.rdims[d] = std::max(
+                1LL, (long long)std::ceil((hi[d] - lo[d]) / (double)ds[d]));
+          }
+          plans.push_back(pp);
           continue;
         }
         /* world overlap -> local interval per view */
@@ -205,9 +268,11 @@ int main(int argc, char **argv) {
         PairPlan pp;
         pp.sa = A.id;
         pp.sb = B.id;
+        pp.ga = &vgroups[i];
+        pp.gb = &vgroups[j];
         pp.ma = ra->second;
         pp.mb = rb->second;
-        pp.pd.view_a = A.id;
+        pp.pd.view_a = A.id; /* replaced by the combined id at prep */
         pp.pd.view_b = B.id;
         common_level(A.id, B.id, pp.f, &pp.lva, &pp.lvb);
         for (int d = 0; d < 3; ++d)
@@ -260,9 +325,128 @@ int main(int argc, char **argv) {
     if (args.has("dryRun") || plans.empty()) continue;
     if (!ensure_ctx()) return 1;
 
-    for (auto &pp : plans)
-      if (!ensure_view(pp.sa, pp.lva) || !ensure_view(pp.sb, pp.lvb))
+    /* prepare each group's stitched image: upload members at the
+     * pair's pyramid level and aggregate [PIN-GROUP] (illum action
+     * first, then channel action — ref :204-208). Singleton groups
+     * pass through. Combined views get ids >= 1<<20. */
+    int next_tmp = 1 << 20;
+    std::map<std::string, int> combined; /* "rep:level" -> view id */
+    auto pick_brightest = [&](const std::vector<int> &ids) -> int {
+      int best = ids[0];
+      uint64_t bestSum = 0;
+      for (size_t q = 0; q < ids.size(); ++q) {
+        uint64_t sum = 0;
+        if (bs_view_sum(ctx, ids[q], &sum) != BS_OK) return -1;
+        if (q == 0 || sum > bestSum) {
+          bestSum = sum;
+          best = ids[q];
+        }
+      }
+      return best;
+    };
+    auto combine_ids = [&](std::vector<int> ids,
+                           const std::string &action) -> int {
+      if (ids.size() == 1) return ids[0];
+      if (action == "PICK_BRIGHTEST") return pick_brightest(ids);
+      std::vector<int32_t> in(ids.begin(), ids.end());
+      int out_id = next_tmp++;
+      if (bs_view_combine_avg(ctx, out_id, in.data(), (int)in.size()) !=
+          BS_OK) {
+        fprintf(stderr, "combine failed: %s\n", bs_last_error(ctx));
+        return -1;
+      }
+      return out_id;
+    };
+    auto prepare_group = [&](const VGroup &g, int level,
+                             int *out_id) -> bool {
+      if (g.members.size() == 1) {
+        if (!ensure_view(g.rep, level)) return false;
+        *out_id = g.rep;
+        return true;
+      }
+      std::string key = std::to_string(g.rep) + ":" + std::to_string(level);
+      auto itc = combined.find(key);
+      if (itc != combined.end()) {
+        *out_id = itc->second;
+        return true;
+      }
+      for (auto *m : g.members)
+        if (!ensure_view(m->id, level)) return false;
+      /* by channel, illum-ascending */
+      std::map<int, std::vector<int>> by_ch;
+      for (auto *m : g.members) by_ch[m->channel].push_back(m->id);
+      std::vector<int> ch_ids;
+      for (auto &kv : by_ch) {
+        int cid = combine_ids(kv.second, ilComb);
+        if (cid < 0) return false;
+        ch_ids.push_back(cid);
+      }
+      int result = combine_ids(ch_ids, chComb);
+      if (result < 0) return false;
+      printf("group of %zu views (rep %d): combined into view %d "
+             "(illum %s, channel %s)\n",
+             g.members.size(), g.rep, result, ilComb.c_str(),
+             chComb.c_str());
+      combined[key] = result;
+      *out_id = result;
+      return true;
+    };
+    for (auto &pp : plans) {
+      if (pp.nonequal) {
+        /* resample both groups onto the overlap box at ds spacing via
+         * the fusion sampler, then stitch the resampled volumes 1:1 */
+        int ga_id, gb_id;
+        if (!prepare_group(*pp.ga, 0, &ga_id) ||
+            !prepare_group(*pp.gb, 0, &gb_id))
+          return 1;
+        for (int side = 0; side < 2; ++side) {
+          const M34 &m = side == 0 ? pp.ma : pp.mb;
+          bs_fuse_view fv{};
+          fv.view_id = side == 0 ? ga_id : gb_id;
+          for (int r2 = 0; r2 < 3; ++r2) {
+            for (int c2 = 0; c2 < 3; ++c2)
+              fv.affine[r2 * 4 + c2] = m[r2 * 4 + c2] / (double)ds[r2];
+            fv.affine[r2 * 4 + 3] =
+                (m[r2 * 4 + 3] - pp.lo[r2]) / (double)ds[r2];
+          }
+          bs_fuse_params fprm{};
+          fprm.fusion_type = BS_FUSION_AVG;
+          fprm.out_dtype = BS_OUT_UINT16;
+          fprm.min_intensity = 0;
+          fprm.max_intensity = 65535;
+          fprm.interp = 1;
+          int64_t vmin[3] = {0, 0, 0};
+          int64_t vdim[3] = {pp.rdims[0], pp.rdims[1], pp.rdims[2]};
+          int32_t l0[3] = {1, 1, 1};
+          int64_t ld[3];
+          std::vector<uint16_t> host((size_t)pp.rdims[0] * pp.rdims[1] *
+                                     pp.rdims[2]);
+          void *bufs[1] = {host.data()};
+          if (bs_fuse_volume(ctx, &fv, 1, vmin, vdim, &fprm, 1, l0, ld,
+                             bufs) != BS_OK) {
+            fprintf(stderr, "resample failed: %s\n", bs_last_error(ctx));
+            return 1;
+          }
+          int rid = next_tmp++;
+          if (bs_view_upload(ctx, rid, host.data(), vdim) != BS_OK)
+            return 1;
+          int64_t *off = side == 0 ? pp.pd.off_a : pp.pd.off_b;
+          int64_t *size = side == 0 ? pp.pd.size_a : pp.pd.size_b;
+          for (int d = 0; d < 3; ++d) {
+            off[d] = 0;
+            size[d] = pp.rdims[d];
+          }
+          (side == 0 ? pp.pd.view_a : pp.pd.view_b) = rid;
+        }
+        continue;
+      }
+      int ga_id, gb_id;
+      if (!prepare_group(*pp.ga, pp.lva, &ga_id) ||
+          !prepare_group(*pp.gb, pp.lvb, &gb_id))
         return 1;
+      pp.pd.view_a = ga_id;
+      pp.pd.view_b = gb_id;
+    }
     /* one bs_stitch_batch per distinct ds-remainder (pairs from
      * different pyramid depths stitch at different residual factors) */
     std::vector<bs_shift_result> res(plans.size());
@@ -302,14 +486,19 @@ int main(int argc, char **argv) {
         printf("pair (%d,%d): no shift found\n", pp.sa, pp.sb);
         continue;
       }
-      /* world shift = L x local shift; filters per
-       * FilteredStitchingResults (ref :369-380) */
-      double L[9], t[3];
-      bscli::decompose(pp.ma, L, t);
+      /* world shift = L x local shift (equal-transform path) or
+       * ds x grid shift (the fallback's grid axes ARE world axes);
+       * filters per FilteredStitchingResults (ref :369-380) */
       double ws[3];
-      for (int d = 0; d < 3; ++d)
-        ws[d] = L[d * 3 + 0] * r.shift[0] + L[d * 3 + 1] * r.shift[1] +
-                L[d * 3 + 2] * r.shift[2];
+      if (pp.nonequal) {
+        for (int d = 0; d < 3; ++d) ws[d] = r.shift[d] * (double)ds[d];
+      } else {
+        double L[9], t[3];
+        bscli::decompose(pp.ma, L, t);
+        for (int d = 0; d < 3; ++d)
+          ws[d] = L[d * 3 + 0] * r.shift[0] + L[d * 3 + 1] * r.shift[1] +
+                  L[d * 3 + 2] * r.shift[2];
+      }
       printf("pair (%d,%d): shift=(%.3f, %.3f, %.3f) r=%.4f\n", pp.sa,
              pp.sb, ws[0], ws[1], ws[2], r.r);
       if (r.r < minR || r.r > maxR) continue;
@@ -320,8 +509,8 @@ int main(int argc, char **argv) {
           maxShiftTotal)
         continue;
       bssd::StitchEntry e;
-      e.views_a = {{tp, pp.sa}};
-      e.views_b = {{tp, pp.sb}};
+      for (auto *m : pp.ga->members) e.views_a.push_back({tp, m->id});
+      for (auto *m : pp.gb->members) e.views_b.push_back({tp, m->id});
       for (int d = 0; d < 3; ++d) e.matrix[d * 4 + 3] = ws[d];
       /* bbox: world overlap of the two transformed views */
       double loA[3], hiA[3], loB[3], hiB[3];

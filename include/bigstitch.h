@@ -93,6 +93,20 @@ int bs_view_download(bs_ctx *ctx, int32_t view_id, uint16_t *out);
 int bs_view_set_coefficients(bs_ctx *ctx, int32_t view_id, const float *ab,
                              const int32_t grid_dims[3]);
 
+/* View-group aggregation primitives (GroupedViewAggregator restatement
+ * [PIN-GROUP]; reference SparkPairwiseStitching.java:204-208 applies
+ * illumCombine then channelCombine over the {Illumination, Channel}
+ * group of each Tile):
+ *  - bs_view_combine_avg: out_id := voxelwise mean of n equal-sized
+ *    uploaded views, rounded to nearest (ActionType.AVERAGE);
+ *  - bs_view_sum: exact integer voxel sum (the host implements
+ *    ActionType.PICK_BRIGHTEST by choosing the member with the highest
+ *    mean = sum / voxels; exact u64 sums make the pick deterministic).
+ */
+int bs_view_combine_avg(bs_ctx *ctx, int32_t out_id, const int32_t *in_ids,
+                        int32_t n);
+int bs_view_sum(bs_ctx *ctx, int32_t view_id, uint64_t *sum);
+
 /* --------------------------------------------------------------- stitching */
 
 /* One tile-pair work unit. The overlap interval inside each view is computed

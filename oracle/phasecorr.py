@@ -229,3 +229,22 @@ def phase_correlation_shift(
         "r": r,
         "valid": True,
     }
+
+
+def combine_group(vols, action):
+    """[PIN-GROUP] GroupedViewAggregator action restatement (reference
+    SparkPairwiseStitching.java:204-208; mvrecon artifact un-vendored):
+    AVERAGE = voxelwise float32 mean rounded to nearest uint16 (matches
+    the k_view_avg kernel arithmetic exactly — f32 sums of <=8 uint16
+    are exact); PICK_BRIGHTEST = the member with the largest exact
+    integer voxel sum (ties: first member)."""
+    if len(vols) == 1:
+        return vols[0]
+    if action == "AVERAGE":
+        s = np.zeros(vols[0].shape, np.float32)
+        for v in vols:
+            s += v.astype(np.float32)
+        return np.rint(s * np.float32(1.0 / len(vols))).astype(np.uint16)
+    assert action == "PICK_BRIGHTEST"
+    sums = [int(v.astype(np.uint64).sum()) for v in vols]
+    return vols[int(np.argmax(sums))]

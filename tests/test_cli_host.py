@@ -763,3 +763,124 @@ def test_fusion_mip_level_parity_anisotropy(tmp_path):
                         of.FUSION_AVG_BLEND)
     denom = np.maximum(np.abs(ref), 1.0)
     assert np.max(np.abs(fused.astype(np.float64) - ref) / denom) < 1e-4
+
+
+def make_grouped_dataset(tmp, size=64, overlap=24, err=(2.5, -1.5, 1.0),
+                         seed=5, nch=2):
+    """2x1 tile grid, each tile with `nch` channel views (same scene,
+    different noise). Returns (xml, n5, per-setup volumes)."""
+    n5 = os.path.join(tmp, "input.n5")
+    xml = os.path.join(tmp, "dataset.xml")
+    shape = (size, size, size)
+    posB = size - overlap
+    s = (-(posB + err[0]), -err[1], -err[2])
+    ba, bb = synth.pair_blobs_union(shape, s, seed=seed)
+    vols, setups = {}, []
+    sid = 0
+    for tile, blobs, pos in ((0, ba, 0.0), (1, bb, float(posB))):
+        for ch in range(nch):
+            v = synth.render_tile(shape, blobs, noise_seed=seed * 100
+                                  + 10 * tile + ch)
+            vols[sid] = v
+            n5util.write_dataset(n5, f"setup{sid}/timepoint0/s0", v,
+                                 (32, 32, 32))
+            setups.append(dict(id=sid, dims=shape, pos=(pos, 0.0, 0.0),
+                               attrs=dict(tile=tile, angle=0, channel=ch,
+                                          illumination=0)))
+            sid += 1
+    n5util.make_dataset_xml(xml, "input.n5", setups)
+    return xml, n5, vols, err
+
+
+@pytest.mark.gpu
+def test_cli_stitching_grouped_channels(tmp_path):
+    """channel grouping. Two channels per tile are
+    AVERAGEd ([PIN-GROUP]) before phase correlation; the XML entry
+    carries ALL member ViewIds (the reference's grouped
+    SerializablePairwiseStitchingResult)."""
+    err = (2.5, -1.5, 1.0)
+    xml, n5, vols, _ = make_grouped_dataset(str(tmp_path), err=err)
+    r = run([os.path.join(BIN, "stitching"), "-x", xml, "-ds", "1,1,1",
+             "--minR", "0.3"])
+    assert r.returncode == 0, r.stderr + r.stdout
+    assert "combined into view" in r.stdout
+    tree = ET.parse(xml)
+    prs = tree.getroot().findall(".//PairwiseResult")
+    assert len(prs) == 1
+    assert prs[0].find("ViewIdsA").text == "0,0;0,1"
+    assert prs[0].find("ViewIdsB").text == "0,2;0,3"
+    m = [float(x) for x in prs[0].find("Matrix").text.split()]
+    want = (-(40 + err[0]), -err[1], -err[2])
+    for g, w in zip((m[3], m[7], m[11]), want):
+        assert abs(g - w) < 0.5, ((m[3], m[7], m[11]), want)
+
+
+@pytest.mark.gpu
+def test_cli_stitching_pick_brightest_illum(tmp_path):
+    """illumCombine PICK_BRIGHTEST: the dim garbage illumination is
+    dropped; stitching still lands on the true shift."""
+    size, overlap = 64, 24
+    err = (2.5, -1.5, 1.0)
+    n5 = os.path.join(str(tmp_path), "input.n5")
+    xml = os.path.join(str(tmp_path), "dataset.xml")
+    shape = (size, size, size)
+    posB = size - overlap
+    s = (-(posB + err[0]), -err[1], -err[2])
+    ba, bb = synth.pair_blobs_union(shape, s, seed=5)
+    rng = np.random.default_rng(9)
+    setups, sid = [], 0
+    for tile, blobs, pos in ((0, ba, 0.0), (1, bb, float(posB))):
+        for il in range(2):
+            if il == 0:  # dim garbage illumination
+                v = rng.integers(0, 40, size=shape).astype(np.uint16)
+            else:
+                v = synth.render_tile(shape, blobs, noise_seed=7 + tile)
+            n5util.write_dataset(n5, f"setup{sid}/timepoint0/s0", v,
+                                 (32, 32, 32))
+            setups.append(dict(id=sid, dims=shape, pos=(pos, 0.0, 0.0),
+                               attrs=dict(tile=tile, angle=0, channel=0,
+                                          illumination=il)))
+            sid += 1
+    n5util.make_dataset_xml(xml, "input.n5", setups)
+    r = run([os.path.join(BIN, "stitching"), "-x", xml, "-ds", "1,1,1",
+             "--minR", "0.3"])
+    assert r.returncode == 0, r.stderr + r.stdout
+    tree = ET.parse(xml)
+    prs = tree.getroot().findall(".//PairwiseResult")
+    assert len(prs) == 1
+    m = [float(x) for x in prs[0].find("Matrix").text.split()]
+    want = (-(40 + err[0]), -err[1], -err[2])
+    for g, w in zip((m[3], m[7], m[11]), want):
+        assert abs(g - w) < 0.5, ((m[3], m[7], m[11]), want)
+
+
+@pytest.mark.gpu
+def test_cli_stitching_nonequal_transforms(tmp_path):
+    """computeStitchingNonEqualTransformations restatement
+    ([PIN-NONEQ], ref :259-267): a pair whose linear parts differ is
+    resampled onto the world overlap box (fusion sampler) and
+    phase-correlated there."""
+    err = (2.5, -1.5, 1.0)
+    xml, n5, _e, (a, b) = make_grid_dataset(str(tmp_path), err=err)
+    # perturb B's linear part so nonTranslationsEqual fails (1e-5 shear,
+    # geometrically still ~a translation)
+    tree = ET.parse(xml)
+    vrs = tree.getroot().findall(".//ViewRegistration")
+    for vr in vrs:
+        if vr.get("setup") == "1":
+            aff = vr.find(".//affine")
+            m = [float(x) for x in aff.text.split()]
+            m[1] = 1e-5
+            aff.text = " ".join(str(x) for x in m)
+    tree.write(xml)
+    r = run([os.path.join(BIN, "stitching"), "-x", xml, "-ds", "1,1,1",
+             "--minR", "0.3"])
+    assert r.returncode == 0, r.stderr + r.stdout
+    assert "virtually fused views" in r.stdout
+    tree = ET.parse(xml)
+    prs = tree.getroot().findall(".//PairwiseResult")
+    assert len(prs) == 1
+    m = [float(x) for x in prs[0].find("Matrix").text.split()]
+    want = (-(40 + err[0]), -err[1], -err[2])
+    for g, w in zip((m[3], m[7], m[11]), want):
+        assert abs(g - w) < 1.0, ((m[3], m[7], m[11]), want)

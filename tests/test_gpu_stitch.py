@@ -252,3 +252,28 @@ def test_stitch_differing_tile_sizes(ctx):
         assert np.all(np.abs(got["shift"] - ref["shift"]) < 1e-3), (
             got["shift"], ref["shift"])
         assert got["r"] == pytest.approx(ref["r"], abs=1e-9)
+
+
+@pytest.mark.gpu
+def test_view_combine_avg_parity():
+    """bs_view_combine_avg == oracle.combine_group AVERAGE bit-exact
+    ([PIN-GROUP]: f32 sums of <=8 uint16 are exact; same rint)."""
+    from oracle.phasecorr import combine_group
+    rng = np.random.default_rng(21)
+    vols = [rng.integers(0, 65536, size=(9, 17, 23)).astype(np.uint16)
+            for _ in range(3)]
+    with Context(0) as ctx:
+        for i, v in enumerate(vols):
+            ctx.upload(100 + i, v)
+        import ctypes as C
+        ids = (C.c_int32 * 3)(100, 101, 102)
+        rc = ctx._lib.bs_view_combine_avg(ctx._h, 200, ids, 3)
+        assert rc == 0
+        got = ctx.download(200, vols[0].shape)
+        ref = combine_group(vols, "AVERAGE")
+        assert np.array_equal(got, ref)
+        # PICK_BRIGHTEST support: exact sums
+        s = C.c_uint64()
+        rc = ctx._lib.bs_view_sum(ctx._h, 100, C.byref(s))
+        assert rc == 0
+        assert s.value == int(vols[0].astype(np.uint64).sum())
