@@ -181,8 +181,9 @@ int main(int argc, char **argv) {
       for (auto &sv : sd.setups) {
         if (!selset.count({tp, sv.id}) || !sd.regs.count({tp, sv.id}))
           continue;
-        auto key = sv.tile >= 0 ? std::make_pair(sv.tile, sv.angle)
-                                : std::make_pair(-1 - (solo++), -1);
+        auto key = sv.tile >= 0
+                       ? std::make_pair(sv.tile, sv.angle)
+                       : std::make_pair((1 << 20) + (solo++), -1);
         gm[key].members.push_back(&sv);
       }
       for (auto &kv : gm) {
