@@ -707,10 +707,11 @@ def test_stitching_mip_level_parity(tmp_path):
     assert len(prs) == 1
     m = [float(x) for x in prs[0].find("Matrix").text.split()]
     got = (m[3], m[7], m[11])
-    # ds=2 stitching has ~ds precision; content err target ~(-(40+2.5),1.5,-1)
-    want = (-(40 + err[0]), -err[1], -err[2])
+    # stored world shift corrects B's position: expected -err, with
+    # ~ds-level precision at -ds 2,2,2
+    want = (-err[0], -err[1], -err[2])
     for g, w in zip(got, want):
-        assert abs(g - w) < 2.5, (got, want)
+        assert abs(g - w) < 2.0, (got, want)
 
 
 @pytest.mark.gpu
@@ -810,9 +811,9 @@ def test_cli_stitching_grouped_channels(tmp_path):
     assert prs[0].find("ViewIdsA").text == "0,0;0,1"
     assert prs[0].find("ViewIdsB").text == "0,2;0,3"
     m = [float(x) for x in prs[0].find("Matrix").text.split()]
-    want = (-(40 + err[0]), -err[1], -err[2])
+    want = (-err[0], -err[1], -err[2])
     for g, w in zip((m[3], m[7], m[11]), want):
-        assert abs(g - w) < 0.5, ((m[3], m[7], m[11]), want)
+        assert abs(g - w) < 0.75, ((m[3], m[7], m[11]), want)
 
 
 @pytest.mark.gpu
@@ -849,9 +850,9 @@ def test_cli_stitching_pick_brightest_illum(tmp_path):
     prs = tree.getroot().findall(".//PairwiseResult")
     assert len(prs) == 1
     m = [float(x) for x in prs[0].find("Matrix").text.split()]
-    want = (-(40 + err[0]), -err[1], -err[2])
+    want = (-err[0], -err[1], -err[2])
     for g, w in zip((m[3], m[7], m[11]), want):
-        assert abs(g - w) < 0.5, ((m[3], m[7], m[11]), want)
+        assert abs(g - w) < 0.75, ((m[3], m[7], m[11]), want)
 
 
 @pytest.mark.gpu
@@ -881,6 +882,6 @@ def test_cli_stitching_nonequal_transforms(tmp_path):
     prs = tree.getroot().findall(".//PairwiseResult")
     assert len(prs) == 1
     m = [float(x) for x in prs[0].find("Matrix").text.split()]
-    want = (-(40 + err[0]), -err[1], -err[2])
+    want = (-err[0], -err[1], -err[2])
     for g, w in zip((m[3], m[7], m[11]), want):
         assert abs(g - w) < 1.0, ((m[3], m[7], m[11]), want)

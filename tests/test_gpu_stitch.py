@@ -258,6 +258,7 @@ def test_stitch_differing_tile_sizes(ctx):
 def test_view_combine_avg_parity():
     """bs_view_combine_avg == oracle.combine_group AVERAGE bit-exact
     ([PIN-GROUP]: f32 sums of <=8 uint16 are exact; same rint)."""
+    from bigstitcher_spark_amd import Context
     from oracle.phasecorr import combine_group
     rng = np.random.default_rng(21)
     vols = [rng.integers(0, 65536, size=(9, 17, 23)).astype(np.uint16)
