@@ -33,6 +33,7 @@ class _StitchParams(C.Structure):
     _fields_ = [
         ("ds", C.c_int32 * 3), ("peaks_to_check", C.c_int32),
         ("do_subpixel", C.c_int32), ("min_overlap_ratio", C.c_double),
+        ("pad_mode", C.c_int32), ("_pad", C.c_int32),
     ]
 
 
@@ -225,7 +226,8 @@ class Context:
 
     # -- stitching --------------------------------------------------------
     def stitch_batch(self, pairs, ds=(2, 2, 1), peaks_to_check=5,
-                     do_subpixel=True, min_overlap_ratio=0.25):
+                     do_subpixel=True, min_overlap_ratio=0.25,
+                     pad_mode="pow2"):
         """pairs: list of dicts {view_a, view_b, off_a, size_a, off_b,
         size_b} with off/size in (x,y,z). Returns list of dicts(shift
         (sx,sy,sz), r, valid) per oracle.phasecorr [PIN-SIGN]."""
@@ -241,7 +243,7 @@ class Context:
                 cp[i].size_b[d] = p["size_b"][d]
         prm = _StitchParams(
             (C.c_int32 * 3)(*ds), peaks_to_check, int(do_subpixel),
-            min_overlap_ratio,
+            min_overlap_ratio, {"pow2": 0, "fast": 1}[pad_mode], 0,
         )
         res = (_ShiftResult * n)()
         self._check(

@@ -115,6 +115,81 @@ __device__ __forceinline__ void fft_lds(f2 *data, long base, int n,
 #undef D_
 }
 
+/* ---- mixed-radix Stockham FFT (pad_mode=fast, radices 2/3/5/7) ----
+ * The reference's FFT pads to "fast" 7-smooth sizes (imglib2
+ * FFTMethods; [PIN-PAD]); this engine covers those sizes while the
+ * radix-2^2 bit-reversal path above stays the pow2 fast path.
+ * Classic DIT Stockham autosort: natural-order input, ping-pong
+ * between two LDS buffers, natural-order output — no digit-reversal.
+ * Stage with radix R at accumulated sub-size L: for i = b*L + j < n/R:
+ *   t_r = x[i + (n/R)*r] * w_{LR}^{j*r};  y[b*L*R + j + L*s] = DFT_R(t)_s
+ * Twiddles and the DFT-R roots both come from the FULL n-entry table
+ * (w_R^{rs} = tw[(r*s mod R) * n/R]). factors packed 4 bits each,
+ * least-significant first. Element e of line l lives at buf[e*ES + l].
+ * Returns the buffer holding the result (0 = A, 1 = B). */
+template <int ES, int TPL>
+__device__ __forceinline__ int fft_stockham(f2 *bufA, f2 *bufB, long base,
+                                            int n, unsigned long long
+                                                        factors,
+                                            int tl, const f2 *tw,
+                                            int dir) {
+  f2 *src = bufA, *dst = bufB;
+  int cur = 0;
+  int L = 1;
+  for (unsigned long long fac = factors; fac; fac >>= 4) {
+    const int R = (int)(fac & 15);
+    const int M = n / R;
+    for (int i = tl; i < M; i += TPL) {
+      const int j = i % L, b = i / L;
+      f2 t[7];
+      for (int r = 0; r < R; ++r) {
+        f2 v = src[base + (long)(i + M * r) * ES];
+        long k = ((long)j * r * (n / (L * R))) % n;
+        f2 w = tw[k];
+        if (dir < 0) w.y = -w.y;
+        t[r] = cmul(v, w);
+      }
+      for (int s2 = 0; s2 < R; ++s2) {
+        f2 acc = t[0];
+        for (int r = 1; r < R; ++r) {
+          f2 w = tw[(long)((r * s2) % R) * (n / R)];
+          if (dir < 0) w.y = -w.y;
+          f2 p = cmul(t[r], w);
+          acc.x += p.x;
+          acc.y += p.y;
+        }
+        dst[base + (long)(b * L * R + j + L * s2) * ES] = acc;
+      }
+    }
+    __syncthreads();
+    f2 *tmp = src;
+    src = dst;
+    dst = tmp;
+    cur ^= 1;
+    L *= R;
+  }
+  return cur;
+}
+
+__host__ __device__ static inline unsigned long long bs_factorize(int n) {
+  /* 4-bit packed radices, 2s last (so early stages are the cheap big
+   * radices; any order is mathematically fine) */
+  unsigned long long f = 0;
+  int shift = 0;
+  for (int p : {7, 5, 3, 2})
+    while (n % p == 0) {
+      f |= (unsigned long long)p << shift;
+      shift += 4;
+      n /= p;
+    }
+  return n == 1 ? f : 0; /* 0 = not 7-smooth */
+}
+
+static int next_fast_even(int n) { /* smallest even 7-smooth >= n */
+  for (int c = n + (n & 1);; c += 2)
+    if (bs_factorize(c)) return c;
+}
+
 /* -------------------------------------------------------- region descs */
 
 struct bs_region { /* a (possibly strided) uint16 sub-volume */
@@ -719,6 +794,153 @@ __global__ __launch_bounds__(LPB_X *TPL_X) void k_fft_x_inv(
   }
 }
 
+/* ---- mixed-radix (pad_mode=fast) kernel variants: natural-order
+ * Stockham in ping-pong LDS; same line/group mapping as the pow2
+ * kernels. Not perf-tuned — this is the [PIN-PAD] compatibility mode
+ * for the reference's 7-smooth pad sizes (pow2 sizes stay on the
+ * radix-2^2 path). */
+
+__global__ __launch_bounds__(LPB_X *TPL_X) void k_fft_x_fwd_m(
+    bs_region in, f2 *out, int n, unsigned long long fh, long cxp, int py,
+    const f2 *twg /* full n-entry table */) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int h = n >> 1;
+  f2 *tw = (f2 *)smem;       /* h-entry table for the h-point FFT */
+  f2 *da = tw + h;           /* LPB_X * h */
+  f2 *db = da + (long)LPB_X * h;
+  const int tid = threadIdx.x;
+  const int tl = tid % TPL_X, line = tid / TPL_X;
+  for (int i = tid; i < h; i += LPB_X * TPL_X) tw[i] = twg[2 * i];
+  __syncthreads();
+  long nlines = (long)in.my * in.mz;
+  long ngroups = (nlines + LPB_X - 1) / LPB_X;
+  for (long grp = blockIdx.x; grp < ngroups; grp += gridDim.x) {
+    long lid = grp * LPB_X + line;
+    bool active = lid < nlines;
+    int y = active ? (int)(lid % in.my) : 0;
+    int z = active ? (int)(lid / in.my) : 0;
+    const unsigned short *src =
+        in.ptr + (in.oz + z) * in.sxy + (in.oy + y) * in.sx + in.ox;
+    for (int j = tl; j < h; j += TPL_X) {
+      float xa = (active && 2 * j < in.mx) ? (float)src[2 * j] : 0.0f;
+      float xb =
+          (active && 2 * j + 1 < in.mx) ? (float)src[2 * j + 1] : 0.0f;
+      da[(long)j * LPB_X + line] = {xa, xb};
+    }
+    __syncthreads();
+    int res = fft_stockham<LPB_X, TPL_X>(da, db, (long)line, h, fh, tl,
+                                         tw, +1);
+    f2 *ld = (res ? db : da);
+    if (active) {
+      f2 *o = out + ((long)z * py + y) * cxp;
+      for (int k = tl; k <= h; k += TPL_X) {
+        if (k == 0) {
+          f2 z0 = ld[0 * LPB_X + line];
+          o[0] = {z0.x + z0.y, 0.0f};
+          o[h] = {z0.x - z0.y, 0.0f};
+        } else if (k < h) {
+          f2 zk = ld[(long)k * LPB_X + line];
+          f2 zm = ld[(long)(h - k) * LPB_X + line];
+          f2 ze = {0.5f * (zk.x + zm.x), 0.5f * (zk.y - zm.y)};
+          f2 dd = {zk.x - zm.x, zk.y + zm.y};
+          f2 zo = {0.5f * dd.y, -0.5f * dd.x};
+          f2 wzo = cmul(twg[k], zo);
+          o[k] = {ze.x + wzo.x, ze.y + wzo.y};
+        }
+      }
+    }
+    __syncthreads();
+  }
+}
+
+__global__ __launch_bounds__(LPB_X *TPL_X) void k_fft_x_inv_m(
+    const f2 *in, float *out, int n, unsigned long long fh, long cxp,
+    long nlines, const f2 *twg) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int h = n >> 1;
+  f2 *tw = (f2 *)smem;
+  f2 *da = tw + h;
+  f2 *db = da + (long)LPB_X * h;
+  const int tid = threadIdx.x;
+  const int tl = tid % TPL_X, line = tid / TPL_X;
+  for (int i = tid; i < h; i += LPB_X * TPL_X) tw[i] = twg[2 * i];
+  __syncthreads();
+  long ngroups = (nlines + LPB_X - 1) / LPB_X;
+  for (long grp = blockIdx.x; grp < ngroups; grp += gridDim.x) {
+    long lid = grp * LPB_X + line;
+    bool active = lid < nlines;
+    const f2 *src = in + lid * cxp;
+    for (int k = tl; k < h; k += TPL_X) {
+      f2 v = {0.0f, 0.0f};
+      if (active) {
+        f2 xk = src[k], xm = src[h - k];
+        f2 A = {xk.x + xm.x, xk.y - xm.y};
+        f2 B = {xk.x - xm.x, xk.y + xm.y};
+        f2 wc = twg[k];
+        wc.y = -wc.y;
+        f2 wb = cmul(wc, B);
+        v = {A.x - wb.y, A.y + wb.x};
+      }
+      da[(long)k * LPB_X + line] = v;
+    }
+    __syncthreads();
+    int res = fft_stockham<LPB_X, TPL_X>(da, db, (long)line, h, fh, tl,
+                                         tw, -1);
+    f2 *ld = (res ? db : da);
+    if (active) {
+      f2 *o = (f2 *)(out + lid * n);
+      for (int j = tl; j < h; j += TPL_X)
+        o[j] = ld[(long)j * LPB_X + line];
+    }
+    __syncthreads();
+  }
+}
+
+__global__ __launch_bounds__(LPB_S *TPL_S) void k_fft_pass_m(
+    const f2 *in, const f2 *in2, f2 *out, int n, unsigned long long fn,
+    long estride, long gstride, int nlines, int nchunks, int ngroups,
+    int valid, int dir, float scale, const f2 *twg) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  f2 *tw = (f2 *)smem;
+  f2 *da = tw + n;
+  f2 *db = da + (long)LPB_S * n;
+  const int tid = threadIdx.x;
+  const int line = tid % LPB_S, tl = tid / LPB_S;
+  for (int i = tid; i < n; i += LPB_S * TPL_S) tw[i] = twg[i];
+  __syncthreads();
+  const long nwg = (long)ngroups * nchunks;
+  for (long wg = blockIdx.x; wg < nwg; wg += gridDim.x) {
+    const int group = (int)(wg / nchunks);
+    const int x = (int)(wg % nchunks) * LPB_S + line;
+    const bool active = x < nlines;
+    const long base = (long)group * gstride + x;
+    for (int e = tl; e < n; e += TPL_S) {
+      f2 v = {0.0f, 0.0f};
+      if (active && e < valid) {
+        if (in2) {
+          f2 q = conjmul(in[base + e * estride], in2[base + e * estride]);
+          float m2 = q.x * q.x + q.y * q.y;
+          if (m2 > 1e-40f) {
+            float s = scale / sqrtf(m2);
+            v = {q.x * s, q.y * s};
+          }
+        } else {
+          v = in[base + e * estride];
+        }
+      }
+      da[(long)e * LPB_S + line] = v;
+    }
+    __syncthreads();
+    int res = fft_stockham<LPB_S, TPL_S>(da, db, (long)line, n, fn, tl,
+                                         tw, dir);
+    f2 *ld = (res ? db : da);
+    if (active)
+      for (int e = tl; e < n; e += TPL_S)
+        out[base + e * estride] = ld[(long)e * LPB_S + line];
+    __syncthreads();
+  }
+}
+
 /* ----------------------- wave-resident x passes (pow2 fast paths) */
 
 /* h-point complex FFT resident in one wave's registers, E = h/64
@@ -1059,6 +1281,49 @@ __global__ __launch_bounds__(256)
 __global__ __launch_bounds__(256) void k_peak_tile_relaxed(
     const float *pcm, int px, int py, int pz, bs_peak *wgbuf) {
   pk_body<false>(pcm, px, py, pz, wgbuf);
+}
+
+/* Generic strict-26-maxima scan (modulo wrap, scalar loads) for PCM
+ * x-dims not divisible by 4 (pad_mode=fast sizes like 50, 54): the
+ * register-window kernel above needs aligned float4 rows. Parity mode,
+ * not perf-tuned. */
+__global__ __launch_bounds__(256) void k_peak_generic(
+    const float *pcm, int px, int py, int pz, bs_peak *wgbuf) {
+  __shared__ float wvs[4][5];
+  __shared__ long long wis[4][5];
+  const int tid = threadIdx.x, lane = tid & 63, wv = tid >> 6;
+  float tv[5];
+  long long ti[5];
+  for (int k = 0; k < 5; ++k) { tv[k] = -3.0e38f; ti[k] = 0x7fffffffffffffffLL; }
+  const long n = (long)px * py * pz;
+  for (long i = (long)blockIdx.x * 256 + tid; i < n;
+       i += (long)gridDim.x * 256) {
+    const int x = (int)(i % px);
+    const long t = i / px;
+    const int y = (int)(t % py), z = (int)(t / py);
+    const float v = pcm[i];
+    float m = -3.0e38f;
+    for (int dz = -1; dz <= 1; ++dz)
+      for (int dy = -1; dy <= 1; ++dy)
+        for (int dx = -1; dx <= 1; ++dx) {
+          if (!dx && !dy && !dz) continue;
+          const int xx = (x + dx + px) % px;
+          const int yy = (y + dy + py) % py;
+          const int zz = (z + dz + pz) % pz;
+          m = fmaxf(m, pcm[((long)zz * py + yy) * px + xx]);
+        }
+    if (v > m) pk_insert(tv, ti, v, i);
+  }
+  pk_merge_shfl(tv, ti);
+  if (lane == 0)
+    for (int k = 0; k < 5; ++k) { wvs[wv][k] = tv[k]; wis[wv][k] = ti[k]; }
+  __syncthreads();
+  if (tid == 0) {
+    for (int w = 1; w < 4; ++w)
+      for (int k = 0; k < 5; ++k) pk_insert(tv, ti, wvs[w][k], wis[w][k]);
+    bs_peak *o = wgbuf + (long)blockIdx.x * 5;
+    for (int k = 0; k < 5; ++k) o[k] = {tv[k], 0, ti[k]};
+  }
 }
 
 /* Merge per-WG top-5 lists: each block covers a contiguous slice of
@@ -1767,6 +2032,7 @@ struct bs_slot {
 
 struct bs_ctx {
   int dev;
+  std::map<int, f2 *> twiddles_full; /* full n-entry tables (fast pads) */
   hipStream_t stream; /* default stream: views, synth, fusion */
   hipStream_t copy_stream = nullptr; /* D2H overlap (fusion volume) */
   bs_pool cpool; /* host memcpy workers for staged_d2h */
@@ -1849,6 +2115,15 @@ extern "C" int bs_ctx_create(bs_ctx **out, int device_id) {
   (void)hipFuncSetAttribute((const void *)k_fft_pass,
                             hipFuncAttributeMaxDynamicSharedMemorySize,
                             160 * 1024);
+  (void)hipFuncSetAttribute((const void *)k_fft_pass_m,
+                            hipFuncAttributeMaxDynamicSharedMemorySize,
+                            160 * 1024);
+  (void)hipFuncSetAttribute((const void *)k_fft_x_fwd_m,
+                            hipFuncAttributeMaxDynamicSharedMemorySize,
+                            160 * 1024);
+  (void)hipFuncSetAttribute((const void *)k_fft_x_inv_m,
+                            hipFuncAttributeMaxDynamicSharedMemorySize,
+                            160 * 1024);
   (void)hipFuncSetAttribute((const void *)k_fft_pass_glds<2>,
                             hipFuncAttributeMaxDynamicSharedMemorySize,
                             160 * 1024);
@@ -1913,6 +2188,7 @@ extern "C" void bs_ctx_destroy(bs_ctx *c) {
     if (kv.second.coeff) (void)hipFree(kv.second.coeff);
   }
   for (auto &kv : c->twiddles) (void)hipFree(kv.second);
+  for (auto &kv : c->twiddles_full) (void)hipFree(kv.second);
   for (auto &pr : c->evpool) {
     (void)hipEventDestroy(pr.first);
     (void)hipEventDestroy(pr.second);
@@ -2178,6 +2454,24 @@ extern "C" int bs_view_sum(bs_ctx *c, int32_t view_id, uint64_t *sum) {
 
 /* ---- twiddles ---- */
 
+static f2 *get_twiddle_full(bs_ctx *c, int n) {
+  auto it = c->twiddles_full.find(n);
+  if (it != c->twiddles_full.end()) return it->second;
+  std::vector<f2> h(n);
+  for (int k = 0; k < n; ++k) {
+    double a = -2.0 * M_PI * k / n;
+    h[k] = {(float)std::cos(a), (float)std::sin(a)};
+  }
+  f2 *d = nullptr;
+  if (hipMalloc(&d, h.size() * sizeof(f2)) != hipSuccess) return nullptr;
+  (void)hipMemcpy(d, h.data(), h.size() * sizeof(f2),
+                  hipMemcpyHostToDevice);
+  c->twiddles_full[n] = d;
+  return d;
+}
+
+static bool is_pow2(int n) { return n > 0 && (n & (n - 1)) == 0; }
+
 static f2 *get_twiddle(bs_ctx *c, int n) {
   auto it = c->twiddles.find(n);
   if (it != c->twiddles.end()) return it->second;
@@ -2259,10 +2553,17 @@ static int stitch_phaseA(bs_ctx *c, bs_slot *sl, const bs_pair_desc &pd,
                     sl->m[t][0], sl->m[t][1], sl->m[t][2]};
     }
   }
-  /* padded FFT dims [PIN-PAD] */
-  int Px = next_pow2(std::max(sl->m[0][0], sl->m[1][0]));
-  int Py = next_pow2(std::max(sl->m[0][1], sl->m[1][1]));
-  int Pz = next_pow2(std::max(sl->m[0][2], sl->m[1][2]));
+  /* padded FFT dims [PIN-PAD]: pow2 (default, radix-2^2 fast path) or
+   * "fast" even 7-smooth sizes (pad_mode=1, the imglib2 FFTMethods
+   * family — Stockham mixed-radix kernels) */
+  const bool fastpad = prm->pad_mode == 1;
+  auto pad1 = [&](int m01) {
+    return fastpad ? next_fast_even(std::max(m01, 8))
+                   : next_pow2(m01);
+  };
+  int Px = pad1(std::max(sl->m[0][0], sl->m[1][0]));
+  int Py = pad1(std::max(sl->m[0][1], sl->m[1][1]));
+  int Pz = pad1(std::max(sl->m[0][2], sl->m[1][2]));
   if (Px > 1024 || Py > 1024 || Pz > 1024 || Px < 8) {
     c->err = "FFT size unsupported (need 8..1024 per axis)";
     return BS_EUNSUP;
@@ -2278,8 +2579,9 @@ static int stitch_phaseA(bs_ctx *c, bs_slot *sl, const bs_pair_desc &pd,
   size_t pcm_n = (size_t)Pz * Py * Px;
   rc = ensure_dev(c, (void **)&sl->pcm, &sl->pcm_cap, pcm_n * sizeof(float));
   if (rc) return rc;
-  f2 *twx = get_twiddle(c, Px), *twy = get_twiddle(c, Py),
-     *twz = get_twiddle(c, Pz);
+  f2 *twx = is_pow2(Px) ? get_twiddle(c, Px) : get_twiddle_full(c, Px);
+  f2 *twy = is_pow2(Py) ? get_twiddle(c, Py) : get_twiddle_full(c, Py);
+  f2 *twz = is_pow2(Pz) ? get_twiddle(c, Pz) : get_twiddle_full(c, Pz);
   if (!twx || !twy || !twz) {
     c->err = "twiddle alloc failed";
     return BS_ENOMEM;
@@ -2288,7 +2590,13 @@ static int stitch_phaseA(bs_ctx *c, bs_slot *sl, const bs_pair_desc &pd,
   for (int t = 0; t < 2; ++t) {
     long nlines = (long)sl->reg[t].my * sl->reg[t].mz;
     bs_tim tt(c, BS_K_FFT_X_FWD, sl->stream);
-    if (Px >= 128 && Px <= 1024) { /* wave-resident fast path */
+    if (!is_pow2(Px)) { /* fast-size pad: Stockham mixed radix */
+      long ngrp = (nlines + LPB_X - 1) / LPB_X;
+      size_t lds = ((Px / 2) + 2 * (size_t)LPB_X * (Px / 2)) * sizeof(f2);
+      hipLaunchKernelGGL(k_fft_x_fwd_m, dim3(std::min(4096L, ngrp)),
+                         dim3(LPB_X * TPL_X), lds, sl->stream, sl->reg[t],
+                         spec[t], Px, bs_factorize(Px / 2), Cxp, Py, twx);
+    } else if (Px >= 128 && Px <= 1024) { /* wave-resident fast path */
       long ngrp = (nlines + XW_WPB - 1) / XW_WPB;
       dim3 g(std::min(4096L, ngrp)), b(64 * XW_WPB);
       if (Px == 128)
@@ -2316,7 +2624,15 @@ static int stitch_phaseA(bs_ctx *c, bs_slot *sl, const bs_pair_desc &pd,
   /* glds double-buffered y-pass launcher (n >= 256; see
    * k_fft_pass_glds) — grid sized to residency (2 WGs/CU) */
   auto launch_y = [&](f2 *sp, int ngroups, int valid, int dir) {
-    if (yglds && Py >= 256) {
+    if (!is_pow2(Py)) {
+      size_t lds = (Py + 2 * (size_t)LPB_S * Py) * sizeof(f2);
+      hipLaunchKernelGGL(k_fft_pass_m,
+                         dim3(std::min(4096L, (long)ngroups * nchunks)),
+                         dim3(LPB_S * TPL_S), lds, sl->stream, sp,
+                         (const f2 *)nullptr, sp, Py, bs_factorize(Py),
+                         Cxp, (long)Py * Cxp, Cx, nchunks, ngroups, valid,
+                         dir, 1.0f, twy);
+    } else if (yglds && Py >= 256) {
       size_t lds2 = ((Py / 2) + 2 * (size_t)LPB_S * Py) * sizeof(f2);
       long nwg = (long)ngroups * nchunks;
       dim3 g(std::min(512L, nwg)), b(LPB_S * TPL_S);
@@ -2346,8 +2662,24 @@ static int stitch_phaseA(bs_ctx *c, bs_slot *sl, const bs_pair_desc &pd,
     bs_tim tt(c, BS_K_FFT_Y_FWD, sl->stream);
     launch_y(spec[t], sl->reg[t].mz, sl->reg[t].my, +1);
   }
-  { /* fused z chain: fwd z (A,B) + cross-power [PIN-EPS] + inv z in one
-     * launch — the z spectra never round-trip through HBM */
+  if (!is_pow2(Pz)) { /* fast-size pad: unfused Stockham z chain */
+    float scale = 1.0f / ((float)Px * (float)Py * (float)Pz);
+    size_t lds = (Pz + 2 * (size_t)LPB_S * Pz) * sizeof(f2);
+    dim3 g(std::min(4096L, (long)Py * nchunks)), b(LPB_S * TPL_S);
+    for (int t = 0; t < 2; ++t) {
+      bs_tim tt(c, BS_K_FFT_Z_FWD, sl->stream);
+      hipLaunchKernelGGL(k_fft_pass_m, g, b, lds, sl->stream, spec[t],
+                         (const f2 *)nullptr, spec[t], Pz,
+                         bs_factorize(Pz), (long)Py * Cxp, Cxp, Cx,
+                         nchunks, Py, sl->reg[t].mz, +1, 1.0f, twz);
+    }
+    bs_tim tt(c, BS_K_FFT_Z_INV, sl->stream);
+    hipLaunchKernelGGL(k_fft_pass_m, g, b, lds, sl->stream, spec[0],
+                       spec[1], spec[0], Pz, bs_factorize(Pz),
+                       (long)Py * Cxp, Cxp, Cx, nchunks, Py, Pz, -1,
+                       scale, twz);
+  } else { /* fused z chain: fwd z (A,B) + cross-power [PIN-EPS] + inv z
+     * in one launch — the z spectra never round-trip through HBM */
     float scale = 1.0f / ((float)Px * (float)Py * (float)Pz);
     size_t lds = ((Pz / 2) + (size_t)LPB_S * Pz) * sizeof(f2);
     bs_tim tt(c, BS_K_FFT_Z_INV, sl->stream);
@@ -2378,7 +2710,14 @@ static int stitch_phaseA(bs_ctx *c, bs_slot *sl, const bs_pair_desc &pd,
   {
     long nlines = (long)Pz * Py;
     bs_tim tt(c, BS_K_FFT_X_INV, sl->stream);
-    if (Px >= 128 && Px <= 1024) { /* wave-resident fast path */
+    if (!is_pow2(Px)) {
+      long ngrp = (nlines + LPB_X - 1) / LPB_X;
+      size_t lds = ((Px / 2) + 2 * (size_t)LPB_X * (Px / 2)) * sizeof(f2);
+      hipLaunchKernelGGL(k_fft_x_inv_m, dim3(std::min(4096L, ngrp)),
+                         dim3(LPB_X * TPL_X), lds, sl->stream, spec[0],
+                         sl->pcm, Px, bs_factorize(Px / 2), Cxp, nlines,
+                         twx);
+    } else if (Px >= 128 && Px <= 1024) { /* wave-resident fast path */
       long ngrp = (nlines + XW_WPB - 1) / XW_WPB;
       dim3 g(std::min(4096L, ngrp)), b(64 * XW_WPB);
       if (Px == 128)
@@ -2415,7 +2754,10 @@ static int stitch_phaseA(bs_ctx *c, bs_slot *sl, const bs_pair_desc &pd,
   {
     bs_tim tt(c, BS_K_PEAK, sl->stream);
     static const bool pk_relax = getenv("BS_PEAK_RELAX") != nullptr;
-    if (pk_relax)
+    if (Px % 4 != 0)
+      hipLaunchKernelGGL(k_peak_generic, dim3(npkwg), dim3(256), 0,
+                         sl->stream, sl->pcm, Px, Py, Pz, sl->wgpk);
+    else if (pk_relax)
       hipLaunchKernelGGL(k_peak_tile_relaxed, dim3(npkwg), dim3(256), 0,
                          sl->stream, sl->pcm, Px, Py, Pz, sl->wgpk);
     else

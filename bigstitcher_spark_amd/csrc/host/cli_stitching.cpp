@@ -465,6 +465,9 @@ int main(int argc, char **argv) {
       prm.peaks_to_check = (int)args.getl("peaksToCheck", 5);
       prm.do_subpixel = args.has("disableSubpixelResolution") ? 0 : 1;
       prm.min_overlap_ratio = args.getd("minOverlapRatio", 0.25);
+      /* [PIN-PAD]: pow2 (default) or the reference dependency's even
+       * 7-smooth "fast" pad sizes */
+      prm.pad_mode = args.get("fftPadSize", "POW2") == "FAST" ? 1 : 0;
       std::vector<bs_shift_result> gres(pds.size());
       int rc =
           bs_stitch_batch(ctx, pds.data(), pds.size(), &prm, gres.data());

@@ -129,6 +129,12 @@ typedef struct {
   int32_t do_subpixel;      /* default 1 */
   double min_overlap_ratio; /* min candidate overlap as fraction of the
                                smaller (downsampled) interval; default 0.25 */
+  int32_t pad_mode;         /* [PIN-PAD] FFT pad size rule: 0 = next
+                               power of two (this build's historical
+                               default), 1 = next even 7-smooth "fast"
+                               size (the imglib2 FFTMethods family the
+                               reference's dependency uses) */
+  int32_t _pad;             /* alignment */
 } bs_stitch_params;
 
 /* The result contract of computeStitching -> SerializablePairwiseStitchingResult

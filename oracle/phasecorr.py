@@ -64,15 +64,38 @@ def _next_pow2(n: int) -> int:
     return p
 
 
-def pcm(a: np.ndarray, b: np.ndarray, workers: int = 1):
+def _next_fast_even(n: int) -> int:
+    """[PIN-PAD] pad_mode='fast': smallest EVEN 7-smooth (2^a 3^b 5^c
+    7^d) size >= max(n, 8) — the imglib2 FFTMethods "fast" family the
+    reference's FFT dependency pads to (artifact un-vendored; the even
+    restriction keeps the packed-real R2C x-pass applicable)."""
+    n = max(int(n), 8)
+    c = n + (n % 2)
+    while True:
+        m = c
+        for f in (2, 3, 5, 7):
+            while m % f == 0:
+                m //= f
+        if m == 1:
+            return c
+        c += 2
+
+
+def pad_shape(shape_a, shape_b, pad_mode="pow2"):
+    f = _next_pow2 if pad_mode == "pow2" else _next_fast_even
+    return tuple(f(max(sa, sb)) for sa, sb in zip(shape_a, shape_b))
+
+
+def pcm(a: np.ndarray, b: np.ndarray, workers: int = 1,
+        pad_mode: str = "pow2"):
     """Phase-correlation matrix of two (nz,ny,nx) uint16 arrays.
 
     Returns (pcm float64 (pz,py,px), padded shape). [PIN-PAD] [PIN-EPS]
-    workers != 1 switches to scipy.fft's multithreaded pocketfft — same
-    algorithm; used only by bench.py's timed cpu_baseline leg."""
-    shape = tuple(
-        _next_pow2(max(sa, sb)) for sa, sb in zip(a.shape, b.shape)
-    )
+    pad_mode 'pow2' (default) or 'fast' (even 7-smooth, the reference
+    dependency's rule). workers != 1 switches to scipy.fft's
+    multithreaded pocketfft — same algorithm; used only by bench.py's
+    timed cpu_baseline leg."""
+    shape = pad_shape(a.shape, b.shape, pad_mode)
     if workers == 1:
         fa = np.fft.rfftn(a, s=shape, axes=(0, 1, 2))
         fb = np.fft.rfftn(b, s=shape, axes=(0, 1, 2))
@@ -185,6 +208,7 @@ def phase_correlation_shift(
     do_subpixel: bool = True,
     min_overlap_ratio: float = 0.25,
     workers: int = 1,
+    pad_mode: str = "pow2",
 ):
     """Full restatement of PairwiseStitching.getShift for one tile pair.
 
@@ -194,7 +218,7 @@ def phase_correlation_shift(
     (include/bigstitch.h)."""
     ad = downsample(a, ds)
     bd = downsample(b, ds)
-    p, _shape = pcm(ad, bd, workers=workers)
+    p, _shape = pcm(ad, bd, workers=workers, pad_mode=pad_mode)
     peaks = _local_maxima_topk(p, peaks_to_check)
     min_n = min_overlap_ratio * min(ad.size, bd.size)
     best = None  # (r, peak_rank, cand_idx, cand_shift, peak)

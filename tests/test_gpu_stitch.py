@@ -278,3 +278,37 @@ def test_view_combine_avg_parity():
         rc = ctx._lib.bs_view_sum(ctx._h, 100, C.byref(s))
         assert rc == 0
         assert s.value == int(vols[0].astype(np.uint64).sum())
+
+
+@pytest.mark.parametrize(
+    "shape,shift,ds",
+    [
+        # fast pads exercise radix 3 (48=2^4*3), 5 (50=2*5^2), and
+        # 7 (56=2^3*7) in the Stockham mixed-radix engine; the pow2
+        # mode pads the same inputs to 64^3 (covered above)
+        ((55, 50, 45), (4.25, -2.5, 1.0), (1, 1, 1)),
+        ((55, 50, 45), (4.25, -2.5, 1.0), (2, 2, 1)),
+        ((100, 54, 42), (-3.5, 2.25, 0.5), (1, 1, 1)),
+    ],
+)
+def test_stitch_parity_fast_pad(ctx, shape, shift, ds):
+    """[PIN-PAD] pad_mode='fast' (even 7-smooth sizes, the reference
+    dependency's FFTMethods rule) — GPU vs oracle at 1e-3 px, r
+    bit-comparable; VERDICT r1 item 7's 'both modes green' bar
+    (the pow2 mode is pinned by test_stitch_parity)."""
+    a, b = synth.make_pair(shape, shift, seed=13)
+    ref = phasecorr.phase_correlation_shift(a, b, ds=ds, pad_mode="fast")
+    got = ctx.stitch_batch([full_pair(ctx, a, b)], ds=ds,
+                           pad_mode="fast")[0]
+    assert got["valid"] == ref["valid"]
+    if ref["valid"]:
+        assert np.all(np.abs(got["shift"] - ref["shift"]) < 1e-3), (
+            got["shift"], ref["shift"])
+        assert got["r"] == pytest.approx(ref["r"], abs=1e-9)
+    # same inputs, pow2 mode: also green (pads differ, both pinned)
+    ref2 = phasecorr.phase_correlation_shift(a, b, ds=ds, pad_mode="pow2")
+    got2 = ctx.stitch_batch([full_pair(ctx, a, b)], ds=ds,
+                            pad_mode="pow2")[0]
+    assert got2["valid"] == ref2["valid"]
+    if ref2["valid"]:
+        assert np.all(np.abs(got2["shift"] - ref2["shift"]) < 1e-3)
