@@ -739,6 +739,15 @@ __global__ __launch_bounds__(LPB_S *TPL_S) void k_fft_z_fused(
 /* no-register-prefetch variant, VGPR-capped for 4 blocks/CU (34 KB LDS
  * is the other limit at n=512): __launch_bounds__ min-blocks 4 */
 template <int NE>
+__global__ __launch_bounds__(LPB_S *TPL_S) void k_fft_z_fused_np2(
+    f2 *a, f2 *b, int n, int log2n, long estride, long gstride, int nlines,
+    int nchunks, int ngroups, int valid_a, int valid_b, float scale,
+    const f2 *twg) {
+  zf_body<NE, false>(a, b, n, log2n, estride, gstride, nlines, nchunks,
+                     ngroups, valid_a, valid_b, scale, twg);
+}
+
+template <int NE>
 __global__ __launch_bounds__(LPB_S *TPL_S) __attribute__((amdgpu_waves_per_eu(8))) void k_fft_z_fused_np(
     f2 *a, f2 *b, int n, int log2n, long estride, long gstride, int nlines,
     int nchunks, int ngroups, int valid_a, int valid_b, float scale,
@@ -1019,13 +1028,13 @@ __global__ __launch_bounds__(64 * XW_WPB) void k_fft_x_fwd_w(
     if (al4) {
       /* contiguous u32 loads (2 cachelines per wave-load instead of a
        * 16-line bit-reversed gather), bit reversal via one LDS
-       * round-trip; all waves iterate niter times so the barrier is
-       * uniform */
+       * round-trip. stg[wv] is WAVE-PRIVATE: no cross-wave barrier —
+       * the compiler's own lgkm waits order the same-wave
+       * ds_write -> ds_read dependency. */
 #pragma unroll
       for (int e = 0; e < E; ++e)
         stg[wv][e * 64 + lane] = ((const unsigned *)src)[e * 64 + lane];
     }
-    __syncthreads();
 #pragma unroll
     for (int e = 0; e < E; ++e) {
       const int j = (int)brev_n((unsigned)(e * 64 + lane), log2h);
@@ -1063,7 +1072,6 @@ __global__ __launch_bounds__(64 * XW_WPB) void k_fft_x_fwd_w(
         o[k] = {ze.x + wzo.x, ze.y + wzo.y};
       }
     }
-    __syncthreads(); /* stg reused next iteration */
   }
 }
 
@@ -2028,10 +2036,13 @@ struct bs_slot {
   int stage = 0; /* 0 idle, 1 fft+peak issued, 2 rtest issued */
 };
 
-#define BS_NSLOTS 3
+#define BS_NSLOTS 4
 
 struct bs_ctx {
   int dev;
+  int nslots = 2; /* pipeline slots in use (<= BS_NSLOTS, BS_SLOTS env;
+                      2 measured best with the round-2 kernel mix: less
+                      cross-stream HBM contention) */
   std::map<int, f2 *> twiddles_full; /* full n-entry tables (fast pads) */
   hipStream_t stream; /* default stream: views, synth, fusion */
   hipStream_t copy_stream = nullptr; /* D2H overlap (fusion volume) */
@@ -2103,6 +2114,10 @@ extern "C" int bs_ctx_create(bs_ctx **out, int device_id) {
   }
   bs_ctx *c = new bs_ctx();
   c->dev = device_id;
+  if (const char *e = getenv("BS_SLOTS")) {
+    int v = atoi(e);
+    if (v >= 1 && v <= BS_NSLOTS) c->nslots = v;
+  }
   if (hipSetDevice(device_id) != hipSuccess ||
       hipStreamCreate(&c->stream) != hipSuccess ||
       hipStreamCreate(&c->copy_stream) != hipSuccess) {
@@ -2692,14 +2707,15 @@ static int stitch_phaseA(bs_ctx *c, bs_slot *sl, const bs_pair_desc &pd,
     /* n>=512: the occupancy-capped no-prefetch variant wins (A/B on
      * hardware: 0.734 vs 0.842 ms at 512^3 — 4 blocks/CU beats the
      * B-register prefetch at 2 blocks/CU) */
+    static const bool z_ns = getenv("BS_Z_NOSPILL") != nullptr;
     if (Pz <= 128)
       zf(k_fft_z_fused<1>);
     else if (Pz == 256)
       zf(k_fft_z_fused<2>);
     else if (Pz == 512)
-      zf(k_fft_z_fused_np<4>);
+      z_ns ? zf(k_fft_z_fused_np2<4>) : zf(k_fft_z_fused_np<4>);
     else
-      zf(k_fft_z_fused_np<8>);
+      z_ns ? zf(k_fft_z_fused_np2<8>) : zf(k_fft_z_fused_np<8>);
   }
   {
     size_t lds = ((Py / 2) + (size_t)LPB_S * Py) * sizeof(f2);
@@ -2853,8 +2869,12 @@ static int stitch_phaseB(bs_ctx *c, bs_slot *sl,
       for (auto &h : sl->hc)
         maxz = std::max(maxz, h.gc.loz + h.gc.nz);
       bs_tim tt(c, BS_K_CORR, sl->stream);
-      hipLaunchKernelGGL(k_rtest, dim3((unsigned)maxz, 4), dim3(256), 0,
-                         sl->stream, sl->reg[0], sl->reg[1], sl->dcands,
+      static const int corr_gy = [] {
+        const char *e = getenv("BS_CORR_GY");
+        return e ? atoi(e) : 8; /* measured: 8 > 4 > 2 >> 1 at 512^3 */
+      }();
+      hipLaunchKernelGGL(k_rtest, dim3((unsigned)maxz, corr_gy), dim3(256),
+                         0, sl->stream, sl->reg[0], sl->reg[1], sl->dcands,
                          (int)gc.size(), sl->dsums);
     }
     CHK(c, hipMemcpyAsync(sl->hsums, sl->dsums,
@@ -2946,15 +2966,15 @@ extern "C" int bs_stitch_batch(bs_ctx *c, const bs_pair_desc *pairs, size_t np,
   auto t0 = std::chrono::steady_clock::now();
   int rc = BS_OK;
   for (size_t i = 0; i < np && rc == BS_OK; ++i) {
-    bs_slot *sl = &c->slot[i % BS_NSLOTS];
+    bs_slot *sl = &c->slot[i % c->nslots];
     if (sl->stage == 1) rc = stitch_phaseB(c, sl, prm);
     if (rc == BS_OK && sl->stage == 2) rc = stitch_phaseC(c, sl, prm, out);
     if (rc == BS_OK) rc = stitch_phaseA(c, sl, pairs[i], prm, i);
   }
   /* drain the open slots, oldest first */
-  for (size_t k = np >= BS_NSLOTS ? np - BS_NSLOTS : 0;
+  for (size_t k = np >= (size_t)c->nslots ? np - c->nslots : 0;
        k < np && rc == BS_OK; ++k) {
-    bs_slot *sl = &c->slot[k % BS_NSLOTS];
+    bs_slot *sl = &c->slot[k % c->nslots];
     if (sl->stage == 1) rc = stitch_phaseB(c, sl, prm);
     if (rc == BS_OK && sl->stage == 2) rc = stitch_phaseC(c, sl, prm, out);
   }
