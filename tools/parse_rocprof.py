@@ -38,7 +38,9 @@ NAME_MAP = {
     "k_fft_x_inv": ["fft_x_inv"],
     "k_fft_x_fwd_w": ["fft_x_fwd"],
     "k_fft_x_inv_w": ["fft_x_inv"],
-    "k_fft_pass": ["fft_y_fwd", "fft_z_fwd", "fft_z_inv", "fft_y_inv"],
+    "k_fft_pass": ["fft_y_fwd", "fft_y_inv"],
+    "k_fft_z_fused_np": ["fft_z_inv"],  # round-2 fused z chain
+    "k_fft_z_fused": ["fft_z_inv"],
     "k_peak_tile": ["peak"],
     "k_rtest": ["corr"],
     "k_fuse": ["fuse"],
