@@ -2707,15 +2707,14 @@ static int stitch_phaseA(bs_ctx *c, bs_slot *sl, const bs_pair_desc &pd,
     /* n>=512: the occupancy-capped no-prefetch variant wins (A/B on
      * hardware: 0.734 vs 0.842 ms at 512^3 — 4 blocks/CU beats the
      * B-register prefetch at 2 blocks/CU) */
-    static const bool z_ns = getenv("BS_Z_NOSPILL") != nullptr;
     if (Pz <= 128)
       zf(k_fft_z_fused<1>);
     else if (Pz == 256)
       zf(k_fft_z_fused<2>);
     else if (Pz == 512)
-      z_ns ? zf(k_fft_z_fused_np2<4>) : zf(k_fft_z_fused_np<4>);
+      zf(k_fft_z_fused_np<4>);
     else
-      z_ns ? zf(k_fft_z_fused_np2<8>) : zf(k_fft_z_fused_np<8>);
+      zf(k_fft_z_fused_np<8>);
   }
   {
     size_t lds = ((Py / 2) + (size_t)LPB_S * Py) * sizeof(f2);
