@@ -9,9 +9,12 @@
  * AVG/AVG_BLEND/MAX_INTENSITY; --masks writes coverage masks instead of
  * fused intensities (reference SparkAffineFusion.java:112-115, :565-578
  * via fusion/GenerateComputeBlockMasks.java). */
+#include <atomic>
 #include <cstdio>
 #include <cstring>
 #include <set>
+#include <thread>
+#include <vector>
 
 #include "../../../include/bigstitch.h"
 #include "bs_cli_util.h"
@@ -380,36 +383,53 @@ int main(int argc, char **argv) {
       }
       long long lx = ldims[l * 3], ly = ldims[l * 3 + 1],
                 lz = ldims[l * 3 + 2];
-      std::vector<char> blk((size_t)bx * by * bz * esz2);
-      long long nwritten = 0;
-      for (long long gz = 0; gz * bz < lz; ++gz)
-        for (long long gy = 0; gy * by < ly; ++gy)
-          for (long long gx = 0; gx * bx < lx; ++gx) {
-            int cx = (int)std::min((long long)bx, lx - gx * bx);
-            int cy = (int)std::min((long long)by, ly - gy * by);
-            int cz = (int)std::min((long long)bz, lz - gz * bz);
-            const char *src = hostlvl[l].data();
-            for (int z = 0; z < cz; ++z)
-              for (int y = 0; y < cy; ++y)
-                memcpy(&blk[((size_t)z * cy + y) * cx * esz2],
-                       src + (((gz * bz + z) * ly + gy * by + y) * lx +
-                              gx * bx) * esz2,
-                       (size_t)cx * esz2);
-            bool ok;
-            if (zarr)
-              ok = zr.write_chunk(dsnames[l], za, {ti, 0, gz, gy, gx},
-                                  blk.data(), {1, 1, cz, cy, cx});
-            else
-              ok = n5.write_block(dsnames[l], da, {gx, gy, gz}, blk.data(),
-                                  {cx, cy, cz});
-            if (!ok) {
-              fprintf(stderr, "block write failed\n");
-              return 1;
-            }
-            ++nwritten;
-          }
+      /* chunk compression (zstd/gzip) + file writes are independent
+       * per chunk: spread over a thread pool — the reference gets the
+       * same parallelism from its Spark executors writing blocks
+       * concurrently (N5Utils.saveBlock per task) */
+      const long long ngx = (lx + bx - 1) / bx, ngy = (ly + by - 1) / by,
+                      ngz = (lz + bz - 1) / bz;
+      const long long nchunks = ngx * ngy * ngz;
+      const int NW = (int)std::min<long long>(
+          nchunks, std::max(1u, std::thread::hardware_concurrency() / 2));
+      std::atomic<long long> next(0);
+      std::atomic<bool> failed(false);
+      auto worker = [&]() {
+        std::vector<char> blk((size_t)bx * by * bz * esz2);
+        for (;;) {
+          long long i = next.fetch_add(1);
+          if (i >= nchunks || failed.load()) return;
+          const long long gx = i % ngx, gy = (i / ngx) % ngy,
+                          gz = i / (ngx * ngy);
+          int cx = (int)std::min((long long)bx, lx - gx * bx);
+          int cy = (int)std::min((long long)by, ly - gy * by);
+          int cz = (int)std::min((long long)bz, lz - gz * bz);
+          const char *src = hostlvl[l].data();
+          for (int z = 0; z < cz; ++z)
+            for (int y = 0; y < cy; ++y)
+              memcpy(&blk[((size_t)z * cy + y) * cx * esz2],
+                     src + (((gz * bz + z) * ly + gy * by + y) * lx +
+                            gx * bx) * esz2,
+                     (size_t)cx * esz2);
+          bool ok;
+          if (zarr)
+            ok = zr.write_chunk(dsnames[l], za, {ti, 0, gz, gy, gx},
+                                blk.data(), {1, 1, cz, cy, cx});
+          else
+            ok = n5.write_block(dsnames[l], da, {gx, gy, gz}, blk.data(),
+                                {cx, cy, cz});
+          if (!ok) failed.store(true);
+        }
+      };
+      std::vector<std::thread> ws;
+      for (int w = 0; w < NW; ++w) ws.emplace_back(worker);
+      for (auto &w : ws) w.join();
+      if (failed.load()) {
+        fprintf(stderr, "block write failed\n");
+        return 1;
+      }
       printf("level %d (%s): wrote %lld blocks (%lldx%lldx%lld)\n", l,
-             dsnames[l].c_str(), nwritten, lx, ly, lz);
+             dsnames[l].c_str(), nchunks, lx, ly, lz);
     }
     for (auto &fv : fviews) bs_view_release(ctx, fv.view_id);
   }
