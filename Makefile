@@ -55,3 +55,20 @@ clean-cli:
 	rm -f $(HOSTOBJS) $(BINDIR)/*
 
 .PHONY: cli clean-cli
+
+# ---- host-side AddressSanitizer build (sanitizer config)
+# Builds the GPU-free host binaries (create-fusion-container, solver)
+# with ASAN and runs the CPU host tests against them:
+#   make asan && BS_BIN=bigstitcher_spark_amd/bin-asan python -m pytest \
+#       tests/test_cli_host.py -q -m "not gpu"
+ASANDIR = bigstitcher_spark_amd/bin-asan
+ASANFLAGS = -O1 -g -std=c++17 -fsanitize=address -fno-omit-frame-pointer -Wall
+
+asan:
+	@mkdir -p $(ASANDIR)
+	$(CXX_HOST) $(ASANFLAGS) -Iinclude $(HOSTDIR)/cli_container.cpp $(HOSTDIR)/bs_json.cpp $(HOSTDIR)/bs_n5.cpp $(HOSTDIR)/bs_zarr.cpp $(HOSTDIR)/bs_xml.cpp $(HOSTDIR)/bs_spimdata.cpp -o $(ASANDIR)/create-fusion-container -lz -l:libzstd.so.1
+	$(CXX_HOST) $(ASANFLAGS) -Iinclude $(HOSTDIR)/cli_solver.cpp $(HOSTDIR)/bs_json.cpp $(HOSTDIR)/bs_n5.cpp $(HOSTDIR)/bs_zarr.cpp $(HOSTDIR)/bs_xml.cpp $(HOSTDIR)/bs_spimdata.cpp -o $(ASANDIR)/solver -lz -l:libzstd.so.1
+	$(CXX_HOST) $(ASANFLAGS) -Iinclude $(HOSTDIR)/cli_stitching.cpp $(HOSTDIR)/bs_json.cpp $(HOSTDIR)/bs_n5.cpp $(HOSTDIR)/bs_zarr.cpp $(HOSTDIR)/bs_xml.cpp $(HOSTDIR)/bs_spimdata.cpp -o $(ASANDIR)/stitching -Lbigstitcher_spark_amd -lbigstitch -lz -l:libzstd.so.1 -L/opt/rocm/lib -lamdhip64 -Wl,-rpath,'$$ORIGIN/..' -Wl,-rpath,/opt/rocm/lib
+	$(CXX_HOST) $(ASANFLAGS) -Iinclude $(HOSTDIR)/cli_resave.cpp $(HOSTDIR)/bs_json.cpp $(HOSTDIR)/bs_n5.cpp $(HOSTDIR)/bs_zarr.cpp $(HOSTDIR)/bs_xml.cpp $(HOSTDIR)/bs_spimdata.cpp -o $(ASANDIR)/resave -Lbigstitcher_spark_amd -lbigstitch -lz -l:libzstd.so.1 -L/opt/rocm/lib -lamdhip64 -Wl,-rpath,'$$ORIGIN/..' -Wl,-rpath,/opt/rocm/lib
+
+.PHONY: asan

@@ -15,7 +15,8 @@ from oracle import phasecorr, synth
 from tests import n5util
 
 ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
-BIN = os.path.join(ROOT, "bigstitcher_spark_amd", "bin")
+BIN = os.environ.get("BS_BIN", os.path.join(ROOT, "bigstitcher_spark_amd", "bin"))
+BIN = BIN if os.path.isabs(BIN) else os.path.join(ROOT, BIN)
 
 
 def run(cmd, **kw):
