@@ -148,7 +148,9 @@ def test_cli_fusion_end_to_end(tmp_path):
     s1, _ = n5util.read_dataset(out, "ch0tp0/s1")
     assert s1.shape == (32, 32, 52)
     ref1 = of.downsample_level(fused, (2, 2, 2))
-    assert np.max(np.abs(s1 - ref1)) < 1e-3
+    # relative: s0 itself is pinned at 1e-4 relative, so the box-mean
+    # inherits ulp-level reassociation differences
+    assert np.max(np.abs(s1 - ref1) / np.maximum(np.abs(ref1), 1.0)) < 1e-4
 
 
 @pytest.mark.gpu
