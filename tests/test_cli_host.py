@@ -888,3 +888,89 @@ def test_cli_stitching_nonequal_transforms(tmp_path):
     want = (-err[0], -err[1], -err[2])
     for g, w in zip((m[3], m[7], m[11]), want):
         assert abs(g - w) < 1.0, ((m[3], m[7], m[11]), want)
+
+
+def make_two_tp_dataset(tmp, size=48, overlap=16, seed=3):
+    """2x1 grid at two timepoints (independent content per tp)."""
+    n5 = os.path.join(tmp, "input.n5")
+    xml = os.path.join(tmp, "dataset.xml")
+    shape = (size, size, size)
+    posB = size - overlap
+    setups = [dict(id=0, dims=shape, pos=(0.0, 0.0, 0.0)),
+              dict(id=1, dims=shape, pos=(float(posB), 0.0, 0.0))]
+    vols = {}
+    for tp in (0, 1):
+        s = (-(posB + 1.5 + tp), -0.5, 0.25)
+        ba, bb = synth.pair_blobs_union(shape, s, seed=seed + tp)
+        for sid, blobs in ((0, ba), (1, bb)):
+            v = synth.render_tile(shape, blobs, noise_seed=50 * tp + sid)
+            vols[(tp, sid)] = v
+            n5util.write_dataset(n5, f"setup{sid}/timepoint{tp}/s0", v,
+                                 (32, 32, 32))
+    n5util.make_dataset_xml(xml, "input.n5", setups)
+    # extend to two timepoints: duplicate registrations for tp 1
+    t = ET.parse(xml)
+    root = t.getroot()
+    pat = root.find(".//Timepoints/integerpattern")
+    pat.text = "0,1"
+    vrs = root.find("ViewRegistrations")
+    for vr in list(vrs.findall("ViewRegistration")):
+        import copy as _copy
+        vr2 = _copy.deepcopy(vr)
+        vr2.set("timepoint", "1")
+        vrs.append(vr2)
+    t.write(xml)
+    return xml, n5, vols
+
+
+def test_two_timepoints_dryrun(tmp_path):
+    xml, n5, _ = make_two_tp_dataset(str(tmp_path))
+    r = run([os.path.join(BIN, "stitching"), "-x", xml, "--dryRun"])
+    assert r.returncode == 0, r.stderr
+    assert "timepoint 0: 1 overlapping pairs" in r.stdout
+    assert "timepoint 1: 1 overlapping pairs" in r.stdout
+    # --timepointId restricts
+    r = run([os.path.join(BIN, "stitching"), "-x", xml, "--dryRun",
+             "--timepointId", "1"])
+    assert r.returncode == 0, r.stderr
+    assert "timepoint 0: 0 overlapping pairs" in r.stdout
+    assert "timepoint 1: 1 overlapping pairs" in r.stdout
+
+
+@pytest.mark.gpu
+def test_two_timepoints_stitch_fuse(tmp_path):
+    """Multi-timepoint surface: stitching writes one entry per tp pair;
+    the container carries NumTimepoints=2 and fusion fills both
+    ch0tp0/s0 and ch0tp1/s0 with their own timepoint's content."""
+    xml, n5, vols = make_two_tp_dataset(str(tmp_path))
+    r = run([os.path.join(BIN, "stitching"), "-x", xml, "-ds", "1,1,1",
+             "--minOverlapRatio", "0.05"])
+    assert r.returncode == 0, r.stderr + r.stdout
+    prs = ET.parse(xml).getroot().findall(".//PairwiseResult")
+    assert len(prs) == 2
+    tps = sorted(pr.find("ViewIdsA").text.split(",")[0] for pr in prs)
+    assert tps == ["0", "1"]
+    out = os.path.join(str(tmp_path), "fused.n5")
+    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml,
+             "-s", "N5", "-o", out, "--blockSize", "32,32,32",
+             "-d", "FLOAT32"])
+    assert r.returncode == 0, r.stderr
+    attrs = n5util.root_attrs(out)["Bigstitcher-Spark"]
+    assert attrs["NumTimepoints"] == 2
+    r = run([os.path.join(BIN, "affine-fusion"), "-o", out,
+             "-f", "AVG", "--blendingRange", "0"])
+    assert r.returncode == 0, r.stderr + r.stdout
+    bbmin = attrs["Boundingbox_min"]
+    for tp in (0, 1):
+        fused, _ = n5util.read_dataset(out, f"ch0tp{tp}/s0")
+        ident = np.hstack([np.eye(3), np.zeros((3, 1))])
+        affB = np.hstack([np.eye(3), np.array([[32.0], [0.0], [0.0]])])
+        views = [dict(data=vols[(tp, 0)], affine=ident,
+                      border=(0, 0, 0), range=(0, 0, 0)),
+                 dict(data=vols[(tp, 1)], affine=affB,
+                      border=(0, 0, 0), range=(0, 0, 0))]
+        ref = of.fuse_block(views, tuple(bbmin), fused.shape[::-1],
+                            of.FUSION_AVG)
+        denom = np.maximum(np.abs(ref), 1.0)
+        assert np.max(np.abs(fused.astype(np.float64) - ref) / denom) \
+            < 1e-4, f"tp{tp}"
