@@ -9,8 +9,11 @@
  * it. The pyramid levels are computed on the GPU (the K8 box-mean
  * kernel via bs_fuse_volume with the single identity view, which is an
  * exact uint16 round-trip at minI=0/maxI=65535). */
+#include <atomic>
 #include <cstdio>
 #include <cstring>
+#include <thread>
+#include <vector>
 
 #include "../../../include/bigstitch.h"
 #include "bs_cli_util.h"
@@ -132,26 +135,46 @@ int main(int argc, char **argv) {
                             ladder[l][0], ladder[l][1], ladder[l][2]}));
         long long lx = ldims[l * 3], ly = ldims[l * 3 + 1],
                   lz = ldims[l * 3 + 2];
-        std::vector<char> blk((size_t)bs[0] * bs[1] * bs[2] * 2);
-        for (long long gz = 0; gz * bs[2] < lz; ++gz)
-          for (long long gy = 0; gy * bs[1] < ly; ++gy)
-            for (long long gx = 0; gx * bs[0] < lx; ++gx) {
-              int cx = (int)std::min(bs[0], lx - gx * bs[0]);
-              int cy = (int)std::min(bs[1], ly - gy * bs[1]);
-              int cz = (int)std::min(bs[2], lz - gz * bs[2]);
-              const char *src = hostlvl[l].data();
-              for (int z = 0; z < cz; ++z)
-                for (int y = 0; y < cy; ++y)
-                  memcpy(&blk[((size_t)z * cy + y) * cx * 2],
-                         src + (((gz * bs[2] + z) * ly + gy * bs[1] + y) *
-                                    lx + gx * bs[0]) * 2,
-                         (size_t)cx * 2);
-              if (!out_n5.write_block(dsn, da, {gx, gy, gz}, blk.data(),
-                                      {cx, cy, cz})) {
-                fprintf(stderr, "block write failed\n");
-                return 1;
-              }
-            }
+        /* independent chunks: compression + file writes on a pool (as
+         * the reference's Spark executors write blocks concurrently) */
+        const long long ngx = (lx + bs[0] - 1) / bs[0];
+        const long long ngy = (ly + bs[1] - 1) / bs[1];
+        const long long ngz = (lz + bs[2] - 1) / bs[2];
+        const long long nchunks = ngx * ngy * ngz;
+        const int NW = (int)std::min<long long>(
+            nchunks,
+            std::max(1u, std::thread::hardware_concurrency() / 2));
+        std::atomic<long long> next(0);
+        std::atomic<bool> failed(false);
+        auto worker = [&]() {
+          std::vector<char> blk((size_t)bs[0] * bs[1] * bs[2] * 2);
+          for (;;) {
+            long long i = next.fetch_add(1);
+            if (i >= nchunks || failed.load()) return;
+            const long long gx = i % ngx, gy = (i / ngx) % ngy,
+                            gz = i / (ngx * ngy);
+            int cx = (int)std::min(bs[0], lx - gx * bs[0]);
+            int cy = (int)std::min(bs[1], ly - gy * bs[1]);
+            int cz = (int)std::min(bs[2], lz - gz * bs[2]);
+            const char *src = hostlvl[l].data();
+            for (int z = 0; z < cz; ++z)
+              for (int y = 0; y < cy; ++y)
+                memcpy(&blk[((size_t)z * cy + y) * cx * 2],
+                       src + (((gz * bs[2] + z) * ly + gy * bs[1] + y) *
+                                  lx + gx * bs[0]) * 2,
+                       (size_t)cx * 2);
+            if (!out_n5.write_block(dsn, da, {gx, gy, gz}, blk.data(),
+                                    {cx, cy, cz}))
+              failed.store(true);
+          }
+        };
+        std::vector<std::thread> ws;
+        for (int w = 0; w < NW; ++w) ws.emplace_back(worker);
+        for (auto &w : ws) w.join();
+        if (failed.load()) {
+          fprintf(stderr, "block write failed\n");
+          return 1;
+        }
       }
       printf("resaved setup %d tp %d: %d level(s)\n", s.id, tp, nlevels);
     }
