@@ -51,6 +51,12 @@ int main(int argc, char **argv) {
             "--illuminationId/--timepointId '0,1,..']\n");
     return 2;
   }
+  if (args.has("dryRun")) {
+    /* reference CreateFusionContainer.java:126-130: dry-run is not
+     * supported for this command — print and exit cleanly */
+    printf("dry-run not supported for CreateFusionContainer.\n");
+    return 0;
+  }
   if (args.has("bdv")) {
     fprintf(stderr, "--bdv (BDV N5/HDF5 container variants) is not "
                     "supported by this build\n");

@@ -200,6 +200,13 @@ int main(int argc, char **argv) {
   float brange = (float)args.getd("blendingRange", 40.0);
   float bborder = (float)args.getd("blendingBorder", 0.0);
 
+  if (args.has("dryRun")) {
+    printf("dry run: would fuse container %s (FusionFormat %s, bbox "
+           "[%lld,%lld,%lld]..[%lld,%lld,%lld], %s)\n",
+           args.get("n5Path").c_str(), fmt->str.c_str(), bbmin[0],
+           bbmin[1], bbmin[2], bbmax[0], bbmax[1], bbmax[2], dt.c_str());
+    return 0;
+  }
   bssd::SpimData sd;
   std::string err;
   if (!sd.load(xml, &err)) {

@@ -974,3 +974,24 @@ def test_two_timepoints_stitch_fuse(tmp_path):
         denom = np.maximum(np.abs(ref), 1.0)
         assert np.max(np.abs(fused.astype(np.float64) - ref) / denom) \
             < 1e-4, f"tp{tp}"
+
+
+def test_dryrun_flags(tmp_path):
+    """--dryRun semantics: container prints the reference's
+    not-supported notice and creates NOTHING; fusion prints the plan
+    without touching the GPU."""
+    xml, n5, _err, _ = make_grid_dataset(str(tmp_path))
+    out = os.path.join(str(tmp_path), "x.n5")
+    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml,
+             "-o", out, "--dryRun"])
+    assert r.returncode == 0
+    assert "dry-run not supported" in r.stdout
+    assert not os.path.exists(out)
+    # a real container, then fusion --dryRun (no GPU needed)
+    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml,
+             "-s", "N5", "-o", out, "--blockSize", "32,32,32",
+             "-d", "UINT16"])
+    assert r.returncode == 0, r.stderr
+    r = run([os.path.join(BIN, "affine-fusion"), "-o", out, "--dryRun"])
+    assert r.returncode == 0, r.stderr + r.stdout
+    assert "dry run" in r.stdout
