@@ -36,9 +36,11 @@ int main(int argc, char **argv) {
       {"-ds", "--downsampling"},
       {"--downsamplings", "--downsampling"}, /* pre-round-2 spelling */
       {"-tp", "--numTimepoints"}, {"-ch", "--numChannels"},
+      {"-xo", "--xmlout"},
       {"-vi", "--vi"}};
   if (!args.parse(argc, argv, alias,
-                  {"preserveAnisotropy", "multiRes", "dryRun", "bdv"}) ||
+                  {"preserveAnisotropy", "multiRes", "dryRun", "bdv",
+                   "localSparkBindAddress"}) ||
       !args.has("xml") || !args.has("outputPath")) {
     fprintf(stderr,
             "usage: create-fusion-container -x dataset.xml -o out.zarr "
@@ -57,6 +59,10 @@ int main(int argc, char **argv) {
     printf("dry-run not supported for CreateFusionContainer.\n");
     return 0;
   }
+  for (const char *f : {"s3Region", "localSparkBindAddress"})
+    if (args.has(f))
+      fprintf(stderr, "note: --%s accepted for compatibility (no-op in "
+                      "this build)\n", f);
   if (args.has("bdv")) {
     fprintf(stderr, "--bdv (BDV N5/HDF5 container variants) is not "
                     "supported by this build\n");

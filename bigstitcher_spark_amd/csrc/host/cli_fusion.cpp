@@ -91,8 +91,11 @@ int main(int argc, char **argv) {
       {"-o", "--n5Path"},
       {"-f", "--fusion"}, /* reference spelling, SparkAffineFusion.java:124 */
       {"--fusionType", "--fusion"}, /* pre-round-2 spelling */
-      {"-vi", "--vi"}};
-  if (!args.parse(argc, argv, alias, {"masks", "prefetch", "dryRun"}) ||
+      {"-vi", "--vi"},
+      {"-s", "--storage"}};
+  if (!args.parse(argc, argv, alias,
+                  {"masks", "prefetch", "dryRun",
+                   "localSparkBindAddress"}) ||
       !args.has("n5Path")) {
     fprintf(stderr,
             "usage: affine-fusion -o out.n5 [-x dataset.xml] "
@@ -106,6 +109,16 @@ int main(int argc, char **argv) {
             "[--intensityN5Dataset intensity]]\n");
     return 2;
   }
+  /* compatibility no-ops: --blockScale shapes Spark super-blocks (the
+   * volume-mode fusion supersedes it), -s/--storage and
+   * --intensityN5Storage override format GUESSING (this build reads
+   * the container's own FusionFormat attribute), --prefetch is a
+   * cloud-latency hint (views are HBM-resident here) */
+  for (const char *f : {"blockScale", "storage", "intensityN5Storage",
+                        "s3Region", "localSparkBindAddress", "prefetch"})
+    if (args.has(f))
+      fprintf(stderr, "note: --%s accepted for compatibility (no-op in "
+                      "this build)\n", f);
   bsn5::Container n5(args.get("n5Path"));
   bszarr::Container zr(args.get("n5Path"));
   auto geta = [&](const std::string &k) {

@@ -29,7 +29,8 @@ int main(int argc, char **argv) {
       {"-x", "--xml"}, {"-ds", "--downsampling"}, {"-p", "--peaksToCheck"},
       {"-vi", "--vi"}};
   if (!args.parse(argc, argv, alias,
-                  {"disableSubpixelResolution", "dryRun"}) ||
+                  {"disableSubpixelResolution", "dryRun",
+                   "localSparkBindAddress"}) ||
       !args.has("xml")) {
     fprintf(stderr,
             "usage: stitching -x dataset.xml [-ds 2,2,1] [-p 5] "
@@ -42,6 +43,12 @@ int main(int argc, char **argv) {
             "[--dryRun]\n");
     return 2;
   }
+  /* Spark-infrastructure flags accepted for drop-in compatibility;
+   * meaningless here (no Spark, no S3) */
+  for (const char *f : {"s3Region", "localSparkBindAddress"})
+    if (args.has(f))
+      fprintf(stderr, "note: --%s accepted for compatibility (no-op in "
+                      "this build)\n", f);
   const std::string chComb = args.get("channelCombine", "AVERAGE");
   const std::string ilComb = args.get("illumCombine", "PICK_BRIGHTEST");
   for (auto &a : {chComb, ilComb})
