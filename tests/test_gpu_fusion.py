@@ -296,3 +296,27 @@ def test_fuse_view_table_spill_over_64(ctx):
                           out_dtype=np.float32)[0]
     denom = np.maximum(np.abs(ref), 1.0)
     assert np.max(np.abs(got - ref) / denom) < 1e-4
+
+
+@pytest.mark.gpu
+def test_fuse_block_no_covering_views():
+    """A block fully outside every view: all voxels uncovered -> zeros
+    (reference BlkAffineFusion semantics for empty overlap)."""
+    from bigstitcher_spark_amd import Context
+    import numpy as np
+    from oracle import fusion as of
+    with Context(0) as ctx:
+        vol = np.full((8, 8, 8), 1234, np.uint16)
+        aff = np.hstack([np.eye(3), np.zeros((3, 1))])
+        ctx.upload(90, vol)
+        out = ctx.fuse_blocks(
+            [dict(view_id=90, affine=aff, border=(0, 0, 0),
+                  range=(0, 0, 0))],
+            [((100, 100, 100), (8, 8, 8))], [[0]],
+            out_dtype=np.float32)[0]
+        assert np.all(out == 0.0)
+        ref = of.fuse_block(
+            [dict(data=vol, affine=aff, border=(0, 0, 0),
+                  range=(0, 0, 0))],
+            (100, 100, 100), (8, 8, 8), of.FUSION_AVG_BLEND)
+        assert np.array_equal(out, ref.astype(np.float32))

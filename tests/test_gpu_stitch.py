@@ -312,3 +312,15 @@ def test_stitch_parity_fast_pad(ctx, shape, shift, ds):
     assert got2["valid"] == ref2["valid"]
     if ref2["valid"]:
         assert np.all(np.abs(got2["shift"] - ref2["shift"]) < 1e-3)
+
+
+@pytest.mark.gpu
+def test_stitch_constant_tiles_invalid(ctx):
+    """Degenerate content: constant tiles have zero variance, every
+    candidate's r denominator is 0 -> no valid result (oracle and GPU
+    agree on validity)."""
+    a = np.full((32, 32, 32), 500, np.uint16)
+    b = np.full((32, 32, 32), 700, np.uint16)
+    ref = phasecorr.phase_correlation_shift(a, b, ds=(1, 1, 1))
+    got = ctx.stitch_batch([full_pair(ctx, a, b, 50, 51)], ds=(1, 1, 1))[0]
+    assert got["valid"] == ref["valid"] == False  # noqa: E712
