@@ -3140,12 +3140,21 @@ extern "C" int bs_fuse_blocks(bs_ctx *c, const bs_fuse_view *views,
   for (size_t i = 0; i < nb; ++i)
     maxvox = std::max(maxvox, (size_t)blocks[i].size[0] * blocks[i].size[1] *
                                   blocks[i].size[2]);
-  rc = ensure_dev(c, (void **)&c->fuse_out, &c->fuse_cap, maxvox * esz);
+  /* double-buffered output so block i+1's kernel overlaps block i's
+   * D2H (copy_stream gated on a per-block event; round 1 serialized
+   * kernel -> copy -> kernel on one stream) */
+  rc = ensure_dev(c, (void **)&c->fuse_out, &c->fuse_cap, 2 * maxvox * esz);
   if (rc) return rc;
+  hipEvent_t bev[2] = {nullptr, nullptr};
+  CHK(c, hipEventCreateWithFlags(&bev[0], hipEventDisableTiming));
+  CHK(c, hipEventCreateWithFlags(&bev[1], hipEventDisableTiming));
   for (size_t i = 0; i < nb; ++i) {
     const bs_block_desc &bd = blocks[i];
     long nvox = (long)bd.size[0] * bd.size[1] * bd.size[2];
     int nvb = (int)(view_idx_offsets[i + 1] - view_idx_offsets[i]);
+    void *obuf = (char *)c->fuse_out + (i & 1) * maxvox * esz;
+    if (i >= 2) /* buffer reused: its previous D2H must be done */
+      CHK(c, hipStreamWaitEvent(c->stream, bev[i & 1], 0));
     {
       long nrows_f = (long)bd.size[1] * bd.size[2];
       long gfb = std::max(8L, std::min(4096L, nrows_f) & ~7L);
@@ -3160,7 +3169,7 @@ extern "C" int bs_fuse_blocks(bs_ctx *c, const bs_fuse_view *views,
                            (int)bd.size[1], (int)bd.size[2], prm->out_dtype,
                            (float)prm->mask_offset[0],
                            (float)prm->mask_offset[1],
-                           (float)prm->mask_offset[2], c->fuse_out, 0L,
+                           (float)prm->mask_offset[2], obuf, 0L,
                            (long)bd.size[0], (long)bd.size[0] * bd.size[1]);
       else
         hipLaunchKernelGGL(k_fuse, dim3(gfb),
@@ -3169,15 +3178,21 @@ extern "C" int bs_fuse_blocks(bs_ctx *c, const bs_fuse_view *views,
                            bd.min[1], bd.min[2], (int)bd.size[0],
                            (int)bd.size[1], (int)bd.size[2],
                            prm->fusion_type, prm->out_dtype,
-                           (float)prm->min_intensity, invRange, c->fuse_out,
+                           (float)prm->min_intensity, invRange, obuf,
                            0L, (long)bd.size[0],
                            (long)bd.size[0] * bd.size[1],
                            getenv("BS_FUSE_NORIV") ? 0 : 1);
     }
-    CHK(c, hipMemcpyAsync(out_blocks[i], c->fuse_out, nvox * esz,
-                          hipMemcpyDeviceToHost, c->stream));
+    CHK(c, hipEventRecord(bev[i & 1], c->stream));
+    CHK(c, hipStreamWaitEvent(c->copy_stream, bev[i & 1], 0));
+    CHK(c, hipMemcpyAsync(out_blocks[i], obuf, nvox * esz,
+                          hipMemcpyDeviceToHost, c->copy_stream));
+    CHK(c, hipEventRecord(bev[i & 1], c->copy_stream));
   }
   CHK(c, hipStreamSynchronize(c->stream));
+  CHK(c, hipStreamSynchronize(c->copy_stream));
+  (void)hipEventDestroy(bev[0]);
+  (void)hipEventDestroy(bev[1]);
   c->stats.blocks += (long long)nb;
   flush_stats(c);
   return BS_OK;
