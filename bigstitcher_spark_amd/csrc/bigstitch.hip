@@ -392,6 +392,93 @@ __global__ __launch_bounds__(LPB_S *TPL_S) void k_fft_pass(
   }
 }
 
+template <int ES, int TPL>
+__device__ __forceinline__ void fft_lds_r8(f2 *data, long base, int n,
+                                        int log2n, int tl, const f2 *tw,
+                                        int dir) {
+#define D_(e) data[base + (long)(e) * ES]
+#define BF_(u, v, w, lo, hi)                                                \
+  {                                                                         \
+    f2 t_ = cmul(v, w);                                                     \
+    lo = {u.x + t_.x, u.y + t_.y};                                          \
+    hi = {u.x - t_.x, u.y - t_.y};                                          \
+  }
+  int h = 1;
+  const int lead = log2n % 3;
+  if (lead == 1) {
+    for (int bf = tl; bf < (n >> 1); bf += TPL) {
+      f2 u = D_(2 * bf), v = D_(2 * bf + 1);
+      D_(2 * bf) = {u.x + v.x, u.y + v.y};
+      D_(2 * bf + 1) = {u.x - v.x, u.y - v.y};
+    }
+    h = 2;
+    __syncthreads();
+  } else if (lead == 2) {
+    const int s2 = n >> 2;
+    for (int g = tl; g < (n >> 2); g += TPL) {
+      int i = g * 4;
+      f2 a = D_(i), b = D_(i + 1), c = D_(i + 2), d = D_(i + 3);
+      f2 A = {a.x + b.x, a.y + b.y}, B = {a.x - b.x, a.y - b.y};
+      f2 Cc = {c.x + d.x, c.y + d.y}, Dd = {c.x - d.x, c.y - d.y};
+      f2 w2b = tw[s2];
+      if (dir < 0) w2b.y = -w2b.y;
+      f2 u2 = cmul(Dd, w2b);
+      D_(i) = {A.x + Cc.x, A.y + Cc.y};
+      D_(i + 2) = {A.x - Cc.x, A.y - Cc.y};
+      D_(i + 1) = {B.x + u2.x, B.y + u2.y};
+      D_(i + 3) = {B.x - u2.x, B.y - u2.y};
+    }
+    h = 4;
+    __syncthreads();
+  }
+  for (; h < n; h <<= 3) {
+    const int q = n >> 3;
+    const int s1 = n / (2 * h), s2 = n / (4 * h), s3 = n / (8 * h);
+    for (int g = tl; g < q; g += TPL) {
+      int off = g % h, blk = g / h;
+      int i = blk * 8 * h + off;
+      f2 a0 = D_(i), a1 = D_(i + h), a2 = D_(i + 2 * h), a3 = D_(i + 3 * h);
+      f2 a4 = D_(i + 4 * h), a5 = D_(i + 5 * h), a6 = D_(i + 6 * h),
+         a7 = D_(i + 7 * h);
+      f2 w1 = tw[off * s1];
+      f2 w2a = tw[off * s2], w2b = tw[(off + h) * s2];
+      f2 w3a = tw[off * s3], w3b = tw[(off + h) * s3],
+         w3c = tw[(off + 2 * h) * s3], w3d = tw[(off + 3 * h) * s3];
+      if (dir < 0) {
+        w1.y = -w1.y; w2a.y = -w2a.y; w2b.y = -w2b.y;
+        w3a.y = -w3a.y; w3b.y = -w3b.y; w3c.y = -w3c.y; w3d.y = -w3d.y;
+      }
+      f2 b0, b1, b2, b3, b4, b5, b6, b7;
+      BF_(a0, a1, w1, b0, b1);
+      BF_(a2, a3, w1, b2, b3);
+      BF_(a4, a5, w1, b4, b5);
+      BF_(a6, a7, w1, b6, b7);
+      f2 c0, c1, c2, c3, c4, c5, c6, c7;
+      BF_(b0, b2, w2a, c0, c2);
+      BF_(b1, b3, w2b, c1, c3);
+      BF_(b4, b6, w2a, c4, c6);
+      BF_(b5, b7, w2b, c5, c7);
+      f2 d0, d1, d2, d3, d4, d5, d6, d7;
+      BF_(c0, c4, w3a, d0, d4);
+      BF_(c1, c5, w3b, d1, d5);
+      BF_(c2, c6, w3c, d2, d6);
+      BF_(c3, c7, w3d, d3, d7);
+      D_(i) = d0;
+      D_(i + h) = d1;
+      D_(i + 2 * h) = d2;
+      D_(i + 3 * h) = d3;
+      D_(i + 4 * h) = d4;
+      D_(i + 5 * h) = d5;
+      D_(i + 6 * h) = d6;
+      D_(i + 7 * h) = d7;
+    }
+    __syncthreads();
+  }
+#undef BF_
+#undef D_
+}
+
+
 /* ---- glds (async global->LDS DMA) double-buffered strided pass ---- */
 
 typedef __attribute__((address_space(3))) unsigned bs_lds_u32;
@@ -566,7 +653,7 @@ __global__ __launch_bounds__(LPB_S *TPL_S) void k_fft_pass_glds(
  * same buffer for the inverse DIT. NE = elements per thread
  * (compile-time so the staging arrays live in REGISTERS, not scratch);
  * n <= 128*NE, max 1024. */
-template <int NE, bool PB>
+template <int NE, bool PB, bool R8 = false>
 __device__ __forceinline__ void zf_body(
     f2 *a, f2 *b, int n, int log2n, long estride, long gstride, int nlines,
     int nchunks, int ngroups, int valid_a, int valid_b, float scale,
@@ -624,7 +711,10 @@ __device__ __forceinline__ void zf_body(
       }
     }
     __syncthreads();
-    fft_lds<LPB_S, TPL_S>(da, (long)line, n, log2n, tl, tw, +1);
+    if (R8)
+      fft_lds_r8<LPB_S, TPL_S>(da, (long)line, n, log2n, tl, tw, +1);
+    else
+      fft_lds<LPB_S, TPL_S>(da, (long)line, n, log2n, tl, tw, +1);
     { /* park A, refill with bit-reversed B, FFT B */
 #pragma unroll
       for (int k = 0; k < NE; ++k) {
@@ -666,7 +756,10 @@ __device__ __forceinline__ void zf_body(
         }
       }
       __syncthreads();
-      fft_lds<LPB_S, TPL_S>(da, (long)line, n, log2n, tl, tw, +1);
+      if (R8)
+        fft_lds_r8<LPB_S, TPL_S>(da, (long)line, n, log2n, tl, tw, +1);
+      else
+        fft_lds<LPB_S, TPL_S>(da, (long)line, n, log2n, tl, tw, +1);
     }
     { /* cross-power normalise [PIN-EPS] into regs, rewrite bit-reversed */
 #pragma unroll
@@ -701,7 +794,10 @@ __device__ __forceinline__ void zf_body(
       }
       __syncthreads();
     }
-    fft_lds<LPB_S, TPL_S>(da, (long)line, n, log2n, tl, tw, -1);
+    if (R8)
+      fft_lds_r8<LPB_S, TPL_S>(da, (long)line, n, log2n, tl, tw, -1);
+    else
+      fft_lds<LPB_S, TPL_S>(da, (long)line, n, log2n, tl, tw, -1);
     if (pair_ok) {
 #pragma unroll
       for (int k = 0; k < NE; ++k) {
@@ -734,6 +830,15 @@ __global__ __launch_bounds__(LPB_S *TPL_S) void k_fft_z_fused(
     const f2 *twg) {
   zf_body<NE, true>(a, b, n, log2n, estride, gstride, nlines, nchunks,
                     ngroups, valid_a, valid_b, scale, twg);
+}
+
+template <int NE>
+__global__ __launch_bounds__(LPB_S *TPL_S) void k_fft_z_fused_r8(
+    f2 *a, f2 *b, int n, int log2n, long estride, long gstride, int nlines,
+    int nchunks, int ngroups, int valid_a, int valid_b, float scale,
+    const f2 *twg) {
+  zf_body<NE, false, true>(a, b, n, log2n, estride, gstride, nlines,
+                           nchunks, ngroups, valid_a, valid_b, scale, twg);
 }
 
 /* no-register-prefetch variant, VGPR-capped for 4 blocks/CU (34 KB LDS
@@ -2707,14 +2812,15 @@ static int stitch_phaseA(bs_ctx *c, bs_slot *sl, const bs_pair_desc &pd,
     /* n>=512: the occupancy-capped no-prefetch variant wins (A/B on
      * hardware: 0.734 vs 0.842 ms at 512^3 — 4 blocks/CU beats the
      * B-register prefetch at 2 blocks/CU) */
+    static const bool z_r8 = getenv("BS_Z_R8") != nullptr;
     if (Pz <= 128)
       zf(k_fft_z_fused<1>);
     else if (Pz == 256)
       zf(k_fft_z_fused<2>);
     else if (Pz == 512)
-      zf(k_fft_z_fused_np<4>);
+      z_r8 ? zf(k_fft_z_fused_r8<4>) : zf(k_fft_z_fused_np<4>);
     else
-      zf(k_fft_z_fused_np<8>);
+      z_r8 ? zf(k_fft_z_fused_r8<8>) : zf(k_fft_z_fused_np<8>);
   }
   {
     size_t lds = ((Py / 2) + (size_t)LPB_S * Py) * sizeof(f2);
