@@ -9,6 +9,7 @@
  * AVG/AVG_BLEND/MAX_INTENSITY; --masks writes coverage masks instead of
  * fused intensities (reference SparkAffineFusion.java:112-115, :565-578
  * via fusion/GenerateComputeBlockMasks.java). */
+#include <algorithm>
 #include <atomic>
 #include <cstdio>
 #include <cstring>
@@ -92,6 +93,8 @@ int main(int argc, char **argv) {
       {"-f", "--fusion"}, /* reference spelling, SparkAffineFusion.java:124 */
       {"--fusionType", "--fusion"}, /* pre-round-2 spelling */
       {"-vi", "--vi"},
+      {"-t", "--timepointIndex"},
+      {"-c", "--channelIndex"},
       {"-s", "--storage"}};
   if (!args.parse(argc, argv, alias,
                   {"masks", "prefetch", "dryRun",
@@ -243,16 +246,35 @@ int main(int argc, char **argv) {
     return 1;
   }
   int num_tp = ntp_a ? (int)ntp_a->inum : (int)sd.timepoints.size();
-
+  auto nch_a = geta("NumChannels");
+  int num_ch = nch_a ? (int)nch_a->inum : 1;
+  /* distinct channel attribute ids of the dataset (sorted) — output
+   * channel c takes the views whose channel attr equals the c-th id
+   * (reference: -c/--channelIndex + view selection decide the content;
+   * with one channel everything passes) */
+  std::vector<int> chan_ids;
+  for (auto &s2 : sd.setups)
+    if (std::find(chan_ids.begin(), chan_ids.end(), s2.channel) ==
+        chan_ids.end())
+      chan_ids.push_back(s2.channel);
+  std::sort(chan_ids.begin(), chan_ids.end());
+  const long tpIdx = args.getl("timepointIndex", -1);
+  const long chIdx = args.getl("channelIndex", -1);
   for (int ti = 0; ti < num_tp; ++ti) {
+   if (tpIdx >= 0 && ti != tpIdx) continue;
+   for (int ci = 0; ci < num_ch; ++ci) {
+    if (chIdx >= 0 && ci != chIdx) continue;
+    const int want_chan =
+        num_ch > 1 ? chan_ids[ci % chan_ids.size()] : -2 /* any */;
     int tp = sd.timepoints[ti % sd.timepoints.size()];
-    /* upload + describe views of this timepoint */
+    /* upload + describe views of this timepoint (+channel) */
     std::vector<bs_fuse_view> fviews;
     std::vector<const bssd::ViewSetup *> fsetups;
     for (auto &s : sd.setups) {
       auto r = sd.regs.find({tp, s.id});
       if (r == sd.regs.end()) continue;
       if (!selset.count({tp, s.id})) continue;
+      if (want_chan != -2 && s.channel != want_chan) continue;
       bs_fuse_view fv{};
       fv.view_id = s.id;
       /* world coords; bs_fuse_volume shifts by -vol_min itself.
@@ -324,9 +346,11 @@ int main(int argc, char **argv) {
       fviews.push_back(fv);
       fsetups.push_back(&s);
     }
-    /* levels from the container's MultiResolutionInfos */
+    /* levels from the container's MultiResolutionInfos (one entry
+     * list per output volume, t-major then channel — the container's
+     * write order) */
     auto mri = geta("MultiResolutionInfos");
-    int vol_idx = ti; /* ch0 only this round */
+    int vol_idx = ti * num_ch + ci;
     if (!mri || (size_t)vol_idx >= mri->arr.size()) {
       fprintf(stderr, "missing MultiResolutionInfos for volume %d\n",
               vol_idx);
@@ -452,6 +476,7 @@ int main(int argc, char **argv) {
              dsnames[l].c_str(), nchunks, lx, ly, lz);
     }
     for (auto &fv : fviews) bs_view_release(ctx, fv.view_id);
+   }
   }
   bs_ctx_destroy(ctx);
   printf("affine-fusion done.\n");

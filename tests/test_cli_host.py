@@ -1014,3 +1014,36 @@ def test_n5_block_roundtrip_property(tmp_path):
         back, attrs = n5util.read_dataset(root, "ds")
         assert np.array_equal(back, vol), (dims, blk, codec)
         assert attrs["compression"]["type"] == codec
+
+
+@pytest.mark.gpu
+def test_cli_fusion_two_channels(tmp_path):
+    """NumChannels=2 container: each output channel volume gets the
+    views whose channel attribute matches (t-major vol ordering in
+    MultiResolutionInfos); -c/--channelIndex restricts to one."""
+    xml, n5, vols, err = make_grouped_dataset(str(tmp_path), nch=2)
+    out = os.path.join(str(tmp_path), "fused.n5")
+    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml,
+             "-s", "N5", "-o", out, "--blockSize", "32,32,32",
+             "-d", "FLOAT32", "-ch", "2"])
+    assert r.returncode == 0, r.stderr
+    attrs = n5util.root_attrs(out)["Bigstitcher-Spark"]
+    assert attrs["NumChannels"] == 2
+    r = run([os.path.join(BIN, "affine-fusion"), "-o", out, "-f", "AVG",
+             "--blendingRange", "0"])
+    assert r.returncode == 0, r.stderr + r.stdout
+    bbmin = attrs["Boundingbox_min"]
+    ident = np.hstack([np.eye(3), np.zeros((3, 1))])
+    affB = np.hstack([np.eye(3), np.array([[40.0], [0.0], [0.0]])])
+    # setups: 0=tileA ch0, 1=tileA ch1, 2=tileB ch0, 3=tileB ch1
+    for ci, (sa, sb) in enumerate([(0, 2), (1, 3)]):
+        fused, _ = n5util.read_dataset(out, f"ch{ci}tp0/s0")
+        views = [dict(data=vols[sa], affine=ident, border=(0, 0, 0),
+                      range=(0, 0, 0)),
+                 dict(data=vols[sb], affine=affB, border=(0, 0, 0),
+                      range=(0, 0, 0))]
+        ref = of.fuse_block(views, tuple(bbmin), fused.shape[::-1],
+                            of.FUSION_AVG)
+        denom = np.maximum(np.abs(ref), 1.0)
+        assert np.max(np.abs(fused.astype(np.float64) - ref) /
+                      denom) < 1e-4, f"ch{ci}"
