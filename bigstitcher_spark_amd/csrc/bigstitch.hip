@@ -3408,17 +3408,231 @@ extern "C" int bs_fuse_volume(bs_ctx *c, const bs_fuse_view *views,
     lbytes[l] = (size_t)ldims[l][0] * ldims[l][1] * ldims[l][2] * esz;
     total += (lbytes[l] + 255) & ~(size_t)255;
   }
+  /* fit check: when level 0 exceeds the device budget, fall to the
+   * Z-SLAB path below — level 0 lives in a double-buffered slab arena
+   * while the (8x+ smaller) levels >= 1 stay fully resident, each slab
+   * D2H overlapping the next slab's fusion. Budget = free HBM - margin
+   * (BS_FUSE_BUDGET_MB overrides, used by the parity test). */
+  size_t budget;
+  {
+    size_t freeb = 0, totb = 0;
+    (void)hipMemGetInfo(&freeb, &totb);
+    budget = freeb > (10UL << 30) ? freeb - (6UL << 30) : freeb / 2;
+    if (const char *e = getenv("BS_FUSE_BUDGET_MB"))
+      budget = (size_t)atoll(e) << 20;
+  }
+  long slab_z = ldims[0][2];
+  const size_t plane0 = (size_t)ldims[0][0] * ldims[0][1] * esz;
+  size_t rest = 0;
+  for (int l = 1; l < nlevels; ++l) rest += (lbytes[l] + 255) & ~(size_t)255;
+  if (total > budget) {
+    /* slab multiple of the fusion block z AND every level's absolute z
+     * factor so per-slab pyramid ranges stay aligned */
+    long step = 128;
+    for (int l = 0; l < nlevels; ++l) {
+      long f = abs_ds[l * 3 + 2];
+      while (step % f) step += 128;
+    }
+    long maxz = (long)((budget > rest ? (budget - rest) / 2 : 0) /
+                       std::max((size_t)1, plane0));
+    slab_z = std::max(step, (maxz / step) * step);
+    if (2 * plane0 * slab_z + rest > budget && slab_z == step) {
+      c->err = "device volume alloc failed (even one slab exceeds HBM)";
+      return BS_ENOMEM;
+    }
+    total = 2 * ((plane0 * slab_z + 255) & ~(size_t)255) + rest;
+  }
   rc = ensure_dev(c, &c->dvol_arena, &c->dvol_cap, total);
   if (rc) {
     c->err = "device volume alloc failed (volume too large this round)";
     return rc;
   }
+  const bool slabbed = slab_z < ldims[0][2];
+  void *slabbuf[2] = {nullptr, nullptr};
   {
     size_t off = 0;
-    for (int l = 0; l < nlevels; ++l) {
-      dlvl[l] = (char *)c->dvol_arena + off;
-      off += (lbytes[l] + 255) & ~(size_t)255;
+    if (slabbed) {
+      size_t sb = (plane0 * slab_z + 255) & ~(size_t)255;
+      slabbuf[0] = (char *)c->dvol_arena;
+      slabbuf[1] = (char *)c->dvol_arena + sb;
+      dlvl[0] = nullptr; /* level 0 never whole on device */
+      off = 2 * sb;
+      for (int l = 1; l < nlevels; ++l) {
+        dlvl[l] = (char *)c->dvol_arena + off;
+        off += (lbytes[l] + 255) & ~(size_t)255;
+      }
+    } else {
+      for (int l = 0; l < nlevels; ++l) {
+        dlvl[l] = (char *)c->dvol_arena + off;
+        off += (lbytes[l] + 255) & ~(size_t)255;
+      }
     }
+  }
+  if (slabbed) {
+    /* -------- z-slab path: output bigger than the HBM budget --------
+     * Level 0 is produced slab by slab into a double-buffered arena;
+     * each slab's pyramid contribution lands in the fully-resident
+     * level>=1 buffers (slab_z is a multiple of every level's absolute
+     * z factor, so ranges stay aligned); the PREVIOUS slab's D2H runs
+     * while the current slab fuses. */
+    const long FBs[3] = {256, 128, 128};
+    const long vrow0 = ldims[0][0];
+    const long vslice0 = ldims[0][0] * ldims[0][1];
+    hipEvent_t sev[2] = {nullptr, nullptr};
+    CHK(c, hipEventCreateWithFlags(&sev[0], hipEventDisableTiming));
+    CHK(c, hipEventCreateWithFlags(&sev[1], hipEventDisableTiming));
+    long nslabs = (ldims[0][2] + slab_z - 1) / slab_z;
+    long long nblocks = 0;
+    for (long si = 0; si < nslabs; ++si) {
+      const int sbi = (int)(si & 1);
+      const long zs = si * slab_z;
+      const long ze = std::min(ldims[0][2], zs + slab_z);
+      /* enqueue this slab's fusion blocks */
+      std::vector<int32_t> flat;
+      std::vector<std::array<long, 6>> fbl;
+      std::vector<int64_t> offs2;
+      for (long z0 = zs; z0 < ze; z0 += FBs[2])
+        for (long y0 = 0; y0 < ldims[0][1]; y0 += FBs[1])
+          for (long x0 = 0; x0 < ldims[0][0]; x0 += FBs[0]) {
+            long bmin[3] = {x0, y0, z0};
+            long bsz[3] = {std::min(FBs[0], ldims[0][0] - x0),
+                           std::min(FBs[1], ldims[0][1] - y0),
+                           std::min(FBs[2], ze - z0)};
+            offs2.push_back((int64_t)flat.size());
+            for (size_t v = 0; v < nviews; ++v)
+              if (view_overlaps_block(views[v], dv[v], vol_min, bmin, bsz))
+                flat.push_back((int32_t)v);
+            fbl.push_back(
+                {bmin[0], bmin[1], bmin[2], bsz[0], bsz[1], bsz[2]});
+          }
+      offs2.push_back((int64_t)flat.size());
+      rc = ensure_dev(c, (void **)&c->dvidx, &c->dvidx_cap,
+                      std::max((size_t)1, flat.size()) * sizeof(int32_t));
+      if (rc) return rc;
+      if (!flat.empty())
+        CHK(c, hipMemcpyAsync(c->dvidx, flat.data(),
+                              flat.size() * sizeof(int32_t),
+                              hipMemcpyHostToDevice, c->stream));
+      for (size_t b = 0; b < fbl.size(); ++b) {
+        auto &fb = fbl[b];
+        int nvb = (int)(offs2[b + 1] - offs2[b]);
+        long nrows_f = fb[4] * fb[5];
+        long gfb = std::max(8L, std::min(4096L, nrows_f) & ~7L);
+        bs_tim tt(c, BS_K_FUSE, c->stream);
+        const long oo =
+            (fb[2] - zs) * vslice0 + fb[1] * vrow0 + fb[0];
+        if (prm->masks)
+          hipLaunchKernelGGL(k_mask, dim3(gfb), dim3(256), 0, c->stream,
+                             c->dviews, c->dvidx + offs2[b], nvb, fb[0],
+                             fb[1], fb[2], (int)fb[3], (int)fb[4],
+                             (int)fb[5], prm->out_dtype,
+                             (float)prm->mask_offset[0],
+                             (float)prm->mask_offset[1],
+                             (float)prm->mask_offset[2], slabbuf[sbi], oo,
+                             vrow0, vslice0);
+        else
+          hipLaunchKernelGGL(k_fuse, dim3(gfb), dim3(256), 0, c->stream,
+                             c->dviews, c->dvidx + offs2[b], nvb, fb[0],
+                             fb[1], fb[2], (int)fb[3], (int)fb[4],
+                             (int)fb[5], prm->fusion_type, prm->out_dtype,
+                             (float)prm->min_intensity, invRange,
+                             slabbuf[sbi], oo, vrow0, vslice0,
+                             getenv("BS_FUSE_NORIV") ? 0 : 1);
+      }
+      nblocks += (long long)fbl.size();
+      /* this slab's pyramid contributions (level l from level l-1) */
+      for (int l = 1; l < nlevels; ++l) {
+        const long az_prev = abs_ds[(l - 1) * 3 + 2];
+        const long az = abs_ds[l * 3 + 2];
+        int rx = abs_ds[l * 3 + 0] / abs_ds[(l - 1) * 3 + 0];
+        int ry = abs_ds[l * 3 + 1] / abs_ds[(l - 1) * 3 + 1];
+        int rz = (int)(az / az_prev);
+        const long zprev0 = zs / az_prev;
+        const long zprev1 = std::min(
+            ldims[l - 1][2], (ze + az_prev - 1) / az_prev);
+        const long zl0 = zs / az;
+        const long zl1 = std::min(ldims[l][2], (ze + az - 1) / az);
+        const void *src =
+            l == 1 ? slabbuf[sbi]
+                   : (const void *)((const char *)dlvl[l - 1] +
+                                    (size_t)zprev0 * ldims[l - 1][0] *
+                                        ldims[l - 1][1] * esz);
+        void *dst = (char *)dlvl[l] +
+                    (size_t)zl0 * ldims[l][0] * ldims[l][1] * esz;
+        long nrows = ldims[l][1] * (zl1 - zl0);
+        if (nrows <= 0) continue;
+        bs_tim tt(c, BS_K_PYRAMID, c->stream);
+        if (esz == 4)
+          hipLaunchKernelGGL(k_pyr<float>, dim3(std::min(4096L, nrows)),
+                             dim3(256), 0, c->stream, (const float *)src,
+                             (float *)dst, (int)ldims[l - 1][0],
+                             (int)ldims[l - 1][1], (int)(zprev1 - zprev0),
+                             (int)ldims[l][0], (int)ldims[l][1],
+                             (int)(zl1 - zl0), rx, ry, rz);
+        else if (esz == 2)
+          hipLaunchKernelGGL(k_pyr<unsigned short>,
+                             dim3(std::min(4096L, nrows)), dim3(256), 0,
+                             c->stream, (const unsigned short *)src,
+                             (unsigned short *)dst, (int)ldims[l - 1][0],
+                             (int)ldims[l - 1][1], (int)(zprev1 - zprev0),
+                             (int)ldims[l][0], (int)ldims[l][1],
+                             (int)(zl1 - zl0), rx, ry, rz);
+        else
+          hipLaunchKernelGGL(k_pyr<unsigned char>,
+                             dim3(std::min(4096L, nrows)), dim3(256), 0,
+                             c->stream, (const unsigned char *)src,
+                             (unsigned char *)dst, (int)ldims[l - 1][0],
+                             (int)ldims[l - 1][1], (int)(zprev1 - zprev0),
+                             (int)ldims[l][0], (int)ldims[l][1],
+                             (int)(zl1 - zl0), rx, ry, rz);
+      }
+      CHK(c, hipEventRecord(sev[sbi], c->stream));
+      /* drain the PREVIOUS slab while this one computes */
+      if (si > 0) {
+        const int pbi = (int)((si - 1) & 1);
+        const long pzs = (si - 1) * slab_z;
+        const long pze = std::min(ldims[0][2], pzs + slab_z);
+        bool gated = false;
+        auto gate = [&](size_t, size_t) {
+          if (!gated) {
+            (void)hipStreamWaitEvent(c->copy_stream, sev[pbi], 0);
+            gated = true;
+          }
+        };
+        rc = staged_d2h(c, slabbuf[pbi],
+                        (char *)level_buffers[0] + (size_t)pzs * plane0,
+                        (size_t)(pze - pzs) * plane0, c->copy_stream,
+                        gate);
+        if (rc) return rc;
+      }
+    }
+    { /* last slab */
+      const int pbi = (int)((nslabs - 1) & 1);
+      const long pzs = (nslabs - 1) * slab_z;
+      const long pze = ldims[0][2];
+      bool gated = false;
+      auto gate = [&](size_t, size_t) {
+        if (!gated) {
+          (void)hipStreamWaitEvent(c->copy_stream, sev[pbi], 0);
+          gated = true;
+        }
+      };
+      rc = staged_d2h(c, slabbuf[pbi],
+                      (char *)level_buffers[0] + (size_t)pzs * plane0,
+                      (size_t)(pze - pzs) * plane0, c->copy_stream, gate);
+      if (rc) return rc;
+    }
+    for (int l = 1; l < nlevels; ++l) {
+      rc = staged_d2h(c, dlvl[l], level_buffers[l], lbytes[l], c->stream);
+      if (rc) return rc;
+    }
+    CHK(c, hipStreamSynchronize(c->stream));
+    CHK(c, hipStreamSynchronize(c->copy_stream));
+    (void)hipEventDestroy(sev[0]);
+    (void)hipEventDestroy(sev[1]);
+    c->stats.blocks += nblocks;
+    flush_stats(c);
+    return BS_OK;
   }
   /* fusion of level 0 over an internal 256x128x128 grid with culling */
   const long FB[3] = {256, 128, 128};

@@ -320,3 +320,43 @@ def test_fuse_block_no_covering_views():
                   range=(0, 0, 0))],
             (100, 100, 100), (8, 8, 8), of.FUSION_AVG_BLEND)
         assert np.array_equal(out, ref.astype(np.float32))
+
+
+@pytest.mark.gpu
+def test_fuse_volume_slab_path_parity():
+    """Outputs bigger than the HBM budget fuse in z-slabs (double-
+    buffered, D2H overlapped). Forcing a tiny budget via
+    BS_FUSE_BUDGET_MB must give BIT-IDENTICAL volumes and pyramid
+    levels to the resident path (slab boundaries are multiples of
+    every level's z factor)."""
+    import os
+    from bigstitcher_spark_amd import Context
+    import numpy as np
+    from oracle import synth
+    shape = (200, 64, 64)  # (nz, ny, nx)
+    rng = np.random.default_rng(31)
+    with Context(0) as ctx:
+        vols = []
+        for i in range(2):
+            v = rng.integers(0, 60000, size=shape).astype(np.uint16)
+            vols.append(v)
+            ctx.upload(70 + i, v)
+        views = [
+            dict(view_id=70, affine=np.hstack([np.eye(3),
+                 np.zeros((3, 1))]), border=(0, 0, 0), range=(9, 9, 9)),
+            dict(view_id=71, affine=np.hstack([np.eye(3),
+                 np.array([[0.0], [0.0], [180.0]])]),
+                 border=(0, 0, 0), range=(9, 9, 9)),
+        ]
+        kw = dict(downsamplings=[(1, 1, 1), (2, 2, 2)],
+                  fusion_type=1, out_dtype=np.float32,
+                  min_intensity=0, max_intensity=65535)
+        dims = (64, 64, 380)
+        ref = ctx.fuse_volume(views, (0, 0, 0), dims, **kw)
+        os.environ["BS_FUSE_BUDGET_MB"] = "6"
+        try:
+            got = ctx.fuse_volume(views, (0, 0, 0), dims, **kw)
+        finally:
+            del os.environ["BS_FUSE_BUDGET_MB"]
+        for l in range(2):
+            assert np.array_equal(got[l], ref[l]), f"level {l}"
