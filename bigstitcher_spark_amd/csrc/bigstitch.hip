@@ -2397,6 +2397,18 @@ extern "C" int bs_reset_stats(bs_ctx *c) {
 
 /* ---- views ---- */
 
+extern "C" int bs_device_mem(bs_ctx *c, uint64_t *free_bytes,
+                             uint64_t *total_bytes) {
+  if (!c || !free_bytes || !total_bytes) return BS_EINVAL;
+  std::lock_guard<std::mutex> g(c->mu);
+  CHK(c, hipSetDevice(c->dev));
+  size_t f = 0, t = 0;
+  CHK(c, hipMemGetInfo(&f, &t));
+  *free_bytes = f;
+  *total_bytes = t;
+  return BS_OK;
+}
+
 extern "C" int bs_view_upload(bs_ctx *c, int32_t id, const uint16_t *data,
                               const int64_t dims[3]) {
   if (!c || !data || !dims) return BS_EINVAL;

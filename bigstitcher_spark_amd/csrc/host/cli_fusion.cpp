@@ -267,15 +267,25 @@ int main(int argc, char **argv) {
     const int want_chan =
         num_ch > 1 ? chan_ids[ci % chan_ids.size()] : -2 /* any */;
     int tp = sd.timepoints[ti % sd.timepoints.size()];
-    /* upload + describe views of this timepoint (+channel) */
-    std::vector<bs_fuse_view> fviews;
-    std::vector<const bssd::ViewSetup *> fsetups;
+    /* ---- PLAN: per selected view, pick the pyramid level, build the
+     * adjusted affine and its container-frame z extent — no voxel
+     * reads yet (the z-band loop below uploads a sliding window) */
+    struct VPlan {
+      bs_fuse_view fv;
+      int setup = 0, tp = 0, level = 0;
+      double zlo = 0, zhi = 0; /* container-frame z extent (+affine) */
+      std::vector<long long> ldims3;
+    };
+    std::vector<VPlan> plans;
     for (auto &s : sd.setups) {
       auto r = sd.regs.find({tp, s.id});
       if (r == sd.regs.end()) continue;
       if (!selset.count({tp, s.id})) continue;
       if (want_chan != -2 && s.channel != want_chan) continue;
-      bs_fuse_view fv{};
+      VPlan pl;
+      pl.setup = s.id;
+      pl.tp = tp;
+      bs_fuse_view &fv = pl.fv;
       fv.view_id = s.id;
       /* world coords; bs_fuse_volume shifts by -vol_min itself.
        * [PIN-ANISO]: world z divided by the anisotropy factor
@@ -283,11 +293,8 @@ int main(int argc, char **argv) {
       for (int i = 0; i < 12; ++i) fv.affine[i] = r->second[i];
       if (anisoF != 1.0)
         for (int i = 8; i < 12; ++i) fv.affine[i] /= anisoF;
-      /* multi-resolution input: pick the coarsest level whose sampling
-       * step under the adjusted transform stays <= ~1 px (ViewUtil's
-       * forBestResolution rule, bs_cli_util.h) and fold the [PIN-MIP]
-       * mipmap transform (x0 = f*xl + (f-1)/2) into the view affine;
-       * blending borders/ranges shrink into the level grid. */
+      /* multi-resolution input: ViewUtil forBestResolution rule +
+       * [PIN-MIP] mipmap fold (see bs_mip.h / bs_cli_util.h) */
       auto levels = bscli::read_levels(in_n5, s.id, tp);
       bscli::M34 model;
       for (int i = 0; i < 12; ++i) model[i] = fv.affine[i];
@@ -306,49 +313,34 @@ int main(int argc, char **argv) {
           }
           fv.affine[rr * 4 + 3] += off;
         }
-      }
-      std::vector<uint16_t> vox;
-      std::vector<long long> vdims;
-      if (!in_n5.read_volume_u16(
-              bssd::SpimData::image_dataset(
-                  s.id, tp, lvi < (int)levels.size() && lvi > 0
-                                ? levels[lvi].level
-                                : 0),
-              &vox, &vdims)) {
-        fprintf(stderr, "cannot read view tp=%d setup=%d\n", tp, s.id);
-        return 1;
-      }
-      int64_t d[3] = {vdims[0], vdims[1], vdims[2]};
-      if (bs_view_upload(ctx, s.id, vox.data(), d) != BS_OK) {
-        fprintf(stderr, "upload failed: %s\n", bs_last_error(ctx));
-        return 1;
+        pl.level = levels[lvi].level;
       }
       for (int d2 = 0; d2 < 3; ++d2) {
         fv.blend_border[d2] = bborder / (float)lf[d2];
         fv.blend_range[d2] = brange / (float)lf[d2];
       }
-      if (args.has("intensityN5Path")) {
-        bsn5::Container coeff_n5(args.get("intensityN5Path"));
-        std::vector<float> ab;
-        int32_t gd[3];
-        if (load_coefficients(coeff_n5, args.get("intensityN5Group", ""),
-                              args.get("intensityN5Dataset", "intensity"),
-                              s.id, tp, &ab, gd)) {
-          if (bs_view_set_coefficients(ctx, s.id, ab.data(), gd) != BS_OK) {
-            fprintf(stderr, "set_coefficients failed: %s\n",
-                    bs_last_error(ctx));
-            return 1;
-          }
-          printf("loaded intensity coefficients for setup %d (%dx%dx%d)\n",
-                 s.id, gd[0], gd[1], gd[2]);
-        }
+      /* dims of the chosen level (from metadata; fallback: ceil) */
+      if (pl.level < (int)levels.size() &&
+          levels[pl.level].dims.size() == 3) {
+        pl.ldims3 = levels[pl.level].dims;
+      } else {
+        pl.ldims3 = {(s.dims[0] + lf[0] - 1) / lf[0],
+                     (s.dims[1] + lf[1] - 1) / lf[1],
+                     (s.dims[2] + lf[2] - 1) / lf[2]};
       }
-      fviews.push_back(fv);
-      fsetups.push_back(&s);
+      {
+        bscli::M34 m2;
+        for (int i2 = 0; i2 < 12; ++i2) m2[i2] = fv.affine[i2];
+        double lo[3], hi[3];
+        long long ld3[3] = {pl.ldims3[0], pl.ldims3[1], pl.ldims3[2]};
+        bscli::tbbox(m2, ld3, lo, hi);
+        pl.zlo = lo[2] - (double)bbmin[2];
+        pl.zhi = hi[2] - (double)bbmin[2];
+      }
+      plans.push_back(pl);
     }
     /* levels from the container's MultiResolutionInfos (one entry
-     * list per output volume, t-major then channel — the container's
-     * write order) */
+     * list per output volume, t-major then channel) */
     auto mri = geta("MultiResolutionInfos");
     int vol_idx = ti * num_ch + ci;
     if (!mri || (size_t)vol_idx >= mri->arr.size()) {
@@ -367,6 +359,31 @@ int main(int argc, char **argv) {
       for (int d = 0; d < 3; ++d)
         abs_ds[l * 3 + d] = (int32_t)ad->arr[d]->inum;
     }
+    /* per-level container attrs (chunk sizes) up front */
+    std::vector<bsn5::DatasetAttrs> das(nlevels);
+    std::vector<bszarr::ArrayAttrs> zas(nlevels);
+    int cbx = 128, cby = 128, cbz = 128;
+    for (int l = 0; l < nlevels; ++l) {
+      if (zarr) {
+        if (!zr.get_array_attrs(dsnames[l], &zas[l])) {
+          fprintf(stderr, "missing array %s in container\n",
+                  dsnames[l].c_str());
+          return 1;
+        }
+        cbx = zas[l].chunks[4];
+        cby = zas[l].chunks[3];
+        cbz = zas[l].chunks[2];
+      } else {
+        if (!n5.get_dataset_attrs(dsnames[l], &das[l])) {
+          fprintf(stderr, "missing dataset %s in container\n",
+                  dsnames[l].c_str());
+          return 1;
+        }
+        cbx = das[l].block[0];
+        cby = das[l].block[1];
+        cbz = das[l].block[2];
+      }
+    }
     bs_fuse_params prm{};
     prm.fusion_type = fusion_type;
     prm.out_dtype = out_dtype;
@@ -381,101 +398,199 @@ int main(int argc, char **argv) {
     }
     size_t esz2 = out_dtype == BS_OUT_FLOAT32 ? 4
                   : out_dtype == BS_OUT_UINT16 ? 2 : 1;
-    std::vector<std::vector<char>> hostlvl(nlevels);
-    std::vector<void *> lvlptr(nlevels);
-    std::vector<int64_t> ldims(3 * nlevels);
+    /* ---- z-band sizing: one band when everything fits; otherwise the
+     * output is produced in chunk-aligned z bands with a SLIDING view
+     * window (upload views intersecting the band, release ones fully
+     * above it) — datasets whose views exceed HBM fuse on one GPU */
+    long bandz = dims[2];
+    long band_step = cbz;
     for (int l = 0; l < nlevels; ++l) {
-      long long b = esz2;
-      for (int d = 0; d < 3; ++d)
-        b *= (dims[d] + abs_ds[l * 3 + d] - 1) / abs_ds[l * 3 + d];
-      hostlvl[l].resize((size_t)b);
-      lvlptr[l] = hostlvl[l].data();
+      long f = abs_ds[l * 3 + 2];
+      while (band_step % (f * cbz)) band_step += cbz;
     }
-    int64_t vmin[3] = {bbmin[0], bbmin[1], bbmin[2]};
-    int64_t vdim[3] = {dims[0], dims[1], dims[2]};
-    if (bs_fuse_volume(ctx, fviews.data(), fviews.size(), vmin, vdim, &prm,
-                       nlevels, abs_ds.data(), ldims.data(),
-                       lvlptr.data()) != BS_OK) {
-      fprintf(stderr, "fusion failed: %s\n", bs_last_error(ctx));
-      return 1;
-    }
-    /* write every level's chunks (N5 3-D datasets or OME-ZARR 5-D
-     * arrays — the reference's 3-D-block-into-5-D lift,
-     * SparkAffineFusion.java:630-643) */
-    for (int l = 0; l < nlevels; ++l) {
-      bsn5::DatasetAttrs da;
-      bszarr::ArrayAttrs za;
-      int bx, by, bz;
-      if (zarr) {
-        if (!zr.get_array_attrs(dsnames[l], &za)) {
-          fprintf(stderr, "missing array %s in container\n",
-                  dsnames[l].c_str());
-          return 1;
-        }
-        bx = za.chunks[4];
-        by = za.chunks[3];
-        bz = za.chunks[2];
-      } else {
-        if (!n5.get_dataset_attrs(dsnames[l], &da)) {
-          fprintf(stderr, "missing dataset %s in container\n",
-                  dsnames[l].c_str());
-          return 1;
-        }
-        bx = da.block[0];
-        by = da.block[1];
-        bz = da.block[2];
+    {
+      size_t view_bytes = 0;
+      for (auto &pl : plans)
+        view_bytes +=
+            (size_t)pl.ldims3[0] * pl.ldims3[1] * pl.ldims3[2] * 2;
+      size_t out_bytes = 0;
+      for (int l = 0; l < nlevels; ++l) {
+        size_t b = esz2;
+        for (int d = 0; d < 3; ++d)
+          b *= (dims[d] + abs_ds[l * 3 + d] - 1) / abs_ds[l * 3 + d];
+        out_bytes += b;
       }
-      long long lx = ldims[l * 3], ly = ldims[l * 3 + 1],
-                lz = ldims[l * 3 + 2];
-      /* chunk compression (zstd/gzip) + file writes are independent
-       * per chunk: spread over a thread pool — the reference gets the
-       * same parallelism from its Spark executors writing blocks
-       * concurrently (N5Utils.saveBlock per task) */
-      const long long ngx = (lx + bx - 1) / bx, ngy = (ly + by - 1) / by,
-                      ngz = (lz + bz - 1) / bz;
-      const long long nchunks = ngx * ngy * ngz;
-      const int NW = (int)std::min<long long>(
-          nchunks, std::max(1u, std::thread::hardware_concurrency() / 2));
-      std::atomic<long long> next(0);
-      std::atomic<bool> failed(false);
-      auto worker = [&]() {
-        std::vector<char> blk((size_t)bx * by * bz * esz2);
-        for (;;) {
-          long long i = next.fetch_add(1);
-          if (i >= nchunks || failed.load()) return;
-          const long long gx = i % ngx, gy = (i / ngx) % ngy,
-                          gz = i / (ngx * ngy);
-          int cx = (int)std::min((long long)bx, lx - gx * bx);
-          int cy = (int)std::min((long long)by, ly - gy * by);
-          int cz = (int)std::min((long long)bz, lz - gz * bz);
-          const char *src = hostlvl[l].data();
-          for (int z = 0; z < cz; ++z)
-            for (int y = 0; y < cy; ++y)
-              memcpy(&blk[((size_t)z * cy + y) * cx * esz2],
-                     src + (((gz * bz + z) * ly + gy * by + y) * lx +
-                            gx * bx) * esz2,
-                     (size_t)cx * esz2);
-          bool ok;
-          if (zarr)
-            ok = zr.write_chunk(dsnames[l], za, {ti, 0, gz, gy, gx},
-                                blk.data(), {1, 1, cz, cy, cx});
-          else
-            ok = n5.write_block(dsnames[l], da, {gx, gy, gz}, blk.data(),
-                                {cx, cy, cz});
-          if (!ok) failed.store(true);
+      uint64_t freeb = 0, totb = 0;
+      (void)bs_device_mem(ctx, &freeb, &totb);
+      size_t budget = freeb > (8UL << 30) ? freeb - (4UL << 30)
+                                          : (size_t)freeb * 3 / 4;
+      if (view_bytes + out_bytes > budget) {
+        const long step = band_step;
+        const size_t plane =
+            (size_t)dims[0] * dims[1] * esz2 * 5 / 4; /* + pyramid */
+        long maxz = (long)(budget / 3 / std::max((size_t)1, plane));
+        bandz = std::max(step, (maxz / step) * step);
+        if (bandz < dims[2])
+          printf("z-band mode: %lld-deep bands (views %.1f GB + output "
+                 "%.1f GB exceed the %.1f GB budget)\n",
+                 (long long)bandz, view_bytes / 1e9, out_bytes / 1e9,
+                 budget / 1e9);
+      }
+    }
+    if (const char *e = getenv("BS_CLI_BAND_Z")) {
+      /* test hook: force a band depth (rounded up to the chunk/level
+       * alignment step) */
+      long v = std::max(1L, (long)atol(e));
+      bandz = ((v + band_step - 1) / band_step) * band_step;
+    }
+    std::set<int> resident;
+    for (long bz0 = 0; bz0 < (long)dims[2]; bz0 += bandz) {
+      const long bz1 = std::min((long)dims[2], bz0 + bandz);
+      /* slide the view window */
+      for (auto it = resident.begin(); it != resident.end();) {
+        bool still = false;
+        for (auto &pl : plans)
+          if (pl.setup == *it && pl.zhi >= bz0 - 2) still = true;
+        if (!still) {
+          bs_view_release(ctx, *it);
+          it = resident.erase(it);
+        } else {
+          ++it;
         }
-      };
-      std::vector<std::thread> ws;
-      for (int w = 0; w < NW; ++w) ws.emplace_back(worker);
-      for (auto &w : ws) w.join();
-      if (failed.load()) {
-        fprintf(stderr, "block write failed\n");
+      }
+      std::vector<bs_fuse_view> fviews;
+      for (auto &pl : plans) {
+        if (pl.zhi < bz0 - 2 || pl.zlo > bz1 + 2) continue;
+        if (!resident.count(pl.setup)) {
+          std::vector<uint16_t> vox;
+          std::vector<long long> vdims;
+          if (!in_n5.read_volume_u16(
+                  bssd::SpimData::image_dataset(pl.setup, pl.tp,
+                                                pl.level),
+                  &vox, &vdims)) {
+            fprintf(stderr, "cannot read view tp=%d setup=%d\n", pl.tp,
+                    pl.setup);
+            return 1;
+          }
+          int64_t d[3] = {vdims[0], vdims[1], vdims[2]};
+          if (bs_view_upload(ctx, pl.setup, vox.data(), d) != BS_OK) {
+            fprintf(stderr, "upload failed: %s\n", bs_last_error(ctx));
+            return 1;
+          }
+          if (args.has("intensityN5Path")) {
+            bsn5::Container coeff_n5(args.get("intensityN5Path"));
+            std::vector<float> ab;
+            int32_t gd[3];
+            if (load_coefficients(
+                    coeff_n5, args.get("intensityN5Group", ""),
+                    args.get("intensityN5Dataset", "intensity"),
+                    pl.setup, pl.tp, &ab, gd)) {
+              if (bs_view_set_coefficients(ctx, pl.setup, ab.data(),
+                                           gd) != BS_OK) {
+                fprintf(stderr, "set_coefficients failed: %s\n",
+                        bs_last_error(ctx));
+                return 1;
+              }
+              printf("loaded intensity coefficients for setup %d "
+                     "(%dx%dx%d)\n",
+                     pl.setup, gd[0], gd[1], gd[2]);
+            }
+          }
+          resident.insert(pl.setup);
+        }
+        fviews.push_back(pl.fv);
+      }
+      /* fuse this band (bs_fuse_volume may additionally z-slab
+       * internally when even the band exceeds free HBM) */
+      std::vector<std::vector<char>> hostlvl(nlevels);
+      std::vector<void *> lvlptr(nlevels);
+      std::vector<int64_t> ldims(3 * nlevels);
+      int64_t vmin[3] = {bbmin[0], bbmin[1], bbmin[2] + bz0};
+      int64_t vdim[3] = {dims[0], dims[1], bz1 - bz0};
+      for (int l = 0; l < nlevels; ++l) {
+        long long b = esz2;
+        for (int d = 0; d < 3; ++d)
+          b *= (vdim[d] + abs_ds[l * 3 + d] - 1) / abs_ds[l * 3 + d];
+        hostlvl[l].resize((size_t)b);
+        lvlptr[l] = hostlvl[l].data();
+      }
+      if (fviews.empty()) {
+        for (int l = 0; l < nlevels; ++l)
+          std::fill(hostlvl[l].begin(), hostlvl[l].end(), 0);
+        for (int l = 0; l < nlevels; ++l)
+          for (int d = 0; d < 3; ++d)
+            ldims[l * 3 + d] =
+                (vdim[d] + abs_ds[l * 3 + d] - 1) / abs_ds[l * 3 + d];
+      } else if (bs_fuse_volume(ctx, fviews.data(), fviews.size(), vmin,
+                                vdim, &prm, nlevels, abs_ds.data(),
+                                ldims.data(), lvlptr.data()) != BS_OK) {
+        fprintf(stderr, "fusion failed: %s\n", bs_last_error(ctx));
         return 1;
       }
-      printf("level %d (%s): wrote %lld blocks (%lldx%lldx%lld)\n", l,
-             dsnames[l].c_str(), nchunks, lx, ly, lz);
+      /* write the band's chunks (N5 3-D datasets or OME-ZARR 5-D
+       * arrays — the 3-D-block-into-5-D lift,
+       * SparkAffineFusion.java:630-643); band starts are chunk-aligned
+       * per level, so global chunk z = band offset + local */
+      for (int l = 0; l < nlevels; ++l) {
+        const int bx = cbx, by = cby, bz = cbz;
+        long long lx = ldims[l * 3], ly = ldims[l * 3 + 1],
+                  lz = ldims[l * 3 + 2];
+        if (fviews.empty()) {
+          lx = (vdim[0] + abs_ds[l * 3] - 1) / abs_ds[l * 3];
+          ly = (vdim[1] + abs_ds[l * 3 + 1] - 1) / abs_ds[l * 3 + 1];
+          lz = (vdim[2] + abs_ds[l * 3 + 2] - 1) / abs_ds[l * 3 + 2];
+        }
+        const long long gz_off = (bz0 / abs_ds[l * 3 + 2]) / bz;
+        const long long ngx = (lx + bx - 1) / bx,
+                        ngy = (ly + by - 1) / by,
+                        ngz = (lz + bz - 1) / bz;
+        const long long nchunks = ngx * ngy * ngz;
+        const int NW = (int)std::min<long long>(
+            nchunks,
+            std::max(1u, std::thread::hardware_concurrency() / 2));
+        std::atomic<long long> next(0);
+        std::atomic<bool> failed(false);
+        auto worker = [&]() {
+          std::vector<char> blk((size_t)bx * by * bz * esz2);
+          for (;;) {
+            long long i = next.fetch_add(1);
+            if (i >= nchunks || failed.load()) return;
+            const long long gx = i % ngx, gy = (i / ngx) % ngy,
+                            gz = i / (ngx * ngy);
+            int cx = (int)std::min((long long)bx, lx - gx * bx);
+            int cy = (int)std::min((long long)by, ly - gy * by);
+            int cz = (int)std::min((long long)bz, lz - gz * bz);
+            const char *src = hostlvl[l].data();
+            for (int z = 0; z < cz; ++z)
+              for (int y = 0; y < cy; ++y)
+                memcpy(&blk[((size_t)z * cy + y) * cx * esz2],
+                       src + (((gz * bz + z) * ly + gy * by + y) * lx +
+                              gx * bx) * esz2,
+                       (size_t)cx * esz2);
+            bool ok;
+            if (zarr)
+              ok = zr.write_chunk(dsnames[l], zas[l],
+                                  {ti, ci, gz_off + gz, gy, gx},
+                                  blk.data(), {1, 1, cz, cy, cx});
+            else
+              ok = n5.write_block(dsnames[l], das[l],
+                                  {gx, gy, gz_off + gz}, blk.data(),
+                                  {cx, cy, cz});
+            if (!ok) failed.store(true);
+          }
+        };
+        std::vector<std::thread> ws;
+        for (int w = 0; w < NW; ++w) ws.emplace_back(worker);
+        for (auto &w : ws) w.join();
+        if (failed.load()) {
+          fprintf(stderr, "block write failed\n");
+          return 1;
+        }
+        printf("level %d (%s): wrote %lld blocks (band z %ld..%ld)\n",
+               l, dsnames[l].c_str(), nchunks, bz0, bz1);
+      }
     }
-    for (auto &fv : fviews) bs_view_release(ctx, fv.view_id);
+    for (int id : resident) bs_view_release(ctx, id);
    }
   }
   bs_ctx_destroy(ctx);

@@ -61,6 +61,10 @@ void bs_ctx_destroy(bs_ctx *ctx);
 /* Last error message for a failed call on this ctx (valid until next call). */
 const char *bs_last_error(const bs_ctx *ctx);
 
+/* Free/total device memory of the ctx's GPU (host-side capacity
+ * planning: the fusion CLI sizes its z-band view window from this). */
+int bs_device_mem(bs_ctx *ctx, uint64_t *free_bytes, uint64_t *total_bytes);
+
 /* ------------------------------------------------------------------ views */
 
 /* Upload a view's voxels to device HBM and register them under view_id.

@@ -1047,3 +1047,33 @@ def test_cli_fusion_two_channels(tmp_path):
         denom = np.maximum(np.abs(ref), 1.0)
         assert np.max(np.abs(fused.astype(np.float64) - ref) /
                       denom) < 1e-4, f"ch{ci}"
+
+
+@pytest.mark.gpu
+def test_cli_fusion_zband_parity(tmp_path):
+    """The z-band mode (sliding view window for datasets larger than
+    HBM) must produce a bit-identical container to the single-band
+    path: same dataset fused with BS_CLI_BAND_Z=32 (two bands, views
+    uploaded/released per band) vs default."""
+    import shutil
+    xml, n5, err, (a, b) = make_grid_dataset(str(tmp_path))
+    outs = {}
+    for mode, env in (("one", {}), ("band", {"BS_CLI_BAND_Z": "32"})):
+        out = os.path.join(str(tmp_path), f"fused_{mode}.n5")
+        r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml,
+                 "-s", "N5", "-o", out, "--blockSize", "16,16,16",
+                 "-d", "FLOAT32", "-ds", "1,1,1", "-ds", "2,2,2"])
+        assert r.returncode == 0, r.stderr
+        e2 = dict(os.environ)
+        e2.update(env)
+        r = subprocess.run([os.path.join(BIN, "affine-fusion"), "-o", out,
+                            "-f", "AVG_BLEND", "--blendingRange", "8"],
+                           capture_output=True, text=True, env=e2)
+        assert r.returncode == 0, r.stderr + r.stdout
+        if env:
+            assert "band z 32..64" in r.stdout
+        outs[mode] = out
+    for ds in ("ch0tp0/s0", "ch0tp0/s1"):
+        va, _ = n5util.read_dataset(outs["one"], ds)
+        vb, _ = n5util.read_dataset(outs["band"], ds)
+        assert np.array_equal(va, vb), ds
