@@ -399,7 +399,49 @@ int main(int argc, char **argv) {
       *out_id = result;
       return true;
     };
-    for (auto &pp : plans) {
+    /* memory-bounded windows: pairs are prepared and stitched in
+     * chunks sized so the resident views (this window's + still-needed
+     * earlier ones) stay under the free-HBM budget; views whose last
+     * use has passed are released after each chunk. Large tile grids
+     * (views >> HBM) stitch on one GPU. */
+    uint64_t freeb = 0, totb = 0;
+    (void)bs_device_mem(ctx, &freeb, &totb);
+    size_t budget = freeb > (8UL << 30) ? (size_t)freeb - (4UL << 30)
+                                        : (size_t)freeb * 3 / 4;
+    if (const char *e = getenv("BS_STITCH_BUDGET_MB"))
+      budget = (size_t)atoll(e) << 20;
+    std::map<int, size_t> sbytes;     /* conservative: full-res bytes */
+    std::map<int, size_t> last_use;   /* setup -> last plan index */
+    for (size_t k = 0; k < plans.size(); ++k)
+      for (const VGroup *g2 : {plans[k].ga, plans[k].gb})
+        for (auto *m : g2->members) {
+          sbytes[m->id] =
+              (size_t)m->dims[0] * m->dims[1] * m->dims[2] * 2;
+          last_use[m->id] = k;
+        }
+    std::vector<bs_shift_result> res(plans.size());
+    size_t k0 = 0;
+    while (k0 < plans.size()) {
+      /* grow the window under the budget */
+      std::set<int> counted;
+      for (auto &kv : uploaded) counted.insert(kv.first);
+      size_t bytes = 0;
+      for (int id : counted) bytes += sbytes.count(id) ? sbytes[id] : 0;
+      size_t k1 = k0;
+      while (k1 < plans.size()) {
+        size_t add = 0;
+        for (const VGroup *g2 : {plans[k1].ga, plans[k1].gb})
+          for (auto *m : g2->members)
+            if (!counted.count(m->id)) add += sbytes[m->id];
+        if (k1 > k0 && bytes + add > budget) break;
+        bytes += add;
+        for (const VGroup *g2 : {plans[k1].ga, plans[k1].gb})
+          for (auto *m : g2->members) counted.insert(m->id);
+        ++k1;
+      }
+      const int tmp_first = next_tmp;
+      for (size_t k = k0; k < k1; ++k) {
+        auto &pp = plans[k];
       if (pp.nonequal) {
         /* resample both groups onto the overlap box at ds spacing via
          * the fusion sampler, then stitch the resampled volumes 1:1 */
@@ -454,12 +496,11 @@ int main(int argc, char **argv) {
         return 1;
       pp.pd.view_a = ga_id;
       pp.pd.view_b = gb_id;
-    }
+      }
     /* one bs_stitch_batch per distinct ds-remainder (pairs from
      * different pyramid depths stitch at different residual factors) */
-    std::vector<bs_shift_result> res(plans.size());
     std::map<std::array<int, 3>, std::vector<size_t>> groups;
-    for (size_t k = 0; k < plans.size(); ++k)
+    for (size_t k = k0; k < k1; ++k)
       groups[{plans[k].rem[0], plans[k].rem[1], plans[k].rem[2]}]
           .push_back(k);
     for (auto &g : groups) {
@@ -489,6 +530,23 @@ int main(int argc, char **argv) {
           gres[i].shift[d] *= (double)plans[g.second[i]].f[d];
         res[g.second[i]] = gres[i];
       }
+      }
+      /* chunk epilogue: drop views whose last use has passed, the
+       * chunk's temp views (combined / resampled), and the combine
+       * cache (later chunks recombine on demand) */
+      for (auto it = uploaded.begin(); it != uploaded.end();) {
+        auto lu = last_use.find(it->first);
+        if (lu == last_use.end() || lu->second < k1) {
+          (void)bs_view_release(ctx, it->first);
+          it = uploaded.erase(it);
+        } else {
+          ++it;
+        }
+      }
+      for (int id = tmp_first; id < next_tmp; ++id)
+        (void)bs_view_release(ctx, id);
+      combined.clear();
+      k0 = k1;
     }
     for (size_t k = 0; k < plans.size(); ++k) {
       const auto &pp = plans[k];

@@ -1077,3 +1077,58 @@ def test_cli_fusion_zband_parity(tmp_path):
         va, _ = n5util.read_dataset(outs["one"], ds)
         vb, _ = n5util.read_dataset(outs["band"], ds)
         assert np.array_equal(va, vb), ds
+
+
+@pytest.mark.gpu
+def test_cli_stitching_windowed_parity(tmp_path):
+    """Memory-bounded stitching windows: forcing a tiny view budget
+    (BS_STITCH_BUDGET_MB) makes the CLI stitch in chunks with view
+    eviction; the resulting XML links must be identical to the
+    unconstrained run."""
+    from tests.test_cli_solver import model_translations  # noqa: F401
+    import shutil
+    size, ov = 48, 16
+    step = size - ov
+    rng = np.random.default_rng(99)
+    setups, tiles, true_pos = [], {}, {}
+    sid = 0
+    for gy in range(2):
+        for gx in range(3):
+            nominal = np.array([gx * step, gy * step, 0.0])
+            errv = np.zeros(3) if sid == 0 else rng.uniform(-2.5, 2.5, 3)
+            true_pos[sid] = nominal + errv
+            setups.append(dict(id=sid, dims=(size, size, size),
+                               pos=tuple(nominal)))
+            sid += 1
+    world_rng = np.random.default_rng(5)
+    scene = synth.make_scene((size, size + step, size + 2 * step),
+                             world_rng, margin=10.0)
+    for s in range(6):
+        local = scene.copy()
+        for d in range(3):
+            local[:, d] -= np.float32(true_pos[s][d])
+        tiles[s] = synth.render_tile((size, size, size), local,
+                                     noise_seed=1000 + s)
+    results = {}
+    for mode, env in (("all", {}), ("win", {"BS_STITCH_BUDGET_MB": "1"})):
+        d2 = os.path.join(str(tmp_path), mode)
+        os.makedirs(d2)
+        n5 = os.path.join(d2, "input.n5")
+        xml = os.path.join(d2, "dataset.xml")
+        for s in range(6):
+            n5util.write_dataset(n5, f"setup{s}/timepoint0/s0", tiles[s],
+                                 (32, 32, 32))
+        n5util.make_dataset_xml(xml, "input.n5", setups)
+        e2 = dict(os.environ)
+        e2.update(env)
+        r = subprocess.run([os.path.join(BIN, "stitching"), "-x", xml,
+                            "-ds", "1,1,1", "--minOverlapRatio", "0.05"],
+                           capture_output=True, text=True, env=e2)
+        assert r.returncode == 0, r.stderr + r.stdout
+        prs = ET.parse(xml).getroot().findall(".//PairwiseResult")
+        results[mode] = sorted(
+            (pr.find("ViewIdsA").text, pr.find("ViewIdsB").text,
+             pr.find("Matrix").text, pr.find("Correlation").text)
+            for pr in prs)
+    assert len(results["all"]) >= 7
+    assert results["all"] == results["win"]
