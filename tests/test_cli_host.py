@@ -1050,18 +1050,18 @@ def test_cli_fusion_two_channels(tmp_path):
 
 
 @pytest.mark.gpu
-def test_cli_fusion_zband_parity(tmp_path):
+@pytest.mark.parametrize("storage", ["N5", "ZARR"])
+def test_cli_fusion_zband_parity(tmp_path, storage):
     """The z-band mode (sliding view window for datasets larger than
     HBM) must produce a bit-identical container to the single-band
     path: same dataset fused with BS_CLI_BAND_Z=32 (two bands, views
-    uploaded/released per band) vs default."""
-    import shutil
+    uploaded/released per band) vs default — for both storages."""
     xml, n5, err, (a, b) = make_grid_dataset(str(tmp_path))
     outs = {}
     for mode, env in (("one", {}), ("band", {"BS_CLI_BAND_Z": "32"})):
-        out = os.path.join(str(tmp_path), f"fused_{mode}.n5")
+        out = os.path.join(str(tmp_path), f"fused_{mode}")
         r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml,
-                 "-s", "N5", "-o", out, "--blockSize", "16,16,16",
+                 "-s", storage, "-o", out, "--blockSize", "16,16,16",
                  "-d", "FLOAT32", "-ds", "1,1,1", "-ds", "2,2,2"])
         assert r.returncode == 0, r.stderr
         e2 = dict(os.environ)
@@ -1073,10 +1073,31 @@ def test_cli_fusion_zband_parity(tmp_path):
         if env:
             assert "band z 32..64" in r.stdout
         outs[mode] = out
-    for ds in ("ch0tp0/s0", "ch0tp0/s1"):
-        va, _ = n5util.read_dataset(outs["one"], ds)
-        vb, _ = n5util.read_dataset(outs["band"], ds)
+    dss = (("ch0tp0/s0", "ch0tp0/s1") if storage == "N5"
+           else ("s0", "s1"))
+    rd = (n5util.read_dataset if storage == "N5" else n5util.read_zarr)
+    for ds in dss:
+        va, _ = rd(outs["one"], ds)
+        vb, _ = rd(outs["band"], ds)
         assert np.array_equal(va, vb), ds
+
+
+@pytest.mark.gpu
+def test_cli_stitching_fast_pad_e2e(tmp_path):
+    """--fftPadSize FAST through the CLI: a 60^3-tile dataset pads to
+    60/64 (fast) instead of 64 (pow2); the link still lands on the
+    injected shift."""
+    err = (2.5, -1.5, 1.0)
+    xml, n5, _e, _ = make_grid_dataset(str(tmp_path), size=60, overlap=22,
+                                       err=err)
+    r = run([os.path.join(BIN, "stitching"), "-x", xml, "-ds", "1,1,1",
+             "--fftPadSize", "FAST", "--minOverlapRatio", "0.05"])
+    assert r.returncode == 0, r.stderr + r.stdout
+    prs = ET.parse(xml).getroot().findall(".//PairwiseResult")
+    assert len(prs) == 1
+    m = [float(x) for x in prs[0].find("Matrix").text.split()]
+    for g, w in zip((m[3], m[7], m[11]), (-err[0], -err[1], -err[2])):
+        assert abs(g - w) < 0.75, ((m[3], m[7], m[11]), err)
 
 
 @pytest.mark.gpu
