@@ -5,10 +5,13 @@
  * per block (OverlappingViews, :536-537), fuses every block on the GPU
  * (bs_fuse_blocks = the BlkAffineFusion replacement, :602-615) and
  * writes the voxel blocks into the container (N5Utils.saveBlock, :670).
- * Round-1 scope: plain-N5 containers, level s0, FusionTypes
- * AVG/AVG_BLEND/MAX_INTENSITY; --masks writes coverage masks instead of
- * fused intensities (reference SparkAffineFusion.java:112-115, :565-578
- * via fusion/GenerateComputeBlockMasks.java). */
+ * Round-2 surface: N5 and OME-ZARR containers, multi-resolution
+ * pyramids, all six FusionTypes, --masks coverage masks
+ * (:112-115, :565-578), --intensityN5Path coefficients, anisotropy
+ * ([PIN-ANISO], :271-272/:486-491), per-view pyramid-level picks
+ * ([PIN-MIP]), multi-timepoint/channel output volumes
+ * (-t/--timepointIndex, -c/--channelIndex), and z-band processing with
+ * a sliding view window for datasets beyond HBM. */
 #include <algorithm>
 #include <atomic>
 #include <cstdio>

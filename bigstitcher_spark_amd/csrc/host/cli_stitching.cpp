@@ -5,12 +5,13 @@
  * FilteredStitchingResults filters (:369-380: minR/maxR/maxShift*), and
  * writes <StitchingResults> back into dataset.xml (:389).
  *
- * Round-1 scope (documented limitations vs reference):
- *  - one group per view (channel/illum grouping flags accepted, only
- *    defaults supported),
- *  - pairs whose registrations differ in their non-translational part
- *    are skipped (the reference falls back to
- *    computeStitchingNonEqualTransformations, :259-267). */
+ * Round-2 surface: view grouping over {illumination, channel} per tile
+ * with AVERAGE / PICK_BRIGHTEST aggregation on the GPU ([PIN-GROUP]),
+ * the computeStitchingNonEqualTransformations resample fallback
+ * ([PIN-NONEQ], :259-267), multi-resolution input levels
+ * ([PIN-MIPSEL]), view-selection flags, --fftPadSize POW2|FAST
+ * ([PIN-PAD]), and memory-bounded pair windows with view eviction for
+ * tile sets larger than HBM. */
 #include <cinttypes>
 #include <cstdio>
 #include <set>
