@@ -63,9 +63,10 @@ int main(int argc, char **argv) {
     if (args.has(f))
       fprintf(stderr, "note: --%s accepted for compatibility (no-op in "
                       "this build)\n", f);
-  if (args.has("bdv")) {
-    fprintf(stderr, "--bdv (BDV N5/HDF5 container variants) is not "
-                    "supported by this build\n");
+  const bool bdv = args.has("bdv");
+  if (bdv && !args.has("xmlout")) {
+    /* reference CreateFusionContainer.java:132-136 */
+    printf("Please specify the output XML for the BDV dataset: -xo\n");
     return 2;
   }
   bssd::SpimData sd;
@@ -154,6 +155,11 @@ int main(int argc, char **argv) {
       (long)std::max<size_t>(1, sd.timepoints.size()));
   int numCh = (int)args.getl("numChannels", 1);
   std::string storage = args.get("storage", "ZARR"); /* reference default */
+  if (bdv && storage != "N5") {
+    fprintf(stderr, "--bdv requires -s N5 in this build (no HDF5 "
+                    "libraries in this environment)\n");
+    return 2;
+  }
   if (storage != "ZARR" && storage != "N5") {
     fprintf(stderr,
             "unsupported --storage %s (supported: ZARR, N5; HDF5 is not "
@@ -175,7 +181,8 @@ int main(int argc, char **argv) {
     else
       n5.set_attr("", "Bigstitcher-Spark/" + k, v);
   };
-  set("FusionFormat", bsj::Value::mkstr(zarr ? "OME-ZARR" : "N5"));
+  set("FusionFormat", bsj::Value::mkstr(zarr ? "OME-ZARR"
+                                        : bdv ? "BDV/N5" : "N5"));
   set("InputXML", bsj::Value::mkstr(args.get("xml")));
   set("NumTimepoints", bsj::Value::mkint(numTp));
   set("NumChannels", bsj::Value::mkint(numCh));
@@ -319,7 +326,13 @@ int main(int argc, char **argv) {
         auto levels = bsj::Value::mkarr();
         for (size_t l = 0; l < ladder.size(); ++l) {
           char dsname[64];
-          snprintf(dsname, sizeof dsname, "ch%dtp%d/s%zu", ch, t, l);
+          /* BDV/N5: the bdv.n5 layout (setup = output channel) the
+           * viewer opens through the -xo XML; plain N5: ch{c}tp{t} */
+          if (bdv)
+            snprintf(dsname, sizeof dsname, "setup%d/timepoint%d/s%zu",
+                     ch, t, l);
+          else
+            snprintf(dsname, sizeof dsname, "ch%dtp%d/s%zu", ch, t, l);
           bsn5::DatasetAttrs da;
           da.dims = {(dims[0] + ladder[l][0] - 1) / ladder[l][0],
                      (dims[1] + ladder[l][1] - 1) / ladder[l][1],
@@ -332,6 +345,10 @@ int main(int argc, char **argv) {
             fprintf(stderr, "cannot create dataset %s\n", dsname);
             return 1;
           }
+          if (bdv)
+            n5.set_attr(dsname, "downsamplingFactors",
+                        bsj::Value::mkints(std::vector<long long>{
+                            ladder[l][0], ladder[l][1], ladder[l][2]}));
           auto lv = bsj::Value::mkobj();
           lv->obj["dataset"] = bsj::Value::mkstr(dsname);
           lv->obj["dimensions"] = bsj::Value::mkints(da.dims);
@@ -347,6 +364,34 @@ int main(int argc, char **argv) {
       }
   }
   set("MultiResolutionInfos", mri_all);
+  if (bdv) {
+    /* BDV project XML for the fused dataset: numCh setups of the fused
+     * dims, identity registrations (the fused volume IS the world
+     * frame), bdv.n5 loader pointing at the container */
+    std::vector<bssd::ViewSetup> fsetups(numCh);
+    std::map<bssd::ViewId, std::array<double, 12>> fregs;
+    std::vector<int> ftps;
+    for (int t = 0; t < numTp; ++t) ftps.push_back(t);
+    for (int ch = 0; ch < numCh; ++ch) {
+      fsetups[ch].id = ch;
+      fsetups[ch].name = "fused channel " + std::to_string(ch);
+      for (int d = 0; d < 3; ++d) fsetups[ch].dims[d] = dims[d];
+      for (int t = 0; t < numTp; ++t)
+        fregs[{t, ch}] = {1, 0, 0, 0, 0, 1, 0, 0, 0, 0, 1, 0};
+    }
+    auto xr = bssd::make_dataset_xml(args.get("outputPath"), fsetups,
+                                     ftps, fregs);
+    /* absolute container path */
+    auto seq = xr->child("SequenceDescription");
+    auto il = seq ? seq->child("ImageLoader") : nullptr;
+    auto n5n = il ? il->child("n5") : nullptr;
+    if (n5n) n5n->attrs["type"] = "absolute";
+    if (!bsx::save_file(args.get("xmlout"), xr)) {
+      fprintf(stderr, "cannot write %s\n", args.get("xmlout").c_str());
+      return 1;
+    }
+    printf("BDV project XML: %s\n", args.get("xmlout").c_str());
+  }
   printf("created %s: %d tp x %d ch, bbox [%lld,%lld,%lld]..[%lld,%lld,%lld]"
          ", %zu level(s), %s %s\n",
          args.get("outputPath").c_str(), numTp, numCh, bbmin[0], bbmin[1],

@@ -149,10 +149,11 @@ int main(int argc, char **argv) {
   /* exact-format whitelist: BDV-prefixed and HDF5 containers have different
    * dataset layouts (reference SparkAffineFusion.java:239-307) — refuse
    * them explicitly rather than guessing a layout */
-  if (fmt->str != "N5" && fmt->str != "OME-ZARR") {
+  if (fmt->str != "N5" && fmt->str != "OME-ZARR" &&
+      fmt->str != "BDV/N5") {
     fprintf(stderr,
-            "unsupported FusionFormat '%s' (supported: N5, OME-ZARR; "
-            "BDV/* and HDF5 containers are not supported by this build)\n",
+            "unsupported FusionFormat '%s' (supported: N5, OME-ZARR, "
+            "BDV/N5; HDF5 containers are not supported by this build)\n",
             fmt->str.c_str());
     return 1;
   }
