@@ -69,6 +69,7 @@ asan:
 	$(CXX_HOST) $(ASANFLAGS) -Iinclude $(HOSTDIR)/cli_container.cpp $(HOSTDIR)/bs_json.cpp $(HOSTDIR)/bs_n5.cpp $(HOSTDIR)/bs_zarr.cpp $(HOSTDIR)/bs_xml.cpp $(HOSTDIR)/bs_spimdata.cpp -o $(ASANDIR)/create-fusion-container -lz -l:libzstd.so.1
 	$(CXX_HOST) $(ASANFLAGS) -Iinclude $(HOSTDIR)/cli_solver.cpp $(HOSTDIR)/bs_json.cpp $(HOSTDIR)/bs_n5.cpp $(HOSTDIR)/bs_zarr.cpp $(HOSTDIR)/bs_xml.cpp $(HOSTDIR)/bs_spimdata.cpp -o $(ASANDIR)/solver -lz -l:libzstd.so.1
 	$(CXX_HOST) $(ASANFLAGS) -Iinclude $(HOSTDIR)/cli_stitching.cpp $(HOSTDIR)/bs_json.cpp $(HOSTDIR)/bs_n5.cpp $(HOSTDIR)/bs_zarr.cpp $(HOSTDIR)/bs_xml.cpp $(HOSTDIR)/bs_spimdata.cpp -o $(ASANDIR)/stitching -Lbigstitcher_spark_amd -lbigstitch -lz -l:libzstd.so.1 -L/opt/rocm/lib -lamdhip64 -Wl,-rpath,'$$ORIGIN/..' -Wl,-rpath,/opt/rocm/lib
+	$(CXX_HOST) $(ASANFLAGS) -Iinclude $(HOSTDIR)/cli_fusion.cpp $(HOSTDIR)/bs_json.cpp $(HOSTDIR)/bs_n5.cpp $(HOSTDIR)/bs_zarr.cpp $(HOSTDIR)/bs_xml.cpp $(HOSTDIR)/bs_spimdata.cpp -o $(ASANDIR)/affine-fusion -Lbigstitcher_spark_amd -lbigstitch -lz -l:libzstd.so.1 -L/opt/rocm/lib -lamdhip64 -Wl,-rpath,'$$ORIGIN/..' -Wl,-rpath,/opt/rocm/lib
 	$(CXX_HOST) $(ASANFLAGS) -Iinclude $(HOSTDIR)/cli_resave.cpp $(HOSTDIR)/bs_json.cpp $(HOSTDIR)/bs_n5.cpp $(HOSTDIR)/bs_zarr.cpp $(HOSTDIR)/bs_xml.cpp $(HOSTDIR)/bs_spimdata.cpp -o $(ASANDIR)/resave -Lbigstitcher_spark_amd -lbigstitch -lz -l:libzstd.so.1 -L/opt/rocm/lib -lamdhip64 -Wl,-rpath,'$$ORIGIN/..' -Wl,-rpath,/opt/rocm/lib
 
 .PHONY: asan
