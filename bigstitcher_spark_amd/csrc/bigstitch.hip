@@ -3276,8 +3276,12 @@ extern "C" int bs_fuse_blocks(bs_ctx *c, const bs_fuse_view *views,
     {
       long nrows_f = (long)bd.size[1] * bd.size[2];
       long gfb = std::max(8L, std::min(4096L, nrows_f) & ~7L);
-      static const bool noswiz = getenv("BS_FUSE_NOSWIZ") != nullptr;
-      if (noswiz && gfb > 8) gfb -= 1; /* non-x8 grid = identity map */
+      /* XCD row swizzle and per-row interval culling both measured
+       * slightly NEGATIVE on the 2x2x2 bench (446 GB/s with both off
+       * vs 428 with both on, same box) — opt-in via BS_FUSE_SWIZ /
+       * BS_FUSE_RIV for many-view workloads */
+      static const bool swiz = getenv("BS_FUSE_SWIZ") != nullptr;
+      if (!swiz && gfb > 8) gfb -= 1; /* non-x8 grid = identity map */
       bs_tim tt(c, BS_K_FUSE);
       if (prm->masks)
         hipLaunchKernelGGL(k_mask, dim3(gfb),
@@ -3299,7 +3303,7 @@ extern "C" int bs_fuse_blocks(bs_ctx *c, const bs_fuse_view *views,
                            (float)prm->min_intensity, invRange, obuf,
                            0L, (long)bd.size[0],
                            (long)bd.size[0] * bd.size[1],
-                           getenv("BS_FUSE_NORIV") ? 0 : 1);
+                           getenv("BS_FUSE_RIV") ? 1 : 0);
     }
     CHK(c, hipEventRecord(bev[i & 1], c->stream));
     CHK(c, hipStreamWaitEvent(c->copy_stream, bev[i & 1], 0));
@@ -3549,7 +3553,7 @@ extern "C" int bs_fuse_volume(bs_ctx *c, const bs_fuse_view *views,
                              (int)fb[5], prm->fusion_type, prm->out_dtype,
                              (float)prm->min_intensity, invRange,
                              slabbuf[sbi], oo, vrow0, vslice0,
-                             getenv("BS_FUSE_NORIV") ? 0 : 1);
+                             getenv("BS_FUSE_RIV") ? 1 : 0);
       }
       nblocks += (long long)fbl.size();
       /* this slab's pyramid contributions (level l from level l-1) */
@@ -3689,8 +3693,8 @@ extern "C" int bs_fuse_volume(bs_ctx *c, const bs_fuse_view *views,
     int nvb = (int)(offs[b + 1] - offs[b]);
     long nrows_f = fb[4] * fb[5];
     long gfb = std::max(8L, std::min(4096L, nrows_f) & ~7L);
-    static const bool noswiz = getenv("BS_FUSE_NOSWIZ") != nullptr;
-    if (noswiz && gfb > 8) gfb -= 1; /* non-x8 grid = identity map */
+    static const bool swiz = getenv("BS_FUSE_SWIZ") != nullptr;
+    if (!swiz && gfb > 8) gfb -= 1; /* non-x8 grid = identity map */
     bs_tim tt(c, BS_K_FUSE, c->stream);
     if (prm->masks)
       hipLaunchKernelGGL(k_mask, dim3(gfb), dim3(256),
@@ -3709,7 +3713,7 @@ extern "C" int bs_fuse_volume(bs_ctx *c, const bs_fuse_view *views,
                          (int)fb[5], prm->fusion_type, prm->out_dtype,
                          (float)prm->min_intensity, invRange, dlvl[0],
                          (long)fb[2] * vslice + fb[1] * vrow + fb[0], vrow,
-                         vslice, getenv("BS_FUSE_NORIV") ? 0 : 1);
+                         vslice, getenv("BS_FUSE_RIV") ? 1 : 0);
     if (b + 1 == fblocks.size() || fblocks[b + 1][2] != fb[2]) {
       hipEvent_t ev;
       CHK(c, hipEventCreateWithFlags(&ev, hipEventDisableTiming));
