@@ -388,3 +388,44 @@ def test_subpixel_offset_exact_parabola():
     p3[1, 0, 0] = 0.4
     off3 = _subpixel_offset(p3, (0, 0, 0))
     assert -0.5 <= off3[0] <= 0.5 and off3[0] != 0.0
+
+
+def test_fast_pad_sizes():
+    """[PIN-PAD] fast mode: smallest EVEN 7-smooth size >= max(n, 8)."""
+    from oracle.phasecorr import _next_fast_even
+    assert _next_fast_even(1) == 8
+    assert _next_fast_even(45) == 48      # 2^4*3
+    assert _next_fast_even(49) == 50      # 2*5^2
+    assert _next_fast_even(51) == 54      # 2*3^3
+    assert _next_fast_even(55) == 56      # 2^3*7
+    assert _next_fast_even(512) == 512
+    assert _next_fast_even(513) == 540    # 2^2*3^3*5
+    # every returned size is even and 7-smooth
+    for n in range(8, 200):
+        m = _next_fast_even(n)
+        assert m >= n and m % 2 == 0
+        x = m
+        for f in (2, 3, 5, 7):
+            while x % f == 0:
+                x //= f
+        assert x == 1, (n, m)
+
+
+def test_known_shift_fast_pad():
+    """Injected sub-pixel shift recovered in fast-pad mode at 7-smooth
+    pad sizes (analytic ground truth — the oracle pins itself for the
+    mode, mirroring the pow2-mode known-answer tests)."""
+    from oracle import phasecorr, synth
+    shape = (55, 50, 45)  # pads to (56, 50, 48): radices 7, 5, 3
+    shift = (3.25, -2.5, 1.0)
+    a, b = synth.make_pair(shape, shift, seed=23)
+    r = phasecorr.phase_correlation_shift(a, b, ds=(1, 1, 1),
+                                          pad_mode="fast")
+    assert r["valid"]
+    import numpy as np
+    got = np.asarray(r["shift"])
+    assert np.all(np.abs(got - np.asarray(shift)) < 0.35), got
+    # and the two modes agree with each other within subpixel noise
+    r2 = phasecorr.phase_correlation_shift(a, b, ds=(1, 1, 1),
+                                           pad_mode="pow2")
+    assert np.all(np.abs(got - np.asarray(r2["shift"])) < 0.5)
