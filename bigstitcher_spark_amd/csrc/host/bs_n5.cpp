@@ -5,9 +5,12 @@
 #include <zlib.h>
 #include <cerrno>
 
+#include <atomic>
 #include <cstdio>
 #include <cstring>
 #include <fstream>
+#include <thread>
+#include <vector>
 
 namespace bsio {
 
@@ -329,23 +332,39 @@ bool Container::read_volume_u16(const std::string &name,
   long long nx = a.dims[0], ny = a.dims[1], nz = a.dims[2];
   out->assign((size_t)(nx * ny * nz), 0);
   int bx = a.block[0], by = a.block[1], bz = a.block[2];
-  std::vector<uint16_t> blk((size_t)bx * by * bz);
-  for (long long gz = 0; gz * bz < nz; ++gz)
-    for (long long gy = 0; gy * by < ny; ++gy)
-      for (long long gx = 0; gx * bx < nx; ++gx) {
-        std::vector<int> cd;
-        if (!read_block(name, a, {gx, gy, gz}, blk.data(), &cd))
-          continue; /* missing chunk = zeros (N5 semantics) */
-        int cx = cd[0], cy = cd[1], cz = cd[2];
-        for (int z = 0; z < cz; ++z)
-          for (int y = 0; y < cy; ++y) {
-            long long dst =
-                ((gz * bz + z) * ny + (gy * by + y)) * nx + gx * bx;
-            memcpy(&(*out)[dst], &blk[((size_t)z * cy + y) * cx],
-                   (size_t)cx * 2);
-          }
-      }
-  return true;
+  /* chunk decode (gzip/zstd) dominates large reads: thread pool, one
+   * scratch block per worker; chunks write disjoint output regions */
+  const long long ngx = (nx + bx - 1) / bx, ngy = (ny + by - 1) / by,
+                  ngz = (nz + bz - 1) / bz;
+  const long long nchunks = ngx * ngy * ngz;
+  const int NW = (int)std::min<long long>(
+      nchunks, std::max(1u, std::thread::hardware_concurrency() / 2));
+  std::atomic<long long> next(0);
+  std::atomic<bool> failed(false);
+  auto worker = [&]() {
+    std::vector<uint16_t> blk((size_t)bx * by * bz);
+    for (;;) {
+      long long i = next.fetch_add(1);
+      if (i >= nchunks || failed.load()) return;
+      const long long gx = i % ngx, gy = (i / ngx) % ngy,
+                      gz = i / (ngx * ngy);
+      std::vector<int> cd;
+      if (!read_block(name, a, {gx, gy, gz}, blk.data(), &cd))
+        continue; /* missing chunk = zeros (N5 semantics) */
+      int cx = cd[0], cy = cd[1], cz = cd[2];
+      for (int z = 0; z < cz; ++z)
+        for (int y = 0; y < cy; ++y) {
+          long long dst =
+              ((gz * bz + z) * ny + (gy * by + y)) * nx + gx * bx;
+          memcpy(&(*out)[dst], &blk[((size_t)z * cy + y) * cx],
+                 (size_t)cx * 2);
+        }
+    }
+  };
+  std::vector<std::thread> ws;
+  for (int w = 0; w < NW; ++w) ws.emplace_back(worker);
+  for (auto &w : ws) w.join();
+  return !failed.load();
 }
 
 }  // namespace bsn5
