@@ -5,7 +5,7 @@
  *      registrations (Solver.java:404-415 semantics; hash rule
  *      [PIN-HASH] in bs_spimdata.h),
  *   2. solves the global translation adjustment minimising
- *      sum_links || (t_B - t_A) - d_link ||^2 with the first (or
+ *      sum_links r_link * || (t_B - t_A) - d_link ||^2 with the first (or
  *      --fixedViews) view fixed — the translation case of the
  *      reference's GlobalOpt tile optimisation (Solver.java:301-337;
  *      the mpicbg TranslationModel3D path) as a direct graph-Laplacian
@@ -57,6 +57,10 @@ int main(int argc, char **argv) {
   struct Link {
     int a, b;
     double d[3];
+    double w = 1.0; /* [PIN-WEIGHT]: the reference's point matches
+      carry the link's correlation as the mpicbg weight
+      (ImageCorrelationPointMatchCreator -> PointMatch weight;
+      artifact un-vendored, restated as w = r) */
   };
   std::vector<Link> links;
   size_t dropped_hash = 0, dropped_r = 0;
@@ -81,6 +85,7 @@ int main(int argc, char **argv) {
     l.a = ia->second;
     l.b = ib->second;
     for (int d = 0; d < 3; ++d) l.d[d] = -e.matrix[d * 4 + 3];
+    l.w = std::max(0.0, e.r);
     links.push_back(l);
   }
   printf("solver: %zu usable links (%zu stale-hash, %zu below minR)\n",
@@ -116,12 +121,12 @@ int main(int argc, char **argv) {
   for (int ax = 0; ax < 3; ++ax) {
     std::vector<double> A((size_t)n * n, 0.0), rhs(n, 0.0);
     for (auto &l : links) {
-      A[(size_t)l.a * n + l.a] += 1.0;
-      A[(size_t)l.b * n + l.b] += 1.0;
-      A[(size_t)l.a * n + l.b] -= 1.0;
-      A[(size_t)l.b * n + l.a] -= 1.0;
-      rhs[l.a] -= l.d[ax];
-      rhs[l.b] += l.d[ax];
+      A[(size_t)l.a * n + l.a] += l.w;
+      A[(size_t)l.b * n + l.b] += l.w;
+      A[(size_t)l.a * n + l.b] -= l.w;
+      A[(size_t)l.b * n + l.a] -= l.w;
+      rhs[l.a] -= l.w * l.d[ax];
+      rhs[l.b] += l.w * l.d[ax];
     }
     for (int f : fixed) { /* pin t_f = 0 */
       for (int j = 0; j < n; ++j) A[(size_t)f * n + j] = 0.0;
