@@ -133,6 +133,8 @@ def main():
                     help="distinct synthesized pairs per GPU")
     ap.add_argument("--overlap", type=float, default=0.1)
     ap.add_argument("--no-cpu-baseline", action="store_true")
+    ap.add_argument("--pad-mode", default="pow2", choices=["pow2", "fast"],
+                    help="[PIN-PAD] FFT pad rule (fast = even 7-smooth)")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -176,7 +178,8 @@ def main():
 
     def step():
         return ctx.stitch_batch(pairs, ds=(1, 1, 1), peaks_to_check=5,
-                                do_subpixel=True, min_overlap_ratio=0.05)
+                                do_subpixel=True, min_overlap_ratio=0.05,
+                                pad_mode=args.pad_mode)
 
     for _ in range(args.warmup):
         res = step()
