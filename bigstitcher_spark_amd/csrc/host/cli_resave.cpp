@@ -24,12 +24,13 @@ int main(int argc, char **argv) {
   bscli::Args args;
   std::map<std::string, std::string> alias = {{"-x", "--xml"},
                                               {"-o", "--n5Path"},
+                                              {"-c", "--compression"},
                                               {"-xo", "--xmlOut"}};
   if (!args.parse(argc, argv, alias, {}) || !args.has("xml") ||
       !args.has("n5Path")) {
     fprintf(stderr,
             "usage: resave -x dataset.xml -o out.n5 [-xo out.xml] "
-            "[--blockSize 128,128,64] "
+            "[--blockSize 128,128,64] [-c Gzip|Zstandard|Raw] "
             "[--downsamplings \"1,1,1;2,2,1;4,4,2\"] [--device N]\n");
     return 2;
   }
@@ -124,7 +125,11 @@ int main(int argc, char **argv) {
         da.dims = {ldims[l * 3], ldims[l * 3 + 1], ldims[l * 3 + 2]};
         da.block = {(int)bs[0], (int)bs[1], (int)bs[2]};
         da.dtype = "uint16";
-        da.compression = "gzip";
+        {
+          std::string cn = args.get("compression", "Gzip");
+          da.compression = cn == "Zstandard" ? "zstd"
+                           : cn == "Raw" ? "raw" : "gzip";
+        }
         if (!out_n5.create_dataset(dsn, da)) {
           fprintf(stderr, "cannot create %s\n", dsn.c_str());
           return 1;
