@@ -844,15 +844,6 @@ __global__ __launch_bounds__(LPB_S *TPL_S) void k_fft_z_fused_r8(
 /* no-register-prefetch variant, VGPR-capped for 4 blocks/CU (34 KB LDS
  * is the other limit at n=512): __launch_bounds__ min-blocks 4 */
 template <int NE>
-__global__ __launch_bounds__(LPB_S *TPL_S) void k_fft_z_fused_np2(
-    f2 *a, f2 *b, int n, int log2n, long estride, long gstride, int nlines,
-    int nchunks, int ngroups, int valid_a, int valid_b, float scale,
-    const f2 *twg) {
-  zf_body<NE, false>(a, b, n, log2n, estride, gstride, nlines, nchunks,
-                     ngroups, valid_a, valid_b, scale, twg);
-}
-
-template <int NE>
 __global__ __launch_bounds__(LPB_S *TPL_S) __attribute__((amdgpu_waves_per_eu(8))) void k_fft_z_fused_np(
     f2 *a, f2 *b, int n, int log2n, long estride, long gstride, int nlines,
     int nchunks, int ngroups, int valid_a, int valid_b, float scale,
