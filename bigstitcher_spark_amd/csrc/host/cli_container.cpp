@@ -16,7 +16,9 @@
  * -ds/--downsampling (";"-splittable), --preserveAnisotropy,
  * --anisotropyFactor, plus view-selection (-vi/--angleId/...).
  * Unsupported storage/codec values exit with an explicit error.
- * HDF5/--bdv variants are not built (documented gap). */
+ * --bdv writes the bdv.n5 layout (setup{c}/timepoint{t}/s{l}) plus the
+ * BDV project XML (-xo); HDF5 storage is not built (no HDF5 libs in
+ * this image — documented gap). */
 #include <cmath>
 #include <cstdio>
 #include <set>
