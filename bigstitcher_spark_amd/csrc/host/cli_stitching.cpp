@@ -59,6 +59,12 @@ int main(int argc, char **argv) {
               a.c_str());
       return 2;
     }
+  const std::string padSz = args.get("fftPadSize", "POW2");
+  if (padSz != "POW2" && padSz != "FAST") {
+    fprintf(stderr, "unsupported --fftPadSize %s (POW2|FAST)\n",
+            padSz.c_str());
+    return 2;
+  }
   auto ds = bscli::parse_ints(args.get("downsampling", "2,2,1"));
   if (ds.size() != 3) {
     fprintf(stderr, "bad -ds\n");
@@ -516,7 +522,7 @@ int main(int argc, char **argv) {
       prm.min_overlap_ratio = args.getd("minOverlapRatio", 0.25);
       /* [PIN-PAD]: pow2 (default) or the reference dependency's even
        * 7-smooth "fast" pad sizes */
-      prm.pad_mode = args.get("fftPadSize", "POW2") == "FAST" ? 1 : 0;
+      prm.pad_mode = padSz == "FAST" ? 1 : 0;
       std::vector<bs_shift_result> gres(pds.size());
       int rc =
           bs_stitch_batch(ctx, pds.data(), pds.size(), &prm, gres.data());

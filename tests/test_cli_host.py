@@ -1215,3 +1215,56 @@ def test_cli_fusion_bdv_n5_end_to_end(tmp_path):
     # the output XML is itself a valid dataset for this repo's tools
     r = run([os.path.join(BIN, "stitching"), "-x", xo, "--dryRun"])
     assert r.returncode == 0, r.stderr
+
+
+def test_stitching_rejects_bad_pad_size(tmp_path):
+    """--fftPadSize takes the reference enum's two values; anything else
+    exits with an explicit error (picocli enum-conversion behavior)."""
+    xml, _n5, _err, _ = make_grid_dataset(str(tmp_path))
+    r = run([os.path.join(BIN, "stitching"), "-x", xml,
+             "--fftPadSize", "LARGE", "--dryRun"])
+    assert r.returncode == 2 and "fftPadSize" in r.stderr
+
+
+def test_container_cli_anisotropy_explicit_factor(tmp_path):
+    """--anisotropyFactor given explicitly overrides the voxel-size
+    computation (CreateFusionContainer.java:189-211: the NaN sentinel
+    triggers the average; a real value is used as-is)."""
+    n5 = os.path.join(str(tmp_path), "input.n5")
+    xml = os.path.join(str(tmp_path), "dataset.xml")
+    rng = np.random.default_rng(4)
+    vol = rng.integers(0, 60000, size=(16, 32, 32)).astype(np.uint16)
+    n5util.write_dataset(n5, "setup0/timepoint0/s0", vol, (16, 16, 16))
+    n5util.make_dataset_xml(
+        xml, "input.n5",
+        [dict(id=0, dims=(32, 32, 16), pos=(0.0, 0.0, 0.0),
+              voxel=(0.5, 0.5, 2.0))])  # computed factor would be 4.0
+    out = os.path.join(str(tmp_path), "fused.n5")
+    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml,
+             "-s", "N5", "-o", out, "--blockSize", "16,16,16",
+             "-d", "UINT16", "--preserveAnisotropy",
+             "--anisotropyFactor", "2.0"])
+    assert r.returncode == 0, r.stderr
+    attrs = n5util.root_attrs(out)["Bigstitcher-Spark"]
+    assert attrs["PreserveAnisotropy"] is True
+    assert abs(attrs["AnisotropyFactor"] - 2.0) < 1e-12
+    # z extent [0,15] -> [floor(0/2), ceil(15/2)] = [0, 8]
+    assert attrs["Boundingbox_max"][2] == 8
+    assert attrs["MultiResolutionInfos"][0][0]["dimensions"] == [32, 32, 9]
+
+
+@pytest.mark.gpu
+def test_cli_resave_zstd_codec(tmp_path):
+    """resave -c Zstandard: the re-chunked container stores zstd chunks
+    (SparkResaveN5's compression option; Zstandard is the project-wide
+    default codec) and s0 still round-trips exactly."""
+    xml, n5, err, (a, b) = make_grid_dataset(str(tmp_path))
+    out = os.path.join(str(tmp_path), "resaved.n5")
+    xo = os.path.join(str(tmp_path), "resaved.xml")
+    r = run([os.path.join(BIN, "resave"), "-x", xml, "-o", out, "-xo", xo,
+             "--blockSize", "32,32,16", "-c", "Zstandard",
+             "--downsamplings", "1,1,1"])
+    assert r.returncode == 0, r.stderr + r.stdout
+    s0, attrs = n5util.read_dataset(out, "setup0/timepoint0/s0")
+    assert attrs["compression"]["type"] == "zstd"
+    assert np.array_equal(s0, a)
