@@ -182,3 +182,62 @@ def test_solver_missing_file_clean_error(tmp_path):
     r = run([os.path.join(BIN, "solver"), "-x",
              os.path.join(str(tmp_path), "nope.xml")])
     assert r.returncode != 0 and r.returncode < 128
+
+
+def test_solver_correlation_weighted(tmp_path):
+    """[PIN-WEIGHT] w = max(0, r): an inconsistent low-correlation cycle
+    link pulls the solution by exactly the weighted-Laplacian amount
+    (checked against an independent numpy solve), and NOT by the
+    unweighted amount."""
+    xml = os.path.join(str(tmp_path), "dataset.xml")
+    n5util.make_dataset_xml(
+        xml, "input.n5",
+        [dict(id=0, dims=(64, 64, 64), pos=(0.0, 0.0, 0.0)),
+         dict(id=1, dims=(64, 64, 64), pos=(40.0, 0.0, 0.0)),
+         dict(id=2, dims=(64, 64, 64), pos=(80.0, 0.0, 0.0))],
+    )
+    # links as (a, b, ws_x, r, hash = 3+posA + 3+posB of the current
+    # identity+translation registrations)
+    links = [("0,0", "0,1", -1.0, 0.9, 3.0 + 43.0),
+             ("0,1", "0,2", -1.0, 0.9, 43.0 + 83.0),
+             ("0,0", "0,2", -5.0, 0.35, 3.0 + 83.0)]  # above minR=0.3
+    text = open(xml).read()
+    sr = "  <StitchingResults>\n"
+    for a, b, ws, r_, h in links:
+        sr += (
+            "    <PairwiseResult>\n"
+            f"      <ViewIdsA>{a}</ViewIdsA>\n      <ViewIdsB>{b}</ViewIdsB>\n"
+            f"      <Matrix>1 0 0 {ws} 0 1 0 0 0 0 1 0</Matrix>\n"
+            "      <BoundingBoxMin>0 0 0</BoundingBoxMin>\n"
+            "      <BoundingBoxMax>1 1 1</BoundingBoxMax>\n"
+            f"      <Correlation>{r_}</Correlation>\n"
+            f"      <Hash>{h}</Hash>\n    </PairwiseResult>\n"
+        )
+    sr += "  </StitchingResults>\n"
+    open(xml, "w").write(text.replace("</SpimData>", sr + "</SpimData>"))
+    r = run([os.path.join(BIN, "solver"), "-x", xml])
+    assert r.returncode == 0, r.stderr + r.stdout
+    t = model_translations(xml)
+
+    def solve(ws_list):
+        # minimize sum w (d_b - d_a - m)^2, d_0 fixed at 0; m = -ws
+        A = np.zeros((2, 2))
+        rhs = np.zeros(2)
+        for (ia, ib, ws, w) in ws_list:
+            m = -ws
+            for (i, s) in ((ia, -1.0), (ib, 1.0)):
+                if i == 0:
+                    continue
+                A[i - 1, i - 1] += w
+                other = ib if i == ia else ia
+                if other != 0:
+                    A[i - 1, other - 1] -= w
+                rhs[i - 1] += s * w * m
+        return np.linalg.solve(A, rhs)
+
+    lw = [(0, 1, -1.0, 0.9), (1, 2, -1.0, 0.9), (0, 2, -5.0, 0.35)]
+    dw = solve(lw)
+    du = solve([(a, b, ws, 1.0) for (a, b, ws, _w) in lw])
+    assert abs(dw[0] - du[0]) > 0.1  # the weight matters for this graph
+    assert np.allclose(t[1], [40.0 + dw[0], 0, 0], atol=1e-9)
+    assert np.allclose(t[2], [80.0 + dw[1], 0, 0], atol=1e-9)
