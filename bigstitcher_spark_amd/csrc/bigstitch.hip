@@ -653,7 +653,7 @@ __global__ __launch_bounds__(LPB_S *TPL_S) void k_fft_pass_glds(
  * same buffer for the inverse DIT. NE = elements per thread
  * (compile-time so the staging arrays live in REGISTERS, not scratch);
  * n <= 128*NE, max 1024. */
-template <int NE, bool PB, bool R8 = false>
+template <int NE, bool PB, bool R8 = false, bool NT = false>
 __device__ __forceinline__ void zf_body(
     f2 *a, f2 *b, int n, int log2n, long estride, long gstride, int nlines,
     int nchunks, int ngroups, int valid_a, int valid_b, float scale,
@@ -803,8 +803,17 @@ __device__ __forceinline__ void zf_body(
       for (int k = 0; k < NE; ++k) {
         const int e = t2 + k * ESTR;
         if (e >= n) break;
-        *(float4 *)&a[base2 + e * estride] =
-            *(const float4 *)&da[(long)e * LPB_S + 2 * pl];
+        const float4 v = *(const float4 *)&da[(long)e * LPB_S + 2 * pl];
+        if (NT) { /* skip the L2 write-allocate RFO: each WG writes only
+                     64 of each 128 B output line (profiles/README.md,
+                     the one traffic gap) and nothing re-reads Q before
+                     the inverse-y pass streams the whole volume */
+          typedef float vf4 __attribute__((ext_vector_type(4)));
+          __builtin_nontemporal_store(*(const vf4 *)&v,
+                                      (vf4 *)&a[base2 + e * estride]);
+        }
+        else
+          *(float4 *)&a[base2 + e * estride] = v;
       }
     } else {
       for (int l = 0; l < 2; ++l) {
@@ -850,6 +859,17 @@ __global__ __launch_bounds__(LPB_S *TPL_S) __attribute__((amdgpu_waves_per_eu(8)
     const f2 *twg) {
   zf_body<NE, false>(a, b, n, log2n, estride, gstride, nlines, nchunks,
                      ngroups, valid_a, valid_b, scale, twg);
+}
+
+/* _np + nontemporal Q stores (BS_Z_NT A/B) */
+template <int NE>
+__global__ __launch_bounds__(LPB_S *TPL_S) __attribute__((amdgpu_waves_per_eu(8))) void k_fft_z_fused_nt(
+    f2 *a, f2 *b, int n, int log2n, long estride, long gstride, int nlines,
+    int nchunks, int ngroups, int valid_a, int valid_b, float scale,
+    const f2 *twg) {
+  zf_body<NE, false, false, true>(a, b, n, log2n, estride, gstride, nlines,
+                                  nchunks, ngroups, valid_a, valid_b, scale,
+                                  twg);
 }
 
 /* Inverse x pass (C2R, packed): per-line Hermitian half-line -> the real
@@ -2816,14 +2836,17 @@ static int stitch_phaseA(bs_ctx *c, bs_slot *sl, const bs_pair_desc &pd,
      * hardware: 0.734 vs 0.842 ms at 512^3 — 4 blocks/CU beats the
      * B-register prefetch at 2 blocks/CU) */
     static const bool z_r8 = getenv("BS_Z_R8") != nullptr;
+    static const bool z_nt = getenv("BS_Z_NT") != nullptr;
     if (Pz <= 128)
       zf(k_fft_z_fused<1>);
     else if (Pz == 256)
       zf(k_fft_z_fused<2>);
     else if (Pz == 512)
-      z_r8 ? zf(k_fft_z_fused_r8<4>) : zf(k_fft_z_fused_np<4>);
+      z_r8 ? zf(k_fft_z_fused_r8<4>)
+           : z_nt ? zf(k_fft_z_fused_nt<4>) : zf(k_fft_z_fused_np<4>);
     else
-      z_r8 ? zf(k_fft_z_fused_r8<8>) : zf(k_fft_z_fused_np<8>);
+      z_r8 ? zf(k_fft_z_fused_r8<8>)
+           : z_nt ? zf(k_fft_z_fused_nt<8>) : zf(k_fft_z_fused_np<8>);
   }
   {
     size_t lds = ((Py / 2) + (size_t)LPB_S * Py) * sizeof(f2);
