@@ -46,13 +46,17 @@ def main():
     ap.add_argument("--grid", type=int, nargs=3, default=[4, 4, 2])
     ap.add_argument("--out", default="/tmp/bs_e2e")
     ap.add_argument("--overlap", type=float, default=0.1)
+    ap.add_argument("--storage", default="N5", choices=["N5", "ZARR"],
+                    help="fused container format (ZARR = OME-ZARR, the "
+                         "configs[4] surface)")
     args = ap.parse_args()
     size, (gx, gy, gz) = args.size, args.grid
     step = int(size * (1.0 - args.overlap))
     os.makedirs(args.out, exist_ok=True)
     n5 = os.path.join(args.out, "input.n5")
     xml = os.path.join(args.out, "dataset.xml")
-    fused = os.path.join(args.out, "fused.n5")
+    fused = os.path.join(
+        args.out, "fused.zarr" if args.storage == "ZARR" else "fused.n5")
 
     world = [step * (g - 1) + size for g in (gx, gy, gz)]
     rng = np.random.default_rng(42)
@@ -105,7 +109,7 @@ def main():
     t_sv, _ = run([os.path.join(BIN, "solver"), "-x", xml])
     print(f"solver: {t_sv:.1f}s", flush=True)
     t_cc, _ = run([os.path.join(BIN, "create-fusion-container"),
-                   "-x", xml, "-s", "N5", "-o", fused,
+                   "-x", xml, "-s", args.storage, "-o", fused,
                    "--blockSize", "128,128,128", "-d", "UINT16",
                    "--minIntensity", "0", "--maxIntensity", "65535"])
     t_fu, _ = run([os.path.join(BIN, "affine-fusion"), "-o", fused,
@@ -130,7 +134,8 @@ def main():
         nvox *= d
     line = {
         "metric": "configs[2] e2e wall seconds (stitch+solve+container+fusion)",
-        "grid": f"{gx}x{gy}x{gz} x {size}^3 uint16, zstd N5",
+        "grid": (f"{gx}x{gy}x{gz} x {size}^3 uint16, zstd "
+                 + ("OME-ZARR" if args.storage == "ZARR" else "N5")),
         "stages_s": {"input_gen_untimed": round(t_gen, 1),
                      "stitching": round(t_st, 1),
                      "solver": round(t_sv, 1),
