@@ -1268,3 +1268,21 @@ def test_cli_resave_zstd_codec(tmp_path):
     s0, attrs = n5util.read_dataset(out, "setup0/timepoint0/s0")
     assert attrs["compression"]["type"] == "zstd"
     assert np.array_equal(s0, a)
+
+
+@pytest.mark.gpu
+def test_n5_missing_chunk_reads_zero(tmp_path):
+    """N5 semantics: an absent chunk file is all-zeros (the reference's
+    datasets are sparse at tile borders). Delete one chunk of a view,
+    resave it, and the round trip shows exactly that chunk zeroed."""
+    xml, n5, err, (a, b) = make_grid_dataset(str(tmp_path))
+    os.remove(os.path.join(n5, "setup0", "timepoint0", "s0", "1", "0", "0"))
+    out = os.path.join(str(tmp_path), "resaved.n5")
+    xo = os.path.join(str(tmp_path), "resaved.xml")
+    r = run([os.path.join(BIN, "resave"), "-x", xml, "-o", out, "-xo", xo,
+             "--blockSize", "32,32,32", "--downsamplings", "1,1,1"])
+    assert r.returncode == 0, r.stderr + r.stdout
+    s0, _ = n5util.read_dataset(out, "setup0/timepoint0/s0")
+    want = a.copy()
+    want[0:32, 0:32, 32:64] = 0  # chunk grid index (x=1,y=0,z=0)
+    assert np.array_equal(s0, want)
