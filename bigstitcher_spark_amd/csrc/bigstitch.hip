@@ -1523,9 +1523,10 @@ __device__ __forceinline__ u64 wave_sum_u64(u64 v) {
  * sums are order-independent -> bit-exact [PIN-R]. The per-candidate
  * row walk uses the adaptive row width (rows here are single y-lines,
  * no division anywhere). */
-__global__ __launch_bounds__(256) void k_rtest(bs_region a, bs_region b,
-                                                const bs_cand *cands,
-                                                int nc, u64 *sums) {
+template <bool R4>
+__device__ __forceinline__ void rtest_body(bs_region a, bs_region b,
+                                           const bs_cand *cands,
+                                           int nc, u64 *sums) {
   __shared__ u64 ws[4][5];
   const int tid = threadIdx.x;
   const int z = blockIdx.x;
@@ -1587,6 +1588,38 @@ __global__ __launch_bounds__(256) void k_rtest(bs_region a, bs_region b,
         }
       };
       int r = r00;
+      if (R4) /* 4 rows in flight: deeper MLP for the latency-bound
+                 dual-stream scan; u64 sums keep it bit-exact [PIN-R] */
+        for (; r + 3 * rstride < c.ny; r += 4 * rstride) {
+          if (h && lx == 0)
+#pragma unroll
+            for (int q = 0; q < 4; ++q)
+              add1(ar[q * astep], br[q * bstep]);
+          if (tail && lx == rw - 1)
+#pragma unroll
+            for (int q = 0; q < 4; ++q)
+              add1(ar[q * astep + c.nx - 1], br[q * bstep + c.nx - 1]);
+          for (int p = lx; p < np2; p += rw) {
+            unsigned av[4], bv[4];
+#pragma unroll
+            for (int q = 0; q < 4; ++q) {
+              av[q] = ((const unsigned *)(ar + q * astep + h))[p];
+              bv[q] = ((const unsigned *)(br + q * bstep + h))[p];
+            }
+#pragma unroll
+            for (int q = 0; q < 4; ++q) {
+              const unsigned al = av[q] & 0xFFFFu, ah = av[q] >> 16;
+              const unsigned bl = bv[q] & 0xFFFFu, bh = bv[q] >> 16;
+              pa += al + ah;
+              pb += bl + bh;
+              paa += (u64)(al * al) + (u64)(ah * ah);
+              pbb += (u64)(bl * bl) + (u64)(bh * bh);
+              pab += (u64)(al * bl) + (u64)(ah * bh);
+            }
+          }
+          ar += 4 * astep;
+          br += 4 * bstep;
+        }
       for (; r + rstride < c.ny; r += 2 * rstride) {
         if (h && lx == 0) {
           add1(ar[0], br[0]);
@@ -1655,6 +1688,19 @@ __global__ __launch_bounds__(256) void k_rtest(bs_region a, bs_region b,
     }
     __syncthreads(); /* ws reused next candidate */
   }
+}
+
+__global__ __launch_bounds__(256) void k_rtest(bs_region a, bs_region b,
+                                               const bs_cand *cands,
+                                               int nc, u64 *sums) {
+  rtest_body<false>(a, b, cands, nc, sums);
+}
+
+/* 4-rows-in-flight A/B (BS_CORR_R4) */
+__global__ __launch_bounds__(256) void k_rtest_r4(bs_region a, bs_region b,
+                                                  const bs_cand *cands,
+                                                  int nc, u64 *sums) {
+  rtest_body<true>(a, b, cands, nc, sums);
 }
 
 /* ---------------------------------------------------------------- fusion */
@@ -3004,7 +3050,9 @@ static int stitch_phaseB(bs_ctx *c, bs_slot *sl,
         const char *e = getenv("BS_CORR_GY");
         return e ? atoi(e) : 8; /* measured: 8 > 4 > 2 >> 1 at 512^3 */
       }();
-      hipLaunchKernelGGL(k_rtest, dim3((unsigned)maxz, corr_gy), dim3(256),
+      static const bool corr_r4 = getenv("BS_CORR_R4") != nullptr;
+      hipLaunchKernelGGL(corr_r4 ? k_rtest_r4 : k_rtest,
+                         dim3((unsigned)maxz, corr_gy), dim3(256),
                          0, sl->stream, sl->reg[0], sl->reg[1], sl->dcands,
                          (int)gc.size(), sl->dsums);
     }
