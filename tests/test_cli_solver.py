@@ -241,3 +241,21 @@ def test_solver_correlation_weighted(tmp_path):
     assert abs(dw[0] - du[0]) > 0.1  # the weight matters for this graph
     assert np.allclose(t[1], [40.0 + dw[0], 0, 0], atol=1e-9)
     assert np.allclose(t[2], [80.0 + dw[1], 0, 0], atol=1e-9)
+
+
+def test_solver_fixed_views_flag(tmp_path):
+    """--fixedViews tp,setup: anchoring view 1 instead of the default
+    first view moves view 0 by the opposite amount."""
+    xml = os.path.join(str(tmp_path), "dataset.xml")
+    e = (2.5, -1.5, 1.0)
+    hash_ok = 3.0 + (3.0 + 40.0)
+    write_xml_with_links(
+        xml,
+        [dict(a="0,0", b="0,1", ws=(-e[0], -e[1], -e[2]), hash=hash_ok)],
+    )
+    r = run([os.path.join(BIN, "solver"), "-x", xml,
+             "--fixedViews", "0,1"])
+    assert r.returncode == 0, r.stderr + r.stdout
+    t = model_translations(xml)
+    assert np.allclose(t[1], [40, 0, 0], atol=1e-9)  # anchored
+    assert np.allclose(t[0], [-e[0], -e[1], -e[2]], atol=1e-6), t[0]
