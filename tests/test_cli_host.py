@@ -1286,3 +1286,51 @@ def test_n5_missing_chunk_reads_zero(tmp_path):
     want = a.copy()
     want[0:32, 0:32, 32:64] = 0  # chunk grid index (x=1,y=0,z=0)
     assert np.array_equal(s0, want)
+
+
+@pytest.mark.gpu
+def test_cli_stitching_filters_and_peak_flags(tmp_path):
+    """FilteredStitchingResults sub-flags (SparkPairwiseStitching.java:
+    76-107): a tight --maxShiftX drops the link; --maxR below the pair's
+    r drops it too; --disableSubpixelResolution --peaksToCheck 3 keeps
+    an integer-valued shift near the truth."""
+    err = (2.5, -1.5, 1.0)
+    xml, n5, _err, _ = make_grid_dataset(str(tmp_path), err=err)
+    r = run([os.path.join(BIN, "stitching"), "-x", xml, "-ds", "1,1,1",
+             "--minOverlapRatio", "0.05", "--maxShiftX", "1.0"])
+    assert r.returncode == 0, r.stderr + r.stdout
+    assert not ET.parse(xml).getroot().findall(".//PairwiseResult")
+    r = run([os.path.join(BIN, "stitching"), "-x", xml, "-ds", "1,1,1",
+             "--minOverlapRatio", "0.05", "--maxR", "0.2"])
+    assert r.returncode == 0, r.stderr + r.stdout
+    assert not ET.parse(xml).getroot().findall(".//PairwiseResult")
+    r = run([os.path.join(BIN, "stitching"), "-x", xml, "-ds", "1,1,1",
+             "--minOverlapRatio", "0.05", "--peaksToCheck", "3",
+             "--disableSubpixelResolution"])
+    assert r.returncode == 0, r.stderr + r.stdout
+    prs = ET.parse(xml).getroot().findall(".//PairwiseResult")
+    assert len(prs) == 1
+    m = [float(x) for x in prs[0].find("Matrix").text.split()]
+    got = (m[3], m[7], m[11])
+    for g, w in zip(got, (-err[0], -err[1], -err[2])):
+        assert g == round(g)  # integer (no subpixel fit)
+        assert abs(g - w) <= 1.0, (got, err)
+
+
+@pytest.mark.gpu
+def test_cli_stitching_channel_combine_flag(tmp_path):
+    """--channelCombine PICK_BRIGHTEST (non-default for the channel
+    axis): the brighter channel is selected per group, and the link is
+    still found on the selected members."""
+    err = (2.5, -1.5, 1.0)
+    xml, n5, vols, _ = make_grouped_dataset(str(tmp_path), err=err)
+    r = run([os.path.join(BIN, "stitching"), "-x", xml, "-ds", "1,1,1",
+             "--minR", "0.3", "--channelCombine", "PICK_BRIGHTEST"])
+    assert r.returncode == 0, r.stderr + r.stdout
+    prs = ET.parse(xml).getroot().findall(".//PairwiseResult")
+    assert len(prs) == 1
+    assert prs[0].find("ViewIdsA").text == "0,0;0,1"
+    m = [float(x) for x in prs[0].find("Matrix").text.split()]
+    want = (-err[0], -err[1], -err[2])
+    for g, w in zip((m[3], m[7], m[11]), want):
+        assert abs(g - w) < 1.0, (m, want)
