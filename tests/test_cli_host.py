@@ -1334,3 +1334,23 @@ def test_cli_stitching_channel_combine_flag(tmp_path):
     want = (-err[0], -err[1], -err[2])
     for g, w in zip((m[3], m[7], m[11]), want):
         assert abs(g - w) < 1.0, (m, want)
+
+
+def test_container_cli_bbox_and_compression_level(tmp_path):
+    """--bbMin/--bbMax explicit output bounding box and
+    -cl/--compressionLevel (CreateFusionContainer's compression-level
+    option) land in the container attributes and codec config."""
+    xml, n5, _err, _ = make_grid_dataset(str(tmp_path))
+    out = os.path.join(str(tmp_path), "fused.n5")
+    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml,
+             "-s", "N5", "-o", out, "--blockSize", "32,32,32",
+             "-d", "UINT16", "-c", "Zstandard", "-cl", "7",
+             "--bbMin", "8,4,2", "--bbMax", "71,59,49"])
+    assert r.returncode == 0, r.stderr
+    attrs = n5util.root_attrs(out)["Bigstitcher-Spark"]
+    assert attrs["Boundingbox_min"] == [8, 4, 2]
+    assert attrs["Boundingbox_max"] == [71, 59, 49]
+    assert attrs["MultiResolutionInfos"][0][0]["dimensions"] == [64, 56, 48]
+    _, dattrs = n5util.read_dataset(out, "ch0tp0/s0")
+    assert dattrs["compression"]["type"] == "zstd"
+    assert dattrs["compression"]["level"] == 7
