@@ -1354,3 +1354,32 @@ def test_container_cli_bbox_and_compression_level(tmp_path):
     _, dattrs = n5util.read_dataset(out, "ch0tp0/s0")
     assert dattrs["compression"]["type"] == "zstd"
     assert dattrs["compression"]["level"] == 7
+
+
+@pytest.mark.gpu
+def test_cli_fusion_blending_border(tmp_path):
+    """--blendingBorder b shrinks the blend support by b voxels per
+    face before the ramp ([PIN-BLEND]; reference defaultBlendingBorder)
+    — parity vs the oracle with border=(b,b,b)."""
+    xml, n5, err, (a, b) = make_grid_dataset(str(tmp_path))
+    out = os.path.join(str(tmp_path), "fused.n5")
+    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml,
+             "-s", "N5", "-o", out, "--blockSize", "32,32,32",
+             "--dataType", "FLOAT32"])
+    assert r.returncode == 0, r.stderr
+    r = run([os.path.join(BIN, "affine-fusion"), "-o", out,
+             "-f", "AVG_BLEND", "--blendingRange", "8",
+             "--blendingBorder", "2"])
+    assert r.returncode == 0, r.stderr + r.stdout
+    fused, _ = n5util.read_dataset(out, "ch0tp0/s0")
+    ident = np.hstack([np.eye(3), np.zeros((3, 1))])
+    affB = ident.copy()
+    affB[0, 3] = 40.0
+    views = [
+        dict(data=a, affine=ident, border=(2, 2, 2), range=(8, 8, 8)),
+        dict(data=b, affine=affB, border=(2, 2, 2), range=(8, 8, 8)),
+    ]
+    ref = of.fuse_block(views, (0, 0, 0), (104, 64, 64),
+                        of.FUSION_AVG_BLEND, out_dtype=np.float32)
+    denom = np.maximum(np.abs(ref), 1.0)
+    assert np.max(np.abs(fused - ref) / denom) < 1e-4
