@@ -86,9 +86,13 @@ bool SpimData::load(const std::string &path, std::string *err) {
   auto loader = seq->child("ImageLoader");
   if (loader) {
     auto n5 = loader->child("n5");
-    if (n5) {
-      bool rel = !n5->attrs.count("type") || n5->attrs["type"] == "relative";
-      n5_path = rel ? base_dir + "/" + n5->text : n5->text;
+    auto zr = loader->child("zarr"); /* [PIN-OMEZARR-BDV] bs_imgio.h */
+    auto src = n5 ? n5 : zr;
+    if (src) {
+      bool rel =
+          !src->attrs.count("type") || src->attrs["type"] == "relative";
+      n5_path = rel ? base_dir + "/" + src->text : src->text;
+      zarr_loader = (src == zr);
     }
   }
   auto vss = seq->child("ViewSetups");

@@ -52,7 +52,10 @@ struct StitchEntry {
 struct SpimData {
   std::string xml_path, base_dir;
   bsx::NodePtr root;
-  std::string n5_path; /* resolved image container (bdv.n5 loader) */
+  std::string n5_path; /* resolved image container (bdv.n5 or zarr) */
+  bool zarr_loader = false; /* ImageLoader holds <zarr> (the OME-ZARR
+    BDV layout `resave` writes by default; [PIN-OMEZARR-BDV],
+    bs_imgio.h) instead of <n5> */
   std::vector<ViewSetup> setups;
   std::vector<int> timepoints;
   /* (tp,setup) -> concatenated 3x4 model, row-major, world = M x local */

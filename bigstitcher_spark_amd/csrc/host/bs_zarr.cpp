@@ -1,6 +1,8 @@
 #include "bs_zarr.h"
 
+#include <atomic>
 #include <cstring>
+#include <thread>
 
 #include "bs_n5.h" /* bsio helpers */
 
@@ -26,6 +28,51 @@ bool Container::create() {
   std::string existing;
   if (read_file(root_ + "/.zgroup", &existing)) return true;
   return write_file(root_ + "/.zgroup", bsj::dump(g));
+}
+
+bool Container::create_group(const std::string &path) {
+  /* every path component needs its own .zgroup for zarr validity */
+  std::string cur = root_;
+  std::string rest = path + "/";
+  std::string seg;
+  for (char c : rest) {
+    if (c == '/') {
+      if (!seg.empty()) {
+        cur += "/" + seg;
+        if (!mkdirs(cur)) return false;
+        std::string existing;
+        if (!read_file(cur + "/.zgroup", &existing)) {
+          auto g = bsj::Value::mkobj();
+          g->obj["zarr_format"] = bsj::Value::mkint(2);
+          if (!write_file(cur + "/.zgroup", bsj::dump(g))) return false;
+        }
+        seg.clear();
+      }
+    } else {
+      seg += c;
+    }
+  }
+  return true;
+}
+
+bool Container::set_group_attr(const std::string &group,
+                               const std::string &key, bsj::ValuePtr v) {
+  std::string p = root_ + "/" + group + "/.zattrs";
+  std::string text;
+  bsj::ValuePtr a;
+  if (read_file(p, &text)) a = bsj::parse(text);
+  if (!a || a->type != bsj::Value::OBJ) a = bsj::Value::mkobj();
+  bsj::set_path(a, key, v);
+  return write_file(p, bsj::dump(a));
+}
+
+bsj::ValuePtr Container::get_group_attr(const std::string &group,
+                                        const std::string &key) const {
+  std::string text;
+  if (!read_file(root_ + "/" + group + "/.zattrs", &text)) return nullptr;
+  auto a = bsj::parse(text);
+  if (!a) return nullptr;
+  return bsj::get_path(a, key);
 }
 
 bool Container::set_root_attr(const std::string &key, bsj::ValuePtr v) {
@@ -150,6 +197,106 @@ bool Container::write_chunk(const std::string &name, const ArrayAttrs &a,
     key += std::to_string(grid_pos[d]);
   }
   return write_file(root_ + "/" + name + "/" + key, body);
+}
+
+bool Container::read_chunk(const std::string &name, const ArrayAttrs &a,
+                           const std::vector<long long> &grid_pos,
+                           void *data) const {
+  size_t esz = dtype_size(a.dtype);
+  size_t nd = a.shape.size();
+  if (!esz || grid_pos.size() != nd) return false;
+  size_t full = esz;
+  for (int c : a.chunks) full *= (size_t)c;
+  std::string key;
+  for (size_t d = 0; d < nd; ++d) {
+    if (d) key += ".";
+    key += std::to_string(grid_pos[d]);
+  }
+  std::string raw;
+  if (!read_file(root_ + "/" + name + "/" + key, &raw)) {
+    /* "/"-separated chunk keys are also legal zarr v2 */
+    std::string k2;
+    for (size_t d = 0; d < nd; ++d) {
+      if (d) k2 += "/";
+      k2 += std::to_string(grid_pos[d]);
+    }
+    if (!read_file(root_ + "/" + name + "/" + k2, &raw)) {
+      memset(data, 0, full); /* missing chunk = fill_value 0 */
+      return true;
+    }
+  }
+  std::string payload;
+  if (a.codec == "gzip") {
+    if (!bsio::gzip_inflate((const unsigned char *)raw.data(), raw.size(),
+                            &payload, full))
+      return false;
+  } else if (a.codec == "zstd") {
+    if (!bsio::zstd_decompress((const unsigned char *)raw.data(), raw.size(),
+                               &payload, full))
+      return false;
+  } else {
+    payload = raw;
+  }
+  if (payload.size() != full) return false;
+  memcpy(data, payload.data(), full);
+  return true;
+}
+
+bool Container::read_volume_u16(const std::string &name,
+                                std::vector<uint16_t> *out,
+                                std::vector<long long> *dims_xyz) const {
+  ArrayAttrs a;
+  if (!get_array_attrs(name, &a) || a.dtype != "<u2" || a.shape.size() < 3)
+    return false;
+  const size_t nd = a.shape.size();
+  for (size_t d = 0; d + 3 < nd; ++d)
+    if (a.shape[d] < 1 || a.chunks[d] < 1) return false;
+  const long long nz = a.shape[nd - 3], ny = a.shape[nd - 2],
+                  nx = a.shape[nd - 1];
+  const int bz = a.chunks[nd - 3], by = a.chunks[nd - 2],
+            bx = a.chunks[nd - 1];
+  *dims_xyz = {nx, ny, nz};
+  out->assign((size_t)(nx * ny * nz), 0);
+  const long long ngx = (nx + bx - 1) / bx, ngy = (ny + by - 1) / by,
+                  ngz = (nz + bz - 1) / bz;
+  const long long nchunks = ngx * ngy * ngz;
+  const int NW = (int)std::min<long long>(
+      nchunks, std::max(1u, std::thread::hardware_concurrency() / 2));
+  std::atomic<long long> next(0);
+  std::atomic<bool> failed(false);
+  auto worker = [&]() {
+    size_t full = 1;
+    for (int c : a.chunks) full *= (size_t)c;
+    std::vector<uint16_t> blk(full);
+    std::vector<long long> gp(nd, 0);
+    for (;;) {
+      long long i = next.fetch_add(1);
+      if (i >= nchunks || failed.load()) return;
+      const long long gx = i % ngx, gy = (i / ngx) % ngy,
+                      gz = i / (ngx * ngy);
+      gp[nd - 3] = gz;
+      gp[nd - 2] = gy;
+      gp[nd - 1] = gx;
+      if (!read_chunk(name, a, gp, blk.data())) {
+        failed.store(true);
+        return;
+      }
+      const int cx = (int)std::min<long long>(bx, nx - gx * bx);
+      const int cy = (int)std::min<long long>(by, ny - gy * by);
+      const int cz = (int)std::min<long long>(bz, nz - gz * bz);
+      for (int z = 0; z < cz; ++z)
+        for (int y = 0; y < cy; ++y) {
+          long long dst =
+              ((gz * bz + z) * ny + (gy * by + y)) * nx + gx * bx;
+          memcpy(&(*out)[dst], &blk[((size_t)z * by + y) * bx],
+                 (size_t)cx * 2);
+        }
+    }
+  };
+  std::vector<std::thread> ws;
+  for (int w = 0; w < NW; ++w) ws.emplace_back(worker);
+  for (auto &w : ws) w.join();
+  return !failed.load();
 }
 
 }  // namespace bszarr

@@ -39,8 +39,14 @@ class Container {
  public:
   explicit Container(const std::string &root) : root_(root) {}
   bool create();                                    /* .zgroup */
+  bool create_group(const std::string &path);       /* nested .zgroup s */
   bool set_root_attr(const std::string &key, bsj::ValuePtr v); /* .zattrs */
   bsj::ValuePtr get_root_attr(const std::string &key) const;
+  /* group-level .zattrs (e.g. per-view multiscales in the BDV layout) */
+  bool set_group_attr(const std::string &group, const std::string &key,
+                      bsj::ValuePtr v);
+  bsj::ValuePtr get_group_attr(const std::string &group,
+                               const std::string &key) const;
   bool create_array(const std::string &name, const ArrayAttrs &a);
   bool get_array_attrs(const std::string &name, ArrayAttrs *out) const;
   /* data: the chunk's full buffer in C order; clipped dims give the
@@ -48,6 +54,16 @@ class Container {
   bool write_chunk(const std::string &name, const ArrayAttrs &a,
                    const std::vector<long long> &grid_pos, const void *data,
                    const std::vector<int> &clipped);
+  /* data: FULL-size chunk buffer (zarr chunks are never clipped); a
+   * missing chunk file fills with fill_value 0. */
+  bool read_chunk(const std::string &name, const ArrayAttrs &a,
+                  const std::vector<long long> &grid_pos, void *data) const;
+  /* convenience mirror of bsn5::Container::read_volume_u16 for the
+   * bdv-omezarr input path: read a "<u2" array's trailing 3 dims
+   * (leading dims — t,c for the 5-D BDV layout — taken at index 0)
+   * into a contiguous x-fastest buffer; dims_xyz = {x,y,z}. */
+  bool read_volume_u16(const std::string &name, std::vector<uint16_t> *out,
+                       std::vector<long long> *dims_xyz) const;
   const std::string &root() const { return root_; }
 
  private:

@@ -22,6 +22,7 @@
 
 #include "../../../include/bigstitch.h"
 #include "bs_cli_util.h"
+#include "bs_imgio.h"
 #include "bs_n5.h"
 #include "bs_mip.h"
 #include "bs_zarr.h"
@@ -243,7 +244,7 @@ int main(int argc, char **argv) {
     return 1;
   }
   std::set<std::pair<int, int>> selset(selected.begin(), selected.end());
-  bsn5::Container in_n5(sd.n5_path);
+  bsimg::Input in_n5(sd);
   bs_ctx *ctx = nullptr;
   if (bs_ctx_create(&ctx, (int)args.getl("device", 0)) != BS_OK) {
     fprintf(stderr, "error: %s\n", bs_last_error(nullptr));
@@ -299,7 +300,7 @@ int main(int argc, char **argv) {
         for (int i = 8; i < 12; ++i) fv.affine[i] /= anisoF;
       /* multi-resolution input: ViewUtil forBestResolution rule +
        * [PIN-MIP] mipmap fold (see bs_mip.h / bs_cli_util.h) */
-      auto levels = bscli::read_levels(in_n5, s.id, tp);
+      auto levels = in_n5.read_levels(s.id, tp);
       bscli::M34 model;
       for (int i = 0; i < 12; ++i) model[i] = fv.affine[i];
       int lvi = bscli::pick_level_for_transform(model, levels);
@@ -468,10 +469,8 @@ int main(int argc, char **argv) {
         if (!resident.count(pl.setup)) {
           std::vector<uint16_t> vox;
           std::vector<long long> vdims;
-          if (!in_n5.read_volume_u16(
-                  bssd::SpimData::image_dataset(pl.setup, pl.tp,
-                                                pl.level),
-                  &vox, &vdims)) {
+          if (!in_n5.read_volume_u16(pl.setup, pl.tp, pl.level,
+                                     &vox, &vdims)) {
             fprintf(stderr, "cannot read view tp=%d setup=%d\n", pl.tp,
                     pl.setup);
             return 1;

@@ -1,14 +1,23 @@
 /* `resave` — drop-in for the reference's SparkResaveN5 (reference
  * SparkResaveN5.java:189-207, 298-383: per-view s0 re-blocking via
  * N5ApiTools.resaveS0Block + pyramid via writeDownsampledBlock).
- * SURVEY.md §8(f) row 2. Round-1 scope: input = the dataset's bdv.n5
+ * SURVEY.md §8(f) row 2. Input = the dataset's bdv.n5 or OME-ZARR BDV
  * container (the reference's TIFF/CZI ingest readers are out of scope —
- * no image libraries in this environment); output = a new bdv.n5
- * container with per-(setup,timepoint) multi-resolution pyramids
- * "setup{s}/timepoint{t}/s{l}" plus a rewritten dataset.xml pointing at
- * it. The pyramid levels are computed on the GPU (the K8 box-mean
- * kernel via bs_fuse_volume with the single identity view, which is an
- * exact uint16 round-trip at minI=0/maxI=65535). */
+ * no image libraries in this environment); output = OME-ZARR by
+ * default with --N5 opting into bdv.n5, exactly the reference's
+ * polarity (SparkResaveN5.java:85 "--N5: Export as N5 (default:
+ * OMEZARR)"). N5 levels are "s{l}", ZARR levels "{l}" (pinned by the
+ * reference's own log lines :331/:347 `useN5 ? "N5 s0" : "OME-ZARR
+ * 0"`); the OME-ZARR group layout "setup{s}/timepoint{t}" +
+ * multiscales restates un-vendored N5ApiTools.setupBdvDatasetsOMEZARR
+ * ([PIN-OMEZARR-BDV], bs_imgio.h). Flag surface mirrors
+ * SparkResaveN5.java:80-104: -xo default = overwrite input keeping a
+ * "~1" backup, -o default = '<folder of xml>/dataset.n5|.ome.zarr',
+ * -c default Zstandard, repeated -ds/--downsampling with an
+ * automatically computed default ladder ([PIN-MULTIRES]), --blockSize
+ * default 128,128,64. The pyramid levels are computed on the GPU (the
+ * K8 box-mean kernel via bs_fuse_volume with the single identity view,
+ * an exact uint16 round-trip at minI=0/maxI=65535). */
 #include <atomic>
 #include <cstdio>
 #include <cstring>
@@ -17,55 +26,116 @@
 
 #include "../../../include/bigstitch.h"
 #include "bs_cli_util.h"
+#include "bs_imgio.h"
 #include "bs_n5.h"
 #include "bs_spimdata.h"
+#include "bs_zarr.h"
 
 int main(int argc, char **argv) {
   bscli::Args args;
-  std::map<std::string, std::string> alias = {{"-x", "--xml"},
-                                              {"-o", "--n5Path"},
-                                              {"-c", "--compression"},
-                                              {"-xo", "--xmlOut"}};
-  if (!args.parse(argc, argv, alias, {}) || !args.has("xml") ||
-      !args.has("n5Path")) {
+  std::map<std::string, std::string> alias = {
+      {"-x", "--xml"},           {"-o", "--n5Path"},
+      {"-c", "--compression"},   {"-cl", "--compressionLevel"},
+      {"-xo", "--xmlout"},       {"--xmlOut", "--xmlout"},
+      {"-ds", "--downsampling"}};
+  if (!args.parse(argc, argv, alias, {"N5"}) || !args.has("xml")) {
     fprintf(stderr,
-            "usage: resave -x dataset.xml -o out.n5 [-xo out.xml] "
-            "[--blockSize 128,128,64] [-c Gzip|Zstandard|Raw] "
-            "[--downsamplings \"1,1,1;2,2,1;4,4,2\"] [--device N]\n");
+            "usage: resave -x dataset.xml [-o out.n5|out.ome.zarr] "
+            "[-xo out.xml] [--N5] [--blockSize 128,128,64] "
+            "[-c Gzip|Zstandard|Raw] [-cl N] "
+            "[-ds 1,1,1 -ds 2,2,1 | -ds \"1,1,1;2,2,1\"] [--device N]\n");
     return 2;
   }
+  const bool useN5 = args.has("N5");
   bssd::SpimData sd;
   std::string err;
   if (!sd.load(args.get("xml"), &err)) {
     fprintf(stderr, "error: %s\n", err.c_str());
     return 1;
   }
+  /* -xo default: overwrite the input XML, keep a "~1" backup
+   * (SparkResaveN5.java:80) */
+  std::string xo = args.get("xmlout");
+  if (xo.empty()) xo = args.get("xml");
+  /* -o default: '<folder of the xml>/dataset.n5' | 'dataset.ome.zarr'
+   * (SparkResaveN5.java:104,166) */
+  std::string outPath = args.get("n5Path");
+  if (outPath.empty()) {
+    std::string dir = xo;
+    size_t sl = dir.find_last_of('/');
+    dir = sl == std::string::npos ? "." : dir.substr(0, sl);
+    outPath = dir + (useN5 ? "/dataset.n5" : "/dataset.ome.zarr");
+  }
   auto bs = bscli::parse_ints(args.get("blockSize", "128,128,64"));
+  if (bs.size() != 3) {
+    fprintf(stderr, "bad --blockSize\n");
+    return 2;
+  }
+  const std::string cn = args.get("compression", "Zstandard");
+  if (cn != "Gzip" && cn != "Zstandard" && cn != "Raw") {
+    fprintf(stderr, "unsupported -c %s (Gzip|Zstandard|Raw)\n", cn.c_str());
+    return 2;
+  }
+  const std::string codec =
+      cn == "Zstandard" ? "zstd" : cn == "Raw" ? "raw" : "gzip";
+  const int clevel = (int)args.getl("compressionLevel", 0);
+
   std::vector<std::array<int, 3>> ladder;
   {
-    std::string spec = args.get("downsamplings", "1,1,1;2,2,1");
-    std::string cur;
-    for (char c : spec + ";") {
-      if (c == ';') {
-        auto f = bscli::parse_ints(cur);
-        if (f.size() == 3)
-          ladder.push_back({(int)f[0], (int)f[1], (int)f[2]});
-        cur.clear();
-      } else {
-        cur += c;
+    std::vector<std::string> specs = args.getall("downsampling");
+    {
+      std::string legacy = args.get("downsamplings");
+      if (!legacy.empty()) specs.push_back(legacy);
+    }
+    std::string joined;
+    for (auto &sp : specs) joined += (joined.empty() ? "" : ";") + sp;
+    if (!joined.empty()) {
+      std::string cur;
+      for (char c : joined + ";") {
+        if (c == ';') {
+          auto f = bscli::parse_ints(cur);
+          if (f.size() == 3)
+            ladder.push_back({(int)f[0], (int)f[1], (int)f[2]});
+          cur.clear();
+        } else {
+          cur += c;
+        }
+      }
+      if (ladder.empty() || ladder[0] != std::array<int, 3>{1, 1, 1}) {
+        fprintf(stderr, "--downsampling must start with 1,1,1\n");
+        return 2;
+      }
+    } else {
+      /* default: automatically computed (SparkResaveN5.java:94) — the
+       * [PIN-MULTIRES] estimate over the first view's dims: halve every
+       * axis whose current extent exceeds its block size until all fit
+       * one block or 8 levels */
+      long long d0[3] = {64, 64, 64};
+      if (!sd.setups.empty())
+        for (int d = 0; d < 3; ++d) d0[d] = sd.setups[0].dims[d];
+      long long f[3] = {1, 1, 1};
+      ladder.push_back({1, 1, 1});
+      for (int l = 0; l < 7; ++l) {
+        bool any = false;
+        long long nf[3];
+        for (int d = 0; d < 3; ++d) {
+          long long ext = (d0[d] + f[d] - 1) / f[d];
+          nf[d] = ext > bs[d] ? f[d] * 2 : f[d];
+          any |= nf[d] != f[d];
+        }
+        if (!any) break;
+        for (int d = 0; d < 3; ++d) f[d] = nf[d];
+        ladder.push_back({(int)f[0], (int)f[1], (int)f[2]});
       }
     }
   }
-  if (ladder.empty() || ladder[0] != std::array<int, 3>{1, 1, 1}) {
-    fprintf(stderr, "--downsamplings must start with 1,1,1\n");
-    return 2;
-  }
   int nlevels = (int)ladder.size();
 
-  bsn5::Container in_n5(sd.n5_path);
-  bsn5::Container out_n5(args.get("n5Path"));
-  if (!out_n5.create()) {
-    fprintf(stderr, "cannot create %s\n", args.get("n5Path").c_str());
+  bsimg::Input in(sd);
+  bsn5::Container out_n5(outPath);
+  bszarr::Container out_zr(outPath);
+  if (useN5 ? !out_n5.create() : !out_zr.create()) {
+    fprintf(stderr, "cannot create %s\n", outPath.c_str());
     return 1;
   }
   bs_ctx *ctx = nullptr;
@@ -81,8 +151,7 @@ int main(int argc, char **argv) {
     for (auto &s : sd.setups) {
       std::vector<uint16_t> vox;
       std::vector<long long> dims;
-      if (!in_n5.read_volume_u16(bssd::SpimData::image_dataset(s.id, tp),
-                                 &vox, &dims)) {
+      if (!in.read_volume_u16(s.id, tp, 0, &vox, &dims)) {
         fprintf(stderr, "cannot read view tp=%d setup=%d\n", tp, s.id);
         return 1;
       }
@@ -119,27 +188,66 @@ int main(int argc, char **argv) {
         return 1;
       }
       bs_view_release(ctx, s.id);
-      for (int l = 0; l < nlevels; ++l) {
-        std::string dsn = bssd::SpimData::image_dataset(s.id, tp, l);
-        bsn5::DatasetAttrs da;
-        da.dims = {ldims[l * 3], ldims[l * 3 + 1], ldims[l * 3 + 2]};
-        da.block = {(int)bs[0], (int)bs[1], (int)bs[2]};
-        da.dtype = "uint16";
-        {
-          std::string cn = args.get("compression", "Gzip");
-          da.compression = cn == "Zstandard" ? "zstd"
-                           : cn == "Raw" ? "raw" : "gzip";
-        }
-        if (!out_n5.create_dataset(dsn, da)) {
-          fprintf(stderr, "cannot create %s\n", dsn.c_str());
+      const std::string zgroup = "setup" + std::to_string(s.id) +
+                                 "/timepoint" + std::to_string(tp);
+      bsj::ValuePtr msets;
+      if (!useN5) {
+        if (!out_zr.create_group(zgroup)) {
+          fprintf(stderr, "cannot create group %s\n", zgroup.c_str());
           return 1;
         }
-        /* downsampling-factors attribute per level (bdv.n5 layout) */
-        out_n5.set_attr(dsn, "downsamplingFactors",
-                        bsj::Value::mkints(std::vector<int>{
-                            ladder[l][0], ladder[l][1], ladder[l][2]}));
+        msets = bsj::Value::mkarr();
+      }
+      for (int l = 0; l < nlevels; ++l) {
         long long lx = ldims[l * 3], ly = ldims[l * 3 + 1],
                   lz = ldims[l * 3 + 2];
+        std::string dsn;
+        bsn5::DatasetAttrs da;
+        bszarr::ArrayAttrs za;
+        if (useN5) {
+          dsn = bssd::SpimData::image_dataset(s.id, tp, l);
+          da.dims = {lx, ly, lz};
+          da.block = {(int)bs[0], (int)bs[1], (int)bs[2]};
+          da.dtype = "uint16";
+          da.compression = codec;
+          da.level = clevel;
+          if (!out_n5.create_dataset(dsn, da)) {
+            fprintf(stderr, "cannot create %s\n", dsn.c_str());
+            return 1;
+          }
+          /* downsampling-factors attribute per level (bdv.n5 layout) */
+          out_n5.set_attr(dsn, "downsamplingFactors",
+                          bsj::Value::mkints(std::vector<int>{
+                              ladder[l][0], ladder[l][1], ladder[l][2]}));
+        } else {
+          /* [PIN-OMEZARR-BDV] 5-D [t,c,z,y,x] array "{l}" per level */
+          dsn = zgroup + "/" + std::to_string(l);
+          za.shape = {1, 1, lz, ly, lx};
+          za.chunks = {1, 1, (int)bs[2], (int)bs[1], (int)bs[0]};
+          za.dtype = "<u2";
+          za.codec = codec;
+          za.level = clevel;
+          za.gzip = codec == "gzip";
+          if (!out_zr.create_array(dsn, za)) {
+            fprintf(stderr, "cannot create %s\n", dsn.c_str());
+            return 1;
+          }
+          auto dset = bsj::Value::mkobj();
+          dset->obj["path"] = bsj::Value::mkstr(std::to_string(l));
+          auto cts = bsj::Value::mkarr();
+          auto sc = bsj::Value::mkobj();
+          sc->obj["type"] = bsj::Value::mkstr("scale");
+          auto scale = bsj::Value::mkarr();
+          scale->arr.push_back(bsj::Value::mknum(1.0));
+          scale->arr.push_back(bsj::Value::mknum(1.0));
+          scale->arr.push_back(bsj::Value::mknum((double)ladder[l][2]));
+          scale->arr.push_back(bsj::Value::mknum((double)ladder[l][1]));
+          scale->arr.push_back(bsj::Value::mknum((double)ladder[l][0]));
+          sc->obj["scale"] = scale;
+          cts->arr.push_back(sc);
+          dset->obj["coordinateTransformations"] = cts;
+          msets->arr.push_back(dset);
+        }
         /* independent chunks: compression + file writes on a pool (as
          * the reference's Spark executors write blocks concurrently) */
         const long long ngx = (lx + bs[0] - 1) / bs[0];
@@ -168,9 +276,12 @@ int main(int argc, char **argv) {
                        src + (((gz * bs[2] + z) * ly + gy * bs[1] + y) *
                                   lx + gx * bs[0]) * 2,
                        (size_t)cx * 2);
-            if (!out_n5.write_block(dsn, da, {gx, gy, gz}, blk.data(),
-                                    {cx, cy, cz}))
-              failed.store(true);
+            bool ok =
+                useN5 ? out_n5.write_block(dsn, da, {gx, gy, gz},
+                                           blk.data(), {cx, cy, cz})
+                      : out_zr.write_chunk(dsn, za, {0, 0, gz, gy, gx},
+                                           blk.data(), {1, 1, cz, cy, cx});
+            if (!ok) failed.store(true);
           }
         };
         std::vector<std::thread> ws;
@@ -181,24 +292,50 @@ int main(int argc, char **argv) {
           return 1;
         }
       }
+      if (!useN5) {
+        /* OME-NGFF v0.4 multiscales on the view group */
+        auto ms = bsj::Value::mkobj();
+        ms->obj["version"] = bsj::Value::mkstr("0.4");
+        ms->obj["name"] = bsj::Value::mkstr(zgroup);
+        auto axes = bsj::Value::mkarr();
+        const char *axn[5] = {"t", "c", "z", "y", "x"};
+        const char *axt[5] = {"time", "channel", "space", "space", "space"};
+        for (int i = 0; i < 5; ++i) {
+          auto ax = bsj::Value::mkobj();
+          ax->obj["name"] = bsj::Value::mkstr(axn[i]);
+          ax->obj["type"] = bsj::Value::mkstr(axt[i]);
+          axes->arr.push_back(ax);
+        }
+        ms->obj["axes"] = axes;
+        ms->obj["datasets"] = msets;
+        auto msl = bsj::Value::mkarr();
+        msl->arr.push_back(ms);
+        out_zr.set_group_attr(zgroup, "multiscales", msl);
+      }
       printf("resaved setup %d tp %d: %d level(s)\n", s.id, tp, nlevels);
     }
   }
   bs_ctx_destroy(ctx);
   /* rewritten XML pointing at the new container */
-  std::string xo = args.get("xmlOut");
-  if (xo.empty()) xo = args.get("xml");
   auto seq = sd.root->child("SequenceDescription");
   auto il = seq ? seq->child("ImageLoader") : nullptr;
-  auto n5node = il ? il->child("n5") : nullptr;
-  if (n5node) {
-    n5node->attrs["type"] = "absolute";
-    n5node->text = args.get("n5Path");
+  if (il) {
+    /* replace the loader with the output container's kind */
+    il->attrs["format"] = useN5 ? "bdv.n5" : "bdv.ome.zarr";
+    il->attrs["version"] = "1.0";
+    il->children.clear();
+    auto node = il->add_text(useN5 ? "n5" : "zarr", outPath);
+    node->attrs["type"] = "absolute";
+  }
+  if (xo == args.get("xml")) {
+    /* overwrite-in-place: keep a "~1" backup of the original */
+    std::string orig;
+    if (bsio::read_file(xo, &orig)) bsio::write_file(xo + "~1", orig);
   }
   if (!bsx::save_file(xo, sd.root)) {
     fprintf(stderr, "cannot write %s\n", xo.c_str());
     return 1;
   }
-  printf("resave done -> %s, %s\n", args.get("n5Path").c_str(), xo.c_str());
+  printf("resave done -> %s, %s\n", outPath.c_str(), xo.c_str());
   return 0;
 }

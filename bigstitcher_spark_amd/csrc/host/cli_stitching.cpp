@@ -18,6 +18,7 @@
 
 #include "../../../include/bigstitch.h"
 #include "bs_cli_util.h"
+#include "bs_imgio.h"
 #include "bs_n5.h"
 #include "bs_mip.h"
 #include "bs_spimdata.h"
@@ -94,7 +95,7 @@ int main(int argc, char **argv) {
   }
   std::set<std::pair<int, int>> selset(selected.begin(), selected.end());
 
-  bsn5::Container n5(sd.n5_path);
+  bsimg::Input n5(sd);
   /* ctx created lazily: --dryRun only enumerates pairs (no GPU) */
   bs_ctx *ctx = nullptr;
   auto ensure_ctx = [&]() -> bool {
@@ -117,7 +118,7 @@ int main(int argc, char **argv) {
       auto it = setup_levels.find(setup);
       if (it == setup_levels.end())
         it = setup_levels
-                 .emplace(setup, bscli::read_levels(n5, setup, tp))
+                 .emplace(setup, n5.read_levels(setup, tp))
                  .first;
       return it->second;
     };
@@ -163,9 +164,7 @@ int main(int argc, char **argv) {
       }
       std::vector<uint16_t> vox;
       std::vector<long long> dims;
-      if (!n5.read_volume_u16(
-              bssd::SpimData::image_dataset(setup, tp, level), &vox,
-              &dims)) {
+      if (!n5.read_volume_u16(setup, tp, level, &vox, &dims)) {
         fprintf(stderr, "cannot read view tp=%d setup=%d s%d from %s\n",
                 tp, setup, level, sd.n5_path.c_str());
         return false;
