@@ -1462,3 +1462,30 @@ def test_cli_resave_default_paths_and_backup(tmp_path):
     assert open(xml + "~1").read() == orig
     il = ET.parse(xml).getroot().find(".//ImageLoader")
     assert il.get("format") == "bdv.ome.zarr"
+
+
+@pytest.mark.gpu
+def test_cli_resave_omezarr_mip_selection(tmp_path):
+    """The zarr input path's level enumeration (factors derived from
+    level dims — bs_imgio.h) feeds [PIN-MIPSEL]: stitching at
+    -ds 2,2,1 on a resaved OME-ZARR picks level 1 and still lands on
+    the injected shift."""
+    err = (2.5, -1.5, 1.0)
+    xml, n5, _e, _ = make_grid_dataset(str(tmp_path), err=err)
+    out = os.path.join(str(tmp_path), "re.ome.zarr")
+    xo = os.path.join(str(tmp_path), "re.xml")
+    r = run([os.path.join(BIN, "resave"), "-x", xml, "-o", out, "-xo", xo,
+             "-ds", "1,1,1", "-ds", "2,2,1"])
+    assert r.returncode == 0, r.stderr + r.stdout
+    r = run([os.path.join(BIN, "stitching"), "-x", xo, "-ds", "2,2,1",
+             "--dryRun"])
+    assert r.returncode == 0, r.stderr
+    assert "reading pyramid level s1" in r.stdout
+    r = run([os.path.join(BIN, "stitching"), "-x", xo, "-ds", "2,2,1",
+             "--minOverlapRatio", "0.05"])
+    assert r.returncode == 0, r.stderr + r.stdout
+    prs = ET.parse(xo).getroot().findall(".//StitchingResults/PairwiseResult")
+    assert len(prs) == 1
+    m = [float(x) for x in prs[0].find("Matrix").text.split()]
+    for g, w in zip((m[3], m[7], m[11]), (-err[0], -err[1], -err[2])):
+        assert abs(g - w) < 1.0, (m, err)
