@@ -157,11 +157,6 @@ int main(int argc, char **argv) {
       (long)std::max<size_t>(1, sd.timepoints.size()));
   int numCh = (int)args.getl("numChannels", 1);
   std::string storage = args.get("storage", "ZARR"); /* reference default */
-  if (bdv && storage != "N5") {
-    fprintf(stderr, "--bdv requires -s N5 in this build (no HDF5 "
-                    "libraries in this environment)\n");
-    return 2;
-  }
   if (storage != "ZARR" && storage != "N5") {
     fprintf(stderr,
             "unsupported --storage %s (supported: ZARR, N5; HDF5 is not "
@@ -183,8 +178,12 @@ int main(int argc, char **argv) {
     else
       n5.set_attr("", "Bigstitcher-Spark/" + k, v);
   };
-  set("FusionFormat", bsj::Value::mkstr(zarr ? "OME-ZARR"
-                                        : bdv ? "BDV/N5" : "N5"));
+  /* reference CreateFusionContainer.java:391-400: BDV flavors carry a
+   * "BDV/" FusionFormat prefix and the OutputXML attribute */
+  set("FusionFormat",
+      bsj::Value::mkstr(zarr ? (bdv ? "BDV/OME-ZARR" : "OME-ZARR")
+                             : bdv ? "BDV/N5" : "N5"));
+  if (bdv) set("OutputXML", bsj::Value::mkstr(args.get("xmlout")));
   set("InputXML", bsj::Value::mkstr(args.get("xml")));
   set("NumTimepoints", bsj::Value::mkint(numTp));
   set("NumChannels", bsj::Value::mkint(numCh));
@@ -383,11 +382,20 @@ int main(int argc, char **argv) {
     }
     auto xr = bssd::make_dataset_xml(args.get("outputPath"), fsetups,
                                      ftps, fregs);
-    /* absolute container path */
+    /* absolute container path; ZARR: the bdv.ome.zarr loader
+     * ([PIN-OMEZARR-BDV]) — the fused 5-D "s{l}" arrays hold all
+     * (t,c) slices, as the reference's OMEZARREntry{c,t} indices do
+     * (CreateFusionContainer.java:437-451) */
     auto seq = xr->child("SequenceDescription");
     auto il = seq ? seq->child("ImageLoader") : nullptr;
     auto n5n = il ? il->child("n5") : nullptr;
-    if (n5n) n5n->attrs["type"] = "absolute";
+    if (n5n) {
+      n5n->attrs["type"] = "absolute";
+      if (zarr) {
+        n5n->tag = "zarr";
+        il->attrs["format"] = "bdv.ome.zarr";
+      }
+    }
     if (!bsx::save_file(args.get("xmlout"), xr)) {
       fprintf(stderr, "cannot write %s\n", args.get("xmlout").c_str());
       return 1;

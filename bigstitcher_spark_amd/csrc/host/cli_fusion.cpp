@@ -147,18 +147,19 @@ int main(int argc, char **argv) {
     auto x = geta("InputXML");
     if (x) xml = x->str;
   }
-  /* exact-format whitelist: BDV-prefixed and HDF5 containers have different
+  /* exact-format whitelist: HDF5 containers have different
    * dataset layouts (reference SparkAffineFusion.java:239-307) — refuse
    * them explicitly rather than guessing a layout */
   if (fmt->str != "N5" && fmt->str != "OME-ZARR" &&
-      fmt->str != "BDV/N5") {
+      fmt->str != "BDV/N5" && fmt->str != "BDV/OME-ZARR") {
     fprintf(stderr,
             "unsupported FusionFormat '%s' (supported: N5, OME-ZARR, "
             "BDV/N5; HDF5 containers are not supported by this build)\n",
             fmt->str.c_str());
     return 1;
   }
-  const bool zarr = fmt->str == "OME-ZARR";
+  const bool zarr =
+      fmt->str == "OME-ZARR" || fmt->str == "BDV/OME-ZARR";
   auto bbmin_a = geta("Boundingbox_min"), bbmax_a = geta("Boundingbox_max");
   auto bs_a = geta("BlockSize");
   auto dt_a = geta("DataType");

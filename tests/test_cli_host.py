@@ -1182,10 +1182,20 @@ def test_container_cli_bdv_n5(tmp_path):
     tree = ET.parse(xo)
     assert tree.getroot().find(".//ImageLoader/n5").text == out
     assert tree.getroot().find(".//ViewSetup/size").text.startswith("104")
-    # zarr + bdv rejected
+    # zarr + bdv: BDV/OME-ZARR (reference CreateFusionContainer.java:398)
+    zout = os.path.join(str(tmp_path), "fused.zarr")
+    zxo = os.path.join(str(tmp_path), "fusedz.xml")
     r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml,
-             "-o", os.path.join(str(tmp_path), "z"), "--bdv", "-xo", xo])
-    assert r.returncode != 0
+             "-o", zout, "--bdv", "-xo", zxo,
+             "--blockSize", "32,32,32", "-d", "UINT16",
+             "-ds", "1,1,1", "-ds", "2,2,2"])
+    assert r.returncode == 0, r.stderr + r.stdout
+    za = n5util.zarr_root_attrs(zout)["Bigstitcher-Spark"]
+    assert za["FusionFormat"] == "BDV/OME-ZARR"
+    assert za["OutputXML"] == zxo
+    il = ET.parse(zxo).getroot().find(".//ImageLoader")
+    assert il.get("format") == "bdv.ome.zarr"
+    assert il.find("zarr").text == zout
 
 
 @pytest.mark.gpu
@@ -1489,3 +1499,33 @@ def test_cli_resave_omezarr_mip_selection(tmp_path):
     m = [float(x) for x in prs[0].find("Matrix").text.split()]
     for g, w in zip((m[3], m[7], m[11]), (-err[0], -err[1], -err[2])):
         assert abs(g - w) < 1.0, (m, err)
+
+
+@pytest.mark.gpu
+def test_cli_fusion_bdv_omezarr_end_to_end(tmp_path):
+    """BDV/OME-ZARR container (--bdv -s ZARR): affine-fusion writes the
+    shared 5-D s{l} arrays and the result matches the oracle (the BDV
+    XML references (c,t) slices of the same arrays —
+    CreateFusionContainer.java:437-451)."""
+    xml, n5, err, (a, b) = make_grid_dataset(str(tmp_path))
+    out = os.path.join(str(tmp_path), "fused.zarr")
+    xo = os.path.join(str(tmp_path), "fused.xml")
+    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml,
+             "-o", out, "--bdv", "-xo", xo,
+             "--blockSize", "32,32,32", "-d", "FLOAT32"])
+    assert r.returncode == 0, r.stderr
+    r = run([os.path.join(BIN, "affine-fusion"), "-o", out,
+             "-f", "AVG_BLEND", "--blendingRange", "8"])
+    assert r.returncode == 0, r.stderr + r.stdout
+    fused5, za = n5util.read_zarr(out, "s0")
+    assert za["shape"][:2] == [1, 1]
+    fused = fused5[0, 0]
+    ident = np.hstack([np.eye(3), np.zeros((3, 1))])
+    affB = ident.copy()
+    affB[0, 3] = 40.0
+    views = [dict(data=a, affine=ident, border=(0, 0, 0), range=(8, 8, 8)),
+             dict(data=b, affine=affB, border=(0, 0, 0), range=(8, 8, 8))]
+    ref = of.fuse_block(views, (0, 0, 0), (104, 64, 64),
+                        of.FUSION_AVG_BLEND)
+    denom = np.maximum(np.abs(ref), 1.0)
+    assert np.max(np.abs(fused.astype(np.float64) - ref) / denom) < 1e-4
