@@ -249,8 +249,10 @@ bool Container::read_volume_u16(const std::string &name,
   if (!get_array_attrs(name, &a) || a.dtype != "<u2" || a.shape.size() < 3)
     return false;
   const size_t nd = a.shape.size();
+  /* the blk indexing below assumes singleton leading (t,c,...) chunk
+   * dims — true for the BDV 5-D layout {1,1,bz,by,bx} */
   for (size_t d = 0; d + 3 < nd; ++d)
-    if (a.shape[d] < 1 || a.chunks[d] < 1) return false;
+    if (a.shape[d] < 1 || a.chunks[d] != 1) return false;
   const long long nz = a.shape[nd - 3], ny = a.shape[nd - 2],
                   nx = a.shape[nd - 1];
   const int bz = a.chunks[nd - 3], by = a.chunks[nd - 2],
