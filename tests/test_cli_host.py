@@ -1529,3 +1529,39 @@ def test_cli_fusion_bdv_omezarr_end_to_end(tmp_path):
                         of.FUSION_AVG_BLEND)
     denom = np.maximum(np.abs(ref), 1.0)
     assert np.max(np.abs(fused.astype(np.float64) - ref) / denom) < 1e-4
+
+
+def test_view_selection_combined_filters(tmp_path):
+    """Combined attribute filters intersect (Import.getViewIds(data, a,
+    c, i, ti, tp): a view must match EVERY provided id list)."""
+    n5 = os.path.join(str(tmp_path), "input.n5")
+    xml = os.path.join(str(tmp_path), "dataset.xml")
+    rng = np.random.default_rng(5)
+    for sid in range(4):
+        vol = rng.integers(0, 60000, size=(16, 16, 16)).astype(np.uint16)
+        n5util.write_dataset(n5, f"setup{sid}/timepoint0/s0", vol,
+                             (16, 16, 16))
+    # tiles 0/1 x angles 0/1
+    n5util.make_dataset_xml(
+        xml, "input.n5",
+        [dict(id=s, dims=(16, 16, 16), pos=(8.0 * s, 0.0, 0.0),
+              attrs=dict(tile=s % 2, angle=s // 2, channel=0,
+                         illumination=0))
+         for s in range(4)])
+    out = os.path.join(str(tmp_path), "sel.n5")
+    # tile 1 AND angle 1 -> only setup 3 (pos 24, dims 16 -> max x 39)
+    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml,
+             "-s", "N5", "-o", out, "--blockSize", "16,16,16",
+             "-d", "UINT16", "--tileId", "1", "--angleId", "1"])
+    assert r.returncode == 0, r.stderr
+    attrs = n5util.root_attrs(out)["Bigstitcher-Spark"]
+    assert attrs["Boundingbox_min"][0] == 24
+    assert attrs["Boundingbox_max"][0] == 39
+
+
+def test_resave_rejects_bad_codec(tmp_path):
+    """resave -c takes Gzip|Zstandard|Raw; anything else exits with an
+    explicit error before any work."""
+    xml, n5, _err, _ = make_grid_dataset(str(tmp_path))
+    r = run([os.path.join(BIN, "resave"), "-x", xml, "-c", "Lz4"])
+    assert r.returncode == 2 and "Lz4" in r.stderr
