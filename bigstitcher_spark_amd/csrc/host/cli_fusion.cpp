@@ -14,6 +14,7 @@
  * a sliding view window for datasets beyond HBM. */
 #include <algorithm>
 #include <atomic>
+#include <chrono>
 #include <cstdio>
 #include <cstring>
 #include <set>
@@ -450,8 +451,14 @@ int main(int argc, char **argv) {
       bandz = ((v + band_step - 1) / band_step) * band_step;
     }
     std::set<int> resident;
+    const bool timing = getenv("BS_TIMING") != nullptr;
+    auto now = [] { return std::chrono::steady_clock::now(); };
+    auto secs = [](auto a, auto b) {
+      return std::chrono::duration<double>(b - a).count();
+    };
     for (long bz0 = 0; bz0 < (long)dims[2]; bz0 += bandz) {
       const long bz1 = std::min((long)dims[2], bz0 + bandz);
+      auto tb0 = now();
       /* slide the view window */
       for (auto it = resident.begin(); it != resident.end();) {
         bool still = false;
@@ -504,6 +511,7 @@ int main(int argc, char **argv) {
         }
         fviews.push_back(pl.fv);
       }
+      auto tb1 = now();
       /* fuse this band (bs_fuse_volume may additionally z-slab
        * internally when even the band exceeds free HBM) */
       std::vector<std::vector<char>> hostlvl(nlevels);
@@ -531,6 +539,7 @@ int main(int argc, char **argv) {
         fprintf(stderr, "fusion failed: %s\n", bs_last_error(ctx));
         return 1;
       }
+      auto tb2 = now();
       /* write the band's chunks (N5 3-D datasets or OME-ZARR 5-D
        * arrays — the 3-D-block-into-5-D lift,
        * SparkAffineFusion.java:630-643); band starts are chunk-aligned
@@ -593,6 +602,10 @@ int main(int argc, char **argv) {
         printf("level %d (%s): wrote %lld blocks (band z %ld..%ld)\n",
                l, dsnames[l].c_str(), nchunks, bz0, bz1);
       }
+      if (timing)
+        printf("band z %ld..%ld: read+upload %.2fs, alloc+fuse %.2fs, "
+               "chunk-write %.2fs\n",
+               bz0, bz1, secs(tb0, tb1), secs(tb1, tb2), secs(tb2, now()));
     }
     for (int id : resident) bs_view_release(ctx, id);
    }
