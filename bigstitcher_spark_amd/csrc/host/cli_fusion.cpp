@@ -17,6 +17,8 @@
 #include <chrono>
 #include <cstdio>
 #include <cstring>
+#include <future>
+#include <memory>
 #include <set>
 #include <thread>
 #include <vector>
@@ -472,19 +474,47 @@ int main(int argc, char **argv) {
         }
       }
       std::vector<bs_fuse_view> fviews;
-      for (auto &pl : plans) {
-        if (pl.zhi < bz0 - 2 || pl.zlo > bz1 + 2) continue;
-        if (!resident.count(pl.setup)) {
-          std::vector<uint16_t> vox;
-          std::vector<long long> vdims;
-          if (!in_n5.read_volume_u16(pl.setup, pl.tp, pl.level,
-                                     &vox, &vdims)) {
-            fprintf(stderr, "cannot read view tp=%d setup=%d\n", pl.tp,
-                    pl.setup);
-            return 1;
+      /* pipelined loading: one reader ahead of the upload (the chunk
+       * decode of view N+1 overlaps view N's PCIe H2D) */
+      struct RV {
+        std::vector<uint16_t> vox;
+        std::vector<long long> vdims;
+        bool ok;
+      };
+      std::vector<const VPlan *> need;
+      {
+        std::set<int> queued;
+        for (auto &pl : plans) {
+          if (pl.zhi < bz0 - 2 || pl.zlo > bz1 + 2) continue;
+          fviews.push_back(pl.fv);
+          if (!resident.count(pl.setup) && !queued.count(pl.setup)) {
+            queued.insert(pl.setup);
+            need.push_back(&pl);
           }
-          int64_t d[3] = {vdims[0], vdims[1], vdims[2]};
-          if (bs_view_upload(ctx, pl.setup, vox.data(), d) != BS_OK) {
+        }
+      }
+      auto read_one = [&](const VPlan *pl) {
+        return std::async(std::launch::async, [pl, &in_n5] {
+          RV r;
+          r.ok = in_n5.read_volume_u16(pl->setup, pl->tp, pl->level,
+                                       &r.vox, &r.vdims);
+          return r;
+        });
+      };
+      std::future<RV> fut;
+      if (!need.empty()) fut = read_one(need[0]);
+      for (size_t ni = 0; ni < need.size(); ++ni) {
+        const VPlan &pl = *need[ni];
+        RV r = fut.get();
+        if (ni + 1 < need.size()) fut = read_one(need[ni + 1]);
+        if (!r.ok) {
+          fprintf(stderr, "cannot read view tp=%d setup=%d\n", pl.tp,
+                  pl.setup);
+          return 1;
+        }
+        {
+          int64_t d[3] = {r.vdims[0], r.vdims[1], r.vdims[2]};
+          if (bs_view_upload(ctx, pl.setup, r.vox.data(), d) != BS_OK) {
             fprintf(stderr, "upload failed: %s\n", bs_last_error(ctx));
             return 1;
           }
@@ -509,12 +539,15 @@ int main(int argc, char **argv) {
           }
           resident.insert(pl.setup);
         }
-        fviews.push_back(pl.fv);
       }
       auto tb1 = now();
       /* fuse this band (bs_fuse_volume may additionally z-slab
        * internally when even the band exceeds free HBM) */
-      std::vector<std::vector<char>> hostlvl(nlevels);
+      /* uninitialized buffers: every byte is either written by the
+       * staged D2H in bs_fuse_volume or memset in the empty-band
+       * branch — a value-initializing resize would memset ~GBs */
+      std::vector<std::unique_ptr<char[]>> hostlvl(nlevels);
+      std::vector<size_t> hostlvl_bytes(nlevels);
       std::vector<void *> lvlptr(nlevels);
       std::vector<int64_t> ldims(3 * nlevels);
       int64_t vmin[3] = {bbmin[0], bbmin[1], bbmin[2] + bz0};
@@ -523,12 +556,13 @@ int main(int argc, char **argv) {
         long long b = esz2;
         for (int d = 0; d < 3; ++d)
           b *= (vdim[d] + abs_ds[l * 3 + d] - 1) / abs_ds[l * 3 + d];
-        hostlvl[l].resize((size_t)b);
-        lvlptr[l] = hostlvl[l].data();
+        hostlvl[l].reset(new char[(size_t)b]);
+        hostlvl_bytes[l] = (size_t)b;
+        lvlptr[l] = hostlvl[l].get();
       }
       if (fviews.empty()) {
         for (int l = 0; l < nlevels; ++l)
-          std::fill(hostlvl[l].begin(), hostlvl[l].end(), 0);
+          memset(hostlvl[l].get(), 0, hostlvl_bytes[l]);
         for (int l = 0; l < nlevels; ++l)
           for (int d = 0; d < 3; ++d)
             ldims[l * 3 + d] =
@@ -573,7 +607,7 @@ int main(int argc, char **argv) {
             int cx = (int)std::min((long long)bx, lx - gx * bx);
             int cy = (int)std::min((long long)by, ly - gy * by);
             int cz = (int)std::min((long long)bz, lz - gz * bz);
-            const char *src = hostlvl[l].data();
+            const char *src = hostlvl[l].get();
             for (int z = 0; z < cz; ++z)
               for (int y = 0; y < cy; ++y)
                 memcpy(&blk[((size_t)z * cy + y) * cx * esz2],
