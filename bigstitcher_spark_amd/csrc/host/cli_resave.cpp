@@ -20,6 +20,8 @@
  * an exact uint16 round-trip at minI=0/maxI=65535). */
 #include <atomic>
 #include <cstdio>
+#include <future>
+#include <memory>
 #include <cstring>
 #include <thread>
 #include <vector>
@@ -147,14 +149,38 @@ int main(int argc, char **argv) {
   for (int l = 0; l < nlevels; ++l)
     for (int d = 0; d < 3; ++d) abs_ds[l * 3 + d] = ladder[l][d];
 
-  for (int tp : sd.timepoints) {
-    for (auto &s : sd.setups) {
-      std::vector<uint16_t> vox;
-      std::vector<long long> dims;
-      if (!in.read_volume_u16(s.id, tp, 0, &vox, &dims)) {
+  /* flatten the (tp, setup) work list so the NEXT view's chunk decode
+   * overlaps the current view's upload + pyramid + chunk writes */
+  std::vector<std::pair<int, const bssd::ViewSetup *>> work;
+  for (int tp : sd.timepoints)
+    for (auto &s : sd.setups) work.push_back({tp, &s});
+  struct RV {
+    std::vector<uint16_t> vox;
+    std::vector<long long> dims;
+    bool ok;
+  };
+  auto read_one = [&](size_t wi) {
+    return std::async(std::launch::async, [wi, &work, &in] {
+      RV r;
+      r.ok = in.read_volume_u16(work[wi].second->id, work[wi].first, 0,
+                                &r.vox, &r.dims);
+      return r;
+    });
+  };
+  std::future<RV> fut;
+  if (!work.empty()) fut = read_one(0);
+  for (size_t wi = 0; wi < work.size(); ++wi) {
+    {
+      const int tp = work[wi].first;
+      const auto &s = *work[wi].second;
+      RV rv = fut.get();
+      if (wi + 1 < work.size()) fut = read_one(wi + 1);
+      if (!rv.ok) {
         fprintf(stderr, "cannot read view tp=%d setup=%d\n", tp, s.id);
         return 1;
       }
+      std::vector<uint16_t> &vox = rv.vox;
+      std::vector<long long> &dims = rv.dims;
       int64_t d3[3] = {dims[0], dims[1], dims[2]};
       if (bs_view_upload(ctx, s.id, vox.data(), d3) != BS_OK) {
         fprintf(stderr, "upload failed: %s\n", bs_last_error(ctx));
@@ -170,15 +196,16 @@ int main(int argc, char **argv) {
       prm.min_intensity = 0;
       prm.max_intensity = 65535;
       prm.interp = 1;
-      std::vector<std::vector<char>> hostlvl(nlevels);
+      /* uninitialized: bs_fuse_volume's staged D2H writes every byte */
+      std::vector<std::unique_ptr<char[]>> hostlvl(nlevels);
       std::vector<void *> lvlptr(nlevels);
       std::vector<int64_t> ldims(3 * nlevels);
       for (int l = 0; l < nlevels; ++l) {
         long long b = 2;
         for (int d = 0; d < 3; ++d)
           b *= (dims[d] + ladder[l][d] - 1) / ladder[l][d];
-        hostlvl[l].resize((size_t)b);
-        lvlptr[l] = hostlvl[l].data();
+        hostlvl[l].reset(new char[(size_t)b]);
+        lvlptr[l] = hostlvl[l].get();
       }
       int64_t vmin[3] = {0, 0, 0};
       if (bs_fuse_volume(ctx, &fv, 1, vmin, d3, &prm, nlevels,
@@ -269,7 +296,7 @@ int main(int argc, char **argv) {
             int cx = (int)std::min(bs[0], lx - gx * bs[0]);
             int cy = (int)std::min(bs[1], ly - gy * bs[1]);
             int cz = (int)std::min(bs[2], lz - gz * bs[2]);
-            const char *src = hostlvl[l].data();
+            const char *src = hostlvl[l].get();
             for (int z = 0; z < cz; ++z)
               for (int y = 0; y < cy; ++y)
                 memcpy(&blk[((size_t)z * cy + y) * cx * 2],
