@@ -14,6 +14,7 @@
  * tile sets larger than HBM. */
 #include <cinttypes>
 #include <cstdio>
+#include <future>
 #include <set>
 
 #include "../../../include/bigstitch.h"
@@ -446,6 +447,54 @@ int main(int argc, char **argv) {
         ++k1;
       }
       const int tmp_first = next_tmp;
+      { /* pre-read this window's views with a one-ahead pipeline: the
+           chunk decode of view i+1 overlaps view i's H2D upload (same
+           pattern as the fusion CLI band loop) */
+        std::vector<std::pair<int, int>> toload; /* (setup, level) */
+        std::set<int> seen;
+        for (auto &kv : uploaded) seen.insert(kv.first);
+        for (size_t k = k0; k < k1; ++k) {
+          auto &pp = plans[k];
+          const int la = pp.nonequal ? 0 : pp.lva;
+          const int lb = pp.nonequal ? 0 : pp.lvb;
+          for (auto *m : pp.ga->members)
+            if (seen.insert(m->id).second) toload.push_back({m->id, la});
+          for (auto *m : pp.gb->members)
+            if (seen.insert(m->id).second) toload.push_back({m->id, lb});
+        }
+        struct RV {
+          std::vector<uint16_t> vox;
+          std::vector<long long> dims;
+          bool ok;
+        };
+        auto read_one = [&](size_t i) {
+          return std::async(std::launch::async, [i, &toload, &n5, tp] {
+            RV r;
+            r.ok = n5.read_volume_u16(toload[i].first, tp,
+                                      toload[i].second, &r.vox, &r.dims);
+            return r;
+          });
+        };
+        std::future<RV> fu;
+        if (!toload.empty()) fu = read_one(0);
+        for (size_t i = 0; i < toload.size(); ++i) {
+          RV r = fu.get();
+          if (i + 1 < toload.size()) fu = read_one(i + 1);
+          if (!r.ok) {
+            fprintf(stderr, "cannot read view tp=%d setup=%d s%d\n", tp,
+                    toload[i].first, toload[i].second);
+            return 1;
+          }
+          int64_t d[3] = {r.dims[0], r.dims[1], r.dims[2]};
+          if (bs_view_upload(ctx, toload[i].first, r.vox.data(), d) !=
+              BS_OK) {
+            fprintf(stderr, "upload failed: %s\n", bs_last_error(ctx));
+            return 1;
+          }
+          uploaded[toload[i].first] = toload[i].second;
+          updims[toload[i].first] = r.dims;
+        }
+      }
       for (size_t k = k0; k < k1; ++k) {
         auto &pp = plans[k];
       if (pp.nonequal) {
