@@ -17,6 +17,7 @@
 #include <chrono>
 #include <cstdio>
 #include <cstring>
+#include <deque>
 #include <future>
 #include <memory>
 #include <set>
@@ -501,12 +502,17 @@ int main(int argc, char **argv) {
           return r;
         });
       };
-      std::future<RV> fut;
-      if (!need.empty()) fut = read_one(need[0]);
+      std::deque<std::future<RV>> futs; /* depth 2: two view decodes
+        in flight (each uses its own chunk pool) while one uploads */
+      const size_t DEPTH = 2;
+      for (size_t ni = 0; ni < need.size() && ni < DEPTH; ++ni)
+        futs.push_back(read_one(need[ni]));
       for (size_t ni = 0; ni < need.size(); ++ni) {
         const VPlan &pl = *need[ni];
-        RV r = fut.get();
-        if (ni + 1 < need.size()) fut = read_one(need[ni + 1]);
+        RV r = futs.front().get();
+        futs.pop_front();
+        if (ni + DEPTH < need.size())
+          futs.push_back(read_one(need[ni + DEPTH]));
         if (!r.ok) {
           fprintf(stderr, "cannot read view tp=%d setup=%d\n", pl.tp,
                   pl.setup);

@@ -20,6 +20,7 @@
  * an exact uint16 round-trip at minI=0/maxI=65535). */
 #include <atomic>
 #include <cstdio>
+#include <deque>
 #include <future>
 #include <memory>
 #include <cstring>
@@ -167,14 +168,17 @@ int main(int argc, char **argv) {
       return r;
     });
   };
-  std::future<RV> fut;
-  if (!work.empty()) fut = read_one(0);
+  std::deque<std::future<RV>> futs;
+  const size_t DEPTH = 2;
+  for (size_t wi = 0; wi < work.size() && wi < DEPTH; ++wi)
+    futs.push_back(read_one(wi));
   for (size_t wi = 0; wi < work.size(); ++wi) {
     {
       const int tp = work[wi].first;
       const auto &s = *work[wi].second;
-      RV rv = fut.get();
-      if (wi + 1 < work.size()) fut = read_one(wi + 1);
+      RV rv = futs.front().get();
+      futs.pop_front();
+      if (wi + DEPTH < work.size()) futs.push_back(read_one(wi + DEPTH));
       if (!rv.ok) {
         fprintf(stderr, "cannot read view tp=%d setup=%d\n", tp, s.id);
         return 1;

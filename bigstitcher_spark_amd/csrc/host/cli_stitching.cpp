@@ -14,6 +14,7 @@
  * tile sets larger than HBM. */
 #include <cinttypes>
 #include <cstdio>
+#include <deque>
 #include <future>
 #include <set>
 
@@ -475,11 +476,15 @@ int main(int argc, char **argv) {
             return r;
           });
         };
-        std::future<RV> fu;
-        if (!toload.empty()) fu = read_one(0);
+        std::deque<std::future<RV>> fus;
+        const size_t DEPTH = 2;
+        for (size_t i = 0; i < toload.size() && i < DEPTH; ++i)
+          fus.push_back(read_one(i));
         for (size_t i = 0; i < toload.size(); ++i) {
-          RV r = fu.get();
-          if (i + 1 < toload.size()) fu = read_one(i + 1);
+          RV r = fus.front().get();
+          fus.pop_front();
+          if (i + DEPTH < toload.size())
+            fus.push_back(read_one(i + DEPTH));
           if (!r.ok) {
             fprintf(stderr, "cannot read view tp=%d setup=%d s%d\n", tp,
                     toload[i].first, toload[i].second);
