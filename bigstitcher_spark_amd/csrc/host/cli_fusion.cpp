@@ -447,12 +447,19 @@ int main(int argc, char **argv) {
                  budget / 1e9);
       }
     }
+    /* even when everything fits, split tall outputs into ~3 bands so
+       each band's chunk encode/write (async below) overlaps the NEXT
+       band's read+fuse */
+    if (bandz >= (long)dims[2] && (long)dims[2] >= 3 * band_step)
+      bandz = std::max(band_step, ((long)dims[2] / 3 + band_step - 1) /
+                                      band_step * band_step);
     if (const char *e = getenv("BS_CLI_BAND_Z")) {
       /* test hook: force a band depth (rounded up to the chunk/level
        * alignment step) */
       long v = std::max(1L, (long)atol(e));
       bandz = ((v + band_step - 1) / band_step) * band_step;
     }
+    std::future<bool> wfut; /* one band's writes in flight */
     std::set<int> resident;
     const bool timing = getenv("BS_TIMING") != nullptr;
     auto now = [] { return std::chrono::steady_clock::now(); };
@@ -580,19 +587,26 @@ int main(int argc, char **argv) {
         return 1;
       }
       auto tb2 = now();
+      /* wait out the PREVIOUS band's writes (bounded pipeline depth 1),
+       * then hand this band's buffers to an async writer: the chunk
+       * encode/write overlaps the next band's read+fuse */
+      if (wfut.valid() && !wfut.get()) {
+        fprintf(stderr, "block write failed\n");
+        return 1;
+      }
+      auto hl = std::make_shared<std::vector<std::unique_ptr<char[]>>>(
+          std::move(hostlvl));
+      std::vector<int64_t> ldims_c(ldims.begin(), ldims.end());
       /* write the band's chunks (N5 3-D datasets or OME-ZARR 5-D
        * arrays — the 3-D-block-into-5-D lift,
        * SparkAffineFusion.java:630-643); band starts are chunk-aligned
        * per level, so global chunk z = band offset + local */
+      auto write_band = [=, &n5, &zr, &das, &zas, &dsnames,
+                         &abs_ds]() -> bool {
       for (int l = 0; l < nlevels; ++l) {
         const int bx = cbx, by = cby, bz = cbz;
-        long long lx = ldims[l * 3], ly = ldims[l * 3 + 1],
-                  lz = ldims[l * 3 + 2];
-        if (fviews.empty()) {
-          lx = (vdim[0] + abs_ds[l * 3] - 1) / abs_ds[l * 3];
-          ly = (vdim[1] + abs_ds[l * 3 + 1] - 1) / abs_ds[l * 3 + 1];
-          lz = (vdim[2] + abs_ds[l * 3 + 2] - 1) / abs_ds[l * 3 + 2];
-        }
+        long long lx = ldims_c[l * 3], ly = ldims_c[l * 3 + 1],
+                  lz = ldims_c[l * 3 + 2];
         const long long gz_off = (bz0 / abs_ds[l * 3 + 2]) / bz;
         const long long ngx = (lx + bx - 1) / bx,
                         ngy = (ly + by - 1) / by,
@@ -613,7 +627,7 @@ int main(int argc, char **argv) {
             int cx = (int)std::min((long long)bx, lx - gx * bx);
             int cy = (int)std::min((long long)by, ly - gy * by);
             int cz = (int)std::min((long long)bz, lz - gz * bz);
-            const char *src = hostlvl[l].get();
+            const char *src = (*hl)[l].get();
             for (int z = 0; z < cz; ++z)
               for (int y = 0; y < cy; ++y)
                 memcpy(&blk[((size_t)z * cy + y) * cx * esz2],
@@ -635,17 +649,21 @@ int main(int argc, char **argv) {
         std::vector<std::thread> ws;
         for (int w = 0; w < NW; ++w) ws.emplace_back(worker);
         for (auto &w : ws) w.join();
-        if (failed.load()) {
-          fprintf(stderr, "block write failed\n");
-          return 1;
-        }
+        if (failed.load()) return false;
         printf("level %d (%s): wrote %lld blocks (band z %ld..%ld)\n",
                l, dsnames[l].c_str(), nchunks, bz0, bz1);
       }
+      return true;
+      };
+      wfut = std::async(std::launch::async, write_band);
       if (timing)
-        printf("band z %ld..%ld: read+upload %.2fs, alloc+fuse %.2fs, "
-               "chunk-write %.2fs\n",
-               bz0, bz1, secs(tb0, tb1), secs(tb1, tb2), secs(tb2, now()));
+        printf("band z %ld..%ld: read+upload %.2fs, alloc+fuse %.2fs "
+               "(chunk writes async)\n",
+               bz0, bz1, secs(tb0, tb1), secs(tb1, tb2));
+    }
+    if (wfut.valid() && !wfut.get()) {
+      fprintf(stderr, "block write failed\n");
+      return 1;
     }
     for (int id : resident) bs_view_release(ctx, id);
    }
