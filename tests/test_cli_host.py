@@ -1565,3 +1565,36 @@ def test_resave_rejects_bad_codec(tmp_path):
     xml, n5, _err, _ = make_grid_dataset(str(tmp_path))
     r = run([os.path.join(BIN, "resave"), "-x", xml, "-c", "Lz4"])
     assert r.returncode == 2 and "Lz4" in r.stderr
+
+
+@pytest.mark.gpu
+def test_cli_fusion_zband_gap_zeros(tmp_path):
+    """Async band writes with an EMPTY middle band: two tiles separated
+    by a z gap; the uncovered bands are written as zeros and the
+    covered ones match the oracle."""
+    n5 = os.path.join(str(tmp_path), "input.n5")
+    xml = os.path.join(str(tmp_path), "dataset.xml")
+    rng = np.random.default_rng(11)
+    a = rng.integers(0, 60000, size=(32, 32, 32)).astype(np.uint16)
+    b = rng.integers(0, 60000, size=(32, 32, 32)).astype(np.uint16)
+    n5util.write_dataset(n5, "setup0/timepoint0/s0", a, (16, 16, 16))
+    n5util.write_dataset(n5, "setup1/timepoint0/s0", b, (16, 16, 16))
+    n5util.make_dataset_xml(
+        xml, "input.n5",
+        [dict(id=0, dims=(32, 32, 32), pos=(0.0, 0.0, 0.0)),
+         dict(id=1, dims=(32, 32, 32), pos=(0.0, 0.0, 128.0))])
+    out = os.path.join(str(tmp_path), "fused.n5")
+    r = run([os.path.join(BIN, "create-fusion-container"), "-x", xml,
+             "-s", "N5", "-o", out, "--blockSize", "16,16,16",
+             "-d", "FLOAT32"])
+    assert r.returncode == 0, r.stderr
+    env = dict(os.environ, BS_CLI_BAND_Z="16")
+    r = subprocess.run([os.path.join(BIN, "affine-fusion"), "-o", out,
+                        "-f", "AVG"], capture_output=True, text=True,
+                       env=env)
+    assert r.returncode == 0, r.stderr + r.stdout
+    fused, _ = n5util.read_dataset(out, "ch0tp0/s0")
+    assert fused.shape == (160, 32, 32)
+    assert np.array_equal(fused[0:32], a.astype(np.float32))
+    assert np.all(fused[32:128] == 0.0)  # the gap bands
+    assert np.array_equal(fused[128:160], b.astype(np.float32))
