@@ -509,9 +509,9 @@ int main(int argc, char **argv) {
           return r;
         });
       };
-      std::deque<std::future<RV>> futs; /* depth 2: two view decodes
+      std::deque<std::future<RV>> futs; /* depth 3: view decodes
         in flight (each uses its own chunk pool) while one uploads */
-      const size_t DEPTH = 2;
+      const size_t DEPTH = 3;
       for (size_t ni = 0; ni < need.size() && ni < DEPTH; ++ni)
         futs.push_back(read_one(need[ni]));
       for (size_t ni = 0; ni < need.size(); ++ni) {

@@ -169,7 +169,7 @@ int main(int argc, char **argv) {
     });
   };
   std::deque<std::future<RV>> futs;
-  const size_t DEPTH = 2;
+  const size_t DEPTH = 3;
   for (size_t wi = 0; wi < work.size() && wi < DEPTH; ++wi)
     futs.push_back(read_one(wi));
   for (size_t wi = 0; wi < work.size(); ++wi) {

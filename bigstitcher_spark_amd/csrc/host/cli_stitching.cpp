@@ -477,7 +477,7 @@ int main(int argc, char **argv) {
           });
         };
         std::deque<std::future<RV>> fus;
-        const size_t DEPTH = 2;
+        const size_t DEPTH = 3;
         for (size_t i = 0; i < toload.size() && i < DEPTH; ++i)
           fus.push_back(read_one(i));
         for (size_t i = 0; i < toload.size(); ++i) {
