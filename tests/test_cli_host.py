@@ -1598,3 +1598,15 @@ def test_cli_fusion_zband_gap_zeros(tmp_path):
     assert np.array_equal(fused[0:32], a.astype(np.float32))
     assert np.all(fused[32:128] == 0.0)  # the gap bands
     assert np.array_equal(fused[128:160], b.astype(np.float32))
+
+
+def test_stitching_mixed_pyramid_fallback(tmp_path):
+    """A pair where only ONE view carries a pyramid: the common-level
+    rule falls back to the largest level present in BOTH ladders (s0
+    here), so mismatched ladders never read incompatible grids."""
+    xml, n5, _err, (a, b) = make_grid_dataset(str(tmp_path))
+    _write_pyramid_view(n5, 0, a)  # setup 0 has s0+s1; setup 1 only s0
+    r = run([os.path.join(BIN, "stitching"), "-x", xml, "-ds", "2,2,2",
+             "--dryRun"])
+    assert r.returncode == 0, r.stderr
+    assert "reading pyramid level" not in r.stdout  # fell back to s0
