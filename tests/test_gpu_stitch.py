@@ -324,3 +324,19 @@ def test_stitch_constant_tiles_invalid(ctx):
     ref = phasecorr.phase_correlation_shift(a, b, ds=(1, 1, 1))
     got = ctx.stitch_batch([full_pair(ctx, a, b, 50, 51)], ds=(1, 1, 1))[0]
     assert got["valid"] == ref["valid"] == False  # noqa: E712
+
+
+@pytest.mark.gpu
+def test_stitch_identical_tiles_zero_shift(ctx):
+    """Identical tiles: the PCM peak sits at the origin, shift (0,0,0)
+    and r = 1 (kept by the default maxR=1.0 filter semantics — the
+    reference omits only r > maxR)."""
+    a, _ = synth.make_pair((64, 64, 64), (5.0, 0.0, 0.0), seed=21)
+    ctx.upload(0, a)
+    ctx.upload(1, a)
+    pair = dict(view_a=0, view_b=1, off_a=(0, 0, 0), size_a=(64, 64, 64),
+                off_b=(0, 0, 0), size_b=(64, 64, 64))
+    got = ctx.stitch_batch([pair], ds=(1, 1, 1))[0]
+    assert got["valid"]
+    assert np.all(np.abs(got["shift"]) < 1e-3), got["shift"]
+    assert got["r"] == pytest.approx(1.0, abs=1e-12)
